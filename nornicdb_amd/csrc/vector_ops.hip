@@ -1,0 +1,392 @@
+// Tier-1 vector kernels for NornicDB-AMD: L2 normalize, synthetic corpus
+// generation, fused cosine-score + top-k (GEMV path for small query batches),
+// and the cross-block top-k merge.
+//
+// Re-designs (not ports of) the reference's CUDA kernels
+// (reference: pkg/gpu/cuda/cuda_kernels.cu:185-460 — 1-thread-per-vector
+// norms and a <<<1,1>>> top-k): here every kernel is wave64-shaped,
+// vector-loaded (short8 = 8 x bf16 per lane) and fused so the embedding
+// matrix is read exactly once per query batch.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// L2 normalize rows, in-place. bf16 [N][D], D % 8 == 0.
+// One wave per row; lane l covers elements [l*8, l*8+8) striding WAVE*8.
+// ---------------------------------------------------------------------------
+__global__ void k_l2_normalize_bf16(unsigned short* __restrict__ x,
+                                    long long n, int d, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const long long wave_global = (long long)blockIdx.x * waves_per_block + wid;
+  const long long total_waves = (long long)gridDim.x * waves_per_block;
+
+  for (long long row = wave_global; row < n; row += total_waves) {
+    unsigned short* rp = x + row * d;
+    float ss = 0.0f;
+    for (int c = lane * 8; c < d; c += WAVE * 8) {
+      short8v v = *reinterpret_cast<const short8v*>(rp + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_bits_to_f32((unsigned short)v[j]);
+        ss += f * f;
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    float inv = rsqrtf(ss + eps);
+    for (int c = lane * 8; c < d; c += WAVE * 8) {
+      short8v v = *reinterpret_cast<const short8v*>(rp + c);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_bits_to_f32((unsigned short)v[j]) * inv;
+        o[j] = (short)f32_to_bf16_bits(f);
+      }
+      *reinterpret_cast<short8v*>(rp + c) = o;
+    }
+  }
+}
+
+__global__ void k_l2_normalize_f32(float* __restrict__ x, long long n, int d,
+                                   float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const long long wave_global = (long long)blockIdx.x * waves_per_block + wid;
+  const long long total_waves = (long long)gridDim.x * waves_per_block;
+
+  for (long long row = wave_global; row < n; row += total_waves) {
+    float* rp = x + row * d;
+    float ss = 0.0f;
+    for (int c = lane * 4; c < d; c += WAVE * 4) {
+      float4v v = *reinterpret_cast<const float4v*>(rp + c);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) ss += v[j] * v[j];
+    }
+    ss = wave_reduce_sum(ss);
+    float inv = rsqrtf(ss + eps);
+    for (int c = lane * 4; c < d; c += WAVE * 4) {
+      float4v v = *reinterpret_cast<const float4v*>(rp + c);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] *= inv;
+      *reinterpret_cast<float4v*>(rp + c) = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fill rows with deterministic pseudo-gaussian values, L2-normalized.
+// Used to generate benchmark corpora at HBM speed (no curand round trip).
+// Row identity is (row_base + row) so multi-GPU shards are globally
+// consistent and reproducible.
+// ---------------------------------------------------------------------------
+__global__ void k_fill_random_unit_bf16(unsigned short* __restrict__ x,
+                                        long long n, int d,
+                                        long long row_base,
+                                        unsigned long long seed) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const long long wave_global = (long long)blockIdx.x * waves_per_block + wid;
+  const long long total_waves = (long long)gridDim.x * waves_per_block;
+
+  for (long long row = wave_global; row < n; row += total_waves) {
+    const unsigned long long rk =
+        seed ^ (0x100000001b3ULL * (unsigned long long)(row_base + row));
+    float ss = 0.0f;
+    // pass 1: sum of squares
+    for (int c = lane * 2; c < d; c += WAVE * 2) {
+      float2v g = hash_gauss2(rk + (unsigned long long)c);
+      ss += g.x * g.x + g.y * g.y;
+    }
+    ss = wave_reduce_sum(ss);
+    float inv = rsqrtf(ss + 1e-12f);
+    // pass 2: regenerate, scale, store
+    unsigned short* rp = x + row * d;
+    for (int c = lane * 2; c < d; c += WAVE * 2) {
+      float2v g = hash_gauss2(rk + (unsigned long long)c);
+      unsigned int lo = f32_to_bf16_bits(g.x * inv);
+      unsigned int hi = f32_to_bf16_bits(g.y * inv);
+      *reinterpret_cast<unsigned int*>(rp + c) = lo | (hi << 16);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused cosine-score + per-block top-k, GEMV path (Q <= 16).
+//
+// Each wave owns one DB row at a time (grid-stride). Lanes split the D
+// dimension (short8 loads); per-lane partial dot for every query is kept in
+// registers, reduced with xor-shuffles; lane q then folds score(row, q) into
+// a private register top-K list (descending, statically unrolled insert so it
+// stays in VGPRs — see guide rule #20). At the end, each (wave, q) writes its
+// K candidates to cand[(wave_global)][q][K].
+//
+// Queries are staged in LDS as bf16.
+// ---------------------------------------------------------------------------
+template <int QMAX, int K>
+__global__ void k_knn_gemv_bf16(const unsigned short* __restrict__ db,
+                                const unsigned short* __restrict__ qs,
+                                long long n, int d, int q_count,
+                                long long row_base,
+                                float* __restrict__ cand_score,
+                                long long* __restrict__ cand_idx) {
+  extern __shared__ unsigned short s_q[];  // [q_count][d]
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+
+  // stage queries
+  for (int i = threadIdx.x * 8; i < q_count * d; i += blockDim.x * 8) {
+    *reinterpret_cast<short8v*>(s_q + i) =
+        *reinterpret_cast<const short8v*>(qs + i);
+  }
+  __syncthreads();
+
+  const long long wave_global = (long long)blockIdx.x * waves_per_block + wid;
+  const long long total_waves = (long long)gridDim.x * waves_per_block;
+
+  // private top-K (valid in lane q for q < q_count)
+  float tv[K];
+  long long ti[K];
+#pragma unroll
+  for (int i = 0; i < K; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  for (long long row = wave_global; row < n; row += total_waves) {
+    const unsigned short* rp = db + row * d;
+    float acc[QMAX];
+#pragma unroll
+    for (int q = 0; q < QMAX; ++q) acc[q] = 0.0f;
+
+    for (int c = lane * 8; c < d; c += WAVE * 8) {
+      short8v v = *reinterpret_cast<const short8v*>(rp + c);
+      float xv[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xv[j] = bf16_bits_to_f32((unsigned short)v[j]);
+#pragma unroll
+      for (int q = 0; q < QMAX; ++q) {
+        if (q >= q_count) break;
+        short8v qv = *reinterpret_cast<const short8v*>(s_q + q * d + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[q] += xv[j] * bf16_bits_to_f32((unsigned short)qv[j]);
+      }
+    }
+    // reduce each query's partials across the wave; result in all lanes.
+#pragma unroll
+    for (int q = 0; q < QMAX; ++q) {
+      if (q >= q_count) break;
+      acc[q] = wave_reduce_sum(acc[q]);
+    }
+    // lane q keeps the score for query q.
+    if (lane < q_count) {
+      float s = acc[0];
+#pragma unroll
+      for (int q = 1; q < QMAX; ++q)
+        if (lane == q) s = acc[q];
+      if (s > tv[K - 1]) {
+        float cs = s; long long ci = row_base + row;
+#pragma unroll
+        for (int i = 0; i < K; ++i) {
+          bool ins = cs > tv[i];
+          float ts = tv[i]; long long tj = ti[i];
+          tv[i] = ins ? cs : tv[i];
+          ti[i] = ins ? ci : ti[i];
+          cs = ins ? ts : cs; ci = ins ? tj : ci;
+        }
+      }
+    }
+  }
+
+  if (lane < q_count) {
+    long long base = (wave_global * q_count + lane) * K;
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      cand_score[base + i] = tv[i];
+      cand_idx[base + i] = ti[i];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Merge per-block candidate lists into final top-k.
+// cand_*: [W][Q][K]; one block per query; K rounds of packed argmax-reduce.
+// Packs (monotonic_score_bits << 32 | slot) into uint64 for the reduce.
+// ---------------------------------------------------------------------------
+DEV_INLINE unsigned int f32_mono(float f) {
+  unsigned int u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+DEV_INLINE float mono_f32(unsigned int m) {
+  unsigned int u = (m & 0x80000000u) ? (m & 0x7fffffffu) : ~m;
+  return __uint_as_float(u);
+}
+
+template <int K>
+__global__ void k_topk_merge(const float* __restrict__ cand_score,
+                             const long long* __restrict__ cand_idx,
+                             long long w, int q_count, int k_out,
+                             float* __restrict__ out_score,
+                             long long* __restrict__ out_idx) {
+  const int q = blockIdx.x;
+  const long long total = w * K;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  __shared__ unsigned long long s_best[8];
+
+  // each thread scans strided candidates into a private top-K
+  float tv[K];
+  long long ti[K];
+#pragma unroll
+  for (int i = 0; i < K; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+  for (long long j = threadIdx.x; j < total; j += blockDim.x) {
+    long long src = ((j / K) * q_count + q) * K + (j % K);
+    float s = cand_score[src];
+    if (s > tv[K - 1]) {
+      long long ci = cand_idx[src];
+      float cs = s;
+#pragma unroll
+      for (int i = 0; i < K; ++i) {
+        bool ins = cs > tv[i];
+        float ts = tv[i]; long long tj = ti[i];
+        tv[i] = ins ? cs : tv[i];
+        ti[i] = ins ? ci : ti[i];
+        cs = ins ? ts : cs; ci = ins ? tj : ci;
+      }
+    }
+  }
+
+  // k_out rounds: global argmax over every thread's current head element.
+  int head = 0;
+  for (int r = 0; r < k_out; ++r) {
+    float hv = -1e30f;
+    // select current head value via static unroll (keeps tv in registers)
+#pragma unroll
+    for (int i = 0; i < K; ++i)
+      if (i == head) hv = tv[i];
+    if (head >= K) hv = -1e30f;
+    unsigned long long packed =
+        ((unsigned long long)f32_mono(hv) << 32) | (unsigned int)threadIdx.x;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      unsigned long long o = __shfl_xor(packed, off, WAVE);
+      if (o > packed) packed = o;
+    }
+    if (lane == 0) s_best[wid] = packed;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned long long b = s_best[0];
+      for (int i = 1; i < blockDim.x / WAVE; ++i)
+        if (s_best[i] > b) b = s_best[i];
+      s_best[0] = b;
+    }
+    __syncthreads();
+    unsigned long long b = s_best[0];
+    int winner = (int)(b & 0xffffffffu);
+    if (threadIdx.x == winner) {
+      float wv = tv[0];
+      long long wi = ti[0];
+#pragma unroll
+      for (int i = 0; i < K; ++i)
+        if (i == head) { wv = tv[i]; wi = ti[i]; }
+      out_score[(long long)q * k_out + r] = wv;
+      out_idx[(long long)q * k_out + r] = wi;
+      head++;
+    }
+    __syncthreads();
+    // re-broadcast head increment handled per-thread (winner only)
+  }
+}
+
+// ===========================================================================
+// Host wrappers
+// ===========================================================================
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void l2_normalize_(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous(),
+              "l2_normalize_: need contiguous 2D CUDA tensor");
+  long long n = x.size(0);
+  int d = (int)x.size(1);
+  int blocks = (int)std::min<long long>((n + 3) / 4, 8192);
+  if (blocks == 0) return;
+  if (x.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(d % 8 == 0, "bf16 l2_normalize_ needs D % 8 == 0");
+    hipLaunchKernelGGL(k_l2_normalize_bf16, dim3(blocks), dim3(256), 0,
+                       cur_stream(), (unsigned short*)x.data_ptr(), n, d,
+                       1e-12f);
+  } else if (x.scalar_type() == at::kFloat) {
+    TORCH_CHECK(d % 4 == 0, "f32 l2_normalize_ needs D % 4 == 0");
+    hipLaunchKernelGGL(k_l2_normalize_f32, dim3(blocks), dim3(256), 0,
+                       cur_stream(), x.data_ptr<float>(), n, d, 1e-12f);
+  } else {
+    TORCH_CHECK(false, "l2_normalize_: dtype must be bf16 or f32");
+  }
+  HIP_CHECK_LAST();
+}
+
+void fill_random_unit_(at::Tensor x, long long row_base, long long seed) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+                  x.scalar_type() == at::kBFloat16,
+              "fill_random_unit_: need contiguous 2D bf16 CUDA tensor");
+  long long n = x.size(0);
+  int d = (int)x.size(1);
+  TORCH_CHECK(d % 2 == 0, "fill_random_unit_ needs D % 2 == 0");
+  int blocks = (int)std::min<long long>((n + 3) / 4, 16384);
+  if (blocks == 0) return;
+  hipLaunchKernelGGL(k_fill_random_unit_bf16, dim3(blocks), dim3(256), 0,
+                     cur_stream(), (unsigned short*)x.data_ptr(), n, d,
+                     row_base, (unsigned long long)seed);
+  HIP_CHECK_LAST();
+}
+
+constexpr int KNN_K = 16;
+
+std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
+                                            long long row_base, int k_out) {
+  TORCH_CHECK(db.is_cuda() && db.dim() == 2 && db.is_contiguous() &&
+                  db.scalar_type() == at::kBFloat16,
+              "knn_gemv: db must be contiguous 2D bf16 CUDA");
+  TORCH_CHECK(q.is_cuda() && q.dim() == 2 && q.is_contiguous() &&
+                  q.scalar_type() == at::kBFloat16,
+              "knn_gemv: q must be contiguous 2D bf16 CUDA");
+  long long n = db.size(0);
+  int d = (int)db.size(1);
+  int qc = (int)q.size(0);
+  TORCH_CHECK(q.size(1) == d, "dim mismatch");
+  TORCH_CHECK(qc >= 1 && qc <= 16, "knn_gemv supports 1..16 queries");
+  TORCH_CHECK(d % 8 == 0, "knn_gemv needs D % 8 == 0");
+  TORCH_CHECK(k_out >= 1 && k_out <= KNN_K, "k_out must be <= ", KNN_K);
+
+  int blocks = (int)std::min<long long>((n + 255) / 256, 1280);
+  blocks = std::max(blocks, 1);
+  long long waves = (long long)blocks * 4;
+
+  auto opts_f = db.options().dtype(at::kFloat);
+  auto opts_i = db.options().dtype(at::kLong);
+  at::Tensor cand_s = at::empty({waves, qc, KNN_K}, opts_f);
+  at::Tensor cand_i = at::empty({waves, qc, KNN_K}, opts_i);
+  size_t lds = (size_t)qc * d * sizeof(unsigned short);
+  TORCH_CHECK(lds <= 64 * 1024, "query LDS tile too large");
+
+  hipLaunchKernelGGL((k_knn_gemv_bf16<16, KNN_K>), dim3(blocks), dim3(256),
+                     lds, cur_stream(), (const unsigned short*)db.data_ptr(),
+                     (const unsigned short*)q.data_ptr(), n, d, qc, row_base,
+                     cand_s.data_ptr<float>(), cand_i.data_ptr<long long>());
+  HIP_CHECK_LAST();
+
+  at::Tensor out_s = at::empty({qc, k_out}, opts_f);
+  at::Tensor out_i = at::empty({qc, k_out}, opts_i);
+  hipLaunchKernelGGL((k_topk_merge<KNN_K>), dim3(qc), dim3(256), 0,
+                     cur_stream(), cand_s.data_ptr<float>(),
+                     cand_i.data_ptr<long long>(), waves, qc, k_out,
+                     out_s.data_ptr<float>(), out_i.data_ptr<long long>());
+  HIP_CHECK_LAST();
+  return {out_s, out_i};
+}

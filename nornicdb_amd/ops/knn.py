@@ -1,0 +1,81 @@
+"""Brute-force kNN (cosine / inner-product) over a local shard.
+
+Replaces the reference's GPU scoring path (pkg/gpu/gpu.go:1224 EmbeddingIndex
++ pkg/gpu/cuda/cuda_kernels.cu cosine/topk kernels, which score with one
+thread per vector and a <<<1,1>>> top-k). Here:
+
+- small query batches (Q <= 16): fused HIP kernel `knn_gemv` — wave-per-row
+  short8 vector loads, per-lane register top-k, one read of the shard total;
+- large query batches: tiled hipBLASLt GEMM (torch.matmul) + fused block
+  top-k merge (interim until the MFMA fused score+topk kernel lands);
+- CPU: exact fp32 torch reference (also the numerics oracle for GPU tests).
+
+Scores are inner products — callers are expected to store L2-normalized
+vectors for cosine semantics (same contract as the reference,
+pkg/gpu/gpu.go normalized EmbeddingIndex).
+"""
+
+from typing import Tuple
+
+import torch
+
+from . import native_or_none, require_native
+
+
+def knn_search_exact(
+    db: torch.Tensor, q: torch.Tensor, k: int, row_base: int = 0
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """fp32 exact reference: returns (scores [Q,k] fp32, indices [Q,k] int64)."""
+    scores = q.float() @ db.float().T
+    s, i = torch.topk(scores, k, dim=-1)
+    return s, i + row_base
+
+
+def _knn_gemm_chunked(
+    db: torch.Tensor, q: torch.Tensor, k: int, row_base: int, chunk_rows: int = 4 << 20
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Tiled GEMM + top-k. Avoids materializing the full [N, Q] score matrix."""
+    n = db.shape[0]
+    qf = q.to(db.dtype)
+    best_s = None
+    best_i = None
+    for start in range(0, n, chunk_rows):
+        stop = min(start + chunk_rows, n)
+        scores = (qf @ db[start:stop].T).float()  # [Q, chunk]
+        kk = min(k, stop - start)
+        s, i = torch.topk(scores, kk, dim=-1)
+        i = i + (row_base + start)
+        if best_s is None:
+            best_s, best_i = s, i
+        else:
+            cs = torch.cat([best_s, s], dim=-1)
+            ci = torch.cat([best_i, i], dim=-1)
+            best_s, sel = torch.topk(cs, min(k, cs.shape[-1]), dim=-1)
+            best_i = torch.gather(ci, -1, sel)
+    return best_s, best_i
+
+
+def knn_search(
+    db: torch.Tensor, q: torch.Tensor, k: int, row_base: int = 0
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Top-k inner-product search of q (Q x D) against db (N x D).
+
+    Returns (scores [Q,k] fp32, global indices [Q,k] int64).
+    """
+    assert db.dim() == 2 and q.dim() == 2 and db.shape[1] == q.shape[1]
+    k = min(k, db.shape[0])
+    if not db.is_cuda:
+        return knn_search_exact(db, q, k, row_base)
+
+    nat = native_or_none()
+    if nat is None:
+        require_native()  # raises: no eager fallback on GPU
+    if (
+        q.shape[0] <= 16
+        and k <= 16
+        and db.dtype == torch.bfloat16
+        and db.shape[1] % 8 == 0
+    ):
+        qq = q.to(torch.bfloat16).contiguous()
+        return nat.knn_gemv(db.contiguous(), qq, row_base, k)
+    return _knn_gemm_chunked(db, q, k, row_base)

@@ -1,0 +1,40 @@
+"""Build the NornicDB-AMD native kernel extension in-tree.
+
+Usage:  python setup.py build_ext --inplace
+
+Targets MI355X (gfx950) only. Cross-compiles fine on a GPU-less box.
+"""
+import os
+import glob
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "nornicdb_amd", "csrc")
+
+sources = sorted(
+    s
+    for s in glob.glob(os.path.join(CSRC, "*.cpp")) + glob.glob(os.path.join(CSRC, "*.hip"))
+    # torch's hipify step writes generated copies named *_hip.hip; skip them.
+    if not s.endswith("_hip.hip")
+)
+
+setup(
+    name="nornicdb_amd",
+    version="0.1.0",
+    packages=["nornicdb_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="nornicdb_amd._C",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

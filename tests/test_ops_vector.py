@@ -1,0 +1,113 @@
+"""Numerics tests for tier-1 vector kernels.
+
+GPU tests compare every HIP kernel against a plain PyTorch fp32 reference
+(same contract as the reference repo's GPU-vs-CPU parity tests,
+e.g. pkg/gpu/cuda/cuda_test.go).
+"""
+
+import pytest
+import torch
+
+from nornicdb_amd.ops import knn_search, knn_search_exact, l2_normalize_, fill_random_unit_
+
+
+def test_l2_normalize_cpu():
+    x = torch.randn(64, 128)
+    l2_normalize_(x)
+    n = torch.linalg.vector_norm(x, dim=-1)
+    assert torch.allclose(n, torch.ones_like(n), atol=1e-4)
+
+
+def test_knn_cpu_matches_exact():
+    torch.manual_seed(0)
+    db = torch.randn(500, 64)
+    q = torch.randn(7, 64)
+    s, i = knn_search(db, q, 5)
+    s2, i2 = knn_search_exact(db, q, 5)
+    assert torch.equal(i, i2)
+    assert torch.allclose(s, s2)
+
+
+def test_fill_random_unit_cpu():
+    x = torch.empty(16, 64, dtype=torch.bfloat16)
+    fill_random_unit_(x, row_base=0, seed=7)
+    n = torch.linalg.vector_norm(x.float(), dim=-1)
+    assert torch.allclose(n, torch.ones_like(n), atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_l2_normalize_gpu_bf16():
+    torch.manual_seed(1)
+    x = torch.randn(4096, 1024, device="cuda").to(torch.bfloat16)
+    ref = x.float()
+    ref = ref / torch.linalg.vector_norm(ref, dim=-1, keepdim=True)
+    l2_normalize_(x)
+    torch.cuda.synchronize()
+    err = (x.float() - ref).abs().max().item()
+    assert err < 2e-2  # bf16 storage error
+    n = torch.linalg.vector_norm(x.float(), dim=-1)
+    assert (n - 1).abs().max().item() < 1e-2
+
+
+@pytest.mark.gpu
+def test_l2_normalize_gpu_f32():
+    torch.manual_seed(2)
+    x = torch.randn(1000, 512, device="cuda")
+    ref = x / torch.linalg.vector_norm(x, dim=-1, keepdim=True)
+    l2_normalize_(x)
+    torch.cuda.synchronize()
+    assert torch.allclose(x, ref, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_fill_random_unit_gpu():
+    x = torch.empty(10000, 1024, device="cuda", dtype=torch.bfloat16)
+    fill_random_unit_(x, row_base=1234, seed=42)
+    torch.cuda.synchronize()
+    n = torch.linalg.vector_norm(x.float(), dim=-1)
+    assert (n - 1).abs().max().item() < 2e-2
+    # determinism + row_base consistency: regenerating a slice with shifted
+    # base must reproduce the same rows
+    y = torch.empty(100, 1024, device="cuda", dtype=torch.bfloat16)
+    fill_random_unit_(y, row_base=1234 + 500, seed=42)
+    torch.cuda.synchronize()
+    assert torch.equal(x[500:600], y)
+    # rows are distinct
+    assert not torch.equal(x[0], x[1])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("q_count", [1, 3, 16])
+def test_knn_gemv_gpu_matches_fp32(q_count):
+    torch.manual_seed(3)
+    db = torch.randn(50_000, 256, device="cuda")
+    db = db / torch.linalg.vector_norm(db, dim=-1, keepdim=True)
+    q = db[:q_count] + 0.1 * torch.randn(q_count, 256, device="cuda")
+    q = q / torch.linalg.vector_norm(q, dim=-1, keepdim=True)
+    dbb = db.to(torch.bfloat16)
+    qb = q.to(torch.bfloat16)
+
+    s, i = knn_search(dbb, qb, 10, row_base=777)
+    torch.cuda.synchronize()
+    s_ref, i_ref = knn_search_exact(dbb.float(), qb.float(), 10, row_base=777)
+
+    # exact index match demanded at fp32-identical inputs (bf16 both paths)
+    match = (i == i_ref).float().mean().item()
+    assert match > 0.95, f"top-k overlap too low: {match}"
+    assert torch.allclose(s, s_ref, atol=1e-2)
+    # scores must be sorted descending
+    assert (s[:, :-1] >= s[:, 1:] - 1e-6).all()
+
+
+@pytest.mark.gpu
+def test_knn_large_q_gemm_path():
+    torch.manual_seed(4)
+    db = torch.randn(100_000, 128, device="cuda").to(torch.bfloat16)
+    q = torch.randn(64, 128, device="cuda").to(torch.bfloat16)
+    s, i = knn_search(db, q, 10)
+    torch.cuda.synchronize()
+    s_ref, i_ref = knn_search_exact(db.float(), q.float(), 10)
+    overlap = 0.0
+    for r in range(64):
+        overlap += len(set(i[r].tolist()) & set(i_ref[r].tolist())) / 10
+    assert overlap / 64 > 0.95
