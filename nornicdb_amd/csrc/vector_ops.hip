@@ -378,15 +378,15 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
   hipLaunchKernelGGL((k_knn_gemv_bf16<16, KNN_K>), dim3(blocks), dim3(256),
                      lds, cur_stream(), (const unsigned short*)db.data_ptr(),
                      (const unsigned short*)q.data_ptr(), n, d, qc, row_base,
-                     cand_s.data_ptr<float>(), cand_i.data_ptr<long long>());
+                     cand_s.data_ptr<float>(), reinterpret_cast<long long*>(cand_i.data_ptr<int64_t>()));
   HIP_CHECK_LAST();
 
   at::Tensor out_s = at::empty({qc, k_out}, opts_f);
   at::Tensor out_i = at::empty({qc, k_out}, opts_i);
   hipLaunchKernelGGL((k_topk_merge<KNN_K>), dim3(qc), dim3(256), 0,
                      cur_stream(), cand_s.data_ptr<float>(),
-                     cand_i.data_ptr<long long>(), waves, qc, k_out,
-                     out_s.data_ptr<float>(), out_i.data_ptr<long long>());
+                     reinterpret_cast<long long*>(cand_i.data_ptr<int64_t>()), waves, qc, k_out,
+                     out_s.data_ptr<float>(), reinterpret_cast<long long*>(out_i.data_ptr<int64_t>()));
   HIP_CHECK_LAST();
   return {out_s, out_i};
 }

@@ -1,5 +1,7 @@
 """Quick GPU micro-benchmarks for tier-1 kernels. Prints GB/s and ms."""
 import sys
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import torch
