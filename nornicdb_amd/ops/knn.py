@@ -78,4 +78,38 @@ def knn_search(
     ):
         qq = q.to(torch.bfloat16).contiguous()
         return nat.knn_gemv(db.contiguous(), qq, row_base, k)
+    if (
+        q.shape[0] <= 256
+        and k <= 16
+        and db.dtype == torch.bfloat16
+        and db.shape[1] % 64 == 0
+        and db.shape[0] >= 128
+    ):
+        return _knn_mfma(nat, db, q, k, row_base)
     return _knn_gemm_chunked(db, q, k, row_base)
+
+
+def _knn_mfma(nat, db, q, k, row_base):
+    """Fused MFMA score+topk over full 128-row panels + torch-scored tail."""
+    qn = q.shape[0]
+    qq = q.to(torch.bfloat16)
+    if qn < 256:
+        qq = torch.cat(
+            [qq, torch.zeros(256 - qn, q.shape[1], dtype=torch.bfloat16, device=q.device)]
+        )
+    qq = qq.contiguous()
+    n = db.shape[0]
+    n_main = (n // 128) * 128
+    s, i = nat.knn_mfma(db.narrow(0, 0, n_main), qq, row_base, k)
+    s, i = s[:qn], i[:qn]
+    if n_main < n:
+        tail = db.narrow(0, n_main, n - n_main)
+        ts = (qq[:qn] @ tail.T).float()
+        kk = min(k, n - n_main)
+        bs, bi = torch.topk(ts, kk, dim=-1)
+        bi = bi + (row_base + n_main)
+        cs = torch.cat([s, bs], -1)
+        ci = torch.cat([i, bi], -1)
+        s, sel = torch.topk(cs, k, dim=-1)
+        i = torch.gather(ci, -1, sel)
+    return s, i

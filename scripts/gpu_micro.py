@@ -42,7 +42,10 @@ def main():
 
     q256 = x[:256].clone()
     dt = t(lambda: ops.knn_search(x, q256, 10))
-    print(f"knn gemm-path Q=256: {dt*1e3:.2f} ms  {gb/dt:.0f} GB/s(db)  {256/dt:.0f} qps")
+    print(f"knn fused-MFMA Q=256: {dt*1e3:.2f} ms  {gb/dt:.0f} GB/s(db)  {256/dt:.0f} qps")
+
+    gb4 = n * d * 2 / 1e9
+    print(f"  (mfma fused path active for Q=256: reads {gb4:.1f} GB once)")
 
     # raw matmul reference for the same shape
     w = x[:1 << 20]

@@ -111,3 +111,26 @@ def test_knn_large_q_gemm_path():
     for r in range(64):
         overlap += len(set(i[r].tolist()) & set(i_ref[r].tolist())) / 10
     assert overlap / 64 > 0.95
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n,q_count", [(50_000, 256), (12_345, 100), (128, 256), (100_001, 256)])
+def test_knn_mfma_gpu_matches_fp32(n, q_count):
+    """Random asymmetric inputs: catches any MFMA fragment-layout transpose
+    (guide G9: symmetric inputs cannot detect row/col swaps)."""
+    torch.manual_seed(n + q_count)
+    db = torch.randn(n, 1024, device="cuda")
+    db = db / torch.linalg.vector_norm(db, dim=-1, keepdim=True)
+    q = torch.randn(q_count, 1024, device="cuda")
+    q = q / torch.linalg.vector_norm(q, dim=-1, keepdim=True)
+    dbb = db.to(torch.bfloat16)
+    qb = q.to(torch.bfloat16)
+
+    s, i = knn_search(dbb, qb, 10, row_base=1000)
+    torch.cuda.synchronize()
+    s_ref, i_ref = knn_search_exact(dbb.float(), qb.float(), 10, row_base=1000)
+
+    match = (i == i_ref).float().mean().item()
+    assert match > 0.93, f"top-k index overlap too low: {match}"
+    assert torch.allclose(s, s_ref, atol=2e-2), (s - s_ref).abs().max()
+    assert (s[:, :-1] >= s[:, 1:] - 1e-6).all()
