@@ -124,6 +124,7 @@ def main():
     p.add_argument("--k", type=int, default=10)
     p.add_argument("--layers", type=int, default=24)
     p.add_argument("--skip-recall", action="store_true")
+    p.add_argument("--timing", action="store_true", help="print phase breakdown")
     p.add_argument("--recall-queries", type=int, default=8)
     p.add_argument("--device", default=None)
     args = p.parse_args()
@@ -200,6 +201,24 @@ def main():
         one_step()
     if device.type == "cuda":
         torch.cuda.synchronize()
+
+    if args.timing and device.type == "cuda":
+        def _t(fn, iters=5):
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(iters):
+                fn()
+            torch.cuda.synchronize()
+            return (time.time() - t0) / iters * 1000
+
+        tok = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
+        with torch.no_grad():
+            qf = model(tok).to(shard.dtype)
+        ms_embed = _t(lambda: model(tok))
+        ms_search = _t(lambda: search_step(shard, qf, args.k, row_base, world))
+        log(rank, f"[timing] embed {ms_embed:.1f} ms/batch ({args.batch/ms_embed*1000:.0f} docs/s)  "
+                  f"search {ms_search:.1f} ms/batch ({args.batch/ms_search*1000:.0f} qps; "
+                  f"{shard_rows*args.dim*2/ms_search/1e6:.2f} TB/s shard read)")
 
     # --- recall check (untimed): production path vs fp32 exact over full corpus ---
     recall = None

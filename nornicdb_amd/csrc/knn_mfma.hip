@@ -6,13 +6,14 @@
 // exactly once per query batch.
 //
 // Structure: m97-template GEMM from the CDNA4 guide (§5) —
-//   * 128x256 output tile, 8 waves (512 thr), wave-tile 64x64,
+//   * 64x256 output tile, 4 waves (256 thr), wave-tile 64x64,
 //     mfma_f32_16x16x32_bf16 fragments, BK=64 K-steps
 //   * global -> LDS staging via __builtin_amdgcn_global_load_lds width 16
-//   * 2-barrier K-loop (single-buffered LDS)
-//   * epilogue: C chunks bounce through LDS; 2 threads per query column
-//     scan rows into private register top-K lists (statically unrolled
-//     insertion so they stay in VGPRs), candidates merged by k_topk_merge.
+//   * 2-barrier K-loop; LDS score buffer UNIONed over the staging tiles
+//     (40 KB/block -> 3 blocks/CU occupancy)
+//   * epilogue: C chunks bounce through LDS; 1 thread per query column
+//     scans rows into a private register top-K list (statically unrolled
+//     insertion so it stays in VGPRs), candidates merged by k_topk_merge_i32.
 //
 // Replaces the reference's cublasSgemv + single-thread top-k scan
 // (reference: pkg/gpu/cuda/cuda_kernels.cu:340-480).
@@ -27,35 +28,34 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define L_AS __attribute__((address_space(3)))
 
 // tile geometry
-#define BM 128
+#define BM 64
 #define BN 256
 #define BK 64
-#define NTHREADS 512
-#define KCAND 16
+#define NTHREADS 256
+#define KCAND 12
 // LDS score-scan chunk: rows per chunk
 #define SCH 32
 #define S_STRIDE (BN + 4)
 
-__global__ __launch_bounds__(NTHREADS, 1) void k_knn_mfma(
+__global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
     long long n_panels,  // number of full BM-row panels
     int d,               // inner dim, % BK == 0
     long long row_base, float* __restrict__ cand_score,
     int* __restrict__ cand_idx) {
-  __shared__ unsigned short sA[BM * BK];        // 16 KB
-  __shared__ unsigned short sB[BN * BK];        // 32 KB
-  __shared__ float sS[SCH * S_STRIDE];          // 33 KB score chunk
+  // 40 KB union: K-loop uses sA (8 KB) + sB (32 KB); epilogue reuses the
+  // same space as the 32x260 fp32 score chunk (33.3 KB). Barriers separate
+  // the two lifetimes.
+  __shared__ __align__(16) char smem[(BM * BK + BN * BK) * 2];
+  unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
+  float* sS = (float*)smem;
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
-  const int wid = tid / WAVE;        // 0..7
-  const int wr = wid >> 2;           // wave row 0..1  (64 rows each)
-  const int wc = wid & 3;            // wave col 0..3  (64 cols each)
+  const int wc = tid / WAVE;  // wave col 0..3 (64 cols each); all waves row 0
 
-  // per-thread top-K state: this thread owns query column (tid & 255),
-  // row-half (tid >> 8) of each score chunk.
-  const int own_q = tid & (BN - 1);
-  const int own_half = tid >> 8;  // 0 or 1
+  // per-thread top-K state: this thread owns query column tid.
   float tv[KCAND];
   int ti[KCAND];
 #pragma unroll
@@ -73,48 +73,40 @@ __global__ __launch_bounds__(NTHREADS, 1) void k_knn_mfma(
       for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
 
     for (int kt = 0; kt < d; kt += BK) {
-      // ---- stage A (BM x BK = 16 KB) and B (BN x BK = 32 KB) ----
-      // Each global_load_lds issue: one wave writes 64 lanes x 16 B = 1 KB,
-      // LDS destination is wave-uniform base + lane*16 (linear layout).
-      {
-        // A: 16 chunks of 1 KB; 8 waves x 2 issues
+      // ---- stage A (64 x 64 = 8 KB) and B (256 x 64 = 32 KB) ----
+      // One global_load_lds issue = one wave writes 64 lanes x 16 B = 1 KB
+      // at a wave-uniform LDS base (linear row-major tiles).
 #pragma unroll
-        for (int it = 0; it < 2; ++it) {
-          int chunk = wid * 2 + it;
-          int byte_off = chunk * 1024 + lane * 16;   // within tile
-          int r = byte_off / (BK * 2);
-          int cb = byte_off % (BK * 2);
-          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
-              (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
-          L_AS unsigned int* lp = (L_AS unsigned int*)(
-              (char*)sA + chunk * 1024);
-          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-        }
-        // B: 32 chunks of 1 KB; 8 waves x 4 issues
-#pragma unroll
-        for (int it = 0; it < 4; ++it) {
-          int chunk = wid * 4 + it;
-          int byte_off = chunk * 1024 + lane * 16;
-          int r = byte_off / (BK * 2);
-          int cb = byte_off % (BK * 2);
-          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
-              (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
-          L_AS unsigned int* lp = (L_AS unsigned int*)(
-              (char*)sB + chunk * 1024);
-          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-        }
+      for (int it = 0; it < 2; ++it) {  // A: 8 chunks, 4 waves x 2
+        int chunk = wc * 2 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
-      asm volatile("s_waitcnt vmcnt(0)");
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {  // B: 32 chunks, 4 waves x 8
+        int chunk = wc * 8 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
       __syncthreads();
 
       // ---- MFMA over the staged tile: 2 k-steps of 32 ----
 #pragma unroll
       for (int ks = 0; ks < BK / 32; ++ks) {
-        // A fragment: row = wr*64 + m*16 + (lane&15), k = (lane>>4)*8 + ks*32
         bf16x8 af[4], bf[4];
 #pragma unroll
         for (int m = 0; m < 4; ++m) {
-          int r = wr * 64 + m * 16 + (lane & 15);
+          int r = m * 16 + (lane & 15);
           int k = ks * 32 + (lane >> 4) * 8;
           af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(sA + r * BK + k));
         }
@@ -134,32 +126,27 @@ __global__ __launch_bounds__(NTHREADS, 1) void k_knn_mfma(
       __syncthreads();
     }
 
-    // ---- epilogue: 4 chunks of 32 rows bounce through LDS, then scan ----
+    // ---- epilogue: 2 chunks of 32 rows through LDS (aliases sA/sB) ----
 #pragma unroll
-    for (int h = 0; h < 4; ++h) {
-      // waves with wr == h/2 own these rows; fragment m = (h&1)*2 + {0,1}
-      if (wr == (h >> 1)) {
+    for (int h = 0; h < 2; ++h) {
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
-          int m = (h & 1) * 2 + mi;
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
 #pragma unroll
-          for (int nn = 0; nn < 4; ++nn) {
+        for (int nn = 0; nn < 4; ++nn) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              int row = m * 16 + (lane >> 4) * 4 + r;   // within 64-row wave tile
-              int srow = row - (h & 1) * 32;            // within 32-row chunk
-              int col = wc * 64 + nn * 16 + (lane & 15);
-              sS[srow * S_STRIDE + col] = acc[m][nn][r];
-            }
+          for (int r = 0; r < 4; ++r) {
+            int srow = mi * 16 + (lane >> 4) * 4 + r;  // within 32-row chunk
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            sS[srow * S_STRIDE + col] = acc[m][nn][r];
           }
         }
       }
       __syncthreads();
-      // scan: thread owns column own_q, rows [own_half*16, +16)
-      const long long grow0 = prow + (long long)h * SCH + own_half * 16;
+      const long long grow0 = prow + (long long)h * SCH;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        float s = sS[(own_half * 16 + r) * S_STRIDE + own_q];
+      for (int r = 0; r < SCH; ++r) {
+        float s = sS[r * S_STRIDE + tid];
         if (s > tv[KCAND - 1]) {
           float cs = s;
           int ci = (int)(grow0 + r);
@@ -177,8 +164,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void k_knn_mfma(
     }
   }
 
-  // ---- write candidates: slot = (blockIdx*2 + half) ----
-  long long slot = ((long long)blockIdx.x * 2 + own_half) * BN + own_q;
+  // ---- write candidates ----
+  long long slot = (long long)blockIdx.x * BN + tid;
 #pragma unroll
   for (int i = 0; i < KCAND; ++i) {
     cand_score[slot * KCAND + i] = tv[i];
@@ -264,8 +251,8 @@ __global__ void k_topk_merge_i32(const float* __restrict__ cand_score,
 }
 
 // ---------------------------------------------------------------------------
-// host wrapper: full-panel part of the shard only (n_panels * 128 rows).
-// The python side handles the <128-row tail and q-padding to 256.
+// host wrapper: full-panel part of the shard only (n_panels * BM rows).
+// The python side handles the tail rows and q-padding to 256.
 // ---------------------------------------------------------------------------
 std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out) {
@@ -280,20 +267,19 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
   TORCH_CHECK(q.size(0) == BN, "knn_mfma: q must be padded to ", BN, " rows");
   TORCH_CHECK(q.size(1) == d, "dim mismatch");
   TORCH_CHECK(d % BK == 0, "knn_mfma needs D % 64 == 0");
-  TORCH_CHECK(n % BM == 0, "knn_mfma needs N % 128 == 0 (python pads/tails)");
-  TORCH_CHECK(n / BM < (1LL << 31), "shard too large for int32 local rows");
+  TORCH_CHECK(n % BM == 0, "knn_mfma needs N % ", BM, " == 0 (python pads/tails)");
+  TORCH_CHECK(n < (1LL << 31), "shard too large for int32 local rows");
   TORCH_CHECK(k_out >= 1 && k_out <= KCAND);
 
   long long n_panels = n / BM;
-  int grid = (int)std::min<long long>(n_panels, 1024);
+  int grid = (int)std::min<long long>(n_panels, 2048);
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   auto opts_f = db.options().dtype(at::kFloat);
   auto opts_i32 = db.options().dtype(at::kInt);
   auto opts_i64 = db.options().dtype(at::kLong);
-  long long slots = (long long)grid * 2;
-  at::Tensor cand_s = at::empty({slots, BN, KCAND}, opts_f);
-  at::Tensor cand_i = at::empty({slots, BN, KCAND}, opts_i32);
+  at::Tensor cand_s = at::empty({grid, BN, KCAND}, opts_f);
+  at::Tensor cand_i = at::empty({grid, BN, KCAND}, opts_i32);
 
   hipLaunchKernelGGL(k_knn_mfma, dim3(grid), dim3(NTHREADS), 0, stream,
                      (const unsigned short*)db.data_ptr(),
@@ -305,7 +291,7 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
   at::Tensor out_s = at::empty({BN, k_out}, opts_f);
   at::Tensor out_i = at::empty({BN, k_out}, opts_i64);
   hipLaunchKernelGGL((k_topk_merge_i32<KCAND>), dim3(BN), dim3(256), 0, stream,
-                     cand_s.data_ptr<float>(), cand_i.data_ptr<int>(), slots,
+                     cand_s.data_ptr<float>(), cand_i.data_ptr<int>(), grid,
                      BN, k_out, row_base, out_s.data_ptr<float>(),
                      reinterpret_cast<long long*>(out_i.data_ptr<int64_t>()));
   HIP_CHECK_LAST();

@@ -80,7 +80,7 @@ def knn_search(
         return nat.knn_gemv(db.contiguous(), qq, row_base, k)
     if (
         q.shape[0] <= 256
-        and k <= 16
+        and k <= 12
         and db.dtype == torch.bfloat16
         and db.shape[1] % 64 == 0
         and db.shape[0] >= 128
