@@ -214,11 +214,12 @@ def main():
         tok = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
         with torch.no_grad():
             qf = model(tok).to(shard.dtype)
-        ms_embed = _t(lambda: model(tok))
-        ms_search = _t(lambda: search_step(shard, qf, args.k, row_base, world))
+
+            ms_embed = _t(lambda: model(tok))
+            ms_search = _t(lambda: search_step(shard, qf, args.k, row_base, world))
         log(rank, f"[timing] embed {ms_embed:.1f} ms/batch ({args.batch/ms_embed*1000:.0f} docs/s)  "
                   f"search {ms_search:.1f} ms/batch ({args.batch/ms_search*1000:.0f} qps; "
-                  f"{shard_rows*args.dim*2/ms_search/1e6:.2f} TB/s shard read)")
+                  f"{shard_rows*args.dim*2/ms_search/1e9:.2f} TB/s shard read)")
 
     # --- recall check (untimed): production path vs fp32 exact over full corpus ---
     recall = None
