@@ -1,0 +1,21 @@
+"""Storage engines.
+
+Stack (mirrors reference assembly pkg/nornicdb/db.go:762-915):
+    PersistentEngine (RAM state + WAL + snapshots, replaces Badger+WALEngine)
+      -> AsyncEngine (write-behind, optional)
+        -> NamespacedEngine (multi-tenant prefixing)
+"""
+
+from .types import (ConstraintViolation, Edge, Engine, EventType, Node,
+                    NotFoundError, StorageError, new_id)
+from .memory import MemoryEngine
+from .persistent import PersistentEngine, Transaction
+from .async_engine import AsyncEngine
+from .namespaced import NamespacedEngine
+from .wal import WAL, WALCorruption
+
+__all__ = [
+    "Node", "Edge", "Engine", "EventType", "StorageError", "NotFoundError",
+    "ConstraintViolation", "new_id", "MemoryEngine", "PersistentEngine",
+    "Transaction", "AsyncEngine", "NamespacedEngine", "WAL", "WALCorruption",
+]
