@@ -1,0 +1,219 @@
+"""Cypher AST node definitions."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+
+# ---- expressions ----
+@dataclass
+class Lit:
+    value: Any
+
+
+@dataclass
+class Param:
+    name: str
+
+
+@dataclass
+class Var:
+    name: str
+
+
+@dataclass
+class Prop:
+    expr: Any
+    key: str
+
+
+@dataclass
+class BinOp:
+    op: str
+    left: Any
+    right: Any
+
+
+@dataclass
+class UnOp:
+    op: str
+    expr: Any
+
+
+@dataclass
+class FuncCall:
+    name: str
+    args: List[Any]
+    distinct: bool = False
+    star: bool = False  # count(*)
+
+
+@dataclass
+class ListLit:
+    items: List[Any]
+
+
+@dataclass
+class MapLit:
+    items: List[Tuple[str, Any]]
+
+
+@dataclass
+class Index:
+    expr: Any
+    index: Any           # single index
+    slice: Optional[Tuple[Any, Any]] = None  # (start, end)
+
+
+@dataclass
+class Case:
+    test: Optional[Any]                 # simple CASE operand or None
+    whens: List[Tuple[Any, Any]]
+    default: Optional[Any]
+
+
+@dataclass
+class ListComp:
+    var: str
+    source: Any
+    where: Optional[Any]
+    projection: Optional[Any]
+
+
+@dataclass
+class Quantifier:  # ANY/ALL/NONE/SINGLE(x IN list WHERE p)
+    kind: str
+    var: str
+    source: Any
+    where: Any
+
+
+@dataclass
+class PatternPredicate:  # EXISTS((n)-[:R]->()) / bare pattern in WHERE
+    pattern: "PatternPath"
+
+
+@dataclass
+class CountSubquery:
+    pattern: "PatternPath"
+    where: Optional[Any] = None
+
+
+# ---- patterns ----
+@dataclass
+class NodePattern:
+    var: Optional[str]
+    labels: List[str]
+    props: Optional[Any]  # MapLit / Param
+
+
+@dataclass
+class RelPattern:
+    var: Optional[str]
+    types: List[str]
+    props: Optional[Any]
+    direction: str        # "out", "in", "both"
+    min_hops: int = 1
+    max_hops: int = 1
+    var_length: bool = False
+
+
+@dataclass
+class PatternPath:
+    elements: List[Any]   # alternating NodePattern, RelPattern
+    var: Optional[str] = None  # path variable p = (...)
+
+
+# ---- clauses ----
+@dataclass
+class MatchClause:
+    patterns: List[PatternPath]
+    optional: bool = False
+    where: Optional[Any] = None
+
+
+@dataclass
+class CreateClause:
+    patterns: List[PatternPath]
+
+
+@dataclass
+class MergeClause:
+    pattern: PatternPath
+    on_create: List["SetItem"] = field(default_factory=list)
+    on_match: List["SetItem"] = field(default_factory=list)
+
+
+@dataclass
+class SetItem:
+    target: Any           # Prop or Var
+    value: Any
+    op: str = "="         # "=", "+=", "label"
+    labels: List[str] = field(default_factory=list)
+
+
+@dataclass
+class SetClause:
+    items: List[SetItem]
+
+
+@dataclass
+class RemoveClause:
+    items: List[Any]      # Prop / (Var, labels)
+
+
+@dataclass
+class DeleteClause:
+    exprs: List[Any]
+    detach: bool = False
+
+
+@dataclass
+class ReturnItem:
+    expr: Any
+    alias: Optional[str]
+
+
+@dataclass
+class ReturnClause:
+    items: List[ReturnItem]
+    star: bool = False
+    distinct: bool = False
+    order_by: List[Tuple[Any, bool]] = field(default_factory=list)  # (expr, asc)
+    skip: Optional[Any] = None
+    limit: Optional[Any] = None
+
+
+@dataclass
+class WithClause(ReturnClause):
+    where: Optional[Any] = None
+
+
+@dataclass
+class UnwindClause:
+    expr: Any
+    alias: str
+
+
+@dataclass
+class CallClause:
+    proc: str
+    args: List[Any]
+    yields: List[Tuple[str, Optional[str]]]  # (name, alias)
+    where: Optional[Any] = None
+
+
+@dataclass
+class ForeachClause:
+    var: str
+    source: Any
+    updates: List[Any]
+
+
+@dataclass
+class Query:
+    clauses: List[Any]
+    union: Optional[Tuple[str, "Query"]] = None  # ("UNION"/"UNION ALL", next)
+    explain: bool = False
+    profile: bool = False

@@ -1,0 +1,208 @@
+"""Cypher scalar + aggregate function registry.
+
+Parity: reference pkg/cypher/functions.go + fn/registry.go. APOC functions
+register into the same table under their dotted names (apoc/ registry).
+"""
+
+from __future__ import annotations
+
+import math
+import random
+import re
+import time
+import uuid
+from typing import Any, Callable, Dict, List
+
+from ..storage.types import Edge, Node
+
+
+class CypherRuntimeError(Exception):
+    pass
+
+
+FUNCTIONS: Dict[str, Callable] = {}
+
+
+def register(name: str, fn: Callable = None):
+    if fn is not None:
+        FUNCTIONS[name.lower()] = fn
+        return fn
+
+    def deco(f):
+        FUNCTIONS[name.lower()] = f
+        return f
+    return deco
+
+
+def _num(x):
+    if isinstance(x, bool) or not isinstance(x, (int, float)):
+        raise CypherRuntimeError(f"expected number, got {type(x).__name__}")
+    return x
+
+
+# ---- entity functions ----
+register("id", lambda x: None if x is None else getattr(x, "id", None))
+register("elementid", lambda x: None if x is None else getattr(x, "id", None))
+register("labels", lambda n: None if n is None else list(n.labels))
+register("type", lambda e: None if e is None else e.type)
+register("properties", lambda x: None if x is None else
+         (dict(x.properties) if isinstance(x, (Node, Edge)) else dict(x)))
+register("keys", lambda x: None if x is None else
+         (sorted(x.properties.keys()) if isinstance(x, (Node, Edge)) else sorted(x.keys())))
+register("startnode", lambda e: None if e is None else e._start_ref if hasattr(e, "_start_ref") else e.start_node)
+register("endnode", lambda e: None if e is None else e._end_ref if hasattr(e, "_end_ref") else e.end_node)
+
+
+# ---- scalar ----
+register("coalesce", lambda *a: next((x for x in a if x is not None), None))
+register("head", lambda l: None if not l else l[0])
+register("last", lambda l: None if not l else l[-1])
+register("tail", lambda l: None if l is None else list(l[1:]))
+register("size", lambda x: None if x is None else len(x))
+register("length", lambda x: None if x is None else len(x))
+register("reverse", lambda x: None if x is None else
+         (x[::-1] if isinstance(x, str) else list(reversed(x))))
+register("range", lambda a, b, step=1: list(range(int(a), int(b) + (1 if step > 0 else -1), int(step))))
+register("abs", lambda x: None if x is None else abs(_num(x)))
+register("sign", lambda x: None if x is None else (0 if x == 0 else math.copysign(1, _num(x))))
+register("rand", lambda: random.random())
+register("randomuuid", lambda: str(uuid.uuid4()))
+register("timestamp", lambda: int(time.time() * 1000))
+register("toInteger".lower(), lambda x: _to_int(x))
+register("tofloat", lambda x: _to_float(x))
+register("tostring", lambda x: None if x is None else
+         (str(x).lower() if isinstance(x, bool) else str(x)))
+register("toboolean", lambda x: _to_bool(x))
+
+
+def _to_int(x):
+    if x is None:
+        return None
+    try:
+        if isinstance(x, str):
+            return int(float(x)) if ("." in x or "e" in x.lower()) else int(x)
+        if isinstance(x, bool):
+            return 1 if x else 0
+        return int(x)
+    except (ValueError, TypeError):
+        return None
+
+
+def _to_float(x):
+    if x is None:
+        return None
+    try:
+        return float(x)
+    except (ValueError, TypeError):
+        return None
+
+
+def _to_bool(x):
+    if x is None or isinstance(x, bool):
+        return x
+    if isinstance(x, str):
+        return {"true": True, "false": False}.get(x.lower())
+    return None
+
+
+# ---- math ----
+for n, f in dict(
+    ceil=math.ceil, floor=math.floor, round=round, sqrt=math.sqrt,
+    exp=math.exp, log=math.log, log10=math.log10, sin=math.sin, cos=math.cos,
+    tan=math.tan, asin=math.asin, acos=math.acos, atan=math.atan,
+    degrees=math.degrees, radians=math.radians,
+).items():
+    register(n, (lambda f: lambda x: None if x is None else f(_num(x)))(f))
+register("atan2", lambda y, x: math.atan2(_num(y), _num(x)))
+register("pi", lambda: math.pi)
+register("e", lambda: math.e)
+register("haversin", lambda x: None if x is None else (1 - math.cos(_num(x))) / 2)
+
+
+# ---- strings ----
+register("toupper", lambda s: None if s is None else s.upper())
+register("tolower", lambda s: None if s is None else s.lower())
+register("upper", lambda s: None if s is None else s.upper())
+register("lower", lambda s: None if s is None else s.lower())
+register("trim", lambda s: None if s is None else s.strip())
+register("ltrim", lambda s: None if s is None else s.lstrip())
+register("rtrim", lambda s: None if s is None else s.rstrip())
+register("replace", lambda s, a, b: None if s is None else s.replace(a, b))
+register("split", lambda s, d: None if s is None else s.split(d))
+register("substring", lambda s, start, length=None:
+         None if s is None else (s[start:start + length] if length is not None else s[start:]))
+register("left", lambda s, n: None if s is None else s[:n])
+register("right", lambda s, n: None if s is None else s[-n:] if n else "")
+
+
+# ---- list/aggregation helpers ----
+register("__haslabels", lambda n, labels: n is not None and all(lb in n.labels for lb in labels))
+register("nodes", lambda p: None if p is None else p.nodes)
+register("relationships", lambda p: None if p is None else p.edges)
+register("reduce", None)  # handled in evaluator (needs lazy eval)
+
+
+# ---- aggregates (handled by executor; names listed for detection) ----
+AGGREGATES = {"count", "sum", "avg", "min", "max", "collect", "stdev",
+              "stdevp", "percentilecont", "percentiledisc"}
+
+
+def is_aggregate(name: str) -> bool:
+    return name.lower() in AGGREGATES
+
+
+class Aggregator:
+    def __init__(self, name: str, distinct: bool = False):
+        self.name = name.lower()
+        self.distinct = distinct
+        self.values: List[Any] = []
+        self.seen = set()
+        self.count = 0
+
+    def add(self, v):
+        if self.name == "count" and v is None:
+            return
+        if self.distinct:
+            key = repr(v)
+            if key in self.seen:
+                return
+            self.seen.add(key)
+        if v is not None or self.name == "collect":
+            self.values.append(v)
+        self.count += 1
+
+    def result(self, extra=None):
+        vals = [v for v in self.values if v is not None]
+        n = self.name
+        if n == "count":
+            return self.count
+        if n == "collect":
+            return [v for v in self.values if v is not None]
+        if n == "sum":
+            return sum(vals) if vals else 0
+        if n == "avg":
+            return sum(vals) / len(vals) if vals else None
+        if n == "min":
+            return min(vals) if vals else None
+        if n == "max":
+            return max(vals) if vals else None
+        if n in ("stdev", "stdevp"):
+            if len(vals) < 2:
+                return 0.0
+            m = sum(vals) / len(vals)
+            var = sum((v - m) ** 2 for v in vals)
+            var /= (len(vals) - 1) if n == "stdev" else len(vals)
+            return math.sqrt(var)
+        if n in ("percentilecont", "percentiledisc"):
+            if not vals:
+                return None
+            p = extra if extra is not None else 0.5
+            s = sorted(vals)
+            if n == "percentiledisc":
+                return s[min(int(p * len(s)), len(s) - 1)]
+            idx = p * (len(s) - 1)
+            lo, hi = int(math.floor(idx)), int(math.ceil(idx))
+            if lo == hi:
+                return s[lo]
+            return s[lo] + (s[hi] - s[lo]) * (idx - lo)
+        raise CypherRuntimeError(f"unknown aggregate {n}")

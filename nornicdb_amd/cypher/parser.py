@@ -1,0 +1,736 @@
+"""Recursive-descent Cypher parser producing the AST in ast.py.
+
+Coverage target: the clause set the reference executes via keyword dispatch
+(reference pkg/cypher/executor.go:490 + match.go/create.go/merge.go/
+match_with.go/executor_mutations.go/executor_subqueries.go), expressed as a
+single grammar instead of string dispatch.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+from . import ast as A
+from .lexer import CypherSyntaxError, Token, tokenize
+
+
+class Parser:
+    def __init__(self, text: str):
+        self.toks = tokenize(text)
+        self.i = 0
+
+    # ---- token helpers ----
+    def peek(self, k=0) -> Token:
+        return self.toks[min(self.i + k, len(self.toks) - 1)]
+
+    def next(self) -> Token:
+        t = self.toks[self.i]
+        if t.kind != "EOF":
+            self.i += 1
+        return t
+
+    def at_kw(self, *kws) -> bool:
+        t = self.peek()
+        return t.kind == "KW" and t.value in kws
+
+    def at_op(self, *ops) -> bool:
+        t = self.peek()
+        return t.kind == "OP" and t.value in ops
+
+    def eat_kw(self, kw) -> Token:
+        if not self.at_kw(kw):
+            raise CypherSyntaxError(f"expected {kw}, got {self.peek().value!r} at {self.peek().pos}")
+        return self.next()
+
+    def eat_op(self, op) -> Token:
+        if not self.at_op(op):
+            raise CypherSyntaxError(f"expected {op!r}, got {self.peek().value!r} at {self.peek().pos}")
+        return self.next()
+
+    def try_kw(self, *kws) -> bool:
+        if self.at_kw(*kws):
+            self.next()
+            return True
+        return False
+
+    def try_op(self, op) -> bool:
+        if self.at_op(op):
+            self.next()
+            return True
+        return False
+
+    def ident(self) -> str:
+        t = self.peek()
+        if t.kind == "IDENT":
+            return self.next().value
+        # allow non-reserved keywords as identifiers (e.g. count, all)
+        if t.kind == "KW" and t.value in ("COUNT", "ALL", "ANY", "NONE", "SINGLE",
+                                          "EXISTS", "INDEX", "UNIQUE", "ON", "BY",
+                                          "DATABASE", "SHOW"):
+            return self.next().value.lower()
+        raise CypherSyntaxError(f"expected identifier, got {t.value!r} at {t.pos}")
+
+    # ---- entry ----
+    def parse(self) -> A.Query:
+        q = self._query()
+        if self.peek().kind != "EOF":
+            t = self.peek()
+            raise CypherSyntaxError(f"unexpected {t.value!r} at {t.pos}")
+        return q
+
+    def _query(self) -> A.Query:
+        explain = profile = False
+        if self.try_kw("EXPLAIN"):
+            explain = True
+        elif self.try_kw("PROFILE"):
+            profile = True
+        clauses = []
+        while True:
+            c = self._clause()
+            if c is None:
+                break
+            clauses.append(c)
+            if self.try_op(";"):
+                break
+        if not clauses:
+            raise CypherSyntaxError("empty query")
+        q = A.Query(clauses, explain=explain, profile=profile)
+        if self.at_kw("UNION"):
+            self.next()
+            all_ = self.try_kw("ALL")
+            rest = self._query()
+            q.union = ("UNION ALL" if all_ else "UNION", rest)
+        return q
+
+    def _clause(self):
+        if self.at_kw("MATCH") or self.at_kw("OPTIONAL"):
+            return self._match()
+        if self.at_kw("CREATE"):
+            return self._create()
+        if self.at_kw("MERGE"):
+            return self._merge()
+        if self.at_kw("SET"):
+            return self._set()
+        if self.at_kw("REMOVE"):
+            return self._remove()
+        if self.at_kw("DELETE") or self.at_kw("DETACH"):
+            return self._delete()
+        if self.at_kw("RETURN"):
+            return self._return(A.ReturnClause)
+        if self.at_kw("WITH"):
+            return self._with()
+        if self.at_kw("UNWIND"):
+            return self._unwind()
+        if self.at_kw("CALL"):
+            return self._call()
+        if self.at_kw("FOREACH"):
+            return self._foreach()
+        return None
+
+    # ---- clauses ----
+    def _match(self):
+        optional = self.try_kw("OPTIONAL")
+        self.eat_kw("MATCH")
+        pats = [self._pattern_path()]
+        while self.try_op(","):
+            pats.append(self._pattern_path())
+        where = None
+        if self.try_kw("WHERE"):
+            where = self._expr()
+        return A.MatchClause(pats, optional=optional, where=where)
+
+    def _create(self):
+        self.eat_kw("CREATE")
+        # CREATE INDEX / CONSTRAINT handled by executor as schema commands
+        if self.at_kw("INDEX") or self.at_kw("CONSTRAINT") or self.at_kw("UNIQUE"):
+            raise CypherSyntaxError("schema commands go through the schema API")
+        pats = [self._pattern_path()]
+        while self.try_op(","):
+            pats.append(self._pattern_path())
+        return A.CreateClause(pats)
+
+    def _merge(self):
+        self.eat_kw("MERGE")
+        pat = self._pattern_path()
+        on_create, on_match = [], []
+        while self.at_kw("ON"):
+            self.next()
+            if self.try_kw("CREATE"):
+                self.eat_kw("SET")
+                on_create.extend(self._set_items())
+            elif self.try_kw("MATCH"):
+                self.eat_kw("SET")
+                on_match.extend(self._set_items())
+            else:
+                raise CypherSyntaxError("expected ON CREATE/ON MATCH")
+        return A.MergeClause(pat, on_create, on_match)
+
+    def _set(self):
+        self.eat_kw("SET")
+        return A.SetClause(self._set_items())
+
+    def _set_items(self):
+        items = [self._set_item()]
+        while self.try_op(","):
+            items.append(self._set_item())
+        return items
+
+    def _set_item(self):
+        target = self._expr_atom_chain()
+        if self.at_op(":"):
+            labels = []
+            while self.try_op(":"):
+                labels.append(self.ident())
+            return A.SetItem(target, None, op="label", labels=labels)
+        if self.try_op("+="):
+            return A.SetItem(target, self._expr(), op="+=")
+        self.eat_op("=")
+        return A.SetItem(target, self._expr(), op="=")
+
+    def _remove(self):
+        self.eat_kw("REMOVE")
+        items = []
+        while True:
+            target = self._expr_atom_chain()
+            if self.at_op(":"):
+                labels = []
+                while self.try_op(":"):
+                    labels.append(self.ident())
+                items.append(A.SetItem(target, None, op="label", labels=labels))
+            else:
+                items.append(target)
+            if not self.try_op(","):
+                break
+        return A.RemoveClause(items)
+
+    def _delete(self):
+        detach = self.try_kw("DETACH")
+        self.eat_kw("DELETE")
+        exprs = [self._expr()]
+        while self.try_op(","):
+            exprs.append(self._expr())
+        return A.DeleteClause(exprs, detach=detach)
+
+    def _return(self, cls):
+        self.next()  # RETURN or WITH
+        distinct = self.try_kw("DISTINCT")
+        star = False
+        items = []
+        if self.try_op("*"):
+            star = True
+            if self.try_op(","):
+                items = self._return_items()
+        else:
+            items = self._return_items()
+        order, skip, limit = [], None, None
+        if self.try_kw("ORDER"):
+            self.eat_kw("BY")
+            while True:
+                e = self._expr()
+                asc = True
+                if self.try_kw("DESC") or self.try_kw("DESCENDING"):
+                    asc = False
+                else:
+                    self.try_kw("ASC") or self.try_kw("ASCENDING")
+                order.append((e, asc))
+                if not self.try_op(","):
+                    break
+        if self.try_kw("SKIP"):
+            skip = self._expr()
+        if self.try_kw("LIMIT"):
+            limit = self._expr()
+        c = cls(items=items, star=star, distinct=distinct,
+                order_by=order, skip=skip, limit=limit)
+        return c
+
+    def _return_items(self):
+        items = [self._return_item()]
+        while self.try_op(","):
+            items.append(self._return_item())
+        return items
+
+    def _return_item(self):
+        e = self._expr()
+        alias = None
+        if self.try_kw("AS"):
+            t = self.peek()
+            if t.kind in ("IDENT",):
+                alias = self.next().value
+            elif t.kind == "KW":
+                alias = self.next().value.lower()
+            elif t.kind == "STRING":
+                alias = self.next().value
+            else:
+                raise CypherSyntaxError(f"expected alias at {t.pos}")
+        return A.ReturnItem(e, alias)
+
+    def _with(self):
+        c = self._return(A.WithClause)
+        if self.try_kw("WHERE"):
+            c.where = self._expr()
+        return c
+
+    def _unwind(self):
+        self.eat_kw("UNWIND")
+        e = self._expr()
+        self.eat_kw("AS")
+        return A.UnwindClause(e, self.ident())
+
+    def _call(self):
+        self.eat_kw("CALL")
+        name = self.ident()
+        while self.try_op("."):
+            name += "." + self.ident()
+        args = []
+        if self.try_op("("):
+            if not self.at_op(")"):
+                args.append(self._expr())
+                while self.try_op(","):
+                    args.append(self._expr())
+            self.eat_op(")")
+        yields = []
+        where = None
+        if self.try_kw("YIELD"):
+            while True:
+                y = self.ident()
+                alias = None
+                if self.try_kw("AS"):
+                    alias = self.ident()
+                yields.append((y, alias))
+                if not self.try_op(","):
+                    break
+            if self.try_kw("WHERE"):
+                where = self._expr()
+        return A.CallClause(name, args, yields, where)
+
+    def _foreach(self):
+        self.eat_kw("FOREACH")
+        self.eat_op("(")
+        var = self.ident()
+        self.eat_kw("IN")
+        src = self._expr()
+        self.eat_op("|")
+        updates = []
+        while not self.at_op(")"):
+            c = self._clause()
+            if c is None:
+                break
+            updates.append(c)
+        self.eat_op(")")
+        return A.ForeachClause(var, src, updates)
+
+    # ---- patterns ----
+    def _pattern_path(self) -> A.PatternPath:
+        var = None
+        if (self.peek().kind == "IDENT" and self.peek(1).kind == "OP"
+                and self.peek(1).value == "="
+                and ((self.peek(2).kind == "OP" and self.peek(2).value == "(")
+                     or (self.peek(2).kind == "IDENT"
+                         and self.peek(2).value.lower() in ("shortestpath", "allshortestpaths")))):
+            var = self.next().value
+            self.next()  # =
+        # shortestPath(...) / allShortestPaths(...)
+        if (self.peek().kind == "IDENT"
+                and self.peek().value.lower() in ("shortestpath", "allshortestpaths")
+                and self.peek(1).kind == "OP" and self.peek(1).value == "("):
+            fn = self.next().value.lower()
+            self.eat_op("(")
+            inner = self._pattern_path()
+            self.eat_op(")")
+            inner.var = var or inner.var
+            # mark via attribute; executor reads it
+            inner.shortest = fn  # type: ignore[attr-defined]
+            return inner
+        elems = [self._node_pattern()]
+        while self.at_op("-", "<-", "<"):
+            rel = self._rel_pattern()
+            node = self._node_pattern()
+            elems.append(rel)
+            elems.append(node)
+        return A.PatternPath(elems, var=var)
+
+    def _node_pattern(self) -> A.NodePattern:
+        self.eat_op("(")
+        var = None
+        t = self.peek()
+        if t.kind == "IDENT":
+            var = self.next().value
+        labels = []
+        while self.try_op(":"):
+            labels.append(self.ident())
+        props = None
+        if self.at_op("{"):
+            props = self._map_lit()
+        elif self.peek().kind == "PARAM":
+            props = A.Param(self.next().value)
+        self.eat_op(")")
+        return A.NodePattern(var, labels, props)
+
+    def _rel_pattern(self) -> A.RelPattern:
+        direction = "both"
+        if self.try_op("<-"):
+            direction = "in"
+            left_arrow = True
+        elif self.try_op("<"):
+            self.eat_op("-")
+            direction = "in"
+        else:
+            self.eat_op("-")
+        var, types, props = None, [], None
+        min_h, max_h, var_len = 1, 1, False
+        if self.try_op("["):
+            t = self.peek()
+            if t.kind == "IDENT":
+                var = self.next().value
+            while self.try_op(":"):
+                types.append(self.ident())
+                while self.try_op("|"):
+                    self.try_op(":")
+                    types.append(self.ident())
+            if self.try_op("*"):
+                var_len = True
+                min_h, max_h = 1, 15
+                if self.peek().kind == "INT":
+                    min_h = int(self.next().value)
+                    max_h = min_h
+                if self.try_op(".."):
+                    max_h = 15
+                    if self.peek().kind == "INT":
+                        max_h = int(self.next().value)
+            if self.at_op("{"):
+                props = self._map_lit()
+            elif self.peek().kind == "PARAM":
+                props = A.Param(self.next().value)
+            self.eat_op("]")
+        if self.try_op("->"):
+            if direction == "in":
+                raise CypherSyntaxError("relationship cannot have two arrows")
+            direction = "out"
+        else:
+            self.eat_op("-")
+        return A.RelPattern(var, types, props, direction, min_h, max_h, var_len)
+
+    def _map_lit(self) -> A.MapLit:
+        self.eat_op("{")
+        items = []
+        if not self.at_op("}"):
+            while True:
+                t = self.peek()
+                if t.kind in ("IDENT", "STRING"):
+                    k = self.next().value
+                elif t.kind == "KW":
+                    k = self.next().value.lower()
+                else:
+                    raise CypherSyntaxError(f"bad map key at {t.pos}")
+                self.eat_op(":")
+                items.append((k, self._expr()))
+                if not self.try_op(","):
+                    break
+        self.eat_op("}")
+        return A.MapLit(items)
+
+    # ---- expressions (precedence climbing) ----
+    def _expr(self):
+        return self._or()
+
+    def _or(self):
+        e = self._xor()
+        while self.at_kw("OR"):
+            self.next()
+            e = A.BinOp("OR", e, self._xor())
+        return e
+
+    def _xor(self):
+        e = self._and()
+        while self.at_kw("XOR"):
+            self.next()
+            e = A.BinOp("XOR", e, self._and())
+        return e
+
+    def _and(self):
+        e = self._not()
+        while self.at_kw("AND"):
+            self.next()
+            e = A.BinOp("AND", e, self._not())
+        return e
+
+    def _not(self):
+        if self.try_kw("NOT"):
+            return A.UnOp("NOT", self._not())
+        return self._comparison()
+
+    def _comparison(self):
+        e = self._addsub()
+        ops = []
+        while True:
+            t = self.peek()
+            if t.kind == "OP" and t.value in ("=", "<>", "<", ">", "<=", ">=", "=~"):
+                op = self.next().value
+                rhs = self._addsub()
+                ops.append((op, rhs))
+            elif self.at_kw("IN"):
+                self.next()
+                ops.append(("IN", self._addsub()))
+            elif self.at_kw("STARTS"):
+                self.next()
+                self.eat_kw("WITH")
+                ops.append(("STARTS WITH", self._addsub()))
+            elif self.at_kw("ENDS"):
+                self.next()
+                self.eat_kw("WITH")
+                ops.append(("ENDS WITH", self._addsub()))
+            elif self.at_kw("CONTAINS"):
+                self.next()
+                ops.append(("CONTAINS", self._addsub()))
+            elif self.at_kw("IS"):
+                self.next()
+                neg = self.try_kw("NOT")
+                self.eat_kw("NULL")
+                e = A.UnOp("IS NOT NULL" if neg else "IS NULL", e)
+                continue
+            else:
+                break
+        if not ops:
+            return e
+        # chained comparisons: a < b < c  ==  a<b AND b<c
+        result = None
+        left = e
+        for op, rhs in ops:
+            cmp_ = A.BinOp(op, left, rhs)
+            result = cmp_ if result is None else A.BinOp("AND", result, cmp_)
+            left = rhs
+        return result
+
+    def _addsub(self):
+        e = self._muldiv()
+        while self.at_op("+", "-"):
+            op = self.next().value
+            e = A.BinOp(op, e, self._muldiv())
+        return e
+
+    def _muldiv(self):
+        e = self._power()
+        while self.at_op("*", "/", "%"):
+            op = self.next().value
+            e = A.BinOp(op, e, self._power())
+        return e
+
+    def _power(self):
+        e = self._unary()
+        if self.at_op("^"):
+            self.next()
+            return A.BinOp("^", e, self._power())
+        return e
+
+    def _unary(self):
+        if self.at_op("-"):
+            self.next()
+            return A.UnOp("-", self._unary())
+        if self.at_op("+"):
+            self.next()
+            return self._unary()
+        return self._postfix()
+
+    def _postfix(self):
+        e = self._atom()
+        while True:
+            if self.try_op("."):
+                e = A.Prop(e, self.ident())
+            elif self.at_op("["):
+                self.next()
+                if self.try_op(".."):
+                    hi = None if self.at_op("]") else self._expr()
+                    e = A.Index(e, None, slice=(None, hi))
+                else:
+                    idx = self._expr()
+                    if self.try_op(".."):
+                        hi = None if self.at_op("]") else self._expr()
+                        e = A.Index(e, None, slice=(idx, hi))
+                    else:
+                        e = A.Index(e, idx)
+                self.eat_op("]")
+            elif self.at_op(":") and isinstance(e, (A.Var,)) and self._label_predicate_ok():
+                # n:Label predicate inside expressions
+                labels = []
+                while self.try_op(":"):
+                    labels.append(self.ident())
+                e = A.FuncCall("__haslabels", [e, A.Lit(labels)])
+            else:
+                return e
+
+    def _label_predicate_ok(self):
+        # ':' inside expression context means label predicate only when
+        # followed by an identifier (avoid map-literal confusion)
+        return self.peek(1).kind in ("IDENT",)
+
+    def _atom(self):
+        t = self.peek()
+        if t.kind == "INT":
+            self.next()
+            return A.Lit(int(t.value))
+        if t.kind == "FLOAT":
+            self.next()
+            return A.Lit(float(t.value))
+        if t.kind == "STRING":
+            self.next()
+            return A.Lit(t.value)
+        if t.kind == "PARAM":
+            self.next()
+            return A.Param(t.value)
+        if t.kind == "KW":
+            if t.value == "NULL":
+                self.next()
+                return A.Lit(None)
+            if t.value == "TRUE":
+                self.next()
+                return A.Lit(True)
+            if t.value == "FALSE":
+                self.next()
+                return A.Lit(False)
+            if t.value == "COUNT":
+                self.next()
+                self.eat_op("(")
+                if self.try_op("*"):
+                    self.eat_op(")")
+                    return A.FuncCall("count", [], star=True)
+                distinct = self.try_kw("DISTINCT")
+                arg = self._expr()
+                self.eat_op(")")
+                return A.FuncCall("count", [arg], distinct=distinct)
+            if t.value == "CASE":
+                return self._case()
+            if t.value == "EXISTS":
+                self.next()
+                self.eat_op("(")
+                if self.at_op("("):
+                    pat = self._pattern_path()
+                    self.eat_op(")")
+                    return A.PatternPredicate(pat)
+                inner = self._expr()
+                self.eat_op(")")
+                return A.UnOp("IS NOT NULL", inner)
+            if t.value in ("ANY", "ALL", "NONE", "SINGLE"):
+                kind = t.value
+                if self.peek(1).kind == "OP" and self.peek(1).value == "(":
+                    self.next()
+                    self.eat_op("(")
+                    var = self.ident()
+                    self.eat_kw("IN")
+                    src = self._expr()
+                    self.eat_kw("WHERE")
+                    wh = self._expr()
+                    self.eat_op(")")
+                    return A.Quantifier(kind, var, src, wh)
+        if t.kind == "OP" and t.value == "(":
+            # pattern predicate like (n)-[:R]->(m) in boolean context
+            save = self.i
+            try:
+                pat = self._pattern_path()
+                if len(pat.elements) > 1:
+                    return A.PatternPredicate(pat)
+                self.i = save
+            except CypherSyntaxError:
+                self.i = save
+            self.next()
+            e = self._expr()
+            self.eat_op(")")
+            return e
+        if t.kind == "OP" and t.value == "[":
+            self.next()
+            # list comprehension? [x IN src WHERE p | proj]
+            if (self.peek().kind == "IDENT" and self.peek(1).kind == "KW"
+                    and self.peek(1).value == "IN"):
+                var = self.next().value
+                self.next()  # IN
+                src = self._expr()
+                where = None
+                proj = None
+                if self.try_kw("WHERE"):
+                    where = self._expr()
+                if self.try_op("|"):
+                    proj = self._expr()
+                self.eat_op("]")
+                return A.ListComp(var, src, where, proj)
+            items = []
+            if not self.at_op("]"):
+                items.append(self._expr())
+                while self.try_op(","):
+                    items.append(self._expr())
+            self.eat_op("]")
+            return A.ListLit(items)
+        if t.kind == "OP" and t.value == "{":
+            return self._map_lit()
+        if t.kind == "IDENT":
+            # function call?
+            if self.peek(1).kind == "OP" and self.peek(1).value == "(":
+                name = self.next().value
+                while self.try_op("."):
+                    name += "." + self.ident()
+                # qualified name may end before '('
+                self.eat_op("(")
+                distinct = self.try_kw("DISTINCT")
+                args = []
+                if not self.at_op(")"):
+                    args.append(self._expr())
+                    while self.try_op(","):
+                        args.append(self._expr())
+                self.eat_op(")")
+                return A.FuncCall(name.lower(), args, distinct=distinct)
+            # dotted function name: ns.fn(...)
+            if (self.peek(1).kind == "OP" and self.peek(1).value == "."
+                    and self._is_dotted_call()):
+                name = self.next().value
+                while self.try_op("."):
+                    name += "." + self.ident()
+                self.eat_op("(")
+                distinct = self.try_kw("DISTINCT")
+                args = []
+                if not self.at_op(")"):
+                    args.append(self._expr())
+                    while self.try_op(","):
+                        args.append(self._expr())
+                self.eat_op(")")
+                return A.FuncCall(name.lower(), args, distinct=distinct)
+            return A.Var(self.next().value)
+        raise CypherSyntaxError(f"unexpected token {t.value!r} at {t.pos}")
+
+    def _is_dotted_call(self):
+        """Lookahead: IDENT(.IDENT)+( — distinguishes apoc.coll.max(...) from n.prop"""
+        j = self.i
+        toks = self.toks
+        if toks[j].kind != "IDENT":
+            return False
+        j += 1
+        seen_dot = False
+        while (j + 1 < len(toks) and toks[j].kind == "OP" and toks[j].value == "."
+               and toks[j + 1].kind in ("IDENT", "KW")):
+            seen_dot = True
+            j += 2
+        return seen_dot and j < len(toks) and toks[j].kind == "OP" and toks[j].value == "("
+
+    def _case(self):
+        self.eat_kw("CASE")
+        test = None
+        if not self.at_kw("WHEN"):
+            test = self._expr()
+        whens = []
+        while self.try_kw("WHEN"):
+            cond = self._expr()
+            self.eat_kw("THEN")
+            whens.append((cond, self._expr()))
+        default = None
+        if self.try_kw("ELSE"):
+            default = self._expr()
+        self.eat_kw("END")
+        return A.Case(test, whens, default)
+
+    def _expr_atom_chain(self):
+        """Left side of SET: variable with optional property chain."""
+        e = A.Var(self.ident())
+        while self.try_op("."):
+            e = A.Prop(e, self.ident())
+        return e
+
+
+def parse(text: str) -> A.Query:
+    return Parser(text).parse()

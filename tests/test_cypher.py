@@ -1,0 +1,278 @@
+"""Cypher engine behavioral tests.
+
+Mirrors the reference's Cypher test strategy (pkg/cypher/*_test.go with
+SetupTestExecutor fixtures): every test runs real queries against a fresh
+MemoryEngine.
+"""
+
+import pytest
+
+from nornicdb_amd.cypher import CypherSyntaxError, Executor
+from nornicdb_amd.storage import MemoryEngine
+
+
+@pytest.fixture
+def ex():
+    return Executor(MemoryEngine())
+
+
+def rows(res):
+    return res.rows
+
+
+class TestCreateMatch:
+    def test_create_and_match(self, ex):
+        ex.execute("CREATE (n:Person {name: 'Ada', age: 36})")
+        r = ex.execute("MATCH (n:Person) RETURN n.name, n.age")
+        assert r.columns == ["n.name", "n.age"]
+        assert r.rows == [["Ada", 36]]
+
+    def test_create_returns(self, ex):
+        r = ex.execute("CREATE (n:Person {name: 'Bob'}) RETURN n.name AS name")
+        assert r.rows == [["Bob"]]
+        assert r.stats["nodes_created"] == 1
+
+    def test_create_relationship(self, ex):
+        ex.execute("CREATE (a:P {name:'a'})-[:KNOWS {since: 2020}]->(b:P {name:'b'})")
+        r = ex.execute("MATCH (a:P)-[r:KNOWS]->(b:P) RETURN a.name, r.since, b.name")
+        assert r.rows == [["a", 2020, "b"]]
+
+    def test_match_direction(self, ex):
+        ex.execute("CREATE (a:P {n:'a'})-[:R]->(b:P {n:'b'})")
+        assert ex.execute("MATCH (x:P)<-[:R]-(y:P) RETURN x.n, y.n").rows == [["b", "a"]]
+        r = ex.execute("MATCH (x:P {n:'a'})-[:R]-(y:P) RETURN y.n")
+        assert r.rows == [["b"]]
+
+    def test_multi_label_and_props_filter(self, ex):
+        ex.execute("CREATE (:A:B {x: 1}), (:A {x: 2}), (:B {x: 3})")
+        assert len(ex.execute("MATCH (n:A:B) RETURN n").rows) == 1
+        assert ex.execute("MATCH (n:A {x: 2}) RETURN n.x").rows == [[2]]
+
+    def test_where(self, ex):
+        ex.execute("UNWIND range(1, 10) AS i CREATE (:Num {v: i})")
+        r = ex.execute("MATCH (n:Num) WHERE n.v > 7 RETURN n.v ORDER BY n.v")
+        assert r.rows == [[8], [9], [10]]
+
+    def test_where_and_or_not(self, ex):
+        ex.execute("UNWIND range(1, 10) AS i CREATE (:N {v: i})")
+        r = ex.execute(
+            "MATCH (n:N) WHERE (n.v > 8 OR n.v < 3) AND NOT n.v = 9 "
+            "RETURN n.v ORDER BY n.v")
+        assert r.rows == [[1], [2], [10]]
+
+    def test_parameters(self, ex):
+        ex.execute("CREATE (:P {name: $name, age: $age})",
+                   {"name": "Eve", "age": 30})
+        r = ex.execute("MATCH (n:P {name: $name}) RETURN n.age", {"name": "Eve"})
+        assert r.rows == [[30]]
+
+    def test_optional_match(self, ex):
+        ex.execute("CREATE (:A {n: 1})")
+        r = ex.execute("MATCH (a:A) OPTIONAL MATCH (a)-[:R]->(b) RETURN a.n, b")
+        assert r.rows == [[1, None]]
+
+    def test_var_length(self, ex):
+        ex.execute("CREATE (a:V {n:1})-[:R]->(b:V {n:2})-[:R]->(c:V {n:3})")
+        r = ex.execute("MATCH (a:V {n:1})-[:R*1..2]->(x) RETURN x.n ORDER BY x.n")
+        assert r.rows == [[2], [3]]
+
+    def test_shortest_path(self, ex):
+        ex.execute("CREATE (a:S {n:1})-[:R]->(b:S {n:2})-[:R]->(c:S {n:3})")
+        ex.execute("MATCH (a:S {n:1}), (c:S {n:3}) CREATE (a)-[:R]->(c)")
+        r = ex.execute(
+            "MATCH p = shortestPath((a:S {n:1})-[:R*1..5]->(c:S {n:3})) "
+            "RETURN length(p)")
+        assert r.rows == [[1]]
+
+
+class TestAggregation:
+    def test_count_sum_avg(self, ex):
+        ex.execute("UNWIND range(1, 4) AS i CREATE (:N {v: i})")
+        r = ex.execute("MATCH (n:N) RETURN count(n), sum(n.v), avg(n.v), min(n.v), max(n.v)")
+        assert r.rows == [[4, 10, 2.5, 1, 4]]
+
+    def test_count_star_empty(self, ex):
+        assert ex.execute("MATCH (n:Nope) RETURN count(*)").rows == [[0]]
+
+    def test_group_by(self, ex):
+        ex.execute("UNWIND [['a',1],['a',2],['b',5]] AS p CREATE (:G {k: p[0], v: p[1]})")
+        r = ex.execute("MATCH (n:G) RETURN n.k AS k, sum(n.v) AS s ORDER BY k")
+        assert r.rows == [["a", 3], ["b", 5]]
+
+    def test_collect_distinct(self, ex):
+        ex.execute("UNWIND [1,2,2,3] AS v CREATE (:C {v: v})")
+        r = ex.execute("MATCH (n:C) RETURN collect(DISTINCT n.v) AS vs")
+        assert sorted(r.rows[0][0]) == [1, 2, 3]
+
+    def test_count_in_expression(self, ex):
+        ex.execute("UNWIND range(1,5) AS i CREATE (:E)")
+        r = ex.execute("MATCH (n:E) RETURN count(n) * 2 + 1")
+        assert r.rows == [[11]]
+
+
+class TestWriteClauses:
+    def test_set_and_remove(self, ex):
+        ex.execute("CREATE (:P {name: 'x'})")
+        ex.execute("MATCH (n:P) SET n.age = 5, n.name = 'y'")
+        assert ex.execute("MATCH (n:P) RETURN n.name, n.age").rows == [["y", 5]]
+        ex.execute("MATCH (n:P) REMOVE n.age")
+        assert ex.execute("MATCH (n:P) RETURN n.age").rows == [[None]]
+
+    def test_set_labels(self, ex):
+        ex.execute("CREATE (:A)")
+        ex.execute("MATCH (n:A) SET n:B:C")
+        r = ex.execute("MATCH (n:B) RETURN labels(n)")
+        assert sorted(r.rows[0][0]) == ["A", "B", "C"]
+        ex.execute("MATCH (n:A) REMOVE n:C")
+        assert ex.execute("MATCH (n:C) RETURN count(n)").rows == [[0]]
+
+    def test_set_plus_equals(self, ex):
+        ex.execute("CREATE (:M {a: 1, b: 2})")
+        ex.execute("MATCH (n:M) SET n += {b: 3, c: 4}")
+        r = ex.execute("MATCH (n:M) RETURN n.a, n.b, n.c")
+        assert r.rows == [[1, 3, 4]]
+
+    def test_delete_and_detach(self, ex):
+        ex.execute("CREATE (a:D {n:1})-[:R]->(b:D {n:2})")
+        with pytest.raises(Exception):
+            ex.execute("MATCH (n:D {n:1}) DELETE n")
+        ex.execute("MATCH (n:D {n:1}) DETACH DELETE n")
+        assert ex.execute("MATCH (n:D) RETURN count(n)").rows == [[1]]
+
+    def test_merge_creates_then_matches(self, ex):
+        ex.execute("MERGE (n:U {name: 'solo'})")
+        ex.execute("MERGE (n:U {name: 'solo'})")
+        assert ex.execute("MATCH (n:U) RETURN count(n)").rows == [[1]]
+
+    def test_merge_on_create_on_match(self, ex):
+        ex.execute("MERGE (n:W {k: 1}) ON CREATE SET n.created = true")
+        ex.execute("MERGE (n:W {k: 1}) ON MATCH SET n.matched = true")
+        r = ex.execute("MATCH (n:W) RETURN n.created, n.matched")
+        assert r.rows == [[True, True]]
+
+    def test_merge_relationship(self, ex):
+        ex.execute("CREATE (:MA {n:1}), (:MB {n:2})")
+        for _ in range(2):
+            ex.execute("MATCH (a:MA), (b:MB) MERGE (a)-[:L]->(b)")
+        assert ex.execute("MATCH (:MA)-[r:L]->(:MB) RETURN count(r)").rows == [[1]]
+
+
+class TestWithUnwind:
+    def test_with_filter_aggregate(self, ex):
+        ex.execute("UNWIND range(1, 10) AS i CREATE (:T {v: i})")
+        r = ex.execute(
+            "MATCH (n:T) WITH n.v AS v WHERE v % 2 = 0 "
+            "RETURN sum(v) AS total")
+        assert r.rows == [[30]]
+
+    def test_with_order_limit(self, ex):
+        ex.execute("UNWIND range(1, 5) AS i CREATE (:T2 {v: i})")
+        r = ex.execute(
+            "MATCH (n:T2) WITH n ORDER BY n.v DESC LIMIT 2 RETURN n.v")
+        assert sorted(x[0] for x in r.rows) == [4, 5]
+
+    def test_unwind_nested(self, ex):
+        r = ex.execute("UNWIND [[1,2],[3]] AS l UNWIND l AS x RETURN x")
+        assert [x[0] for x in r.rows] == [1, 2, 3]
+
+    def test_union(self, ex):
+        r = ex.execute("RETURN 1 AS x UNION ALL RETURN 1 AS x")
+        assert r.rows == [[1], [1]]
+        r = ex.execute("RETURN 1 AS x UNION RETURN 1 AS x")
+        assert r.rows == [[1]]
+
+
+class TestExpressions:
+    def test_arithmetic_and_strings(self, ex):
+        r = ex.execute("RETURN 2 + 3 * 4, 'a' + 'b', 10 / 4.0, 7 % 3")
+        assert r.rows == [[14, "ab", 2.5, 1]]
+
+    def test_string_predicates(self, ex):
+        r = ex.execute(
+            "RETURN 'hello' STARTS WITH 'he', 'hello' ENDS WITH 'lo', "
+            "'hello' CONTAINS 'ell', 'hello' =~ 'h.*o'")
+        assert r.rows == [[True, True, True, True]]
+
+    def test_in_and_lists(self, ex):
+        r = ex.execute("RETURN 3 IN [1,2,3], [1,2,3][1], [1,2,3][0..2], size([1,2])")
+        assert r.rows == [[True, 2, [1, 2], 2]]
+
+    def test_case(self, ex):
+        r = ex.execute(
+            "UNWIND [1,2,3] AS x "
+            "RETURN CASE WHEN x = 1 THEN 'one' WHEN x = 2 THEN 'two' ELSE 'many' END")
+        assert [x[0] for x in r.rows] == ["one", "two", "many"]
+
+    def test_null_semantics(self, ex):
+        r = ex.execute("RETURN null = null, null IS NULL, 1 + null, coalesce(null, 5)")
+        assert r.rows == [[None, True, None, 5]]
+
+    def test_functions(self, ex):
+        r = ex.execute(
+            "RETURN toUpper('ab'), toInteger('42'), abs(-3), size('abcd'), "
+            "split('a,b', ','), trim('  x ')")
+        assert r.rows == [["AB", 42, 3, 4, ["a", "b"], "x"]]
+
+    def test_list_comprehension(self, ex):
+        r = ex.execute("RETURN [x IN range(1,5) WHERE x % 2 = 0 | x * 10] AS l")
+        assert r.rows == [[[20, 40]]]
+
+    def test_quantifiers(self, ex):
+        r = ex.execute(
+            "RETURN any(x IN [1,2] WHERE x > 1), all(x IN [1,2] WHERE x > 0), "
+            "none(x IN [1,2] WHERE x > 5), single(x IN [1,2] WHERE x = 2)")
+        assert r.rows == [[True, True, True, True]]
+
+    def test_entity_functions(self, ex):
+        ex.execute("CREATE (a:F {x:1})-[:REL]->(b:F)")
+        r = ex.execute("MATCH (a:F {x:1})-[r]->(b) RETURN type(r), labels(a), "
+                       "startNode(r).x")
+        assert r.rows[0][0] == "REL"
+        assert r.rows[0][1] == ["F"]
+        assert r.rows[0][2] == 1
+
+    def test_pattern_predicate(self, ex):
+        ex.execute("CREATE (a:PP {n:1})-[:R]->(:PP {n:2})")
+        ex.execute("CREATE (:PP {n:3})")
+        r = ex.execute(
+            "MATCH (a:PP) WHERE (a)-[:R]->() RETURN a.n")
+        assert r.rows == [[1]]
+        r = ex.execute("MATCH (a:PP) WHERE NOT (a)-[:R]->() RETURN a.n ORDER BY a.n")
+        assert r.rows == [[2], [3]]
+
+
+class TestOrderSkipLimit:
+    def test_order_by_multiple(self, ex):
+        ex.execute("UNWIND [['b',1],['a',2],['a',1]] AS p CREATE (:O {k:p[0], v:p[1]})")
+        r = ex.execute("MATCH (n:O) RETURN n.k, n.v ORDER BY n.k, n.v DESC")
+        assert r.rows == [["a", 2], ["a", 1], ["b", 1]]
+
+    def test_skip_limit(self, ex):
+        ex.execute("UNWIND range(1, 10) AS i CREATE (:SL {v: i})")
+        r = ex.execute("MATCH (n:SL) RETURN n.v ORDER BY n.v SKIP 3 LIMIT 2")
+        assert r.rows == [[4], [5]]
+
+    def test_distinct(self, ex):
+        ex.execute("UNWIND [1,1,2] AS v CREATE (:DI {v: v})")
+        r = ex.execute("MATCH (n:DI) RETURN DISTINCT n.v ORDER BY n.v")
+        assert r.rows == [[1], [2]]
+
+
+class TestErrors:
+    def test_syntax_error(self, ex):
+        with pytest.raises(CypherSyntaxError):
+            ex.execute("MATCH (n RETURN n")
+
+    def test_unknown_function(self, ex):
+        with pytest.raises(Exception):
+            ex.execute("RETURN nosuchfn(1)")
+
+    def test_undefined_variable(self, ex):
+        with pytest.raises(Exception):
+            ex.execute("RETURN zzz")
+
+
+class TestForeach:
+    def test_foreach_create(self, ex):
+        ex.execute("FOREACH (i IN range(1, 3) | CREATE (:FE {v: i}))")
+        assert ex.execute("MATCH (n:FE) RETURN count(n)").rows == [[3]]
