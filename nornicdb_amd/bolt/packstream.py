@@ -1,0 +1,271 @@
+"""PackStream binary serialization (Bolt wire format).
+
+Parity: reference pkg/bolt/packstream.go (1.3K LoC). Implements PackStream
+v1/v2 markers plus the Bolt graph structures (Node 'N', Relationship 'R',
+UnboundRelationship 'r', Path 'P').
+"""
+
+from __future__ import annotations
+
+import struct
+from typing import Any, Dict, List, Tuple
+
+
+class Structure:
+    def __init__(self, tag: int, fields: List[Any]):
+        self.tag = tag
+        self.fields = fields
+
+    def __repr__(self):
+        return f"Structure(0x{self.tag:02x}, {self.fields!r})"
+
+    def __eq__(self, o):
+        return isinstance(o, Structure) and o.tag == self.tag and o.fields == self.fields
+
+
+class PackStreamError(Exception):
+    pass
+
+
+# ---------------------------------------------------------------- packing
+def pack(value: Any) -> bytes:
+    out = bytearray()
+    _pack_into(out, value)
+    return bytes(out)
+
+
+def _pack_into(out: bytearray, v: Any) -> None:
+    if v is None:
+        out.append(0xC0)
+    elif v is True:
+        out.append(0xC3)
+    elif v is False:
+        out.append(0xC2)
+    elif isinstance(v, int):
+        _pack_int(out, v)
+    elif isinstance(v, float):
+        out.append(0xC1)
+        out += struct.pack(">d", v)
+    elif isinstance(v, str):
+        b = v.encode("utf-8")
+        n = len(b)
+        if n < 0x10:
+            out.append(0x80 + n)
+        elif n < 0x100:
+            out += bytes((0xD0, n))
+        elif n < 0x10000:
+            out.append(0xD1)
+            out += struct.pack(">H", n)
+        else:
+            out.append(0xD2)
+            out += struct.pack(">I", n)
+        out += b
+    elif isinstance(v, (bytes, bytearray)):
+        n = len(v)
+        if n < 0x100:
+            out += bytes((0xCC, n))
+        elif n < 0x10000:
+            out.append(0xCD)
+            out += struct.pack(">H", n)
+        else:
+            out.append(0xCE)
+            out += struct.pack(">I", n)
+        out += v
+    elif isinstance(v, (list, tuple)):
+        n = len(v)
+        if n < 0x10:
+            out.append(0x90 + n)
+        elif n < 0x100:
+            out += bytes((0xD4, n))
+        elif n < 0x10000:
+            out.append(0xD5)
+            out += struct.pack(">H", n)
+        else:
+            out.append(0xD6)
+            out += struct.pack(">I", n)
+        for item in v:
+            _pack_into(out, item)
+    elif isinstance(v, dict):
+        n = len(v)
+        if n < 0x10:
+            out.append(0xA0 + n)
+        elif n < 0x100:
+            out += bytes((0xD8, n))
+        elif n < 0x10000:
+            out.append(0xD9)
+            out += struct.pack(">H", n)
+        else:
+            out.append(0xDA)
+            out += struct.pack(">I", n)
+        for k, item in v.items():
+            _pack_into(out, str(k))
+            _pack_into(out, item)
+    elif isinstance(v, Structure):
+        n = len(v.fields)
+        if n < 0x10:
+            out.append(0xB0 + n)
+        else:
+            raise PackStreamError("struct too large")
+        out.append(v.tag)
+        for f in v.fields:
+            _pack_into(out, f)
+    else:
+        raise PackStreamError(f"cannot pack {type(v).__name__}")
+
+
+def _pack_int(out: bytearray, v: int) -> None:
+    if -16 <= v < 128:
+        out += struct.pack(">b", v)
+    elif -128 <= v < 128:
+        out.append(0xC8)
+        out += struct.pack(">b", v)
+    elif -32768 <= v < 32768:
+        out.append(0xC9)
+        out += struct.pack(">h", v)
+    elif -2147483648 <= v < 2147483648:
+        out.append(0xCA)
+        out += struct.pack(">i", v)
+    else:
+        out.append(0xCB)
+        out += struct.pack(">q", v)
+
+
+# -------------------------------------------------------------- unpacking
+class Unpacker:
+    def __init__(self, data: bytes, offset: int = 0):
+        self.data = data
+        self.i = offset
+
+    def _take(self, n: int) -> bytes:
+        if self.i + n > len(self.data):
+            raise PackStreamError("truncated data")
+        b = self.data[self.i:self.i + n]
+        self.i += n
+        return b
+
+    def unpack(self) -> Any:
+        m = self._take(1)[0]
+        if m <= 0x7F:
+            return m
+        if m >= 0xF0:
+            return m - 0x100
+        if 0x80 <= m <= 0x8F:
+            return self._take(m & 0x0F).decode("utf-8")
+        if 0x90 <= m <= 0x9F:
+            return [self.unpack() for _ in range(m & 0x0F)]
+        if 0xA0 <= m <= 0xAF:
+            return {self.unpack(): self.unpack() for _ in range(m & 0x0F)}
+        if 0xB0 <= m <= 0xBF:
+            n = m & 0x0F
+            tag = self._take(1)[0]
+            return Structure(tag, [self.unpack() for _ in range(n)])
+        if m == 0xC0:
+            return None
+        if m == 0xC1:
+            return struct.unpack(">d", self._take(8))[0]
+        if m == 0xC2:
+            return False
+        if m == 0xC3:
+            return True
+        if m == 0xC8:
+            return struct.unpack(">b", self._take(1))[0]
+        if m == 0xC9:
+            return struct.unpack(">h", self._take(2))[0]
+        if m == 0xCA:
+            return struct.unpack(">i", self._take(4))[0]
+        if m == 0xCB:
+            return struct.unpack(">q", self._take(8))[0]
+        if m == 0xCC:
+            return bytes(self._take(self._take(1)[0]))
+        if m == 0xCD:
+            return bytes(self._take(struct.unpack(">H", self._take(2))[0]))
+        if m == 0xCE:
+            return bytes(self._take(struct.unpack(">I", self._take(4))[0]))
+        if m == 0xD0:
+            return self._take(self._take(1)[0]).decode("utf-8")
+        if m == 0xD1:
+            return self._take(struct.unpack(">H", self._take(2))[0]).decode("utf-8")
+        if m == 0xD2:
+            return self._take(struct.unpack(">I", self._take(4))[0]).decode("utf-8")
+        if m == 0xD4:
+            return [self.unpack() for _ in range(self._take(1)[0])]
+        if m == 0xD5:
+            return [self.unpack() for _ in range(struct.unpack(">H", self._take(2))[0])]
+        if m == 0xD6:
+            return [self.unpack() for _ in range(struct.unpack(">I", self._take(4))[0])]
+        if m == 0xD8:
+            return {self.unpack(): self.unpack() for _ in range(self._take(1)[0])}
+        if m == 0xD9:
+            return {self.unpack(): self.unpack()
+                    for _ in range(struct.unpack(">H", self._take(2))[0])}
+        if m == 0xDA:
+            return {self.unpack(): self.unpack()
+                    for _ in range(struct.unpack(">I", self._take(4))[0])}
+        if m in (0xDC, 0xDD):
+            n = self._take(1)[0] if m == 0xDC else struct.unpack(">H", self._take(2))[0]
+            tag = self._take(1)[0]
+            return Structure(tag, [self.unpack() for _ in range(n)])
+        raise PackStreamError(f"unknown marker 0x{m:02x}")
+
+
+def unpack(data: bytes) -> Any:
+    return Unpacker(data).unpack()
+
+
+# ----------------------------------------------------- Bolt graph structs
+NODE_TAG = 0x4E
+REL_TAG = 0x52
+UNBOUND_REL_TAG = 0x72
+PATH_TAG = 0x50
+
+
+class IdMap:
+    """Stable string-id <-> int-id mapping for Bolt's integer entity ids."""
+
+    def __init__(self):
+        self._s2i: Dict[str, int] = {}
+        self._i2s: Dict[int, str] = {}
+
+    def to_int(self, s: str) -> int:
+        i = self._s2i.get(s)
+        if i is None:
+            i = len(self._s2i) + 1
+            self._s2i[s] = i
+            self._i2s[i] = s
+        return i
+
+    def to_str(self, i: int) -> str:
+        return self._i2s.get(i, str(i))
+
+
+def node_struct(node, ids: IdMap, bolt5: bool = False) -> Structure:
+    fields = [ids.to_int(node.id), list(node.labels), dict(node.properties)]
+    if bolt5:
+        fields.append(node.id)
+    return Structure(NODE_TAG, fields)
+
+
+def rel_struct(edge, ids: IdMap, bolt5: bool = False) -> Structure:
+    fields = [ids.to_int(edge.id), ids.to_int(edge.start_node),
+              ids.to_int(edge.end_node), edge.type, dict(edge.properties)]
+    if bolt5:
+        fields += [edge.id, edge.start_node, edge.end_node]
+    return Structure(REL_TAG, fields)
+
+
+def path_struct(path, ids: IdMap, bolt5: bool = False) -> Structure:
+    nodes = [node_struct(n, ids, bolt5) for n in path.nodes]
+    rels = []
+    for e in path.edges:
+        f = [ids.to_int(e.id), e.type, dict(e.properties)]
+        if bolt5:
+            f.append(e.id)
+        rels.append(Structure(UNBOUND_REL_TAG, f))
+    # sequence: alternating rel index (1-based, negative=reversed), node index
+    seq = []
+    for i, e in enumerate(path.edges):
+        prev = path.nodes[i]
+        sign = 1 if e.start_node == prev.id else -1
+        seq.append(sign * (i + 1))
+        seq.append(i + 1)
+    return Structure(PATH_TAG, [nodes, rels, seq])

@@ -1,0 +1,182 @@
+"""Bolt server tests: packstream round trips and a minimal in-process
+Bolt 4.4 client exercising handshake + HELLO/RUN/PULL/RESET over TCP.
+
+Models reference pkg/bolt tests (packstream_bench_test.go, integration_test.go).
+"""
+
+import asyncio
+import struct
+
+import pytest
+
+from nornicdb_amd.bolt import packstream as ps
+from nornicdb_amd.bolt.server import (BOLT_MAGIC, BoltServer, M_HELLO, M_PULL,
+                                      M_RUN, R_FAILURE, R_RECORD, R_SUCCESS)
+from nornicdb_amd.cypher import Executor
+from nornicdb_amd.storage import MemoryEngine
+
+
+class TestPackStream:
+    @pytest.mark.parametrize("v", [
+        None, True, False, 0, 1, -1, 127, -16, -17, 128, 32767, -32768,
+        2 ** 31, -2 ** 31 - 1, 1.5, -0.25, "", "hello", "x" * 300,
+        [1, [2, 3], "a"], {"k": 1, "nested": {"x": [True, None]}},
+        b"\x00\x01\x02", list(range(20)),
+        {"m": "x" * 70000},
+    ])
+    def test_roundtrip(self, v):
+        assert ps.unpack(ps.pack(v)) == v
+
+    def test_struct_roundtrip(self):
+        s = ps.Structure(0x4E, [1, ["A"], {"k": "v"}])
+        assert ps.unpack(ps.pack(s)) == s
+
+    def test_int_boundaries(self):
+        for v in (-9223372036854775808, 9223372036854775807):
+            assert ps.unpack(ps.pack(v)) == v
+
+
+class _Client:
+    """Minimal Bolt 4.4 test client."""
+
+    def __init__(self, reader, writer):
+        self.reader = reader
+        self.writer = writer
+
+    async def handshake(self):
+        self.writer.write(struct.pack(">I", BOLT_MAGIC))
+        self.writer.write(bytes([0, 0, 4, 4, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0]))
+        await self.writer.drain()
+        resp = await self.reader.readexactly(4)
+        return resp[3], resp[2]  # major, minor
+
+    async def send(self, tag, *fields):
+        data = ps.pack(ps.Structure(tag, list(fields)))
+        self.writer.write(struct.pack(">H", len(data)) + data + b"\x00\x00")
+        await self.writer.drain()
+
+    async def recv(self):
+        buf = bytearray()
+        while True:
+            size = struct.unpack(">H", await self.reader.readexactly(2))[0]
+            if size == 0:
+                if buf:
+                    return ps.unpack(bytes(buf))
+                continue
+            buf += await self.reader.readexactly(size)
+
+
+@pytest.fixture
+def bolt_server_port(unused_tcp_port_factory=None):
+    return 0  # unused; we pick ephemeral below
+
+
+async def _start_server():
+    eng = MemoryEngine()
+    ex = Executor(eng)
+    srv = BoltServer(lambda db: ex, host="127.0.0.1", port=0)
+    await srv.start()
+    port = srv._server.sockets[0].getsockname()[1]
+    return srv, port
+
+
+def test_bolt_end_to_end():
+    async def run():
+        srv, port = await _start_server()
+        try:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            c = _Client(reader, writer)
+            major, minor = await c.handshake()
+            assert (major, minor) == (4, 4)
+
+            await c.send(M_HELLO, {"user_agent": "test/1.0", "scheme": "none"})
+            resp = await c.recv()
+            assert resp.tag == R_SUCCESS
+            assert "NornicDB-AMD" in resp.fields[0]["server"]
+
+            await c.send(M_RUN, "CREATE (n:Person {name: $n}) RETURN n.name",
+                         {"n": "Ada"}, {})
+            resp = await c.recv()
+            assert resp.tag == R_SUCCESS
+            assert resp.fields[0]["fields"] == ["n.name"]
+
+            await c.send(M_PULL, {"n": -1})
+            rec = await c.recv()
+            assert rec.tag == R_RECORD
+            assert rec.fields[0] == ["Ada"]
+            done = await c.recv()
+            assert done.tag == R_SUCCESS
+            assert done.fields[0]["stats"]["nodes-created"] == 1
+
+            # node struct returned for entity values
+            await c.send(M_RUN, "MATCH (n:Person) RETURN n", {}, {})
+            await c.recv()
+            await c.send(M_PULL, {"n": -1})
+            rec = await c.recv()
+            node = rec.fields[0][0]
+            assert isinstance(node, ps.Structure) and node.tag == 0x4E
+            assert node.fields[1] == ["Person"]
+            assert node.fields[2]["name"] == "Ada"
+            await c.recv()
+
+            # syntax error -> FAILURE, then RUN ignored until RESET
+            await c.send(M_RUN, "MATCH (n RETURN", {}, {})
+            fail = await c.recv()
+            assert fail.tag == R_FAILURE
+            assert "SyntaxError" in fail.fields[0]["code"]
+            await c.send(M_RUN, "RETURN 1", {}, {})
+            ign = await c.recv()
+            assert ign.tag == 0x7E  # IGNORED
+            await c.send(0x0F)  # RESET
+            ok = await c.recv()
+            assert ok.tag == R_SUCCESS
+            await c.send(M_RUN, "RETURN 1 AS one", {}, {})
+            assert (await c.recv()).tag == R_SUCCESS
+            await c.send(M_PULL, {"n": -1})
+            assert (await c.recv()).fields[0] == [1]
+            await c.recv()
+
+            writer.close()
+        finally:
+            srv.close()
+
+    asyncio.get_event_loop_policy().new_event_loop().run_until_complete(
+        asyncio.wait_for(run(), timeout=15))
+
+
+def test_bolt_auth_required():
+    class Auth:
+        def login(self, user, pw):
+            if (user, pw) != ("neo4j", "secret"):
+                raise PermissionError("bad credentials")
+
+    async def run():
+        eng = MemoryEngine()
+        ex = Executor(eng)
+        srv = BoltServer(lambda db: ex, host="127.0.0.1", port=0,
+                         authenticator=Auth())
+        await srv.start()
+        port = srv._server.sockets[0].getsockname()[1]
+        try:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            c = _Client(reader, writer)
+            await c.handshake()
+            await c.send(M_HELLO, {"scheme": "basic", "principal": "neo4j",
+                                   "credentials": "wrong"})
+            resp = await c.recv()
+            assert resp.tag == R_FAILURE
+            writer.close()
+
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            c = _Client(reader, writer)
+            await c.handshake()
+            await c.send(M_HELLO, {"scheme": "basic", "principal": "neo4j",
+                                   "credentials": "secret"})
+            resp = await c.recv()
+            assert resp.tag == R_SUCCESS
+            writer.close()
+        finally:
+            srv.close()
+
+    asyncio.get_event_loop_policy().new_event_loop().run_until_complete(
+        asyncio.wait_for(run(), timeout=15))
