@@ -1,0 +1,225 @@
+"""Search layer tests: HNSW recall, BM25 ranking, RRF/MMR, k-means
+routing, embedding index, hybrid service with storage sync.
+
+Models reference pkg/search tests (hnsw_recall_test.go recall floor,
+kmeans_benchmark_test.go, search service tests).
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from nornicdb_amd.search import (ClusterIndex, EmbeddingIndex, FulltextIndex,
+                                 HNSWIndex, SearchService, kmeans,
+                                 mmr_diversify, optimal_k, rrf_fuse)
+from nornicdb_amd.storage import MemoryEngine, Node
+
+
+def unit_rows(n, d, seed=0):
+    rng = np.random.default_rng(seed)
+    x = rng.standard_normal((n, d)).astype(np.float32)
+    return x / np.linalg.norm(x, axis=1, keepdims=True)
+
+
+class TestHNSW:
+    def test_recall_floor(self):
+        """HNSW recall@10 vs exact must clear 0.9 on 2000x64."""
+        x = unit_rows(2000, 64, seed=1)
+        idx = HNSWIndex(64)
+        for i, v in enumerate(x):
+            idx.add(f"v{i}", v)
+        rng = np.random.default_rng(2)
+        hits = total = 0
+        for _ in range(20):
+            q = x[rng.integers(0, 2000)] + 0.05 * rng.standard_normal(64).astype(np.float32)
+            q /= np.linalg.norm(q)
+            exact = np.argsort(-(x @ q))[:10]
+            exact_ids = {f"v{i}" for i in exact}
+            got = {i for i, _ in idx.search(q, 10)}
+            hits += len(got & exact_ids)
+            total += 10
+        assert hits / total >= 0.9, f"recall {hits/total}"
+
+    def test_delete_tombstone(self):
+        x = unit_rows(100, 16)
+        idx = HNSWIndex(16)
+        for i, v in enumerate(x):
+            idx.add(f"v{i}", v)
+        assert idx.remove("v0")
+        assert not idx.remove("v0")
+        res = idx.search(x[0], 5)
+        assert "v0" not in [i for i, _ in res]
+        assert len(idx) == 99
+
+    def test_rebuild_after_many_deletes(self):
+        x = unit_rows(200, 16)
+        idx = HNSWIndex(16)
+        for i, v in enumerate(x):
+            idx.add(f"v{i}", v)
+        for i in range(100):
+            idx.remove(f"v{i}")
+        assert len(idx) == 100
+        res = idx.search(x[150], 5)
+        assert res[0][0] == "v150"
+
+
+class TestBM25:
+    def test_ranking(self):
+        ft = FulltextIndex()
+        ft.index("d1", "the quick brown fox jumps over the lazy dog")
+        ft.index("d2", "quick quick quick brown")
+        ft.index("d3", "completely unrelated text about databases")
+        res = ft.search("quick brown", 3)
+        assert res[0][0] == "d2"
+        assert {r[0] for r in res[:2]} == {"d1", "d2"}
+
+    def test_remove(self):
+        ft = FulltextIndex()
+        ft.index("d1", "hello world")
+        ft.remove("d1")
+        assert ft.search("hello") == []
+
+    def test_idf_favors_rare_terms(self):
+        ft = FulltextIndex()
+        for i in range(10):
+            ft.index(f"c{i}", "common words everywhere common")
+        ft.index("rare", "common words plus zebra")
+        assert ft.search("zebra")[0][0] == "rare"
+
+
+class TestFusion:
+    def test_rrf(self):
+        a = [("x", 1.0), ("y", 0.9), ("z", 0.8)]
+        b = [("y", 5.0), ("x", 4.0)]
+        fused = rrf_fuse([a, b])
+        assert {fused[0][0], fused[1][0]} == {"x", "y"}
+        assert fused[2][0] == "z"
+
+    def test_mmr_diversifies(self):
+        vecs = {
+            "a": np.array([1.0, 0.0], np.float32),
+            "a2": np.array([0.999, 0.04], np.float32),
+            "b": np.array([0.0, 1.0], np.float32),
+        }
+        for k in vecs:
+            vecs[k] = vecs[k] / np.linalg.norm(vecs[k])
+        cands = [("a", 1.0), ("a2", 0.99), ("b", 0.8)]
+        out = mmr_diversify(cands, vecs, 2, lambda_=0.5)
+        assert [o[0] for o in out] == ["a", "b"]
+
+
+class TestKMeans:
+    def test_separated_clusters(self):
+        rng = np.random.default_rng(3)
+        c1 = rng.standard_normal((200, 8)).astype(np.float32) * 0.05 + 5
+        c2 = rng.standard_normal((200, 8)).astype(np.float32) * 0.05 - 5
+        x = torch.from_numpy(np.vstack([c1, c2]))
+        cents, assign = kmeans(x, 2, seed=1)
+        a = assign[:200]
+        b = assign[200:]
+        assert len(set(a.tolist())) == 1
+        assert len(set(b.tolist())) == 1
+        assert a[0] != b[0]
+
+    def test_optimal_k(self):
+        assert optimal_k(100_000) == int((100_000 / 2) ** 0.5)
+
+    def test_cluster_index_routing(self):
+        x = unit_rows(500, 16, seed=5)
+        ids = [f"v{i}" for i in range(500)]
+        ci = ClusterIndex(nprobe=2)
+        ci.cluster(ids, torch.from_numpy(x), k=8)
+        cands = ci.candidates(x[7])
+        assert "v7" in cands
+        assert len(cands) < 500
+
+    def test_incremental_add_and_drift(self):
+        x = unit_rows(100, 8, seed=6)
+        ids = [f"v{i}" for i in range(100)]
+        ci = ClusterIndex()
+        ci.cluster(ids, torch.from_numpy(x), k=4)
+        ci.add("new", x[0])
+        assert "new" in ci.candidates(x[0], nprobe=1)
+
+
+class TestEmbeddingIndex:
+    def test_add_search_remove(self):
+        x = unit_rows(300, 32, seed=7)
+        ei = EmbeddingIndex(32, device="cpu")
+        ei.add_batch([f"v{i}" for i in range(300)], x)
+        res = ei.search(x[5], 3)
+        assert res[0][0] == "v5"
+        assert res[0][1] > 0.99
+        ei.remove("v5")
+        res = ei.search(x[5], 3)
+        assert res[0][0] != "v5"
+        assert len(ei) == 299
+
+    def test_score_subset(self):
+        x = unit_rows(50, 16, seed=8)
+        ei = EmbeddingIndex(16, device="cpu")
+        ei.add_batch([f"v{i}" for i in range(50)], x)
+        res = ei.score_subset(x[3], ["v1", "v3", "v9"])
+        assert res[0][0] == "v3"
+
+    def test_update_existing(self):
+        ei = EmbeddingIndex(4, device="cpu")
+        ei.add("a", [1, 0, 0, 0])
+        ei.add("a", [0, 1, 0, 0])
+        assert len(ei) == 1
+        assert ei.search([0, 1, 0, 0], 1)[0][1] > 0.99
+
+
+class TestSearchService:
+    def _svc(self, dims=16):
+        eng = MemoryEngine()
+        svc = SearchService(eng, dims=dims, device="cpu", use_hnsw=True)
+        return eng, svc
+
+    def test_event_sync_and_hybrid(self):
+        eng, svc = self._svc()
+        x = unit_rows(20, 16, seed=9)
+        for i in range(20):
+            eng.create_node(Node(f"n{i}", ["Doc"],
+                                 {"title": f"doc {i} about topic{i % 3}"},
+                                 embedding=list(map(float, x[i]))))
+        # text only
+        res = svc.text_search("topic1", k=5)
+        assert res and all("topic1" in r.node.properties["title"] for r in res)
+        # vector only
+        res = svc.vector_search(x[4], k=3)
+        assert res[0].id == "n4"
+        # hybrid
+        res = svc.search(query="doc topic2", query_vec=x[2], k=5)
+        assert any(r.id == "n2" for r in res)
+
+    def test_delete_removes_from_indexes(self):
+        eng, svc = self._svc()
+        x = unit_rows(5, 16)
+        for i in range(5):
+            eng.create_node(Node(f"n{i}", ["D"], {"title": "hello"},
+                                 embedding=list(map(float, x[i]))))
+        eng.delete_node("n0")
+        assert all(r.id != "n0" for r in svc.vector_search(x[0], k=5))
+        assert len(svc.fulltext) == 4
+
+    def test_label_filter(self):
+        eng, svc = self._svc()
+        x = unit_rows(4, 16)
+        eng.create_node(Node("a", ["A"], {"title": "same text"},
+                             embedding=list(map(float, x[0]))))
+        eng.create_node(Node("b", ["B"], {"title": "same text"},
+                             embedding=list(map(float, x[1]))))
+        res = svc.text_search("same text", k=5, labels=["B"])
+        assert [r.id for r in res] == ["b"]
+
+    def test_build_indexes_scan(self):
+        eng = MemoryEngine()
+        x = unit_rows(3, 16)
+        for i in range(3):
+            eng.create_node(Node(f"n{i}", [], {"title": "prebuilt"},
+                                 embedding=list(map(float, x[i]))))
+        svc = SearchService(eng, dims=16, device="cpu")
+        assert len(svc.fulltext) == 0
+        svc.build_indexes()
+        assert len(svc.fulltext) == 3
