@@ -1,0 +1,334 @@
+"""Embedded database facade.
+
+Parity: reference pkg/nornicdb/db.go — Open (:750), the agent-memory API
+Store/Recall/Remember/Link/Neighbors/Forget (:1365-1776), memory tiers
+(:157-290), embed queue (embed_queue.go), per-DB search services
+(search_services.go), storage->search event wiring (:994-1035) — and
+pkg/multidb DatabaseManager (manager.go:117).
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Sequence
+
+from .cypher import Executor
+from .cypher.procedures import build_procedures
+from .embed import CachedEmbedder, Embedder, MockEmbedder, chunk_text
+from .search import SearchService
+from .storage import (AsyncEngine, Engine, MemoryEngine, NamespacedEngine,
+                      Node, Edge, NotFoundError, PersistentEngine, new_id)
+
+# memory tiers (reference pkg/decay: episodic 7d / semantic 69d / procedural 693d)
+TIER_EPISODIC = "episodic"
+TIER_SEMANTIC = "semantic"
+TIER_PROCEDURAL = "procedural"
+TIERS = (TIER_EPISODIC, TIER_SEMANTIC, TIER_PROCEDURAL)
+
+
+@dataclass
+class Memory:
+    id: str = ""
+    content: str = ""
+    title: str = ""
+    memory_type: str = TIER_EPISODIC
+    importance: float = 0.5
+    tags: List[str] = field(default_factory=list)
+    metadata: Dict[str, Any] = field(default_factory=dict)
+    created_at: float = 0.0
+    access_count: int = 0
+    last_accessed: float = 0.0
+
+
+class EmbedQueue:
+    """Background embedding workers (reference pkg/nornicdb/embed_queue.go:
+    N pull-based workers over pending_embed; debounced k-means trigger)."""
+
+    def __init__(self, db: "NornicDB", workers: int = 2, batch: int = 16,
+                 poll_interval: float = 0.2, recluster_debounce: float = 30.0):
+        self.db = db
+        self.batch = batch
+        self.poll = poll_interval
+        self.debounce = recluster_debounce
+        self._stop = threading.Event()
+        self._threads = [threading.Thread(target=self._loop, daemon=True)
+                         for _ in range(workers)]
+        self._last_empty = 0.0
+        self._pending_recluster = False
+        self._lock = threading.Lock()
+
+    def start(self):
+        self._started = True
+        for t in self._threads:
+            t.start()
+
+    def stop(self):
+        self._stop.set()
+        if getattr(self, "_started", False):
+            for t in self._threads:
+                t.join(timeout=1)
+
+    def drain(self, timeout: float = 30.0):
+        """Synchronously embed everything pending (for tests/CLI)."""
+        t0 = time.time()
+        while self.db.engine.pending_embeddings(1) and time.time() - t0 < timeout:
+            self._process_batch()
+        return not self.db.engine.pending_embeddings(1)
+
+    def _loop(self):
+        while not self._stop.wait(self.poll):
+            try:
+                worked = self._process_batch()
+                if not worked:
+                    with self._lock:
+                        if (self._pending_recluster
+                                and time.time() - self._last_empty > self.debounce):
+                            self._pending_recluster = False
+                            self.db.search.recluster()
+            except Exception:
+                pass
+
+    def _process_batch(self) -> bool:
+        ids = self.db.engine.pending_embeddings(self.batch)
+        if not ids:
+            return False
+        texts, keep = [], []
+        for nid in ids:
+            try:
+                node = self.db.engine.get_node(nid)
+            except NotFoundError:
+                self.db.engine.clear_pending_embedding(nid)
+                continue
+            text = node.properties.get("content") or node.properties.get("title") or ""
+            chunks = chunk_text(text) or [""]
+            texts.append(chunks[0])  # head chunk embeds the node itself
+            keep.append(nid)
+        if not keep:
+            return True
+        vecs = self.db.embedder.embed_batch(texts)
+        for nid, v in zip(keep, vecs):
+            try:
+                if hasattr(self.db.engine, "update_embedding"):
+                    self.db.engine.update_embedding(nid, [float(x) for x in v])
+                else:
+                    n = self.db.engine.get_node(nid)
+                    n.embedding = [float(x) for x in v]
+                    self.db.engine.update_node(n)
+            except NotFoundError:
+                pass
+            self.db.engine.clear_pending_embedding(nid)
+        with self._lock:
+            self._last_empty = time.time()
+            self._pending_recluster = True
+        return True
+
+
+class NornicDB:
+    """One logical database: engine + executor + search + embed queue."""
+
+    def __init__(self, engine: Engine, name: str = "neo4j",
+                 embedder: Embedder = None, dims: int = None,
+                 device: str = None, auto_embed: bool = True,
+                 embed_workers: int = 0):
+        self.name = name
+        self.engine = engine
+        self.embedder = embedder or MockEmbedder(dims or 64)
+        self.dims = dims or self.embedder.dims
+        self.search = SearchService(engine, dims=self.dims, device=device,
+                                    embedder=self.embedder)
+        self.executor = Executor(engine, procedures=build_procedures(self))
+        self.auto_embed = auto_embed
+        self.embed_queue = EmbedQueue(self, workers=max(embed_workers, 1))
+        if embed_workers > 0:
+            self.embed_queue.start()
+
+    # ---- cypher ----
+    def cypher(self, query: str, params: Dict[str, Any] = None):
+        return self.executor.execute(query, params)
+
+    execute_cypher = cypher
+
+    # ---- memory API (reference db.go:1365-1776) ----
+    def store(self, content: str, title: str = "", memory_type: str = TIER_EPISODIC,
+              importance: float = 0.5, tags: Sequence[str] = (),
+              metadata: Dict[str, Any] = None, embed: bool = None) -> Memory:
+        if memory_type not in TIERS:
+            memory_type = TIER_EPISODIC
+        mid = new_id("m")
+        now = time.time()
+        props = {"content": content, "title": title, "importance": importance,
+                 "memory_type": memory_type, "tags": list(tags),
+                 "access_count": 0, "created_at": now, "last_accessed": now}
+        if metadata:
+            props["metadata"] = dict(metadata)
+        node = Node(id=mid, labels=["Memory", memory_type.capitalize()],
+                    properties=props)
+        self.engine.create_node(node)
+        if embed if embed is not None else self.auto_embed:
+            self.engine.mark_pending_embedding(mid)
+        return Memory(id=mid, content=content, title=title,
+                      memory_type=memory_type, importance=importance,
+                      tags=list(tags), metadata=metadata or {}, created_at=now)
+
+    def recall(self, query: str, limit: int = 10,
+               memory_type: str = None) -> List[Memory]:
+        qv = self.embedder.embed_query(query)
+        labels = [memory_type.capitalize()] if memory_type else ["Memory"]
+        res = self.search.search(query=query, query_vec=qv, k=limit, labels=labels)
+        out = []
+        for r in res:
+            m = self._to_memory(r.node)
+            self._touch(r.node.id)
+            out.append(m)
+        return out
+
+    def remember(self, memory_id: str) -> Memory:
+        node = self.engine.get_node(memory_id)
+        self._touch(memory_id)
+        return self._to_memory(node)
+
+    def link(self, from_id: str, to_id: str, rel_type: str = "RELATES_TO",
+             confidence: float = 1.0) -> Edge:
+        e = Edge(id=new_id("e"), type=rel_type, start_node=from_id,
+                 end_node=to_id, properties={"confidence": confidence})
+        return self.engine.create_edge(e)
+
+    def neighbors(self, memory_id: str, depth: int = 1) -> List[Memory]:
+        seen = {memory_id}
+        frontier = [memory_id]
+        out = []
+        for _ in range(depth):
+            nxt = []
+            for nid in frontier:
+                for nb in self.engine.neighbors(nid):
+                    if nb not in seen:
+                        seen.add(nb)
+                        nxt.append(nb)
+                        try:
+                            out.append(self._to_memory(self.engine.get_node(nb)))
+                        except NotFoundError:
+                            pass
+            frontier = nxt
+        return out
+
+    def forget(self, memory_id: str) -> bool:
+        try:
+            self.engine.detach_delete_node(memory_id)
+            return True
+        except NotFoundError:
+            return False
+
+    def _touch(self, node_id: str):
+        try:
+            n = self.engine.get_node(node_id)
+            n.properties["access_count"] = n.properties.get("access_count", 0) + 1
+            n.properties["last_accessed"] = time.time()
+            self.engine.update_node(n)
+        except NotFoundError:
+            pass
+
+    @staticmethod
+    def _to_memory(node: Node) -> Memory:
+        p = node.properties
+        return Memory(id=node.id, content=p.get("content", ""),
+                      title=p.get("title", ""),
+                      memory_type=p.get("memory_type", TIER_EPISODIC),
+                      importance=p.get("importance", 0.5),
+                      tags=list(p.get("tags", [])),
+                      metadata=dict(p.get("metadata", {})),
+                      created_at=p.get("created_at", 0.0),
+                      access_count=p.get("access_count", 0),
+                      last_accessed=p.get("last_accessed", 0.0))
+
+    def close(self):
+        self.embed_queue.stop()
+        self.engine.flush()
+
+
+class DatabaseManager:
+    """Multi-database manager over one shared base engine.
+
+    Parity: reference pkg/multidb/manager.go:117 (CREATE/DROP DATABASE,
+    aliases, metadata) using NamespacedEngine isolation.
+    """
+
+    SYSTEM = "system"
+    DEFAULT = "neo4j"
+
+    def __init__(self, base_engine: Engine, embedder: Embedder = None,
+                 dims: int = None, device: str = None):
+        self._base = base_engine
+        self._embedder = embedder
+        self._dims = dims
+        self._device = device
+        self._lock = threading.Lock()
+        self._dbs: Dict[str, NornicDB] = {}
+        self._aliases: Dict[str, str] = {}
+        for name in (self.DEFAULT, self.SYSTEM):
+            self._open(name)
+
+    def _open(self, name: str) -> NornicDB:
+        eng = NamespacedEngine(self._base, name)
+        db = NornicDB(eng, name=name, embedder=self._embedder,
+                      dims=self._dims, device=self._device)
+        self._dbs[name] = db
+        return db
+
+    def get(self, name: str = None) -> NornicDB:
+        name = self._aliases.get(name or self.DEFAULT, name or self.DEFAULT)
+        with self._lock:
+            db = self._dbs.get(name)
+            if db is None:
+                raise KeyError(f"database {name} does not exist")
+            return db
+
+    def create(self, name: str) -> NornicDB:
+        with self._lock:
+            if name in self._dbs:
+                raise ValueError(f"database {name} already exists")
+            return self._open(name)
+
+    def drop(self, name: str):
+        if name in (self.SYSTEM, self.DEFAULT):
+            raise ValueError(f"cannot drop {name}")
+        with self._lock:
+            db = self._dbs.pop(name, None)
+            if db is None:
+                raise KeyError(f"database {name} does not exist")
+            for node in list(db.engine.all_nodes()):
+                try:
+                    db.engine.detach_delete_node(node.id)
+                except NotFoundError:
+                    pass
+            db.close()
+
+    def alias(self, alias: str, target: str):
+        with self._lock:
+            self._aliases[alias] = target
+
+    def list(self) -> List[str]:
+        with self._lock:
+            return sorted(self._dbs)
+
+    def close(self):
+        for db in self._dbs.values():
+            db.close()
+        self._base.close()
+
+
+def open_db(data_dir: Optional[str] = None, embedder: Embedder = None,
+            dims: int = None, device: str = None, durable_sync: bool = False,
+            **kw) -> DatabaseManager:
+    """Open a NornicDB instance (reference nornicdb.Open, db.go:750).
+
+    data_dir=None -> in-memory. Engine stack: Persistent(WAL+snapshots)
+    [-> Async] -> Namespaced per database.
+    """
+    if data_dir:
+        base = PersistentEngine(data_dir, sync_on_write=durable_sync)
+    else:
+        base = MemoryEngine()
+    return DatabaseManager(base, embedder=embedder, dims=dims, device=device)
