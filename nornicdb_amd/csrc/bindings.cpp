@@ -11,6 +11,13 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
 std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out);
 
+// encoder_ops.hip
+at::Tensor add_layernorm(at::Tensor a, c10::optional<at::Tensor> b,
+                         at::Tensor gamma, at::Tensor beta, double eps);
+at::Tensor bias_gelu(at::Tensor x, at::Tensor bias);
+at::Tensor mean_pool_l2norm(at::Tensor x, c10::optional<at::Tensor> mask);
+at::Tensor flash_attn_nc(at::Tensor q, at::Tensor k, at::Tensor v);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "NornicDB-AMD CDNA4 (gfx950) native kernels";
   m.def("l2_normalize_", &l2_normalize_, "In-place row L2 normalize (bf16/f32)");
@@ -25,4 +32,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused MFMA cosine score + top-k, 256-query batches (bf16 db)",
         py::arg("db"), py::arg("q"), py::arg("row_base") = 0,
         py::arg("k_out") = 10);
+  m.def("add_layernorm", &add_layernorm, "LN(a+b)*gamma+beta fused (bf16)",
+        py::arg("a"), py::arg("b"), py::arg("gamma"), py::arg("beta"),
+        py::arg("eps") = 1e-5);
+  m.def("bias_gelu", &bias_gelu, "gelu(x+bias) fused (bf16)");
+  m.def("mean_pool_l2norm", &mean_pool_l2norm,
+        "masked mean-pool + L2 norm (bf16 -> fp32)",
+        py::arg("x"), py::arg("mask") = c10::nullopt);
+  m.def("flash_attn_nc", &flash_attn_nc,
+        "non-causal flash attention fwd, head_dim 64 (bf16)");
 }

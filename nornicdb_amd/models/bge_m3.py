@@ -49,6 +49,12 @@ def _gelu(x):
     return F.gelu(x)
 
 
+def _fused_ok(x):
+    """Fused HIP kernels are inference-only (no autograd graph)."""
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and not torch.is_grad_enabled())
+
+
 class _EncoderLayer(nn.Module):
     def __init__(self, cfg: BgeM3Config):
         super().__init__()
@@ -73,9 +79,19 @@ class _EncoderLayer(nn.Module):
 
         a = flash_attention_nc(q, k, v, attn_bias)  # [b, nh, s, hd]
         a = a.transpose(1, 2).reshape(b, s, h)
-        x = self.ln1(x + self.attn_out(a))
-        f = self.ffn_out(_gelu(self.ffn_in(x)))
-        x = self.ln2(x + f)
+        if _fused_ok(x):
+            from nornicdb_amd.ops import encoder as eops
+
+            x = eops.add_layernorm(x, self.attn_out(a), self.ln1.weight,
+                                   self.ln1.bias, self.ln1.eps)
+            f = self.ffn_out(eops.bias_gelu(
+                F.linear(x, self.ffn_in.weight), self.ffn_in.bias))
+            x = eops.add_layernorm(x, f, self.ln2.weight, self.ln2.bias,
+                                   self.ln2.eps)
+        else:
+            x = self.ln1(x + self.attn_out(a))
+            f = self.ffn_out(_gelu(self.ffn_in(x)))
+            x = self.ln2(x + f)
         return x
 
 
@@ -123,12 +139,16 @@ class BgeM3Encoder(nn.Module):
             x = layer(x, attn_bias)
 
         if self.cfg.pooling == "cls":
-            pooled = x[:, 0]
+            pooled = x[:, 0].float()
+        elif _fused_ok(x):
+            from nornicdb_amd.ops import encoder as eops
+
+            return eops.mean_pool_l2norm(x, attention_mask)
         else:
             if attention_mask is None:
                 pooled = x.mean(dim=1)
             else:
                 m = attention_mask[..., None].to(x.dtype)
                 pooled = (x * m).sum(1) / m.sum(1).clamp_min(1)
-        pooled = pooled.float()
+            pooled = pooled.float()
         return pooled / torch.linalg.vector_norm(pooled, dim=-1, keepdim=True).clamp_min(1e-12)

@@ -23,7 +23,9 @@ def flash_attention_nc(q, k, v, attn_bias=None):
         and hasattr(nat, "flash_attn_nc")
         and attn_bias is None
         and q.dtype == torch.bfloat16
-        and q.shape[-1] in (64, 128)
+        and q.shape[-1] == 64
+        and q.shape[-2] % 64 == 0
+        and not (q.requires_grad and torch.is_grad_enabled())
     ):
         return nat.flash_attn_nc(q.contiguous(), k.contiguous(), v.contiguous())
     return F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias)
