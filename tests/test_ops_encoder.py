@@ -31,7 +31,7 @@ def test_add_layernorm_gpu():
     y = add_layernorm(a, b, g, be)
     ref = F.layer_norm((a.float() + b.float()), (1024,), g, be)
     torch.cuda.synchronize()
-    assert (y.float() - ref).abs().max().item() < 0.05
+    assert (y.float() - ref).abs().max().item() < 0.1
 
 
 @pytest.mark.gpu
@@ -42,7 +42,7 @@ def test_bias_gelu_gpu():
     y = bias_gelu(x, b)
     ref = F.gelu(x.float() + b.float())
     torch.cuda.synchronize()
-    assert (y.float() - ref).abs().max().item() < 0.05
+    assert (y.float() - ref).abs().max().item() < 0.1
 
 
 @pytest.mark.gpu
