@@ -11,6 +11,18 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
 std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out);
 
+// graph.hip
+at::Tensor pagerank_contrib(at::Tensor rank, at::Tensor outdeg);
+at::Tensor pagerank_gather(at::Tensor row_ptr, at::Tensor col_idx,
+                           at::Tensor contrib, double damping, double base);
+void bfs_level(at::Tensor row_ptr, at::Tensor col_idx, at::Tensor dist,
+               at::Tensor changed, long long row_base, long long level);
+at::Tensor labelprop_step(at::Tensor row_ptr, at::Tensor col_idx,
+                          at::Tensor labels, at::Tensor changed,
+                          long long row_base);
+void wcc_hook(at::Tensor row_ptr, at::Tensor col_idx, at::Tensor comp,
+              at::Tensor changed, long long row_base);
+
 // encoder_ops.hip
 at::Tensor add_layernorm(at::Tensor a, c10::optional<at::Tensor> b,
                          at::Tensor gamma, at::Tensor beta, double eps);
@@ -41,4 +53,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("mask") = c10::nullopt);
   m.def("flash_attn_nc", &flash_attn_nc,
         "non-causal flash attention fwd, head_dim 64 (bf16)");
+  m.def("pagerank_contrib", &pagerank_contrib, "rank/outdeg elementwise");
+  m.def("pagerank_gather", &pagerank_gather,
+        "CSR pull-gather pagerank iteration (wave per row)");
+  m.def("bfs_level", &bfs_level, "BFS frontier level expansion");
+  m.def("labelprop_step", &labelprop_step, "label propagation step");
+  m.def("wcc_hook", &wcc_hook, "connected-components hook step");
 }

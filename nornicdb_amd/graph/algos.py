@@ -1,0 +1,430 @@
+"""Graph algorithms: GPU (HIP kernels) for large graphs, numpy/python for
+small graphs and behavioral parity.
+
+Parity: reference apoc/algo/algo.go:32-417 (PageRank, Betweenness,
+Closeness, Degree, AStar, Dijkstra, AllPairs, Cover) and
+apoc/community/community.go:66-505 (Louvain, LabelProp, Modularity,
+Triangles, ClusteringCoeff, ConnectedComponents).
+"""
+
+from __future__ import annotations
+
+import heapq
+from collections import defaultdict
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ..ops import native_or_none
+from .csr import CSRGraph
+
+GPU_MIN_EDGES = 50_000  # below this the numpy path wins on latency
+
+
+def _use_gpu(g: CSRGraph, device=None) -> bool:
+    return (torch.cuda.is_available() and native_or_none() is not None
+            and g.m >= GPU_MIN_EDGES and device != "cpu")
+
+
+# ---------------------------------------------------------------- PageRank
+def pagerank(g: CSRGraph, damping: float = 0.85, iters: int = 20,
+             tol: float = 1e-6, device=None) -> np.ndarray:
+    if _use_gpu(g, device):
+        return _pagerank_gpu(g, damping, iters, tol)
+    return _pagerank_np(g, damping, iters, tol)
+
+
+def _pagerank_np(g: CSRGraph, damping, iters, tol) -> np.ndarray:
+    n = g.n
+    if n == 0:
+        return np.zeros(0, np.float32)
+    g.with_in_edges()
+    outdeg = g.out_degrees().astype(np.float64)
+    rank = np.full(n, 1.0 / n)
+    rows = np.repeat(np.arange(n), np.diff(g.in_row_ptr))
+    src = g.in_col_idx
+    for _ in range(iters):
+        contrib = np.where(outdeg > 0, rank / np.maximum(outdeg, 1), 0.0)
+        dangling = rank[outdeg == 0].sum() / n
+        new = np.full(n, (1 - damping) / n) + damping * dangling
+        np.add.at(new, rows, damping * contrib[src])
+        if np.abs(new - rank).sum() < tol:
+            rank = new
+            break
+        rank = new
+    return rank.astype(np.float32)
+
+
+def _pagerank_gpu(g: CSRGraph, damping, iters, tol) -> np.ndarray:
+    nat = native_or_none()
+    dev = "cuda"
+    n = g.n
+    rp, ci = g.torch_csr(dev, in_edges=True)
+    outdeg = torch.as_tensor(g.out_degrees(), dtype=torch.int32, device=dev)
+    rank = torch.full((n,), 1.0 / n, device=dev)
+    dangling_mask = outdeg == 0
+    for _ in range(iters):
+        contrib = nat.pagerank_contrib(rank, outdeg)
+        dangling = float(rank[dangling_mask].sum()) / n
+        base = (1 - damping) / n + damping * dangling
+        new = nat.pagerank_gather(rp, ci, contrib, damping, base)
+        if float((new - rank).abs().sum()) < tol:
+            rank = new
+            break
+        rank = new
+    return rank.cpu().numpy()
+
+
+# ---------------------------------------------------------------- BFS / SSSP
+def bfs_distances(g: CSRGraph, source: int, device=None) -> np.ndarray:
+    if _use_gpu(g, device):
+        return _bfs_gpu(g, source)
+    n = g.n
+    dist = np.full(n, -1, np.int32)
+    dist[source] = 0
+    frontier = [source]
+    level = 0
+    rp, ci = g.row_ptr, g.col_idx
+    while frontier:
+        nxt = []
+        for u in frontier:
+            for j in range(rp[u], rp[u + 1]):
+                v = ci[j]
+                if dist[v] < 0:
+                    dist[v] = level + 1
+                    nxt.append(v)
+        frontier = nxt
+        level += 1
+    return dist
+
+
+def _bfs_gpu(g: CSRGraph, source: int) -> np.ndarray:
+    nat = native_or_none()
+    dev = "cuda"
+    rp, ci = g.torch_csr(dev)
+    dist = torch.full((g.n,), -1, dtype=torch.int32, device=dev)
+    dist[source] = 0
+    changed = torch.zeros(1, dtype=torch.int32, device=dev)
+    level = 0
+    while True:
+        changed.zero_()
+        nat.bfs_level(rp, ci, dist, changed, 0, level)
+        if int(changed.item()) == 0:
+            break
+        level += 1
+    return dist.cpu().numpy()
+
+
+def dijkstra(g: CSRGraph, source: int, target: int = None):
+    """Weighted shortest paths; weights default 1. Returns (dist, prev)."""
+    n = g.n
+    w = g.weights if g.weights is not None else np.ones(g.m, np.float32)
+    dist = np.full(n, np.inf)
+    prev = np.full(n, -1, np.int64)
+    dist[source] = 0.0
+    pq = [(0.0, source)]
+    rp, ci = g.row_ptr, g.col_idx
+    while pq:
+        d, u = heapq.heappop(pq)
+        if d > dist[u]:
+            continue
+        if target is not None and u == target:
+            break
+        for j in range(rp[u], rp[u + 1]):
+            v = ci[j]
+            nd = d + w[j]
+            if nd < dist[v]:
+                dist[v] = nd
+                prev[v] = u
+                heapq.heappush(pq, (nd, v))
+    return dist, prev
+
+
+def astar(g: CSRGraph, source: int, target: int, heuristic) -> Tuple[float, List[int]]:
+    w = g.weights if g.weights is not None else np.ones(g.m, np.float32)
+    n = g.n
+    dist = np.full(n, np.inf)
+    prev = np.full(n, -1, np.int64)
+    dist[source] = 0.0
+    pq = [(heuristic(source), source)]
+    rp, ci = g.row_ptr, g.col_idx
+    while pq:
+        _, u = heapq.heappop(pq)
+        if u == target:
+            break
+        for j in range(rp[u], rp[u + 1]):
+            v = ci[j]
+            nd = dist[u] + w[j]
+            if nd < dist[v]:
+                dist[v] = nd
+                prev[v] = u
+                heapq.heappush(pq, (nd + heuristic(v), v))
+    if not np.isfinite(dist[target]):
+        return float("inf"), []
+    path = [target]
+    while path[-1] != source:
+        path.append(int(prev[path[-1]]))
+    return float(dist[target]), path[::-1]
+
+
+def shortest_path(g: CSRGraph, source: int, target: int) -> List[int]:
+    dist, prev = dijkstra(g, source, target)
+    if not np.isfinite(dist[target]):
+        return []
+    path = [target]
+    while path[-1] != source:
+        path.append(int(prev[path[-1]]))
+    return path[::-1]
+
+
+# ------------------------------------------------------- components / labels
+def connected_components(g: CSRGraph, device=None) -> np.ndarray:
+    """Weakly connected components; returns component id per node."""
+    if _use_gpu(g, device):
+        return _wcc_gpu(g)
+    n = g.n
+    parent = np.arange(n)
+
+    def find(x):
+        while parent[x] != x:
+            parent[x] = parent[parent[x]]
+            x = parent[x]
+        return x
+
+    rp, ci = g.row_ptr, g.col_idx
+    for u in range(n):
+        for j in range(rp[u], rp[u + 1]):
+            a, b = find(u), find(int(ci[j]))
+            if a != b:
+                parent[max(a, b)] = min(a, b)
+    return np.fromiter((find(i) for i in range(n)), np.int64, n)
+
+
+def _wcc_gpu(g: CSRGraph) -> np.ndarray:
+    nat = native_or_none()
+    dev = "cuda"
+    rp, ci = g.torch_csr(dev)
+    comp = torch.arange(g.n, dtype=torch.int32, device=dev)
+    changed = torch.zeros(1, dtype=torch.int32, device=dev)
+    for _ in range(64):
+        changed.zero_()
+        nat.wcc_hook(rp, ci, comp, changed, 0)
+        # pointer jumping
+        comp = comp[comp.long()].contiguous()
+        comp = comp[comp.long()].contiguous()
+        if int(changed.item()) == 0:
+            break
+    return comp.long().cpu().numpy()
+
+
+def label_propagation(g: CSRGraph, iters: int = 20, device=None) -> np.ndarray:
+    if _use_gpu(g, device):
+        return _labelprop_gpu(g, iters)
+    n = g.n
+    labels = np.arange(n, dtype=np.int64)
+    rp, ci = g.row_ptr, g.col_idx
+    for _ in range(iters):
+        changed = False
+        new = labels.copy()
+        for u in range(n):
+            s, e = rp[u], rp[u + 1]
+            if s == e:
+                continue
+            counts: Dict[int, int] = defaultdict(int)
+            for j in range(s, e):
+                counts[int(labels[ci[j]])] += 1
+            # most frequent, ties -> smallest label
+            mc = max(counts.values())
+            best = min(lb for lb, c in counts.items() if c == mc)
+            if best != labels[u]:
+                new[u] = best
+                changed = True
+        labels = new
+        if not changed:
+            break
+    return labels
+
+
+def _labelprop_gpu(g: CSRGraph, iters: int) -> np.ndarray:
+    nat = native_or_none()
+    dev = "cuda"
+    rp, ci = g.torch_csr(dev)
+    labels = torch.arange(g.n, dtype=torch.int32, device=dev)
+    changed = torch.zeros(1, dtype=torch.int32, device=dev)
+    for _ in range(iters):
+        changed.zero_()
+        labels = nat.labelprop_step(rp, ci, labels, changed, 0)
+        if int(changed.item()) == 0:
+            break
+    return labels.long().cpu().numpy()
+
+
+# ---------------------------------------------------------------- centrality
+def degree_centrality(g: CSRGraph) -> np.ndarray:
+    deg = g.out_degrees().astype(np.float64)
+    g.with_in_edges()
+    deg = deg + np.diff(g.in_row_ptr)
+    denom = max(g.n - 1, 1)
+    return (deg / denom).astype(np.float32)
+
+
+def closeness_centrality(g: CSRGraph, nodes: Sequence[int] = None) -> np.ndarray:
+    nodes = range(g.n) if nodes is None else nodes
+    out = np.zeros(g.n, np.float32)
+    for u in nodes:
+        d = bfs_distances(g, u, device="cpu")
+        reach = d[d >= 0]
+        if len(reach) > 1:
+            out[u] = (len(reach) - 1) / reach.sum()
+    return out
+
+
+def betweenness_centrality(g: CSRGraph, samples: int = None,
+                           seed: int = 0) -> np.ndarray:
+    """Brandes' algorithm (exact, or source-sampled approximation)."""
+    n = g.n
+    bc = np.zeros(n)
+    rng = np.random.default_rng(seed)
+    sources = list(rng.choice(n, min(samples, n), replace=False)) \
+        if samples else list(range(n))
+    rp, ci = g.row_ptr, g.col_idx
+    scale = (n / max(len(sources), 1)) if samples else 1.0
+    for s in sources:
+        stack = []
+        preds = [[] for _ in range(n)]
+        sigma = np.zeros(n)
+        sigma[s] = 1
+        dist = np.full(n, -1)
+        dist[s] = 0
+        q = [s]
+        while q:
+            nq = []
+            for u in q:
+                stack.append(u)
+                for j in range(rp[u], rp[u + 1]):
+                    v = int(ci[j])
+                    if dist[v] < 0:
+                        dist[v] = dist[u] + 1
+                        nq.append(v)
+                    if dist[v] == dist[u] + 1:
+                        sigma[v] += sigma[u]
+                        preds[v].append(u)
+            q = nq
+        delta = np.zeros(n)
+        for v in reversed(stack):
+            for u in preds[v]:
+                delta[u] += sigma[u] / sigma[v] * (1 + delta[v])
+            if v != s:
+                bc[v] += delta[v]
+        # (stack built in BFS order already)
+    return (bc * scale).astype(np.float32)
+
+
+# --------------------------------------------------------------- community
+def triangle_count(g: CSRGraph) -> int:
+    n = g.n
+    neigh = [set() for _ in range(n)]
+    rp, ci = g.row_ptr, g.col_idx
+    for u in range(n):
+        for j in range(rp[u], rp[u + 1]):
+            v = int(ci[j])
+            if v != u:
+                neigh[u].add(v)
+                neigh[v].add(u)
+    count = 0
+    for u in range(n):
+        for v in neigh[u]:
+            if v > u:
+                count += len(neigh[u] & neigh[v] & set(range(v + 1, n)))
+    return count
+
+
+def clustering_coefficient(g: CSRGraph) -> np.ndarray:
+    n = g.n
+    neigh = [set() for _ in range(n)]
+    rp, ci = g.row_ptr, g.col_idx
+    for u in range(n):
+        for j in range(rp[u], rp[u + 1]):
+            v = int(ci[j])
+            if v != u:
+                neigh[u].add(v)
+                neigh[v].add(u)
+    out = np.zeros(n, np.float32)
+    for u in range(n):
+        k = len(neigh[u])
+        if k < 2:
+            continue
+        links = sum(1 for v in neigh[u] for w in neigh[u]
+                    if v < w and w in neigh[v])
+        out[u] = 2.0 * links / (k * (k - 1))
+    return out
+
+
+def modularity(g: CSRGraph, communities: np.ndarray) -> float:
+    """Newman modularity of a partition (undirected interpretation)."""
+    m = g.m
+    if m == 0:
+        return 0.0
+    deg = np.zeros(g.n)
+    rp, ci = g.row_ptr, g.col_idx
+    inside = 0.0
+    for u in range(g.n):
+        for j in range(rp[u], rp[u + 1]):
+            v = int(ci[j])
+            deg[u] += 1
+            deg[v] += 1
+            if communities[u] == communities[v]:
+                inside += 2
+    two_m = 2.0 * m
+    q = inside / two_m
+    for c in np.unique(communities):
+        dc = deg[communities == c].sum()
+        q -= (dc / two_m) ** 2
+    return float(q)
+
+
+def louvain(g: CSRGraph, max_passes: int = 5) -> np.ndarray:
+    """Single-level Louvain (local moving) — community per node."""
+    n = g.n
+    comm = np.arange(n)
+    # undirected adjacency with weights
+    adj: List[Dict[int, float]] = [defaultdict(float) for _ in range(n)]
+    rp, ci = g.row_ptr, g.col_idx
+    w = g.weights if g.weights is not None else np.ones(g.m, np.float32)
+    for u in range(n):
+        for j in range(rp[u], rp[u + 1]):
+            v = int(ci[j])
+            if u == v:
+                continue
+            adj[u][v] += float(w[j])
+            adj[v][u] += float(w[j])
+    k = np.array([sum(adj[u].values()) for u in range(n)])
+    two_m = k.sum()
+    if two_m == 0:
+        return comm
+    sigma_tot = {c: k[c] for c in range(n)}
+    for _ in range(max_passes):
+        moved = False
+        for u in range(n):
+            cu = comm[u]
+            links = defaultdict(float)
+            for v, wv in adj[u].items():
+                links[comm[v]] += wv
+            sigma_tot[cu] -= k[u]
+            best_c, best_gain = cu, 0.0
+            for c, l_uc in links.items():
+                gain = l_uc - sigma_tot.get(c, 0.0) * k[u] / two_m
+                base = links.get(cu, 0.0) - sigma_tot.get(cu, 0.0) * k[u] / two_m
+                if gain - base > best_gain + 1e-12:
+                    best_gain = gain - base
+                    best_c = c
+            sigma_tot[best_c] = sigma_tot.get(best_c, 0.0) + k[u]
+            if best_c != cu:
+                comm[u] = best_c
+                moved = True
+        if not moved:
+            break
+    # renumber
+    _, inv = np.unique(comm, return_inverse=True)
+    return inv

@@ -1,0 +1,210 @@
+"""Graph analytics tests: algorithm correctness on small graphs (CPU),
+GPU kernel parity, and 2-process gloo sharded PageRank/BFS."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from nornicdb_amd.graph import (betweenness_centrality, bfs_distances,
+                                clustering_coefficient, connected_components,
+                                degree_centrality, dijkstra, from_edges,
+                                label_propagation, louvain, modularity,
+                                pagerank, random_graph, shortest_path,
+                                triangle_count)
+
+
+def line_graph(n):
+    return from_edges(n, [(i, i + 1) for i in range(n - 1)])
+
+
+def two_cliques():
+    """Two 4-cliques joined by one edge."""
+    edges = []
+    for base in (0, 4):
+        for i in range(4):
+            for j in range(i + 1, 4):
+                edges.append((base + i, base + j))
+    edges.append((0, 4))
+    return from_edges(8, edges, undirected=True)
+
+
+class TestPageRank:
+    def test_star_graph(self):
+        # all point at node 0 -> node 0 has the highest rank
+        g = from_edges(5, [(i, 0) for i in range(1, 5)])
+        r = pagerank(g, iters=50)
+        assert r.argmax() == 0
+        assert abs(r.sum() - 1.0) < 1e-3
+
+    def test_cycle_uniform(self):
+        g = from_edges(4, [(0, 1), (1, 2), (2, 3), (3, 0)])
+        r = pagerank(g, iters=100)
+        assert np.allclose(r, 0.25, atol=1e-3)
+
+    def test_dangling_nodes(self):
+        g = from_edges(3, [(0, 1), (1, 2)])  # 2 dangles
+        r = pagerank(g, iters=100)
+        assert abs(r.sum() - 1.0) < 1e-3
+        assert r[2] > r[1] > r[0]
+
+
+class TestTraversal:
+    def test_bfs(self):
+        g = line_graph(6)
+        d = bfs_distances(g, 0)
+        assert d.tolist() == [0, 1, 2, 3, 4, 5]
+
+    def test_bfs_unreachable(self):
+        g = from_edges(4, [(0, 1)])
+        d = bfs_distances(g, 0)
+        assert d[1] == 1 and d[2] == -1 and d[3] == -1
+
+    def test_dijkstra_weighted(self):
+        g = from_edges(4, [(0, 1), (1, 3), (0, 2), (2, 3)],
+                       weights=[1, 5, 2, 1])
+        dist, _ = dijkstra(g, 0)
+        assert dist[3] == 3.0
+        assert shortest_path(g, 0, 3) == [0, 2, 3]
+
+
+class TestComponents:
+    def test_wcc(self):
+        g = from_edges(6, [(0, 1), (1, 2), (3, 4)])
+        c = connected_components(g)
+        assert c[0] == c[1] == c[2]
+        assert c[3] == c[4]
+        assert c[0] != c[3] and c[5] not in (c[0], c[3])
+
+    def test_labelprop_two_cliques(self):
+        g = two_cliques()
+        labels = label_propagation(g, iters=30)
+        assert len(set(labels[:4])) == 1
+        assert len(set(labels[4:])) == 1
+
+    def test_louvain_and_modularity(self):
+        g = two_cliques()
+        comm = louvain(g)
+        assert comm[0] == comm[1] == comm[2] == comm[3]
+        assert comm[4] == comm[5] == comm[6] == comm[7]
+        q = modularity(g, comm)
+        q_bad = modularity(g, np.arange(8) % 2)  # interleaved partition
+        assert q > q_bad
+
+
+class TestCentrality:
+    def test_degree(self):
+        g = from_edges(3, [(0, 1), (0, 2)])
+        d = degree_centrality(g)
+        assert d.argmax() == 0
+
+    def test_betweenness_line(self):
+        g = from_edges(3, [(0, 1), (1, 0), (1, 2), (2, 1)])
+        bc = betweenness_centrality(g)
+        assert bc.argmax() == 1
+
+    def test_triangles_and_clustering(self):
+        g = from_edges(3, [(0, 1), (1, 2), (0, 2)])
+        assert triangle_count(g) == 1
+        cc = clustering_coefficient(g)
+        assert np.allclose(cc, 1.0)
+
+
+@pytest.mark.gpu
+class TestGPUGraphKernels:
+    def test_pagerank_gpu_matches_cpu(self):
+        g = random_graph(20000, 8, seed=1)
+        r_cpu = pagerank(g, iters=15, device="cpu")
+        r_gpu = pagerank(g, iters=15)
+        assert np.abs(r_cpu - r_gpu).max() < 1e-5
+
+    def test_bfs_gpu_matches_cpu(self):
+        g = random_graph(20000, 8, seed=2)
+        d_cpu = bfs_distances(g, 0, device="cpu")
+        d_gpu = bfs_distances(g, 0)
+        assert (d_cpu == d_gpu).all()
+
+    def test_wcc_gpu(self):
+        g = random_graph(20000, 4, seed=3)
+        c_cpu = connected_components(g, device="cpu")
+        c_gpu = connected_components(g)
+        # same partition (compare via canonical relabeling)
+        import numpy as np
+        _, a = np.unique(c_cpu, return_inverse=True)
+        _, b = np.unique(c_gpu, return_inverse=True)
+        # components equal iff grouping identical
+        m = {}
+        ok = True
+        for x, y in zip(a, b):
+            if x in m and m[x] != y:
+                ok = False
+                break
+            m[x] = y
+        assert ok and len(set(a)) == len(set(b))
+
+    def test_labelprop_gpu_runs(self):
+        g = random_graph(20000, 8, seed=4)
+        lb = label_propagation(g)
+        assert lb.shape == (20000,)
+
+
+# ---------------- distributed (gloo, 2 processes) ----------------
+def _dist_worker(rank, world, port, fn_name, result_q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from nornicdb_amd.graph import from_edges, pagerank
+        from nornicdb_amd.parallel import bfs_sharded, pagerank_sharded, shard_rows
+
+        # deterministic graph on all ranks
+        rng = np.random.default_rng(7)
+        n = 200
+        edges = [(int(a), int(b)) for a, b in rng.integers(0, n, (1200, 2))]
+        g = from_edges(n, edges)
+        g.with_in_edges()
+        lo, hi = shard_rows(n, rank, world)
+
+        # local in-edge CSR rows
+        rp = torch.as_tensor(g.in_row_ptr[lo:hi + 1] - g.in_row_ptr[lo])
+        ci = torch.as_tensor(g.in_col_idx[g.in_row_ptr[lo]:g.in_row_ptr[hi]],
+                             dtype=torch.int32)
+        outdeg = torch.as_tensor(g.out_degrees(), dtype=torch.int32)
+        r = pagerank_sharded(rp, ci, outdeg, n, lo, iters=30)
+
+        # BFS on out-edges
+        rp2 = torch.as_tensor(g.row_ptr[lo:hi + 1] - g.row_ptr[lo])
+        ci2 = torch.as_tensor(g.col_idx[g.row_ptr[lo]:g.row_ptr[hi]],
+                              dtype=torch.int32)
+        d = bfs_sharded(rp2, ci2, n, lo, source=0)
+
+        if rank == 0:
+            ref_r = pagerank(g, iters=30, device="cpu")
+            ref_d = bfs_distances(g, 0, device="cpu")
+            result_q.put((
+                float(np.abs(r.numpy() - ref_r).max()),
+                bool((d.numpy() == ref_d).all()),
+            ))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_pagerank_bfs_gloo_2proc():
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29731
+    procs = [ctx.Process(target=_dist_worker, args=(r, 2, port, "pr", q))
+             for r in range(2)]
+    [p.start() for p in procs]
+    try:
+        err, bfs_ok = q.get(timeout=120)
+    finally:
+        [p.join(timeout=30) for p in procs]
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+    assert err < 1e-4, f"sharded pagerank diverges from single-proc: {err}"
+    assert bfs_ok
