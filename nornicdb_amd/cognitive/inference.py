@@ -1,0 +1,150 @@
+"""Automatic relationship inference ("cognitive" auto-linking).
+
+Parity: reference pkg/inference/inference.go — Engine.OnStore derives
+suggested edges from (a) embedding similarity, (b) co-access, (c) temporal
+proximity, (d) transitive closure; with per-pair cooldown (cooldown.go),
+evidence accumulation (evidence.go) and edge decay (edge_decay.go).
+"""
+
+from __future__ import annotations
+
+import time
+from collections import defaultdict
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from ..storage.types import Edge, Engine, Node, NotFoundError, new_id
+from .linkpredict import adamic_adar
+from .temporal import AccessTracker
+
+
+@dataclass
+class InferenceConfig:
+    similarity_threshold: float = 0.82
+    max_suggestions_per_store: int = 5
+    min_confidence: float = 0.5
+    cooldown_s: float = 300.0
+    temporal_window_s: float = 600.0
+    evidence_required: int = 1
+    edge_type: str = "RELATES_TO"
+    edge_decay_per_day: float = 0.02
+    prune_below: float = 0.1
+
+
+class InferenceEngine:
+    def __init__(self, engine: Engine, search_service=None,
+                 tracker: AccessTracker = None,
+                 config: InferenceConfig = None, now_fn=time.time):
+        self.engine = engine
+        self.search = search_service
+        self.tracker = tracker or AccessTracker(now_fn=now_fn)
+        self.cfg = config or InferenceConfig()
+        self.now = now_fn
+        self._cooldown: Dict[Tuple[str, str], float] = {}
+        self._evidence: Dict[Tuple[str, str], int] = defaultdict(int)
+
+    # ---- main hook ----
+    def on_store(self, node: Node) -> List[Edge]:
+        """Run inference for a freshly stored/embedded node; returns created edges."""
+        suggestions = self.suggest(node)
+        created = []
+        for target, confidence, reason in suggestions:
+            e = self._maybe_link(node.id, target, confidence, reason)
+            if e is not None:
+                created.append(e)
+        return created
+
+    def suggest(self, node: Node) -> List[Tuple[str, float, str]]:
+        out: List[Tuple[str, float, str]] = []
+        seen = set()
+        # (a) embedding similarity
+        if self.search is not None and node.embedding is not None:
+            for r in self.search.vector_search(node.embedding,
+                                               self.cfg.max_suggestions_per_store + 1):
+                if r.id == node.id or r.id in seen:
+                    continue
+                if r.score >= self.cfg.similarity_threshold:
+                    seen.add(r.id)
+                    out.append((r.id, float(r.score), "similarity"))
+        # (b) co-access
+        for a, b, count in self.tracker.co_accessed(self.cfg.temporal_window_s)[:10]:
+            other = b if a == node.id else a if b == node.id else None
+            if other and other not in seen:
+                seen.add(other)
+                conf = min(0.5 + 0.1 * count, 0.9)
+                out.append((other, conf, "co-access"))
+        # (c) temporal proximity: nodes created within the window
+        t0 = node.properties.get("created_at", self.now())
+        for other in self.engine.get_nodes_by_label("Memory"):
+            if other.id == node.id or other.id in seen:
+                continue
+            dt = abs(other.properties.get("created_at", 0) - t0)
+            if 0 < dt <= self.cfg.temporal_window_s:
+                seen.add(other.id)
+                conf = 0.5 + 0.3 * (1 - dt / self.cfg.temporal_window_s)
+                out.append((other.id, conf, "temporal"))
+        # (d) transitive topology (2-hop closure score)
+        for cand, score in self._transitive(node.id):
+            if cand not in seen:
+                seen.add(cand)
+                out.append((cand, min(0.5 + score / 4.0, 0.95), "transitive"))
+        out.sort(key=lambda t: -t[1])
+        return out[: self.cfg.max_suggestions_per_store]
+
+    def _transitive(self, nid: str) -> List[Tuple[str, float]]:
+        direct = set(self.engine.neighbors(nid))
+        scores = []
+        two_hop = set()
+        for nb in direct:
+            two_hop |= set(self.engine.neighbors(nb))
+        two_hop -= direct
+        two_hop.discard(nid)
+        for c in list(two_hop)[:20]:
+            s = adamic_adar(self.engine, nid, c)
+            if s > 0:
+                scores.append((c, s))
+        scores.sort(key=lambda kv: -kv[1])
+        return scores[:5]
+
+    def _maybe_link(self, a: str, b: str, confidence: float,
+                    reason: str) -> Optional[Edge]:
+        if confidence < self.cfg.min_confidence:
+            return None
+        key = tuple(sorted((a, b)))
+        now = self.now()
+        if now - self._cooldown.get(key, 0) < self.cfg.cooldown_s:
+            return None
+        self._evidence[key] += 1
+        if self._evidence[key] < self.cfg.evidence_required:
+            return None
+        # already linked?
+        if b in self.engine.neighbors(a):
+            return None
+        self._cooldown[key] = now
+        e = Edge(id=new_id("inf"), type=self.cfg.edge_type, start_node=a,
+                 end_node=b,
+                 properties={"confidence": confidence, "inferred": True,
+                             "reason": reason, "created_at": now})
+        try:
+            return self.engine.create_edge(e)
+        except NotFoundError:
+            return None
+
+    # ---- edge decay (reference edge_decay.go) ----
+    def decay_inferred_edges(self) -> Dict[str, int]:
+        stats = {"decayed": 0, "pruned": 0}
+        now = self.now()
+        for e in list(self.engine.all_edges()):
+            if not e.properties.get("inferred"):
+                continue
+            age_days = (now - e.properties.get("created_at", now)) / 86400.0
+            conf = e.properties.get("confidence", 0.5) - \
+                self.cfg.edge_decay_per_day * age_days
+            if conf < self.cfg.prune_below:
+                self.engine.delete_edge(e.id)
+                stats["pruned"] += 1
+            else:
+                e.properties["confidence"] = conf
+                self.engine.update_edge(e)
+                stats["decayed"] += 1
+        return stats

@@ -1,0 +1,122 @@
+"""Temporal access tracking: Kalman next-access prediction, session
+boundaries, co-access detection, cyclic pattern detection.
+
+Parity: reference pkg/temporal (tracker.go, pattern_detector.go,
+query_load.go).
+"""
+
+from __future__ import annotations
+
+import math
+import threading
+import time
+from collections import defaultdict, deque
+from typing import Dict, List, Optional, Tuple
+
+from .kalman import Kalman1D
+
+SESSION_GAP = 1800.0  # 30 min
+
+
+class AccessTracker:
+    def __init__(self, max_history: int = 256, now_fn=time.time):
+        self.now = now_fn
+        self._lock = threading.Lock()
+        self._hist: Dict[str, deque] = defaultdict(lambda: deque(maxlen=max_history))
+        self._interval_kf: Dict[str, Kalman1D] = {}
+        self._sessions: List[Tuple[float, float]] = []
+        self._last_access_time: Optional[float] = None
+
+    def record(self, node_id: str, ts: float = None):
+        ts = ts if ts is not None else self.now()
+        with self._lock:
+            h = self._hist[node_id]
+            if h:
+                interval = ts - h[-1]
+                kf = self._interval_kf.setdefault(
+                    node_id, Kalman1D(q=10.0, r=100.0, initial=interval))
+                kf.update(interval)
+            h.append(ts)
+            # session boundary
+            if (self._last_access_time is None
+                    or ts - self._last_access_time > SESSION_GAP):
+                self._sessions.append((ts, ts))
+            else:
+                s, _ = self._sessions[-1]
+                self._sessions[-1] = (s, ts)
+            self._last_access_time = ts
+
+    def predict_next_access(self, node_id: str) -> Optional[float]:
+        with self._lock:
+            h = self._hist.get(node_id)
+            kf = self._interval_kf.get(node_id)
+            if not h or kf is None:
+                return None
+            return h[-1] + max(kf.x, 0.0)
+
+    def access_count(self, node_id: str) -> int:
+        with self._lock:
+            return len(self._hist.get(node_id, ()))
+
+    def co_accessed(self, window: float = 60.0) -> List[Tuple[str, str, int]]:
+        """Pairs accessed within `window` seconds of each other."""
+        with self._lock:
+            events = sorted(
+                (ts, nid) for nid, h in self._hist.items() for ts in h)
+        pairs = defaultdict(int)
+        for i, (ts, nid) in enumerate(events):
+            j = i + 1
+            while j < len(events) and events[j][0] - ts <= window:
+                other = events[j][1]
+                if other != nid:
+                    pairs[tuple(sorted((nid, other)))] += 1
+                j += 1
+        return [(a, b, c) for (a, b), c in
+                sorted(pairs.items(), key=lambda kv: -kv[1])]
+
+    def sessions(self) -> List[Tuple[float, float]]:
+        with self._lock:
+            return list(self._sessions)
+
+    def detect_period(self, node_id: str) -> Optional[float]:
+        """Dominant access period via mean/variance test on intervals."""
+        with self._lock:
+            h = list(self._hist.get(node_id, ()))
+        if len(h) < 4:
+            return None
+        intervals = [b - a for a, b in zip(h, h[1:])]
+        mean = sum(intervals) / len(intervals)
+        if mean <= 0:
+            return None
+        var = sum((x - mean) ** 2 for x in intervals) / len(intervals)
+        cv = math.sqrt(var) / mean
+        return mean if cv < 0.5 else None  # regular enough to be a cycle
+
+
+class QueryLoadTracker:
+    """Adaptive decay pacing by query load (reference query_load.go)."""
+
+    def __init__(self, window: float = 60.0, now_fn=time.time):
+        self.window = window
+        self.now = now_fn
+        self._times: deque = deque()
+        self._lock = threading.Lock()
+
+    def record_query(self):
+        with self._lock:
+            t = self.now()
+            self._times.append(t)
+            while self._times and self._times[0] < t - self.window:
+                self._times.popleft()
+
+    def qps(self) -> float:
+        with self._lock:
+            t = self.now()
+            while self._times and self._times[0] < t - self.window:
+                self._times.popleft()
+            return len(self._times) / self.window
+
+    def decay_interval(self, base: float = 300.0) -> float:
+        """Back off decay cycles under load."""
+        load = self.qps()
+        return base * (1.0 + min(load / 10.0, 10.0))

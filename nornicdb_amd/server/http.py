@@ -1,0 +1,272 @@
+"""HTTP API server (FastAPI).
+
+Parity: reference pkg/server/server_router.go — Neo4j HTTP tx API
+(/db/{name}/tx/commit, server_db.go), /nornicdb/* memory+search routes
+(server_nornicdb.go), auth routes, admin + health + Prometheus /metrics
+(:104-108), GDPR endpoints (server_gdpr.go), MCP mount (/mcp).
+"""
+
+from __future__ import annotations
+
+import json
+import time
+from typing import Any, Dict, List, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Request, Response
+from pydantic import BaseModel
+
+from ..cypher import CypherRuntimeError, CypherSyntaxError
+from ..cypher.executor import Path
+from ..db import DatabaseManager
+from ..storage.types import Edge, Node
+from .metrics import MetricsRegistry
+from .mcp import MCPServer
+
+START_TIME = time.time()
+
+
+def _jsonable(v):
+    if isinstance(v, Node):
+        return {"id": v.id, "labels": v.labels, "properties": v.properties}
+    if isinstance(v, Edge):
+        return {"id": v.id, "type": v.type, "startNode": v.start_node,
+                "endNode": v.end_node, "properties": v.properties}
+    if isinstance(v, Path):
+        return {"nodes": [_jsonable(n) for n in v.nodes],
+                "relationships": [_jsonable(e) for e in v.edges]}
+    if isinstance(v, list):
+        return [_jsonable(x) for x in v]
+    if isinstance(v, dict):
+        return {k: _jsonable(x) for k, x in v.items()}
+    if isinstance(v, float) and (v != v or v in (float("inf"), float("-inf"))):
+        return None
+    return v
+
+
+class Statement(BaseModel):
+    statement: str
+    parameters: Dict[str, Any] = {}
+
+
+class TxRequest(BaseModel):
+    statements: List[Statement] = []
+
+
+class StoreRequest(BaseModel):
+    content: str
+    title: str = ""
+    memory_type: str = "episodic"
+    importance: float = 0.5
+    tags: List[str] = []
+    metadata: Dict[str, Any] = {}
+
+
+class SearchRequest(BaseModel):
+    query: str
+    limit: int = 10
+    labels: Optional[List[str]] = None
+    mmr: bool = False
+
+
+class LoginRequest(BaseModel):
+    username: str
+    password: str
+
+
+def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastAPI:
+    app = FastAPI(title="NornicDB-AMD", version=version)
+    metrics = MetricsRegistry()
+    mcp = MCPServer(mgr)
+    app.state.manager = mgr
+    app.state.metrics = metrics
+
+    def check_auth(request: Request):
+        if auth is None:
+            return None
+        hdr = request.headers.get("authorization", "")
+        if hdr.lower().startswith("bearer "):
+            try:
+                return auth.validate_token(hdr[7:])
+            except Exception:
+                raise HTTPException(401, "invalid token")
+        if hdr.lower().startswith("basic "):
+            import base64
+            try:
+                user, pw = base64.b64decode(hdr[6:]).decode().split(":", 1)
+                auth.login(user, pw)
+                return user
+            except HTTPException:
+                raise
+            except Exception:
+                raise HTTPException(401, "invalid credentials")
+        raise HTTPException(401, "authentication required")
+
+    # ---- health / status / metrics ----
+    @app.get("/health")
+    def health():
+        return {"status": "ok"}
+
+    @app.get("/status")
+    def status():
+        db = mgr.get()
+        return {
+            "version": version,
+            "uptime_s": round(time.time() - START_TIME, 1),
+            "databases": mgr.list(),
+            "nodes": db.engine.node_count(),
+            "relationships": db.engine.edge_count(),
+        }
+
+    @app.get("/metrics")
+    def prom_metrics():
+        db = mgr.get()
+        metrics.gauge("nornicdb_nodes", db.engine.node_count())
+        metrics.gauge("nornicdb_relationships", db.engine.edge_count())
+        metrics.gauge("nornicdb_uptime_seconds", time.time() - START_TIME)
+        return Response(metrics.render(), media_type="text/plain; version=0.0.4")
+
+    # ---- auth ----
+    @app.post("/auth/login")
+    def login(req: LoginRequest):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        try:
+            token = auth.issue_token(req.username, req.password)
+        except Exception as e:
+            raise HTTPException(401, str(e))
+        return {"token": token}
+
+    # ---- Neo4j HTTP transaction API ----
+    @app.post("/db/{db_name}/tx/commit")
+    def tx_commit(db_name: str, req: TxRequest, request: Request,
+                  _user=Depends(check_auth)):
+        try:
+            db = mgr.get(db_name)
+        except KeyError:
+            raise HTTPException(404, f"database {db_name} not found")
+        results, errors = [], []
+        for stmt in req.statements:
+            t0 = time.time()
+            try:
+                r = db.cypher(stmt.statement, stmt.parameters)
+                metrics.observe("nornicdb_query_seconds", time.time() - t0)
+                results.append({
+                    "columns": r.columns,
+                    "data": [{"row": [_jsonable(v) for v in row],
+                              "meta": []} for row in r.rows],
+                    "stats": r.stats,
+                })
+            except (CypherSyntaxError,) as e:
+                errors.append({"code": "Neo.ClientError.Statement.SyntaxError",
+                               "message": str(e)})
+                break
+            except Exception as e:
+                errors.append({"code": "Neo.ClientError.Statement.ExecutionFailed",
+                               "message": str(e)})
+                break
+        return {"results": results, "errors": errors}
+
+    # ---- multi-database management ----
+    @app.post("/admin/databases/{name}")
+    def create_db(name: str, _user=Depends(check_auth)):
+        try:
+            mgr.create(name)
+        except ValueError as e:
+            raise HTTPException(409, str(e))
+        return {"created": name}
+
+    @app.delete("/admin/databases/{name}")
+    def drop_db(name: str, _user=Depends(check_auth)):
+        try:
+            mgr.drop(name)
+        except (KeyError, ValueError) as e:
+            raise HTTPException(400, str(e))
+        return {"dropped": name}
+
+    @app.get("/admin/databases")
+    def list_dbs(_user=Depends(check_auth)):
+        return {"databases": mgr.list()}
+
+    @app.post("/admin/backup")
+    def backup(_user=Depends(check_auth)):
+        base = mgr._base
+        if hasattr(base, "snapshot"):
+            base.snapshot()
+            return {"status": "snapshot written"}
+        return {"status": "in-memory engine, nothing to snapshot"}
+
+    # ---- NornicDB memory/search API ----
+    @app.post("/nornicdb/store")
+    def store(req: StoreRequest, db: str = None, _user=Depends(check_auth)):
+        m = mgr.get(db).store(req.content, title=req.title,
+                              memory_type=req.memory_type,
+                              importance=req.importance, tags=req.tags,
+                              metadata=req.metadata)
+        return {"id": m.id}
+
+    @app.post("/nornicdb/search")
+    def search(req: SearchRequest, db: str = None, _user=Depends(check_auth)):
+        d = mgr.get(db)
+        qv = d.embedder.embed_query(req.query)
+        res = d.search.search(query=req.query, query_vec=qv, k=req.limit,
+                              labels=req.labels, mmr=req.mmr)
+        return {"results": [{"id": r.id, "score": r.score,
+                             "node": _jsonable(r.node)} for r in res]}
+
+    @app.get("/nornicdb/similar/{node_id}")
+    def similar(node_id: str, limit: int = 10, db: str = None,
+                _user=Depends(check_auth)):
+        d = mgr.get(db)
+        try:
+            node = d.engine.get_node(node_id)
+        except Exception:
+            raise HTTPException(404, "node not found")
+        if node.embedding is None:
+            raise HTTPException(400, "node has no embedding")
+        res = d.search.vector_search(node.embedding, limit + 1)
+        return {"results": [{"id": r.id, "score": r.score}
+                            for r in res if r.id != node_id][:limit]}
+
+    @app.post("/nornicdb/embed")
+    def embed(body: Dict[str, Any], db: str = None, _user=Depends(check_auth)):
+        texts = body.get("texts") or [body.get("text", "")]
+        vecs = mgr.get(db).embedder.embed_batch(texts)
+        return {"embeddings": [v.tolist() for v in vecs]}
+
+    @app.post("/nornicdb/decay/run")
+    def decay_run(db: str = None, _user=Depends(check_auth)):
+        from ..cognitive.decay import DecayManager
+        d = mgr.get(db)
+        dm = getattr(d, "_decay", None) or DecayManager(d.engine)
+        stats = dm.run_cycle()
+        return stats
+
+    # ---- GDPR (reference server_gdpr.go / pkg/retention) ----
+    @app.get("/gdpr/export/{subject}")
+    def gdpr_export(subject: str, db: str = None, _user=Depends(check_auth)):
+        d = mgr.get(db)
+        nodes = [
+            _jsonable(n) for n in d.engine.all_nodes()
+            if n.properties.get("subject") == subject
+            or n.properties.get("user") == subject
+        ]
+        return {"subject": subject, "nodes": nodes}
+
+    @app.delete("/gdpr/delete/{subject}")
+    def gdpr_delete(subject: str, db: str = None, _user=Depends(check_auth)):
+        d = mgr.get(db)
+        deleted = 0
+        for n in list(d.engine.all_nodes()):
+            if (n.properties.get("subject") == subject
+                    or n.properties.get("user") == subject):
+                d.engine.detach_delete_node(n.id)
+                deleted += 1
+        return {"subject": subject, "deleted": deleted}
+
+    # ---- MCP (JSON-RPC 2.0) ----
+    @app.post("/mcp")
+    async def mcp_endpoint(request: Request):
+        body = await request.json()
+        return mcp.handle(body)
+
+    return app

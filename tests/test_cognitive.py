@@ -1,0 +1,202 @@
+"""Cognitive layer tests: decay scoring/archival, Kalman, temporal
+tracking, link prediction, inference auto-linking.
+
+Models reference pkg/decay, pkg/filter, pkg/temporal, pkg/linkpredict,
+pkg/inference tests."""
+
+import numpy as np
+import pytest
+
+from nornicdb_amd.cognitive import (AccessTracker, DecayConfig, DecayManager,
+                                    HALF_LIVES, InferenceConfig,
+                                    InferenceEngine, Kalman1D,
+                                    QueryLoadTracker, adamic_adar,
+                                    common_neighbors, jaccard, predict_links)
+from nornicdb_amd.search import SearchService
+from nornicdb_amd.storage import Edge, MemoryEngine, Node
+
+DAY = 86400.0
+
+
+def mem(id, created, accessed=None, tier="episodic", importance=0.5, count=0):
+    return Node(id, ["Memory"], {
+        "memory_type": tier, "importance": importance,
+        "created_at": created, "last_accessed": accessed or created,
+        "access_count": count, "content": f"content {id}"})
+
+
+class TestDecay:
+    def test_halflife_recency(self):
+        eng = MemoryEngine()
+        now = [1000.0 * DAY]
+        dm = DecayManager(eng, now_fn=lambda: now[0])
+        fresh = mem("fresh", now[0])
+        week_old = mem("old", now[0] - 7 * DAY)
+        eng.create_node(fresh); eng.create_node(week_old)
+        s_fresh = dm.score(eng.get_node("fresh"))
+        s_old = dm.score(eng.get_node("old"))
+        assert s_fresh > s_old
+        # one episodic half-life halves the recency component
+        assert abs((s_fresh - s_old) - 0.5 * 0.5) < 0.02
+
+    def test_tier_halflives(self):
+        eng = MemoryEngine()
+        now = [1000.0 * DAY]
+        dm = DecayManager(eng, now_fn=lambda: now[0])
+        for tier in ("episodic", "semantic", "procedural"):
+            eng.create_node(mem(tier, now[0] - 30 * DAY, tier=tier))
+        se = dm.score(eng.get_node("episodic"))
+        ss = dm.score(eng.get_node("semantic"))
+        sp = dm.score(eng.get_node("procedural"))
+        assert se < ss < sp
+
+    def test_cycle_archives_and_deletes(self):
+        eng = MemoryEngine()
+        now = [1000.0 * DAY]
+        dm = DecayManager(eng, DecayConfig(archive_threshold=0.3,
+                                           delete_threshold=0.05),
+                          now_fn=lambda: now[0])
+        eng.create_node(mem("dead", now[0] - 400 * DAY, importance=0.0))
+        eng.create_node(mem("fading", now[0] - 40 * DAY, importance=0.4))
+        eng.create_node(mem("alive", now[0], importance=0.9, count=10))
+        stats = dm.run_cycle()
+        assert stats["deleted"] == 1
+        assert not eng.has_node("dead")
+        assert "Archived" in eng.get_node("fading").labels
+        assert "Archived" not in eng.get_node("alive").labels
+
+    def test_reinforce(self):
+        eng = MemoryEngine()
+        eng.create_node(mem("m", 0))
+        dm = DecayManager(eng)
+        imp = dm.reinforce("m")
+        assert imp == 0.6
+        assert eng.get_node("m").properties["access_count"] == 1
+
+
+class TestKalman:
+    def test_converges_to_constant(self):
+        kf = Kalman1D(q=0.001, r=0.5, initial=0.0)
+        for _ in range(100):
+            kf.update(5.0)
+        assert abs(kf.x - 5.0) < 0.05
+
+    def test_smooths_noise(self):
+        rng = np.random.default_rng(0)
+        kf = Kalman1D(q=0.01, r=1.0)
+        xs = [kf.update(3.0 + rng.normal(0, 0.5)) for _ in range(200)]
+        assert abs(np.mean(xs[-50:]) - 3.0) < 0.2
+        assert np.std(xs[-50:]) < 0.2
+
+
+class TestTemporal:
+    def test_next_access_prediction(self):
+        now = [0.0]
+        t = AccessTracker(now_fn=lambda: now[0])
+        for i in range(10):
+            now[0] = i * 100.0
+            t.record("n")
+        pred = t.predict_next_access("n")
+        assert pred is not None
+        assert abs(pred - 1000.0) < 30
+
+    def test_sessions(self):
+        now = [0.0]
+        t = AccessTracker(now_fn=lambda: now[0])
+        t.record("a"); now[0] = 60; t.record("b")
+        now[0] = 10000; t.record("c")
+        assert len(t.sessions()) == 2
+
+    def test_co_access(self):
+        now = [0.0]
+        t = AccessTracker(now_fn=lambda: now[0])
+        t.record("x"); now[0] = 5; t.record("y")
+        now[0] = 5000; t.record("z")
+        pairs = t.co_accessed(window=60)
+        assert pairs[0][:2] == ("x", "y")
+
+    def test_period_detection(self):
+        now = [0.0]
+        t = AccessTracker(now_fn=lambda: now[0])
+        for i in range(8):
+            now[0] = i * 3600.0
+            t.record("daily")
+        p = t.detect_period("daily")
+        assert p is not None and abs(p - 3600) < 1
+
+    def test_query_load(self):
+        now = [0.0]
+        q = QueryLoadTracker(window=10, now_fn=lambda: now[0])
+        for _ in range(50):
+            q.record_query()
+        assert q.qps() == 5.0
+        assert q.decay_interval(100) > 100
+
+
+class TestLinkPredict:
+    def _triangle_plus(self):
+        eng = MemoryEngine()
+        for n in "abcd":
+            eng.create_node(Node(n, [], {}))
+        eng.create_edge(Edge("e1", "R", "a", "b"))
+        eng.create_edge(Edge("e2", "R", "b", "c"))
+        eng.create_edge(Edge("e3", "R", "a", "d"))
+        eng.create_edge(Edge("e4", "R", "d", "c"))
+        return eng
+
+    def test_scores(self):
+        eng = self._triangle_plus()
+        assert common_neighbors(eng, "a", "c") == 2.0
+        assert 0 < jaccard(eng, "a", "c") <= 1
+        assert adamic_adar(eng, "a", "c") > 0
+
+    def test_predict_links_ranks_closure(self):
+        eng = self._triangle_plus()
+        preds = predict_links(eng, "a", method="common_neighbors")
+        assert preds and preds[0][0] == "c"
+
+
+class TestInference:
+    def test_similarity_autolink(self):
+        eng = MemoryEngine()
+        svc = SearchService(eng, dims=8, device="cpu", use_hnsw=False)
+        v = np.zeros(8); v[0] = 1.0
+        a = Node("a", ["Memory"], {"created_at": 0.0}, embedding=list(v))
+        v2 = v + 0.01
+        b = Node("b", ["Memory"], {"created_at": 1e9}, embedding=list(v2 / np.linalg.norm(v2)))
+        eng.create_node(a)
+        eng.create_node(b)
+        inf = InferenceEngine(eng, svc, config=InferenceConfig(
+            similarity_threshold=0.9, temporal_window_s=1))
+        created = inf.on_store(eng.get_node("b"))
+        assert created and created[0].type == "RELATES_TO"
+        assert created[0].properties["reason"] == "similarity"
+        assert "a" in eng.neighbors("b")
+
+    def test_cooldown_prevents_duplicates(self):
+        eng = MemoryEngine()
+        svc = SearchService(eng, dims=8, device="cpu", use_hnsw=False)
+        v = [1.0] + [0.0] * 7
+        eng.create_node(Node("a", ["Memory"], {"created_at": 0.0}, embedding=v))
+        eng.create_node(Node("b", ["Memory"], {"created_at": 1e9}, embedding=v))
+        inf = InferenceEngine(eng, svc, config=InferenceConfig(
+            similarity_threshold=0.9, temporal_window_s=1))
+        first = inf.on_store(eng.get_node("b"))
+        second = inf.on_store(eng.get_node("b"))
+        assert len(first) == 1 and len(second) == 0
+
+    def test_edge_decay_prunes(self):
+        eng = MemoryEngine()
+        now = [0.0]
+        inf = InferenceEngine(eng, None, now_fn=lambda: now[0],
+                              config=InferenceConfig(edge_decay_per_day=0.1,
+                                                     prune_below=0.2))
+        eng.create_node(Node("a", [], {}))
+        eng.create_node(Node("b", [], {}))
+        eng.create_edge(Edge("e", "RELATES_TO", "a", "b",
+                             {"inferred": True, "confidence": 0.5,
+                              "created_at": 0.0}))
+        now[0] = 10 * 86400.0  # conf -> 0.5 - 1.0 < prune
+        stats = inf.decay_inferred_edges()
+        assert stats["pruned"] == 1
+        assert eng.edge_count() == 0

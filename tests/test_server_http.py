@@ -1,0 +1,142 @@
+"""HTTP + MCP server tests via FastAPI TestClient.
+
+Models reference pkg/server + pkg/mcp e2e tests."""
+
+import pytest
+from fastapi.testclient import TestClient
+
+from nornicdb_amd.db import open_db
+from nornicdb_amd.embed import MockEmbedder
+from nornicdb_amd.server import create_app
+
+
+@pytest.fixture
+def client():
+    mgr = open_db(embedder=MockEmbedder(32), dims=32)
+    app = create_app(mgr)
+    with TestClient(app) as c:
+        yield c
+    mgr.close()
+
+
+class TestCore:
+    def test_health_status(self, client):
+        assert client.get("/health").json()["status"] == "ok"
+        s = client.get("/status").json()
+        assert "neo4j" in s["databases"]
+
+    def test_metrics_prometheus(self, client):
+        text = client.get("/metrics").text
+        assert "nornicdb_nodes" in text and "# TYPE" in text
+
+
+class TestTxAPI:
+    def test_cypher_roundtrip(self, client):
+        r = client.post("/db/neo4j/tx/commit", json={"statements": [
+            {"statement": "CREATE (n:City {name: $n}) RETURN n",
+             "parameters": {"n": "Oslo"}}]}).json()
+        assert r["errors"] == []
+        res = r["results"][0]
+        assert res["columns"] == ["n"]
+        assert res["data"][0]["row"][0]["properties"]["name"] == "Oslo"
+        assert res["stats"]["nodes_created"] == 1
+
+    def test_syntax_error_reported(self, client):
+        r = client.post("/db/neo4j/tx/commit", json={"statements": [
+            {"statement": "MATCH (n RETURN"}]}).json()
+        assert r["errors"] and "SyntaxError" in r["errors"][0]["code"]
+
+    def test_unknown_db_404(self, client):
+        assert client.post("/db/nope/tx/commit",
+                           json={"statements": []}).status_code == 404
+
+    def test_multi_statement(self, client):
+        r = client.post("/db/neo4j/tx/commit", json={"statements": [
+            {"statement": "CREATE (:X {v: 1})"},
+            {"statement": "MATCH (n:X) RETURN n.v"}]}).json()
+        assert r["results"][1]["data"][0]["row"] == [1]
+
+
+class TestNornicRoutes:
+    def test_store_search_similar(self, client):
+        rid = client.post("/nornicdb/store", json={
+            "content": "the quick brown fox", "title": "fox"}).json()["id"]
+        # drain embed queue synchronously through the manager
+        mgr = client.app.state.manager
+        mgr.get().embed_queue.drain()
+        res = client.post("/nornicdb/search",
+                          json={"query": "quick brown fox"}).json()["results"]
+        assert res and res[0]["id"] == rid
+        sim = client.get(f"/nornicdb/similar/{rid}").json()
+        assert "results" in sim
+
+    def test_embed_endpoint(self, client):
+        r = client.post("/nornicdb/embed", json={"texts": ["a", "b"]}).json()
+        assert len(r["embeddings"]) == 2
+        assert len(r["embeddings"][0]) == 32
+
+    def test_decay_run(self, client):
+        client.post("/nornicdb/store", json={"content": "m"})
+        r = client.post("/nornicdb/decay/run").json()
+        assert r["scored"] >= 1
+
+
+class TestAdmin:
+    def test_db_management(self, client):
+        assert client.post("/admin/databases/t1").status_code == 200
+        assert "t1" in client.get("/admin/databases").json()["databases"]
+        assert client.delete("/admin/databases/t1").status_code == 200
+
+    def test_gdpr(self, client):
+        client.post("/db/neo4j/tx/commit", json={"statements": [
+            {"statement": "CREATE (:P {subject: 'alice', d: 1})"}]})
+        exp = client.get("/gdpr/export/alice").json()
+        assert len(exp["nodes"]) == 1
+        dele = client.delete("/gdpr/delete/alice").json()
+        assert dele["deleted"] == 1
+
+
+class TestMCP:
+    def _rpc(self, client, method, params=None, id=1):
+        return client.post("/mcp", json={
+            "jsonrpc": "2.0", "id": id, "method": method,
+            "params": params or {}}).json()
+
+    def test_initialize_and_list(self, client):
+        r = self._rpc(client, "initialize")
+        assert r["result"]["serverInfo"]["name"] == "nornicdb-amd"
+        tools = self._rpc(client, "tools/list")["result"]["tools"]
+        names = {t["name"] for t in tools}
+        assert names == {"store", "recall", "discover", "link", "task", "tasks"}
+
+    def test_store_recall_link_flow(self, client):
+        import json as J
+        r1 = self._rpc(client, "tools/call", {
+            "name": "store", "arguments": {"content": "mcp memory one"}})
+        id1 = J.loads(r1["result"]["content"][0]["text"])["id"]
+        r2 = self._rpc(client, "tools/call", {
+            "name": "store", "arguments": {"content": "mcp memory two"}})
+        id2 = J.loads(r2["result"]["content"][0]["text"])["id"]
+        client.app.state.manager.get().embed_queue.drain()
+        rec = self._rpc(client, "tools/call", {
+            "name": "recall", "arguments": {"query": "mcp memory one"}})
+        found = J.loads(rec["result"]["content"][0]["text"])
+        assert any(m["id"] == id1 for m in found)
+        self._rpc(client, "tools/call", {
+            "name": "link", "arguments": {"from": id1, "to": id2}})
+        disc = self._rpc(client, "tools/call", {
+            "name": "discover", "arguments": {"id": id1}})
+        linked = J.loads(disc["result"]["content"][0]["text"])
+        assert any(m["id"] == id2 for m in linked)
+
+    def test_tasks(self, client):
+        import json as J
+        self._rpc(client, "tools/call", {
+            "name": "task", "arguments": {"title": "write tests"}})
+        r = self._rpc(client, "tools/call", {"name": "tasks", "arguments": {}})
+        tasks = J.loads(r["result"]["content"][0]["text"])
+        assert tasks and tasks[0]["title"] == "write tests"
+
+    def test_unknown_method(self, client):
+        r = self._rpc(client, "nope/nope")
+        assert r["error"]["code"] == -32601
