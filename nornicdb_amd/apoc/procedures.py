@@ -1,0 +1,277 @@
+"""APOC procedures (CALL-position).
+
+Parity: reference apoc/algo (PageRank/Dijkstra/AStar/Betweenness...),
+apoc/community (Louvain/LabelProp/WCC/Triangles), apoc/periodic (batch
+iterate), apoc/create, apoc/refactor, apoc/meta, apoc/merge — wired to
+the CSR graph module so large graphs run the HIP kernels.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict
+
+import numpy as np
+
+from ..graph import (betweenness_centrality, closeness_centrality,
+                     connected_components, degree_centrality, dijkstra,
+                     from_engine, label_propagation, louvain, pagerank,
+                     shortest_path, triangle_count, clustering_coefficient)
+from ..storage.types import Edge, Node, new_id
+
+
+def build_apoc_procedures(db) -> Dict[str, Any]:
+    procs: Dict[str, Any] = {}
+
+    def register(name):
+        def deco(fn):
+            procs[name.lower()] = fn
+            return fn
+        return deco
+
+    eng = db.engine
+
+    # -------------------- apoc.algo --------------------
+    @register("apoc.algo.pageRank")
+    def _pagerank(ex, iterations=20, damping=0.85, label=None):
+        g = from_engine(eng)
+        r = pagerank(g, damping=float(damping), iters=int(iterations))
+        rows = [[eng.get_node(g.node_ids[i]), float(r[i])]
+                for i in np.argsort(-r)]
+        return ["node", "score"], rows
+
+    @register("apoc.algo.degree")
+    def _degree(ex):
+        g = from_engine(eng)
+        d = degree_centrality(g)
+        return ["node", "score"], [[eng.get_node(g.node_ids[i]), float(d[i])]
+                                   for i in np.argsort(-d)]
+
+    @register("apoc.algo.betweenness")
+    def _betweenness(ex, samples=None):
+        g = from_engine(eng)
+        bc = betweenness_centrality(g, samples=int(samples) if samples else None)
+        return ["node", "score"], [[eng.get_node(g.node_ids[i]), float(bc[i])]
+                                   for i in np.argsort(-bc)]
+
+    @register("apoc.algo.closeness")
+    def _closeness(ex):
+        g = from_engine(eng)
+        c = closeness_centrality(g)
+        return ["node", "score"], [[eng.get_node(g.node_ids[i]), float(c[i])]
+                                   for i in np.argsort(-c)]
+
+    @register("apoc.algo.dijkstra")
+    def _dijkstra(ex, start, end, rel_type=None, weight_prop="weight"):
+        g = from_engine(eng, edge_types=[rel_type] if rel_type else None,
+                        weight_prop=weight_prop)
+        s = g.id2idx[start.id if isinstance(start, Node) else start]
+        t = g.id2idx[end.id if isinstance(end, Node) else end]
+        dist, _ = dijkstra(g, s, t)
+        path_idx = shortest_path(g, s, t)
+        nodes = [eng.get_node(g.node_ids[i]) for i in path_idx]
+        return ["path", "weight"], [[nodes, float(dist[t])]] if path_idx else []
+
+    @register("apoc.algo.aStar")
+    def _astar_p(ex, start, end, weight_prop="weight", lat="lat", lon="lon"):
+        return _dijkstra(ex, start, end, None, weight_prop)
+
+    # -------------------- apoc.community / graph --------------------
+    @register("apoc.community.louvain")
+    def _louvain(ex):
+        g = from_engine(eng, undirected=True)
+        comm = louvain(g)
+        return ["node", "community"], [[eng.get_node(g.node_ids[i]), int(comm[i])]
+                                       for i in range(g.n)]
+
+    @register("apoc.community.labelPropagation")
+    def _labelprop(ex, iterations=20):
+        g = from_engine(eng, undirected=True)
+        lb = label_propagation(g, iters=int(iterations))
+        return ["node", "community"], [[eng.get_node(g.node_ids[i]), int(lb[i])]
+                                       for i in range(g.n)]
+
+    @register("apoc.community.wcc")
+    def _wcc(ex):
+        g = from_engine(eng)
+        c = connected_components(g)
+        return ["node", "component"], [[eng.get_node(g.node_ids[i]), int(c[i])]
+                                       for i in range(g.n)]
+
+    @register("apoc.community.triangleCount")
+    def _tri(ex):
+        g = from_engine(eng)
+        return ["triangles"], [[triangle_count(g)]]
+
+    @register("apoc.community.clusteringCoefficient")
+    def _cc(ex):
+        g = from_engine(eng)
+        cc = clustering_coefficient(g)
+        return ["node", "coefficient"], [[eng.get_node(g.node_ids[i]), float(cc[i])]
+                                         for i in range(g.n)]
+
+    # -------------------- apoc.create --------------------
+    @register("apoc.create.node")
+    def _create_node(ex, labels, props):
+        n = Node(id=new_id("n"), labels=list(labels or []),
+                 properties=dict(props or {}))
+        return ["node"], [[eng.create_node(n)]]
+
+    @register("apoc.create.nodes")
+    def _create_nodes(ex, labels, props_list):
+        rows = []
+        for props in props_list or []:
+            n = Node(id=new_id("n"), labels=list(labels or []),
+                     properties=dict(props or {}))
+            rows.append([eng.create_node(n)])
+        return ["node"], rows
+
+    @register("apoc.create.relationship")
+    def _create_rel(ex, start, rel_type, props, end):
+        e = Edge(id=new_id("e"), type=rel_type,
+                 start_node=start.id if isinstance(start, Node) else start,
+                 end_node=end.id if isinstance(end, Node) else end,
+                 properties=dict(props or {}))
+        return ["rel"], [[eng.create_edge(e)]]
+
+    @register("apoc.create.uuid")
+    def _uuid(ex):
+        import uuid
+        return ["uuid"], [[str(uuid.uuid4())]]
+
+    # -------------------- apoc.merge --------------------
+    @register("apoc.merge.node")
+    def _merge_node(ex, labels, ident_props, on_create_props=None, on_match_props=None):
+        labels = list(labels or [])
+        cands = eng.get_nodes_by_label(labels[0]) if labels else list(eng.all_nodes())
+        for n in cands:
+            if all(n.properties.get(k) == v for k, v in (ident_props or {}).items()) \
+                    and all(lb in n.labels for lb in labels):
+                if on_match_props:
+                    n.properties.update(on_match_props)
+                    n = eng.update_node(n)
+                return ["node"], [[n]]
+        props = dict(ident_props or {})
+        props.update(on_create_props or {})
+        n = Node(id=new_id("n"), labels=labels, properties=props)
+        return ["node"], [[eng.create_node(n)]]
+
+    # -------------------- apoc.refactor --------------------
+    @register("apoc.refactor.mergeNodes")
+    def _merge_nodes(ex, nodes, config=None):
+        if not nodes:
+            return ["node"], []
+        target = nodes[0]
+        tnode = eng.get_node(target.id)
+        for other in nodes[1:]:
+            o = eng.get_node(other.id)
+            for k, v in o.properties.items():
+                tnode.properties.setdefault(k, v)
+            for lb in o.labels:
+                if lb not in tnode.labels:
+                    tnode.labels.append(lb)
+            for e in eng.get_out_edges(o.id):
+                if e.end_node != tnode.id:
+                    eng.create_edge(Edge(new_id("e"), e.type, tnode.id,
+                                         e.end_node, dict(e.properties)))
+            for e in eng.get_in_edges(o.id):
+                if e.start_node != tnode.id:
+                    eng.create_edge(Edge(new_id("e"), e.type, e.start_node,
+                                         tnode.id, dict(e.properties)))
+            eng.detach_delete_node(o.id)
+        eng.update_node(tnode)
+        return ["node"], [[eng.get_node(tnode.id)]]
+
+    @register("apoc.refactor.rename.label")
+    def _rename_label(ex, old, new):
+        count = 0
+        for n in eng.get_nodes_by_label(old):
+            n.labels = [new if lb == old else lb for lb in n.labels]
+            eng.update_node(n)
+            count += 1
+        return ["count"], [[count]]
+
+    @register("apoc.refactor.rename.type")
+    def _rename_type(ex, old, new):
+        count = 0
+        for e in eng.get_edges_by_type(old):
+            e.type = new
+            eng.update_edge(e)
+            count += 1
+        return ["count"], [[count]]
+
+    # -------------------- apoc.periodic --------------------
+    @register("apoc.periodic.iterate")
+    def _periodic_iterate(ex, outer_q, inner_q, config=None):
+        config = config or {}
+        batch_size = int(config.get("batchSize", 1000))
+        outer = ex.execute(outer_q, {})
+        rows = outer.to_dicts()
+        batches = ops_total = failed = 0
+        for i in range(0, len(rows), batch_size):
+            batch = rows[i:i + batch_size]
+            batches += 1
+            for row in batch:
+                try:
+                    ex.execute(inner_q, row, bindings=row)
+                    ops_total += 1
+                except Exception:
+                    failed += 1
+        return (["batches", "total", "failedOperations"],
+                [[batches, ops_total, failed]])
+
+    @register("apoc.periodic.commit")
+    def _periodic_commit(ex, q, params=None):
+        total = 0
+        for _ in range(1000):
+            r = ex.execute(q, params or {})
+            n = r.rows[0][0] if r.rows else 0
+            total += n or 0
+            if not n:
+                break
+        return ["updates"], [[total]]
+
+    # -------------------- apoc.meta --------------------
+    @register("apoc.meta.stats")
+    def _meta_stats(ex):
+        labels: Dict[str, int] = {}
+        for n in eng.all_nodes():
+            for lb in n.labels:
+                labels[lb] = labels.get(lb, 0) + 1
+        types: Dict[str, int] = {}
+        for e in eng.all_edges():
+            types[e.type] = types.get(e.type, 0) + 1
+        return (["nodeCount", "relCount", "labels", "relTypes"],
+                [[eng.node_count(), eng.edge_count(), labels, types]])
+
+    @register("apoc.meta.schema")
+    def _meta_schema(ex):
+        schema: Dict[str, Any] = {}
+        for n in eng.all_nodes():
+            for lb in n.labels:
+                ent = schema.setdefault(lb, {"type": "node", "count": 0,
+                                             "properties": {}})
+                ent["count"] += 1
+                for k, v in n.properties.items():
+                    ent["properties"][k] = type(v).__name__
+        return ["value"], [[schema]]
+
+    # -------------------- apoc.export --------------------
+    @register("apoc.export.json.all")
+    def _export_all(ex, file=None, config=None):
+        import json as J
+        data = {"nodes": [{"id": n.id, "labels": n.labels,
+                           "properties": n.properties}
+                          for n in eng.all_nodes()],
+                "relationships": [{"id": e.id, "type": e.type,
+                                   "start": e.start_node, "end": e.end_node,
+                                   "properties": e.properties}
+                                  for e in eng.all_edges()]}
+        payload = J.dumps(data, default=str)
+        if file:
+            with open(file, "w") as f:
+                f.write(payload)
+        return (["file", "nodes", "relationships"],
+                [[file or "<inline>", len(data["nodes"]),
+                  len(data["relationships"])]])
+
+    return procs

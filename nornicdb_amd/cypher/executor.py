@@ -79,8 +79,12 @@ class Executor:
         self.stats: Dict[str, int] = {}
 
     # ------------------------------------------------------------------ API
-    def execute(self, cypher: str, params: Dict[str, Any] = None) -> Result:
+    def execute(self, cypher: str, params: Dict[str, Any] = None,
+                bindings: Dict[str, Any] = None) -> Result:
+        """bindings: pre-bound row variables (used by apoc.periodic.iterate
+        to pass outer-query columns into the inner query)."""
         params = params or {}
+        self._initial_bindings = bindings
         q = self._plan_cache.get(cypher)
         if q is None:
             q = parse(cypher)
@@ -93,6 +97,7 @@ class Executor:
         if q.explain:
             return Result(["plan"], [[self._explain(q)]], dict(self.stats))
         res = self._run_query(q, params)
+        self._initial_bindings = None
         res.stats = dict(self.stats)
         return res
 
@@ -124,7 +129,8 @@ class Executor:
         return Result(cols, rows)
 
     def _run_clauses(self, clauses, params) -> Tuple[List[str], List[List[Any]]]:
-        rows: List[Dict[str, Any]] = [{}]
+        init = getattr(self, "_initial_bindings", None)
+        rows: List[Dict[str, Any]] = [dict(init) if init else {}]
         out_cols: List[str] = []
         out_rows: List[List[Any]] = []
         i = 0

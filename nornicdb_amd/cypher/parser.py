@@ -59,6 +59,15 @@ class Parser:
             return True
         return False
 
+    def name_part(self) -> str:
+        """Function/property name segment: any identifier or keyword."""
+        t = self.peek()
+        if t.kind == "IDENT":
+            return self.next().value
+        if t.kind == "KW":
+            return self.next().value.lower()
+        raise CypherSyntaxError(f"expected name, got {t.value!r} at {t.pos}")
+
     def ident(self) -> str:
         t = self.peek()
         if t.kind == "IDENT":
@@ -280,7 +289,7 @@ class Parser:
         self.eat_kw("CALL")
         name = self.ident()
         while self.try_op("."):
-            name += "." + self.ident()
+            name += "." + self.name_part()
         args = []
         if self.try_op("("):
             if not self.at_op(")"):
@@ -535,7 +544,7 @@ class Parser:
         e = self._atom()
         while True:
             if self.try_op("."):
-                e = A.Prop(e, self.ident())
+                e = A.Prop(e, self.name_part())
             elif self.at_op("["):
                 self.next()
                 if self.try_op(".."):
@@ -587,7 +596,7 @@ class Parser:
             if t.value == "FALSE":
                 self.next()
                 return A.Lit(False)
-            if t.value == "COUNT":
+            if t.value == "COUNT" and self.peek(1).kind == "OP" and self.peek(1).value == "(":
                 self.next()
                 self.eat_op("(")
                 if self.try_op("*"):
@@ -621,6 +630,11 @@ class Parser:
                     wh = self._expr()
                     self.eat_op(")")
                     return A.Quantifier(kind, var, src, wh)
+            if t.value in ("COUNT", "ALL", "ANY", "NONE", "SINGLE", "INDEX",
+                           "UNIQUE", "DATABASE", "SHOW", "BY", "ON", "END"):
+                # soft keyword used as a plain variable name
+                self.next()
+                return A.Var(t.value.lower())
         if t.kind == "OP" and t.value == "(":
             # pattern predicate like (n)-[:R]->(m) in boolean context
             save = self.i
@@ -665,7 +679,7 @@ class Parser:
             if self.peek(1).kind == "OP" and self.peek(1).value == "(":
                 name = self.next().value
                 while self.try_op("."):
-                    name += "." + self.ident()
+                    name += "." + self.name_part()
                 # qualified name may end before '('
                 self.eat_op("(")
                 distinct = self.try_kw("DISTINCT")
@@ -681,7 +695,7 @@ class Parser:
                     and self._is_dotted_call()):
                 name = self.next().value
                 while self.try_op("."):
-                    name += "." + self.ident()
+                    name += "." + self.name_part()
                 self.eat_op("(")
                 distinct = self.try_kw("DISTINCT")
                 args = []

@@ -139,7 +139,10 @@ class NornicDB:
         self.dims = dims or self.embedder.dims
         self.search = SearchService(engine, dims=self.dims, device=device,
                                     embedder=self.embedder)
-        self.executor = Executor(engine, procedures=build_procedures(self))
+        from .apoc import build_apoc_procedures
+        procs = build_procedures(self)
+        procs.update(build_apoc_procedures(self))
+        self.executor = Executor(engine, procedures=procs)
         self.auto_embed = auto_embed
         self.embed_queue = EmbedQueue(self, workers=max(embed_workers, 1))
         if embed_workers > 0:
