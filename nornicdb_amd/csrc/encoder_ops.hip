@@ -170,18 +170,24 @@ __global__ void k_mean_pool_l2norm(const unsigned short* __restrict__ x,
 __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
     const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
-    int bh, int s, float scale) {
+    int n_heads, int s, float scale,
+    long long q_bs, long long q_ss,   // Q batch/seq strides (elements)
+    long long k_bs, long long k_ss,
+    long long v_bs, long long v_ss) {
+  // Layout: Q/K/V are [B, S, H, 64] views with arbitrary batch/seq strides
+  // (head stride == 64, last dim contiguous) so the qkv projection output
+  // feeds in with ZERO transposes; O is contiguous [B, S, H, 64].
   __shared__ unsigned short sK[FA_KT * FA_D];          // 8 KB, linear
   __shared__ unsigned short sVT[FA_D * FA_VSTRIDE];    // 9 KB, transposed
   __shared__ unsigned short sP[4][16 * FA_PSTRIDE];    // 4 x 2.25 KB
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  const int bh_i = blockIdx.x;
+  const int b = blockIdx.x / n_heads;
+  const int h = blockIdx.x % n_heads;
   const int q0 = blockIdx.y * FA_QT;
-  if (bh_i >= bh) return;
+  const long long hoff = (long long)h * FA_D;
 
-  const long long base = (long long)bh_i * s * FA_D;
   const int lq = lane & 15;           // row-in-16 for A-frags / col for C
   const int lk8 = (lane >> 4) * 8;    // k-offset for A/B frags
 
@@ -189,7 +195,8 @@ __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
   // A-frag for QK^T: lane holds Q[q=lq][d = lk8 + ks*32 .. +8], ks = 0,1
   bf16x8 qf[2];
   {
-    const unsigned short* qp = Q + base + (long long)(q0 + wid * 16 + lq) * FA_D;
+    const unsigned short* qp =
+        Q + b * q_bs + (long long)(q0 + wid * 16 + lq) * q_ss + hoff;
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks)
       qf[ks] = (bf16x8)(*reinterpret_cast<const short8v*>(qp + ks * 32 + lk8));
@@ -206,27 +213,31 @@ __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
   const float log2e = 1.44269504f;
 
   for (int k0 = 0; k0 < s; k0 += FA_KT) {
-    // ---- stage K tile linear: 8 KB contiguous; 8 chunks of 1 KB ----
+    // ---- stage K tile: 64 rows x 128 B (strided source, linear LDS) ----
     {
-      const unsigned short* kp = K + base + (long long)k0 * FA_D;
+      const unsigned short* kp = K + b * k_bs + hoff;
 #pragma unroll
       for (int it = 0; it < 2; ++it) {
         int chunk = wid * 2 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (FA_D * 2);
+        int cb = byte_off % (FA_D * 2);
         const G_AS unsigned int* gp = (const G_AS unsigned int*)(
-            (const char*)kp + chunk * 1024 + lane * 16);
+            (const char*)(kp + (long long)(k0 + r) * k_ss) + cb);
         L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sK + chunk * 1024);
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
     }
     // ---- stage V transposed: thread t covers k = t % 64, d-block t/64*8 --
     {
-      const unsigned short* vp = V + base + (long long)k0 * FA_D;
+      const unsigned short* vp = V + b * v_bs + hoff;
       int kk = threadIdx.x & 63;
       int db = (threadIdx.x >> 6) * 16;  // 4 groups x 16 d each
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
         int dd = db + half * 8;
-        short8v v = *reinterpret_cast<const short8v*>(vp + kk * FA_D + dd);
+        short8v v = *reinterpret_cast<const short8v*>(
+            vp + (long long)(k0 + kk) * v_ss + dd);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           sVT[(dd + j) * FA_VSTRIDE + kk] = (unsigned short)v[j];
@@ -316,15 +327,17 @@ __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
     __syncthreads();
   }
 
-  // ---- epilogue: normalize by l, write O [16q][64d] bf16 ----
-  const long long obase = base + (long long)(q0 + wid * 16) * FA_D;
+  // ---- epilogue: normalize by l, write O [B,S,H,64] contiguous ----
+  const long long o_ss = (long long)n_heads * FA_D;
+  const long long obase =
+      ((long long)b * s + q0 + wid * 16) * o_ss + hoff;
 #pragma unroll
   for (int nn = 0; nn < 4; ++nn) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = (lane >> 4) * 4 + r;
       float v = acc[nn][r] / fmaxf(l_run[r], 1e-20f);
-      O[obase + (long long)row * FA_D + nn * 16 + lq] = f32_to_bf16_bits(v);
+      O[obase + row * o_ss + nn * 16 + lq] = f32_to_bf16_bits(v);
     }
   }
 }
@@ -403,21 +416,29 @@ at::Tensor mean_pool_l2norm(at::Tensor x, c10::optional<at::Tensor> mask) {
 }
 
 at::Tensor flash_attn_nc(at::Tensor q, at::Tensor k, at::Tensor v) {
+  // q,k,v: [B, S, H, 64] bf16 VIEWS — head stride must be 64 and the last
+  // dim contiguous; batch/seq strides are free (so qkv.unbind(2) feeds in
+  // without any transpose/copy). Returns contiguous [B, S, H, 64].
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.scalar_type() == at::kBFloat16,
-              "flash_attn_nc: q [B,H,S,D] bf16");
-  int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2),
+              "flash_attn_nc: q [B,S,H,D] bf16");
+  int B = (int)q.size(0), S = (int)q.size(1), H = (int)q.size(2),
       D = (int)q.size(3);
   TORCH_CHECK(D == 64, "flash_attn_nc supports head_dim 64");
   TORCH_CHECK(S % 64 == 0, "flash_attn_nc needs S % 64 == 0");
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
-  at::Tensor o = at::empty_like(qc);
+  for (auto* t : {&q, &k, &v}) {
+    TORCH_CHECK(t->stride(3) == 1 && t->stride(2) == D,
+                "flash_attn_nc: head stride must be D, last dim contiguous");
+  }
+  at::Tensor o = at::empty({B, S, H, D}, q.options());
   float scale = 1.0f / sqrtf((float)D);
   dim3 grid(B * H, S / 64);
   hipLaunchKernelGGL(k_flash_attn_nc, grid, dim3(256), 0, enc_stream(),
-                     (const unsigned short*)qc.data_ptr(),
-                     (const unsigned short*)kc.data_ptr(),
-                     (const unsigned short*)vc.data_ptr(),
-                     (unsigned short*)o.data_ptr(), B * H, S, scale);
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (unsigned short*)o.data_ptr(), H, S, scale,
+                     q.stride(0), q.stride(1), k.stride(0), k.stride(1),
+                     v.stride(0), v.stride(1));
   HIP_CHECK_LAST();
   return o;
 }

@@ -71,14 +71,17 @@ class _EncoderLayer(nn.Module):
     def forward(self, x, attn_bias):
         b, s, h = x.shape
         qkv = self.qkv(x).view(b, s, 3, self.num_heads, self.head_dim)
-        q, k, v = qkv.unbind(dim=2)  # [b, s, nh, hd]
-        q = q.transpose(1, 2)
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
-        from nornicdb_amd.ops.attention import flash_attention_nc
+        q, k, v = qkv.unbind(dim=2)  # [b, s, nh, hd] strided views
+        from nornicdb_amd.ops.attention import (flash_attention_bshd,
+                                                flash_attention_nc)
 
-        a = flash_attention_nc(q, k, v, attn_bias)  # [b, nh, s, hd]
-        a = a.transpose(1, 2).reshape(b, s, h)
+        if attn_bias is None and _fused_ok(x):
+            # zero-transpose path: stride-aware HIP flash attention
+            a = flash_attention_bshd(q, k, v).reshape(b, s, h)
+        else:
+            a = flash_attention_nc(q.transpose(1, 2), k.transpose(1, 2),
+                                   v.transpose(1, 2), attn_bias)
+            a = a.transpose(1, 2).reshape(b, s, h)
         if _fused_ok(x):
             from nornicdb_amd.ops import encoder as eops
 
