@@ -1,0 +1,254 @@
+// Standalone ablation probe for the fused kNN MFMA kernel (no torch deps).
+// Compile: hipcc --offload-arch=gfx950 -O3 -std=c++17 knn_ablate.hip -o knn_ablate
+// Variants (template<int V>):
+//   0 = full kernel (stage A+B, MFMA, epilogue scan)
+//   1 = no epilogue scan (C still written to LDS, kept live)
+//   2 = no B staging (stale B tile)
+//   3 = no A staging (stale A tile)
+//   4 = MFMA only (no staging, no epilogue)
+//   5 = full + threshold-gated epilogue (skip column scan when the LDS
+//       chunk's max can't beat this thread's current kth best)
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+
+#define WAVE 64
+typedef short short8v __attribute__((ext_vector_type(8)));
+typedef float float4v __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+#define G_AS __attribute__((address_space(1)))
+#define L_AS __attribute__((address_space(3)))
+
+#define BM 64
+#define BN 256
+#define BK 64
+#define NT 256
+#define KC 12
+#define SCH 32
+#define SSTR (BN + 4)
+
+__global__ void fill_rand(unsigned short* x, long long n) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned long long h = i * 0x9e3779b97f4a7c15ULL;
+    h ^= h >> 33; h *= 0xff51afd7ed558ccdULL; h ^= h >> 33;
+    float f = ((float)(unsigned)(h & 0xffffff) / 8388608.0f - 1.0f) * 0.03f;
+    union { unsigned u; float ff; } v; v.ff = f;
+    unsigned r = v.u + 0x7fff + ((v.u >> 16) & 1);
+    x[i] = (unsigned short)(r >> 16);
+  }
+}
+
+template <int V>
+__global__ __launch_bounds__(NT, 3) void k_knn(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  __shared__ __align__(16) char smem[(BM * BK + BN * BK) * 2];
+  unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
+  float* sS = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[KC];
+  int ti[KC];
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM;
+    float4v acc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < d; kt += BK) {
+      if (V != 3 && V != 4) {
+#pragma unroll
+        for (int it = 0; it < 2; ++it) {
+          int chunk = wc * 2 + it;
+          int byte_off = chunk * 1024 + lane * 16;
+          int r = byte_off / (BK * 2);
+          int cb = byte_off % (BK * 2);
+          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+              (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+          L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+      }
+      if (V != 2 && V != 4) {
+#pragma unroll
+        for (int it = 0; it < 8; ++it) {
+          int chunk = wc * 8 + it;
+          int byte_off = chunk * 1024 + lane * 16;
+          int r = byte_off / (BK * 2);
+          int cb = byte_off % (BK * 2);
+          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+              (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+          L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+      }
+      __syncthreads();
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = m * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(sA + r * BK + k));
+        }
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(sB + c * BK + k));
+        }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+
+    if (V == 1 || V == 4) {
+      // keep acc live without the epilogue
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+          asm volatile("" ::"v"(acc[m][nn][0]), "v"(acc[m][nn][3]));
+      continue;
+    }
+
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int srow = mi * 16 + (lane >> 4) * 4 + r;
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            sS[srow * SSTR + col] = acc[m][nn][r];
+          }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * SCH;
+      bool scan = true;
+      if (V == 5) {
+        // wave-cooperative chunk max: each lane maxes its column set
+        float cmax = -1e30f;
+        for (int r = 0; r < SCH; ++r)
+          cmax = fmaxf(cmax, sS[r * SSTR + tid]);
+        scan = cmax > tv[KC - 1];
+        if (scan) {
+#pragma unroll
+          for (int r = 0; r < SCH; ++r) {
+            float s = sS[r * SSTR + tid];
+            if (s > tv[KC - 1]) {
+              float cs = s; int ci = (int)(grow0 + r);
+#pragma unroll
+              for (int i = 0; i < KC; ++i) {
+                bool ins = cs > tv[i];
+                float ts2 = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < SCH; ++r) {
+          float s = sS[r * SSTR + tid];
+          if (s > tv[KC - 1]) {
+            float cs = s; int ci = (int)(grow0 + r);
+#pragma unroll
+            for (int i = 0; i < KC; ++i) {
+              bool ins = cs > tv[i];
+              float ts2 = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * BN + tid;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[slot * KC + i] = tv[i];
+    cand_idx[slot * KC + i] = ti[i];
+  }
+}
+
+template <int V>
+float run(const unsigned short* db, const unsigned short* qs, long long n,
+          int d, float* cs, int* ci, int iters) {
+  long long panels = n / BM;
+  int grid = (int)std::min<long long>(panels, 2048);
+  // warmup
+  hipLaunchKernelGGL((k_knn<V>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn<V>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+  hipEventRecord(t1);
+  hipEventSynchronize(t1);
+  float ms;
+  hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+int main() {
+  const long long n = 4 << 20;
+  const int d = 1024;
+  unsigned short *db, *qs;
+  float* cs;
+  int* ci;
+  hipMalloc(&db, n * d * 2);
+  hipMalloc(&qs, (long long)BN * d * 2);
+  hipMalloc(&cs, 2048LL * BN * KC * 4);
+  hipMalloc(&ci, 2048LL * BN * KC * 4);
+  hipLaunchKernelGGL(fill_rand, dim3(4096), dim3(256), 0, 0, db, n * d);
+  hipLaunchKernelGGL(fill_rand, dim3(64), dim3(256), 0, 0, qs, (long long)BN * d);
+  hipDeviceSynchronize();
+  double flops = 2.0 * n * d * BN;
+  double bytes = (double)n * d * 2;
+  const char* names[] = {"full", "no-epilogue", "no-B-stage", "no-A-stage",
+                         "mfma-only", "gated-epilogue"};
+  float ms;
+  ms = run<0>(db, qs, n, d, cs, ci, 10);
+  printf("V0 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[0], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<1>(db, qs, n, d, cs, ci, 10);
+  printf("V1 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[1], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<2>(db, qs, n, d, cs, ci, 10);
+  printf("V2 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[2], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<3>(db, qs, n, d, cs, ci, 10);
+  printf("V3 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[3], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<4>(db, qs, n, d, cs, ci, 10);
+  printf("V4 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[4], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<5>(db, qs, n, d, cs, ci, 10);
+  printf("V5 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[5], ms, flops/ms/1e9, bytes/ms/1e9);
+  return 0;
+}
