@@ -8,6 +8,10 @@
 //   4 = MFMA only (no staging, no epilogue)
 //   5 = full + threshold-gated epilogue (skip column scan when the LDS
 //       chunk's max can't beat this thread's current kth best)
+//   6 = full + TRANSPOSED epilogue chunk [col][row] -> float4 writes and
+//       contiguous float4 scans per thread
+//   7 = V6 + double-buffered LDS 2-phase loop (stage next tile before MFMA,
+//       counted barrier per K-step)
 #include <hip/hip_runtime.h>
 #include <cstdio>
 #include <cstdlib>
@@ -27,6 +31,7 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define KC 12
 #define SCH 32
 #define SSTR (BN + 4)
+#define TSTR 36   // transposed chunk row stride (32 + 4 pad)
 
 __global__ void fill_rand(unsigned short* x, long long n) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -46,7 +51,9 @@ __global__ __launch_bounds__(NT, 3) void k_knn(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
     long long n_panels, int d, float* __restrict__ cand_score,
     int* __restrict__ cand_idx) {
-  __shared__ __align__(16) char smem[(BM * BK + BN * BK) * 2];
+  constexpr bool DBUF = (V == 7);
+  __shared__ __align__(16) char smem[DBUF ? (BM * BK + BN * BK) * 4
+                                          : (BM * BK + BN * BK) * 2];
   unsigned short* sA = (unsigned short*)smem;
   unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
   float* sS = (float*)smem;
@@ -70,6 +77,75 @@ __global__ __launch_bounds__(NT, 3) void k_knn(
 #pragma unroll
       for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
 
+    auto stage = [&](int kt, int buf) {
+      unsigned short* dA = sA + buf * (BM * BK + BN * BK);
+      unsigned short* dB = sB + buf * (BM * BK + BN * BK);
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int chunk = wc * 2 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)dA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {
+        int chunk = wc * 8 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)dB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+    };
+    auto mfma_tile = [&](int buf) {
+      const unsigned short* rA = sA + buf * (BM * BK + BN * BK);
+      const unsigned short* rB = sB + buf * (BM * BK + BN * BK);
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = m * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(rA + r * BK + k));
+        }
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(rB + c * BK + k));
+        }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+      }
+    };
+
+    if (DBUF) {
+      int nt = d / BK;
+      stage(0, 0);
+      asm volatile("s_waitcnt vmcnt(0)");
+      __syncthreads();
+      int cur = 0;
+      for (int t = 0; t < nt - 1; ++t) {
+        stage((t + 1) * BK, cur ^ 1);
+        mfma_tile(cur);
+        asm volatile("s_waitcnt vmcnt(0)");
+        __syncthreads();
+        cur ^= 1;
+      }
+      mfma_tile(cur);
+      __syncthreads();
+    } else
     for (int kt = 0; kt < d; kt += BK) {
       if (V != 3 && V != 4) {
 #pragma unroll
@@ -130,6 +206,46 @@ __global__ __launch_bounds__(NT, 3) void k_knn(
 #pragma unroll
         for (int nn = 0; nn < 4; ++nn)
           asm volatile("" ::"v"(acc[m][nn][0]), "v"(acc[m][nn][3]));
+      continue;
+    }
+
+    if (V == 6 || V == 7) {
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int m = h * 2 + mi;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            int s0 = mi * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sS + col * TSTR + s0) = acc[m][nn];
+          }
+        }
+        __syncthreads();
+        const long long grow0 = prow + (long long)h * SCH;
+        const float* myrow = sS + tid * TSTR;
+#pragma unroll
+        for (int rb = 0; rb < SCH / 4; ++rb) {
+          float4v v4 = *reinterpret_cast<const float4v*>(myrow + rb * 4);
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            float s = v4[j];
+            if (s > tv[KC - 1]) {
+              float cs = s; int ci = (int)(grow0 + rb * 4 + j);
+#pragma unroll
+              for (int i = 0; i < KC; ++i) {
+                bool ins = cs > tv[i];
+                float ts2 = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+        __syncthreads();
+      }
+      long long slot0 = (long long)blockIdx.x * BN + tid;
       continue;
     }
 
@@ -250,5 +366,9 @@ int main() {
   printf("V4 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[4], ms, flops/ms/1e9, bytes/ms/1e9);
   ms = run<5>(db, qs, n, d, cs, ci, 10);
   printf("V5 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", names[5], ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<6>(db, qs, n, d, cs, ci, 10);
+  printf("V6 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "transposed-epi", ms, flops/ms/1e9, bytes/ms/1e9);
+  ms = run<7>(db, qs, n, d, cs, ci, 10);
+  printf("V7 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "v6+dbuf-2ph", ms, flops/ms/1e9, bytes/ms/1e9);
   return 0;
 }
