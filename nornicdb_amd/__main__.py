@@ -1,0 +1,190 @@
+"""NornicDB-AMD command-line interface.
+
+Parity: reference cmd/nornicdb/main.go cobra commands:
+    serve | init | import | shell | decay  (main.go:71-210)
+
+Usage:
+    python -m nornicdb_amd serve [--data-dir DIR] [--config FILE]
+    python -m nornicdb_amd init --data-dir DIR
+    python -m nornicdb_amd import --data-dir DIR --file export.json
+    python -m nornicdb_amd shell [--data-dir DIR]
+    python -m nornicdb_amd decay --data-dir DIR
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import sys
+
+
+def _open(args, cfg):
+    from .db import open_db
+    from .embed import create_embedder
+
+    provider = "mock" if cfg.embedder == "mock" else cfg.embedder
+    try:
+        emb = create_embedder(provider, dims=cfg.embedding_dims)
+    except Exception as e:
+        print(f"embedder '{provider}' unavailable ({e}); using mock",
+              file=sys.stderr)
+        emb = create_embedder("mock", dims=cfg.embedding_dims)
+    return open_db(args.data_dir or cfg.data_dir or None, embedder=emb,
+                   dims=cfg.embedding_dims)
+
+
+def cmd_serve(args):
+    from .bolt import BoltServer
+    from .server import create_app
+    from .auth import Authenticator
+    from .utils import load_config
+    import uvicorn
+
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    auth = None
+    if cfg.auth_enabled or args.auth:
+        auth = Authenticator(mgr.get("system").engine)
+        pw = auth.ensure_admin(password=cfg.initial_admin_password or None)
+        if pw:
+            print(f"initial admin user 'neo4j' password: {pw}")
+
+    app = create_app(mgr, auth=auth)
+    bolt = BoltServer(lambda db: mgr.get(db).executor,
+                      host=cfg.bolt_host, port=args.bolt_port or cfg.bolt_port,
+                      authenticator=auth)
+
+    async def main():
+        await bolt.start()
+        print(f"Bolt listening on {cfg.bolt_host}:{bolt.port}")
+        config = uvicorn.Config(app, host=cfg.http_host,
+                                port=args.http_port or cfg.http_port,
+                                log_level="warning")
+        server = uvicorn.Server(config)
+        print(f"HTTP listening on {cfg.http_host}:{config.port}")
+        await server.serve()
+
+    try:
+        asyncio.run(main())
+    except KeyboardInterrupt:
+        pass
+    finally:
+        mgr.close()
+
+
+def cmd_init(args):
+    from .utils import load_config
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    db = mgr.get()
+    print(f"initialized database at {args.data_dir or '(memory)'}: "
+          f"{db.engine.node_count()} nodes")
+    mgr.close()
+
+
+def cmd_import(args):
+    from .utils import load_config
+    from .storage import Edge, Node
+
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    eng = mgr.get(args.database).engine
+    with open(args.file) as f:
+        data = json.load(f)
+    n_nodes = n_edges = 0
+    for nd in data.get("nodes", []):
+        try:
+            eng.create_node(Node(id=str(nd.get("id", n_nodes)),
+                                 labels=nd.get("labels", []),
+                                 properties=nd.get("properties", {})))
+            n_nodes += 1
+        except Exception:
+            pass
+    for ed in data.get("relationships", data.get("edges", [])):
+        try:
+            eng.create_edge(Edge(id=str(ed.get("id", f"e{n_edges}")),
+                                 type=ed.get("type", "RELATED"),
+                                 start_node=str(ed.get("start", ed.get("startNode"))),
+                                 end_node=str(ed.get("end", ed.get("endNode"))),
+                                 properties=ed.get("properties", {})))
+            n_edges += 1
+        except Exception:
+            pass
+    print(f"imported {n_nodes} nodes, {n_edges} relationships")
+    mgr.close()
+
+
+def cmd_shell(args):
+    from .utils import load_config
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    db = mgr.get(args.database)
+    print("NornicDB-AMD shell — Cypher queries; :quit to exit")
+    while True:
+        try:
+            line = input("nornicdb> ").strip()
+        except (EOFError, KeyboardInterrupt):
+            break
+        if not line:
+            continue
+        if line in (":quit", ":exit", "quit", "exit"):
+            break
+        try:
+            r = db.cypher(line)
+            print("\t".join(r.columns))
+            for row in r.rows[:100]:
+                print("\t".join(str(v) for v in row))
+            if len(r.rows) > 100:
+                print(f"... {len(r.rows) - 100} more rows")
+        except Exception as e:
+            print(f"error: {e}")
+    mgr.close()
+
+
+def cmd_decay(args):
+    from .cognitive import DecayManager
+    from .utils import load_config
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    dm = DecayManager(mgr.get(args.database).engine)
+    stats = dm.run_cycle()
+    print(json.dumps(stats))
+    mgr.close()
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(prog="nornicdb-amd")
+    p.add_argument("--config", default=None)
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    sp = sub.add_parser("serve")
+    sp.add_argument("--data-dir", default=None)
+    sp.add_argument("--bolt-port", type=int, default=None)
+    sp.add_argument("--http-port", type=int, default=None)
+    sp.add_argument("--auth", action="store_true")
+    sp.set_defaults(fn=cmd_serve)
+
+    for name, fn in (("init", cmd_init), ("decay", cmd_decay)):
+        sp = sub.add_parser(name)
+        sp.add_argument("--data-dir", default=None)
+        sp.add_argument("--database", default=None)
+        sp.set_defaults(fn=fn)
+
+    sp = sub.add_parser("import")
+    sp.add_argument("--data-dir", default=None)
+    sp.add_argument("--database", default=None)
+    sp.add_argument("--file", required=True)
+    sp.set_defaults(fn=cmd_import)
+
+    sp = sub.add_parser("shell")
+    sp.add_argument("--data-dir", default=None)
+    sp.add_argument("--database", default=None)
+    sp.set_defaults(fn=cmd_shell)
+
+    args = p.parse_args(argv)
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()

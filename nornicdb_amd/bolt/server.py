@@ -334,6 +334,8 @@ class BoltServer:
     async def start(self):
         self._server = await asyncio.start_server(
             self._on_conn, self.host, self.port)
+        if self.port == 0 and self._server.sockets:
+            self.port = self._server.sockets[0].getsockname()[1]
         return self
 
     async def _on_conn(self, reader, writer):

@@ -46,6 +46,340 @@ __global__ void fill_rand(unsigned short* x, long long n) {
   }
 }
 
+// V8: B-fragments read directly from global (L2-resident queries, no sB);
+//     transposed epilogue. V9 = V8 + tile-max gate on the epilogue.
+template <int V>
+__global__ __launch_bounds__(NT, 3) void k_knn_direct(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  __shared__ __align__(16) char smem[48 * 1024];
+  unsigned short* sA = (unsigned short*)smem;
+  float* sS = (float*)smem + (8 * 1024) / 4 * 2;  // after 8KB A tile... bytes!
+  sS = (float*)(smem + 8 * 1024);
+  __shared__ float s_minthr[1];
+  __shared__ float s_red[4];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[KC];
+  int ti[KC];
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+  if (tid == 0) s_minthr[0] = -1e30f;
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM;
+    float4v acc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < d; kt += BK) {
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int chunk = wc * 2 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+      __syncthreads();
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = m * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(sA + r * BK + k));
+        }
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          int k = kt + ks * 32 + (lane >> 4) * 8;
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(qs + (long long)c * d + k));
+        }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+
+    bool do_scan = true;
+    if (V == 9) {
+      float vmax = -1e30f;
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) vmax = fmaxf(vmax, acc[m][nn][r]);
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1)
+        vmax = fmaxf(vmax, __shfl_xor(vmax, off, WAVE));
+      if (lane == 0) s_red[wc] = vmax;
+      __syncthreads();
+      float tmax = fmaxf(fmaxf(s_red[0], s_red[1]), fmaxf(s_red[2], s_red[3]));
+      do_scan = tmax >= s_minthr[0];
+    }
+
+    if (do_scan) {
+      float newthr = 1e30f;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int m = h * 2 + mi;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            int s0 = mi * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sS + col * TSTR + s0) = acc[m][nn];
+          }
+        }
+        __syncthreads();
+        const long long grow0 = prow + (long long)h * SCH;
+        const float* myrow = sS + tid * TSTR;
+#pragma unroll
+        for (int rb = 0; rb < SCH / 4; ++rb) {
+          float4v v4 = *reinterpret_cast<const float4v*>(myrow + rb * 4);
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            float s = v4[j];
+            if (s > tv[KC - 1]) {
+              float cs = s; int ci = (int)(grow0 + rb * 4 + j);
+#pragma unroll
+              for (int i = 0; i < KC; ++i) {
+                bool ins = cs > tv[i];
+                float ts2 = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+        __syncthreads();
+      }
+      if (V == 9) {
+        // refresh block-min threshold: min over threads of tv[KC-1]
+        float myth = tv[KC - 1];
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1)
+          myth = fminf(myth, __shfl_xor(myth, off, WAVE));
+        if (lane == 0) s_red[wc] = myth;
+        __syncthreads();
+        if (tid == 0)
+          s_minthr[0] = fminf(fminf(s_red[0], s_red[1]),
+                              fminf(s_red[2], s_red[3]));
+        __syncthreads();
+      }
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * BN + tid;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[slot * KC + i] = tv[i];
+    cand_idx[slot * KC + i] = ti[i];
+  }
+}
+
+// V10/V11: counted-vmcnt pipeline (T3/T4): raw s_barrier (no forced
+// vmcnt-0 drain) + asm s_waitcnt vmcnt(N) so next-tile loads stay in
+// flight across barriers. V10 = A+B double-buffered (80 KB LDS,
+// 2 blocks/CU); V11 = A-only double-buffer (48 KB, 3 blocks/CU),
+// B staged synchronously per K-step.
+#define RAW_BAR() do { __builtin_amdgcn_s_barrier(); \
+                       __builtin_amdgcn_sched_barrier(0); } while (0)
+#define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+
+template <int V>
+__global__ __launch_bounds__(NT, 3) void k_knn_pipe(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  constexpr bool FULL_DBUF = (V == 10);
+  __shared__ __align__(16) char smem[FULL_DBUF ? 80 * 1024 : 48 * 1024];
+  // layout: V10: A0 8K | B0 32K | A1 8K | B1 32K
+  //         V11: A0 8K | A1 8K | B 32K
+  unsigned short* A0 = (unsigned short*)smem;
+  unsigned short* B0 = (unsigned short*)(smem + (FULL_DBUF ? 8192 : 16384));
+  unsigned short* A1 = (unsigned short*)(smem + (FULL_DBUF ? 40960 : 8192));
+  unsigned short* B1 = FULL_DBUF ? (unsigned short*)(smem + 49152) : B0;
+  float* sS = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[KC];
+  int ti[KC];
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM;
+    float4v acc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+#define STAGE_A(kt, dst) do { \
+      _Pragma("unroll") \
+      for (int it = 0; it < 2; ++it) { \
+        int chunk = wc * 2 + it; \
+        int byte_off = chunk * 1024 + lane * 16; \
+        int r = byte_off / (BK * 2); \
+        int cb = byte_off % (BK * 2); \
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)( \
+            (const char*)db + (prow + r) * d2 + (long long)(kt) * 2 + cb); \
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)(dst) + chunk * 1024); \
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0); \
+      } } while (0)
+#define STAGE_B(kt, dst) do { \
+      _Pragma("unroll") \
+      for (int it = 0; it < 8; ++it) { \
+        int chunk = wc * 8 + it; \
+        int byte_off = chunk * 1024 + lane * 16; \
+        int r = byte_off / (BK * 2); \
+        int cb = byte_off % (BK * 2); \
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)( \
+            (const char*)qs + (long long)r * d2 + (long long)(kt) * 2 + cb); \
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)(dst) + chunk * 1024); \
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0); \
+      } } while (0)
+#define MFMA_TILE(rA, rB) do { \
+      _Pragma("unroll") \
+      for (int ks = 0; ks < BK / 32; ++ks) { \
+        bf16x8 af[4], bf[4]; \
+        _Pragma("unroll") \
+        for (int m = 0; m < 4; ++m) { \
+          int r = m * 16 + (lane & 15); \
+          int k = ks * 32 + (lane >> 4) * 8; \
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>((rA) + r * BK + k)); \
+        } \
+        _Pragma("unroll") \
+        for (int nn = 0; nn < 4; ++nn) { \
+          int c = wc * 64 + nn * 16 + (lane & 15); \
+          int k = ks * 32 + (lane >> 4) * 8; \
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>((rB) + c * BK + k)); \
+        } \
+        _Pragma("unroll") \
+        for (int m = 0; m < 4; ++m) \
+          _Pragma("unroll") \
+          for (int nn = 0; nn < 4; ++nn) \
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16( \
+                af[m], bf[nn], acc[m][nn], 0, 0, 0); \
+      } } while (0)
+
+    const int nt = d / BK;  // assume even (d % 128 == 0)
+    if (FULL_DBUF) {
+      STAGE_A(0, A0); STAGE_B(0, B0);
+      for (int t = 0; t < nt; t += 2) {
+        if (t + 1 < nt) { STAGE_A((t + 1) * BK, A1); STAGE_B((t + 1) * BK, B1); }
+        VMCNT(10);          // tile t landed; tile t+1 (10 loads) in flight
+        RAW_BAR();
+        MFMA_TILE(A0, B0);
+        RAW_BAR();
+        if (t + 2 < nt) {
+          STAGE_A((t + 2) * BK, A0); STAGE_B((t + 2) * BK, B0);
+          VMCNT(10);
+        } else {
+          VMCNT(0);
+        }
+        RAW_BAR();
+        MFMA_TILE(A1, B1);
+        RAW_BAR();
+      }
+    } else {
+      STAGE_A(0, A0);
+      for (int t = 0; t < nt; t += 2) {
+        if (t + 1 < nt) STAGE_A((t + 1) * BK, A1);  // 2 loads, stays in flight
+        STAGE_B(t * BK, B0);                        // 8 loads (L2-hot)
+        VMCNT(2);           // drain A(t)+B(t); A(t+1)'s 2 keep flying
+        RAW_BAR();
+        MFMA_TILE(A0, B0);
+        RAW_BAR();
+        STAGE_B((t + 1) * BK, B0);
+        if (t + 2 < nt) {
+          STAGE_A((t + 2) * BK, A0);
+          VMCNT(2);
+        } else {
+          VMCNT(0);
+        }
+        RAW_BAR();
+        MFMA_TILE(A1, B0);
+        RAW_BAR();
+      }
+    }
+    VMCNT(0);
+    __syncthreads();
+
+    // transposed epilogue (V6 form)
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int col = wc * 64 + nn * 16 + (lane & 15);
+          int s0 = mi * 16 + (lane >> 4) * 4;
+          *reinterpret_cast<float4v*>(sS + col * TSTR + s0) = acc[m][nn];
+        }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * SCH;
+      const float* myrow = sS + tid * TSTR;
+#pragma unroll
+      for (int rb = 0; rb < SCH / 4; ++rb) {
+        float4v v4 = *reinterpret_cast<const float4v*>(myrow + rb * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float s = v4[j];
+          if (s > tv[KC - 1]) {
+            float cs = s; int ci = (int)(grow0 + rb * 4 + j);
+#pragma unroll
+            for (int i = 0; i < KC; ++i) {
+              bool ins = cs > tv[i];
+              float ts2 = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * BN + tid;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[slot * KC + i] = tv[i];
+    cand_idx[slot * KC + i] = ti[i];
+  }
+#undef STAGE_A
+#undef STAGE_B
+#undef MFMA_TILE
+}
+
 template <int V>
 __global__ __launch_bounds__(NT, 3) void k_knn(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
@@ -370,5 +704,42 @@ int main() {
   printf("V6 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "transposed-epi", ms, flops/ms/1e9, bytes/ms/1e9);
   ms = run<7>(db, qs, n, d, cs, ci, 10);
   printf("V7 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "v6+dbuf-2ph", ms, flops/ms/1e9, bytes/ms/1e9);
+  {
+    long long panels = n / BM;
+    int grid = (int)std::min<long long>(panels, 2048);
+    hipLaunchKernelGGL((k_knn_direct<8>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipDeviceSynchronize();
+    hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+    hipEventRecord(t0);
+    for (int i = 0; i < 10; ++i)
+      hipLaunchKernelGGL((k_knn_direct<8>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipEventRecord(t1); hipEventSynchronize(t1);
+    hipEventElapsedTime(&ms, t0, t1); ms /= 10;
+    printf("V8 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "B-direct-L2", ms, flops/ms/1e9, bytes/ms/1e9);
+    hipLaunchKernelGGL((k_knn_direct<9>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipDeviceSynchronize();
+    hipEventRecord(t0);
+    for (int i = 0; i < 10; ++i)
+      hipLaunchKernelGGL((k_knn_direct<9>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipEventRecord(t1); hipEventSynchronize(t1);
+    hipEventElapsedTime(&ms, t0, t1); ms /= 10;
+    printf("V9 %-15s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "V8+tilemax-gate", ms, flops/ms/1e9, bytes/ms/1e9);
+    hipLaunchKernelGGL((k_knn_pipe<10>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipDeviceSynchronize();
+    hipEventRecord(t0);
+    for (int i = 0; i < 10; ++i)
+      hipLaunchKernelGGL((k_knn_pipe<10>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipEventRecord(t1); hipEventSynchronize(t1);
+    hipEventElapsedTime(&ms, t0, t1); ms /= 10;
+    printf("V10 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "pipe-full-dbuf", ms, flops/ms/1e9, bytes/ms/1e9);
+    hipLaunchKernelGGL((k_knn_pipe<11>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipDeviceSynchronize();
+    hipEventRecord(t0);
+    for (int i = 0; i < 10; ++i)
+      hipLaunchKernelGGL((k_knn_pipe<11>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+    hipEventRecord(t1); hipEventSynchronize(t1);
+    hipEventElapsedTime(&ms, t0, t1); ms /= 10;
+    printf("V11 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "pipe-A-dbuf", ms, flops/ms/1e9, bytes/ms/1e9);
+  }
   return 0;
 }
