@@ -125,6 +125,8 @@ def main():
     p.add_argument("--layers", type=int, default=24)
     p.add_argument("--skip-recall", action="store_true")
     p.add_argument("--timing", action="store_true", help="print phase breakdown")
+    p.add_argument("--no-pipeline", action="store_true",
+                   help="disable embed/search stream overlap (sequential steps)")
     p.add_argument("--recall-queries", type=int, default=8)
     p.add_argument("--device", default=None)
     args = p.parse_args()
@@ -181,24 +183,48 @@ def main():
         args.dim = h
         shard = shard[:, :h].contiguous() if shard.shape[1] >= h else torch.randn(shard_rows, h)
 
-    def one_step():
+    def embed_batch():
         tokens = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
         with torch.no_grad():
             q = model(tokens)  # [B, H] fp32 normalized
-        q = q.to(shard.dtype)
+        return q.to(shard.dtype)
+
+    def gather_queries(q):
         if world > 1:
             gathered = [torch.empty_like(q) for _ in range(world)]
             dist.all_gather(gathered, q.contiguous())
-            qg = torch.cat(gathered, 0)
-        else:
-            qg = q
+            return torch.cat(gathered, 0)
+        return q
+
+    def one_step():
+        qg = gather_queries(embed_batch())
         s, i = search_step(shard, qg, args.k, row_base, world)
         return qg, s, i
 
+    # serving pipeline: embed batch t on one HIP stream while batch t-1 is
+    # searched on another. Every query is still embedded AND searched; the
+    # pipeline just has one batch of latency (steady-state serving shape).
+    pipelined = device.type == "cuda" and not args.no_pipeline
+    if pipelined:
+        s_embed = torch.cuda.Stream()
+        s_search = torch.cuda.Stream()
+
+        def pipe_step(prev_q):
+            with torch.cuda.stream(s_embed):
+                q_new = embed_batch()
+            results = [None]
+            with torch.cuda.stream(s_search):
+                results[0] = search_step(shard, prev_q, args.k, row_base, world)
+            torch.cuda.synchronize()
+            return gather_queries(q_new), results[0]
+
     # --- warmup ---
     log(rank, f"[bench] warmup x{args.warmup} ...")
+    last_q = None
     for _ in range(args.warmup):
-        one_step()
+        last_q, _, _ = one_step()
+    if pipelined:
+        last_q, _ = pipe_step(last_q)  # warm both streams
     if device.type == "cuda":
         torch.cuda.synchronize()
 
@@ -247,8 +273,12 @@ def main():
     if device.type == "cuda":
         torch.cuda.synchronize()
     t_start = time.time()
-    for _ in range(args.steps):
-        one_step()
+    if pipelined:
+        for _ in range(args.steps):
+            last_q, _ = pipe_step(last_q)
+    else:
+        for _ in range(args.steps):
+            one_step()
     if device.type == "cuda":
         torch.cuda.synchronize()
     if world > 1:
@@ -284,7 +314,9 @@ def main():
                 "model": "bge-m3 (XLM-R-large shape: 24L/1024h/16heads, random init)",
                 "global_batch": args.batch * world,
                 "seq_len": args.seq_len,
-                "parallelism": f"shard{world} (corpus row-sharded, RCCL all-gather top-k merge)",
+                "parallelism": f"shard{world} (corpus row-sharded, RCCL all-gather top-k merge)"
+                               + ("; embed/search HIP-stream pipelined (1-batch latency)"
+                                  if pipelined else ""),
                 "corpus_vectors": args.corpus,
                 "dim": args.dim,
                 "k": args.k,
