@@ -269,4 +269,18 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         body = await request.json()
         return mcp.handle(body)
 
+    # ---- GraphQL (reference pkg/graphql) ----
+    from .graphql import GraphQLExecutor
+
+    @app.post("/graphql")
+    async def graphql_endpoint(request: Request, db: str = None):
+        body = await request.json()
+        gql = GraphQLExecutor(mgr.get(db))
+        return gql.execute(body.get("query", ""), body.get("variables"))
+
+    # ---- Qdrant-compatible REST (reference pkg/qdrantgrpc) ----
+    from .qdrant import QdrantRegistry, qdrant_router
+    app.state.qdrant = QdrantRegistry()
+    app.include_router(qdrant_router(app.state.qdrant))
+
     return app

@@ -1,0 +1,91 @@
+"""Qdrant REST compat + GraphQL endpoint tests."""
+
+import pytest
+from fastapi.testclient import TestClient
+
+from nornicdb_amd.db import open_db
+from nornicdb_amd.embed import MockEmbedder
+from nornicdb_amd.server import create_app
+
+
+@pytest.fixture
+def client():
+    mgr = open_db(embedder=MockEmbedder(16), dims=16)
+    with TestClient(create_app(mgr)) as c:
+        yield c
+    mgr.close()
+
+
+class TestQdrant:
+    def test_collection_lifecycle(self, client):
+        r = client.put("/collections/docs", json={
+            "vectors": {"size": 4, "distance": "Cosine"}})
+        assert r.json()["status"] == "ok"
+        assert client.put("/collections/docs", json={
+            "vectors": {"size": 4}}).status_code == 409
+        got = client.get("/collections/docs").json()["result"]
+        assert got["config"]["params"]["vectors"]["size"] == 4
+        names = [c["name"] for c in
+                 client.get("/collections").json()["result"]["collections"]]
+        assert "docs" in names
+
+    def test_upsert_search_delete(self, client):
+        client.put("/collections/c1", json={"vectors": {"size": 4,
+                                                        "distance": "Cosine"}})
+        pts = [{"id": i, "vector": v, "payload": {"tag": f"t{i}"}}
+               for i, v in [(1, [1, 0, 0, 0]), (2, [0, 1, 0, 0]),
+                            (3, [0.9, 0.1, 0, 0])]]
+        r = client.put("/collections/c1/points", json={"points": pts})
+        assert r.json()["result"]["status"] == "completed"
+        res = client.post("/collections/c1/points/search",
+                          json={"vector": [1, 0, 0, 0], "limit": 2}).json()["result"]
+        assert res[0]["id"] == 1 and res[1]["id"] == 3
+        assert res[0]["payload"]["tag"] == "t1"
+        client.post("/collections/c1/points/delete", json={"points": [1]})
+        res = client.post("/collections/c1/points/search",
+                          json={"vector": [1, 0, 0, 0], "limit": 2}).json()["result"]
+        assert res[0]["id"] == 3
+
+    def test_scroll(self, client):
+        client.put("/collections/c2", json={"vectors": {"size": 2}})
+        client.put("/collections/c2/points", json={"points": [
+            {"id": i, "vector": [i, 0], "payload": {}} for i in range(5)]})
+        page = client.post("/collections/c2/points/scroll",
+                           json={"limit": 3}).json()["result"]
+        assert len(page["points"]) == 3
+        assert page["next_page_offset"] is not None
+
+
+class TestGraphQL:
+    def _gql(self, client, q):
+        return client.post("/graphql", json={"query": q}).json()
+
+    def test_nodes_query(self, client):
+        mgr = client.app.state.manager
+        mgr.get().cypher("CREATE (:Person {name: 'Ada'}), (:Person {name: 'Bob'})")
+        r = self._gql(client, '{ nodes(label: "Person") { id properties } }')
+        assert len(r["data"]["nodes"]) == 2
+        names = {n["properties"]["name"] for n in r["data"]["nodes"]}
+        assert names == {"Ada", "Bob"}
+
+    def test_node_with_relationships(self, client):
+        db = client.app.state.manager.get()
+        db.cypher("CREATE (a:G {name:'a'})-[:KNOWS]->(b:G {name:'b'})")
+        nid = db.cypher("MATCH (a:G {name:'a'}) RETURN a").rows[0][0].id
+        r = self._gql(client,
+                      '{ node(id: "%s") { id relationships { type } } }' % nid)
+        assert r["data"]["node"]["relationships"][0]["type"] == "KNOWS"
+
+    def test_mutation_create(self, client):
+        r = self._gql(client,
+                      'mutation { createNode(labels: ["M"], '
+                      'properties: "{\\"v\\": 7}") { id properties } }')
+        assert r["data"]["createNode"]["properties"]["v"] == 7
+
+    def test_cypher_field(self, client):
+        r = self._gql(client, '{ cypher(query: "RETURN 1 + 1") { columns rows } }')
+        assert r["data"]["cypher"]["rows"] == [[2]]
+
+    def test_error_reported(self, client):
+        r = self._gql(client, '{ nosuch }')
+        assert "errors" in r
