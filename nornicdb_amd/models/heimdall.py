@@ -1,0 +1,179 @@
+"""Heimdall generative model: Qwen2-style causal decoder, MI355X-native.
+
+Replaces the reference's llama.cpp Qwen-2.5-0.5B GGUF generation path
+(reference pkg/localllm/llama.go:195-240, pkg/heimdall/generator_cgo.go)
+with a from-scratch PyTorch-ROCm decoder: RMSNorm, RoPE, GQA attention
+with KV cache, SwiGLU MLP. Weights are random-init in this offline
+environment (a state_dict can be loaded when available); the contract is
+architecture + decode-throughput parity.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class HeimdallConfig:
+    # Qwen-2.5-0.5B shape
+    vocab_size: int = 151_936
+    hidden_size: int = 896
+    num_layers: int = 24
+    num_heads: int = 14
+    num_kv_heads: int = 2
+    intermediate_size: int = 4864
+    max_position: int = 4096
+    rope_theta: float = 1_000_000.0
+    rms_eps: float = 1e-6
+
+    @classmethod
+    def tiny(cls, **kw):
+        d = dict(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, intermediate_size=128, max_position=256)
+        d.update(kw)
+        return cls(**d)
+
+
+def _rope(x, cos, sin):
+    # x: [B, H, S, D]
+    d = x.shape[-1]
+    x1, x2 = x[..., : d // 2], x[..., d // 2:]
+    rot = torch.cat([-x2, x1], dim=-1)
+    return x * cos + rot * sin
+
+
+class _DecoderLayer(nn.Module):
+    def __init__(self, cfg: HeimdallConfig):
+        super().__init__()
+        h = cfg.hidden_size
+        self.hd = h // cfg.num_heads
+        self.nh = cfg.num_heads
+        self.nkv = cfg.num_kv_heads
+        self.q_proj = nn.Linear(h, self.nh * self.hd, bias=True)
+        self.k_proj = nn.Linear(h, self.nkv * self.hd, bias=True)
+        self.v_proj = nn.Linear(h, self.nkv * self.hd, bias=True)
+        self.o_proj = nn.Linear(self.nh * self.hd, h, bias=False)
+        self.gate_proj = nn.Linear(h, cfg.intermediate_size, bias=False)
+        self.up_proj = nn.Linear(h, cfg.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, h, bias=False)
+        self.ln1 = nn.RMSNorm(h, eps=cfg.rms_eps) if hasattr(nn, "RMSNorm") \
+            else nn.LayerNorm(h, eps=cfg.rms_eps)
+        self.ln2 = nn.RMSNorm(h, eps=cfg.rms_eps) if hasattr(nn, "RMSNorm") \
+            else nn.LayerNorm(h, eps=cfg.rms_eps)
+
+    def forward(self, x, cos, sin, kv_cache=None, pos0: int = 0):
+        b, s, h = x.shape
+        r = self.ln1(x)
+        q = self.q_proj(r).view(b, s, self.nh, self.hd).transpose(1, 2)
+        k = self.k_proj(r).view(b, s, self.nkv, self.hd).transpose(1, 2)
+        v = self.v_proj(r).view(b, s, self.nkv, self.hd).transpose(1, 2)
+        q = _rope(q, cos, sin)
+        k = _rope(k, cos, sin)
+        if kv_cache is not None:
+            pk, pv = kv_cache
+            if pk is not None:
+                k = torch.cat([pk, k], dim=2)
+                v = torch.cat([pv, v], dim=2)
+            new_cache = (k, v)
+        else:
+            new_cache = None
+        rep = self.nh // self.nkv
+        ke = k.repeat_interleave(rep, dim=1)
+        ve = v.repeat_interleave(rep, dim=1)
+        causal = kv_cache is None or kv_cache[0] is None
+        a = F.scaled_dot_product_attention(q, ke, ve, is_causal=causal and s > 1)
+        a = a.transpose(1, 2).reshape(b, s, self.nh * self.hd)
+        x = x + self.o_proj(a)
+        r = self.ln2(x)
+        x = x + self.down_proj(F.silu(self.gate_proj(r)) * self.up_proj(r))
+        return x, new_cache
+
+
+class HeimdallModel(nn.Module):
+    def __init__(self, cfg: HeimdallConfig = None):
+        super().__init__()
+        self.cfg = cfg or HeimdallConfig()
+        c = self.cfg
+        self.embed = nn.Embedding(c.vocab_size, c.hidden_size)
+        self.layers = nn.ModuleList(_DecoderLayer(c) for _ in range(c.num_layers))
+        self.norm = nn.RMSNorm(c.hidden_size, eps=c.rms_eps) \
+            if hasattr(nn, "RMSNorm") else nn.LayerNorm(c.hidden_size)
+        self.lm_head = nn.Linear(c.hidden_size, c.vocab_size, bias=False)
+        self.lm_head.weight = self.embed.weight  # tied
+        # rope tables
+        hd = c.hidden_size // c.num_heads
+        inv = 1.0 / (c.rope_theta ** (torch.arange(0, hd, 2).float() / hd))
+        t = torch.arange(c.max_position).float()
+        freqs = torch.outer(t, inv)
+        emb = torch.cat([freqs, freqs], dim=-1)
+        self.register_buffer("rope_cos", emb.cos(), persistent=False)
+        self.register_buffer("rope_sin", emb.sin(), persistent=False)
+
+    @torch.no_grad()
+    def init_small(self, std=0.02):
+        for p in self.parameters():
+            if p.dim() > 1:
+                p.normal_(0, std)
+            else:
+                p.zero_()
+        for m in self.modules():
+            if m.__class__.__name__ in ("RMSNorm", "LayerNorm"):
+                if hasattr(m, "weight") and m.weight is not None:
+                    m.weight.fill_(1.0)
+        return self
+
+    def forward(self, token_ids, kv_caches=None, pos0: int = 0):
+        b, s = token_ids.shape
+        x = self.embed(token_ids)
+        cos = self.rope_cos[pos0:pos0 + s].to(x.dtype)[None, None]
+        sin = self.rope_sin[pos0:pos0 + s].to(x.dtype)[None, None]
+        new_caches = []
+        for i, layer in enumerate(self.layers):
+            cache = kv_caches[i] if kv_caches is not None else None
+            x, nc = layer(x, cos, sin, kv_cache=cache, pos0=pos0)
+            new_caches.append(nc)
+        x = self.norm(x)
+        logits = self.lm_head(x)
+        return logits, (new_caches if kv_caches is not None else None)
+
+    @torch.no_grad()
+    def generate(self, token_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.8, top_k: int = 40,
+                 eos_id: Optional[int] = None, stream_cb=None) -> List[int]:
+        """Greedy/top-k sampled decode with KV cache. Returns new token ids."""
+        self.eval()
+        b, s = token_ids.shape
+        assert b == 1, "generate() is single-sequence"
+        caches = [(None, None)] * len(self.layers)
+        logits, caches = self.forward(token_ids, kv_caches=caches, pos0=0)
+        out = []
+        cur = None
+        pos = s
+        for _ in range(max_new_tokens):
+            last = logits[:, -1, :].float()
+            if temperature <= 0:
+                nxt = int(last.argmax(-1))
+            else:
+                last = last / temperature
+                if top_k:
+                    v, ix = torch.topk(last, min(top_k, last.shape[-1]))
+                    probs = torch.softmax(v, dim=-1)
+                    nxt = int(ix[0, int(torch.multinomial(probs[0], 1))])
+                else:
+                    probs = torch.softmax(last, dim=-1)
+                    nxt = int(torch.multinomial(probs[0], 1))
+            out.append(nxt)
+            if stream_cb:
+                stream_cb(nxt)
+            if eos_id is not None and nxt == eos_id:
+                break
+            cur = torch.tensor([[nxt]], device=token_ids.device)
+            logits, caches = self.forward(cur, kv_caches=caches, pos0=pos)
+            pos += 1
+        return out

@@ -269,6 +269,53 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         body = await request.json()
         return mcp.handle(body)
 
+    # ---- Heimdall / Bifrost (reference pkg/heimdall/bifrost.go SSE) ----
+    _heimdall = [None]
+
+    def get_heimdall():
+        if _heimdall[0] is None:
+            from ..heimdall import HeimdallManager
+            from ..models.heimdall import HeimdallConfig
+            import torch
+            cfg = None if torch.cuda.is_available() else HeimdallConfig.tiny()
+            _heimdall[0] = HeimdallManager(mgr.get(), config=cfg)
+        return _heimdall[0]
+
+    @app.post("/bifrost/generate")
+    def bifrost_generate(body: Dict[str, Any], _user=Depends(check_auth)):
+        h = get_heimdall()
+        text = h.generate(body.get("prompt", ""),
+                          max_tokens=body.get("max_tokens"))
+        return {"text": text, "stats": h.stats}
+
+    @app.post("/bifrost/chat")
+    def bifrost_chat(body: Dict[str, Any], _user=Depends(check_auth)):
+        from ..heimdall import ChatMessage
+        h = get_heimdall()
+        msgs = [ChatMessage(m.get("role", "user"), m.get("content", ""))
+                for m in body.get("messages", [])]
+        return {"text": h.chat(msgs, max_tokens=body.get("max_tokens"))}
+
+    @app.post("/bifrost/stream")
+    def bifrost_stream(body: Dict[str, Any], _user=Depends(check_auth)):
+        from starlette.responses import StreamingResponse
+        h = get_heimdall()
+
+        def sse():
+            for tok in h.generate_stream(body.get("prompt", ""),
+                                         max_tokens=body.get("max_tokens")):
+                yield f"data: {tok}\n\n"
+            yield "data: [DONE]\n\n"
+
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.get("/bifrost/metrics")
+    def bifrost_metrics(_user=Depends(check_auth)):
+        h = get_heimdall()
+        return {"db": h.db_metrics(), "generation": h.stats,
+                "plugins": h.plugin_health(),
+                "tokens_per_second": h.tokens_per_second()}
+
     # ---- GraphQL (reference pkg/graphql) ----
     from .graphql import GraphQLExecutor
 
