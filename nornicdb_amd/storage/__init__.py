@@ -13,9 +13,12 @@ from .persistent import PersistentEngine, Transaction
 from .async_engine import AsyncEngine
 from .namespaced import NamespacedEngine
 from .wal import WAL, WALCorruption
+from .schema import Constraint, SchemaManager, VectorIndexMeta
+from .composite import CompositeEngine
 
 __all__ = [
     "Node", "Edge", "Engine", "EventType", "StorageError", "NotFoundError",
     "ConstraintViolation", "new_id", "MemoryEngine", "PersistentEngine",
     "Transaction", "AsyncEngine", "NamespacedEngine", "WAL", "WALCorruption",
+    "SchemaManager", "Constraint", "VectorIndexMeta", "CompositeEngine",
 ]

@@ -28,6 +28,7 @@ class MemoryEngine(Engine):
         self._in: Dict[str, Set[str]] = {}
         self._pending_embed: Dict[str, float] = {}
         self._callbacks: List[Callable[[str, Any], None]] = []
+        self._validators: List[Callable] = []
         # property indexes: (label, prop) -> value -> set[node_id]
         self._prop_indexes: Dict[tuple, Dict[Any, Set[str]]] = {}
 
@@ -35,6 +36,15 @@ class MemoryEngine(Engine):
     def register_callback(self, cb):
         with self._lock:
             self._callbacks.append(cb)
+
+    def add_validator(self, fn):
+        """fn(node, is_update) raises ConstraintViolation to veto a write."""
+        with self._lock:
+            self._validators.append(fn)
+
+    def _check(self, node, is_update):
+        for v in list(self._validators):
+            v(node, is_update)
 
     def _emit(self, ev: str, obj):
         for cb in list(self._callbacks):
@@ -45,6 +55,7 @@ class MemoryEngine(Engine):
 
     # ---- nodes ----
     def create_node(self, node: Node) -> Node:
+        self._check(node, False)
         with self._lock:
             if node.id in self._nodes:
                 raise ConstraintViolation(f"node {node.id} already exists")
@@ -66,6 +77,7 @@ class MemoryEngine(Engine):
             return n.copy()
 
     def update_node(self, node: Node) -> Node:
+        self._check(node, True)
         with self._lock:
             old = self._nodes.get(node.id)
             if old is None:
