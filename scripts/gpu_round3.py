@@ -1,0 +1,34 @@
+"""k-means timing (BASELINE parity), heimdall decode rate, smoke()."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import time
+import torch
+
+# k-means: reference baseline = 100K points, CUDA 80 ms (75x vs CPU 6 s)
+from nornicdb_amd.search.kmeans import kmeans, optimal_k
+x = torch.randn(100_000, 1024, device="cuda")
+k = optimal_k(100_000)
+torch.cuda.synchronize()
+t0 = time.time()
+c, a = kmeans(x, k, iters=25)
+torch.cuda.synchronize()
+dt = (time.time() - t0) * 1000
+print(f"kmeans 100Kx1024 k={k} 25 iters: {dt:.0f} ms  "
+      f"(reference CUDA baseline: 80 ms @ unknown dims/iters)")
+
+# heimdall decode
+from nornicdb_amd.heimdall import HeimdallManager
+h = HeimdallManager(device="cuda")
+h.generate("warm up", max_tokens=8)
+t0 = time.time()
+out = h.generate("the quick brown fox", max_tokens=64)
+dt = time.time() - t0
+print(f"heimdall (qwen2-0.5B shape bf16) decode: {64/dt:.1f} tok/s")
+
+# smoke()
+sys.path.insert(0, "/root/repo")
+import importlib.util
+spec = importlib.util.spec_from_file_location("ge", "/root/repo/__graft_entry__.py")
+ge = importlib.util.module_from_spec(spec)
+spec.loader.exec_module(ge)
+ge.smoke()

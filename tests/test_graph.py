@@ -208,3 +208,43 @@ def test_sharded_pagerank_bfs_gloo_2proc():
                 p.terminate()
     assert err < 1e-4, f"sharded pagerank diverges from single-proc: {err}"
     assert bfs_ok
+
+
+def _knn_merge_worker(rank, world, port, q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import bench
+        torch.manual_seed(0)
+        full = torch.randn(200, 16)
+        full = full / torch.linalg.vector_norm(full, dim=-1, keepdim=True)
+        shard = full[rank * 100:(rank + 1) * 100]
+        queries = full[:8] + 0.01 * torch.randn(8, 16)
+        s, i = bench.search_step(shard, queries, 5, rank * 100, world)
+        if rank == 0:
+            from nornicdb_amd.ops import knn_search_exact
+            es, ei = knn_search_exact(full, queries, 5)
+            match = (i == ei).float().mean().item()
+            q.put(match)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_knn_merge_gloo_2proc():
+    """bench.py's sharded top-k merge must equal single-machine exact."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_knn_merge_worker, args=(r, 2, 29741, q))
+             for r in range(2)]
+    [p.start() for p in procs]
+    try:
+        match = q.get(timeout=120)
+    finally:
+        [p.join(timeout=30) for p in procs]
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+    assert match > 0.99, f"sharded merge mismatch: {match}"
