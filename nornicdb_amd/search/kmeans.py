@@ -68,17 +68,24 @@ def kmeans(x: torch.Tensor, k: int, iters: int = 25, tol: float = 1e-4,
 
 
 def _kmeanspp(x: torch.Tensor, k: int, g) -> torch.Tensor:
+    """k-means++ seeding; distances via |x|^2+|c|^2-2xc (GEMV on MFMA),
+    sampling stays on-device — no [n,d] broadcast temporaries, no per-step
+    host sync (the reference's init is the same algorithm on CPU,
+    kmeans.go:364)."""
     n = x.shape[0]
+    x_sq = (x * x).sum(-1)
     first = int(torch.randint(n, (1,), generator=g))
-    cents = [x[first]]
-    d2 = ((x - cents[0]) ** 2).sum(-1)
+    idxs = [torch.as_tensor([first], device=x.device)]
+    c = x[first]
+    d2 = (x_sq + (c * c).sum() - 2.0 * (x @ c)).clamp_min_(0)
     for _ in range(1, k):
-        probs = d2.clamp_min(1e-12)
-        idx = int(torch.multinomial(probs.cpu(), 1, generator=g))
-        cents.append(x[idx])
-        nd = ((x - cents[-1]) ** 2).sum(-1)
+        # sampling + gather stay on-device: no host sync inside the loop
+        idx = torch.multinomial(d2.clamp_min(1e-12), 1)
+        idxs.append(idx)
+        c = x[idx[0]]
+        nd = (x_sq + (c * c).sum() - 2.0 * (x @ c)).clamp_min_(0)
         d2 = torch.minimum(d2, nd)
-    return torch.stack(cents)
+    return x[torch.cat(idxs)].clone()
 
 
 class ClusterIndex:
