@@ -109,6 +109,29 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
         return ["node", "coefficient"], [[eng.get_node(g.node_ids[i]), float(cc[i])]
                                          for i in range(g.n)]
 
+    # -------------------- gds compat (reference pkg/cypher/fastrp.go) ----
+    @register("gds.fastRP.stream")
+    def _fastrp(ex, dims=128, iteration_weights=None, seed=42):
+        from ..graph.fastrp import fastrp_embeddings
+        g = from_engine(eng)
+        emb = fastrp_embeddings(g, dims=int(dims),
+                                iteration_weights=iteration_weights or (0.0, 1.0, 1.0),
+                                seed=int(seed))
+        return ["node", "embedding"], [
+            [eng.get_node(g.node_ids[i]), [float(x) for x in emb[i]]]
+            for i in range(g.n)]
+
+    @register("gds.fastRP.write")
+    def _fastrp_write(ex, prop="fastrp", dims=128, seed=42):
+        from ..graph.fastrp import fastrp_embeddings
+        g = from_engine(eng)
+        emb = fastrp_embeddings(g, dims=int(dims), seed=int(seed))
+        for i in range(g.n):
+            n = eng.get_node(g.node_ids[i])
+            n.properties[prop] = [float(x) for x in emb[i]]
+            eng.update_node(n)
+        return ["nodeCount"], [[g.n]]
+
     # -------------------- apoc.create --------------------
     @register("apoc.create.node")
     def _create_node(ex, labels, props):

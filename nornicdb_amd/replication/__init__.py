@@ -6,7 +6,9 @@ from .transport import ChaosConfig, ChaosTransport, InProcTransport, Transport
 from .raft import CANDIDATE, FOLLOWER, LEADER, LogEntry, RaftNode
 from .ha import HAPrimary, HAStandby
 from .adapter import StorageAdapter, command_for
+from .multi_region import Region, RegionReceiver
 
 __all__ = ["Transport", "InProcTransport", "ChaosTransport", "ChaosConfig",
            "RaftNode", "LogEntry", "LEADER", "FOLLOWER", "CANDIDATE",
-           "HAPrimary", "HAStandby", "StorageAdapter", "command_for"]
+           "HAPrimary", "HAStandby", "StorageAdapter", "command_for",
+           "Region", "RegionReceiver"]

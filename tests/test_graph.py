@@ -248,3 +248,32 @@ def test_sharded_knn_merge_gloo_2proc():
             if p.is_alive():
                 p.terminate()
     assert match > 0.99, f"sharded merge mismatch: {match}"
+
+
+class TestFastRP:
+    def test_similar_neighborhoods_embed_close(self):
+        from nornicdb_amd.graph import fastrp_embeddings
+        g = two_cliques()
+        emb = fastrp_embeddings(g, dims=32, seed=1)
+        assert emb.shape == (8, 32)
+        import numpy as np
+        # same-clique nodes more similar than cross-clique on average
+        def cos(a, b):
+            return float(emb[a] @ emb[b])
+        intra = np.mean([cos(i, j) for i in range(4) for j in range(4) if i != j])
+        inter = np.mean([cos(i, j) for i in range(4) for j in range(4, 8)])
+        assert intra > inter
+
+    def test_gds_procedure(self):
+        from nornicdb_amd.db import open_db
+        from nornicdb_amd.embed import MockEmbedder
+        mgr = open_db(embedder=MockEmbedder(8), dims=8)
+        db = mgr.get()
+        db.cypher("CREATE (a:F2)-[:R]->(b:F2)-[:R]->(c:F2)")
+        r = db.cypher("CALL gds.fastRP.stream(16) YIELD node, embedding "
+                      "RETURN size(embedding)")
+        assert r.rows == [[16]] * 3
+        r = db.cypher("CALL gds.fastRP.write('frp', 8) YIELD nodeCount RETURN nodeCount")
+        assert r.rows == [[3]]
+        assert len(db.cypher("MATCH (n:F2) RETURN n.frp LIMIT 1").rows[0][0]) == 8
+        mgr.close()

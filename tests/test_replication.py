@@ -165,3 +165,48 @@ class TestHA:
         assert standby.check_failover()
         assert standby.health()["role"] == "primary"
         bus.close()
+
+
+class TestMultiRegion:
+    def test_cross_region_streaming(self):
+        from nornicdb_amd.replication import (InProcTransport, Region,
+                                              RegionReceiver, StorageAdapter)
+        from nornicdb_amd.storage import MemoryEngine
+        import time as _time
+
+        bus = InProcTransport()
+        eng_us = MemoryEngine()
+        eng_eu = MemoryEngine()
+        recv_eu = RegionReceiver("eu", bus, StorageAdapter(eng_eu).apply)
+        region_us = Region("us", ["us0", "us1", "us2"], bus,
+                           StorageAdapter(eng_us).apply,
+                           remote_regions=["eu"])
+        # elect + replicate
+        t0 = _time.monotonic()
+        while region_us.leader() is None and _time.monotonic() - t0 < 5:
+            region_us.tick_all()
+            _time.sleep(0.01)
+        assert region_us.leader() is not None
+        for i in range(3):
+            assert region_us.propose(node_cmd(i))
+        t0 = _time.monotonic()
+        while eng_eu.node_count() < 3 and _time.monotonic() - t0 < 5:
+            region_us.tick_all()
+            _time.sleep(0.01)
+        assert eng_us.node_count() == 3
+        assert eng_eu.node_count() == 3
+        bus.close()
+
+    def test_out_of_order_xregion(self):
+        from nornicdb_amd.replication import InProcTransport, RegionReceiver, StorageAdapter
+        from nornicdb_amd.storage import MemoryEngine
+        bus = InProcTransport()
+        eng = MemoryEngine()
+        r = RegionReceiver("x", bus, StorageAdapter(eng).apply)
+        r._on_message({"type": "xregion_entry", "region": "y", "seq": 1,
+                       "command": node_cmd(1)})
+        assert eng.node_count() == 0
+        r._on_message({"type": "xregion_entry", "region": "y", "seq": 0,
+                       "command": node_cmd(0)})
+        assert eng.node_count() == 2
+        bus.close()
