@@ -272,7 +272,10 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
   TORCH_CHECK(k_out >= 1 && k_out <= KCAND);
 
   long long n_panels = n / BM;
-  int grid = (int)std::min<long long>(n_panels, 2048);
+  // Enough blocks that each retires quickly (load balance + lets other
+  // streams' kernels co-schedule), few enough that the candidate buffer
+  // and merge stay small.
+  int grid = (int)std::min<long long>(n_panels, 8192);
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   auto opts_f = db.options().dtype(at::kFloat);

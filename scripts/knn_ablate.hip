@@ -202,8 +202,12 @@ __global__ __launch_bounds__(NT, 3) void k_knn_direct(
 // flight across barriers. V10 = A+B double-buffered (80 KB LDS,
 // 2 blocks/CU); V11 = A-only double-buffer (48 KB, 3 blocks/CU),
 // B staged synchronously per K-step.
-#define RAW_BAR() do { __builtin_amdgcn_s_barrier(); \
-                       __builtin_amdgcn_sched_barrier(0); } while (0)
+// compiler memory fence (no instruction) + HW barrier; avoids both the
+// __syncthreads() forced vmcnt(0) drain AND sched_barrier(0) order-pinning
+// (guide m141: order-pinning = 510 TF regression).
+#define RAW_BAR() do { asm volatile("" ::: "memory"); \
+                       __builtin_amdgcn_s_barrier(); \
+                       asm volatile("" ::: "memory"); } while (0)
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
 
 template <int V>
