@@ -79,13 +79,21 @@ def knn_search(
         qq = q.to(torch.bfloat16).contiguous()
         return nat.knn_gemv(db.contiguous(), qq, row_base, k)
     if (
-        q.shape[0] <= 256
-        and k <= 12
+        k <= 12
         and db.dtype == torch.bfloat16
         and db.shape[1] % 64 == 0
         and db.shape[0] >= 128
     ):
-        return _knn_mfma(nat, db, q, k, row_base)
+        if q.shape[0] <= 256:
+            return _knn_mfma(nat, db, q, k, row_base)
+        # large query batches (multi-GPU all-gather): chunk through the
+        # fused kernel 256 queries at a time
+        ss, ii = [], []
+        for s in range(0, q.shape[0], 256):
+            cs, ci = _knn_mfma(nat, db, q[s:s + 256], k, row_base)
+            ss.append(cs)
+            ii.append(ci)
+        return torch.cat(ss, 0), torch.cat(ii, 0)
     return _knn_gemm_chunked(db, q, k, row_base)
 
 
