@@ -143,14 +143,37 @@ class NornicDB:
         procs = build_procedures(self)
         procs.update(build_apoc_procedures(self))
         self.executor = Executor(engine, procedures=procs)
+        # read-query result cache (reference pkg/cache/query_cache.go),
+        # invalidated by any storage write event
+        from .utils.cache import QueryCache
+        self.query_cache = QueryCache(capacity=512, ttl=60.0)
+        engine.register_callback(lambda ev, obj: self.query_cache.invalidate())
         self.auto_embed = auto_embed
         self.embed_queue = EmbedQueue(self, workers=max(embed_workers, 1))
         if embed_workers > 0:
             self.embed_queue.start()
 
     # ---- cypher ----
+    _WRITE_KEYWORDS = ("create", "merge", "set ", "delete", "remove", "drop",
+                       "detach", "foreach")
+
     def cypher(self, query: str, params: Dict[str, Any] = None):
-        return self.executor.execute(query, params)
+        q = query.strip().lower()
+        cacheable = not any(kw in q for kw in self._WRITE_KEYWORDS)
+        key = None
+        if cacheable:
+            try:
+                key = (query, repr(sorted((params or {}).items())))
+            except Exception:
+                key = None
+            if key is not None:
+                hit = self.query_cache.get(key)
+                if hit is not None:
+                    return hit
+        res = self.executor.execute(query, params)
+        if key is not None and not any(res.stats.values()):
+            self.query_cache.put(key, res)
+        return res
 
     execute_cypher = cypher
 

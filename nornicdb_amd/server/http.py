@@ -137,18 +137,22 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         return {"token": token}
 
     # ---- Neo4j HTTP transaction API ----
+    # raw-body endpoint (no pydantic model): this is the hot path and
+    # schema validation costs more than the query itself
     @app.post("/db/{db_name}/tx/commit")
-    def tx_commit(db_name: str, req: TxRequest, request: Request,
-                  _user=Depends(check_auth)):
+    async def tx_commit(db_name: str, request: Request,
+                        _user=Depends(check_auth)):
+        body = await request.json()
         try:
             db = mgr.get(db_name)
         except KeyError:
             raise HTTPException(404, f"database {db_name} not found")
         results, errors = [], []
-        for stmt in req.statements:
+        for stmt in body.get("statements", []):
             t0 = time.time()
             try:
-                r = db.cypher(stmt.statement, stmt.parameters)
+                r = db.cypher(stmt.get("statement", ""),
+                              stmt.get("parameters") or {})
                 metrics.observe("nornicdb_query_seconds", time.time() - t0)
                 results.append({
                     "columns": r.columns,

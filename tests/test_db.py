@@ -163,3 +163,24 @@ class TestCachedEmbedder:
         ce.embed("b")  # evicts "hello"
         ce.embed("hello")
         assert ce.misses == 4
+
+
+class TestQueryCache:
+    def test_read_cached_and_invalidated(self, mgr):
+        db = mgr.get()
+        db.cypher("CREATE (:QC {v: 1})")
+        r1 = db.cypher("MATCH (n:QC) RETURN n.v")
+        assert db.query_cache.hits == 0
+        r2 = db.cypher("MATCH (n:QC) RETURN n.v")
+        assert db.query_cache.hits == 1
+        assert r2.rows == r1.rows
+        # a write invalidates
+        db.cypher("CREATE (:QC {v: 2})")
+        r3 = db.cypher("MATCH (n:QC) RETURN count(n)")
+        assert r3.rows == [[2]]
+
+    def test_write_queries_not_cached(self, mgr):
+        db = mgr.get()
+        db.cypher("CREATE (:W1)")
+        db.cypher("CREATE (:W1)")
+        assert db.cypher("MATCH (n:W1) RETURN count(n)").rows == [[2]]
