@@ -512,8 +512,8 @@ class Executor:
             if rp.props is not None:
                 rel_props = dict(self._eval(rp.props, r, params) or {})
 
-            def try_end(e_list, r2, last_node):
-                """e_list = edges traversed for this hop set."""
+            def try_end(e_list, r2, last_node, mids=()):
+                """e_list = edges traversed; mids = intermediate node objects."""
                 # bind the end node pattern
                 if np.var and np.var in r2:
                     bound = r2[np.var]
@@ -537,7 +537,8 @@ class Executor:
                     if np.var:
                         r3[np.var] = endn
                     yield from walk(i + 2, r3, endn.id,
-                                    nodes + [endn], edges + e_list,
+                                    nodes + list(mids) + [endn],
+                                    edges + e_list,
                                     used_edges | {e.id for e in e_list})
 
             if not rp.var_length:
@@ -553,16 +554,21 @@ class Executor:
                         r2[rp.var] = e
                     yield from try_end([e], r2, other)
             else:
-                # variable-length BFS/DFS up to max_hops
-                def expand(nid, hops, e_acc, visited_edges):
+                # variable-length BFS/DFS up to max_hops; n_acc carries the
+                # intermediate node objects (path var correctness)
+                def expand(nid, hops, e_acc, n_acc, visited_edges):
                     if limit and count[0] >= limit:
                         return
                     if hops >= rp.min_hops:
                         r2 = dict(r)
                         if rp.var:
                             r2[rp.var] = list(e_acc)
-                        yield from try_end(list(e_acc), r2, nid)
+                        yield from try_end(list(e_acc), r2, nid, n_acc)
                     if hops >= rp.max_hops:
+                        return
+                    try:
+                        nid_node = self.engine.get_node(nid)
+                    except NotFoundError:
                         return
                     for e, other in self._edges_from(nid, rp):
                         if e.id in visited_edges or e.id in used_edges:
@@ -570,13 +576,10 @@ class Executor:
                         if rel_props and not self._props_match(e, rel_props):
                             continue
                         yield from expand(other, hops + 1, e_acc + [e],
+                                          n_acc + ([nid_node] if hops > 0 else []),
                                           visited_edges | {e.id})
-                if rp.min_hops == 0:
-                    r2 = dict(r)
-                    if rp.var:
-                        r2[rp.var] = []
-                    yield from try_end([], r2, cur_node)
-                yield from expand(cur_node, 0, [], set())
+                # expand() yields the 0-hop case itself when min_hops == 0
+                yield from expand(cur_node, 0, [], [], set())
 
         start_np: A.NodePattern = elems[0]
         for n0 in self._node_candidates(start_np, row, params):
