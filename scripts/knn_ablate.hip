@@ -654,6 +654,190 @@ __global__ __launch_bounds__(NT, 3) void k_knn(
   }
 }
 
+
+// ===========================================================================
+// V16: 256x256 tile, 8 waves, per-wave-PARTITIONED staging (each wave issues
+// exactly the A-half + B-half it consumes), issue-early during the previous
+// tile, own-loads vmcnt(0) + raw barrier once per K-tile, quadrant-phased
+// MFMA with B fragments hoisted into registers per tile.
+// ===========================================================================
+#define BM16 256
+#define TSTR16 36
+
+__global__ __launch_bounds__(512, 2) void k_knn_v16(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  __shared__ __align__(16) char smem[128 * 1024];
+  unsigned short* sA[2] = {(unsigned short*)smem,
+                           (unsigned short*)(smem + 32768)};
+  unsigned short* sB[2] = {(unsigned short*)(smem + 65536),
+                           (unsigned short*)(smem + 98304)};
+  float* sS = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;       // 0..7
+  const int wr = wid >> 2;        // A half consumed
+  const int wc = wid & 3;
+  const int bh = wc >> 1;         // B half consumed
+  // staging groups: A-half wr staged by the 4 waves with this wr
+  //   (group rank ga = wc in 0..3); B-half bh staged by the 4 waves with
+  //   this bh (group rank gb = (wr<<1) | (wc&1)).
+  const int ga = wc;
+  const int gb = (wr << 1) | (wc & 1);
+
+  float tv[KC];
+  int ti[KC];
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM16;
+    float4v acc[8][4];
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    // one K-tile's staging for THIS wave: 4 chunks of its A-half +
+    // 4 chunks of its B-half (chunk = 64 lanes x 16 B = 1 KB).
+    // A-half wr occupies sA[buf] bytes [wr*16KB, +16KB); within the half,
+    // this wave covers chunks [ga*4, ga*4+4). Same for B with gb.
+#define V16_STAGE_CHUNK(kt, buf, c)  do {                                   \
+      int cc = (c);                                                         \
+      if (cc < 4) {   /* A chunk */                                         \
+        int chunk = wr * 16 + ga * 4 + cc;                                  \
+        int byte_off = chunk * 1024 + lane * 16;                            \
+        int r = byte_off / (BK * 2);                                        \
+        int cb = byte_off % (BK * 2);                                       \
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(           \
+            (const char*)db + (prow + r) * d2 + (long long)(kt) * 2 + cb);  \
+        L_AS unsigned int* lp = (L_AS unsigned int*)(                       \
+            (char*)sA[buf] + chunk * 1024);                                 \
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);                 \
+      } else {        /* B chunk */                                         \
+        int chunk = bh * 16 + gb * 4 + (cc - 4);                            \
+        int byte_off = chunk * 1024 + lane * 16;                            \
+        int r = byte_off / (BK * 2);                                        \
+        int cb = byte_off % (BK * 2);                                       \
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(           \
+            (const char*)qs + (long long)r * d2 + (long long)(kt) * 2 + cb);\
+        L_AS unsigned int* lp = (L_AS unsigned int*)(                       \
+            (char*)sB[buf] + chunk * 1024);                                 \
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);                 \
+      }                                                                     \
+    } while (0)
+
+    const int nt = d / BK;
+    // prologue: stage tile 0, drain, join
+#pragma unroll
+    for (int c = 0; c < 8; ++c) V16_STAGE_CHUNK(0, 0, c);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+
+    for (int t = 0; t < nt; ++t) {
+      const int cur = t & 1;
+      const unsigned short* rA = sA[cur];
+      const unsigned short* rB = sB[cur];
+      const int lq = lane & 15;
+      const int lk8 = (lane >> 4) * 8;
+      // B fragments for this wave's 64 cols, whole K-tile (held in regs)
+      bf16x8 bfr[4][2];
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          bfr[nn][ks] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              rB + (wc * 64 + nn * 16 + lq) * BK + ks * 32 + lk8));
+      // 4 quadrant phases: issue 2 next-tile stages, read 2 A-frag rows,
+      // 16 MFMA
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (t + 1 < nt) {
+          V16_STAGE_CHUNK((t + 1) * BK, cur ^ 1, 2 * q);
+          V16_STAGE_CHUNK((t + 1) * BK, cur ^ 1, 2 * q + 1);
+        }
+        bf16x8 af[2][2];
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            af[mi][ks] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                rA + (wr * 128 + q * 32 + mi * 16 + lq) * BK + ks * 32 + lk8));
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              acc[q * 2 + mi][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mi][ks], bfr[nn][ks], acc[q * 2 + mi][nn], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      // tile boundary: own next-tile loads must land; nothing newer exists
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+    }
+
+    // ---- epilogue: 8 chunks of 32 rows, transposed sS [col][32+4] ----
+#pragma unroll
+    for (int h = 0; h < 8; ++h) {
+      if (wr == (h >> 2)) {
+        int q = h & 3;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int col = 0;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            col = wc * 64 + nn * 16 + (lane & 15);
+            int s0 = mi * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sS + col * TSTR16 + s0) =
+                acc[q * 2 + mi][nn];
+          }
+        }
+      }
+      __syncthreads();
+      const int myq = tid & 255;
+      const int half = tid >> 8;
+      const long long grow0 = prow + (long long)h * 32 + half * 16;
+      const float* myrow = sS + myq * TSTR16 + half * 16;
+#pragma unroll
+      for (int rb = 0; rb < 4; ++rb) {
+        float4v v4 = *reinterpret_cast<const float4v*>(myrow + rb * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float s = v4[j];
+          if (s > tv[KC - 1]) {
+            float cs = s; int ci = (int)(grow0 + rb * 4 + j);
+#pragma unroll
+            for (int i = 0; i < KC; ++i) {
+              bool ins = cs > tv[i];
+              float ts2 = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = ((long long)blockIdx.x * 2 + (tid >> 8)) * BN + (tid & 255);
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[slot * KC + i] = tv[i];
+    cand_idx[slot * KC + i] = ti[i];
+  }
+#undef V16_STAGE_CHUNK
+}
+
 template <int V>
 float run(const unsigned short* db, const unsigned short* qs, long long n,
           int d, float* cs, int* ci, int iters) {
@@ -744,6 +928,68 @@ int main() {
     hipEventRecord(t1); hipEventSynchronize(t1);
     hipEventElapsedTime(&ms, t0, t1); ms /= 10;
     printf("V11 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "pipe-A-dbuf", ms, flops/ms/1e9, bytes/ms/1e9);
+
+    // ---- V16: refcheck at small N, then perf ----
+    {
+      long long nn = 4096;
+      long long panels16 = nn / 256;
+      int g16 = (int)panels16;
+      long long slots = (long long)g16 * 2;
+      hipLaunchKernelGGL(k_knn_v16, dim3(g16), dim3(512), 0, 0, db, qs,
+                         panels16, d, cs, ci);
+      hipDeviceSynchronize();
+      hipError_t err = hipGetLastError();
+      if (err != hipSuccess) { printf("V16 launch err: %s\n", hipGetErrorString(err)); return 1; }
+      // host refcheck: top-1 per query vs brute force
+      unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+      unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+      float* hcs = (float*)malloc(slots * BN * KC * 4);
+      int* hci = (int*)malloc(slots * BN * KC * 4);
+      hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+      hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+      hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+      hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+      auto b2f = [](unsigned short u) { union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f; };
+      int bad = 0;
+      for (int qi = 0; qi < BN; qi += 17) {
+        // brute top-1
+        float best = -1e30f; long long bi = -1;
+        for (long long r = 0; r < nn; ++r) {
+          float acc2 = 0;
+          for (int k2 = 0; k2 < d; ++k2)
+            acc2 += b2f(hdb[r * d + k2]) * b2f(hq[(long long)qi * d + k2]);
+          if (acc2 > best) { best = acc2; bi = r; }
+        }
+        // merged candidate top-1
+        float gbest = -1e30f; int gi = -1;
+        for (long long s2 = 0; s2 < slots; ++s2)
+          for (int k2 = 0; k2 < KC; ++k2) {
+            float v = hcs[(s2 * BN + qi) * KC + k2];
+            if (v > gbest) { gbest = v; gi = hci[(s2 * BN + qi) * KC + k2]; }
+          }
+        if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+          if (bad < 3) printf("V16 MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                              qi, gi, gbest, bi, best);
+          bad++;
+        }
+      }
+      printf("V16 refcheck: %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+      free(hdb); free(hq); free(hcs); free(hci);
+      if (!bad) {
+        long long panels = n / 256;
+        int grid = (int)std::min<long long>(panels, 4096);
+        hipLaunchKernelGGL(k_knn_v16, dim3(grid), dim3(512), 0, 0, db, qs, panels, d, cs, ci);
+        hipDeviceSynchronize();
+        hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+        hipEventRecord(t0);
+        for (int i = 0; i < 10; ++i)
+          hipLaunchKernelGGL(k_knn_v16, dim3(grid), dim3(512), 0, 0, db, qs, panels, d, cs, ci);
+        hipEventRecord(t1); hipEventSynchronize(t1);
+        float ms16; hipEventElapsedTime(&ms16, t0, t1); ms16 /= 10;
+        printf("V16 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "256-tile-partn",
+               ms16, flops/ms16/1e9, bytes/ms16/1e9);
+      }
+    }
   }
   return 0;
 }
