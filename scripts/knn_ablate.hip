@@ -669,10 +669,9 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
     long long n_panels, int d, float* __restrict__ cand_score,
     int* __restrict__ cand_idx) {
   __shared__ __align__(16) char smem[128 * 1024];
-  unsigned short* sA[2] = {(unsigned short*)smem,
-                           (unsigned short*)(smem + 32768)};
-  unsigned short* sB[2] = {(unsigned short*)(smem + 65536),
-                           (unsigned short*)(smem + 98304)};
+  // layout: A0 | A1 | B0 | B1 (32 KB each); sS epilogue buffer aliases
+#define V16_A(buf) ((unsigned short*)(smem + (buf) * 32768))
+#define V16_B(buf) ((unsigned short*)(smem + 65536 + (buf) * 32768))
   float* sS = (float*)smem;
 
   const int tid = threadIdx.x;
@@ -716,7 +715,7 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
         const G_AS unsigned int* gp = (const G_AS unsigned int*)(           \
             (const char*)db + (prow + r) * d2 + (long long)(kt) * 2 + cb);  \
         L_AS unsigned int* lp = (L_AS unsigned int*)(                       \
-            (char*)sA[buf] + chunk * 1024);                                 \
+            (char*)V16_A(buf) + chunk * 1024);                              \
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);                 \
       } else {        /* B chunk */                                         \
         int chunk = bh * 16 + gb * 4 + (cc - 4);                            \
@@ -726,7 +725,7 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
         const G_AS unsigned int* gp = (const G_AS unsigned int*)(           \
             (const char*)qs + (long long)r * d2 + (long long)(kt) * 2 + cb);\
         L_AS unsigned int* lp = (L_AS unsigned int*)(                       \
-            (char*)sB[buf] + chunk * 1024);                                 \
+            (char*)V16_B(buf) + chunk * 1024);                              \
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);                 \
       }                                                                     \
     } while (0)
@@ -741,8 +740,8 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
 
     for (int t = 0; t < nt; ++t) {
       const int cur = t & 1;
-      const unsigned short* rA = sA[cur];
-      const unsigned short* rB = sB[cur];
+      const unsigned short* rA = V16_A(cur);
+      const unsigned short* rB = V16_B(cur);
       const int lq = lane & 15;
       const int lk8 = (lane >> 4) * 8;
       // B fragments for this wave's 64 cols, whole K-tile (held in regs)
@@ -836,6 +835,8 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
     cand_idx[slot * KC + i] = ti[i];
   }
 #undef V16_STAGE_CHUNK
+#undef V16_A
+#undef V16_B
 }
 
 template <int V>
@@ -866,8 +867,8 @@ int main() {
   int* ci;
   hipMalloc(&db, n * d * 2);
   hipMalloc(&qs, (long long)BN * d * 2);
-  hipMalloc(&cs, 2048LL * BN * KC * 4);
-  hipMalloc(&ci, 2048LL * BN * KC * 4);
+  hipMalloc(&cs, 8192LL * BN * KC * 4);
+  hipMalloc(&ci, 8192LL * BN * KC * 4);
   hipLaunchKernelGGL(fill_rand, dim3(4096), dim3(256), 0, 0, db, n * d);
   hipLaunchKernelGGL(fill_rand, dim3(64), dim3(256), 0, 0, qs, (long long)BN * d);
   hipDeviceSynchronize();
