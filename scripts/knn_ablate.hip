@@ -839,6 +839,280 @@ __global__ __launch_bounds__(512, 2) void k_knn_v16(
 #undef V16_B
 }
 
+#define KO4 10
+template <int V>
+__global__ __launch_bounds__(NT, 4) void k_knn_occ4(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  constexpr bool DBUF = (V == 7);
+  __shared__ __align__(16) char smem[DBUF ? (BM * BK + BN * BK) * 4
+                                          : (BM * BK + BN * BK) * 2];
+  unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
+  float* sS = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[KO4];
+  int ti[KO4];
+#pragma unroll
+  for (int i = 0; i < KO4; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM;
+    float4v acc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    auto stage = [&](int kt, int buf) {
+      unsigned short* dA = sA + buf * (BM * BK + BN * BK);
+      unsigned short* dB = sB + buf * (BM * BK + BN * BK);
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int chunk = wc * 2 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)dA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {
+        int chunk = wc * 8 + it;
+        int byte_off = chunk * 1024 + lane * 16;
+        int r = byte_off / (BK * 2);
+        int cb = byte_off % (BK * 2);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)dB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+    };
+    auto mfma_tile = [&](int buf) {
+      const unsigned short* rA = sA + buf * (BM * BK + BN * BK);
+      const unsigned short* rB = sB + buf * (BM * BK + BN * BK);
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = m * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(rA + r * BK + k));
+        }
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(rB + c * BK + k));
+        }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+      }
+    };
+
+    if (DBUF) {
+      int nt = d / BK;
+      stage(0, 0);
+      asm volatile("s_waitcnt vmcnt(0)");
+      __syncthreads();
+      int cur = 0;
+      for (int t = 0; t < nt - 1; ++t) {
+        stage((t + 1) * BK, cur ^ 1);
+        mfma_tile(cur);
+        asm volatile("s_waitcnt vmcnt(0)");
+        __syncthreads();
+        cur ^= 1;
+      }
+      mfma_tile(cur);
+      __syncthreads();
+    } else
+    for (int kt = 0; kt < d; kt += BK) {
+      if (V != 3 && V != 4) {
+#pragma unroll
+        for (int it = 0; it < 2; ++it) {
+          int chunk = wc * 2 + it;
+          int byte_off = chunk * 1024 + lane * 16;
+          int r = byte_off / (BK * 2);
+          int cb = byte_off % (BK * 2);
+          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+              (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+          L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+      }
+      if (V != 2 && V != 4) {
+#pragma unroll
+        for (int it = 0; it < 8; ++it) {
+          int chunk = wc * 8 + it;
+          int byte_off = chunk * 1024 + lane * 16;
+          int r = byte_off / (BK * 2);
+          int cb = byte_off % (BK * 2);
+          const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+              (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+          L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+      }
+      __syncthreads();
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = m * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(sA + r * BK + k));
+        }
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          int k = ks * 32 + (lane >> 4) * 8;
+          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(sB + c * BK + k));
+        }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+
+    if (V == 1 || V == 4) {
+      // keep acc live without the epilogue
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+          asm volatile("" ::"v"(acc[m][nn][0]), "v"(acc[m][nn][3]));
+      continue;
+    }
+
+    if (V == 6 || V == 7) {
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int m = h * 2 + mi;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            int s0 = mi * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sS + col * TSTR + s0) = acc[m][nn];
+          }
+        }
+        __syncthreads();
+        const long long grow0 = prow + (long long)h * SCH;
+        const float* myrow = sS + tid * TSTR;
+#pragma unroll
+        for (int rb = 0; rb < SCH / 4; ++rb) {
+          float4v v4 = *reinterpret_cast<const float4v*>(myrow + rb * 4);
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            float s = v4[j];
+            if (s > tv[KO4 - 1]) {
+              float cs = s; int ci = (int)(grow0 + rb * 4 + j);
+#pragma unroll
+              for (int i = 0; i < KO4; ++i) {
+                bool ins = cs > tv[i];
+                float ts2 = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+        __syncthreads();
+      }
+      long long slot0 = (long long)blockIdx.x * BN + tid;
+      continue;
+    }
+
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int srow = mi * 16 + (lane >> 4) * 4 + r;
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            sS[srow * SSTR + col] = acc[m][nn][r];
+          }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * SCH;
+      bool scan = true;
+      if (V == 5) {
+        // wave-cooperative chunk max: each lane maxes its column set
+        float cmax = -1e30f;
+        for (int r = 0; r < SCH; ++r)
+          cmax = fmaxf(cmax, sS[r * SSTR + tid]);
+        scan = cmax > tv[KO4 - 1];
+        if (scan) {
+#pragma unroll
+          for (int r = 0; r < SCH; ++r) {
+            float s = sS[r * SSTR + tid];
+            if (s > tv[KO4 - 1]) {
+              float cs = s; int ci = (int)(grow0 + r);
+#pragma unroll
+              for (int i = 0; i < KO4; ++i) {
+                bool ins = cs > tv[i];
+                float ts2 = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < SCH; ++r) {
+          float s = sS[r * SSTR + tid];
+          if (s > tv[KO4 - 1]) {
+            float cs = s; int ci = (int)(grow0 + r);
+#pragma unroll
+            for (int i = 0; i < KO4; ++i) {
+              bool ins = cs > tv[i];
+              float ts2 = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i]; ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts2 : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * BN + tid;
+#pragma unroll
+  for (int i = 0; i < KO4; ++i) {
+    cand_score[slot * KO4 + i] = tv[i];
+    cand_idx[slot * KO4 + i] = ti[i];
+  }
+}
+
+
+#undef KO4
+
 template <int V>
 float run(const unsigned short* db, const unsigned short* qs, long long n,
           int d, float* cs, int* ci, int iters) {
@@ -929,6 +1203,21 @@ int main() {
     hipEventRecord(t1); hipEventSynchronize(t1);
     hipEventElapsedTime(&ms, t0, t1); ms /= 10;
     printf("V11 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "pipe-A-dbuf", ms, flops/ms/1e9, bytes/ms/1e9);
+
+    {
+      long long panels = n / BM;
+      int grid = (int)std::min<long long>(panels, 2048);
+      hipLaunchKernelGGL((k_knn_occ4<0>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+      hipDeviceSynchronize();
+      hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+      hipEventRecord(t0);
+      for (int i = 0; i < 10; ++i)
+        hipLaunchKernelGGL((k_knn_occ4<0>), dim3(grid), dim3(NT), 0, 0, db, qs, panels, d, cs, ci);
+      hipEventRecord(t1); hipEventSynchronize(t1);
+      float mso; hipEventElapsedTime(&mso, t0, t1); mso /= 10;
+      printf("V17 %-14s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "occ4-k10",
+             mso, flops/mso/1e9, bytes/mso/1e9);
+    }
 
     // ---- V16: refcheck at small N, then perf ----
     {
