@@ -16,7 +16,26 @@ from ..graph import (betweenness_centrality, closeness_centrality,
                      connected_components, degree_centrality, dijkstra,
                      from_engine, label_propagation, louvain, pagerank,
                      shortest_path, triangle_count, clustering_coefficient)
+import threading
+
 from ..storage.types import Edge, Node, new_id
+
+_ATOMIC_LOCK = threading.Lock()
+_NODE_LOCKS = {}
+
+
+def _parse_rel_filter(rel_filter):
+    """'KNOWS|WORKS_AT>' -> ({KNOWS, WORKS_AT}, 'out'); None -> any/both."""
+    if not rel_filter:
+        return set(), "both"
+    direction = "both"
+    f = rel_filter
+    if f.endswith(">"):
+        direction, f = "out", f[:-1]
+    elif f.endswith("<") or f.startswith("<"):
+        direction, f = "in", f.strip("<")
+    types = {t for t in f.split("|") if t}
+    return types, direction
 
 
 def build_apoc_procedures(db) -> Dict[str, Any]:
@@ -296,5 +315,161 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
         return (["file", "nodes", "relationships"],
                 [[file or "<inline>", len(data["nodes"]),
                   len(data["relationships"])]])
+
+
+
+    # -------------------- apoc.path --------------------
+    @register("apoc.path.expand")
+    def _path_expand(ex, start, rel_filter=None, label_filter=None,
+                     min_level=1, max_level=3):
+        """BFS path expansion (reference apoc/path). rel_filter like
+        'KNOWS|WORKS_AT>' (> = outgoing only, < = incoming only)."""
+        sid = start.id if isinstance(start, Node) else start
+        types, direction = _parse_rel_filter(rel_filter)
+        allowed = set((label_filter or "").replace("+", "").split("|")) \
+            if label_filter else None
+        out_rows = []
+
+        def edges_of(nid):
+            es = []
+            if direction in ("out", "both"):
+                es += [(e, e.end_node) for e in eng.get_out_edges(nid)]
+            if direction in ("in", "both"):
+                es += [(e, e.start_node) for e in eng.get_in_edges(nid)]
+            return [(e, o) for e, o in es if not types or e.type in types]
+
+        def walk(nid, depth, nodes, rels, seen):
+            if depth >= int(min_level):
+                out_rows.append([list(nodes), list(rels)])
+            if depth >= int(max_level):
+                return
+            for e, other in edges_of(nid):
+                if e.id in seen:
+                    continue
+                try:
+                    onode = eng.get_node(other)
+                except Exception:
+                    continue
+                if allowed and not (set(onode.labels) & allowed):
+                    continue
+                walk(other, depth + 1, nodes + [onode], rels + [e],
+                     seen | {e.id})
+
+        walk(sid, 0, [eng.get_node(sid)], [], set())
+        return ["nodes", "relationships"], out_rows
+
+    @register("apoc.path.subgraphNodes")
+    def _subgraph_nodes(ex, start, max_level=3):
+        sid = start.id if isinstance(start, Node) else start
+        seen = {sid}
+        frontier = [sid]
+        rows = [[eng.get_node(sid)]]
+        for _ in range(int(max_level)):
+            nxt = []
+            for nid in frontier:
+                for nb in eng.neighbors(nid):
+                    if nb not in seen:
+                        seen.add(nb)
+                        nxt.append(nb)
+                        rows.append([eng.get_node(nb)])
+            frontier = nxt
+        return ["node"], rows
+
+    # -------------------- apoc.atomic --------------------
+    @register("apoc.atomic.add")
+    def _atomic_add(ex, node, prop, value):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            n.properties[prop] = (n.properties.get(prop) or 0) + value
+            n = eng.update_node(n)
+        return ["node", "value"], [[n, n.properties[prop]]]
+
+    @register("apoc.atomic.subtract")
+    def _atomic_sub(ex, node, prop, value):
+        return _atomic_add(ex, node, prop, -value)
+
+    @register("apoc.atomic.update")
+    def _atomic_update(ex, node, prop, value):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            n.properties[prop] = value
+            n = eng.update_node(n)
+        return ["node"], [[n]]
+
+    # -------------------- apoc.lock --------------------
+    @register("apoc.lock.nodes")
+    def _lock_nodes(ex, nodes):
+        # cooperative advisory locks (reference apoc/lock); engine ops are
+        # already serialized, so this is ordering-only
+        ids = sorted(n.id if isinstance(n, Node) else n for n in nodes or [])
+        for i in ids:
+            _NODE_LOCKS.setdefault(i, threading.Lock()).acquire()
+        for i in reversed(ids):
+            _NODE_LOCKS[i].release()
+        return ["locked"], [[len(ids)]]
+
+    # -------------------- apoc.trigger --------------------
+    @register("apoc.trigger.add")
+    def _trigger_add(ex, name, statement, selector=None):
+        phase = (selector or {}).get("phase", "after")
+        db.triggers[name] = {"statement": statement, "phase": phase,
+                             "paused": False}
+        return ["name", "installed"], [[name, True]]
+
+    @register("apoc.trigger.remove")
+    def _trigger_remove(ex, name):
+        db.triggers.pop(name, None)
+        return ["name", "removed"], [[name, True]]
+
+    @register("apoc.trigger.list")
+    def _trigger_list(ex):
+        return (["name", "statement", "paused"],
+                [[k, v["statement"], v["paused"]]
+                 for k, v in db.triggers.items()])
+
+    @register("apoc.trigger.pause")
+    def _trigger_pause(ex, name):
+        if name in db.triggers:
+            db.triggers[name]["paused"] = True
+        return ["name", "paused"], [[name, True]]
+
+    @register("apoc.trigger.resume")
+    def _trigger_resume(ex, name):
+        if name in db.triggers:
+            db.triggers[name]["paused"] = False
+        return ["name", "paused"], [[name, False]]
+
+    # -------------------- apoc.load / export CSV --------------------
+    @register("apoc.load.json")
+    def _load_json(ex, path):
+        import json as J
+        with open(path) as f:
+            data = J.load(f)
+        rows = data if isinstance(data, list) else [data]
+        return ["value"], [[r] for r in rows]
+
+    @register("apoc.load.csv")
+    def _load_csv(ex, path, config=None):
+        import csv
+        rows = []
+        with open(path, newline="") as f:
+            reader = csv.DictReader(f)
+            for i, rec in enumerate(reader):
+                rows.append([i, dict(rec), list(rec.values())])
+        return ["lineNo", "map", "list"], rows
+
+    @register("apoc.export.csv.all")
+    def _export_csv(ex, file, config=None):
+        import csv
+        props = sorted({k for n in eng.all_nodes() for k in n.properties})
+        with open(file, "w", newline="") as f:
+            w = csv.writer(f)
+            w.writerow(["_id", "_labels"] + props)
+            count = 0
+            for n in eng.all_nodes():
+                w.writerow([n.id, ";".join(n.labels)]
+                           + [n.properties.get(p, "") for p in props])
+                count += 1
+        return ["file", "nodes"], [[file, count]]
 
     return procs

@@ -149,6 +149,8 @@ class NornicDB:
         self.query_cache = QueryCache(capacity=512, ttl=60.0)
         engine.register_callback(lambda ev, obj: self.query_cache.invalidate())
         self.auto_embed = auto_embed
+        # apoc.trigger registry: fired after mutating cypher statements
+        self.triggers: Dict[str, Dict[str, Any]] = {}
         self.embed_queue = EmbedQueue(self, workers=max(embed_workers, 1))
         if embed_workers > 0:
             self.embed_queue.start()
@@ -173,6 +175,14 @@ class NornicDB:
         res = self.executor.execute(query, params)
         if key is not None and not any(res.stats.values()):
             self.query_cache.put(key, res)
+        if self.triggers and any(res.stats.values()):
+            for name, t in list(self.triggers.items()):
+                if t.get("paused"):
+                    continue
+                try:
+                    self.executor.execute(t["statement"], {})
+                except Exception:
+                    pass
         return res
 
     execute_cypher = cypher

@@ -152,3 +152,45 @@ class TestPeriodicMeta:
         import json
         data = json.load(open(f))
         assert data["nodes"][0]["properties"]["v"] == 1
+
+
+class TestPathAtomicTrigger:
+    def test_path_expand(self, db):
+        db.cypher("CREATE (a:PE {n:'a'})-[:R]->(b:PE {n:'b'})-[:R]->(c:PE {n:'c'})")
+        r = db.cypher(
+            "MATCH (a:PE {n:'a'}) CALL apoc.path.expand(a, 'R>', null, 1, 2) "
+            "YIELD nodes RETURN size(nodes)")
+        assert sorted(x[0] for x in r.rows) == [2, 3]
+
+    def test_subgraph_nodes(self, db):
+        db.cypher("CREATE (a:SG {n:1})-[:R]->(:SG {n:2})-[:R]->(:SG {n:3})")
+        r = db.cypher("MATCH (a:SG {n:1}) "
+                      "CALL apoc.path.subgraphNodes(a, 2) YIELD node "
+                      "RETURN count(node)")
+        assert r.rows == [[3]]
+
+    def test_atomic_add(self, db):
+        db.cypher("CREATE (:AT {c: 10})")
+        r = db.cypher("MATCH (n:AT) CALL apoc.atomic.add(n, 'c', 5) "
+                      "YIELD value RETURN value")
+        assert r.rows == [[15]]
+        assert db.cypher("MATCH (n:AT) RETURN n.c").rows == [[15]]
+
+    def test_trigger_fires_on_write(self, db):
+        db.cypher("CALL apoc.trigger.add('audit', "
+                  "'MERGE (c:TriggerCounter) ON CREATE SET c.n = 1 "
+                  "ON MATCH SET c.n = c.n + 1')")
+        db.cypher("CREATE (:TG)")
+        db.cypher("CREATE (:TG)")
+        r = db.cypher("MATCH (c:TriggerCounter) RETURN c.n")
+        assert r.rows[0][0] >= 2
+        db.cypher("CALL apoc.trigger.remove('audit')")
+
+    def test_load_export_csv(self, db, tmp_path):
+        db.cypher("CREATE (:CSV {name: 'x', v: 1}), (:CSV {name: 'y', v: 2})")
+        f = str(tmp_path / "out.csv")
+        r = db.cypher(f"CALL apoc.export.csv.all('{f}') YIELD nodes RETURN nodes")
+        assert r.rows == [[2]]
+        r = db.cypher(f"CALL apoc.load.csv('{f}') YIELD map RETURN map['name'] "
+                      "ORDER BY map['name']")
+        assert [x[0] for x in r.rows] == ["x", "y"]
