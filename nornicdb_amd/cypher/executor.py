@@ -96,10 +96,78 @@ class Executor:
                       "properties_set": 0, "labels_added": 0}
         if q.explain:
             return Result(["plan"], [[self._explain(q)]], dict(self.stats))
+        fast = self._try_fast_count(q, params)
+        if fast is not None:
+            fast.stats = dict(self.stats)
+            return fast
         res = self._run_query(q, params)
         self._initial_bindings = None
         res.stats = dict(self.stats)
         return res
+
+    def _try_fast_count(self, q: A.Query, params):
+        """Fast path for `MATCH (n:L [{props}]) [WHERE ...] RETURN count(..)`
+        over the raw label index — no node copies, no row pipeline
+        (reference executor.go:994 tryFastPathCompoundQuery +
+        storage_fastpaths.go)."""
+        if q.union or len(q.clauses) != 2:
+            return None
+        m, r = q.clauses
+        if not (isinstance(m, A.MatchClause) and not m.optional
+                and len(m.patterns) == 1):
+            return None
+        pat = m.patterns[0]
+        if pat.var or getattr(pat, "shortest", None) or len(pat.elements) != 1:
+            return None
+        np_ = pat.elements[0]
+        if not (isinstance(r, A.ReturnClause) and not r.distinct
+                and not r.order_by and r.skip is None and r.limit is None
+                and not r.star and len(r.items) == 1):
+            return None
+        it = r.items[0].expr
+        if not (isinstance(it, A.FuncCall) and it.name == "count"
+                and not it.distinct):
+            return None
+        if not it.star:
+            if not (len(it.args) == 1 and isinstance(it.args[0], A.Var)
+                    and np_.var and it.args[0].name == np_.var):
+                return None
+        raw_iter = getattr(self.engine, "iter_nodes_raw", None)
+        if raw_iter is None:
+            return None
+        name = r.items[0].alias or self._expr_name(it)
+
+        props = {}
+        if np_.props is not None:
+            try:
+                props = dict(self._eval(np_.props, {}, params) or {})
+            except CypherRuntimeError:
+                return None
+        label = np_.labels[0] if np_.labels else None
+        extra_labels = np_.labels[1:]
+
+        # O(1): bare single-label count
+        if (label and not extra_labels and not props and m.where is None
+                and hasattr(self.engine, "node_count_by_label")):
+            return Result([name], [[self.engine.node_count_by_label(label)]])
+        if (label is None and not props and m.where is None):
+            return Result([name], [[self.engine.node_count()]])
+
+        count = 0
+        try:
+            for n in raw_iter(label):
+                if extra_labels and not all(lb in n.labels for lb in extra_labels):
+                    continue
+                if props and not self._props_match(n, props):
+                    continue
+                if m.where is not None:
+                    row = {np_.var: n} if np_.var else {}
+                    if self._eval(m.where, row, params) is not True:
+                        continue
+                count += 1
+        except CypherRuntimeError:
+            return None  # exotic WHERE -> full pipeline
+        return Result([name], [[count]])
 
     def _explain(self, q: A.Query) -> str:
         lines = []

@@ -27,6 +27,8 @@ class MemoryEngine(Engine):
         self._out: Dict[str, Set[str]] = {}  # node -> edge ids
         self._in: Dict[str, Set[str]] = {}
         self._pending_embed: Dict[str, float] = {}
+        # O(1) per-(label, namespace-prefix) counts; prefix = id up to ':'
+        self._label_ns_counts: Dict[tuple, int] = {}
         self._callbacks: List[Callable[[str, Any], None]] = []
         self._validators: List[Callable] = []
         # property indexes: (label, prop) -> value -> set[node_id]
@@ -45,6 +47,21 @@ class MemoryEngine(Engine):
     def _check(self, node, is_update):
         for v in list(self._validators):
             v(node, is_update)
+
+    @staticmethod
+    def _ns_of(node_id: str) -> str:
+        i = node_id.find(":")
+        return node_id[:i] if i > 0 else ""
+
+    def _count_adjust(self, node: Node, delta: int):
+        ns = self._ns_of(node.id)
+        for lb in node.labels:
+            k = (lb, ns)
+            v = self._label_ns_counts.get(k, 0) + delta
+            if v <= 0:
+                self._label_ns_counts.pop(k, None)
+            else:
+                self._label_ns_counts[k] = v
 
     def _emit(self, ev: str, obj):
         for cb in list(self._callbacks):
@@ -65,6 +82,7 @@ class MemoryEngine(Engine):
             self._nodes[n.id] = n
             for lb in n.labels:
                 self._label_index.setdefault(lb, set()).add(n.id)
+            self._count_adjust(n, +1)
             self._index_node_props(n, add=True)
         self._emit(EventType.NODE_CREATED, n.copy())
         return n.copy()
@@ -84,6 +102,7 @@ class MemoryEngine(Engine):
                 raise NotFoundError(f"node {node.id} not found")
             for lb in old.labels:
                 self._label_index.get(lb, set()).discard(node.id)
+            self._count_adjust(old, -1)
             self._index_node_props(old, add=False)
             n = node.copy()
             n.created_at = old.created_at
@@ -91,6 +110,7 @@ class MemoryEngine(Engine):
             self._nodes[n.id] = n
             for lb in n.labels:
                 self._label_index.setdefault(lb, set()).add(n.id)
+            self._count_adjust(n, +1)
             self._index_node_props(n, add=True)
         self._emit(EventType.NODE_UPDATED, n.copy())
         return n.copy()
@@ -106,6 +126,7 @@ class MemoryEngine(Engine):
             del self._nodes[node_id]
             for lb in n.labels:
                 self._label_index.get(lb, set()).discard(node_id)
+            self._count_adjust(n, -1)
             self._index_node_props(n, add=False)
             self._pending_embed.pop(node_id, None)
         self._emit(EventType.NODE_DELETED, n)
@@ -123,6 +144,24 @@ class MemoryEngine(Engine):
     def node_count(self) -> int:
         with self._lock:
             return len(self._nodes)
+
+    def node_count_by_label(self, label: str, ns: str = None) -> int:
+        with self._lock:
+            if ns is None:
+                return len(self._label_index.get(label, ()))
+            return self._label_ns_counts.get((label, ns), 0)
+
+    def iter_nodes_raw(self, label: str = None):
+        """Yield LIVE node objects without copying — read-only fast paths
+        (reference storage_fastpaths.go). Callers must not mutate."""
+        with self._lock:
+            if label is None:
+                snap = list(self._nodes.values())
+            else:
+                snap = [self._nodes[i]
+                        for i in self._label_index.get(label, ())
+                        if i in self._nodes]
+        return iter(snap)
 
     # ---- edges ----
     def create_edge(self, edge: Edge) -> Edge:
@@ -288,11 +327,13 @@ class MemoryEngine(Engine):
             self._label_index.clear(); self._type_index.clear()
             self._out.clear(); self._in.clear()
             self._pending_embed = dict(state.get("pending", {}))
+            self._label_ns_counts.clear()
             for (nid, labels, props, emb, ca, ua) in state["nodes"]:
                 n = Node(nid, list(labels), dict(props), emb, ca, ua)
                 self._nodes[nid] = n
                 for lb in n.labels:
                     self._label_index.setdefault(lb, set()).add(nid)
+                self._count_adjust(n, +1)
             for (eid, et, s, t, props, ca, ua) in state["edges"]:
                 e = Edge(eid, et, s, t, dict(props), ca, ua)
                 self._edges[eid] = e

@@ -78,6 +78,23 @@ class NamespacedEngine(Engine):
     def node_count(self) -> int:
         return sum(1 for _ in self.all_nodes())
 
+    def node_count_by_label(self, label: str) -> int:
+        fast = getattr(self.inner, "node_count_by_label", None)
+        if fast is not None:
+            try:
+                return fast(label, ns=self.ns)
+            except TypeError:
+                pass
+        return len(self.get_nodes_by_label(label))
+
+    def iter_nodes_raw(self, label: str = None):
+        inner = getattr(self.inner, "iter_nodes_raw", None)
+        if inner is None:
+            return iter(())
+        # raw nodes keep their PREFIXED ids (read-only count/filter paths
+        # only look at labels/properties)
+        return (n for n in inner(label) if self._mine(n.id))
+
     # --- edges ---
     def create_edge(self, edge: Edge) -> Edge:
         return self._unwrap_edge(self.inner.create_edge(self._wrap_edge(edge)))
