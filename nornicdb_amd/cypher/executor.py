@@ -153,19 +153,53 @@ class Executor:
         if (label is None and not props and m.where is None):
             return Result([name], [[self.engine.node_count()]])
 
+        # indexed equality WHERE: var.prop = <value> -> exact index count
+        if (m.where is not None and isinstance(m.where, A.BinOp)
+                and m.where.op == "=" and label and not extra_labels
+                and not props and np_.var):
+            lhs, rhs = m.where.left, m.where.right
+            if (isinstance(lhs, A.Prop) and isinstance(lhs.expr, A.Var)
+                    and lhs.expr.name == np_.var
+                    and isinstance(rhs, (A.Lit, A.Param))):
+                lookup = getattr(self.engine, "lookup_property_index", None)
+                if lookup is not None:
+                    try:
+                        val = self._eval(rhs, {}, params)
+                        hits = lookup(label, lhs.key, val)
+                    except CypherRuntimeError:
+                        hits = None
+                    if hits is not None:
+                        return Result([name], [[len(hits)]])
+
+        npred = None
+        if m.where is not None:
+            from .compiler import compile_node_predicate
+            npred = compile_node_predicate(m.where, np_.var) if np_.var else None
         count = 0
         try:
-            for n in raw_iter(label):
-                if extra_labels and not all(lb in n.labels for lb in extra_labels):
-                    continue
-                if props and not self._props_match(n, props):
-                    continue
-                if m.where is not None:
-                    row = {np_.var: n} if np_.var else {}
-                    if self._eval(m.where, row, params) is not True:
+            if (m.where is not None and npred is not None
+                    and not extra_labels and not props):
+                # tightest loop: raw nodes -> compiled predicate
+                for n in raw_iter(label):
+                    if npred(n, params) is True:
+                        count += 1
+            else:
+                for n in raw_iter(label):
+                    if extra_labels and not all(lb in n.labels
+                                                for lb in extra_labels):
                         continue
-                count += 1
-        except CypherRuntimeError:
+                    if props and not self._props_match(n, props):
+                        continue
+                    if m.where is not None:
+                        if npred is not None:
+                            if npred(n, params) is not True:
+                                continue
+                        else:
+                            row = {np_.var: n} if np_.var else {}
+                            if self._eval(m.where, row, params) is not True:
+                                continue
+                    count += 1
+        except (CypherRuntimeError, KeyError, TypeError):
             return None  # exotic WHERE -> full pipeline
         return Result([name], [[count]])
 
@@ -719,8 +753,17 @@ class Executor:
                     nxt.extend(self._match_path(pat, r, params))
                 matched = nxt
             if c.where is not None:
-                matched = [r for r in matched
-                           if self._eval(c.where, r, params) is True]
+                from .compiler import compile_predicate
+                fn = compile_predicate(c.where)
+                if fn is not None:
+                    try:
+                        matched = [r for r in matched if fn(r, params) is True]
+                    except Exception:
+                        matched = [r for r in matched
+                                   if self._eval(c.where, r, params) is True]
+                else:
+                    matched = [r for r in matched
+                               if self._eval(c.where, r, params) is True]
             if matched:
                 out.extend(matched)
             elif c.optional:
