@@ -32,12 +32,12 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define BN 256
 #define BK 64
 #define NTHREADS 256
-#define KCAND 10
+#define KCAND 12
 // LDS score-scan chunk: rows per chunk
 #define SCH 32
 #define S_STRIDE (BN + 4)
 
-__global__ __launch_bounds__(NTHREADS, 4) void k_knn_mfma(
+__global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
     long long n_panels,  // number of full BM-row panels
     int d,               // inner dim, % BK == 0
