@@ -27,8 +27,25 @@ class PackStreamError(Exception):
     pass
 
 
+try:  # native codec (csrc/packstream.cpp); python fallback below
+    from nornicdb_amd import _C as _native
+    _HAS_NATIVE_PS = hasattr(_native, "ps_pack")
+except ImportError:  # pragma: no cover
+    _native = None
+    _HAS_NATIVE_PS = False
+
+
 # ---------------------------------------------------------------- packing
 def pack(value: Any) -> bytes:
+    if _HAS_NATIVE_PS:
+        return _native.ps_pack(value)
+    out = bytearray()
+    _pack_into(out, value)
+    return bytes(out)
+
+
+def pack_py(value: Any) -> bytes:
+    """Pure-python packer (kept as the oracle for codec tests)."""
     out = bytearray()
     _pack_into(out, value)
     return bytes(out)
@@ -209,6 +226,14 @@ class Unpacker:
 
 
 def unpack(data: bytes) -> Any:
+    # note: the native decoder exists (_C.ps_unpack) but loses to the
+    # python one on typical small Bolt messages because every nested
+    # Structure requires a callback into python; encode is native.
+    return Unpacker(data).unpack()
+
+
+def unpack_py(data: bytes) -> Any:
+    """Pure-python unpacker (codec-test oracle)."""
     return Unpacker(data).unpack()
 
 

@@ -23,6 +23,11 @@ at::Tensor labelprop_step(at::Tensor row_ptr, at::Tensor col_idx,
 void wcc_hook(at::Tensor row_ptr, at::Tensor col_idx, at::Tensor comp,
               at::Tensor changed, long long row_base);
 
+// packstream.cpp (CPU)
+pybind11::bytes ps_pack(pybind11::object v);
+pybind11::object ps_unpack(pybind11::buffer data,
+                           pybind11::object structure_factory);
+
 // encoder_ops.hip
 at::Tensor add_layernorm(at::Tensor a, c10::optional<at::Tensor> b,
                          at::Tensor gamma, at::Tensor beta, double eps);
@@ -59,4 +64,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bfs_level", &bfs_level, "BFS frontier level expansion");
   m.def("labelprop_step", &labelprop_step, "label propagation step");
   m.def("wcc_hook", &wcc_hook, "connected-components hook step");
+  m.def("ps_pack", &ps_pack, "PackStream encode (native)");
+  m.def("ps_unpack", &ps_unpack, "PackStream decode (native)",
+        py::arg("data"), py::arg("structure_factory"));
 }

@@ -180,3 +180,16 @@ def test_bolt_auth_required():
 
     asyncio.get_event_loop_policy().new_event_loop().run_until_complete(
         asyncio.wait_for(run(), timeout=15))
+
+
+class TestNativeCodec:
+    def test_native_pack_matches_python(self):
+        if not ps._HAS_NATIVE_PS:
+            pytest.skip("native codec not built")
+        vals = [None, True, 42, -300, 2 ** 40, 1.5, "hi", b"xy",
+                [1, [2], "a"], {"k": {"n": None}},
+                ps.Structure(0x4E, [1, ["A"], {"x": 1}]), "y" * 300,
+                list(range(20)), {"m": "x" * 70000}]
+        for v in vals:
+            assert ps.pack(v) == ps.pack_py(v)
+            assert ps.unpack_py(ps.pack(v)) == v
