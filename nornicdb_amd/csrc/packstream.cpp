@@ -91,7 +91,7 @@ struct Packer {
       return;
     }
     if (py::isinstance<py::bytes>(v) || py::isinstance<py::bytearray>(v)) {
-      std::string s = py::bytes(v).cast<std::string>();
+      std::string s = v.cast<std::string>();
       pack_len(s.size(), 0xFF, 0xCC, 0xCD, 0xCE);
       out += s;
       return;
