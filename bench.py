@@ -308,7 +308,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(qps / baseline_qps, 3) if device.type == "cuda" else None,
-            "dtype": "bf16",
+            "dtype": "bf16" if device.type == "cuda" else "fp32 (cpu dev mode)",
             "data": "synthetic (deterministic unit-norm corpus; random token ids; random-init bge-m3 weights)",
             "config": {
                 "model": "bge-m3 (XLM-R-large shape: 24L/1024h/16heads, random init)",
