@@ -59,6 +59,16 @@ class HeimdallManager:
         self.stats = {"generations": 0, "tokens_generated": 0,
                       "total_latency_s": 0.0}
         self._torch = torch
+        self._graphed = None
+        if self.device.startswith("cuda"):
+            try:
+                from ..models.heimdall import GraphedDecoder
+                self._graphed = GraphedDecoder(self.model,
+                                               max_len=cfg.max_position
+                                               if cfg.max_position <= 2048
+                                               else 2048).capture()
+            except Exception:
+                self._graphed = None  # eager fallback
 
     # ---- plugins ----
     def register_plugin(self, plugin: HeimdallPlugin):
@@ -92,9 +102,10 @@ class HeimdallManager:
         t = self._torch.as_tensor([toks], device=self.device)
         t0 = time.time()
         with self._lock:
-            out = self.model.generate(t, max_new_tokens=max_tokens
-                                      or self.max_tokens_default,
-                                      temperature=temperature)
+            gen = self._graphed.generate if self._graphed is not None \
+                else self.model.generate
+            out = gen(t, max_new_tokens=max_tokens or self.max_tokens_default,
+                      temperature=temperature)
         dt = time.time() - t0
         self.stats["generations"] += 1
         self.stats["tokens_generated"] += len(out)
@@ -110,10 +121,10 @@ class HeimdallManager:
 
         def worker():
             with self._lock:
-                self.model.generate(t, max_new_tokens=max_tokens
-                                    or self.max_tokens_default,
-                                    temperature=temperature,
-                                    stream_cb=lambda tok: q.put(tok))
+                gen = self._graphed.generate if self._graphed is not None \
+                    else self.model.generate
+                gen(t, max_new_tokens=max_tokens or self.max_tokens_default,
+                    temperature=temperature, stream_cb=lambda tok: q.put(tok))
             q.put(DONE)
 
         th = threading.Thread(target=worker, daemon=True)

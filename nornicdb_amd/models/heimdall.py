@@ -177,3 +177,113 @@ class HeimdallModel(nn.Module):
             logits, caches = self.forward(cur, kv_caches=caches, pos0=pos)
             pos += 1
         return out
+
+
+class GraphedDecoder:
+    """hipGraph-captured single-token decode step.
+
+    The eager decode loop launches ~150 kernels per token and is launch
+    bound on small models; capturing one step as a HIP graph replays the
+    whole token in a single launch. Static KV cache (attention over
+    max_len with a position mask keeps every shape static), static
+    input/position/output buffers.
+    """
+
+    def __init__(self, model: HeimdallModel, max_len: int = 512):
+        self.m = model
+        self.cfg = model.cfg
+        self.max_len = min(max_len, self.cfg.max_position)
+        dev = next(model.parameters()).device
+        dtype = next(model.parameters()).dtype
+        c = self.cfg
+        hd = c.hidden_size // c.num_heads
+        self.hd = hd
+        self.cache_k = [torch.zeros(1, c.num_kv_heads, self.max_len, hd,
+                                    device=dev, dtype=dtype)
+                        for _ in range(c.num_layers)]
+        self.cache_v = [torch.zeros_like(self.cache_k[0])
+                        for _ in range(c.num_layers)]
+        self.tok = torch.zeros(1, 1, device=dev, dtype=torch.long)
+        self.pos = torch.zeros(1, device=dev, dtype=torch.long)
+        self.arange = torch.arange(self.max_len, device=dev)
+        self.graph = None
+        self.out = None
+
+    def _step(self):
+        m, c = self.m, self.cfg
+        x = m.embed(self.tok)                              # [1,1,h]
+        cos = m.rope_cos.index_select(0, self.pos).to(x.dtype)[None, None]
+        sin = m.rope_sin.index_select(0, self.pos).to(x.dtype)[None, None]
+        keep = (self.arange <= self.pos).view(1, 1, 1, self.max_len)
+        rep = c.num_heads // c.num_kv_heads
+        for li, layer in enumerate(m.layers):
+            r = layer.ln1(x)
+            q = layer.q_proj(r).view(1, 1, layer.nh, layer.hd).transpose(1, 2)
+            k = layer.k_proj(r).view(1, 1, layer.nkv, layer.hd).transpose(1, 2)
+            v = layer.v_proj(r).view(1, 1, layer.nkv, layer.hd).transpose(1, 2)
+            q = _rope(q, cos, sin)
+            k = _rope(k, cos, sin)
+            self.cache_k[li].index_copy_(2, self.pos, k)
+            self.cache_v[li].index_copy_(2, self.pos, v)
+            ke = self.cache_k[li].repeat_interleave(rep, dim=1)
+            ve = self.cache_v[li].repeat_interleave(rep, dim=1)
+            a = F.scaled_dot_product_attention(q, ke, ve, attn_mask=keep)
+            a = a.transpose(1, 2).reshape(1, 1, layer.nh * layer.hd)
+            x = x + layer.o_proj(a)
+            r = layer.ln2(x)
+            x = x + layer.down_proj(F.silu(layer.gate_proj(r)) * layer.up_proj(r))
+        return m.lm_head(m.norm(x))[:, -1, :]
+
+    def capture(self):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self.out = self._step()
+        self.graph = g
+        return self
+
+    @torch.no_grad()
+    def generate(self, token_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.8, top_k: int = 40,
+                 eos_id=None, stream_cb=None):
+        """Prefill with the dynamic forward, then graph-replay per token."""
+        assert self.graph is not None, "call capture() first"
+        m = self.m
+        s = token_ids.shape[1]
+        caches = [(None, None)] * len(m.layers)
+        logits, caches = m.forward(token_ids, kv_caches=caches, pos0=0)
+        for li, (k, v) in enumerate(caches):
+            self.cache_k[li][:, :, :s] = k
+            self.cache_v[li][:, :, :s] = v
+        out = []
+        last = logits[:, -1, :].float()
+        pos = s
+        for _ in range(max_new_tokens):
+            if temperature <= 0:
+                nxt = int(last.argmax(-1))
+            else:
+                sc = last / temperature
+                if top_k:
+                    vv, ix = torch.topk(sc, min(top_k, sc.shape[-1]))
+                    probs = torch.softmax(vv, dim=-1)
+                    nxt = int(ix[0, int(torch.multinomial(probs[0], 1))])
+                else:
+                    nxt = int(torch.multinomial(torch.softmax(sc, -1)[0], 1))
+            out.append(nxt)
+            if stream_cb:
+                stream_cb(nxt)
+            if eos_id is not None and nxt == eos_id:
+                break
+            if pos >= self.max_len:
+                break
+            self.tok.fill_(nxt)
+            self.pos.fill_(pos)
+            self.graph.replay()
+            last = self.out.float()
+            pos += 1
+        return out

@@ -98,3 +98,19 @@ class TestBifrostHTTP:
             r = c.post("/bifrost/stream", json={"prompt": "s", "max_tokens": 2})
             assert "data:" in r.text and "[DONE]" in r.text
         mgr.close()
+
+
+@pytest.mark.gpu
+def test_graphed_decode_matches_eager():
+    from nornicdb_amd.models.heimdall import GraphedDecoder
+    torch.manual_seed(5)
+    cfg = HeimdallConfig.tiny(max_position=128)
+    m = HeimdallModel(cfg).init_small().cuda().eval()
+    for p in m.parameters():
+        if p.dim() > 1:
+            torch.nn.init.normal_(p, 0, 0.05)
+    ids = torch.randint(0, cfg.vocab_size, (1, 6), device="cuda")
+    eager = m.generate(ids.clone(), max_new_tokens=10, temperature=0)
+    gd = GraphedDecoder(m, max_len=64).capture()
+    graphed = gd.generate(ids.clone(), max_new_tokens=10, temperature=0)
+    assert eager == graphed, (eager, graphed)
