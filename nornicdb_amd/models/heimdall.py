@@ -260,30 +260,37 @@ class GraphedDecoder:
         for li, (k, v) in enumerate(caches):
             self.cache_k[li][:, :, :s] = k
             self.cache_v[li][:, :, :s] = v
-        out = []
+        # sampling stays ON DEVICE: no host sync per token unless the
+        # caller streams or uses eos early-exit
+        need_sync = stream_cb is not None or eos_id is not None
+        out_t = []
         last = logits[:, -1, :].float()
         pos = s
-        for _ in range(max_new_tokens):
+
+        def pick(lg):
             if temperature <= 0:
-                nxt = int(last.argmax(-1))
-            else:
-                sc = last / temperature
-                if top_k:
-                    vv, ix = torch.topk(sc, min(top_k, sc.shape[-1]))
-                    probs = torch.softmax(vv, dim=-1)
-                    nxt = int(ix[0, int(torch.multinomial(probs[0], 1))])
-                else:
-                    nxt = int(torch.multinomial(torch.softmax(sc, -1)[0], 1))
-            out.append(nxt)
-            if stream_cb:
-                stream_cb(nxt)
-            if eos_id is not None and nxt == eos_id:
-                break
+                return lg.argmax(-1)                      # [1]
+            sc = lg / temperature
+            if top_k:
+                vv, ix = torch.topk(sc, min(top_k, sc.shape[-1]))
+                j = torch.multinomial(torch.softmax(vv, -1)[0], 1)
+                return ix[0].index_select(0, j)           # [1]
+            return torch.multinomial(torch.softmax(sc, -1)[0], 1)
+
+        for _ in range(max_new_tokens):
+            nxt = pick(last)
+            out_t.append(nxt)
+            if need_sync:
+                nv = int(nxt)
+                if stream_cb:
+                    stream_cb(nv)
+                if eos_id is not None and nv == eos_id:
+                    break
             if pos >= self.max_len:
                 break
-            self.tok.fill_(nxt)
+            self.tok.copy_(nxt.view(1, 1))
             self.pos.fill_(pos)
             self.graph.replay()
             last = self.out.float()
             pos += 1
-        return out
+        return [int(t) for t in torch.cat(out_t).tolist()] if out_t else []
