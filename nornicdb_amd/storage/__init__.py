@@ -12,7 +12,7 @@ from .memory import MemoryEngine
 from .persistent import PersistentEngine, Transaction
 from .async_engine import AsyncEngine
 from .namespaced import NamespacedEngine
-from .wal import WAL, WALCorruption
+from .wal import WAL, WALCorruption, WALDegraded
 from .schema import Constraint, SchemaManager, VectorIndexMeta
 from .composite import CompositeEngine
 
@@ -21,4 +21,5 @@ __all__ = [
     "ConstraintViolation", "new_id", "MemoryEngine", "PersistentEngine",
     "Transaction", "AsyncEngine", "NamespacedEngine", "WAL", "WALCorruption",
     "SchemaManager", "Constraint", "VectorIndexMeta", "CompositeEngine",
+    "WALDegraded",
 ]

@@ -47,6 +47,10 @@ OP_DETACH_DELETE = 13
 OP_CHECKPOINT = 14
 
 
+class WALDegraded(Exception):
+    """Writes refused after an I/O failure; reads keep working."""
+
+
 class WALCorruption(Exception):
     def __init__(self, offset, reason):
         super().__init__(f"WAL corruption at byte {offset}: {reason}")
@@ -61,6 +65,7 @@ class WAL:
         self._lock = threading.Lock()
         self._f = open(path, "ab")
         self._sync_on_write = sync_on_write
+        self.degraded = False
         self._sync_interval = sync_interval
         self._dirty = False
         self._stop = threading.Event()
@@ -74,15 +79,23 @@ class WAL:
             self.sync()
 
     def append(self, op: int, payload: dict) -> None:
+        if self.degraded:
+            raise WALDegraded("WAL is in degraded mode (previous write failed)")
         data = msgpack.packb(payload, use_bin_type=True)
         rec = _HDR.pack(MAGIC, op, len(data), zlib.crc32(data)) + data
         with self._lock:
-            self._f.write(rec)
-            self._dirty = True
-            if self._sync_on_write:
-                self._f.flush()
-                os.fsync(self._f.fileno())
-                self._dirty = False
+            try:
+                self._f.write(rec)
+                self._dirty = True
+                if self._sync_on_write:
+                    self._f.flush()
+                    os.fsync(self._f.fileno())
+                    self._dirty = False
+            except OSError as e:
+                # reference wal_degraded.go: stop accepting writes but keep
+                # the process serving reads
+                self.degraded = True
+                raise WALDegraded(f"WAL write failed: {e}") from e
 
     def sync(self):
         with self._lock:

@@ -200,3 +200,38 @@ class TestInference:
         stats = inf.decay_inferred_edges()
         assert stats["pruned"] == 1
         assert eng.edge_count() == 0
+
+
+class TestKalmanDecayAB:
+    def test_kalman_variant_smooths_decay_scores(self):
+        """A/B comparison (reference kalman_adapter_ab_test.go): under a
+        bursty access pattern the Kalman-filtered decay score must vary
+        less step-to-step than the raw score, while tracking its level."""
+        eng_raw, eng_kal = MemoryEngine(), MemoryEngine()
+        now = [0.0]
+        raw = DecayManager(eng_raw, DecayConfig(use_kalman=False),
+                           now_fn=lambda: now[0])
+        kal = DecayManager(eng_kal, DecayConfig(use_kalman=True),
+                           now_fn=lambda: now[0])
+        for e in (eng_raw, eng_kal):
+            e.create_node(mem("m", 0.0))
+        raw_scores, kal_scores = [], []
+        import random
+        rng = random.Random(0)
+        for step in range(40):
+            now[0] += 86400.0 * rng.choice([0.1, 3.0])  # bursty gaps
+            if rng.random() < 0.4:  # sporadic reinforcement
+                for e in (eng_raw, eng_kal):
+                    n = e.get_node("m")
+                    n.properties["last_accessed"] = now[0]
+                    n.properties["access_count"] += 1
+                    e.update_node(n)
+            raw_scores.append(raw.score(eng_raw.get_node("m")))
+            kal_scores.append(kal.score(eng_kal.get_node("m")))
+
+        def roughness(xs):
+            return sum(abs(b - a) for a, b in zip(xs, xs[1:])) / (len(xs) - 1)
+
+        assert roughness(kal_scores) < roughness(raw_scores)
+        # still tracks the same level
+        assert abs(sum(kal_scores) / 40 - sum(raw_scores) / 40) < 0.15

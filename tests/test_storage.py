@@ -288,3 +288,84 @@ class TestNamespacedEngine:
         a.create_node(mk("mine"))
         b.create_node(mk("theirs"))
         assert seen == ["mine"]
+
+
+class TestDegradedWAL:
+    def test_write_failure_degrades_but_reads_work(self, tmp_path):
+        from nornicdb_amd.storage import WALDegraded
+        e = PersistentEngine(str(tmp_path / "db"), sync_on_write=True,
+                             snapshot_interval=3600)
+        e.create_node(mk("a", "X", v=1))
+        # simulate device failure: close the underlying file
+        e._wal._f.close()
+        with pytest.raises(Exception):
+            e.create_node(mk("b", "X"))
+        # reads still served from RAM (reference wal_degraded.go)
+        assert e.get_node("a").properties["v"] == 1
+        assert e.node_count() == 1
+        e._stop.set()
+
+
+class TestConcurrency:
+    def test_concurrent_cypher_writers_and_readers(self):
+        """Race-style test (reference -race suites): concurrent writers
+        against one engine must not corrupt counts or indexes."""
+        import threading
+        from nornicdb_amd.cypher import Executor
+        eng = MemoryEngine()
+        ex = Executor(eng)
+        errors = []
+
+        def writer(base):
+            try:
+                for i in range(50):
+                    ex.execute("CREATE (:RC {k: $k})", {"k": f"{base}-{i}"})
+            except Exception as exn:  # pragma: no cover
+                errors.append(exn)
+
+        def reader():
+            try:
+                for _ in range(50):
+                    ex.execute("MATCH (n:RC) RETURN count(n)")
+            except Exception as exn:  # pragma: no cover
+                errors.append(exn)
+
+        ts = [threading.Thread(target=writer, args=(b,)) for b in range(4)]
+        ts += [threading.Thread(target=reader) for _ in range(2)]
+        [t.start() for t in ts]
+        [t.join() for t in ts]
+        assert not errors
+        assert eng.node_count() == 200
+        assert eng.node_count_by_label("RC") == 200
+
+    def test_concurrent_embed_queue_and_search(self):
+        import threading
+        from nornicdb_amd.db import open_db
+        from nornicdb_amd.embed import MockEmbedder
+        mgr = open_db(embedder=MockEmbedder(16), dims=16)
+        db = mgr.get()
+        errs = []
+
+        def storer():
+            try:
+                for i in range(30):
+                    db.store(f"doc number {i}")
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        def drainer():
+            try:
+                for _ in range(20):
+                    db.embed_queue._process_batch()
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        ts = [threading.Thread(target=storer) for _ in range(2)]
+        ts += [threading.Thread(target=drainer) for _ in range(2)]
+        [t.start() for t in ts]
+        [t.join() for t in ts]
+        db.embed_queue.drain()
+        assert not errs
+        assert db.engine.node_count() == 60
+        assert len(db.search.emb) == 60
+        mgr.close()
