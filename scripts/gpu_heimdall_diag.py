@@ -34,3 +34,20 @@ print(f"graphed sampled: {128/(time.time()-t0):.0f} tok/s")
 t0 = time.time()
 m.generate(ids, max_new_tokens=64, temperature=0)
 print(f"eager greedy: {64/(time.time()-t0):.0f} tok/s")
+
+# 1M x 1024 Q=1 latency (reference parity: A100 1 ms)
+from nornicdb_amd import ops
+db1 = torch.empty(1_000_000, 1024, device="cuda", dtype=torch.bfloat16)
+ops.fill_random_unit_(db1)
+q1 = db1[:1].clone()
+for _ in range(5): ops.knn_search(db1, q1, 10)
+torch.cuda.synchronize()
+t0 = time.time()
+for _ in range(50): ops.knn_search(db1, q1, 10)
+torch.cuda.synchronize()
+print(f"kNN 1Mx1024 Q=1 k=10: {(time.time()-t0)/50*1000:.3f} ms/query")
+q8b = db1[:8].clone()
+t0 = time.time()
+for _ in range(50): ops.knn_search(db1, q8b, 10)
+torch.cuda.synchronize()
+print(f"kNN 1Mx1024 Q=8 k=10: {(time.time()-t0)/50*1000:.3f} ms/batch")
