@@ -142,6 +142,22 @@ def cmd_shell(args):
     mgr.close()
 
 
+def cmd_eval(args):
+    """IR evaluation over JSON cases (reference cmd/eval)."""
+    from .search.eval import EvalHarness
+    from .utils import load_config
+    cfg = load_config(args.config)
+    mgr = _open(args, cfg)
+    db = mgr.get(args.database)
+    db.embed_queue.drain()
+    harness = EvalHarness(lambda q, k: [r.id for r in
+                                        db.search.search(query=q, k=k)])
+    cases = EvalHarness.load_cases(args.cases)
+    report = harness.run(cases)
+    print(json.dumps(report, indent=2))
+    mgr.close()
+
+
 def cmd_decay(args):
     from .cognitive import DecayManager
     from .utils import load_config
@@ -176,6 +192,12 @@ def main(argv=None):
     sp.add_argument("--database", default=None)
     sp.add_argument("--file", required=True)
     sp.set_defaults(fn=cmd_import)
+
+    sp = sub.add_parser("eval")
+    sp.add_argument("--data-dir", default=None)
+    sp.add_argument("--database", default=None)
+    sp.add_argument("--cases", required=True)
+    sp.set_defaults(fn=cmd_eval)
 
     sp = sub.add_parser("shell")
     sp.add_argument("--data-dir", default=None)

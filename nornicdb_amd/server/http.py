@@ -101,6 +101,15 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
                 raise HTTPException(401, "invalid credentials")
         raise HTTPException(401, "authentication required")
 
+    # ---- embedded admin console (reference ui/ React app -> minimal
+    # single-file console served at /) ----
+    @app.get("/")
+    def console():
+        import os
+        path = os.path.join(os.path.dirname(__file__), "static", "console.html")
+        with open(path) as f:
+            return Response(f.read(), media_type="text/html")
+
     # ---- health / status / metrics ----
     @app.get("/health")
     def health():

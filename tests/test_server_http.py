@@ -140,3 +140,9 @@ class TestMCP:
     def test_unknown_method(self, client):
         r = self._rpc(client, "nope/nope")
         assert r["error"]["code"] == -32601
+
+
+def test_console_served(client):
+    r = client.get("/")
+    assert r.status_code == 200
+    assert "NornicDB-AMD console" in r.text or "console" in r.text
