@@ -118,15 +118,18 @@ __global__ void k_fill_random_unit_bf16(unsigned short* __restrict__ x,
 // ---------------------------------------------------------------------------
 // Fused cosine-score + per-block top-k, GEMV path (Q <= 16).
 //
-// Each wave owns one DB row at a time (grid-stride). Lanes split the D
-// dimension (short8 loads); per-lane partial dot for every query is kept in
-// registers, reduced with xor-shuffles; lane q then folds score(row, q) into
-// a private register top-K list (descending, statically unrolled insert so it
-// stays in VGPRs — see guide rule #20). At the end, each (wave, q) writes its
-// K candidates to cand[(wave_global)][q][K].
+// Each 16-LANE GROUP owns one DB row (4 rows per wave in flight), so the
+// cross-lane reduction is 4 shfl steps instead of 6 and a wave retires 4
+// rows per iteration. Lane g*16+l covers elements [l*64, l*64+64) of the
+// row (8 x short8 loads = 128 B/lane, 2 KB contiguous per group). After
+// the 16-lane reduce, lane l of the group holds score(row, q=l) and folds
+// it into a per-(group, q) register top-K list (statically unrolled
+// insert). Candidates land in cand[group_global][q][K].
 //
 // Queries are staged in LDS as bf16.
 // ---------------------------------------------------------------------------
+typedef __bf16 bf16x2v __attribute__((ext_vector_type(2)));
+
 template <int QMAX, int K>
 __global__ void k_knn_gemv_bf16(const unsigned short* __restrict__ db,
                                 const unsigned short* __restrict__ qs,
@@ -136,6 +139,8 @@ __global__ void k_knn_gemv_bf16(const unsigned short* __restrict__ db,
                                 long long* __restrict__ cand_idx) {
   extern __shared__ unsigned short s_q[];  // [q_count][d]
   const int lane = threadIdx.x & (WAVE - 1);
+  const int grp = lane >> 4;        // 4 row-groups per wave
+  const int gl = lane & 15;         // lane within group
   const int wid = threadIdx.x / WAVE;
   const int waves_per_block = blockDim.x / WAVE;
 
@@ -148,45 +153,55 @@ __global__ void k_knn_gemv_bf16(const unsigned short* __restrict__ db,
 
   const long long wave_global = (long long)blockIdx.x * waves_per_block + wid;
   const long long total_waves = (long long)gridDim.x * waves_per_block;
+  const long long group_global = wave_global * 4 + grp;
+  const long long total_groups = total_waves * 4;
+  const int nj = d / 128;           // 16 lanes x 8 elems per j-step
 
-  // private top-K (valid in lane q for q < q_count)
+  // private top-K (valid in lane gl == q for q < q_count)
   float tv[K];
   long long ti[K];
 #pragma unroll
   for (int i = 0; i < K; ++i) { tv[i] = -1e30f; ti[i] = -1; }
 
-  for (long long row = wave_global; row < n; row += total_waves) {
-    const unsigned short* rp = db + row * d;
+  for (long long row = group_global; row < n; row += total_groups) {
+    const unsigned short* rp = db + row * d + (long long)gl * 8;
     float acc[QMAX];
 #pragma unroll
     for (int q = 0; q < QMAX; ++q) acc[q] = 0.0f;
 
-    for (int c = lane * 8; c < d; c += WAVE * 8) {
-      short8v v = *reinterpret_cast<const short8v*>(rp + c);
-      float xv[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) xv[j] = bf16_bits_to_f32((unsigned short)v[j]);
+    // lane gl reads elems [jj*128 + gl*8 .. +8): 16 lanes cover a
+    // contiguous 256 B segment per step (coalesced)
+#pragma unroll 4
+    for (int jj = 0; jj < nj; ++jj) {
+      short8v v = *reinterpret_cast<const short8v*>(rp + jj * 128);
+      const bf16x2v* xa = reinterpret_cast<const bf16x2v*>(&v);
 #pragma unroll
       for (int q = 0; q < QMAX; ++q) {
         if (q >= q_count) break;
-        short8v qv = *reinterpret_cast<const short8v*>(s_q + q * d + c);
+        short8v qv = *reinterpret_cast<const short8v*>(
+            s_q + q * d + jj * 128 + gl * 8);
+        const bf16x2v* qa = reinterpret_cast<const bf16x2v*>(&qv);
+        // v_dot2c_f32_bf16: 2 bf16 MACs/instr, f32 accumulate — no
+        // explicit converts, 4 instrs per 8 elems per query
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          acc[q] += xv[j] * bf16_bits_to_f32((unsigned short)qv[j]);
+        for (int j = 0; j < 4; ++j)
+          acc[q] = __builtin_amdgcn_fdot2_f32_bf16(xa[j], qa[j], acc[q], false);
       }
     }
-    // reduce each query's partials across the wave; result in all lanes.
+    // reduce each query's partials across the 16-lane group
 #pragma unroll
     for (int q = 0; q < QMAX; ++q) {
       if (q >= q_count) break;
-      acc[q] = wave_reduce_sum(acc[q]);
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        acc[q] += __shfl_xor(acc[q], off, WAVE);
     }
-    // lane q keeps the score for query q.
-    if (lane < q_count) {
+    // lane gl keeps the score for query gl
+    if (gl < q_count) {
       float s = acc[0];
 #pragma unroll
       for (int q = 1; q < QMAX; ++q)
-        if (lane == q) s = acc[q];
+        if (gl == q) s = acc[q];
       if (s > tv[K - 1]) {
         float cs = s; long long ci = row_base + row;
 #pragma unroll
@@ -201,8 +216,8 @@ __global__ void k_knn_gemv_bf16(const unsigned short* __restrict__ db,
     }
   }
 
-  if (lane < q_count) {
-    long long base = (wave_global * q_count + lane) * K;
+  if (gl < q_count) {
+    long long base = (group_global * q_count + gl) * K;
 #pragma unroll
     for (int i = 0; i < K; ++i) {
       cand_score[base + i] = tv[i];
@@ -361,12 +376,12 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
   int qc = (int)q.size(0);
   TORCH_CHECK(q.size(1) == d, "dim mismatch");
   TORCH_CHECK(qc >= 1 && qc <= 16, "knn_gemv supports 1..16 queries");
-  TORCH_CHECK(d % 8 == 0, "knn_gemv needs D % 8 == 0");
+  TORCH_CHECK(d % 128 == 0, "knn_gemv needs D % 128 == 0");
   TORCH_CHECK(k_out >= 1 && k_out <= KNN_K, "k_out must be <= ", KNN_K);
 
-  int blocks = (int)std::min<long long>((n + 255) / 256, 1280);
+  int blocks = (int)std::min<long long>((n + 1023) / 1024, 1280);
   blocks = std::max(blocks, 1);
-  long long waves = (long long)blocks * 4;
+  long long waves = (long long)blocks * 4 * 4;  // 4 waves x 4 row-groups
 
   auto opts_f = db.options().dtype(at::kFloat);
   auto opts_i = db.options().dtype(at::kLong);

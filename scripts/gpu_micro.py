@@ -40,6 +40,10 @@ def main():
     dt = t(lambda: ops.knn_search(x, q1, 10))
     print(f"knn_gemv Q=1: {dt*1e3:.2f} ms  {gb/dt:.0f} GB/s  {1/dt:.0f} qps")
 
+    q16 = x[:16].clone()
+    dt = t(lambda: ops.knn_search(x, q16, 10))
+    print(f"knn_gemv Q=16: {dt*1e3:.2f} ms  {gb/dt:.0f} GB/s  {16/dt:.0f} qps")
+
     q256 = x[:256].clone()
     dt = t(lambda: ops.knn_search(x, q256, 10))
     print(f"knn fused-MFMA Q=256: {dt*1e3:.2f} ms  {gb/dt:.0f} GB/s(db)  {256/dt:.0f} qps")
