@@ -263,6 +263,51 @@ class IdMap:
         return self._i2s.get(i, str(i))
 
 
+# ---- temporal structures (Bolt 4.4/5.x PackStream spec) ----
+DATE_TAG = 0x44            # 'D' days since epoch
+TIME_TAG = 0x54            # 'T' nanos-of-day + tz offset seconds
+LOCAL_TIME_TAG = 0x74      # 't'
+DATETIME_TAG = 0x49        # 'I' UTC epoch seconds + nanos + offset (Bolt 5)
+DATETIME_LEGACY_TAG = 0x46  # 'F' local-epoch seconds variant (Bolt 4)
+LOCAL_DATETIME_TAG = 0x64  # 'd'
+DURATION_TAG = 0x45        # 'E' months, days, seconds, nanoseconds
+
+
+def temporal_struct(v, bolt5: bool = False):
+    """Convert a cypher temporal value to its Bolt structure (or None)."""
+    import datetime as _dt
+
+    from ..cypher import temporal as _tp
+
+    if isinstance(v, _tp.CypherDuration):
+        return Structure(DURATION_TAG,
+                         [v.months, v.days, v.seconds, v.nanoseconds])
+    if isinstance(v, _tp.CypherDate):
+        days = (v.date - _dt.date(1970, 1, 1)).days
+        return Structure(DATE_TAG, [days])
+    if isinstance(v, _tp.CypherDateTime):
+        dt = v._v
+        if dt.tzinfo is None:
+            epoch = int(dt.replace(tzinfo=_dt.timezone.utc).timestamp())
+            return Structure(LOCAL_DATETIME_TAG,
+                             [epoch, dt.microsecond * 1000])
+        off = int(dt.utcoffset().total_seconds())
+        secs = int(dt.timestamp())
+        if bolt5:
+            return Structure(DATETIME_TAG, [secs, dt.microsecond * 1000, off])
+        return Structure(DATETIME_LEGACY_TAG,
+                         [secs + off, dt.microsecond * 1000, off])
+    if isinstance(v, _tp.CypherTime):
+        dt = v._v
+        nanos = ((dt.hour * 3600 + dt.minute * 60 + dt.second) * 1_000_000
+                 + dt.microsecond) * 1000
+        if dt.tzinfo is None:
+            return Structure(LOCAL_TIME_TAG, [nanos])
+        return Structure(TIME_TAG,
+                         [nanos, int(dt.utcoffset().total_seconds())])
+    return None
+
+
 def node_struct(node, ids: IdMap, bolt5: bool = False) -> Structure:
     fields = [ids.to_int(node.id), list(node.labels), dict(node.properties)]
     if bolt5:

@@ -118,6 +118,9 @@ class BoltSession:
             return {k: self.to_bolt(x) for k, x in v.items()}
         if isinstance(v, float) and v != v:  # NaN
             return None
+        ts = ps.temporal_struct(v, self.bolt5)
+        if ts is not None:
+            return ts
         return v
 
     # ---- message handlers ----

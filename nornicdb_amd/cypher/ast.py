@@ -100,6 +100,13 @@ class CountSubquery:
     where: Optional[Any] = None
 
 
+@dataclass
+class SubqueryExpr:
+    """EXISTS { ... } / COUNT { ... } expression subquery (Cypher 5)."""
+    kind: str              # "EXISTS" | "COUNT" | "COLLECT"
+    query: "Query"
+
+
 # ---- patterns ----
 @dataclass
 class NodePattern:
@@ -202,6 +209,14 @@ class CallClause:
     args: List[Any]
     yields: List[Tuple[str, Optional[str]]]  # (name, alias)
     where: Optional[Any] = None
+
+
+@dataclass
+class SubqueryCallClause:
+    """CALL { ... } [IN TRANSACTIONS [OF n ROWS]] clause subquery."""
+    query: "Query"
+    in_transactions: bool = False
+    rows_per_tx: int = 1000
 
 
 @dataclass

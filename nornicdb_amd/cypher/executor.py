@@ -262,12 +262,57 @@ class Executor:
                 rows, call_out = self._exec_call(c, rows, params, standalone=last)
                 if call_out is not None:
                     out_cols, out_rows = call_out
+            elif isinstance(c, A.SubqueryCallClause):
+                rows = self._exec_subquery_call(c, rows, params)
             elif isinstance(c, A.ForeachClause):
                 rows = self._exec_foreach(c, rows, params)
             else:
                 raise CypherRuntimeError(f"unsupported clause {type(c).__name__}")
             i += 1
         return out_cols, out_rows
+
+    def _run_subquery(self, q, bindings, params):
+        """Execute a subquery AST with the outer row's bindings visible
+        (superset of Cypher's WITH-import rule). Returns (cols, rows)."""
+        save = getattr(self, "_initial_bindings", None)
+        self._initial_bindings = bindings
+        try:
+            cols, rows = self._run_clauses(q.clauses, params)
+            node = q
+            while node.union is not None:
+                kind, nxt = node.union
+                c2, r2 = self._run_clauses(nxt.clauses, params)
+                rows = rows + r2
+                if kind == "UNION":
+                    seen, uniq = set(), []
+                    for r in rows:
+                        k = tuple(_hkey(v) for v in r)
+                        if k not in seen:
+                            seen.add(k)
+                            uniq.append(r)
+                    rows = uniq
+                node = nxt
+        finally:
+            self._initial_bindings = save
+        return cols, rows
+
+    def _exec_subquery_call(self, c, rows, params):
+        """CALL { ... }: run the inner query once per incoming row; returned
+        columns are appended to the row (cartesian with inner results).
+        A unit subquery (no RETURN) passes rows through unchanged.
+        IN TRANSACTIONS executes eagerly (single-process engine; batching
+        only bounds memory, which list processing already does)."""
+        out = []
+        for row in rows:
+            cols, subrows = self._run_subquery(c.query, dict(row), params)
+            if cols:
+                for sr in subrows:
+                    nr = dict(row)
+                    nr.update(zip(cols, sr))
+                    out.append(nr)
+            else:
+                out.append(row)
+        return out
 
     # -------------------------------------------------------------- helpers
     def _eval(self, e, row, params):
@@ -290,6 +335,11 @@ class Executor:
                 return base.properties.get(e.key)
             if isinstance(base, dict):
                 return base.get(e.key)
+            if hasattr(base, "component"):  # temporal values
+                try:
+                    return base.component(e.key)
+                except KeyError:
+                    return None
             raise CypherRuntimeError(f"cannot access .{e.key} on {type(base).__name__}")
         if isinstance(e, A.BinOp):
             return self._eval_binop(e, row, params)
@@ -360,6 +410,13 @@ class Executor:
             for _ in self._match_path(e.pattern, dict(row), params, limit=1):
                 return True
             return False
+        if isinstance(e, A.SubqueryExpr):
+            cols, rows2 = self._run_subquery(e.query, dict(row), params)
+            if e.kind == "EXISTS":
+                return len(rows2) > 0
+            if e.kind == "COUNT":
+                return len(rows2)
+            return [r[0] for r in rows2]  # COLLECT
         raise CypherRuntimeError(f"cannot evaluate {type(e).__name__}")
 
     def _eval_binop(self, e, row, params):

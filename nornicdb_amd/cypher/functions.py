@@ -68,6 +68,31 @@ register("sign", lambda x: None if x is None else (0 if x == 0 else math.copysig
 register("rand", lambda: random.random())
 register("randomuuid", lambda: str(uuid.uuid4()))
 register("timestamp", lambda: int(time.time() * 1000))
+
+# ---- temporal (see temporal.py) ----
+from . import temporal as _tp  # noqa: E402
+
+register("date", lambda arg=None: _tp.make_date(arg))
+register("datetime", lambda arg=None: _tp.make_datetime(arg))
+register("localdatetime", lambda arg=None: _tp.make_datetime(arg, local=True))
+register("time", lambda arg=None: _tp.make_time(arg))
+register("localtime", lambda arg=None: _tp.make_time(arg, local=True))
+register("duration", lambda arg: _tp.make_duration(arg))
+register("duration.between", lambda a, b: _tp.duration_between(a, b))
+register("duration.indays", lambda a, b: _tp.CypherDuration(
+    0, _tp.duration_between(a, b).days, 0, 0))
+register("duration.inseconds", lambda a, b: _tp.CypherDuration(
+    0, 0, int(_tp.duration_between(a, b).total_seconds_approx()), 0))
+register("duration.inmonths", lambda a, b: _tp.CypherDuration(
+    _tp.duration_between(a, b).days // 30, 0, 0, 0))
+register("datetime.truncate", lambda unit, v=None:
+         _tp.truncate(unit, v if v is not None else _tp.make_datetime(None)))
+register("date.truncate", lambda unit, v=None:
+         _tp.truncate(unit, v if v is not None else _tp.make_datetime(None),
+                      kind="date"))
+register("datetime.fromepoch", lambda s, ns=0: _tp.make_datetime(
+    (float(s) + float(ns) / 1e9) * 1000))
+register("datetime.fromepochmillis", lambda ms: _tp.make_datetime(float(ms)))
 register("toInteger".lower(), lambda x: _to_int(x))
 register("tofloat", lambda x: _to_float(x))
 register("tostring", lambda x: None if x is None else

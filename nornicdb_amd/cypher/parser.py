@@ -287,6 +287,29 @@ class Parser:
 
     def _call(self):
         self.eat_kw("CALL")
+        if self.at_op("{"):
+            q = self._braced_query()
+            in_tx = False
+            rows_per_tx = 1000
+            if self.at_kw("IN"):
+                self.next()
+                t = self.next()  # TRANSACTIONS (not a reserved word)
+                if t.value.upper() != "TRANSACTIONS":
+                    raise CypherSyntaxError(
+                        f"expected TRANSACTIONS, got {t.value!r} at {t.pos}")
+                in_tx = True
+                if (self.peek().kind == "IDENT"
+                        and self.peek().value.upper() == "OF"):
+                    self.next()
+                    n = self._expr()
+                    t = self.next()
+                    if t.value.upper() != "ROWS":
+                        raise CypherSyntaxError(
+                            f"expected ROWS, got {t.value!r} at {t.pos}")
+                    if isinstance(n, A.Lit):
+                        rows_per_tx = int(n.value)
+            return A.SubqueryCallClause(q, in_transactions=in_tx,
+                                        rows_per_tx=rows_per_tx)
         name = self.ident()
         while self.try_op("."):
             name += "." + self.name_part()
@@ -311,6 +334,28 @@ class Parser:
             if self.try_kw("WHERE"):
                 where = self._expr()
         return A.CallClause(name, args, yields, where)
+
+    def _braced_query(self) -> A.Query:
+        """Parse `{ <clauses> }`; a bare pattern (EXISTS shorthand) is
+        desugared to MATCH pattern [WHERE ...] RETURN 1."""
+        self.eat_op("{")
+        t = self.peek()
+        if t.kind == "KW" and t.value in (
+                "MATCH", "OPTIONAL", "WITH", "UNWIND", "CALL", "CREATE",
+                "MERGE", "RETURN", "FOREACH", "SET", "DELETE", "DETACH"):
+            q = self._query()
+        else:
+            pat = self._pattern_path()
+            where = self._expr() if self.try_kw("WHERE") else None
+            q = A.Query([A.MatchClause([pat], where=where)])
+        self.eat_op("}")
+        return q
+
+    @staticmethod
+    def _ensure_return(q: A.Query) -> A.Query:
+        if not any(isinstance(c, A.ReturnClause) for c in q.clauses):
+            q.clauses.append(A.ReturnClause([A.ReturnItem(A.Lit(1), "one")]))
+        return q
 
     def _foreach(self):
         self.eat_kw("FOREACH")
@@ -596,6 +641,10 @@ class Parser:
             if t.value == "FALSE":
                 self.next()
                 return A.Lit(False)
+            if t.value == "COUNT" and self.peek(1).kind == "OP" and self.peek(1).value == "{":
+                self.next()
+                return A.SubqueryExpr(
+                    "COUNT", self._ensure_return(self._braced_query()))
             if t.value == "COUNT" and self.peek(1).kind == "OP" and self.peek(1).value == "(":
                 self.next()
                 self.eat_op("(")
@@ -610,6 +659,9 @@ class Parser:
                 return self._case()
             if t.value == "EXISTS":
                 self.next()
+                if self.at_op("{"):
+                    return A.SubqueryExpr(
+                        "EXISTS", self._ensure_return(self._braced_query()))
                 self.eat_op("(")
                 if self.at_op("("):
                     pat = self._pattern_path()

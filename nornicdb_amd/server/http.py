@@ -40,6 +40,8 @@ def _jsonable(v):
         return {k: _jsonable(x) for k, x in v.items()}
     if isinstance(v, float) and (v != v or v in (float("inf"), float("-inf"))):
         return None
+    if hasattr(v, "component"):  # temporal values -> ISO-8601 strings
+        return str(v)
     return v
 
 

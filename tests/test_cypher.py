@@ -276,3 +276,146 @@ class TestForeach:
     def test_foreach_create(self, ex):
         ex.execute("FOREACH (i IN range(1, 3) | CREATE (:FE {v: i}))")
         assert ex.execute("MATCH (n:FE) RETURN count(n)").rows == [[3]]
+
+
+# ---------------------------------------------------------------- subqueries
+class TestSubqueries:
+    """CALL {} / EXISTS {} / COUNT {} (Cypher 5 subqueries; reference
+    pkg/cypher supports these through its openCypher layer)."""
+
+    def _ex(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        ex.execute("CREATE (a:Person {name:'ann'})-[:KNOWS]->"
+                   "(b:Person {name:'bob'}), (c:Person {name:'cat'})")
+        return ex
+
+    def test_exists_subquery_match(self):
+        ex = self._ex()
+        r = ex.execute("MATCH (p:Person) WHERE EXISTS { MATCH (p)-[:KNOWS]->() } "
+                       "RETURN p.name")
+        assert r.rows == [["ann"]]
+
+    def test_exists_subquery_pattern_shorthand(self):
+        ex = self._ex()
+        r = ex.execute("MATCH (p:Person) WHERE EXISTS { (p)-[:KNOWS]->(q) "
+                       "WHERE q.name = 'bob' } RETURN p.name")
+        assert r.rows == [["ann"]]
+
+    def test_count_subquery(self):
+        ex = self._ex()
+        r = ex.execute("MATCH (p:Person) RETURN p.name, "
+                       "COUNT { MATCH (p)-[:KNOWS]->() } AS deg ORDER BY p.name")
+        assert r.rows == [["ann", 1], ["bob", 0], ["cat", 0]]
+
+    def test_call_subquery_returning(self):
+        ex = self._ex()
+        r = ex.execute("MATCH (p:Person) CALL { WITH p MATCH (p)-[:KNOWS]->(q) "
+                       "RETURN q.name AS friend } RETURN p.name, friend")
+        assert r.rows == [["ann", "bob"]]
+
+    def test_call_subquery_expression(self):
+        ex = self._ex()
+        r = ex.execute("UNWIND [1,2,3] AS x CALL { WITH x RETURN x*10 AS y } "
+                       "RETURN x, y")
+        assert r.rows == [[1, 10], [2, 20], [3, 30]]
+
+    def test_call_subquery_unit_in_transactions(self):
+        ex = self._ex()
+        r = ex.execute("UNWIND range(1,5) AS x CALL { WITH x "
+                       "CREATE (:Batch {v:x}) } IN TRANSACTIONS OF 2 ROWS "
+                       "RETURN count(x) AS n")
+        assert r.rows == [[5]]
+        assert ex.execute("MATCH (b:Batch) RETURN count(*)").rows == [[5]]
+
+    def test_call_subquery_aggregates_per_row(self):
+        ex = self._ex()
+        r = ex.execute("MATCH (p:Person) CALL { WITH p "
+                       "MATCH (p)-[:KNOWS]->(q) RETURN count(q) AS c } "
+                       "RETURN p.name, c ORDER BY p.name")
+        assert r.rows == [["ann", 1], ["bob", 0], ["cat", 0]]
+
+
+# ---------------------------------------------------------------- temporal
+class TestTemporal:
+    def _ex(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        return Executor(MemoryEngine())
+
+    def test_date_accessors(self):
+        r = self._ex().execute(
+            "RETURN date('2026-09-12').year, date('2026-09-12').month, "
+            "date('2026-09-12').dayOfWeek, date('2026-09-12').quarter")
+        assert r.rows == [[2026, 9, 6, 3]]
+
+    def test_date_plus_duration_clamps_month_end(self):
+        r = self._ex().execute(
+            "RETURN toString(date('2026-01-31') + duration('P1M'))")
+        assert r.rows == [["2026-02-28"]]
+
+    def test_datetime_minus_duration_map(self):
+        r = self._ex().execute(
+            "RETURN toString(datetime('2026-09-12T10:00:00Z') - "
+            "duration({hours: 1, minutes: 30}))")
+        assert r.rows == [["2026-09-12T08:30:00+00:00"]]
+
+    def test_duration_between(self):
+        # Neo4j semantics: components don't roll up (P1DT6H -> days 1, hours 6)
+        r = self._ex().execute(
+            "RETURN duration.between(datetime('2026-01-01T00:00:00Z'), "
+            "datetime('2026-01-02T06:00:00Z')).days, "
+            "duration.between(datetime('2026-01-01T00:00:00Z'), "
+            "datetime('2026-01-02T06:00:00Z')).hours")
+        assert r.rows == [[1, 6]]
+
+    def test_datetime_map_constructor_epoch(self):
+        r = self._ex().execute(
+            "RETURN datetime({year: 2026, month: 9, day: 12, hour: 8})"
+            ".epochMillis")
+        assert r.rows == [[1789200000000]]
+
+    def test_truncate(self):
+        r = self._ex().execute(
+            "RETURN toString(datetime.truncate('month', "
+            "datetime('2026-09-12T10:11:12Z')))")
+        assert r.rows == [["2026-09-01T00:00:00+00:00"]]
+
+    def test_comparison_and_scaling(self):
+        ex = self._ex()
+        assert ex.execute(
+            "RETURN date('2026-09-12') < date('2026-10-01')").rows == [[True]]
+        assert ex.execute(
+            "RETURN toString(duration('P1Y2M3DT4H5M6S') * 2)"
+        ).rows == [["P2Y4M6DT8H10M12S"]]
+
+    def test_duration_roundtrip_parse_print(self):
+        ex = self._ex()
+        assert ex.execute("RETURN duration('P3DT4H').seconds").rows == [[14400]]
+        assert ex.execute("RETURN toString(duration({days: 3, hours: 4}))"
+                          ).rows == [["P3DT4H"]]
+
+    def test_bolt_temporal_structs(self):
+        import datetime as dt
+
+        from nornicdb_amd.bolt import packstream as ps
+        from nornicdb_amd.cypher import temporal as tp
+        d = ps.temporal_struct(tp.make_date("2026-09-12"))
+        assert d.tag == ps.DATE_TAG
+        assert d.fields == [(dt.date(2026, 9, 12) - dt.date(1970, 1, 1)).days]
+        dur = ps.temporal_struct(tp.make_duration({"days": 2, "seconds": 5}))
+        assert dur.tag == ps.DURATION_TAG and dur.fields == [0, 2, 5, 0]
+        z = ps.temporal_struct(tp.make_datetime("2026-09-12T08:00:00Z"),
+                               bolt5=True)
+        assert z.tag == ps.DATETIME_TAG and z.fields[2] == 0
+        # structures survive the packer roundtrip (python oracle)
+        blob = ps.pack_py(d)
+        back = ps.unpack(blob)
+        assert back == d
+
+    def test_http_jsonable_temporal(self):
+        from nornicdb_amd.cypher import temporal as tp
+        from nornicdb_amd.server.http import _jsonable
+        assert _jsonable(tp.make_date("2026-09-12")) == "2026-09-12"
+        assert _jsonable({"d": tp.make_duration("PT5S")}) == {"d": "PT5S"}
