@@ -5,15 +5,22 @@
 // [N, Q] score matrix is never materialized in HBM. The DB shard is read
 // exactly once per query batch.
 //
-// Structure: m97-template GEMM from the CDNA4 guide (§5) —
-//   * 64x256 output tile, 4 waves (256 thr), wave-tile 64x64,
-//     mfma_f32_16x16x32_bf16 fragments, BK=64 K-steps
-//   * global -> LDS staging via __builtin_amdgcn_global_load_lds width 16
-//   * 2-barrier K-loop; LDS score buffer UNIONed over the staging tiles
-//     (40 KB/block -> 3 blocks/CU occupancy)
-//   * epilogue: C chunks bounce through LDS; 1 thread per query column
-//     scans rows into a private register top-K list (statically unrolled
-//     insertion so it stays in VGPRs), candidates merged by k_topk_merge_i32.
+// Structure (chosen by ablation, scripts/knn8p.hip — see profiles/README.md):
+//   * 96x256 output tile, 4 waves (256 thr), 3 workgroups/CU. Occupancy is
+//     the LDS-DMA throughput lever on gfx950: global_load_lds ingest
+//     scales with resident waves (1-WG/8-wave kernels cap at ~3 TB/s
+//     aggregate DMA; 12 waves across 3 WGs reach ~8.8 TB/s), so the
+//     "big-tile 1-WG pipelined" template loses to 3 small co-resident WGs.
+//   * row-XOR LDS swizzle: 16-B slot index XORed with (row & 7) inside
+//     each 128-B row; kills the 16-way ds_read bank conflict of
+//     16-consecutive-rows-at-one-column fragment reads (measured:
+//     SQ_LDS_BANK_CONFLICT -> 0, +9% kernel).
+//   * mfma_f32_16x16x32_bf16 fragments, BK=64 K-steps, staging via
+//     __builtin_amdgcn_global_load_lds width 16.
+//   * transposed epilogue: acc fragments land in a [col][36] fp32 LDS
+//     chunk with float4 stores/scans (4 consecutive rows per C-register
+//     quad); each thread owns one query column and keeps a register
+//     top-K list (statically unrolled insertion).
 //
 // Replaces the reference's cublasSgemv + single-thread top-k scan
 // (reference: pkg/gpu/cuda/cuda_kernels.cu:340-480).
@@ -28,14 +35,18 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define L_AS __attribute__((address_space(3)))
 
 // tile geometry
-#define BM 64
+#define BM 96
 #define BN 256
 #define BK 64
 #define NTHREADS 256
-#define KCAND 12
-// LDS score-scan chunk: rows per chunk
-#define SCH 32
-#define S_STRIDE (BN + 4)
+#define KCAND 10
+#define MW (BM / 16)
+
+// row-XOR swizzle (self-inverse): within a 128 B row, XOR the 16 B-slot
+// index with (row & 7).
+__device__ __forceinline__ int swz(int b) {
+  return (b & ~127) | ((b & 127) ^ (((b >> 7) & 7) << 4));
+}
 
 __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
@@ -43,17 +54,16 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
     int d,               // inner dim, % BK == 0
     long long row_base, float* __restrict__ cand_score,
     int* __restrict__ cand_idx) {
-  // 40 KB union: K-loop uses sA (8 KB) + sB (32 KB); epilogue reuses the
-  // same space as the 32x260 fp32 score chunk (33.3 KB). Barriers separate
-  // the two lifetimes.
-  __shared__ __align__(16) char smem[(BM * BK + BN * BK) * 2];
+  // 44 KB: K-loop uses sA (12 KB) + sB (32 KB); epilogue reuses the same
+  // space as the [256][36] fp32 transposed chunk (36.9 KB).
+  __shared__ __align__(16) char smem[BM * BK * 2 + BN * BK * 2];
   unsigned short* sA = (unsigned short*)smem;
   unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
-  float* sS = (float*)smem;
+  float* sT = (float*)smem;  // epilogue alias [256][36]
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
-  const int wc = tid / WAVE;  // wave col 0..3 (64 cols each); all waves row 0
+  const int wc = tid / WAVE;  // wave col 0..3 (64 cols each)
 
   // per-thread top-K state: this thread owns query column tid.
   float tv[KCAND];
@@ -66,35 +76,36 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
   for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
     const long long prow = panel * BM;
 
-    float4v acc[4][4];
+    float4v acc[MW][4];
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < MW; ++m)
 #pragma unroll
       for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
 
     for (int kt = 0; kt < d; kt += BK) {
-      // ---- stage A (64 x 64 = 8 KB) and B (256 x 64 = 32 KB) ----
-      // One global_load_lds issue = one wave writes 64 lanes x 16 B = 1 KB
-      // at a wave-uniform LDS base (linear row-major tiles).
+      // ---- stage A (96x64 = 12 KB) and B (256x64 = 32 KB) ----
+      // One global_load_lds = one wave writes 64 lanes x 16 B = 1 KB at a
+      // wave-uniform LDS base; the global source address is pre-swizzled
+      // so the LDS image is the swizzled layout.
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {  // A: 8 chunks, 4 waves x 2
-        int chunk = wc * 2 + it;
-        int byte_off = chunk * 1024 + lane * 16;
-        int r = byte_off / (BK * 2);
-        int cb = byte_off % (BK * 2);
+      for (int it = 0; it < BM / 32; ++it) {  // A: BM/8 chunks, 4 waves
+        int chunk = wc * (BM / 32) + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz(x);
         const G_AS unsigned int* gp = (const G_AS unsigned int*)(
-            (const char*)db + (prow + r) * d2 + (long long)kt * 2 + cb);
+            (const char*)db + (prow + (p >> 7)) * d2 + (long long)kt * 2 +
+            (p & 127));
         L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
 #pragma unroll
       for (int it = 0; it < 8; ++it) {  // B: 32 chunks, 4 waves x 8
         int chunk = wc * 8 + it;
-        int byte_off = chunk * 1024 + lane * 16;
-        int r = byte_off / (BK * 2);
-        int cb = byte_off % (BK * 2);
+        int x = chunk * 1024 + lane * 16;
+        int p = swz(x);
         const G_AS unsigned int* gp = (const G_AS unsigned int*)(
-            (const char*)qs + (long long)r * d2 + (long long)kt * 2 + cb);
+            (const char*)qs + (long long)(p >> 7) * d2 + (long long)kt * 2 +
+            (p & 127));
         L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
@@ -102,61 +113,61 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
 
       // ---- MFMA over the staged tile: 2 k-steps of 32 ----
 #pragma unroll
-      for (int ks = 0; ks < BK / 32; ++ks) {
-        bf16x8 af[4], bf[4];
-#pragma unroll
-        for (int m = 0; m < 4; ++m) {
-          int r = m * 16 + (lane & 15);
-          int k = ks * 32 + (lane >> 4) * 8;
-          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(sA + r * BK + k));
-        }
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+        bf16x8 bfr[4];
 #pragma unroll
         for (int nn = 0; nn < 4; ++nn) {
           int c = wc * 64 + nn * 16 + (lane & 15);
-          int k = ks * 32 + (lane >> 4) * 8;
-          bf[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(sB + c * BK + k));
+          bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sB + swz(c * 128 + kb)));
         }
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < MW; ++m) {
+          int r = m * 16 + (lane & 15);
+          bf16x8 af = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sA + swz(r * 128 + kb)));
 #pragma unroll
           for (int nn = 0; nn < 4; ++nn)
             acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[m], bf[nn], acc[m][nn], 0, 0, 0);
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
+        }
       }
       __syncthreads();
     }
 
-    // ---- epilogue: 2 chunks of 32 rows through LDS (aliases sA/sB) ----
+    // ---- epilogue: 32-row chunks through the transposed LDS block ----
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int h = 0; h < BM / 32; ++h) {
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
         int m = h * 2 + mi;
 #pragma unroll
         for (int nn = 0; nn < 4; ++nn) {
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            int srow = mi * 16 + (lane >> 4) * 4 + r;  // within 32-row chunk
-            int col = wc * 64 + nn * 16 + (lane & 15);
-            sS[srow * S_STRIDE + col] = acc[m][nn][r];
-          }
+          int col = wc * 64 + nn * 16 + (lane & 15);
+          int srow = mi * 16 + (lane >> 4) * 4;
+          *reinterpret_cast<float4v*>(sT + col * 36 + srow) = acc[m][nn];
         }
       }
       __syncthreads();
-      const long long grow0 = prow + (long long)h * SCH;
+      const long long grow0 = prow + (long long)h * 32;
 #pragma unroll
-      for (int r = 0; r < SCH; ++r) {
-        float s = sS[r * S_STRIDE + tid];
-        if (s > tv[KCAND - 1]) {
-          float cs = s;
-          int ci = (int)(grow0 + r);
+      for (int j = 0; j < 32; j += 4) {
+        float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
 #pragma unroll
-          for (int i = 0; i < KCAND; ++i) {
-            bool ins = cs > tv[i];
-            float ts = tv[i]; int tj = ti[i];
-            tv[i] = ins ? cs : tv[i];
-            ti[i] = ins ? ci : ti[i];
-            cs = ins ? ts : cs; ci = ins ? tj : ci;
+        for (int e = 0; e < 4; ++e) {
+          float s = v[e];
+          if (s > tv[KCAND - 1]) {
+            float cs = s;
+            int ci = (int)(grow0 + j + e);
+#pragma unroll
+            for (int i = 0; i < KCAND; ++i) {
+              bool ins = cs > tv[i];
+              float ts = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i];
+              ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts : cs; ci = ins ? tj : ci;
+            }
           }
         }
       }

@@ -1,0 +1,1100 @@
+// 256-square 8-wave fused kNN probe (guide §5 "256² 8-phase template").
+// Compile: hipcc --offload-arch=gfx950 -O3 -std=c++17 knn8p.hip -o knn8p
+//
+// Variants (template<int V>):
+//   0 = 256x256 tile, 8 waves, whole-K-tile double-buffer, counted
+//       vmcnt(8), 2-barrier per K-tile, quadrant MFMA order + setprio
+//   1 = V0 + st_16x32 LDS swizzle (pre-swizzled global src + swizzled
+//       ds_read addresses)
+//   2 = V1 + fine-grained phase barriers (4 sub-phases per K-tile, each
+//       with its own barrier pair — approximates the 8-phase schedule)
+//
+// Epilogue: 4 rounds (one per wave_n group); writers dump acc fragments
+// into a transposed [64 col][260 row-stride] fp32 LDS block (b128 stores,
+// 4 consecutive rows per fragment register quad); each thread owns ONE
+// query column for the whole kernel and K-inserts 128 rows per round from
+// contiguous b128 reads. Candidates: [grid*2][256][KC].
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <algorithm>
+#include <type_traits>
+
+#define WAVE 64
+typedef short short8v __attribute__((ext_vector_type(8)));
+typedef float float4v __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+#define G_AS __attribute__((address_space(1)))
+#define L_AS __attribute__((address_space(3)))
+
+#define BM 256
+#define BN 256
+#define BK 64
+#define NT 512
+#define KC 10
+#define ESTR 260  // epilogue transposed row stride (256 + 4)
+
+__global__ void fill_rand(unsigned short* x, long long n) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned long long h = i * 0x9e3779b97f4a7c15ULL;
+    h ^= h >> 33; h *= 0xff51afd7ed558ccdULL; h ^= h >> 33;
+    float f = ((float)(unsigned)(h & 0xffffff) / 8388608.0f - 1.0f) * 0.03f;
+    union { unsigned u; float ff; } v; v.ff = f;
+    unsigned r = v.u + 0x7fff + ((v.u >> 16) & 1);
+    x[i] = (unsigned short)(r >> 16);
+  }
+}
+
+// Swizzles (self-inverse). MODE 1: st_16x32 (XOR byte-bit-5 with bit-9).
+// MODE 2: row-XOR — within a 128 B row, XOR the 16 B-slot index with
+// (row & 7): lanes reading 16 consecutive rows at one column spread over
+// 8 bank groups instead of all hitting one (fixes the af/bf frag-read
+// conflict for [row][128 B] tiles).
+template <int MODE>
+__device__ __forceinline__ int swz(int b) {
+  if (MODE == 1) return b ^ (((b >> 9) & 1) << 5);
+  if (MODE == 2) return (b & ~127) | ((b & 127) ^ (((b >> 7) & 7) << 4));
+  return b;
+}
+
+template <int V>
+__global__ __launch_bounds__(NT, 1) void k_knn8p(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  constexpr int SWZ = (V == 0) ? 0 : (V == 1 ? 1 : 2);
+  // top-k indices live in LDS above the staging buffers (128K..148K):
+  // keeping ti[KC] in VGPRs alongside acc[8][4] + staging state spills
+  // ~40 regs to scratch (PMC: WAIT/BUSY 12.3). Scores stay in registers
+  // (the hot guard is tv[KC-1]); index writes only happen on insert.
+  constexpr bool DO_STAGE = (V != 10 && V != 6);  // V10: stale LDS; V6: no LDS
+  constexpr bool DO_MFMA = (V != 11);    // V11: stage only
+  constexpr bool A_DIRECT = (V == 4 || V == 6);   // A frags from global/L2
+  constexpr bool B_DIRECT = (V == 6);    // B frags from global/L2
+  extern __shared__ __align__(16) char smem[];
+  // buf b at smem + b*64K: A tile [256][64] bf16 (32K) then B tile (32K)
+  float* sE = (float*)smem;  // epilogue alias: [64][ESTR] fp32 (66.6K)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int wm = wid >> 2;       // wave row 0..1 (128 rows each)
+  const int wn = wid & 3;        // wave col 0..3 (64 cols each)
+
+  // this thread's fixed query column + row-half for the epilogue
+  const int my_col = tid & 255;
+  const int my_sub = tid >> 8;   // 0 or 1 (rows 0..127 / 128..255)
+  const int my_round = my_col >> 6;
+
+  float tv[KC];
+  int* ti = (int*)(smem + 131072) + tid * KC;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+  const int ntiles = d / BK;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM;
+
+    float4v acc[8][4];
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    // ---- staging: per-lane global byte pointers precomputed ONCE per
+    // panel (A) / kernel (B); per tile the only math is +t*128. Recomputing
+    // addresses (swizzle, row/col split, 64-bit muls) inside the K-loop
+    // costs ~60 VGPRs of temporaries on top of acc's 128 and spilled to
+    // scratch (44-71 spills, WAIT/BUSY 12.3 — measured via PMC).
+    const G_AS char* ap[4];
+    // B shares A's per-lane (row, col) pattern: bp = ap + uniform delta
+    const long long ab_delta =
+        (const char*)qs - ((const char*)db + prow * d2);
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int chunk = wid * 4 + it;
+      int x = chunk * 1024 + lane * 16;
+      int p = swz<SWZ>(x);
+      int r = p >> 7, cb = p & 127;
+      ap[it] = (const G_AS char*)db + (prow + r) * d2 + cb;
+    }
+
+    auto stage = [&](int b, int kt) {
+      char* base = smem + b * 65536;
+      const int ko = kt * (BK * 2);
+#pragma unroll
+      for (int it = 0; it < 4 && !A_DIRECT; ++it) {
+        int chunk = wid * 4 + it;
+        L_AS unsigned int* lp = (L_AS unsigned int*)(base + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(
+            (const G_AS unsigned int*)(ap[it] + ko), lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int chunk = wid * 4 + it;
+        L_AS unsigned int* lp =
+            (L_AS unsigned int*)(base + 32768 + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(
+            (const G_AS unsigned int*)(ap[it] + ab_delta + ko), lp, 16, 0, 0);
+      }
+    };
+
+    if (DO_STAGE) stage(0, 0);
+
+    // One K-tile: stage next tile into buffer NB (compile-time constant so
+    // the waitcnt pass can see the ds_reads of buffer CB don't alias the
+    // in-flight LDS-DMA writes to NB — a runtime buffer index makes LLVM
+    // emit vmcnt(0) before every barrier and serializes the pipeline).
+    auto tile = [&](int t, auto cb_c, auto nb_c) {
+      constexpr int CB = decltype(cb_c)::value;
+      constexpr int NB = decltype(nb_c)::value;
+      if (DO_STAGE && t + 1 < ntiles) stage(NB, t + 1);
+      if (DO_STAGE) {
+        if (t + 1 < ntiles) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+      const unsigned short* sA = (const unsigned short*)(smem + CB * 65536);
+      const unsigned short* sB =
+          (const unsigned short*)(smem + CB * 65536 + 32768);
+#pragma unroll
+      for (int ks = 0; ks < 2 && DO_MFMA; ++ks) {
+        bf16x8 bfr[4];
+        const int kb = (ks * 32 + (lane >> 4) * 8) * 2;  // byte offset in row
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wn * 64 + nn * 16 + (lane & 15);
+          if (B_DIRECT)
+            bfr[nn] = (bf16x8)(*reinterpret_cast<const G_AS short8v*>(
+                (const G_AS char*)qs + (long long)c * d2 + t * (BK * 2) + kb));
+          else
+            bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                (const char*)sB + swz<SWZ>(c * 128 + kb)));
+        }
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          bf16x8 af[4];
+#pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            int r = wm * 128 + h * 64 + m * 16 + (lane & 15);
+            if (A_DIRECT)
+              af[m] = (bf16x8)(*reinterpret_cast<const G_AS short8v*>(
+                  (const G_AS char*)db + (prow + r) * d2 + t * (BK * 2) + kb));
+            else
+              af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                  (const char*)sA + swz<SWZ>(r * 128 + kb)));
+          }
+          if (V >= 3) {
+            __syncthreads();
+            __builtin_amdgcn_s_setprio(1);
+          }
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+#pragma unroll
+            for (int nn = 0; nn < 4; ++nn)
+              acc[h * 4 + m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[m], bfr[nn], acc[h * 4 + m][nn], 0, 0, 0);
+          if (V >= 3) {
+            __builtin_amdgcn_s_setprio(0);
+          }
+        }
+      }
+      if (!DO_MFMA) {
+        float x = *(const float*)((const char*)sA + (lane * 16));
+        acc[0][0][0] += x;
+      }
+      if (DO_STAGE) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+    };
+
+    for (int t = 0; t < ntiles; t += 2) {
+      tile(t, std::integral_constant<int, 0>{},
+           std::integral_constant<int, 1>{});
+      tile(t + 1, std::integral_constant<int, 1>{},
+           std::integral_constant<int, 0>{});
+    }
+
+    // ---- epilogue: 4 rounds over wave_n groups ----
+    if (!DO_STAGE) __syncthreads();  // V6/V10: no K-loop barriers
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (wn == r) {
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = nn * 16 + (lane & 15);            // 0..63 in round
+            int row = wm * 128 + m * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sE + col * ESTR + row) = acc[m][nn];
+          }
+      }
+      __syncthreads();
+      if (my_round == r) {
+        const float* src = sE + (my_col & 63) * ESTR + my_sub * 128;
+        const long long grow0 = prow + my_sub * 128;
+#pragma unroll 4
+        for (int j = 0; j < 128; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(src + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            float s = v[e];
+            if (s > tv[KC - 1]) {
+              float cs = s;
+              int ci = (int)(grow0 + j + e);
+#pragma unroll
+              for (int i = 0; i < KC; ++i) {
+                bool ins = cs > tv[i];
+                float ts = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i];
+                if (ins) ti[i] = ci;
+                cs = ins ? ts : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- candidates: slot = blockIdx*2 + my_sub, query = my_col ----
+  long long slot = (long long)blockIdx.x * 2 + my_sub;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[(slot * BN + my_col) * KC + i] = tv[i];
+    cand_idx[(slot * BN + my_col) * KC + i] = ti[i];
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// BM=128 variant: 8 waves (2x4), wave-tile 64x64, acc[4][4] (64 VGPRs) —
+// fits the register budget with zero spills where the 256-row tile could
+// not (acc[8][4]=128 + staging + top-k state -> 29-45 spills).
+// LDS: dbuf 2 x (A 16K + B 32K) = 96K + ti 20K = 116K.
+// ---------------------------------------------------------------------------
+#define BM1 128
+// V: 0 = normal; 1 = NO barriers/waits (WRONG results; overlap ceiling
+// diagnostic); 2 = vmcnt(0) at tile boundary; 3 = normal without setprio;
+// 5 = B fragments direct from L2 (stage only A: 16 KB DMA per tile);
+// 6 = triple-buffer, prefetch distance 2 (vmcnt(12))
+template <int V>
+__global__ __launch_bounds__(NT, 1) void k_knn128(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  constexpr int SWZ = 2;
+  extern __shared__ __align__(16) char smem[];
+  float* sE = (float*)smem;  // epilogue alias [64][EST1]
+  constexpr int EST1 = 132;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int wm = wid >> 2;       // 0..1 (64 rows each)
+  const int wn = wid & 3;        // 0..3 (64 cols each)
+
+  const int my_col = tid & 255;
+  const int my_sub = tid >> 8;   // rows 0..63 / 64..127
+  const int my_round = my_col >> 6;
+
+  float tv[KC];
+  int ti[KC];  // registers: an LDS-resident ti turns the insert chain into
+               // ~1200 scalar ds ops per panel (measured: 5x LDS instrs)
+#pragma unroll
+  for (int i = 0; i < KC; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+  const int ntiles = d / BK;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BM1;
+
+    float4v acc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    // per-lane staging pointers: A 16 chunks (2/wave), B 32 chunks (4/wave)
+    const G_AS char* ap[2];
+    const G_AS char* bp0;
+    {
+      const long long qbase = (long long)((const char*)qs - (const char*)db);
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int x = (wid * 2 + it) * 1024 + lane * 16;
+        int p = swz<SWZ>(x);
+        ap[it] = (const G_AS char*)db + (prow + (p >> 7)) * d2 + (p & 127);
+      }
+      int x = (wid * 4) * 1024 + lane * 16;  // B chunk base for this wave
+      int p = swz<SWZ>(x);
+      bp0 = (const G_AS char*)db + qbase + (long long)(p >> 7) * d2 + (p & 127);
+    }
+
+    auto stage = [&](int b, int kt) {
+      char* base = smem + b * (V == 5 ? 16384 : 49152);
+      const int ko = kt * (BK * 2);
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        L_AS unsigned int* lp =
+            (L_AS unsigned int*)(base + (wid * 2 + it) * 1024);
+        __builtin_amdgcn_global_load_lds(
+            (const G_AS unsigned int*)(ap[it] + ko), lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 4 && V != 5; ++it) {
+        // B chunks: consecutive rows 8 apart -> +8*d2 per chunk
+        L_AS unsigned int* lp =
+            (L_AS unsigned int*)(base + 16384 + (wid * 4 + it) * 1024);
+        __builtin_amdgcn_global_load_lds(
+            (const G_AS unsigned int*)(bp0 + (long long)it * 8 * d2 + ko),
+            lp, 16, 0, 0);
+      }
+    };
+
+    auto tile = [&](int t, auto cb_c) {
+      constexpr int CB = decltype(cb_c)::value;
+      constexpr int BUFSZ = (V == 5) ? 16384 : 49152;
+      constexpr int PDIST = (V == 6) ? 2 : 1;   // prefetch distance
+      if (t + PDIST < ntiles) stage((CB + PDIST) % (PDIST + 1), t + PDIST);
+      if (V != 1) {
+        constexpr int INFLT = (V == 5) ? 2 : 6;  // issues per stage
+        if (t + 1 < ntiles && V != 2)
+          asm volatile("s_waitcnt vmcnt(%0)" :: "i"(INFLT * PDIST) : "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+      const unsigned short* sA = (const unsigned short*)(smem + CB * BUFSZ);
+      const unsigned short* sB =
+          (const unsigned short*)(smem + CB * BUFSZ + 16384);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wn * 64 + nn * 16 + (lane & 15);
+          if (V == 5)
+            bfr[nn] = (bf16x8)(*reinterpret_cast<const G_AS short8v*>(
+                (const G_AS char*)qs + (long long)c * d2 + t * (BK * 2) + kb));
+          else
+            bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                (const char*)sB + swz<SWZ>(c * 128 + kb)));
+        }
+        bf16x8 af[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = wm * 64 + m * 16 + (lane & 15);
+          af[m] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sA + swz<SWZ>(r * 128 + kb)));
+        }
+        if (V != 3) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bfr[nn], acc[m][nn], 0, 0, 0);
+        if (V != 3) __builtin_amdgcn_s_setprio(0);
+      }
+      if (V != 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+    };
+
+    if (V == 6) {
+      stage(0, 0);
+      stage(1, 1);
+      // ntiles = 16: 15 tiles in fives of 3, then the last separately
+      for (int t = 0; t < ntiles - 1; t += 3) {
+        tile(t, std::integral_constant<int, 0>{});
+        tile(t + 1, std::integral_constant<int, 1>{});
+        tile(t + 2, std::integral_constant<int, 2>{});
+      }
+      tile(ntiles - 1, std::integral_constant<int, 0>{});
+    } else {
+      stage(0, 0);
+      for (int t = 0; t < ntiles; t += 2) {
+        tile(t, std::integral_constant<int, 0>{});
+        tile(t + 1, std::integral_constant<int, 1>{});
+      }
+    }
+    if (V == 1) __syncthreads();
+
+    // ---- epilogue: 4 rounds over wave_n groups ----
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (wn == r) {
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = nn * 16 + (lane & 15);
+            int row = wm * 64 + m * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sE + col * EST1 + row) = acc[m][nn];
+          }
+      }
+      __syncthreads();
+      if (my_round == r) {
+        const float* src = sE + (my_col & 63) * EST1 + my_sub * 64;
+        const long long grow0 = prow + my_sub * 64;
+#pragma unroll 4
+        for (int j = 0; j < 64; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(src + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            float s = v[e];
+            if (s > tv[KC - 1]) {
+              float cs = s;
+              int ci = (int)(grow0 + j + e);
+#pragma unroll
+              for (int i = 0; i < KC; ++i) {
+                bool ins = cs > tv[i];
+                float ts = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i];
+                ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * 2 + my_sub;
+#pragma unroll
+  for (int i = 0; i < KC; ++i) {
+    cand_score[(slot * BN + my_col) * KC + i] = tv[i];
+    cand_idx[(slot * BN + my_col) * KC + i] = ti[i];
+  }
+}
+
+template <int V>
+static float run_v(const unsigned short* db, const unsigned short* qs,
+                   long long n, int d, float* cs, int* ci, int iters,
+                   int maxgrid) {
+  long long panels = n / BM;
+  int grid = (int)std::min<long long>(panels, maxgrid);
+  hipFuncSetAttribute((const void*)&k_knn8p<V>,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, 131072 + NT * KC * 4);
+  hipLaunchKernelGGL((k_knn8p<V>), dim3(grid), dim3(NT), 131072 + NT * KC * 4, 0,
+                     db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess || hipGetLastError() != hipSuccess) {
+    printf("V%d launch err: %s\n", V, hipGetErrorString(err));
+    return -1;
+  }
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn8p<V>), dim3(grid), dim3(NT), 131072 + NT * KC * 4, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+template <int V>
+static bool refcheck(const unsigned short* db, const unsigned short* qs,
+                     int d, float* cs, int* ci) {
+  const long long nn = 4096;
+  long long panels = nn / BM;
+  int grid = (int)panels;
+  long long slots = (long long)grid * 2;
+  hipFuncSetAttribute((const void*)&k_knn8p<V>,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, 131072 + NT * KC * 4);
+  hipLaunchKernelGGL((k_knn8p<V>), dim3(grid), dim3(NT), 131072 + NT * KC * 4, 0,
+                     db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess) {
+    printf("V%d refcheck launch err: %s\n", V, hipGetErrorString(err));
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best = -1e30f; long long bi = -1;
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      if (a > best) { best = a; bi = r; }
+    }
+    float gbest = -1e30f; int gi = -1;
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        if (v > gbest) { gbest = v; gi = hci[(s * BN + qi) * KC + k]; }
+      }
+    if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+      if (bad < 3) printf("V%d MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                          V, qi, gi, gbest, bi, best);
+      bad++;
+    }
+  }
+  printf("V%d refcheck: %s (%d bad)\n", V, bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
+
+template <int V>
+static float run_128(const unsigned short* db, const unsigned short* qs,
+                     long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / BM1;
+  int grid = (int)std::min<long long>(panels, 4096);
+  // V6: 3 buffers; V5: A-only buffers but epilogue sE needs 33.8 KB
+  int smembytes = (V == 6) ? 147456 : (V == 5 ? 36864 : 98304);
+  hipFuncSetAttribute((const void*)&k_knn128<V>,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, smembytes);
+  hipLaunchKernelGGL((k_knn128<V>), dim3(grid), dim3(NT),
+                     smembytes, 0, db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess || hipGetLastError() != hipSuccess) {
+    printf("k128 launch err: %s\n", hipGetErrorString(err));
+    return -1;
+  }
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn128<V>), dim3(grid), dim3(NT),
+                       smembytes, 0, db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+static bool refcheck_128(const unsigned short* db, const unsigned short* qs,
+                         int d, float* cs, int* ci) {
+  const long long nn = 4096;
+  long long panels = nn / BM1;
+  int grid = (int)panels;
+  long long slots = (long long)grid * 2;
+  hipFuncSetAttribute((const void*)&k_knn128<0>,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, 98304);
+  hipLaunchKernelGGL((k_knn128<0>), dim3(grid), dim3(NT),
+                     98304, 0, db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess) {
+    printf("k128 refcheck launch err: %s\n", hipGetErrorString(err));
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best = -1e30f; long long bi = -1;
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      if (a > best) { best = a; bi = r; }
+    }
+    float gbest = -1e30f; int gi = -1;
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        if (v > gbest) { gbest = v; gi = hci[(s * BN + qi) * KC + k]; }
+      }
+    if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+      if (bad < 3) printf("k128 MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                          qi, gi, gbest, bi, best);
+      bad++;
+    }
+  }
+  printf("k128 refcheck: %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
+
+// ---------------------------------------------------------------------------
+// Production-envelope variant: BM x 256, 4 waves, 3 workgroups/CU (the
+// measured DMA-throughput sweet spot — LDS-DMA ingest scales with resident
+// waves: 8-wave/1-WG kernels cap at ~3 TB/s, 12 waves across 3 WGs reach
+// ~8.8 TB/s). Template BMT in {64, 96}; SWZY enables the row-XOR LDS
+// swizzle (kills the 16-way af/bf bank conflict).
+// ---------------------------------------------------------------------------
+template <int BMT, int SWZY, int KCT = KC, int EPI = 1>
+__global__ __launch_bounds__(256, 3) void k_knn96(
+    const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
+    long long n_panels, int d, float* __restrict__ cand_score,
+    int* __restrict__ cand_idx) {
+  constexpr int MW = BMT / 16;        // m-fragments per wave (4 or 6)
+  constexpr int ACH = BMT * BK * 2;   // A tile bytes
+  __shared__ __align__(16) char smem[ACH + BN * BK * 2];
+  unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + ACH);
+  float* sS = (float*)smem;           // epilogue alias [32][ESTR? use 260]
+  constexpr int SST = 260;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;          // wave col 0..3
+
+  float tv[KCT];
+  int ti[KCT];
+#pragma unroll
+  for (int i = 0; i < KCT; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long d2 = (long long)d * 2;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * BMT;
+
+    float4v acc[MW][4];
+#pragma unroll
+    for (int m = 0; m < MW; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < d; kt += BK) {
+#pragma unroll
+      for (int it = 0; it < BMT / 32; ++it) {  // A chunks: BMT/8 over 4 waves
+        int chunk = wc * (BMT / 32) + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz<SWZY>(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + (p >> 7)) * d2 + (long long)kt * 2 +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {  // B: 32 chunks over 4 waves
+        int chunk = wc * 8 + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz<SWZY>(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)(p >> 7) * d2 + (long long)kt * 2 +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+      __syncthreads();
+
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sB + swz<SWZY>(c * 128 + kb)));
+        }
+#pragma unroll
+        for (int m = 0; m < MW; ++m) {
+          int r = m * 16 + (lane & 15);
+          bf16x8 af = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sA + swz<SWZY>(r * 128 + kb)));
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: chunks of 32 rows through LDS ----
+    // EPI==2: transposed [col][36] chunk — float4 stores (4 consecutive
+    // rows per MFMA fragment register quad) and float4 scans; 4x fewer
+    // LDS ops than the row-major b32 layout (epilogue measured at 36%
+    // of kernel time at 552 TF).
+    float* sT = (float*)smem;  // [256][36]
+#pragma unroll
+    for (int h = 0; h < BMT / 32; ++h) {
+      if (EPI == 2) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int m = h * 2 + mi;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int col = wc * 64 + nn * 16 + (lane & 15);
+            int srow = mi * 16 + (lane >> 4) * 4;
+            *reinterpret_cast<float4v*>(sT + col * 36 + srow) = acc[m][nn];
+          }
+        }
+      } else {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          int m = h * 2 + mi;
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn) {
+            int srow = mi * 16 + (lane >> 4) * 4;
+            int col = wc * 64 + nn * 16 + (lane & 15);
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              sS[(srow + r) * SST + col] = acc[m][nn][r];
+          }
+        }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * 32;
+      if (EPI == 2) {
+#pragma unroll
+        for (int j = 0; j < 32; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            float s = v[e];
+            if (s > tv[KCT - 1]) {
+              float cs = s;
+              int ci = (int)(grow0 + j + e);
+#pragma unroll
+              for (int i = 0; i < KCT; ++i) {
+                bool ins = cs > tv[i];
+                float ts = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i];
+                ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts : cs; ci = ins ? tj : ci;
+              }
+            }
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 32 && EPI; ++r) {
+          float s = sS[r * SST + tid];
+          if (s > tv[KCT - 1]) {
+            float cs = s;
+            int ci = (int)(grow0 + r);
+#pragma unroll
+            for (int i = 0; i < KCT; ++i) {
+              bool ins = cs > tv[i];
+              float ts = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i];
+              ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * BN + tid;
+#pragma unroll
+  for (int i = 0; i < KCT; ++i) {
+    cand_score[slot * KC + i] = tv[i];
+    cand_idx[slot * KC + i] = ti[i];
+  }
+}
+
+template <int BMT, int SWZY>
+static float run_96(const unsigned short* db, const unsigned short* qs,
+                    long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / BMT;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<BMT, SWZY>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess || hipGetLastError() != hipSuccess) {
+    printf("k96 launch err: %s\n", hipGetErrorString(err));
+    return -1;
+  }
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<BMT, SWZY>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+template <int BMT, int SWZY>
+static bool refcheck_96(const unsigned short* db, const unsigned short* qs,
+                        int d, float* cs, int* ci) {
+  const long long nn = 192 * 22;  // divisible by 64 and 96
+  long long panels = nn / BMT;
+  if (nn % BMT) return false;
+  int grid = (int)panels;
+  long long slots = (long long)grid;
+  hipLaunchKernelGGL((k_knn96<BMT, SWZY>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipError_t err = hipDeviceSynchronize();
+  if (err != hipSuccess) {
+    printf("k96 refcheck launch err: %s\n", hipGetErrorString(err));
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best = -1e30f; long long bi = -1;
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      if (a > best) { best = a; bi = r; }
+    }
+    float gbest = -1e30f; int gi = -1;
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        if (v > gbest) { gbest = v; gi = hci[(s * BN + qi) * KC + k]; }
+      }
+    if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+      if (bad < 3) printf("k96<%d,%d> MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                          BMT, SWZY, qi, gi, gbest, bi, best);
+      bad++;
+    }
+  }
+  printf("k96<%d,%d> refcheck: %s (%d bad)\n", BMT, SWZY,
+         bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
+
+static float run_96k12(const unsigned short* db, const unsigned short* qs,
+                       long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, 12>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, 12>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+static float run_96t(const unsigned short* db, const unsigned short* qs,
+                     long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 2>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 2>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+static bool refcheck_96t(const unsigned short* db, const unsigned short* qs,
+                         int d, float* cs, int* ci) {
+  const long long nn = 192 * 22;
+  long long panels = nn / 96;
+  int grid = (int)panels;
+  long long slots = (long long)grid;
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 2>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  if (hipDeviceSynchronize() != hipSuccess) {
+    printf("k96t refcheck launch err\n");
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best = -1e30f; long long bi = -1;
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      if (a > best) { best = a; bi = r; }
+    }
+    float gbest = -1e30f; int gi = -1;
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        if (v > gbest) { gbest = v; gi = hci[(s * BN + qi) * KC + k]; }
+      }
+    if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+      if (bad < 3) printf("k96t MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                          qi, gi, gbest, bi, best);
+      bad++;
+    }
+  }
+  printf("k96t refcheck: %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
+static float run_96ne(const unsigned short* db, const unsigned short* qs,
+                      long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 0>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 0>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+int main() {
+  const long long n = 4 << 20;
+  const int d = 1024;
+  unsigned short *db, *qs;
+  float* cs;
+  int* ci;
+  hipMalloc(&db, n * d * 2);
+  hipMalloc(&qs, (long long)BN * d * 2);
+  hipMalloc(&cs, 8192LL * BN * KC * 4);   // up to 4096 blocks x 2 subs
+  hipMalloc(&ci, 8192LL * BN * KC * 4);
+  hipLaunchKernelGGL(fill_rand, dim3(4096), dim3(256), 0, 0, db, n * d);
+  hipLaunchKernelGGL(fill_rand, dim3(64), dim3(256), 0, 0, qs, (long long)BN * d);
+  hipDeviceSynchronize();
+  double flops = 2.0 * n * d * BN;
+  double bytes = (double)n * d * 2;
+  float ms;
+
+  if (refcheck<0>(db, qs, d, cs, ci)) {
+    ms = run_v<0>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V0 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "256sq-dbuf",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  if (refcheck<1>(db, qs, d, cs, ci)) {
+    ms = run_v<1>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V1 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "256sq-dbuf+swz",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  if (refcheck<2>(db, qs, d, cs, ci)) {
+    ms = run_v<2>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V2 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "256sq-dbuf+rowswz",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  ms = run_v<10>(db, qs, n, d, cs, ci, 10, 4096);
+  printf("V10 %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (STALE: compute+epi only)\n",
+         "no-stage", ms, flops / ms / 1e9, bytes / ms / 1e9);
+  ms = run_v<11>(db, qs, n, d, cs, ci, 10, 4096);
+  printf("V11 %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (stage only)\n",
+         "no-mfma", ms, flops / ms / 1e9, bytes / ms / 1e9);
+  {
+    long long n96 = (n / 96) * 96;
+    double fl96 = 2.0 * n96 * d * BN, by96 = (double)n96 * d * 2;
+    if (refcheck_96<64, 0>(db, qs, d, cs, ci)) {
+      ms = run_96<64, 0>(db, qs, n96, d, cs, ci, 10);
+      printf("K64  %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (production repro)\n",
+             "64x256-3wg", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+    }
+    if (refcheck_96<64, 2>(db, qs, d, cs, ci)) {
+      ms = run_96<64, 2>(db, qs, n96, d, cs, ci, 10);
+      printf("K64s %-17s %7.3f ms  %6.0f TF  %5.2f TB/s\n",
+             "64x256-3wg+swz", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+    }
+    if (refcheck_96<96, 0>(db, qs, d, cs, ci)) {
+      ms = run_96<96, 0>(db, qs, n96, d, cs, ci, 10);
+      printf("K96  %-17s %7.3f ms  %6.0f TF  %5.2f TB/s\n",
+             "96x256-3wg", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+    }
+    if (refcheck_96<96, 2>(db, qs, d, cs, ci)) {
+      ms = run_96<96, 2>(db, qs, n96, d, cs, ci, 10);
+      printf("K96s %-17s %7.3f ms  %6.0f TF  %5.2f TB/s\n",
+             "96x256-3wg+swz", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+    }
+    {
+      if (refcheck_96t(db, qs, d, cs, ci)) {
+        ms = run_96t(db, qs, n96, d, cs, ci, 10);
+        printf("K96t %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (transposed epi)\n",
+               "96+swz+tepi", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+      }
+      ms = run_96ne(db, qs, n96, d, cs, ci, 10);
+      printf("K96ne %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (no epilogue diag)\n",
+             "96+swz-noepi", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+    }
+  }
+  if (refcheck_128(db, qs, d, cs, ci)) {
+    ms = run_128<0>(db, qs, n, d, cs, ci, 10);
+    printf("K128 %-17s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "128x256-dbuf",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+    ms = run_128<1>(db, qs, n, d, cs, ci, 10);
+    printf("K128b %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (WRONG: no barriers)\n",
+           "no-bar-diag", ms, flops / ms / 1e9, bytes / ms / 1e9);
+    ms = run_128<2>(db, qs, n, d, cs, ci, 10);
+    printf("K128c %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (vmcnt0)\n",
+           "vm0-diag", ms, flops / ms / 1e9, bytes / ms / 1e9);
+    ms = run_128<3>(db, qs, n, d, cs, ci, 10);
+    printf("K128d %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (no setprio)\n",
+           "noprio", ms, flops / ms / 1e9, bytes / ms / 1e9);
+    ms = run_128<5>(db, qs, n, d, cs, ci, 10);
+    printf("K128e %-16s %7.3f ms  %6.0f TF  %5.2f TB/s\n",
+           "B-direct-L2", ms, flops / ms / 1e9, bytes / ms / 1e9);
+    ms = run_128<6>(db, qs, n, d, cs, ci, 10);
+    printf("K128f %-16s %7.3f ms  %6.0f TF  %5.2f TB/s\n",
+           "tbuf-pd2", ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  if (refcheck<4>(db, qs, d, cs, ci)) {
+    ms = run_v<4>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V4 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "A-direct,B-lds",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  if (refcheck<6>(db, qs, d, cs, ci)) {
+    ms = run_v<6>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V6 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "all-direct-noLDS",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  if (refcheck<3>(db, qs, d, cs, ci)) {
+    ms = run_v<3>(db, qs, n, d, cs, ci, 10, 4096);
+    printf("V3 %-18s %7.3f ms  %6.0f TF  %5.2f TB/s\n", "rowswz+phases",
+           ms, flops / ms / 1e9, bytes / ms / 1e9);
+  }
+  return 0;
+}
