@@ -667,6 +667,31 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
   int ti[KCT];
 #pragma unroll
   for (int i = 0; i < KCT; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+  // EPI==4: 4-deep shift-register candidate buffer. The full insertion
+  // chain is ~40 VALU; with 64 lanes the wave executes it for nearly
+  // every value (any-lane divergence). Buffering makes the common path
+  // a ~9-op shift and amortizes the chain 4x.
+  float b0s = -1e30f, b1s = -1e30f, b2s = -1e30f, b3s = -1e30f;
+  int b0i = -1, b1i = -1, b2i = -1, b3i = -1, bn = 0;
+  auto flush = [&]() {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      float cs = u == 0 ? b0s : u == 1 ? b1s : u == 2 ? b2s : b3s;
+      int ci = u == 0 ? b0i : u == 1 ? b1i : u == 2 ? b2i : b3i;
+      if (cs > tv[KCT - 1]) {
+#pragma unroll
+        for (int i = 0; i < KCT; ++i) {
+          bool ins = cs > tv[i];
+          float ts = tv[i]; int tj = ti[i];
+          tv[i] = ins ? cs : tv[i];
+          ti[i] = ins ? ci : ti[i];
+          cs = ins ? ts : cs; ci = ins ? tj : ci;
+        }
+      }
+    }
+    b0s = b1s = b2s = b3s = -1e30f;
+    bn = 0;
+  };
 
   const long long d2 = (long long)d * 2;
 
@@ -736,7 +761,7 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
     float* sT = (float*)smem;  // [256][36]
 #pragma unroll
     for (int h = 0; h < BMT / 32; ++h) {
-      if (EPI == 2) {
+      if (EPI >= 2) {  // transposed writes for EPI 2/3/4
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi) {
           int m = h * 2 + mi;
@@ -763,7 +788,26 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
       }
       __syncthreads();
       const long long grow0 = prow + (long long)h * 32;
-      if (EPI == 2) {
+      if (EPI == 3) {
+        // diagnostic: keep one read per thread so writes aren't dead
+        tv[KCT - 1] = fmaxf(tv[KCT - 1] - 1e-30f, sT[tid * 36]);
+      } else if (EPI == 4) {
+#pragma unroll
+        for (int j = 0; j < 32; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            float s = v[e];
+            if (s > tv[KCT - 1]) {
+              b3s = b2s; b3i = b2i; b2s = b1s; b2i = b1i;
+              b1s = b0s; b1i = b0i; b0s = s;
+              b0i = (int)(grow0 + j + e);
+              ++bn;
+            }
+            if (bn >= 4) flush();
+          }
+        }
+      } else if (EPI == 2) {
 #pragma unroll
         for (int j = 0; j < 32; j += 4) {
           float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
@@ -806,6 +850,7 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
     }
   }
 
+  if (EPI == 4 && bn > 0) flush();
   long long slot = (long long)blockIdx.x * BN + tid;
 #pragma unroll
   for (int i = 0; i < KCT; ++i) {
@@ -972,6 +1017,104 @@ static bool refcheck_96t(const unsigned short* db, const unsigned short* qs,
   return bad == 0;
 }
 
+
+static float run_96ns(const unsigned short* db, const unsigned short* qs,
+                      long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 3>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 3>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+
+
+static float run_96b(const unsigned short* db, const unsigned short* qs,
+                     long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 4>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 4>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+static bool refcheck_96b(const unsigned short* db, const unsigned short* qs,
+                         int d, float* cs, int* ci) {
+  const long long nn = 192 * 22;
+  long long panels = nn / 96;
+  int grid = (int)panels;
+  long long slots = (long long)grid;
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 4>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  if (hipDeviceSynchronize() != hipSuccess) {
+    printf("k96b refcheck launch err\n");
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  // check top-3 per query (buffered path must preserve exact top-k order)
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best[3] = {-1e30f, -1e30f, -1e30f};
+    long long bidx[3] = {-1, -1, -1};
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      for (int t = 0; t < 3; ++t)
+        if (a > best[t]) {
+          for (int u = 2; u > t; --u) { best[u] = best[u-1]; bidx[u] = bidx[u-1]; }
+          best[t] = a; bidx[t] = r; break;
+        }
+    }
+    // merged candidates -> global top-3
+    float gb[3] = {-1e30f, -1e30f, -1e30f};
+    int gi[3] = {-1, -1, -1};
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        int ii = hci[(s * BN + qi) * KC + k];
+        for (int t = 0; t < 3; ++t)
+          if (v > gb[t]) {
+            for (int u = 2; u > t; --u) { gb[u] = gb[u-1]; gi[u] = gi[u-1]; }
+            gb[t] = v; gi[t] = ii; break;
+          }
+      }
+    for (int t = 0; t < 3; ++t)
+      if (gi[t] != bidx[t]) {
+        if (bad < 3) printf("k96b MISMATCH q=%d rank%d: got %d want %lld\n",
+                            qi, t, gi[t], bidx[t]);
+        bad++;
+      }
+  }
+  printf("k96b refcheck(top3): %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
 static float run_96ne(const unsigned short* db, const unsigned short* qs,
                       long long n, int d, float* cs, int* ci, int iters) {
   long long panels = n / 96;
@@ -1059,6 +1202,14 @@ int main() {
       ms = run_96ne(db, qs, n96, d, cs, ci, 10);
       printf("K96ne %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (no epilogue diag)\n",
              "96+swz-noepi", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+      ms = run_96ns(db, qs, n96, d, cs, ci, 10);
+      printf("K96ns %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (writes+barriers, no scan)\n",
+             "96+swz-noscan", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+      if (refcheck_96b(db, qs, d, cs, ci)) {
+        ms = run_96b(db, qs, n96, d, cs, ci, 10);
+        printf("K96b %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (buffered insert)\n",
+               "96+swz+tepi+buf", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+      }
     }
   }
   if (refcheck_128(db, qs, d, cs, ci)) {
