@@ -212,6 +212,24 @@ class CallClause:
 
 
 @dataclass
+class SchemaCommand:
+    """Schema DDL: CREATE/DROP INDEX|CONSTRAINT, SHOW INDEXES|CONSTRAINTS|
+    DATABASES|PROCEDURES|FUNCTIONS (Neo4j 4/5 syntax + 3.x legacy)."""
+    op: str                      # "create" | "drop" | "show"
+    kind: str                    # "index" | "vector" | "fulltext" | "constraint"
+                                 # | "indexes" | "constraints" | "databases"
+                                 # | "procedures" | "functions"
+    name: Optional[str] = None
+    label: Optional[str] = None
+    props: List[str] = field(default_factory=list)
+    constraint_kind: Optional[str] = None   # "unique" | "exists" | "node_key"
+    options: Optional[Any] = None           # MapLit AST for OPTIONS {...}
+    if_not_exists: bool = False
+    if_exists: bool = False
+    or_replace: bool = False
+
+
+@dataclass
 class SubqueryCallClause:
     """CALL { ... } [IN TRANSACTIONS [OF n ROWS]] clause subquery."""
     query: "Query"

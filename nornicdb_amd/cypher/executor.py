@@ -72,11 +72,14 @@ class Executor:
     """StorageExecutor equivalent (reference pkg/cypher/executor.go:187)."""
 
     def __init__(self, engine: Engine, procedures: Dict[str, Any] = None,
-                 query_cache=None):
+                 query_cache=None, schema=None):
         self.engine = engine
         self.procedures = dict(procedures or {})
         self._plan_cache: Dict[str, A.Query] = {}
         self.stats: Dict[str, int] = {}
+        self.schema = schema            # storage.SchemaManager (DDL target)
+        self.database_lister = None     # set by DatabaseManager for SHOW DATABASES
+        self.current_database = "neo4j"
 
     # ------------------------------------------------------------------ API
     def execute(self, cypher: str, params: Dict[str, Any] = None,
@@ -96,6 +99,10 @@ class Executor:
                       "properties_set": 0, "labels_added": 0}
         if q.explain:
             return Result(["plan"], [[self._explain(q)]], dict(self.stats))
+        if len(q.clauses) == 1 and isinstance(q.clauses[0], A.SchemaCommand):
+            res = self._exec_schema(q.clauses[0], params)
+            res.stats = dict(self.stats)
+            return res
         fast = self._try_fast_count(q, params)
         if fast is not None:
             fast.stats = dict(self.stats)
@@ -270,6 +277,124 @@ class Executor:
                 raise CypherRuntimeError(f"unsupported clause {type(c).__name__}")
             i += 1
         return out_cols, out_rows
+
+    def _exec_schema(self, c: A.SchemaCommand, params) -> Result:
+        """Schema DDL (Neo4j 4/5 + 3.x legacy syntax). Requires a
+        SchemaManager (wired by the NornicDB facade); raw Executor use
+        without one raises."""
+        from ..storage.schema import SchemaManager
+
+        if c.op == "show":
+            if c.kind == "databases":
+                names = (self.database_lister() if self.database_lister
+                         else [self.current_database])
+                return Result(
+                    ["name", "type", "access", "role", "currentStatus",
+                     "default", "home"],
+                    [[n, "standard", "read-write", "primary", "online",
+                      n == "neo4j", n == "neo4j"] for n in names])
+            if c.kind == "procedures":
+                return Result(["name", "description", "mode"],
+                              [[n, (getattr(f, "__doc__", "") or "").strip()
+                                .split("\n")[0], "DEFAULT"]
+                               for n, f in sorted(self.procedures.items())])
+            if c.kind == "functions":
+                from .functions import FUNCTIONS
+                return Result(["name", "category", "description"],
+                              [[n, "builtin", ""] for n in sorted(FUNCTIONS)])
+            if self.schema is None:
+                return Result(["name"], [])
+            if c.kind == "indexes":
+                rows = []
+                for i, (name, kind, label, props) in enumerate(
+                        self.schema.list_indexes()):
+                    rows.append([i + 1, name, "ONLINE", 100.0, kind.upper(),
+                                 "NODE", [label], props])
+                return Result(["id", "name", "state", "populationPercent",
+                               "type", "entityType", "labelsOrTypes",
+                               "properties"], rows)
+            if c.kind == "constraints":
+                kindname = {"unique": "UNIQUENESS",
+                            "exists": "NODE_PROPERTY_EXISTENCE"}
+                return Result(
+                    ["id", "name", "type", "entityType", "labelsOrTypes",
+                     "properties"],
+                    [[i + 1, cc.name, kindname.get(cc.kind, cc.kind.upper()),
+                      "NODE", [cc.label], [cc.prop]]
+                     for i, cc in enumerate(self.schema.list_constraints())])
+            raise CypherRuntimeError(f"cannot SHOW {c.kind}")
+
+        if self.schema is None:
+            raise CypherRuntimeError(
+                "schema commands need a SchemaManager (open via NornicDB)")
+
+        if c.op == "drop":
+            if c.kind == "index":
+                if c.name is None:  # legacy DROP INDEX ON :L(p)
+                    name = f"index_{c.label}_{'_'.join(c.props)}"
+                else:
+                    name = c.name
+                ok = self.schema.drop_index(name)
+                if not ok and not c.if_exists:
+                    raise CypherRuntimeError(f"no such index {name!r}")
+            else:
+                if c.name not in {x.name for x in
+                                  self.schema.list_constraints()} \
+                        and not c.if_exists:
+                    raise CypherRuntimeError(f"no such constraint {c.name!r}")
+                self.schema.drop_constraint(c.name)
+            return Result([], [])
+
+        # create
+        if c.kind == "constraint":
+            name = c.name or f"constraint_{c.label}_{'_'.join(c.props)}"
+            existing = {x.name for x in self.schema.list_constraints()}
+            if name in existing:
+                if c.if_not_exists:
+                    return Result([], [])
+                if c.or_replace:
+                    self.schema.drop_constraint(name)
+                else:
+                    raise CypherRuntimeError(
+                        f"constraint {name!r} already exists")
+            for p in c.props:
+                if c.constraint_kind in ("unique", "node_key"):
+                    self.schema.create_unique_constraint(name, c.label, p)
+                else:
+                    self.schema.create_exists_constraint(name, c.label, p)
+                if c.constraint_kind == "node_key":
+                    self.schema.create_exists_constraint(
+                        name + "_exists", c.label, p)
+            return Result([], [])
+
+        name = c.name or f"index_{c.label}_{'_'.join(c.props)}"
+        existing = {x[0] for x in self.schema.list_indexes()}
+        if name in existing:
+            if c.if_not_exists:
+                return Result([], [])
+            if not c.or_replace:
+                raise CypherRuntimeError(f"index {name!r} already exists")
+            self.schema.drop_index(name)
+        if c.kind == "vector":
+            dims, sim = 0, "cosine"
+            if c.options is not None:
+                try:
+                    opts = self._eval(c.options, {}, params) or {}
+                    icfg = opts.get("indexConfig", opts)
+                    for k, v in icfg.items():
+                        lk = k.lower()
+                        if "dimension" in lk or lk == "dims":
+                            dims = int(v)
+                        if "similarity" in lk:
+                            sim = str(v)
+                except CypherRuntimeError:
+                    pass
+            self.schema.create_vector_index(name, c.label, c.props[0],
+                                            dims, sim)
+        else:
+            self.schema.create_index(c.label, c.props[0], name=name,
+                                     kind=c.kind, props=c.props)
+        return Result([], [])
 
     def _run_subquery(self, q, bindings, params):
         """Execute a subquery AST with the outer row's bindings visible

@@ -134,6 +134,11 @@ class Parser:
             return self._call()
         if self.at_kw("FOREACH"):
             return self._foreach()
+        if self.at_kw("SHOW"):
+            return self._show()
+        if self.at_kw("DROP") or (self.peek().kind == "IDENT"
+                                  and self.peek().value.upper() == "DROP"):
+            return self._drop_schema()
         return None
 
     # ---- clauses ----
@@ -150,13 +155,217 @@ class Parser:
 
     def _create(self):
         self.eat_kw("CREATE")
-        # CREATE INDEX / CONSTRAINT handled by executor as schema commands
-        if self.at_kw("INDEX") or self.at_kw("CONSTRAINT") or self.at_kw("UNIQUE"):
-            raise CypherSyntaxError("schema commands go through the schema API")
+        or_replace = False
+        if self.at_kw("OR") and self.peek(1).kind in ("IDENT", "KW") \
+                and self.peek(1).value.upper() == "REPLACE":
+            self.next()
+            self._eat_word("REPLACE")
+            or_replace = True
+        kindword = None
+        if self._at_ident("VECTOR") or self._at_ident("FULLTEXT") \
+                or self._at_ident("TEXT") or self._at_ident("RANGE") \
+                or self._at_ident("POINT") or self._at_ident("LOOKUP"):
+            nxt = self.peek(1)
+            if nxt.kind == "KW" and nxt.value == "INDEX":
+                kindword = self.next().value.upper()
+        if self.at_kw("INDEX") or kindword is not None:
+            return self._create_index(kindword or "RANGE", or_replace)
+        if self.at_kw("CONSTRAINT"):
+            return self._create_constraint(or_replace)
+        if or_replace:
+            raise CypherSyntaxError("expected INDEX or CONSTRAINT after "
+                                    "CREATE OR REPLACE")
         pats = [self._pattern_path()]
         while self.try_op(","):
             pats.append(self._pattern_path())
         return A.CreateClause(pats)
+
+    # ---- schema DDL ----
+    # (FOR/IF/REQUIRE are not reserved words in the lexer; match by value)
+    def _at_ident(self, word):
+        t = self.peek()
+        return t.kind in ("IDENT", "KW") and t.value.upper() == word
+
+    def _at_word(self, word):
+        t = self.peek()
+        return t.kind in ("IDENT", "KW") and t.value.upper() == word
+
+    def _eat_word(self, word):
+        t = self.next()
+        if t.value.upper() != word:
+            raise CypherSyntaxError(
+                f"expected {word}, got {t.value!r} at {t.pos}")
+
+    def _eat_ident(self, word):
+        t = self.next()
+        if t.value.upper() != word:
+            raise CypherSyntaxError(f"expected {word}, got {t.value!r} at {t.pos}")
+
+    def _if_not_exists(self):
+        if self._at_word("IF"):
+            self.next()
+            if self._at_word("NOT"):
+                self.next()
+                self._eat_word("EXISTS")
+                return (True, False)
+            self._eat_word("EXISTS")
+            return (False, True)
+        return (False, False)
+
+    def _create_index(self, kindword, or_replace):
+        self.next()  # INDEX
+        # legacy 3.x: CREATE INDEX ON :Label(prop)
+        if self.at_kw("ON") and self.peek(1).kind == "OP" and self.peek(1).value == ":":
+            self.next()
+            self.eat_op(":")
+            label = self.ident()
+            self.eat_op("(")
+            props = [self.ident()]
+            while self.try_op(","):
+                props.append(self.ident())
+            self.eat_op(")")
+            return A.SchemaCommand("create", "index", label=label, props=props)
+        name = None
+        if not (self._at_word("IF") or self._at_word("FOR")):
+            name = self.name_part()
+        ine, _ = self._if_not_exists()
+        self._eat_word("FOR")
+        self.eat_op("(")
+        var = self.ident()
+        self.eat_op(":")
+        label = self.ident()
+        while self.try_op("|"):
+            self.ident()  # multi-label fulltext: first label indexed
+        self.eat_op(")")
+        self._eat_word("ON")
+        if self._at_ident("EACH"):   # fulltext: ON EACH [n.p, ...]
+            self.next()
+            self.eat_op("[")
+            props = [self._on_prop(var)]
+            while self.try_op(","):
+                props.append(self._on_prop(var))
+            self.eat_op("]")
+        else:
+            self.eat_op("(")
+            props = [self._on_prop(var)]
+            while self.try_op(","):
+                props.append(self._on_prop(var))
+            self.eat_op(")")
+        options = None
+        if self._at_ident("OPTIONS"):
+            self.next()
+            options = self._expr()
+        kind = {"VECTOR": "vector", "FULLTEXT": "fulltext"}.get(
+            kindword, "index")
+        return A.SchemaCommand("create", kind, name=name, label=label,
+                               props=props, options=options,
+                               if_not_exists=ine, or_replace=or_replace)
+
+    def _on_prop(self, var):
+        v = self.ident()
+        if v != var:
+            raise CypherSyntaxError(f"unknown variable {v!r} in index prop")
+        self.eat_op(".")
+        return self.name_part()
+
+    def _create_constraint(self, or_replace):
+        self.next()  # CONSTRAINT
+        name = None
+        if not (self._at_word("IF") or self._at_word("FOR")
+                or self._at_word("ON")):
+            name = self.name_part()
+        ine, _ = self._if_not_exists()
+        # FOR (n:Label) REQUIRE ...   (5.x)  |  ON (n:Label) ASSERT ... (3.x/4.x)
+        if self._at_word("FOR") or self._at_word("ON"):
+            self.next()
+        else:
+            raise CypherSyntaxError("expected FOR or ON")
+        self.eat_op("(")
+        var = self.ident()
+        self.eat_op(":")
+        label = self.ident()
+        self.eat_op(")")
+        if self._at_ident("REQUIRE") or self._at_ident("ASSERT"):
+            self.next()
+        else:
+            raise CypherSyntaxError("expected REQUIRE or ASSERT")
+        # exists(n.prop)  (legacy)
+        if (self._at_ident("EXISTS") or self.at_kw("EXISTS")) and \
+                self.peek(1).kind == "OP" and self.peek(1).value == "(":
+            self.next()
+            self.eat_op("(")
+            prop = self._on_prop(var)
+            self.eat_op(")")
+            return A.SchemaCommand("create", "constraint", name=name,
+                                   label=label, props=[prop],
+                                   constraint_kind="exists",
+                                   if_not_exists=ine, or_replace=or_replace)
+        props = [self._on_prop(var)]
+        while self.try_op(","):
+            props.append(self._on_prop(var))
+        self.eat_kw("IS")
+        if self._at_word("UNIQUE"):
+            self.next()
+            ck = "unique"
+        elif self._at_word("NOT"):
+            self.next()
+            self._eat_word("NULL")
+            ck = "exists"
+        elif self._at_ident("NODE"):
+            self.next()
+            self._eat_word("KEY")
+            ck = "node_key"
+        else:
+            raise CypherSyntaxError("expected UNIQUE, NOT NULL or NODE KEY")
+        return A.SchemaCommand("create", "constraint", name=name, label=label,
+                               props=props, constraint_kind=ck,
+                               if_not_exists=ine, or_replace=or_replace)
+
+    def _drop_schema(self):
+        self.next()  # DROP
+        if self.at_kw("INDEX"):
+            self.next()
+            # legacy: DROP INDEX ON :Label(prop)
+            if self.at_kw("ON"):
+                self.next()
+                self.eat_op(":")
+                label = self.ident()
+                self.eat_op("(")
+                props = [self.ident()]
+                while self.try_op(","):
+                    props.append(self.ident())
+                self.eat_op(")")
+                return A.SchemaCommand("drop", "index", label=label,
+                                       props=props)
+            name = self.name_part()
+            _, ie = self._if_not_exists()
+            return A.SchemaCommand("drop", "index", name=name, if_exists=ie)
+        if self.at_kw("CONSTRAINT"):
+            self.next()
+            name = self.name_part()
+            _, ie = self._if_not_exists()
+            return A.SchemaCommand("drop", "constraint", name=name,
+                                   if_exists=ie)
+        raise CypherSyntaxError("expected INDEX or CONSTRAINT after DROP")
+
+    def _show(self):
+        self.eat_kw("SHOW")
+        t = self.next()
+        word = t.value.upper()
+        kinds = {"INDEX": "indexes", "INDEXES": "indexes",
+                 "CONSTRAINT": "constraints", "CONSTRAINTS": "constraints",
+                 "DATABASE": "databases", "DATABASES": "databases",
+                 "PROCEDURE": "procedures", "PROCEDURES": "procedures",
+                 "FUNCTION": "functions", "FUNCTIONS": "functions"}
+        if word not in kinds:
+            raise CypherSyntaxError(f"cannot SHOW {t.value!r}")
+        # optional YIELD ... (accepted, ignored: full rows returned)
+        if self.try_kw("YIELD"):
+            while self.peek().kind in ("IDENT", "KW") and not self.at_kw("RETURN"):
+                self.next()
+                if not self.try_op(","):
+                    break
+        return A.SchemaCommand("show", kinds[word])
 
     def _merge(self):
         self.eat_kw("MERGE")

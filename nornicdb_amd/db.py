@@ -140,9 +140,12 @@ class NornicDB:
         self.search = SearchService(engine, dims=self.dims, device=device,
                                     embedder=self.embedder)
         from .apoc import build_apoc_procedures
+        from .storage import SchemaManager
         procs = build_procedures(self)
         procs.update(build_apoc_procedures(self))
-        self.executor = Executor(engine, procedures=procs)
+        self.schema = SchemaManager(engine)
+        self.executor = Executor(engine, procedures=procs, schema=self.schema)
+        self.executor.current_database = name
         # read-query result cache (reference pkg/cache/query_cache.go),
         # invalidated by any storage write event
         from .utils.cache import QueryCache
@@ -175,6 +178,10 @@ class NornicDB:
         res = self.executor.execute(query, params)
         if key is not None and not any(res.stats.values()):
             self.query_cache.put(key, res)
+        elif not cacheable:
+            # schema DDL (CREATE/DROP INDEX...) mutates no engine state, so
+            # no storage event fires — invalidate cached reads explicitly
+            self.query_cache.invalidate()
         if self.triggers and any(res.stats.values()):
             for name, t in list(self.triggers.items()):
                 if t.get("paused"):
@@ -310,6 +317,7 @@ class DatabaseManager:
         eng = NamespacedEngine(self._base, name)
         db = NornicDB(eng, name=name, embedder=self._embedder,
                       dims=self._dims, device=self._device)
+        db.executor.database_lister = lambda: sorted(self._dbs)
         self._dbs[name] = db
         return db
 

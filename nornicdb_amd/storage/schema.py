@@ -39,6 +39,7 @@ class SchemaManager:
         self.constraints: Dict[str, Constraint] = {}
         self.vector_indexes: Dict[str, VectorIndexMeta] = {}
         self.property_indexes: List[tuple] = []
+        self.named_indexes: Dict[str, tuple] = {}  # name -> (kind, label, props)
         if hasattr(engine, "add_validator"):
             engine.add_validator(self._validate)
 
@@ -80,10 +81,43 @@ class SchemaManager:
             return list(self.constraints.values())
 
     # ---- indexes ----
-    def create_index(self, label: str, prop: str):
-        self.property_indexes.append((label, prop))
-        if hasattr(self.engine, "create_property_index"):
-            self.engine.create_property_index(label, prop)
+    def create_index(self, label: str, prop: str, name: str = None,
+                     kind: str = "range", props: list = None):
+        """Register a property index (optionally named, e.g. from Cypher
+        CREATE INDEX). Multi-property indexes index each prop."""
+        plist = props or [prop]
+        nm = name or f"index_{label}_{'_'.join(plist)}"
+        with self._lock:
+            self.named_indexes[nm] = (kind, label, list(plist))
+        for p in plist:
+            self.property_indexes.append((label, p))
+            if hasattr(self.engine, "create_property_index"):
+                self.engine.create_property_index(label, p)
+        return nm
+
+    def drop_index(self, name: str) -> bool:
+        with self._lock:
+            meta = self.named_indexes.pop(name, None)
+            if meta is None:
+                return self.vector_indexes.pop(name, None) is not None
+        kind, label, plist = meta
+        for p in plist:
+            try:
+                self.property_indexes.remove((label, p))
+            except ValueError:
+                pass
+            if hasattr(self.engine, "drop_property_index"):
+                self.engine.drop_property_index(label, p)
+        return True
+
+    def list_indexes(self):
+        """[(name, kind, label, props)] incl. vector indexes."""
+        with self._lock:
+            out = [(n, k, lb, ps) for n, (k, lb, ps) in
+                   self.named_indexes.items()]
+            out += [(v.name, "VECTOR", v.label, [v.prop])
+                    for v in self.vector_indexes.values()]
+        return out
 
     def create_vector_index(self, name: str, label: str, prop: str,
                             dims: int, similarity: str = "cosine"):

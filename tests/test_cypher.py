@@ -419,3 +419,80 @@ class TestTemporal:
         from nornicdb_amd.server.http import _jsonable
         assert _jsonable(tp.make_date("2026-09-12")) == "2026-09-12"
         assert _jsonable({"d": tp.make_duration("PT5S")}) == {"d": "PT5S"}
+
+
+# ---------------------------------------------------------------- schema DDL
+class TestSchemaDDL:
+    """Cypher schema commands (Neo4j 4/5 + 3.x legacy syntax).
+    Parity: reference pkg/cypher schema command handling + pkg/storage
+    schema.go."""
+
+    def _db(self):
+        from nornicdb_amd.db import NornicDB
+        from nornicdb_amd.storage.memory import MemoryEngine
+        return NornicDB(MemoryEngine(), auto_embed=False)
+
+    def test_create_show_drop_index(self):
+        db = self._db()
+        db.cypher("CREATE INDEX idx1 FOR (n:Person) ON (n.name)")
+        rows = db.cypher("SHOW INDEXES").rows
+        assert any(r[1] == "idx1" for r in rows)
+        db.cypher("DROP INDEX idx1")
+        rows = db.cypher("SHOW INDEXES").rows
+        assert not any(r[1] == "idx1" for r in rows)
+
+    def test_legacy_index_syntax(self):
+        db = self._db()
+        db.cypher("CREATE INDEX ON :Person(name)")
+        assert any("Person" in str(r[6]) for r in db.cypher("SHOW INDEXES").rows)
+        db.cypher("DROP INDEX ON :Person(name)")
+
+    def test_unique_constraint_enforced(self):
+        import pytest
+
+        from nornicdb_amd.storage.types import ConstraintViolation
+        db = self._db()
+        db.cypher("CREATE CONSTRAINT c1 FOR (n:User) REQUIRE n.email IS UNIQUE")
+        db.cypher("CREATE (:User {email: 'a@x.com'})")
+        with pytest.raises(ConstraintViolation):
+            db.cypher("CREATE (:User {email: 'a@x.com'})")
+        rows = db.cypher("SHOW CONSTRAINTS").rows
+        assert rows and rows[0][2] == "UNIQUENESS"
+        db.cypher("DROP CONSTRAINT c1")
+        db.cypher("CREATE (:User {email: 'a@x.com'})")  # now allowed
+
+    def test_not_null_constraint(self):
+        import pytest
+
+        from nornicdb_amd.storage.types import ConstraintViolation
+        db = self._db()
+        db.cypher("CREATE CONSTRAINT nn IF NOT EXISTS FOR (n:Doc) "
+                  "REQUIRE n.title IS NOT NULL")
+        with pytest.raises(ConstraintViolation):
+            db.cypher("CREATE (:Doc {body: 'no title'})")
+        # IF NOT EXISTS: re-creating is a no-op
+        db.cypher("CREATE CONSTRAINT nn IF NOT EXISTS FOR (n:Doc) "
+                  "REQUIRE n.title IS NOT NULL")
+
+    def test_vector_index_options(self):
+        db = self._db()
+        db.cypher("CREATE VECTOR INDEX emb FOR (n:Chunk) ON (n.embedding) "
+                  "OPTIONS {indexConfig: {`vector.dimensions`: 1024, "
+                  "`vector.similarity_function`: 'cosine'}}")
+        meta = db.schema.vector_indexes["emb"]
+        assert meta.dims == 1024 and meta.similarity == "cosine"
+        rows = db.cypher("SHOW INDEXES").rows
+        assert any(r[4] == "VECTOR" for r in rows)
+
+    def test_fulltext_index(self):
+        db = self._db()
+        db.cypher("CREATE FULLTEXT INDEX ft FOR (n:Doc|Note) "
+                  "ON EACH [n.title, n.body]")
+        rows = db.cypher("SHOW INDEXES").rows
+        assert any(r[1] == "ft" and r[4] == "FULLTEXT" for r in rows)
+
+    def test_show_procedures_functions_databases(self):
+        db = self._db()
+        assert len(db.cypher("SHOW PROCEDURES").rows) > 10
+        assert len(db.cypher("SHOW FUNCTIONS").rows) > 50
+        assert db.cypher("SHOW DATABASES").rows[0][0]
