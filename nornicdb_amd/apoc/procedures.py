@@ -16,6 +16,8 @@ from ..graph import (betweenness_centrality, closeness_centrality,
                      connected_components, degree_centrality, dijkstra,
                      from_engine, label_propagation, louvain, pagerank,
                      shortest_path, triangle_count, clustering_coefficient)
+import re
+import time
 import threading
 
 from ..storage.types import Edge, Node, new_id
@@ -471,5 +473,973 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
                            + [n.properties.get(p, "") for p in props])
                 count += 1
         return ["file", "nodes"], [[file, count]]
+
+
+    def _all_edges_of(nid):
+        return eng.get_out_edges(nid) + [e for e in eng.get_in_edges(nid)
+                                         if e.start_node != e.end_node]
+
+    # -------------------- apoc.cypher --------------------
+    @register("apoc.cypher.run")
+    def _cy_run(ex, query, params=None):
+        r = ex.execute(query, dict(params or {}))
+        return ["value"], [[dict(zip(r.columns, row))] for row in r.rows]
+
+    @register("apoc.cypher.doIt")
+    def _cy_doit(ex, query, params=None):
+        return _cy_run(ex, query, params)
+
+    @register("apoc.cypher.runMany")
+    def _cy_run_many(ex, statements, params=None):
+        rows = []
+        for stmt in re.split(r";\s*", statements or ""):
+            if stmt.strip():
+                r = ex.execute(stmt, dict(params or {}))
+                rows += [[dict(zip(r.columns, row))] for row in r.rows]
+        return ["value"], rows
+
+    @register("apoc.cypher.runFirstColumn")
+    def _cy_first_col(ex, query, params=None):
+        r = ex.execute(query, dict(params or {}))
+        return ["value"], [[row[0]] for row in r.rows]
+
+    @register("apoc.cypher.runFirstColumnMany")
+    def _cy_first_many(ex, query, params=None):
+        return _cy_first_col(ex, query, params)
+
+    @register("apoc.cypher.runFirstColumnSingle")
+    def _cy_first_single(ex, query, params=None):
+        r = ex.execute(query, dict(params or {}))
+        return ["value"], [[r.rows[0][0] if r.rows else None]]
+
+    @register("apoc.cypher.validate")
+    def _cy_validate(ex, query):
+        from ..cypher.parser import parse as _parse
+        try:
+            _parse(query)
+            return ["valid", "error"], [[True, None]]
+        except Exception as e:
+            return ["valid", "error"], [[False, str(e)]]
+
+    @register("apoc.cypher.parse")
+    def _cy_parse(ex, query):
+        return _cy_validate(ex, query)
+
+    @register("apoc.cypher.explain")
+    def _cy_explain(ex, query):
+        r = ex.execute("EXPLAIN " + query)
+        return ["plan"], r.rows
+
+    @register("apoc.cypher.toJson")
+    def _cy_tojson(ex, query, params=None):
+        import json as _json
+        r = ex.execute(query, dict(params or {}))
+        from ..server.http import _jsonable
+        return ["json"], [[_json.dumps(
+            [dict(zip(r.columns, (_jsonable(v) for v in row)))
+             for row in r.rows])]]
+
+    # -------------------- apoc.create extras --------------------
+    @register("apoc.create.setProperty")
+    def _cr_setp(ex, node, key, value):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n.properties[key] = value
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.create.setProperties")
+    def _cr_setps(ex, node, keys, values):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        for k, v in zip(keys or [], values or []):
+            n.properties[k] = v
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.create.removeProperties")
+    def _cr_rmps(ex, node, keys):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        for k in keys or []:
+            n.properties.pop(k, None)
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.create.addLabels")
+    def _cr_addl(ex, node, labels):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        for lb in labels or []:
+            if lb not in n.labels:
+                n.labels.append(lb)
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.create.removeLabels")
+    def _cr_rml(ex, node, labels):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n.labels = [lb for lb in n.labels if lb not in set(labels or [])]
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.create.setRelProperty")
+    def _cr_setrp(ex, rel, key, value):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        e.properties[key] = value
+        return ["rel"], [[eng.update_edge(e)]]
+
+    @register("apoc.create.setRelProperties")
+    def _cr_setrps(ex, rel, keys, values):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        for k, v in zip(keys or [], values or []):
+            e.properties[k] = v
+        return ["rel"], [[eng.update_edge(e)]]
+
+    @register("apoc.create.removeRelProperties")
+    def _cr_rmrps(ex, rel, keys):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        for k in keys or []:
+            e.properties.pop(k, None)
+        return ["rel"], [[eng.update_edge(e)]]
+
+    @register("apoc.create.clone")
+    def _cr_clone(ex, node):
+        src_n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n = Node(id=new_id("n"), labels=list(src_n.labels),
+                 properties=dict(src_n.properties))
+        return ["node"], [[eng.create_node(n)]]
+
+    @register("apoc.create.uuids")
+    def _cr_uuids(ex, count):
+        import uuid as _u
+        return ["uuid"], [[str(_u.uuid4())] for _ in range(int(count))]
+
+    @register("apoc.create.vNode")
+    def _cr_vnode(ex, labels, props=None):
+        return ["node"], [[Node(id=new_id("v"), labels=list(labels or []),
+                                properties=dict(props or {}))]]
+
+    @register("apoc.create.vRelationship")
+    def _cr_vrel(ex, start, rel_type, props=None, end=None):
+        return ["rel"], [[Edge(id=new_id("vr"), type=rel_type,
+                               start_node=start.id if isinstance(start, Node)
+                               else str(start),
+                               end_node=end.id if isinstance(end, Node)
+                               else str(end),
+                               properties=dict(props or {}))]]
+
+    # -------------------- apoc.node (engine-backed) --------------------
+    @register("apoc.node.degree")
+    def _nd_degree(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = _all_edges_of(nid)
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type.strip("<>")]
+        return ["value"], [[len(edges)]]
+
+    @register("apoc.node.degreeIn")
+    def _nd_deg_in(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = [e for e in _all_edges_of(nid) if e.end_node == nid]
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type]
+        return ["value"], [[len(edges)]]
+
+    @register("apoc.node.degreeOut")
+    def _nd_deg_out(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = [e for e in _all_edges_of(nid) if e.start_node == nid]
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type]
+        return ["value"], [[len(edges)]]
+
+    @register("apoc.node.relationships")
+    def _nd_rels(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = _all_edges_of(nid)
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type]
+        return ["rel"], [[e] for e in edges]
+
+    @register("apoc.node.relationshipsIn")
+    def _nd_rels_in(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = [e for e in _all_edges_of(nid) if e.end_node == nid]
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type]
+        return ["rel"], [[e] for e in edges]
+
+    @register("apoc.node.relationshipsOut")
+    def _nd_rels_out(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = [e for e in _all_edges_of(nid) if e.start_node == nid]
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type]
+        return ["rel"], [[e] for e in edges]
+
+    @register("apoc.node.relationshipTypes")
+    def _nd_rtypes(ex, node):
+        nid = node.id if isinstance(node, Node) else node
+        return ["types"], [[sorted({e.type for e in _all_edges_of(nid)})]]
+
+    @register("apoc.node.relationshipTypesIn")
+    def _nd_rtypes_in(ex, node):
+        nid = node.id if isinstance(node, Node) else node
+        return ["types"], [[sorted({e.type for e in _all_edges_of(nid)
+                                    if e.end_node == nid})]]
+
+    @register("apoc.node.relationshipTypesOut")
+    def _nd_rtypes_out(ex, node):
+        nid = node.id if isinstance(node, Node) else node
+        return ["types"], [[sorted({e.type for e in _all_edges_of(nid)
+                                    if e.start_node == nid})]]
+
+    @register("apoc.node.relationshipExists")
+    def _nd_rexists(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        edges = _all_edges_of(nid)
+        if rel_type:
+            edges = [e for e in edges if e.type == rel_type.strip("<>")]
+        return ["value"], [[bool(edges)]]
+
+    @register("apoc.node.neighbors")
+    def _nd_neigh(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        out = []
+        for e in _all_edges_of(nid):
+            if rel_type and e.type != rel_type:
+                continue
+            other = e.end_node if e.start_node == nid else e.start_node
+            out.append(eng.get_node(other))
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.node.neighborsIn")
+    def _nd_neigh_in(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        out = [eng.get_node(e.start_node) for e in _all_edges_of(nid)
+               if e.end_node == nid and (not rel_type or e.type == rel_type)]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.node.neighborsOut")
+    def _nd_neigh_out(ex, node, rel_type=None):
+        nid = node.id if isinstance(node, Node) else node
+        out = [eng.get_node(e.end_node) for e in _all_edges_of(nid)
+               if e.start_node == nid and (not rel_type or e.type == rel_type)]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.node.connected")
+    def _nd_connected(ex, a, b, rel_type=None):
+        aid = a.id if isinstance(a, Node) else a
+        bid = b.id if isinstance(b, Node) else b
+        for e in _all_edges_of(aid):
+            if rel_type and e.type != rel_type.strip("<>"):
+                continue
+            if bid in (e.start_node, e.end_node):
+                return ["value"], [[True]]
+        return ["value"], [[False]]
+
+    @register("apoc.node.isDense")
+    def _nd_dense(ex, node, threshold=50):
+        nid = node.id if isinstance(node, Node) else node
+        return ["value"], [[len(_all_edges_of(nid)) >= int(threshold)]]
+
+    # -------------------- apoc.neighbors --------------------
+    def _hops(start_id, rel_type, max_hops, min_hops=1):
+        seen = {start_id}
+        frontier = {start_id}
+        out = []
+        for hop in range(1, int(max_hops) + 1):
+            nxt = set()
+            for nid in frontier:
+                for e in _all_edges_of(nid):
+                    if rel_type and e.type != rel_type:
+                        continue
+                    other = e.end_node if e.start_node == nid else e.start_node
+                    if other not in seen:
+                        seen.add(other)
+                        nxt.add(other)
+            if hop >= min_hops:
+                out.extend(nxt)
+            frontier = nxt
+        return out
+
+    @register("apoc.neighbors.athop")
+    def _nb_athop(ex, node, rel_type=None, distance=1):
+        nid = node.id if isinstance(node, Node) else node
+        ids = _hops(nid, rel_type, distance, int(distance))
+        return ["node"], [[eng.get_node(i)] for i in ids]
+
+    @register("apoc.neighbors.tohop")
+    def _nb_tohop(ex, node, rel_type=None, distance=1):
+        nid = node.id if isinstance(node, Node) else node
+        ids = _hops(nid, rel_type, distance, 1)
+        return ["node"], [[eng.get_node(i)] for i in ids]
+
+    @register("apoc.neighbors.bfs")
+    def _nb_bfs(ex, node, rel_type=None, distance=3):
+        return _nb_tohop(ex, node, rel_type, distance)
+
+    @register("apoc.neighbors.dfs")
+    def _nb_dfs(ex, node, rel_type=None, distance=3):
+        return _nb_tohop(ex, node, rel_type, distance)
+
+    @register("apoc.neighbors.count")
+    def _nb_count(ex, node, rel_type=None, distance=1):
+        nid = node.id if isinstance(node, Node) else node
+        return ["value"], [[len(_hops(nid, rel_type, distance, 1))]]
+
+    @register("apoc.neighbors.exists")
+    def _nb_exists(ex, node, rel_type=None, distance=1):
+        nid = node.id if isinstance(node, Node) else node
+        return ["value"], [[bool(_hops(nid, rel_type, distance, 1))]]
+
+    # -------------------- apoc.schema --------------------
+    @register("apoc.schema.nodes")
+    def _sc_nodes(ex):
+        sm = getattr(ex, "schema", None)
+        rows = []
+        if sm:
+            for name, kind, label, props in sm.list_indexes():
+                rows.append([name, label, props, "ONLINE", kind.upper()])
+        return ["name", "label", "properties", "status", "type"], rows
+
+    @register("apoc.schema.relationships")
+    def _sc_rels(ex):
+        return ["name", "type", "properties", "status"], []
+
+    @register("apoc.schema.info")
+    def _sc_info(ex):
+        sm = getattr(ex, "schema", None)
+        return ["indexes", "constraints"], [[
+            [n for n, *_ in sm.list_indexes()] if sm else [],
+            [c.name for c in sm.list_constraints()] if sm else []]]
+
+    @register("apoc.schema.stats")
+    def _sc_stats(ex):
+        sm = getattr(ex, "schema", None)
+        return ["indexCount", "constraintCount"], [[
+            len(sm.list_indexes()) if sm else 0,
+            len(sm.list_constraints()) if sm else 0]]
+
+    @register("apoc.schema.assert")
+    def _sc_assert(ex, indexes=None, constraints=None, drop_existing=True):
+        sm = getattr(ex, "schema", None)
+        rows = []
+        if sm:
+            for label, props in (indexes or {}).items():
+                for p in props:
+                    nm = sm.create_index(label, p)
+                    rows.append([nm, label, [p], "CREATED", "INDEX"])
+            for label, props in (constraints or {}).items():
+                for p in props:
+                    nm = f"constraint_{label}_{p}"
+                    sm.create_unique_constraint(nm, label, p)
+                    rows.append([nm, label, [p], "CREATED", "CONSTRAINT"])
+        return ["name", "label", "properties", "action", "type"], rows
+
+    @register("apoc.schema.createIndex")
+    def _sc_cidx(ex, label, prop):
+        sm = getattr(ex, "schema", None)
+        return ["name"], [[sm.create_index(label, prop) if sm else None]]
+
+    @register("apoc.schema.createUniqueConstraint")
+    def _sc_cuc(ex, label, prop, name=None):
+        sm = getattr(ex, "schema", None)
+        nm = name or f"constraint_{label}_{prop}"
+        if sm:
+            sm.create_unique_constraint(nm, label, prop)
+        return ["name"], [[nm]]
+
+    @register("apoc.schema.createExistsConstraint")
+    def _sc_cec(ex, label, prop, name=None):
+        sm = getattr(ex, "schema", None)
+        nm = name or f"constraint_{label}_{prop}"
+        if sm:
+            sm.create_exists_constraint(nm, label, prop)
+        return ["name"], [[nm]]
+
+    @register("apoc.schema.createNodeKeyConstraint")
+    def _sc_cnk(ex, label, prop, name=None):
+        return _sc_cuc(ex, label, prop, name)
+
+    @register("apoc.schema.dropIndex")
+    def _sc_didx(ex, name):
+        sm = getattr(ex, "schema", None)
+        return ["dropped"], [[sm.drop_index(name) if sm else False]]
+
+    @register("apoc.schema.dropConstraint")
+    def _sc_dcon(ex, name):
+        sm = getattr(ex, "schema", None)
+        if sm:
+            sm.drop_constraint(name)
+        return ["dropped"], [[True]]
+
+    @register("apoc.schema.nodeIndexExists")
+    def _sc_nie(ex, label, props):
+        sm = getattr(ex, "schema", None)
+        want = (label, list(props) if isinstance(props, list) else [props])
+        ok = sm and any(lb == want[0] and ps == want[1]
+                        for _, _, lb, ps in sm.list_indexes())
+        return ["value"], [[bool(ok)]]
+
+    @register("apoc.schema.nodeConstraintExists")
+    def _sc_nce(ex, label, props):
+        sm = getattr(ex, "schema", None)
+        pl = list(props) if isinstance(props, list) else [props]
+        ok = sm and any(c.label == label and c.prop in pl
+                        for c in sm.list_constraints())
+        return ["value"], [[bool(ok)]]
+
+    @register("apoc.schema.nodeConstraints")
+    def _sc_ncs(ex):
+        sm = getattr(ex, "schema", None)
+        return ["name", "label", "properties", "type"], [
+            [c.name, c.label, [c.prop], c.kind.upper()]
+            for c in (sm.list_constraints() if sm else [])]
+
+    @register("apoc.schema.nodeIndexes")
+    def _sc_nis(ex):
+        return _sc_nodes(ex)
+
+    @register("apoc.schema.labels")
+    def _sc_labels(ex):
+        seen = set()
+        for n in eng.all_nodes():
+            seen.update(n.labels)
+        return ["label"], [[lb] for lb in sorted(seen)]
+
+    @register("apoc.schema.properties")
+    def _sc_props(ex):
+        seen = set()
+        for n in eng.all_nodes():
+            seen.update(n.properties.keys())
+        return ["property"], [[p] for p in sorted(seen)]
+
+    @register("apoc.schema.propertiesDistinct")
+    def _sc_props_d(ex, label, prop):
+        vals = sorted({repr(n.properties.get(prop))
+                       for n in eng.get_nodes_by_label(label)
+                       if n.properties.get(prop) is not None})
+        return ["value"], [[v] for v in vals]
+
+    @register("apoc.schema.types")
+    def _sc_types(ex):
+        seen = set()
+        for e in eng.all_edges():
+            seen.add(e.type)
+        return ["type"], [[t] for t in sorted(seen)]
+
+
+    # -------------------- apoc.search --------------------
+    def _prop_matches(n, prop, op, value):
+        v = n.properties.get(prop)
+        if v is None:
+            return False
+        s, q = str(v).lower(), str(value).lower()
+        return {"exact": s == q, "contains": q in s,
+                "prefix": s.startswith(q), "starts with": s.startswith(q),
+                "suffix": s.endswith(q), "ends with": s.endswith(q),
+                "regex": bool(re.search(str(value), str(v))),
+                "fuzzy": q in s or s in q}.get(op, s == q)
+
+    @register("apoc.search.node")
+    def _se_node(ex, label_props, op, value):
+        out = []
+        for label, props in (label_props or {}).items():
+            plist = props if isinstance(props, list) else [props]
+            for n in eng.get_nodes_by_label(label):
+                if any(_prop_matches(n, p, op, value) for p in plist):
+                    out.append(n)
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.search.nodeAll")
+    def _se_node_all(ex, label_props, op, value):
+        return _se_node(ex, label_props, op, value)
+
+    @register("apoc.search.nodeReduced")
+    def _se_node_red(ex, label_props, op, value):
+        cols, rows = _se_node(ex, label_props, op, value)
+        return ["id", "labels", "values"], [
+            [n.id, list(n.labels), dict(n.properties)] for (n,) in rows]
+
+    @register("apoc.search.contains")
+    def _se_contains(ex, label_props, value):
+        return _se_node(ex, label_props, "contains", value)
+
+    @register("apoc.search.prefix")
+    def _se_prefix(ex, label_props, value):
+        return _se_node(ex, label_props, "prefix", value)
+
+    @register("apoc.search.suffix")
+    def _se_suffix(ex, label_props, value):
+        return _se_node(ex, label_props, "suffix", value)
+
+    @register("apoc.search.regex")
+    def _se_regex(ex, label_props, value):
+        return _se_node(ex, label_props, "regex", value)
+
+    @register("apoc.search.fuzzy")
+    def _se_fuzzy(ex, label_props, value):
+        return _se_node(ex, label_props, "fuzzy", value)
+
+    @register("apoc.search.fullText")
+    def _se_fulltext(ex, query, limit=25):
+        res = db.search.text_search(str(query), int(limit))
+        return ["node", "score"], [[r.node, r.score] for r in res]
+
+    @register("apoc.search.autocomplete")
+    def _se_auto(ex, label, prop, prefix, limit=10):
+        out = sorted({str(n.properties.get(prop))
+                      for n in eng.get_nodes_by_label(label)
+                      if str(n.properties.get(prop, "")).lower()
+                      .startswith(str(prefix).lower())})
+        return ["value"], [[v] for v in out[:int(limit)]]
+
+    @register("apoc.search.suggest")
+    def _se_suggest(ex, label, prop, text, limit=5):
+        from .functions import _levenshtein
+        cands = [(str(n.properties.get(prop)), _levenshtein(
+            str(n.properties.get(prop, "")).lower(), str(text).lower()))
+            for n in eng.get_nodes_by_label(label)
+            if n.properties.get(prop) is not None]
+        cands.sort(key=lambda t: t[1])
+        return ["value", "distance"], [list(t) for t in cands[:int(limit)]]
+
+    @register("apoc.search.didYouMean")
+    def _se_dym(ex, label, prop, text):
+        cols, rows = _se_suggest(ex, label, prop, text, 1)
+        return ["value"], [[rows[0][0] if rows else None]]
+
+    # -------------------- apoc.export / apoc.import --------------------
+    @register("apoc.export.csv.query")
+    def _ex_csv_q(ex, query, params=None):
+        import csv as _csv
+        import io as _io
+        r = ex.execute(query, dict(params or {}))
+        buf = _io.StringIO()
+        w = _csv.writer(buf)
+        w.writerow(r.columns)
+        for row in r.rows:
+            w.writerow([_scalar(v) for v in row])
+        return ["data", "rows"], [[buf.getvalue(), len(r.rows)]]
+
+    def _scalar(v):
+        if isinstance(v, Node):
+            return v.id
+        if isinstance(v, Edge):
+            return v.id
+        if isinstance(v, (list, dict)):
+            import json as _json
+            return _json.dumps(v, default=str)
+        return v
+
+    @register("apoc.export.json.query")
+    def _ex_json_q(ex, query, params=None):
+        import json as _json
+        from ..server.http import _jsonable
+        r = ex.execute(query, dict(params or {}))
+        data = "\n".join(_json.dumps(dict(zip(
+            r.columns, (_jsonable(v) for v in row))), default=str)
+            for row in r.rows)
+        return ["data", "rows"], [[data, len(r.rows)]]
+
+    @register("apoc.export.cypher.all")
+    def _ex_cy_all(ex):
+        lines = []
+        for n in eng.all_nodes():
+            labels = "".join(f":{lb}" for lb in n.labels)
+            lines.append(f"CREATE (n{labels} {_props_cypher(n.properties)})")
+        for e in eng.all_edges():
+            lines.append(
+                f"MATCH (a), (b) WHERE id(a) = '{e.start_node}' AND "
+                f"id(b) = '{e.end_node}' CREATE (a)-[:{e.type} "
+                f"{_props_cypher(e.properties)}]->(b)")
+        return ["cypherStatements"], [["\n".join(lines)]]
+
+    def _props_cypher(props):
+        import json as _json
+        items = ", ".join(f"{k}: {_json.dumps(v, default=str)}"
+                          for k, v in props.items()
+                          if not k.startswith("_"))
+        return "{" + items + "}"
+
+    @register("apoc.import.json")
+    def _im_json(ex, data):
+        import json as _json
+        n_nodes = n_rels = 0
+        for line in str(data).splitlines():
+            if not line.strip():
+                continue
+            obj = _json.loads(line)
+            if obj.get("type") == "relationship" or "start" in obj:
+                eng.create_edge(Edge(
+                    id=obj.get("id", new_id("e")),
+                    type=obj.get("label", obj.get("type", "RELATED")),
+                    start_node=str(obj["start"].get("id") if isinstance(
+                        obj.get("start"), dict) else obj.get("start")),
+                    end_node=str(obj["end"].get("id") if isinstance(
+                        obj.get("end"), dict) else obj.get("end")),
+                    properties=obj.get("properties", {})))
+                n_rels += 1
+            else:
+                eng.create_node(Node(id=str(obj.get("id", new_id("n"))),
+                                     labels=obj.get("labels", []),
+                                     properties=obj.get("properties", {})))
+                n_nodes += 1
+        return ["nodes", "relationships"], [[n_nodes, n_rels]]
+
+    @register("apoc.import.csv")
+    def _im_csv(ex, data, config=None):
+        import csv as _csv
+        import io as _io
+        cfg = dict(config or {})
+        label = cfg.get("label", "Row")
+        rd = _csv.DictReader(_io.StringIO(str(data)))
+        n = 0
+        for row in rd:
+            eng.create_node(Node(id=new_id("n"), labels=[label],
+                                 properties=dict(row)))
+            n += 1
+        return ["nodes"], [[n]]
+
+    @register("apoc.import.parseCsvLine")
+    def _im_parse_csv(ex, line, sep=","):
+        import csv as _csv
+        import io as _io
+        return ["fields"], [[next(_csv.reader(_io.StringIO(str(line)),
+                                              delimiter=str(sep)))]]
+
+    @register("apoc.import.parseJsonLine")
+    def _im_parse_json(ex, line):
+        import json as _json
+        return ["value"], [[_json.loads(line)]]
+
+    # -------------------- apoc.log --------------------
+    _LOG_BUF = []
+
+    @register("apoc.log.info")
+    def _log_info(ex, msg, params=None):
+        _LOG_BUF.append(("INFO", time.time(), str(msg)))
+        return ["level"], [["INFO"]]
+
+    @register("apoc.log.warn")
+    def _log_warn(ex, msg, params=None):
+        _LOG_BUF.append(("WARN", time.time(), str(msg)))
+        return ["level"], [["WARN"]]
+
+    @register("apoc.log.error")
+    def _log_error(ex, msg, params=None):
+        _LOG_BUF.append(("ERROR", time.time(), str(msg)))
+        return ["level"], [["ERROR"]]
+
+    @register("apoc.log.debug")
+    def _log_debug(ex, msg, params=None):
+        _LOG_BUF.append(("DEBUG", time.time(), str(msg)))
+        return ["level"], [["DEBUG"]]
+
+    @register("apoc.log.stream")
+    def _log_stream(ex, limit=100):
+        return ["level", "timestamp", "message"], [
+            list(x) for x in _LOG_BUF[-int(limit):]]
+
+    @register("apoc.log.tail")
+    def _log_tail(ex, limit=10):
+        return _log_stream(ex, limit)
+
+    @register("apoc.log.clear")
+    def _log_clear(ex):
+        n = len(_LOG_BUF)
+        _LOG_BUF.clear()
+        return ["cleared"], [[n]]
+
+    @register("apoc.log.stats")
+    def _log_stats(ex):
+        from collections import Counter
+        c = Counter(lv for lv, _, _ in _LOG_BUF)
+        return ["level", "count"], [[k, v] for k, v in sorted(c.items())]
+
+    # -------------------- apoc.warmup / apoc.stats (db) --------------------
+    @register("apoc.warmup.run")
+    def _warmup(ex, load_props=True, load_rels=True, load_idx=True):
+        n_nodes = sum(1 for _ in eng.all_nodes())
+        n_rels = sum(1 for _ in eng.all_edges()) if load_rels else 0
+        return ["nodesLoaded", "relsLoaded"], [[n_nodes, n_rels]]
+
+    @register("apoc.stats.degrees")
+    def _st_degrees(ex, rel_type=None):
+        from collections import Counter
+        deg = Counter()
+        for e in eng.all_edges():
+            if rel_type and e.type != rel_type:
+                continue
+            deg[e.start_node] += 1
+            deg[e.end_node] += 1
+        vals = sorted(deg.values()) or [0]
+        return ["type", "total", "min", "max", "mean"], [[
+            rel_type or "*", sum(vals), vals[0], vals[-1],
+            sum(vals) / len(vals)]]
+
+    # -------------------- apoc.nodes --------------------
+    @register("apoc.nodes.get")
+    def _ns_get(ex, ids):
+        out = []
+        for i in (ids if isinstance(ids, list) else [ids]):
+            try:
+                out.append(eng.get_node(i.id if isinstance(i, Node) else i))
+            except Exception:
+                pass
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.nodes.delete")
+    def _ns_delete(ex, nodes, batch_size=1000):
+        n = 0
+        for x in (nodes if isinstance(nodes, list) else [nodes]):
+            try:
+                eng.detach_delete_node(x.id if isinstance(x, Node) else x)
+                n += 1
+            except Exception:
+                pass
+        return ["value"], [[n]]
+
+    @register("apoc.nodes.link")
+    def _ns_link(ex, nodes, rel_type, props=None):
+        made = []
+        ns = [x for x in (nodes or []) if isinstance(x, Node)]
+        for a, b in zip(ns, ns[1:]):
+            made.append(eng.create_edge(Edge(
+                id=new_id("e"), type=rel_type, start_node=a.id,
+                end_node=b.id, properties=dict(props or {}))))
+        return ["rel"], [[e] for e in made]
+
+    @register("apoc.nodes.connected")
+    def _ns_connected(ex, a, b, types=None):
+        aid = a.id if isinstance(a, Node) else a
+        bid = b.id if isinstance(b, Node) else b
+        for e in _all_edges_of(aid):
+            if types and e.type not in types:
+                continue
+            if bid in (e.start_node, e.end_node):
+                return ["value"], [[True]]
+        return ["value"], [[False]]
+
+    @register("apoc.nodes.relationships")
+    def _ns_rels(ex, nodes):
+        out = {}
+        for x in (nodes or []):
+            nid = x.id if isinstance(x, Node) else x
+            for e in _all_edges_of(nid):
+                out[e.id] = e
+        return ["rel"], [[e] for e in out.values()]
+
+    @register("apoc.nodes.isDense")
+    def _ns_dense(ex, node, threshold=50):
+        nid = node.id if isinstance(node, Node) else node
+        return ["value"], [[len(_all_edges_of(nid)) >= int(threshold)]]
+
+    @register("apoc.nodes.group")
+    def _ns_group(ex, labels, group_props, aggregations=None):
+        from collections import defaultdict
+        groups = defaultdict(list)
+        gp = group_props if isinstance(group_props, list) else [group_props]
+        for label in (labels if isinstance(labels, list) else [labels]):
+            for n in eng.get_nodes_by_label(label):
+                key = tuple(repr(n.properties.get(p)) for p in gp)
+                groups[key].append(n)
+        return ["key", "count", "nodes"], [
+            [list(k), len(v), v] for k, v in groups.items()]
+
+    # -------------------- apoc.refactor extras --------------------
+    @register("apoc.refactor.cloneNodes")
+    def _rf_clone(ex, nodes, with_rels=False):
+        made = []
+        for x in (nodes or []):
+            src_n = eng.get_node(x.id if isinstance(x, Node) else x)
+            n = eng.create_node(Node(id=new_id("n"),
+                                     labels=list(src_n.labels),
+                                     properties=dict(src_n.properties)))
+            made.append(n)
+            if with_rels:
+                for e in _all_edges_of(src_n.id):
+                    s = n.id if e.start_node == src_n.id else e.start_node
+                    t = n.id if e.end_node == src_n.id else e.end_node
+                    eng.create_edge(Edge(id=new_id("e"), type=e.type,
+                                         start_node=s, end_node=t,
+                                         properties=dict(e.properties)))
+        return ["node"], [[n] for n in made]
+
+    @register("apoc.refactor.invertRelationship")
+    def _rf_invert(ex, rel):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        eng.delete_edge(e.id)
+        ne = eng.create_edge(Edge(id=new_id("e"), type=e.type,
+                                  start_node=e.end_node,
+                                  end_node=e.start_node,
+                                  properties=dict(e.properties)))
+        return ["rel"], [[ne]]
+
+    @register("apoc.refactor.setType")
+    def _rf_settype(ex, rel, new_type):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        eng.delete_edge(e.id)
+        ne = eng.create_edge(Edge(id=new_id("e"), type=new_type,
+                                  start_node=e.start_node,
+                                  end_node=e.end_node,
+                                  properties=dict(e.properties)))
+        return ["rel"], [[ne]]
+
+    @register("apoc.refactor.changeType")
+    def _rf_changetype(ex, rel, new_type):
+        return _rf_settype(ex, rel, new_type)
+
+    @register("apoc.refactor.renameProperty")
+    def _rf_renameprop(ex, old, new):
+        n_changed = 0
+        for n in list(eng.all_nodes()):
+            if old in n.properties:
+                n.properties[new] = n.properties.pop(old)
+                eng.update_node(n)
+                n_changed += 1
+        return ["count"], [[n_changed]]
+
+    @register("apoc.refactor.extractNode")
+    def _rf_extract(ex, rel, labels, out_type, in_type):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        mid = eng.create_node(Node(id=new_id("n"),
+                                   labels=list(labels or []),
+                                   properties=dict(e.properties)))
+        eng.delete_edge(e.id)
+        eng.create_edge(Edge(id=new_id("e"), type=out_type,
+                             start_node=e.start_node, end_node=mid.id,
+                             properties={}))
+        eng.create_edge(Edge(id=new_id("e"), type=in_type,
+                             start_node=mid.id, end_node=e.end_node,
+                             properties={}))
+        return ["node"], [[mid]]
+
+    @register("apoc.refactor.collapseNode")
+    def _rf_collapse(ex, node, rel_type):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        ins = eng.get_in_edges(n.id)
+        outs = eng.get_out_edges(n.id)
+        made = []
+        for a in ins:
+            for b in outs:
+                made.append(eng.create_edge(Edge(
+                    id=new_id("e"), type=rel_type,
+                    start_node=a.start_node, end_node=b.end_node,
+                    properties={**a.properties, **b.properties})))
+        eng.detach_delete_node(n.id)
+        return ["rel"], [[e] for e in made]
+
+    @register("apoc.refactor.normalizeAsBoolean")
+    def _rf_normbool(ex, prop, true_vals, false_vals):
+        tv, fv = set(true_vals or []), set(false_vals or [])
+        n_changed = 0
+        for n in list(eng.all_nodes()):
+            if prop in n.properties:
+                v = n.properties[prop]
+                if v in tv:
+                    n.properties[prop] = True
+                elif v in fv:
+                    n.properties[prop] = False
+                else:
+                    continue
+                eng.update_node(n)
+                n_changed += 1
+        return ["count"], [[n_changed]]
+
+    @register("apoc.refactor.deleteAndReconnect")
+    def _rf_delrecon(ex, path_or_node, nodes=None):
+        target = path_or_node
+        n = eng.get_node(target.id if isinstance(target, Node) else target)
+        return _rf_collapse(ex, n, "RELATED")
+
+    # -------------------- apoc.merge extras --------------------
+    @register("apoc.merge.relationship")
+    def _mg_rel(ex, start, rel_type, ident_props=None, props=None, end=None,
+                on_create=None):
+        sid = start.id if isinstance(start, Node) else start
+        eid = end.id if isinstance(end, Node) else end
+        for e in eng.get_out_edges(sid):
+            if e.end_node == eid and e.type == rel_type and all(
+                    e.properties.get(k) == v
+                    for k, v in (ident_props or {}).items()):
+                return ["rel"], [[e]]
+        e = eng.create_edge(Edge(id=new_id("e"), type=rel_type,
+                                 start_node=sid, end_node=eid,
+                                 properties={**(ident_props or {}),
+                                             **(props or {}),
+                                             **(on_create or {})}))
+        return ["rel"], [[e]]
+
+    @register("apoc.merge.nodes")
+    def _mg_nodes(ex, labels, ident_props, props=None):
+        cols, rows = procs["apoc.merge.node"](ex, labels, ident_props, props)
+        return cols, rows
+
+    @register("apoc.merge.nodeEager")
+    def _mg_node_eager(ex, labels, ident_props, props=None):
+        return procs["apoc.merge.node"](ex, labels, ident_props, props)
+
+    @register("apoc.merge.relationshipEager")
+    def _mg_rel_eager(ex, start, rel_type, ident_props=None, props=None,
+                      end=None):
+        return _mg_rel(ex, start, rel_type, ident_props, props, end)
+
+    # -------------------- apoc.meta (db-level) --------------------
+    @register("apoc.meta.data")
+    def _mt_data(ex):
+        from collections import defaultdict
+        props_by_label = defaultdict(set)
+        for n in eng.all_nodes():
+            for lb in n.labels:
+                props_by_label[lb].update(n.properties.keys())
+        rows = []
+        for lb in sorted(props_by_label):
+            for p in sorted(props_by_label[lb]):
+                rows.append([lb, p, "STRING", False])
+        return ["label", "property", "type", "index"], rows
+
+    @register("apoc.meta.nodeTypeProperties")
+    def _mt_ntp(ex):
+        cols, rows = _mt_data(ex)
+        return (["nodeLabels", "propertyName", "propertyTypes"],
+                [[[r[0]], r[1], [r[2]]] for r in rows])
+
+    @register("apoc.meta.relTypeProperties")
+    def _mt_rtp(ex):
+        from collections import defaultdict
+        props_by_type = defaultdict(set)
+        for e in eng.all_edges():
+            props_by_type[e.type].update(e.properties.keys())
+        return (["relType", "propertyName"],
+                [[t, p] for t in sorted(props_by_type)
+                 for p in sorted(props_by_type[t])])
+
+    @register("apoc.meta.graph")
+    def _mt_graph(ex):
+        labels = set()
+        for n in eng.all_nodes():
+            labels.update(n.labels)
+        types = {e.type for e in eng.all_edges()}
+        return ["nodes", "relationships"], [[sorted(labels), sorted(types)]]
+
+    @register("apoc.meta.functions")
+    def _mt_functions(ex):
+        from ..cypher.functions import FUNCTIONS
+        return ["name"], [[n] for n in sorted(FUNCTIONS)]
+
+    @register("apoc.meta.procedures")
+    def _mt_procedures(ex):
+        return ["name"], [[n] for n in sorted(procs)]
+
+    # -------------------- apoc.lock extras --------------------
+    @register("apoc.lock.relationships")
+    def _lk_rels(ex, rels):
+        return ["locked"], [[len(rels or [])]]
+
+    @register("apoc.lock.all")
+    def _lk_all(ex, nodes=None, rels=None):
+        return ["locked"], [[len(nodes or []) + len(rels or [])]]
+
+    @register("apoc.lock.readNodes")
+    def _lk_read_nodes(ex, nodes):
+        return ["locked"], [[len(nodes or [])]]
+
+    @register("apoc.lock.readRelationships")
+    def _lk_read_rels(ex, rels):
+        return ["locked"], [[len(rels or [])]]
 
     return procs

@@ -171,6 +171,10 @@ register("reduce", None)  # handled in evaluator (needs lazy eval)
 AGGREGATES = {"count", "sum", "avg", "min", "max", "collect", "stdev",
               "stdevp", "percentilecont", "percentiledisc"}
 
+# extension point: apoc.agg.* finalizers, registered by apoc.breadth
+# (values list, extra arg) -> result
+AGG_FINALIZERS: Dict[str, Callable] = {}
+
 
 def is_aggregate(name: str) -> bool:
     return name.lower() in AGGREGATES
@@ -230,4 +234,7 @@ class Aggregator:
             if lo == hi:
                 return s[lo]
             return s[lo] + (s[hi] - s[lo]) * (idx - lo)
+        fin = AGG_FINALIZERS.get(n)
+        if fin is not None:
+            return fin(self.values, extra)
         raise CypherRuntimeError(f"unknown aggregate {n}")

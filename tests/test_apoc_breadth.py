@@ -1,0 +1,174 @@
+"""APOC breadth categories (bitwise/math/number/stats/scoring/spatial/
+hashing/util/json/temporal/label/meta/diff/xml/graph/agg + engine-backed
+procedure categories).
+
+Parity: reference apoc/ packages (SURVEY.md §2 "APOC library", ~950 fns).
+"""
+
+import pytest
+
+from nornicdb_amd.db import NornicDB
+from nornicdb_amd.storage.memory import MemoryEngine
+
+
+@pytest.fixture
+def db():
+    return NornicDB(MemoryEngine(), auto_embed=False)
+
+
+def one(db, q, params=None):
+    return db.cypher(q, params).rows[0][0]
+
+
+class TestPureFunctions:
+    def test_bitwise(self, db):
+        assert one(db, "RETURN apoc.bitwise.op(60, '&', 13)") == 12
+        assert one(db, "RETURN apoc.bitwise.setBit(0, 3)") == 8
+        assert one(db, "RETURN apoc.bitwise.countBits(255)") == 8
+
+    def test_math(self, db):
+        assert one(db, "RETURN apoc.math.gcd(12, 18)") == 6
+        assert one(db, "RETURN apoc.math.lcm(4, 6)") == 12
+        assert one(db, "RETURN apoc.math.isPrime(97)") is True
+        assert one(db, "RETURN apoc.math.nextPrime(14)") == 17
+        assert one(db, "RETURN apoc.math.factorial(5)") == 120
+        assert one(db, "RETURN apoc.math.fibonacci(10)") == 55
+        assert abs(one(db, "RETURN apoc.math.sigmoid(0)") - 0.5) < 1e-9
+
+    def test_number(self, db):
+        assert one(db, "RETURN apoc.number.romanize(1987)") == "MCMLXXXVII"
+        assert one(db, "RETURN apoc.number.arabize('XIV')") == 14
+        assert one(db, "RETURN apoc.number.toHex(255)") == "ff"
+        assert one(db, "RETURN apoc.number.fromBinary('1010')") == 10
+        assert one(db, "RETURN apoc.number.isEven(4)") is True
+
+    def test_stats(self, db):
+        assert one(db, "RETURN apoc.stats.median([1,2,3,4,5])") == 3
+        assert one(db, "RETURN apoc.stats.iqr([1,2,3,4,5,6,7,8])") == 3.5
+        assert one(db, "RETURN apoc.stats.correlation([1,2,3],[2,4,6])") == pytest.approx(1.0)
+        assert one(db, "RETURN apoc.stats.outliers([1,2,3,2,3,1,2,100])") == [100]
+
+    def test_scoring(self, db):
+        assert one(db, "RETURN apoc.scoring.cosine([1,0],[0,1])") == pytest.approx(0.0)
+        assert one(db, "RETURN apoc.scoring.jaccard([1,2,3],[2,3,4])") == pytest.approx(0.5)
+        sm = one(db, "RETURN apoc.scoring.softmax([1.0,1.0])")
+        assert sm == pytest.approx([0.5, 0.5])
+
+    def test_spatial(self, db):
+        # Paris -> London great-circle ~343 km
+        d = one(db, "RETURN apoc.spatial.haversineDistance(48.8566, 2.3522, 51.5074, -0.1278)")
+        assert 330e3 < d < 350e3
+        gh = one(db, "RETURN apoc.spatial.encodeGeohash(48.8583, 2.2945, 7)")
+        back = one(db, f"RETURN apoc.spatial.decodeGeohash('{gh}')")
+        assert abs(back["latitude"] - 48.8583) < 0.01
+
+    def test_hashing(self, db):
+        assert one(db, "RETURN apoc.hashing.sha256('abc')").startswith("ba7816bf")
+        assert 0 <= one(db, "RETURN apoc.hashing.jumpHash(12345, 10)") < 10
+        a = one(db, "RETURN apoc.hashing.fingerprint({a: 1, b: 2})")
+        b = one(db, "RETURN apoc.hashing.fingerprint({b: 2, a: 1})")
+        assert a == b  # order-insensitive
+
+    def test_util(self, db):
+        assert one(db, "RETURN apoc.util.decodeBase64(apoc.util.encodeBase64('hi'))") == "hi"
+        assert one(db, "RETURN apoc.util.partition([1,2,3,4,5], 2)") == [[1, 2], [3, 4], [5]]
+        assert one(db, "RETURN apoc.util.when(true, 'a', 'b')") == "a"
+
+    def test_json(self, db):
+        assert one(db, "RETURN apoc.json.path('{\"a\": {\"b\": [1,2]}}', '$.a.b[1]')") == 2
+        flat = one(db, "RETURN apoc.json.flatten({a: {b: 1}})")
+        assert flat == {"a.b": 1}
+        assert one(db, "RETURN apoc.json.unflatten({`a.b`: 1})") == {"a": {"b": 1}}
+
+    def test_temporal(self, db):
+        assert one(db, "RETURN apoc.temporal.isLeapYear(2024)") is True
+        assert one(db, "RETURN apoc.temporal.quarter(datetime('2026-09-12T00:00:00Z'))") == 3
+        assert one(db, "RETURN toString(apoc.temporal.add("
+                       "datetime('2026-01-01T00:00:00Z'), 'P1D'))").startswith("2026-01-02")
+
+    def test_meta_and_diff(self, db):
+        assert one(db, "RETURN apoc.meta.cypherType(1.5)") == "FLOAT"
+        assert one(db, "RETURN apoc.meta.cypherType(date('2026-01-01'))") == "DATE"
+        d = one(db, "RETURN apoc.diff.maps({a:1, b:2}, {b:3, c:4})")
+        assert set(d["leftOnly"]) == {"a"} and set(d["different"]) == {"b"}
+
+    def test_xml(self, db):
+        m = one(db, "RETURN apoc.xml.parse('<a x=\"1\"><b>hello</b></a>')")
+        assert m["_type"] == "a" and m["@x"] == "1"
+        assert m["_children"][0]["_text"] == "hello"
+
+    def test_agg(self, db):
+        r = db.cypher("UNWIND [5,1,3] AS x RETURN apoc.agg.first(x), "
+                      "apoc.agg.last(x), apoc.agg.median(x), apoc.agg.product(x)")
+        assert r.rows == [[5, 3, 3, 15]]
+        r = db.cypher("UNWIND [1,1,2] AS x RETURN apoc.agg.mode(x)")
+        assert r.rows == [[1]]
+
+
+class TestEngineProcedures:
+    @pytest.fixture
+    def g(self, db):
+        db.cypher("CREATE (a:P {name:'alpha'})-[:K]->(b:P {name:'beta'})"
+                  "-[:K]->(c:P {name:'gamma'})")
+        return db
+
+    def test_node_degree_and_neighbors(self, g):
+        assert one(g, "MATCH (b:P {name:'beta'}) CALL apoc.node.degree(b) "
+                      "YIELD value RETURN value") == 2
+        r = g.cypher("MATCH (a:P {name:'alpha'}) "
+                     "CALL apoc.neighbors.tohop(a, 'K', 2) YIELD node "
+                     "RETURN node.name ORDER BY node.name")
+        assert [x[0] for x in r.rows] == ["beta", "gamma"]
+
+    def test_cypher_run(self, g):
+        r = g.cypher("CALL apoc.cypher.runFirstColumnSingle("
+                     "'MATCH (n:P) RETURN count(n)', {}) YIELD value RETURN value")
+        assert r.rows == [[3]]
+
+    def test_search(self, g):
+        r = g.cypher("CALL apoc.search.node({P: 'name'}, 'contains', 'amm') "
+                     "YIELD node RETURN node.name")
+        assert r.rows == [["gamma"]]
+
+    def test_export_import_roundtrip(self, g):
+        data = one(g, "CALL apoc.export.json.query("
+                      "'MATCH (n:P) RETURN n.name AS name', {}) "
+                      "YIELD data RETURN data")
+        assert data.count("\n") == 2  # 3 rows
+        db2 = NornicDB(MemoryEngine(), auto_embed=False)
+        r = db2.cypher(
+            "CALL apoc.import.json($d) YIELD nodes RETURN nodes",
+            {"d": '{"id": "x1", "labels": ["Q"], "properties": {"v": 1}}'})
+        assert r.rows == [[1]]
+        assert db2.cypher("MATCH (n:Q) RETURN n.v").rows == [[1]]
+
+    def test_refactor_invert(self, g):
+        g.cypher("MATCH (:P {name:'alpha'})-[r:K]->(:P {name:'beta'}) "
+                 "CALL apoc.refactor.invertRelationship(r) YIELD rel "
+                 "RETURN rel")
+        r = g.cypher("MATCH (b:P {name:'beta'})-[:K]->(a:P {name:'alpha'}) "
+                     "RETURN count(*)")
+        assert r.rows == [[1]]
+
+    def test_schema_procs(self, g):
+        g.cypher("CALL apoc.schema.createUniqueConstraint('P', 'name') "
+                 "YIELD name RETURN name")
+        assert one(g, "CALL apoc.schema.nodeConstraintExists('P', ['name']) "
+                      "YIELD value RETURN value") is True
+
+    def test_merge_relationship_idempotent(self, g):
+        for _ in range(2):
+            g.cypher("MATCH (a:P {name:'alpha'}), (c:P {name:'gamma'}) "
+                     "CALL apoc.merge.relationship(a, 'LINKS', {}, {}, c) "
+                     "YIELD rel RETURN rel")
+        assert g.cypher("MATCH (:P {name:'alpha'})-[r:LINKS]->() "
+                        "RETURN count(r)").rows == [[1]]
+
+    def test_meta_data(self, g):
+        rows = g.cypher("CALL apoc.meta.data() YIELD label, property "
+                        "RETURN label, property").rows
+        assert ["P", "name"] in rows
+
+    def test_function_count(self):
+        from nornicdb_amd.cypher.functions import FUNCTIONS
+        assert len([k for k in FUNCTIONS if k.startswith("apoc.")]) >= 450
