@@ -686,7 +686,9 @@ _r("apoc.hashing.fingerprintGraph", lambda nodes, rels=None: _hash_of(
                      list(nodes or []) + list(rels or []))))
 
 # ============================== apoc.util ==============================
-_r("apoc.util.md5", lambda l: _hash_of("md5", l))
+# apoc.util.md5/sha* take a LIST and hash the concatenated string forms
+_r("apoc.util.md5", lambda l: hashlib.md5("".join(
+    str(x) for x in (l if isinstance(l, list) else [l])).encode()).hexdigest())
 _r("apoc.util.md5Hex", lambda d: _hash_of("md5", d))
 _r("apoc.util.sha1Hex", lambda d: _hash_of("sha1", d))
 _r("apoc.util.sha256Hex", lambda d: _hash_of("sha256", d))
