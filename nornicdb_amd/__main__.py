@@ -158,6 +158,45 @@ def cmd_eval(args):
     mgr.close()
 
 
+def cmd_train(args):
+    """LoRA fine-tuning of the Heimdall model from DB content or JSONL
+    (reference neural/train.py)."""
+    import torch
+
+    from .embed.tokenizer import HashTokenizer
+    from .models.heimdall import HeimdallConfig, HeimdallModel
+    from .neural import (InstructionDataset, LoRATrainer, TrainConfig,
+                         export_merged, generate_dataset_from_db)
+    from .utils import load_config
+
+    cfg = load_config(args.config)
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    mcfg = HeimdallConfig() if device == "cuda" else HeimdallConfig.tiny()
+    model = HeimdallModel(mcfg).init_small()
+    tok = HashTokenizer(mcfg.vocab_size, mcfg.max_position)
+    if args.dataset:
+        ds = InstructionDataset.from_jsonl(args.dataset, tok)
+    else:
+        mgr = _open(args, cfg)
+        recs = generate_dataset_from_db(mgr.get(args.database))
+        mgr.close()
+        if not recs:
+            print("no training data in database", file=sys.stderr)
+            return
+        ds = InstructionDataset(recs, tok)
+    tr = LoRATrainer(model, TrainConfig(epochs=args.epochs,
+                                        batch_size=args.batch_size,
+                                        lr=args.lr), device=device)
+    hist = tr.train(ds)
+    if hist:
+        print(f"trained {tr.step} steps; loss {hist[0]['loss']:.3f} -> "
+              f"{hist[-1]['loss']:.3f}")
+    merged = tr.merge()
+    if args.out:
+        export_merged(merged, args.out)
+        print(f"exported merged model to {args.out}")
+
+
 def cmd_decay(args):
     from .cognitive import DecayManager
     from .utils import load_config
@@ -192,6 +231,16 @@ def main(argv=None):
     sp.add_argument("--database", default=None)
     sp.add_argument("--file", required=True)
     sp.set_defaults(fn=cmd_import)
+
+    sp = sub.add_parser("train")
+    sp.add_argument("--data-dir", default=None)
+    sp.add_argument("--database", default=None)
+    sp.add_argument("--dataset", default=None, help="JSONL prompt/completion")
+    sp.add_argument("--epochs", type=int, default=1)
+    sp.add_argument("--batch-size", type=int, default=4)
+    sp.add_argument("--lr", type=float, default=2e-4)
+    sp.add_argument("--out", default=None, help="export dir for merged model")
+    sp.set_defaults(fn=cmd_train)
 
     sp = sub.add_parser("eval")
     sp.add_argument("--data-dir", default=None)
