@@ -59,20 +59,56 @@ class WALCorruption(Exception):
 
 
 class WAL:
+    """segment_bytes > 0 enables segment rotation (reference wal.go
+    segment files): the active log is <path>, sealed segments are
+    <path>.<n> in increasing age order; replay walks sealed segments
+    oldest-first then the active file; truncate() (post-snapshot)
+    removes everything."""
+
     def __init__(self, path: str, sync_interval: float = 0.05,
-                 sync_on_write: bool = False):
+                 sync_on_write: bool = False, segment_bytes: int = 0):
         self.path = path
         self._lock = threading.Lock()
         self._f = open(path, "ab")
         self._sync_on_write = sync_on_write
         self.degraded = False
         self._sync_interval = sync_interval
+        self._segment_bytes = segment_bytes
         self._dirty = False
         self._stop = threading.Event()
         self._flusher: Optional[threading.Thread] = None
         if not sync_on_write and sync_interval > 0:
             self._flusher = threading.Thread(target=self._flush_loop, daemon=True)
             self._flusher.start()
+
+    # ---- segments ----
+    @staticmethod
+    def _segments(path) -> list:
+        """Sealed segment paths, oldest first."""
+        d = os.path.dirname(path) or "."
+        base = os.path.basename(path)
+        segs = []
+        for fn in os.listdir(d):
+            if fn.startswith(base + "."):
+                suffix = fn[len(base) + 1:]
+                if suffix.isdigit():
+                    segs.append((int(suffix), os.path.join(d, fn)))
+        return [p for _, p in sorted(segs)]
+
+    def _maybe_rotate_locked(self):
+        if not self._segment_bytes:
+            return
+        if self._f.tell() < self._segment_bytes:
+            return
+        self._f.flush()
+        os.fsync(self._f.fileno())
+        self._f.close()
+        segs = self._segments(self.path)
+        nxt = (int(os.path.basename(segs[-1]).rsplit(".", 1)[1]) + 1
+               if segs else 0)
+        os.rename(self.path, f"{self.path}.{nxt}")
+        self._f = open(self.path, "ab")
+        self._dirty = False
 
     def _flush_loop(self):
         while not self._stop.wait(self._sync_interval):
@@ -87,6 +123,7 @@ class WAL:
             try:
                 self._f.write(rec)
                 self._dirty = True
+                self._maybe_rotate_locked()
                 if self._sync_on_write:
                     self._f.flush()
                     os.fsync(self._f.fileno())
@@ -105,14 +142,26 @@ class WAL:
                 self._dirty = False
 
     def size(self) -> int:
+        """Total bytes across sealed segments + the active file."""
         with self._lock:
             self._f.flush()
-            return os.path.getsize(self.path)
+            total = os.path.getsize(self.path)
+            for seg in self._segments(self.path):
+                try:
+                    total += os.path.getsize(seg)
+                except OSError:
+                    pass
+            return total
 
     def truncate(self):
-        """Reset the log (after a snapshot)."""
+        """Reset the log (after a snapshot) — removes sealed segments too."""
         with self._lock:
             self._f.close()
+            for seg in self._segments(self.path):
+                try:
+                    os.remove(seg)
+                except OSError:
+                    pass
             self._f = open(self.path, "wb")
             self._f.close()
             self._f = open(self.path, "ab")
@@ -128,7 +177,14 @@ class WAL:
 
     @staticmethod
     def replay(path: str, tolerate_corruption: bool = True):
-        """Yield (op, payload) records; handles torn tails and CRC errors."""
+        """Yield (op, payload) records across sealed segments (oldest
+        first) then the active file; handles torn tails and CRC errors."""
+        for seg in WAL._segments(path):
+            yield from WAL._replay_one(seg, tolerate_corruption)
+        yield from WAL._replay_one(path, tolerate_corruption)
+
+    @staticmethod
+    def _replay_one(path: str, tolerate_corruption: bool = True):
         if not os.path.exists(path):
             return
         with open(path, "rb") as f:

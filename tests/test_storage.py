@@ -369,3 +369,34 @@ class TestConcurrency:
         assert db.engine.node_count() == 60
         assert len(db.search.emb) == 60
         mgr.close()
+
+
+class TestWALSegments:
+    """WAL segment rotation (reference pkg/storage/wal.go segment files)."""
+
+    def test_rotation_and_replay(self, tmp_path):
+        from nornicdb_amd.storage.wal import OP_CREATE_NODE, WAL
+        p = str(tmp_path / "seg.wal")
+        w = WAL(p, sync_on_write=True, segment_bytes=256)
+        for i in range(50):
+            w.append(OP_CREATE_NODE, {"id": f"n{i}", "i": i})
+        w.close()
+        import os
+        segs = [f for f in os.listdir(tmp_path) if f.startswith("seg.wal.")]
+        assert len(segs) >= 2  # rotated at least twice
+        got = [payload["i"] for op, payload in WAL.replay(p)]
+        assert got == list(range(50))  # ordered across segments
+
+    def test_truncate_removes_segments(self, tmp_path):
+        from nornicdb_amd.storage.wal import OP_CREATE_NODE, WAL
+        import os
+        p = str(tmp_path / "seg2.wal")
+        w = WAL(p, sync_on_write=True, segment_bytes=128)
+        for i in range(30):
+            w.append(OP_CREATE_NODE, {"id": f"n{i}"})
+        assert w.size() > 128
+        w.truncate()
+        assert w.size() == 0
+        assert not [f for f in os.listdir(tmp_path) if f.startswith("seg2.wal.")]
+        w.close()
+        assert list(WAL.replay(p)) == []
