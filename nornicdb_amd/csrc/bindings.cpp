@@ -34,6 +34,12 @@ at::Tensor add_layernorm(at::Tensor a, c10::optional<at::Tensor> b,
 at::Tensor bias_gelu(at::Tensor x, at::Tensor bias);
 at::Tensor mean_pool_l2norm(at::Tensor x, c10::optional<at::Tensor> mask);
 at::Tensor flash_attn_nc(at::Tensor q, at::Tensor k, at::Tensor v);
+void decode_step(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
+                 at::Tensor attn, at::Tensor h, at::Tensor rope_cos,
+                 at::Tensor rope_sin, long long n_layers, long long hidden,
+                 long long n_heads, long long n_kv, long long hd,
+                 long long inter, long long max_len, double rms_eps,
+                 long long pos);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "NornicDB-AMD CDNA4 (gfx950) native kernels";
@@ -58,6 +64,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("mask") = c10::nullopt);
   m.def("flash_attn_nc", &flash_attn_nc,
         "non-causal flash attention fwd, head_dim 64 (bf16)");
+  m.def("decode_step", &decode_step,
+        "fused cooperative single-token decode (all layers, one launch)",
+        py::arg("layer_ptrs"), py::arg("x"), py::arg("q"), py::arg("attn"),
+        py::arg("h"), py::arg("rope_cos"), py::arg("rope_sin"),
+        py::arg("n_layers"), py::arg("hidden"), py::arg("n_heads"),
+        py::arg("n_kv"), py::arg("hd"), py::arg("inter"),
+        py::arg("max_len"), py::arg("rms_eps"), py::arg("pos"));
   m.def("pagerank_contrib", &pagerank_contrib, "rank/outdeg elementwise");
   m.def("pagerank_gather", &pagerank_gather,
         "CSR pull-gather pagerank iteration (wave per row)");

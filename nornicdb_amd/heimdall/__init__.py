@@ -61,14 +61,21 @@ class HeimdallManager:
         self._torch = torch
         self._graphed = None
         if self.device.startswith("cuda"):
+            # preference: fused cooperative decoder (one launch per token,
+            # csrc/decode_fused.hip) > hipGraph replay > eager
             try:
-                from ..models.heimdall import GraphedDecoder
-                self._graphed = GraphedDecoder(self.model,
-                                               max_len=cfg.max_position
-                                               if cfg.max_position <= 2048
-                                               else 2048).capture()
+                from ..models.heimdall import FusedDecoder
+                self._graphed = FusedDecoder(self.model,
+                                             max_len=min(cfg.max_position,
+                                                         2048))
             except Exception:
-                self._graphed = None  # eager fallback
+                try:
+                    from ..models.heimdall import GraphedDecoder
+                    self._graphed = GraphedDecoder(
+                        self.model, max_len=min(cfg.max_position,
+                                                2048)).capture()
+                except Exception:
+                    self._graphed = None  # eager fallback
 
     # ---- plugins ----
     def register_plugin(self, plugin: HeimdallPlugin):

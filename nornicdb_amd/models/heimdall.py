@@ -294,3 +294,110 @@ class GraphedDecoder:
             last = self.out.float()
             pos += 1
         return [int(t) for t in torch.cat(out_t).tolist()] if out_t else []
+
+
+class FusedDecoder:
+    """Single-launch cooperative decode step (csrc/decode_fused.hip).
+
+    All decoder layers for one token run in ONE hipLaunchCooperativeKernel
+    (grid-wide syncs between GEMV/attention stages); only the final norm,
+    lm_head GEMV and sampling remain as torch ops (~4 launches/token vs
+    ~180 for the eager loop). Falls back is handled by HeimdallManager.
+    Requires head_dim 64 and max_len <= 2048.
+    """
+
+    def __init__(self, model: HeimdallModel, max_len: int = 2048):
+        from ..ops import require_native
+        self.nat = require_native()
+        self.m = model
+        c = model.cfg
+        self.cfg = c
+        dev = next(model.parameters()).device
+        assert dev.type == "cuda", "FusedDecoder is GPU-only"
+        self.dev = dev
+        self.hd = c.hidden_size // c.num_heads
+        assert self.hd == 64, "fused decode supports head_dim 64"
+        self.max_len = min(max_len, c.max_position, 2048)
+        dt = next(model.parameters()).dtype
+        assert dt == torch.bfloat16, "fused decode expects bf16 weights"
+        self.cache_k = [torch.zeros(c.num_kv_heads, self.max_len, self.hd,
+                                    device=dev, dtype=dt)
+                        for _ in range(c.num_layers)]
+        self.cache_v = [torch.zeros_like(self.cache_k[0])
+                        for _ in range(c.num_layers)]
+        ptrs = []
+        for li, layer in enumerate(model.layers):
+            ptrs.append([
+                layer.ln1.weight.data_ptr(),
+                layer.q_proj.weight.data_ptr(), layer.q_proj.bias.data_ptr(),
+                layer.k_proj.weight.data_ptr(), layer.k_proj.bias.data_ptr(),
+                layer.v_proj.weight.data_ptr(), layer.v_proj.bias.data_ptr(),
+                layer.o_proj.weight.data_ptr(),
+                layer.ln2.weight.data_ptr(),
+                layer.gate_proj.weight.data_ptr(),
+                layer.up_proj.weight.data_ptr(),
+                layer.down_proj.weight.data_ptr(),
+                self.cache_k[li].data_ptr(), self.cache_v[li].data_ptr(),
+            ])
+        self.layer_ptrs = torch.tensor(ptrs, dtype=torch.int64, device=dev)
+        self.rope_cos = model.rope_cos.float().contiguous().to(dev)
+        self.rope_sin = model.rope_sin.float().contiguous().to(dev)
+        self.x = torch.zeros(c.hidden_size, device=dev, dtype=torch.float32)
+        self.qs = torch.zeros_like(self.x)
+        self.attn = torch.zeros_like(self.x)
+        self.hbuf = torch.zeros(c.intermediate_size, device=dev,
+                                dtype=torch.float32)
+
+    @torch.no_grad()
+    def step_logits(self, token_id: torch.Tensor, pos: int) -> torch.Tensor:
+        """One fused decode step; returns fp32 logits [vocab]."""
+        m, c = self.m, self.cfg
+        self.x.copy_(m.embed.weight.index_select(
+            0, token_id.view(1)).float().view(-1))
+        self.nat.decode_step(self.layer_ptrs, self.x, self.qs, self.attn,
+                             self.hbuf, self.rope_cos, self.rope_sin,
+                             c.num_layers, c.hidden_size, c.num_heads,
+                             c.num_kv_heads, self.hd, c.intermediate_size,
+                             self.max_len, c.rms_eps, pos)
+        xn = self.x * torch.rsqrt(self.x.pow(2).mean() + c.rms_eps)
+        xn = xn * m.norm.weight.float()
+        return (xn.to(torch.bfloat16) @ m.lm_head.weight.T).float()
+
+    @torch.no_grad()
+    def generate(self, token_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.8, top_k: int = 40,
+                 eos_id=None, stream_cb=None):
+        """Prefill with the dynamic forward, then fused per-token steps."""
+        m = self.m
+        s = token_ids.shape[1]
+        caches = [(None, None)] * len(m.layers)
+        logits, caches = m.forward(token_ids, kv_caches=caches, pos0=0)
+        for li, (k, v) in enumerate(caches):
+            self.cache_k[li][:, :s] = k[0]
+            self.cache_v[li][:, :s] = v[0]
+        last = logits[:, -1, :].float().view(-1)
+        out = []
+        pos = s
+        cur = None
+        for _ in range(max_new_tokens):
+            if temperature <= 0:
+                cur = last.argmax(-1, keepdim=True)
+            else:
+                sc = last / temperature
+                if top_k:
+                    vv, ix = torch.topk(sc, min(top_k, sc.shape[-1]))
+                    j = torch.multinomial(torch.softmax(vv, -1), 1)
+                    cur = ix.index_select(0, j.view(-1))
+                else:
+                    cur = torch.multinomial(torch.softmax(sc, -1), 1)
+            tok = int(cur.item())
+            out.append(tok)
+            if stream_cb is not None:
+                stream_cb(tok)
+            if eos_id is not None and tok == eos_id:
+                break
+            if pos >= self.max_len:
+                break
+            last = self.step_logits(cur, pos)
+            pos += 1
+        return out
