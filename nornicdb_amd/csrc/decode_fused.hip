@@ -29,7 +29,7 @@ typedef __bf16 bf16x2d __attribute__((ext_vector_type(2)));
 typedef __bf16 bf16x8d __attribute__((ext_vector_type(8)));
 
 #define DWG 256          // threads per workgroup
-#define DGRID 96         // workgroups (measured best: grid.sync cost grows with WGs, GEMV stages saturate by ~96)
+#define DGRID 64         // workgroups (measured best: grid.sync cost grows with WGs, GEMV stages saturate by ~64)
 
 // per-layer device pointers (filled host-side into an int64 tensor)
 struct LayerPtrs {
