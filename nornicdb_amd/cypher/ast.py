@@ -101,6 +101,39 @@ class CountSubquery:
 
 
 @dataclass
+class MapProjection:
+    """n {.name, .age, .*, extra: expr} (Cypher map projection)."""
+    expr: Any
+    items: List[Any]   # ("prop", name) | ("all",) | ("kv", key, expr) | ("var", name)
+
+
+@dataclass
+class PatternComprehension:
+    """[(a)-[:R]->(b) WHERE pred | proj]"""
+    pattern: "PatternPath"
+    where: Optional[Any]
+    proj: Any
+
+
+@dataclass
+class Reduce:
+    """reduce(acc = init, x IN list | expr)"""
+    acc: str
+    init: Any
+    var: str
+    source: Any
+    expr: Any
+
+
+@dataclass
+class TypePredicate:
+    """expr IS [NOT] :: TYPE"""
+    expr: Any
+    type_name: str
+    negated: bool = False
+
+
+@dataclass
 class SubqueryExpr:
     """EXISTS { ... } / COUNT { ... } expression subquery (Cypher 5)."""
     kind: str              # "EXISTS" | "COUNT" | "COLLECT"
@@ -113,6 +146,8 @@ class NodePattern:
     var: Optional[str]
     labels: List[str]
     props: Optional[Any]  # MapLit / Param
+    or_labels: bool = False        # :A|B -> match ANY of labels
+    where: Optional[Any] = None    # inline (n:L WHERE expr)
 
 
 @dataclass

@@ -127,6 +127,8 @@ class Executor:
         if pat.var or getattr(pat, "shortest", None) or len(pat.elements) != 1:
             return None
         np_ = pat.elements[0]
+        if getattr(np_, "or_labels", False) or getattr(np_, "where", None):
+            return None
         if not (isinstance(r, A.ReturnClause) and not r.distinct
                 and not r.order_by and r.skip is None and r.limit is None
                 and not r.star and len(r.items) == 1):
@@ -535,6 +537,62 @@ class Executor:
             for _ in self._match_path(e.pattern, dict(row), params, limit=1):
                 return True
             return False
+        if isinstance(e, A.MapProjection):
+            base = self._eval(e.expr, row, params)
+            if base is None:
+                return None
+            props = (dict(base.properties) if isinstance(base, (Node, Edge))
+                     else dict(base or {}))
+            out = {}
+            for it in e.items:
+                if it[0] == "all":
+                    out.update(props)
+                elif it[0] == "prop":
+                    out[it[1]] = props.get(it[1])
+                elif it[0] == "kv":
+                    out[it[1]] = self._eval(it[2], row, params)
+                else:  # var
+                    out[it[1]] = row.get(it[1])
+            return out
+        if isinstance(e, A.PatternComprehension):
+            out = []
+            for bound in self._match_path(e.pattern, dict(row), params):
+                if e.where is not None and \
+                        self._eval(e.where, bound, params) is not True:
+                    continue
+                out.append(self._eval(e.proj, bound, params))
+            return out
+        if isinstance(e, A.Reduce):
+            acc = self._eval(e.init, row, params)
+            src_l = self._eval(e.source, row, params)
+            for v in (src_l or []):
+                r2 = dict(row)
+                r2[e.acc] = acc
+                r2[e.var] = v
+                acc = self._eval(e.expr, r2, params)
+            return acc
+        if isinstance(e, A.TypePredicate):
+            v = self._eval(e.expr, row, params)
+            t = e.type_name
+            ok = {
+                "INTEGER": lambda x: isinstance(x, int) and not isinstance(x, bool),
+                "INT": lambda x: isinstance(x, int) and not isinstance(x, bool),
+                "FLOAT": lambda x: isinstance(x, float),
+                "STRING": lambda x: isinstance(x, str),
+                "BOOLEAN": lambda x: isinstance(x, bool),
+                "BOOL": lambda x: isinstance(x, bool),
+                "LIST": lambda x: isinstance(x, list),
+                "MAP": lambda x: isinstance(x, dict),
+                "NODE": lambda x: isinstance(x, Node),
+                "RELATIONSHIP": lambda x: isinstance(x, Edge),
+                "PATH": lambda x: isinstance(x, Path),
+                "NULL": lambda x: x is None,
+                "NUMBER": lambda x: isinstance(x, (int, float))
+                          and not isinstance(x, bool),
+            }.get(t, lambda x: False)(v)
+            if v is None and t != "NULL":
+                ok = False
+            return (not ok) if e.negated else ok
         if isinstance(e, A.SubqueryExpr):
             cols, rows2 = self._run_subquery(e.query, dict(row), params)
             if e.kind == "EXISTS":
@@ -734,7 +792,13 @@ class Executor:
                         return [n for n in r
                                 if all(lb in n.labels for lb in np.labels)
                                 and self._props_match(n, props)]
-        if np.labels:
+        if np.labels and getattr(np, "or_labels", False):
+            seen = {}
+            for lb in np.labels:
+                for n in self.engine.get_nodes_by_label(lb):
+                    seen[n.id] = n
+            cands = list(seen.values())
+        elif np.labels:
             cands = self.engine.get_nodes_by_label(np.labels[0])
             if len(np.labels) > 1:
                 cands = [n for n in cands if all(lb in n.labels for lb in np.labels)]
@@ -742,6 +806,14 @@ class Executor:
             cands = self.engine.all_nodes()
         if props:
             cands = [n for n in cands if self._props_match(n, props)]
+        if getattr(np, "where", None) is not None and np.var:
+            out = []
+            for n in cands:
+                r2 = dict(row)
+                r2[np.var] = n
+                if self._eval(np.where, r2, params) is True:
+                    out.append(n)
+            cands = out
         return cands
 
     @staticmethod

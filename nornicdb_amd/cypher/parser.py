@@ -619,15 +619,24 @@ class Parser:
         if t.kind == "IDENT":
             var = self.next().value
         labels = []
+        or_labels = False
         while self.try_op(":"):
             labels.append(self.ident())
+            while self.try_op("|"):       # :A|B -> OR semantics
+                labels.append(self.ident())
+                or_labels = True
         props = None
         if self.at_op("{"):
             props = self._map_lit()
         elif self.peek().kind == "PARAM":
             props = A.Param(self.next().value)
+        where = None
+        if self._at_word("WHERE"):
+            self.next()
+            where = self._expr()
         self.eat_op(")")
-        return A.NodePattern(var, labels, props)
+        return A.NodePattern(var, labels, props, or_labels=or_labels,
+                             where=where)
 
     def _rel_pattern(self) -> A.RelPattern:
         direction = "both"
@@ -748,6 +757,15 @@ class Parser:
             elif self.at_kw("IS"):
                 self.next()
                 neg = self.try_kw("NOT")
+                if self.at_op("::") or self.at_op(":"):
+                    if not self.try_op("::"):
+                        self.eat_op(":")
+                        self.eat_op(":")
+                    tname = self.next().value.upper()
+                    if tname == "LOCAL" or tname == "ZONED":
+                        tname += " " + self.next().value.upper()
+                    e = A.TypePredicate(e, tname, negated=neg)
+                    continue
                 self.eat_kw("NULL")
                 e = A.UnOp("IS NOT NULL" if neg else "IS NULL", e)
                 continue
@@ -818,6 +836,26 @@ class Parser:
                 while self.try_op(":"):
                     labels.append(self.ident())
                 e = A.FuncCall("__haslabels", [e, A.Lit(labels)])
+            elif self.at_op("{") and isinstance(e, (A.Var, A.Prop)):
+                # map projection n {.name, .*, key: expr, var}
+                self.next()
+                items = []
+                while not self.at_op("}"):
+                    if self.try_op("."):
+                        if self.try_op("*"):
+                            items.append(("all",))
+                        else:
+                            items.append(("prop", self.name_part()))
+                    else:
+                        name = self.name_part()
+                        if self.try_op(":"):
+                            items.append(("kv", name, self._expr()))
+                        else:
+                            items.append(("var", name))
+                    if not self.try_op(","):
+                        break
+                self.eat_op("}")
+                e = A.MapProjection(e, items)
             else:
                 return e
 
@@ -926,6 +964,20 @@ class Parser:
                     proj = self._expr()
                 self.eat_op("]")
                 return A.ListComp(var, src, where, proj)
+            # pattern comprehension [(a)-[:R]->(b) WHERE p | proj]
+            if self.at_op("("):
+                save = self.i
+                try:
+                    pat = self._pattern_path()
+                    if len(pat.elements) > 1:
+                        where = self._expr() if self.try_kw("WHERE") else None
+                        self.eat_op("|")
+                        proj = self._expr()
+                        self.eat_op("]")
+                        return A.PatternComprehension(pat, where, proj)
+                    self.i = save
+                except CypherSyntaxError:
+                    self.i = save
             items = []
             if not self.at_op("]"):
                 items.append(self._expr())
@@ -936,6 +988,22 @@ class Parser:
         if t.kind == "OP" and t.value == "{":
             return self._map_lit()
         if t.kind == "IDENT":
+            # reduce(acc = init, x IN list | expr)
+            if (t.value.lower() == "reduce" and self.peek(1).kind == "OP"
+                    and self.peek(1).value == "("):
+                self.next()
+                self.eat_op("(")
+                acc = self.ident()
+                self.eat_op("=")
+                init = self._expr()
+                self.eat_op(",")
+                var = self.ident()
+                self.eat_kw("IN")
+                source = self._expr()
+                self.eat_op("|")
+                body = self._expr()
+                self.eat_op(")")
+                return A.Reduce(acc, init, var, source, body)
             # function call?
             if self.peek(1).kind == "OP" and self.peek(1).value == "(":
                 name = self.next().value

@@ -496,3 +496,60 @@ class TestSchemaDDL:
         assert len(db.cypher("SHOW PROCEDURES").rows) > 10
         assert len(db.cypher("SHOW FUNCTIONS").rows) > 50
         assert db.cypher("SHOW DATABASES").rows[0][0]
+
+
+# ---------------------------------------------------------- conformance v5
+class TestCypher5Conformance:
+    """Map projections, pattern comprehensions, reduce(), label-OR,
+    inline WHERE, type predicates (Neo4j 5 surface)."""
+
+    def _ex(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        ex.execute("CREATE (a:P {name:'ann', age: 30})-[:K]->"
+                   "(b:Q {name:'bob', age: 25})")
+        return ex
+
+    def test_map_projection(self):
+        ex = self._ex()
+        assert ex.execute("MATCH (n:P) RETURN n {.name, .age}").rows == \
+            [[{"name": "ann", "age": 30}]]
+        assert ex.execute("MATCH (n:P) RETURN n {.*}").rows == \
+            [[{"name": "ann", "age": 30}]]
+        assert ex.execute("MATCH (n:P) RETURN n {.name, extra: 1+1}").rows == \
+            [[{"name": "ann", "extra": 2}]]
+
+    def test_pattern_comprehension(self):
+        ex = self._ex()
+        assert ex.execute(
+            "MATCH (a:P) RETURN [(a)-[:K]->(b) | b.name]").rows == [[["bob"]]]
+        assert ex.execute(
+            "MATCH (a:P) RETURN [(a)-[:K]->(b) WHERE b.age > 99 | b.name]"
+        ).rows == [[[]]]
+
+    def test_reduce(self):
+        ex = self._ex()
+        assert ex.execute(
+            "RETURN reduce(s = 0, x IN [1,2,3] | s + x)").rows == [[6]]
+        assert ex.execute(
+            "RETURN reduce(acc = '', w IN ['a','b'] | acc + w)").rows == [["ab"]]
+
+    def test_label_or(self):
+        ex = self._ex()
+        assert ex.execute("MATCH (n:P|Q) RETURN count(n)").rows == [[2]]
+
+    def test_inline_where(self):
+        ex = self._ex()
+        assert ex.execute(
+            "MATCH (n:P WHERE n.age > 26) RETURN n.name").rows == [["ann"]]
+        assert ex.execute(
+            "MATCH (n:Q WHERE n.age > 26) RETURN count(n)").rows == [[0]]
+
+    def test_type_predicate(self):
+        ex = self._ex()
+        assert ex.execute(
+            "RETURN 1 IS :: INTEGER, 'a' IS :: STRING, 1.5 IS NOT :: STRING"
+        ).rows == [[True, True, True]]
+        assert ex.execute(
+            "MATCH (n:P) RETURN n IS :: NODE").rows == [[True]]
