@@ -646,14 +646,24 @@ static bool refcheck_128(const unsigned short* db, const unsigned short* qs,
 // ~8.8 TB/s). Template BMT in {64, 96}; SWZY enables the row-XOR LDS
 // swizzle (kills the 16-way af/bf bank conflict).
 // ---------------------------------------------------------------------------
+// monotonic order-preserving f32<->u32 (works for negatives)
+__device__ __forceinline__ unsigned int f32_ord(float f) {
+  unsigned int u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+__device__ __forceinline__ float ord_f32(unsigned int u) {
+  return __uint_as_float((u & 0x80000000u) ? (u & 0x7FFFFFFFu) : ~u);
+}
+
 template <int BMT, int SWZY, int KCT = KC, int EPI = 1>
 __global__ __launch_bounds__(256, 3) void k_knn96(
     const unsigned short* __restrict__ db, const unsigned short* __restrict__ qs,
     long long n_panels, int d, float* __restrict__ cand_score,
-    int* __restrict__ cand_idx) {
+    int* __restrict__ cand_idx, unsigned int* __restrict__ kth_global = nullptr) {
   constexpr int MW = BMT / 16;        // m-fragments per wave (4 or 6)
   constexpr int ACH = BMT * BK * 2;   // A tile bytes
-  __shared__ __align__(16) char smem[ACH + BN * BK * 2];
+  // +8 KB for EPI5 candidate stacks (packed 4 B entries)
+  __shared__ __align__(16) char smem[ACH + BN * BK * 2 + 8192];
   unsigned short* sA = (unsigned short*)smem;
   unsigned short* sB = (unsigned short*)(smem + ACH);
   float* sS = (float*)smem;           // epilogue alias [32][ESTR? use 260]
@@ -665,6 +675,15 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
 
   float tv[KCT];
   int ti[KCT];
+  // EPI==5: per-thread candidate stack in LDS. The insert chain is ~40
+  // VALU and wave divergence runs it for nearly every scanned value;
+  // appending costs ~3 ops and the chain only runs at flush. Entries are
+  // packed u32 = (score bits & ~0x7F) | local_row (panel rows < 128 fit
+  // 7 bits; the 2^-17 relative score truncation is far below bf16 input
+  // noise and is applied consistently on both sides of the merge).
+  constexpr int STK = 8;
+  unsigned int* st = nullptr;
+  int st_n = 0;
 #pragma unroll
   for (int i = 0; i < KCT; ++i) { tv[i] = -1e30f; ti[i] = -1; }
   // EPI==4: 4-deep shift-register candidate buffer. The full insertion
@@ -694,9 +713,17 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
   };
 
   const long long d2 = (long long)d * 2;
+  // EPI==6: cross-block threshold. Any thread's local KCT-th best is a
+  // SAFE prune bound for its query column (10 values >= x imply the true
+  // global 10th >= x), so blocks scheduled later skip the insert chain
+  // for almost every value.
+  float thr = -1e30f;
 
   for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
     const long long prow = panel * BMT;
+    if (EPI == 6) {
+      thr = fmaxf(tv[KCT - 1], ord_f32(kth_global[tid]));
+    }
 
     float4v acc[MW][4];
 #pragma unroll
@@ -759,6 +786,8 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
     // LDS ops than the row-major b32 layout (epilogue measured at 36%
     // of kernel time at 552 TF).
     float* sT = (float*)smem;  // [256][36]
+    if (EPI == 5 && st == nullptr)
+      st = (unsigned int*)(smem + ACH + BN * BK * 2) + tid * STK;
 #pragma unroll
     for (int h = 0; h < BMT / 32; ++h) {
       if (EPI >= 2) {  // transposed writes for EPI 2/3/4
@@ -791,6 +820,59 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
       if (EPI == 3) {
         // diagnostic: keep one read per thread so writes aren't dead
         tv[KCT - 1] = fmaxf(tv[KCT - 1] - 1e-30f, sT[tid * 36]);
+      } else if (EPI == 6) {
+#pragma unroll
+        for (int j = 0; j < 32; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            float s = v[e];
+            if (s > thr) {
+              float cs = s;
+              int ci = (int)(grow0 + j + e);
+#pragma unroll
+              for (int i = 0; i < KCT; ++i) {
+                bool ins = cs > tv[i];
+                float ts = tv[i]; int tj = ti[i];
+                tv[i] = ins ? cs : tv[i];
+                ti[i] = ins ? ci : ti[i];
+                cs = ins ? ts : cs; ci = ins ? tj : ci;
+              }
+              thr = fmaxf(thr, tv[KCT - 1]);
+            }
+          }
+        }
+      } else if (EPI == 5) {
+        const int lrow0 = h * 32;
+#pragma unroll
+        for (int j = 0; j < 32; j += 4) {
+          float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            if (v[e] > tv[KCT - 1]) {
+              st[st_n++] = (__float_as_uint(v[e]) & ~0x7Fu) |
+                           (unsigned)(lrow0 + j + e);
+            }
+          }
+          if (st_n > STK - 4) {
+            for (int u = 0; u < st_n; ++u) {
+              unsigned int ent = st[u];
+              float cs = __uint_as_float(ent & ~0x7Fu);
+              if (cs > tv[KCT - 1]) {
+                int ci = (int)(prow + (ent & 0x7Fu));
+#pragma unroll
+                for (int i = 0; i < KCT; ++i) {
+                  bool ins = cs > tv[i];
+                  float ts = tv[i]; int tj = ti[i];
+                  tv[i] = ins ? cs : tv[i];
+                  ti[i] = ins ? ci : ti[i];
+                  cs = ins ? ts : cs; ci = ins ? tj : ci;
+                }
+              }
+            }
+            st_n = 0;
+          }
+        }
       } else if (EPI == 4) {
 #pragma unroll
         for (int j = 0; j < 32; j += 4) {
@@ -847,6 +929,28 @@ __global__ __launch_bounds__(256, 3) void k_knn96(
         }
       }
       __syncthreads();
+    }
+    if (EPI == 6) {
+      // publish this thread's local kth as a global lower bound
+      atomicMax(&kth_global[tid], f32_ord(tv[KCT - 1]));
+    }
+    if (EPI == 5 && st_n > 0) {
+      for (int u = 0; u < st_n; ++u) {
+        unsigned int ent = st[u];
+        float cs = __uint_as_float(ent & ~0x7Fu);
+        if (cs > tv[KCT - 1]) {
+          int ci = (int)(prow + (ent & 0x7Fu));
+#pragma unroll
+          for (int i = 0; i < KCT; ++i) {
+            bool ins = cs > tv[i];
+            float ts = tv[i]; int tj = ti[i];
+            tv[i] = ins ? cs : tv[i];
+            ti[i] = ins ? ci : ti[i];
+            cs = ins ? ts : cs; ci = ins ? tj : ci;
+          }
+        }
+      }
+      st_n = 0;
     }
   }
 
@@ -1115,6 +1219,156 @@ static bool refcheck_96b(const unsigned short* db, const unsigned short* qs,
   return bad == 0;
 }
 
+
+
+static float run_96g(const unsigned short* db, const unsigned short* qs,
+                     long long n, int d, float* cs, int* ci, int iters,
+                     unsigned int* kth) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipMemset(kth, 0, 256 * 4);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 6>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci, kth);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i) {
+    hipMemsetAsync(kth, 0, 256 * 4, 0);
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 6>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci, kth);
+  }
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+static bool refcheck_96g(const unsigned short* db, const unsigned short* qs,
+                         int d, float* cs, int* ci, unsigned int* kth) {
+  const long long nn = 192 * 22;
+  long long panels = nn / 96;
+  int grid = (int)panels;
+  long long slots = (long long)grid;
+  hipMemset(kth, 0, 256 * 4);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 6>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci, kth);
+  if (hipDeviceSynchronize() != hipSuccess) {
+    printf("k96g refcheck launch err\n");
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  // stricter: verify FULL top-5 (threshold pruning must not lose ranks)
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best[5] = {-1e30f, -1e30f, -1e30f, -1e30f, -1e30f};
+    long long bidx[5] = {-1, -1, -1, -1, -1};
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      for (int t = 0; t < 5; ++t)
+        if (a > best[t]) {
+          for (int u = 4; u > t; --u) { best[u] = best[u-1]; bidx[u] = bidx[u-1]; }
+          best[t] = a; bidx[t] = r; break;
+        }
+    }
+    float gb[5] = {-1e30f, -1e30f, -1e30f, -1e30f, -1e30f};
+    int gi[5] = {-1, -1, -1, -1, -1};
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        int ii = hci[(s * BN + qi) * KC + k];
+        for (int t = 0; t < 5; ++t)
+          if (v > gb[t]) {
+            for (int u = 4; u > t; --u) { gb[u] = gb[u-1]; gi[u] = gi[u-1]; }
+            gb[t] = v; gi[t] = ii; break;
+          }
+      }
+    for (int t = 0; t < 5; ++t)
+      if (gi[t] != bidx[t]) {
+        if (bad < 3) printf("k96g MISMATCH q=%d rank%d: got %d want %lld\n",
+                            qi, t, gi[t], bidx[t]);
+        bad++;
+      }
+  }
+  printf("k96g refcheck(top5): %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
+static float run_96s(const unsigned short* db, const unsigned short* qs,
+                     long long n, int d, float* cs, int* ci, int iters) {
+  long long panels = n / 96;
+  int grid = (int)std::min<long long>(panels, 8192);
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 5>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  hipDeviceSynchronize();
+  hipEvent_t t0, t1; hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((k_knn96<96, 2, KC, 5>), dim3(grid), dim3(256), 0, 0,
+                       db, qs, panels, d, cs, ci);
+  hipEventRecord(t1); hipEventSynchronize(t1);
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return ms / iters;
+}
+static bool refcheck_96s(const unsigned short* db, const unsigned short* qs,
+                         int d, float* cs, int* ci) {
+  const long long nn = 192 * 22;
+  long long panels = nn / 96;
+  int grid = (int)panels;
+  long long slots = (long long)grid;
+  hipLaunchKernelGGL((k_knn96<96, 2, KC, 5>), dim3(grid), dim3(256), 0, 0,
+                     db, qs, panels, d, cs, ci);
+  if (hipDeviceSynchronize() != hipSuccess) {
+    printf("k96s refcheck launch err\n");
+    return false;
+  }
+  unsigned short* hdb = (unsigned short*)malloc(nn * d * 2);
+  unsigned short* hq = (unsigned short*)malloc((long long)BN * d * 2);
+  float* hcs = (float*)malloc(slots * BN * KC * 4);
+  int* hci = (int*)malloc(slots * BN * KC * 4);
+  hipMemcpy(hdb, db, nn * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hq, qs, (long long)BN * d * 2, hipMemcpyDeviceToHost);
+  hipMemcpy(hcs, cs, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  hipMemcpy(hci, ci, slots * BN * KC * 4, hipMemcpyDeviceToHost);
+  auto b2f = [](unsigned short u) {
+    union { unsigned i; float f; } v; v.i = (unsigned)u << 16; return v.f;
+  };
+  int bad = 0;
+  for (int qi = 0; qi < BN; qi += 13) {
+    float best = -1e30f; long long bi = -1;
+    for (long long r = 0; r < nn; ++r) {
+      float a = 0;
+      for (int k = 0; k < d; ++k)
+        a += b2f(hdb[r * d + k]) * b2f(hq[(long long)qi * d + k]);
+      if (a > best) { best = a; bi = r; }
+    }
+    float gbest = -1e30f; int gi = -1;
+    for (long long s = 0; s < slots; ++s)
+      for (int k = 0; k < KC; ++k) {
+        float v = hcs[(s * BN + qi) * KC + k];
+        if (v > gbest) { gbest = v; gi = hci[(s * BN + qi) * KC + k]; }
+      }
+    if (gi != bi || fabsf(gbest - best) > 1e-2f * fmaxf(fabsf(best), 1.f)) {
+      if (bad < 3) printf("k96s MISMATCH q=%d: got (%d, %f) want (%lld, %f)\n",
+                          qi, gi, gbest, bi, best);
+      bad++;
+    }
+  }
+  printf("k96s refcheck: %s (%d bad)\n", bad ? "FAIL" : "PASS", bad);
+  free(hdb); free(hq); free(hcs); free(hci);
+  return bad == 0;
+}
+
 static float run_96ne(const unsigned short* db, const unsigned short* qs,
                       long long n, int d, float* cs, int* ci, int iters) {
   long long panels = n / 96;
@@ -1198,6 +1452,21 @@ int main() {
         ms = run_96t(db, qs, n96, d, cs, ci, 10);
         printf("K96t %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (transposed epi)\n",
                "96+swz+tepi", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+      }
+      {
+        unsigned int* kth;
+        hipMalloc(&kth, 256 * 4);
+        if (refcheck_96g(db, qs, d, cs, ci, kth)) {
+          ms = run_96g(db, qs, n96, d, cs, ci, 10, kth);
+          printf("K96g %-17s %7.3f ms  %6.0f TF  %5.2f TB/s (global threshold)\n",
+                 "96+swz+gthr", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
+        }
+        hipFree(kth);
+      }
+      if (refcheck_96s(db, qs, d, cs, ci)) {
+        ms = run_96s(db, qs, n96, d, cs, ci, 10);
+        printf("K96v5 %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (LDS-stack epi)\n",
+               "96+swz+stack", ms, fl96 / ms / 1e9, by96 / ms / 1e9);
       }
       ms = run_96ne(db, qs, n96, d, cs, ci, 10);
       printf("K96ne %-16s %7.3f ms  %6.0f TF  %5.2f TB/s (no epilogue diag)\n",
