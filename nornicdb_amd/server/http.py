@@ -13,6 +13,7 @@ import time
 from typing import Any, Dict, List, Optional
 
 from fastapi import Depends, FastAPI, HTTPException, Request, Response
+from starlette.responses import StreamingResponse
 from pydantic import BaseModel
 
 from ..cypher import CypherRuntimeError, CypherSyntaxError
@@ -334,11 +335,47 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     # ---- GraphQL (reference pkg/graphql) ----
     from .graphql import GraphQLExecutor
 
+    _gql_cache = {}
+
+    def _gql_for(db):
+        d = mgr.get(db)
+        if d.name not in _gql_cache:
+            _gql_cache[d.name] = GraphQLExecutor(d)
+        return _gql_cache[d.name]
+
     @app.post("/graphql")
     async def graphql_endpoint(request: Request, db: str = None):
         body = await request.json()
-        gql = GraphQLExecutor(mgr.get(db))
-        return gql.execute(body.get("query", ""), body.get("variables"))
+        return _gql_for(db).execute(body.get("query", ""),
+                                    body.get("variables"))
+
+    @app.get("/graphql/stream")
+    async def graphql_stream(request: Request, db: str = None,
+                             labels: str = None):
+        """Subscription surface (reference resolvers/subscription_impl.go)
+        as SSE: nodeCreated/nodeUpdated/nodeDeleted/relationship* events."""
+        import asyncio
+        import queue as _q
+
+        broker = _gql_for(db).broker
+        sub = broker.subscribe(labels.split(",") if labels else None)
+
+        async def gen():
+            try:
+                while True:
+                    if await request.is_disconnected():
+                        break
+                    try:
+                        ev = sub.get_nowait()
+                    except _q.Empty:
+                        await asyncio.sleep(0.05)
+                        continue
+                    yield (f"event: {ev['event']}\n"
+                           f"data: {json.dumps(ev['data'], default=str)}\n\n")
+            finally:
+                broker.unsubscribe(sub)
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
 
     # ---- Qdrant-compatible REST (reference pkg/qdrantgrpc) ----
     from .qdrant import QdrantRegistry, qdrant_router

@@ -89,3 +89,88 @@ class TestGraphQL:
     def test_error_reported(self, client):
         r = self._gql(client, '{ nosuch }')
         assert "errors" in r
+
+
+class TestGraphQLExtended:
+    """Full resolver surface (reference pkg/graphql/resolvers/*_impl.go)."""
+
+    @pytest.fixture
+    def gq(self):
+        from nornicdb_amd.db import NornicDB
+        from nornicdb_amd.server.graphql import GraphQLExecutor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        db = NornicDB(MemoryEngine(), auto_embed=False)
+        return GraphQLExecutor(db)
+
+    def test_input_objects_and_variables(self, gq):
+        r = gq.execute(
+            'mutation($p: JSON) { createNode(input: {labels: ["P"], '
+            'properties: $p}) { id properties } }', {"p": {"name": "ann"}})
+        assert r["data"]["createNode"]["properties"] == {"name": "ann"}
+
+    def test_crud_and_counts(self, gq):
+        a = gq.execute('mutation { createNode(input: {labels: ["P"], '
+                       'properties: {name: "a"}}) { id } }')["data"]["createNode"]["id"]
+        b = gq.execute('mutation { createNode(input: {labels: ["P"], '
+                       'properties: {name: "b"}}) { id } }')["data"]["createNode"]["id"]
+        rel = gq.execute('mutation { createRelationship(input: {from: "%s", '
+                         'to: "%s", type: "K"}) { id } }' % (a, b))
+        assert "errors" not in rel
+        r = gq.execute('{ nodeCount relationshipCount '
+                       'nodesByLabel(label: "P") { id } }')["data"]
+        assert r["nodeCount"] == 2 and r["relationshipCount"] == 1
+        assert len(r["nodesByLabel"]) == 2
+        r = gq.execute('{ relationshipsBetween(from: "%s", to: "%s") '
+                       '{ type } }' % (a, b))["data"]
+        assert r["relationshipsBetween"][0]["type"] == "K"
+        upd = gq.execute('mutation { updateNode(input: {id: "%s", '
+                         'properties: {age: 3}}) { properties } }' % a)
+        assert upd["data"]["updateNode"]["properties"]["age"] == 3
+        assert gq.execute('mutation { deleteRelationship(id: "%s") }'
+                          % rel["data"]["createRelationship"]["id"]
+                          )["data"]["deleteRelationship"] is True
+
+    def test_bulk_and_merge(self, gq):
+        r = gq.execute('mutation { bulkCreateNodes(input: {nodes: ['
+                       '{labels: ["B"], properties: {i: 1}}, '
+                       '{labels: ["B"], properties: {i: 2}}]}) { count } }')
+        assert r["data"]["bulkCreateNodes"]["count"] == 2
+        m1 = gq.execute('mutation { mergeNode(input: {labels: ["B"], '
+                        'properties: {i: 1}, mergeKey: "i"}) { id } }')
+        m2 = gq.execute('mutation { mergeNode(input: {labels: ["B"], '
+                        'properties: {i: 1}, mergeKey: "i"}) { id } }')
+        assert m1["data"]["mergeNode"]["id"] == m2["data"]["mergeNode"]["id"]
+        assert gq.execute('{ nodeCount(label: "B") }')["data"]["nodeCount"] == 2
+
+    def test_neighborhood_and_nested(self, gq):
+        a = gq.execute('mutation { createNode(input: {labels: ["N"]}) '
+                       '{ id } }')["data"]["createNode"]["id"]
+        b = gq.execute('mutation { createNode(input: {labels: ["N"]}) '
+                       '{ id } }')["data"]["createNode"]["id"]
+        gq.execute('mutation { createRelationship(input: {from: "%s", '
+                   'to: "%s", type: "L"}) { id } }' % (a, b))
+        r = gq.execute('{ node(id: "%s") { id outgoing { type endNode } '
+                       'neighbors { id } } }' % a)["data"]["node"]
+        assert r["outgoing"][0]["endNode"] == b
+        assert r["neighbors"][0]["id"] == b
+        nb = gq.execute('{ neighborhood(id: "%s", depth: 1) '
+                        '{ nodes { id } } }' % a)["data"]["neighborhood"]
+        assert {n["id"] for n in nb["nodes"]} == {a, b}
+
+    def test_event_broker(self, gq):
+        sub = gq.broker.subscribe(["EV"])
+        gq.execute('mutation { createNode(input: {labels: ["EV"], '
+                   'properties: {x: 1}}) { id } }')
+        ev = sub.get_nowait()
+        assert ev["event"] == "nodeCreated"
+        assert ev["data"]["properties"] == {"x": 1}
+        # label filter: other labels don't reach this subscriber
+        gq.execute('mutation { createNode(input: {labels: ["OTHER"]}) { id } }')
+        import queue as q
+        with pytest.raises(q.Empty):
+            sub.get_nowait()
+
+    def test_clear_all(self, gq):
+        gq.execute('mutation { createNode(input: {labels: ["X"]}) { id } }')
+        assert gq.execute('mutation { clearAll }')["data"]["clearAll"] is True
+        assert gq.execute('{ nodeCount }')["data"]["nodeCount"] == 0
