@@ -192,7 +192,150 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
                             "vector": c.vectors.get(pid)})
         return ok(out)
 
+
+    # ---- payload ops (reference qdrantgrpc SetPayload/OverwritePayload/
+    # DeletePayload/ClearPayload) ----
+    def _coll(name):
+        try:
+            return reg.get(name)
+        except KeyError:
+            raise HTTPException(404, "not found")
+
+    def _sel_ids(c, body):
+        if "points" in body:
+            return [str(p) for p in body["points"]]
+        flt = body.get("filter")
+        if flt:
+            return [pid for pid in c.payloads if _matches(c.payloads[pid], flt)]
+        return list(c.payloads)
+
+    @r.post("/collections/{name}/points/payload")
+    def set_payload(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        for pid in _sel_ids(c, body):
+            if pid in c.payloads:
+                c.payloads[pid].update(body.get("payload", {}))
+        return ok({"operation_id": 0, "status": "completed"})
+
+    @r.put("/collections/{name}/points/payload")
+    def overwrite_payload(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        for pid in _sel_ids(c, body):
+            if pid in c.payloads:
+                c.payloads[pid] = dict(body.get("payload", {}))
+        return ok({"operation_id": 0, "status": "completed"})
+
+    @r.post("/collections/{name}/points/payload/delete")
+    def delete_payload(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        keys = body.get("keys", [])
+        for pid in _sel_ids(c, body):
+            for k in keys:
+                c.payloads.get(pid, {}).pop(k, None)
+        return ok({"operation_id": 0, "status": "completed"})
+
+    @r.post("/collections/{name}/points/payload/clear")
+    def clear_payload(name: str, body: Dict[str, Any] = None):
+        c = _coll(name)
+        for pid in _sel_ids(c, body or {}):
+            c.payloads[pid] = {}
+        return ok({"operation_id": 0, "status": "completed"})
+
+    # ---- count / exists / query (qdrant >=1.10 universal query) ----
+    @r.post("/collections/{name}/points/count")
+    def count_points(name: str, body: Dict[str, Any] = None):
+        c = _coll(name)
+        flt = (body or {}).get("filter")
+        if flt:
+            n = sum(1 for pid in c.payloads
+                    if _matches(c.payloads[pid], flt))
+        else:
+            n = len(c.payloads)
+        return ok({"count": n})
+
+    @r.get("/collections/{name}/exists")
+    def collection_exists(name: str):
+        return ok({"exists": name in reg.collections})
+
+    @r.post("/collections/{name}/points/query")
+    def query_points(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        q = body.get("query")
+        limit = int(body.get("limit", 10))
+        if isinstance(q, dict) and "nearest" in q:
+            q = q["nearest"]
+        if isinstance(q, list):            # vector query
+            hits = c.index.search(q, limit * 4)
+        elif q is not None and str(q) in c.vectors:  # by point id
+            hits = c.index.search(c.vectors[str(q)], limit * 4 + 1)
+            hits = [(p, s) for p, s in hits if p != str(q)]
+        else:
+            hits = [(pid, 1.0) for pid in sorted(c.payloads)][:limit * 4]
+        flt = body.get("filter")
+        out = []
+        for pid, score in hits:
+            if flt and not _matches(c.payloads.get(pid, {}), flt):
+                continue
+            item = {"id": _maybe_int(pid), "score": score}
+            if body.get("with_payload", True):
+                item["payload"] = c.payloads.get(pid, {})
+            out.append(item)
+            if len(out) >= limit:
+                break
+        return ok({"points": out})
+
+    # ---- vector update/delete ----
+    @r.put("/collections/{name}/points/vectors")
+    def update_vectors(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        for p in body.get("points", []):
+            pid = str(p["id"])
+            vec = p.get("vector")
+            if vec is not None and pid in c.payloads:
+                c.vectors[pid] = list(vec)
+                c.index.remove(pid)
+                c.index.add_batch([pid], [vec])
+        return ok({"operation_id": 0, "status": "completed"})
+
+    @r.post("/collections/{name}/points/vectors/delete")
+    def delete_vectors(name: str, body: Dict[str, Any]):
+        c = _coll(name)
+        for pid in body.get("points", []):
+            c.index.remove(str(pid))
+        return ok({"operation_id": 0, "status": "completed"})
+
     return r
+
+
+def _matches(payload: Dict[str, Any], flt: Dict[str, Any]) -> bool:
+    """Qdrant filter subset: must / must_not / should with match/range."""
+    def cond_ok(cond):
+        key = cond.get("key")
+        if "match" in cond:
+            want = cond["match"].get("value", cond["match"].get("text"))
+            return payload.get(key) == want
+        if "range" in cond:
+            v = payload.get(key)
+            if not isinstance(v, (int, float)):
+                return False
+            rg = cond["range"]
+            return all([
+                v >= rg["gte"] if "gte" in rg else True,
+                v > rg["gt"] if "gt" in rg else True,
+                v <= rg["lte"] if "lte" in rg else True,
+                v < rg["lt"] if "lt" in rg else True])
+        return True
+
+    for cond in flt.get("must", []):
+        if not cond_ok(cond):
+            return False
+    for cond in flt.get("must_not", []):
+        if cond_ok(cond):
+            return False
+    should = flt.get("should", [])
+    if should and not any(cond_ok(c) for c in should):
+        return False
+    return True
 
 
 def _maybe_int(pid):

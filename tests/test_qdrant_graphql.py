@@ -174,3 +174,73 @@ class TestGraphQLExtended:
         gq.execute('mutation { createNode(input: {labels: ["X"]}) { id } }')
         assert gq.execute('mutation { clearAll }')["data"]["clearAll"] is True
         assert gq.execute('{ nodeCount }')["data"]["nodeCount"] == 0
+
+
+class TestQdrantExtended:
+    """Payload ops, count, filters, universal query (reference
+    pkg/qdrantgrpc surface)."""
+
+    @pytest.fixture
+    def qc(self):
+        from starlette.testclient import TestClient
+
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.server.http import create_app
+        from nornicdb_amd.storage.memory import MemoryEngine
+        app = create_app(DatabaseManager(MemoryEngine()))
+        c = TestClient(app)
+        c.put("/collections/t", json={"vectors": {"size": 4,
+                                                  "distance": "Cosine"}})
+        c.put("/collections/t/points", json={"points": [
+            {"id": 1, "vector": [1, 0, 0, 0], "payload": {"city": "oslo", "n": 1}},
+            {"id": 2, "vector": [0, 1, 0, 0], "payload": {"city": "bergen", "n": 2}},
+            {"id": 3, "vector": [0, 0, 1, 0], "payload": {"city": "oslo", "n": 3}},
+        ]})
+        return c
+
+    def test_count_and_exists(self, qc):
+        assert qc.post("/collections/t/points/count", json={}
+                       ).json()["result"]["count"] == 3
+        assert qc.post("/collections/t/points/count", json={
+            "filter": {"must": [{"key": "city", "match": {"value": "oslo"}}]}
+        }).json()["result"]["count"] == 2
+        assert qc.get("/collections/t/exists").json()["result"]["exists"]
+        assert not qc.get("/collections/nope/exists").json()["result"]["exists"]
+
+    def test_payload_ops(self, qc):
+        qc.post("/collections/t/points/payload", json={
+            "points": [1], "payload": {"extra": True}})
+        pts = qc.post("/collections/t/points", json={"ids": [1]}
+                      ).json()["result"]
+        assert pts[0]["payload"]["extra"] is True
+        qc.put("/collections/t/points/payload", json={
+            "points": [1], "payload": {"only": 1}})
+        pts = qc.post("/collections/t/points", json={"ids": [1]}
+                      ).json()["result"]
+        assert pts[0]["payload"] == {"only": 1}
+        qc.post("/collections/t/points/payload/delete", json={
+            "points": [2], "keys": ["n"]})
+        pts = qc.post("/collections/t/points", json={"ids": [2]}
+                      ).json()["result"]
+        assert "n" not in pts[0]["payload"]
+        qc.post("/collections/t/points/payload/clear", json={"points": [3]})
+        pts = qc.post("/collections/t/points", json={"ids": [3]}
+                      ).json()["result"]
+        assert pts[0]["payload"] == {}
+
+    def test_query_with_filter(self, qc):
+        r = qc.post("/collections/t/points/query", json={
+            "query": [1, 0, 0, 0], "limit": 2,
+            "filter": {"must": [{"key": "city", "match": {"value": "oslo"}}]}
+        }).json()["result"]["points"]
+        assert r[0]["id"] == 1
+        assert all(p["payload"]["city"] == "oslo" for p in r)
+
+    def test_query_by_id_and_range(self, qc):
+        r = qc.post("/collections/t/points/query", json={
+            "query": 1, "limit": 2}).json()["result"]["points"]
+        assert all(p["id"] != 1 for p in r)  # excludes the anchor
+        r = qc.post("/collections/t/points/count", json={
+            "filter": {"must": [{"key": "n", "range": {"gte": 2}}]}
+        }).json()["result"]["count"]
+        assert r == 2
