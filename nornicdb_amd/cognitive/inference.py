@@ -148,3 +148,56 @@ class InferenceEngine:
                 self.engine.update_edge(e)
                 stats["decayed"] += 1
         return stats
+
+
+class HeimdallQC:
+    """LLM quality-control of auto-inferred links (reference
+    pkg/inference/heimdall_qc.go): before an auto-link is committed, the
+    SLM is asked whether the connection is plausible; non-affirmative
+    answers veto the edge. Degrades to accept-all when no manager is
+    available (same as the reference without a loaded model)."""
+
+    def __init__(self, manager=None, threshold: float = 0.5):
+        self.manager = manager
+        self.threshold = threshold
+        self.stats = {"checked": 0, "vetoed": 0}
+
+    def check(self, a_node, b_node, reason: str) -> bool:
+        self.stats["checked"] += 1
+        if self.manager is None:
+            return True
+        try:
+            prompt = (f"Should '{a_node.properties.get('name', a_node.id)}' "
+                      f"link to '{b_node.properties.get('name', b_node.id)}' "
+                      f"because {reason}? Answer yes or no.")
+            out = self.manager.generate(prompt, max_tokens=4)
+            ok = "no" not in out.lower().split()
+        except Exception:
+            ok = True
+        if not ok:
+            self.stats["vetoed"] += 1
+        return ok
+
+
+class ClusterIntegration:
+    """k-means cluster assignments as an inference signal (reference
+    pkg/inference/cluster_integration.go): nodes sharing a cluster get a
+    similarity-floor boost; re-clustering is triggered by the embed queue
+    (db.EmbedQueue._pending_recluster)."""
+
+    def __init__(self, search_service, boost: float = 0.1):
+        self.search = search_service
+        self.boost = boost
+        self._assign = {}
+
+    def recluster(self, k: int = None):
+        from ..search.kmeans import kmeans_assign_index
+        self._assign = kmeans_assign_index(self.search.emb, k=k)
+        return len(set(self._assign.values()))
+
+    def same_cluster(self, a: str, b: str) -> bool:
+        ca, cb = self._assign.get(a), self._assign.get(b)
+        return ca is not None and ca == cb
+
+    def boost_for(self, a: str, b: str) -> float:
+        return self.boost if self.same_cluster(a, b) else 0.0

@@ -45,6 +45,7 @@ class Result:
     columns: List[str]
     rows: List[List[Any]]
     stats: Dict[str, int] = field(default_factory=dict)
+    profile: Optional[List[Dict[str, Any]]] = None
 
     def to_dicts(self):
         return [dict(zip(self.columns, r)) for r in self.rows]
@@ -78,6 +79,7 @@ class Executor:
         self._plan_cache: Dict[str, A.Query] = {}
         self.stats: Dict[str, int] = {}
         self.schema = schema            # storage.SchemaManager (DDL target)
+        self._profile_log = None
         self.database_lister = None     # set by DatabaseManager for SHOW DATABASES
         self.current_database = "neo4j"
 
@@ -99,6 +101,8 @@ class Executor:
                       "properties_set": 0, "labels_added": 0}
         if q.explain:
             return Result(["plan"], [[self._explain(q)]], dict(self.stats))
+        if q.profile:
+            return self._profile(q, params)
         if len(q.clauses) == 1 and isinstance(q.clauses[0], A.SchemaCommand):
             res = self._exec_schema(q.clauses[0], params)
             res.stats = dict(self.stats)
@@ -219,6 +223,18 @@ class Executor:
         return " -> ".join(lines)
 
     # ------------------------------------------------------------- pipeline
+    def _profile(self, q: A.Query, params) -> Result:
+        """PROFILE: execute and attach per-clause wall time + row counts
+        (clause-granular: the executor pipeline is clause-at-a-time)."""
+        self._profile_log = []
+        try:
+            res = self._run_query(q, params)
+        finally:
+            plan, self._profile_log = self._profile_log, None
+        res.stats = dict(self.stats)
+        res.profile = plan
+        return res
+
     def _run_query(self, q: A.Query, params) -> Result:
         cols, rows = self._run_clauses(q.clauses, params)
         if q.union:
@@ -246,7 +262,11 @@ class Executor:
         out_rows: List[List[Any]] = []
         i = 0
         n = len(clauses)
+        prof = getattr(self, "_profile_log", None)
         while i < n:
+            if prof is not None:
+                import time as _t
+                _pt0 = _t.perf_counter()
             c = clauses[i]
             if isinstance(c, A.MatchClause):
                 rows = self._exec_match(c, rows, params)
@@ -277,6 +297,12 @@ class Executor:
                 rows = self._exec_foreach(c, rows, params)
             else:
                 raise CypherRuntimeError(f"unsupported clause {type(c).__name__}")
+            if prof is not None:
+                prof.append({
+                    "operator": type(c).__name__,
+                    "rows": len(out_rows) if isinstance(c, A.ReturnClause)
+                            and not isinstance(c, A.WithClause) else len(rows),
+                    "time_ms": round((_t.perf_counter() - _pt0) * 1e3, 4)})
             i += 1
         return out_cols, out_rows
 

@@ -174,3 +174,21 @@ class ClusterIndex:
             for t in top:
                 out.extend(self.members[t])
             return out
+
+
+def kmeans_assign_index(emb_index, k: int = None) -> dict:
+    """Cluster an EmbeddingIndex; returns {id: cluster}. k defaults to
+    ~sqrt(n/2) (the reference's heuristic)."""
+    import math
+
+    ids = [i for i in emb_index._ids
+           if emb_index._id2slot.get(i) is not None
+           and emb_index._id2slot[i] not in emb_index._dead]
+    if len(ids) < 2:
+        return {i: 0 for i in ids}
+    slots = [emb_index._id2slot[i] for i in ids]
+    x = emb_index._buf[slots].float()
+    kk = k or max(1, int(math.sqrt(len(ids) / 2)))
+    kk = min(kk, len(ids))
+    centroids, assign = kmeans(x, kk)
+    return {i: int(c) for i, c in zip(ids, assign.tolist())}

@@ -235,3 +235,53 @@ class TestKalmanDecayAB:
         assert roughness(kal_scores) < roughness(raw_scores)
         # still tracks the same level
         assert abs(sum(kal_scores) / 40 - sum(raw_scores) / 40) < 0.15
+
+
+class TestInferenceQCAndClusters:
+    """HeimdallQC veto + ClusterIntegration (reference
+    pkg/inference/heimdall_qc.go + cluster_integration.go)."""
+
+    def test_qc_accepts_without_manager(self):
+        from nornicdb_amd.cognitive.inference import HeimdallQC
+        from nornicdb_amd.storage.types import Node
+        qc = HeimdallQC(manager=None)
+        a = Node(id="a", labels=[], properties={"name": "x"})
+        b = Node(id="b", labels=[], properties={"name": "y"})
+        assert qc.check(a, b, "similar embeddings") is True
+        assert qc.stats["checked"] == 1
+
+    def test_qc_veto_with_stub_manager(self):
+        from nornicdb_amd.cognitive.inference import HeimdallQC
+        from nornicdb_amd.storage.types import Node
+
+        class StubMgr:
+            def generate(self, prompt, max_tokens=4):
+                return "no , unrelated"
+        qc = HeimdallQC(manager=StubMgr())
+        a = Node(id="a", labels=[], properties={})
+        b = Node(id="b", labels=[], properties={})
+        assert qc.check(a, b, "co-access") is False
+        assert qc.stats["vetoed"] == 1
+
+    def test_cluster_integration(self):
+        import numpy as np
+
+        from nornicdb_amd.cognitive.inference import ClusterIntegration
+        from nornicdb_amd.search.embedding_index import EmbeddingIndex
+
+        class S:
+            pass
+        s = S()
+        s.emb = EmbeddingIndex(dims=4, device="cpu")
+        rng = np.random.default_rng(0)
+        for i in range(10):
+            s.emb.add(f"a{i}", [1, 0, 0, 0] + rng.normal(0, 0.01, 4))
+        for i in range(10):
+            s.emb.add(f"b{i}", [0, 1, 0, 0] + rng.normal(0, 0.01, 4))
+        ci = ClusterIntegration(s)
+        n = ci.recluster(k=2)
+        assert n == 2
+        assert ci.same_cluster("a0", "a5")
+        assert not ci.same_cluster("a0", "b0")
+        assert ci.boost_for("a0", "a1") > 0
+        assert ci.boost_for("a0", "b1") == 0.0

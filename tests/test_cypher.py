@@ -553,3 +553,23 @@ class TestCypher5Conformance:
         ).rows == [[True, True, True]]
         assert ex.execute(
             "MATCH (n:P) RETURN n IS :: NODE").rows == [[True]]
+
+
+class TestProfile:
+    def test_profile_attaches_plan(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        ex.execute("UNWIND range(1,10) AS i CREATE (:N {v: i})")
+        r = ex.execute("PROFILE MATCH (n:N) WHERE n.v > 5 RETURN count(n)")
+        assert r.rows == [[5]]
+        assert r.profile and r.profile[0]["operator"] == "MatchClause"
+        assert r.profile[0]["rows"] == 5
+        assert all("time_ms" in p for p in r.profile)
+
+    def test_explain_still_works(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        r = ex.execute("EXPLAIN MATCH (n) RETURN n")
+        assert r.columns == ["plan"]
