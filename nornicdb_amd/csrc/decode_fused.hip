@@ -180,6 +180,7 @@ __global__ void __launch_bounds__(DWG, 1) k_decode_step(DecodeArgs a) {
     // k pairs [NH*HD/2, NH*HD/2 + NKV*HD/2); v rows as HD/2-sized pairs
     const int q_pairs = q_rows / 2, k_pairs = kv_rows / 2,
               v_pairs = kv_rows / 2;
+#pragma unroll 2
     for (int p = gwave; p < q_pairs + k_pairs + v_pairs; p += n_gwaves) {
       if (p < q_pairs) {
         int head = p / (HD / 2), j = p % (HD / 2);
@@ -278,14 +279,23 @@ __global__ void __launch_bounds__(DWG, 1) k_decode_step(DecodeArgs a) {
     // stage attn into LDS (fp32) once per WG
     for (int i = tid; i < H; i += DWG) r_lds[i] = a.attn[i];
     __syncthreads();
-    for (int row = gwave; row < H; row += n_gwaves) {
-      float d = wave_dot_bf16(L.o_w + (long long)row * H, r_lds, H, lane);
-      if (lane == 0) a.x[row] += d;
+    for (int row = gwave; row < H; row += n_gwaves * 2) {
+      int row2 = row + n_gwaves;
+      if (row2 < H) {
+        float d2[2];
+        wave_dot2_bf16(L.o_w + (long long)row * H,
+                       L.o_w + (long long)row2 * H, r_lds, H, lane, d2);
+        if (lane == 0) { a.x[row] += d2[0]; a.x[row2] += d2[1]; }
+      } else {
+        float d = wave_dot_bf16(L.o_w + (long long)row * H, r_lds, H, lane);
+        if (lane == 0) a.x[row] += d;
+      }
     }
     grid.sync();
 
     // ---- S5: rmsnorm2 + gate/up + SiLU (2 rows x 2 mats in flight) ----
     wg_rmsnorm(a.x, L.ln2_w, r_lds, H, a.rms_eps, tid);
+#pragma unroll 2
     for (int row = gwave; row < a.inter; row += n_gwaves * 2) {
       int row2 = row + n_gwaves;
       float d[4];
