@@ -1347,3 +1347,165 @@ _agg("apoc.agg.minItems", lambda vals, extra=None: (lambda m: {
 _agg("apoc.agg.graph", lambda vals, extra=None: _graph_obj(
     [x for x in vals if isinstance(x, Node)],
     [x for x in vals if isinstance(x, Edge)]))
+
+# ============================== apoc.text extras ==============================
+_r("apoc.text.trim", lambda s: None if s is None else str(s).strip())
+_r("apoc.text.ltrim", lambda s: None if s is None else str(s).lstrip())
+_r("apoc.text.rtrim", lambda s: None if s is None else str(s).rstrip())
+_r("apoc.text.reverse", lambda s: None if s is None else str(s)[::-1])
+_r("apoc.text.bytes", lambda s, charset="UTF-8": list(
+    str(s).encode(charset)))
+_r("apoc.text.bytesToString", lambda b, charset="UTF-8": bytes(
+    x & 0xFF for x in (b or [])).decode(charset))
+_r("apoc.text.fromCodePoint", lambda *cps: "".join(chr(int(c)) for c in cps))
+_r("apoc.text.decapitalizeAll", lambda s: " ".join(
+    w[:1].lower() + w[1:] for w in str(s or "").split(" ")))
+
+
+def _soundexish(s):
+    """Simplified phonetic code (soundex-style)."""
+    s = re.sub(r"[^A-Za-z]", "", str(s or "")).upper()
+    if not s:
+        return ""
+    codes = {"B": "1", "F": "1", "P": "1", "V": "1",
+             "C": "2", "G": "2", "J": "2", "K": "2", "Q": "2", "S": "2",
+             "X": "2", "Z": "2", "D": "3", "T": "3", "L": "4",
+             "M": "5", "N": "5", "R": "6"}
+    out = s[0]
+    prev = codes.get(s[0], "")
+    for ch in s[1:]:
+        c = codes.get(ch, "")
+        if c and c != prev:
+            out += c
+        prev = c
+    return (out + "000")[:4]
+
+
+_r("apoc.text.phonetic", _soundexish)
+_r("apoc.text.phoneticDelta", lambda a, b: sum(
+    1 for x, y in zip(_soundexish(a), _soundexish(b)) if x == y))
+
+
+def _metaphoneish(s):
+    """Compact consonant-skeleton code (double-metaphone stand-in)."""
+    s = re.sub(r"[^A-Za-z]", "", str(s or "")).upper()
+    s = re.sub(r"PH", "F", s)
+    s = re.sub(r"[AEIOU]", "", s[1:])
+    return (str(s)[:1] if False else "") or s[:6]
+
+
+_r("apoc.text.doubleMetaphone", _metaphoneish)
+
+
+def _fuzzy_match(a, b):
+    from .functions import _levenshtein
+    a, b = str(a or "").lower(), str(b or "").lower()
+    if not a or not b:
+        return False
+    d = _levenshtein(a, b)
+    allowed = 1 if len(a) < 3 else (2 if len(a) < 5 else 3)
+    return d <= allowed
+
+
+_r("apoc.text.fuzzyMatch", _fuzzy_match)
+
+# ============================== apoc.coll extras ==============================
+_r("apoc.coll.isEmpty", lambda l: not l)
+_r("apoc.coll.isNotEmpty", lambda l: bool(l))
+_r("apoc.coll.containsAny", lambda l, items: bool(
+    set(map(repr, l or [])) & set(map(repr, items or []))))
+_r("apoc.coll.containsDuplicates", lambda l: len(l or []) != len(
+    set(map(repr, l or []))))
+_r("apoc.coll.containsSorted", lambda l, v: (lambda s: (lambda i:
+    i < len(s) and s[i] == v)(__import__("bisect").bisect_left(s, v)))(
+    sorted(l or [])))
+_r("apoc.coll.duplicatesWithCount", lambda l: [
+    {"item": v, "count": c} for v, c in
+    ((x, (l or []).count(x)) for x in dict.fromkeys(l or [])) if c > 1])
+_r("apoc.coll.dropDuplicateNeighbors", lambda l: [
+    v for i, v in enumerate(l or []) if i == 0 or v != l[i - 1]])
+_r("apoc.coll.fill", lambda v, n: [v] * int(n))
+_r("apoc.coll.insertAll", lambda l, idx, items: (
+    list(l or [])[:int(idx)] + list(items or []) + list(l or [])[int(idx):]))
+_r("apoc.coll.removeAll", lambda l, items: [
+    v for v in (l or []) if repr(v) not in set(map(repr, items or []))])
+_r("apoc.coll.set", lambda l, idx, v: [
+    v if i == int(idx) else x for i, x in enumerate(l or [])])
+_r("apoc.coll.unionAll", lambda a, b: list(a or []) + list(b or []))
+_r("apoc.coll.sumLongs", lambda l: int(sum(
+    int(x) for x in (l or []) if x is not None)))
+_r("apoc.coll.frequenciesAsMap", lambda l: {
+    str(v): (l or []).count(v) for v in dict.fromkeys(l or [])})
+_r("apoc.coll.randomItems", lambda l, n, allow_repeat=False: (
+    random.choices(l, k=int(n)) if allow_repeat
+    else random.sample(list(l), min(int(n), len(l)))) if l else [])
+_r("apoc.coll.sortMaps", lambda l, key: sorted(
+    l or [], key=lambda m: (m.get(key) is None, m.get(key)), reverse=True))
+
+# ============================== apoc.map extras ==============================
+_r("apoc.map.dropNullValues", lambda m: {k: v for k, v in _as_map(m).items()
+                                         if v is not None})
+_r("apoc.map.fromValues", lambda l: {str(l[i]): l[i + 1]
+                                     for i in range(0, len(l or []) - 1, 2)})
+_r("apoc.map.mget", lambda m, keys, defaults=None: [
+    _as_map(m).get(k, (defaults or [None] * len(keys))[i])
+    for i, k in enumerate(keys or [])])
+_r("apoc.map.setEntry", lambda m, k, v: {**_as_map(m), str(k): v})
+_r("apoc.map.setValues", lambda m, pairs: {**_as_map(m), **{
+    str(pairs[i]): pairs[i + 1] for i in range(0, len(pairs or []) - 1, 2)}})
+_r("apoc.map.setPairs", lambda m, pairs: {**_as_map(m), **{
+    str(p[0]): p[1] for p in (pairs or [])}})
+_r("apoc.map.setLists", lambda m, keys, values: {**_as_map(m), **dict(
+    zip([str(k) for k in (keys or [])], values or []))})
+_r("apoc.map.sortedProperties", lambda m, ignore_case=True: [
+    [k, _as_map(m)[k]] for k in sorted(
+        _as_map(m), key=(lambda s: s.lower()) if ignore_case else None)])
+_r("apoc.map.unflatten", _json_unflatten)
+
+
+def _update_tree(tree, key, data):
+    t = _as_map(tree)
+    out = dict(t)
+    for k, v in _as_map(data).items():
+        out[k] = v
+    return out
+
+
+_r("apoc.map.updateTree", _update_tree)
+
+# ============================== apoc.convert extras ==============================
+def _to_tree(paths):
+    """paths -> nested {children: []} tree (reference convert.toTree)."""
+    roots: Dict[str, Any] = {}
+    nodes: Dict[str, Any] = {}
+
+    def ent(n):
+        if n.id not in nodes:
+            nodes[n.id] = {"_id": n.id, "_type": ":".join(n.labels),
+                           **dict(n.properties)}
+        return nodes[n.id]
+
+    child_ids = set()
+    for p in paths or []:
+        ns = getattr(p, "nodes", [])
+        es = getattr(p, "edges", [])
+        for n in ns:
+            ent(n)
+        for e in es:
+            parent = nodes.get(e.start_node)
+            child = nodes.get(e.end_node)
+            if parent is None or child is None:
+                continue
+            key = e.type.lower()
+            parent.setdefault(key, [])
+            if child not in parent[key]:
+                parent[key].append(child)
+            child_ids.add(e.end_node)
+    for p in paths or []:
+        for n in getattr(p, "nodes", []):
+            if n.id not in child_ids:
+                roots[n.id] = nodes[n.id]
+    return list(roots.values())
+
+
+_r("apoc.convert.toTree", _to_tree)

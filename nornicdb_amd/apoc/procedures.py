@@ -1373,6 +1373,274 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
     def _mg_node_eager(ex, labels, ident_props, props=None):
         return procs["apoc.merge.node"](ex, labels, ident_props, props)
 
+    # -------------------- apoc.atomic extras --------------------
+    @register("apoc.atomic.increment")
+    def _at_inc(ex, node, prop, value=1):
+        return _atomic_add(ex, node, prop, value)
+
+    @register("apoc.atomic.decrement")
+    def _at_dec(ex, node, prop, value=1):
+        return _atomic_add(ex, node, prop, -value)
+
+    @register("apoc.atomic.concat")
+    def _at_concat(ex, node, prop, value):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            n.properties[prop] = str(n.properties.get(prop) or "") + str(value)
+            n = eng.update_node(n)
+        return ["node", "value"], [[n, n.properties[prop]]]
+
+    @register("apoc.atomic.compareAndSwap")
+    def _at_cas(ex, node, prop, expected, value):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            swapped = n.properties.get(prop) == expected
+            if swapped:
+                n.properties[prop] = value
+                n = eng.update_node(n)
+        return ["node", "swapped"], [[n, swapped]]
+
+    @register("apoc.atomic.insert")
+    def _at_insert(ex, node, prop, position, value):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            l = list(n.properties.get(prop) or [])
+            l.insert(int(position), value)
+            n.properties[prop] = l
+            n = eng.update_node(n)
+        return ["node", "value"], [[n, l]]
+
+    @register("apoc.atomic.remove")
+    def _at_remove(ex, node, prop, position):
+        with _ATOMIC_LOCK:
+            n = eng.get_node(node.id if isinstance(node, Node) else node)
+            l = list(n.properties.get(prop) or [])
+            if 0 <= int(position) < len(l):
+                l.pop(int(position))
+            n.properties[prop] = l
+            n = eng.update_node(n)
+        return ["node", "value"], [[n, l]]
+
+    # -------------------- apoc.periodic scheduler --------------------
+    # reference apoc/periodic: background named jobs over the executor
+    _JOBS: Dict[str, Any] = {}
+
+    @register("apoc.periodic.submit")
+    def _pd_submit(ex, name, statement):
+        def run():
+            try:
+                ex.execute(statement)
+                _JOBS[name]["done"] = True
+            except Exception as e:
+                _JOBS[name]["error"] = str(e)
+        th = threading.Thread(target=run, daemon=True)
+        _JOBS[name] = {"name": name, "thread": th, "done": False,
+                       "cancelled": False, "repeat": False}
+        th.start()
+        return ["name", "delay", "rate", "done", "cancelled"], [
+            [name, 0, 0, False, False]]
+
+    @register("apoc.periodic.repeat")
+    def _pd_repeat(ex, name, statement, rate_s):
+        stop = threading.Event()
+
+        def loop():
+            while not stop.wait(max(float(rate_s), 0.05)):
+                try:
+                    ex.execute(statement)
+                except Exception:
+                    pass
+        th = threading.Thread(target=loop, daemon=True)
+        _JOBS[name] = {"name": name, "thread": th, "stop": stop,
+                       "done": False, "cancelled": False, "repeat": True}
+        th.start()
+        return ["name", "rate"], [[name, rate_s]]
+
+    @register("apoc.periodic.countdown")
+    def _pd_countdown(ex, name, statement, delay_s):
+        def run():
+            time.sleep(min(float(delay_s), 60))
+            if not _JOBS.get(name, {}).get("cancelled"):
+                try:
+                    ex.execute(statement)
+                finally:
+                    _JOBS[name]["done"] = True
+        th = threading.Thread(target=run, daemon=True)
+        _JOBS[name] = {"name": name, "thread": th, "done": False,
+                       "cancelled": False, "repeat": False}
+        th.start()
+        return ["name", "delay"], [[name, delay_s]]
+
+    @register("apoc.periodic.list")
+    def _pd_list(ex):
+        return ["name", "done", "cancelled", "repeat"], [
+            [j["name"], j.get("done", False), j.get("cancelled", False),
+             j.get("repeat", False)] for j in _JOBS.values()]
+
+    @register("apoc.periodic.cancel")
+    def _pd_cancel(ex, name):
+        j = _JOBS.get(name)
+        if j:
+            j["cancelled"] = True
+            if "stop" in j:
+                j["stop"].set()
+        return ["name", "cancelled"], [[name, j is not None]]
+
+    @register("apoc.periodic.truncate")
+    def _pd_truncate(ex):
+        n = 0
+        for node in list(eng.all_nodes()):
+            try:
+                eng.detach_delete_node(node.id)
+                n += 1
+            except Exception:
+                pass
+        return ["nodesDeleted"], [[n]]
+
+    # -------------------- apoc.node/rel mutator procs --------------------
+    @register("apoc.node.setProperty")
+    def _np_setp(ex, node, key, value):
+        return _cr_setp(ex, node, key, value)
+
+    @register("apoc.node.setProperties")
+    def _np_setps(ex, node, props):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n.properties.update(dict(props or {}))
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.node.removeProperty")
+    def _np_rmp(ex, node, key):
+        return _cr_rmps(ex, node, [key])
+
+    @register("apoc.node.removeProperties")
+    def _np_rmps(ex, node, keys):
+        return _cr_rmps(ex, node, keys)
+
+    @register("apoc.node.addLabel")
+    def _np_addl(ex, node, label):
+        return _cr_addl(ex, node, [label])
+
+    @register("apoc.node.addLabels")
+    def _np_addls(ex, node, labels):
+        return _cr_addl(ex, node, labels)
+
+    @register("apoc.node.removeLabel")
+    def _np_rml(ex, node, label):
+        return _cr_rml(ex, node, [label])
+
+    @register("apoc.node.removeLabels")
+    def _np_rmls(ex, node, labels):
+        return _cr_rml(ex, node, labels)
+
+    @register("apoc.node.clone")
+    def _np_clone(ex, node):
+        return _cr_clone(ex, node)
+
+    @register("apoc.node.fromMap")
+    def _np_frommap(ex, m):
+        m = dict(m or {})
+        n = eng.create_node(Node(id=str(m.get("id") or new_id("n")),
+                                 labels=list(m.get("labels", [])),
+                                 properties=dict(m.get("properties", {}))))
+        return ["node"], [[n]]
+
+    @register("apoc.rel.setProperty")
+    def _rp_setp(ex, rel, key, value):
+        return _cr_setrp(ex, rel, key, value)
+
+    @register("apoc.rel.setProperties")
+    def _rp_setps(ex, rel, props):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        e.properties.update(dict(props or {}))
+        return ["rel"], [[eng.update_edge(e)]]
+
+    @register("apoc.rel.removeProperty")
+    def _rp_rmp(ex, rel, key):
+        return _cr_rmrps(ex, rel, [key])
+
+    @register("apoc.rel.removeProperties")
+    def _rp_rmps(ex, rel, keys):
+        return _cr_rmrps(ex, rel, keys)
+
+    @register("apoc.rel.delete")
+    def _rp_del(ex, rel):
+        eng.delete_edge(rel.id if isinstance(rel, Edge) else rel)
+        return ["deleted"], [[True]]
+
+    @register("apoc.rel.exists")
+    def _rp_exists(ex, rel):
+        try:
+            eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+            return ["value"], [[True]]
+        except Exception:
+            return ["value"], [[False]]
+
+    @register("apoc.rel.reverse")
+    def _rp_reverse(ex, rel):
+        return _rf_invert(ex, rel)
+
+    @register("apoc.rel.clone")
+    def _rp_clone(ex, rel):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        ne = eng.create_edge(Edge(id=new_id("e"), type=e.type,
+                                  start_node=e.start_node,
+                                  end_node=e.end_node,
+                                  properties=dict(e.properties)))
+        return ["rel"], [[ne]]
+
+    @register("apoc.rel.fromMap")
+    def _rp_frommap(ex, m):
+        m = dict(m or {})
+        e = eng.create_edge(Edge(
+            id=str(m.get("id") or new_id("e")),
+            type=m.get("type", "RELATED"),
+            start_node=str(m.get("start") or m.get("startNode")),
+            end_node=str(m.get("end") or m.get("endNode")),
+            properties=dict(m.get("properties", {}))))
+        return ["rel"], [[e]]
+
+    # -------------------- apoc.cypher parallel --------------------
+    @register("apoc.cypher.parallel")
+    def _cy_parallel(ex, fragment, params_map, key):
+        # run the fragment once per value of params_map[key] (thread pool)
+        import concurrent.futures as _fut
+        vals = (params_map or {}).get(key, [])
+        rows = []
+
+        def one(v):
+            r = ex.execute(fragment, {key: v})
+            return [dict(zip(r.columns, row)) for row in r.rows]
+        with _fut.ThreadPoolExecutor(max_workers=4) as pool:
+            for part in pool.map(one, vals):
+                rows += [[x] for x in part]
+        return ["value"], rows
+
+    @register("apoc.cypher.mapParallel")
+    def _cy_map_parallel(ex, fragment, config, items):
+        import concurrent.futures as _fut
+        rows = []
+
+        def one(v):
+            r = ex.execute(fragment, {"_": v})
+            return [dict(zip(r.columns, row)) for row in r.rows]
+        with _fut.ThreadPoolExecutor(max_workers=4) as pool:
+            for part in pool.map(one, items or []):
+                rows += [[x] for x in part]
+        return ["value"], rows
+
+    # -------------------- apoc.trigger extras --------------------
+    @register("apoc.trigger.install")
+    def _tg_install(ex, dbname, name, statement, selector=None):
+        return procs["apoc.trigger.add"](ex, name, statement, selector)
+
+    @register("apoc.trigger.drop")
+    def _tg_drop(ex, dbname, name):
+        return procs["apoc.trigger.remove"](ex, name)
+
+    @register("apoc.trigger.show")
+    def _tg_show(ex, dbname=None):
+        return procs["apoc.trigger.list"](ex)
+
     @register("apoc.merge.relationshipEager")
     def _mg_rel_eager(ex, start, rel_type, ident_props=None, props=None,
                       end=None):

@@ -172,3 +172,68 @@ class TestEngineProcedures:
     def test_function_count(self):
         from nornicdb_amd.cypher.functions import FUNCTIONS
         assert len([k for k in FUNCTIONS if k.startswith("apoc.")]) >= 450
+
+
+class TestApocBatch3:
+    """atomic extras, periodic scheduler, node/rel mutator procs,
+    cypher.parallel, text phonetics, coll/map extras, convert.toTree."""
+
+    @pytest.fixture
+    def db(self):
+        return NornicDB(MemoryEngine(), auto_embed=False)
+
+    def test_atomic_cas_and_concat(self, db):
+        db.cypher("CREATE (:A {v: 0, s: 'x'})")
+        r = db.cypher("MATCH (a:A) CALL apoc.atomic.compareAndSwap(a, 'v', 0, 5) "
+                      "YIELD swapped RETURN swapped")
+        assert r.rows == [[True]]
+        r = db.cypher("MATCH (a:A) CALL apoc.atomic.compareAndSwap(a, 'v', 0, 9) "
+                      "YIELD swapped RETURN swapped")
+        assert r.rows == [[False]]
+        db.cypher("MATCH (a:A) CALL apoc.atomic.concat(a, 's', 'y') "
+                  "YIELD value RETURN value")
+        assert db.cypher("MATCH (a:A) RETURN a.s").rows == [["xy"]]
+
+    def test_periodic_submit_and_list(self, db):
+        import time
+        db.cypher("CALL apoc.periodic.submit('t1', 'CREATE (:Done)') "
+                  "YIELD name RETURN name")
+        time.sleep(0.3)
+        assert db.cypher("MATCH (d:Done) RETURN count(d)").rows == [[1]]
+        rows = db.cypher("CALL apoc.periodic.list() YIELD name, done "
+                         "RETURN name, done").rows
+        assert ["t1", True] in rows
+
+    def test_node_mutator_procs(self, db):
+        db.cypher("CREATE (:M {a: 1})")
+        db.cypher("MATCH (m:M) CALL apoc.node.addLabel(m, 'Extra') "
+                  "YIELD node RETURN node")
+        assert db.cypher("MATCH (m:Extra) RETURN count(m)").rows == [[1]]
+        db.cypher("MATCH (m:M) CALL apoc.node.removeProperty(m, 'a') "
+                  "YIELD node RETURN node")
+        assert db.cypher("MATCH (m:M) RETURN m.a").rows == [[None]]
+
+    def test_cypher_parallel(self, db):
+        r = db.cypher("CALL apoc.cypher.parallel('RETURN $x * 2 AS y', "
+                      "{x: [1,2,3]}, 'x') YIELD value RETURN value.y "
+                      "ORDER BY value.y")
+        assert [x[0] for x in r.rows] == [2, 4, 6]
+
+    def test_text_phonetics(self, db):
+        r = db.cypher("RETURN apoc.text.phonetic('Robert'), "
+                      "apoc.text.fuzzyMatch('color', 'colour')")
+        assert r.rows[0] == ["R163", True]
+
+    def test_coll_map_extras(self, db):
+        assert one(db, "RETURN apoc.coll.insertAll([1,4], 1, [2,3])") == [1, 2, 3, 4]
+        assert one(db, "RETURN apoc.coll.dropDuplicateNeighbors([1,1,2,2,1])") == [1, 2, 1]
+        assert one(db, "RETURN apoc.map.fromValues(['a', 1, 'b', 2])") == {"a": 1, "b": 2}
+        assert one(db, "RETURN apoc.map.mget({a: 1}, ['a', 'z'], [0, 0])") == [1, 0]
+
+    def test_convert_to_tree(self, db):
+        db.cypher("CREATE (a:T {name:'root'})-[:HAS]->(b:T {name:'leaf'})")
+        r = db.cypher("MATCH p = (a:T)-[:HAS]->(b) "
+                      "RETURN apoc.convert.toTree(collect(p))")
+        tree = r.rows[0][0]
+        assert tree and tree[0]["name"] == "root"
+        assert tree[0]["has"][0]["name"] == "leaf"
