@@ -21,6 +21,8 @@ from typing import Optional
 
 import msgpack
 
+from . import codec as _codec
+
 from .memory import MemoryEngine
 from .types import Edge, Node, StorageError
 from . import wal as W
@@ -116,7 +118,8 @@ class PersistentEngine(MemoryEngine):
         snap_path = os.path.join(self.data_dir, self.SNAPSHOT)
         if os.path.exists(snap_path):
             with open(snap_path, "rb") as f:
-                state = msgpack.unpackb(f.read(), raw=False, strict_map_key=False)
+                state = msgpack.unpackb(f.read(), raw=False, strict_map_key=False,
+                                        object_hook=_codec.object_hook)
             self.load_state(state)
         wal_path = os.path.join(self.data_dir, self.WAL_FILE)
         # two-phase: collect committed tx ids, then apply
@@ -180,7 +183,8 @@ class PersistentEngine(MemoryEngine):
             state = self.dump_state()
             tmp = os.path.join(self.data_dir, self.SNAPSHOT + ".tmp")
             with open(tmp, "wb") as f:
-                f.write(msgpack.packb(state, use_bin_type=True))
+                f.write(msgpack.packb(state, use_bin_type=True,
+                                      default=_codec.default))
                 f.flush()
                 os.fsync(f.fileno())
             os.replace(tmp, os.path.join(self.data_dir, self.SNAPSHOT))

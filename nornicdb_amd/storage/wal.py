@@ -27,6 +27,8 @@ from typing import Optional
 
 import msgpack
 
+from . import codec as _codec
+
 MAGIC = 0x4E44
 _HDR = struct.Struct("<HBII")
 
@@ -117,7 +119,8 @@ class WAL:
     def append(self, op: int, payload: dict) -> None:
         if self.degraded:
             raise WALDegraded("WAL is in degraded mode (previous write failed)")
-        data = msgpack.packb(payload, use_bin_type=True)
+        data = msgpack.packb(payload, use_bin_type=True,
+                              default=_codec.default)
         rec = _HDR.pack(MAGIC, op, len(data), zlib.crc32(data)) + data
         with self._lock:
             try:
@@ -206,5 +209,6 @@ class WAL:
                 if tolerate_corruption:
                     break
                 raise WALCorruption(off, "crc mismatch")
-            yield op, msgpack.unpackb(payload, raw=False)
+            yield op, msgpack.unpackb(payload, raw=False,
+                                      object_hook=_codec.object_hook)
             off += _HDR.size + plen

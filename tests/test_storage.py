@@ -400,3 +400,36 @@ class TestWALSegments:
         assert not [f for f in os.listdir(tmp_path) if f.startswith("seg2.wal.")]
         w.close()
         assert list(WAL.replay(p)) == []
+
+
+class TestTemporalPersistence:
+    """Temporal property values survive WAL replay and snapshots
+    (storage/codec.py msgpack hooks)."""
+
+    def test_wal_replay_roundtrip(self, tmp_path):
+        from nornicdb_amd.db import open_db
+        d = str(tmp_path / "tdb")
+        mgr = open_db(d)
+        mgr.get().cypher(
+            "CREATE (:E {d: date('2026-01-01'), "
+            "ts: datetime('2026-06-01T12:00:00Z'), dur: duration('P1DT2H')})")
+        mgr.close()
+        mgr2 = open_db(d)
+        r = mgr2.get().cypher(
+            "MATCH (e:E) RETURN e.d.year, e.ts.month, e.dur.hours")
+        assert r.rows == [[2026, 6, 2]]
+        mgr2.close()
+
+    def test_snapshot_roundtrip(self, tmp_path):
+        from nornicdb_amd.db import open_db
+        d = str(tmp_path / "tdb2")
+        mgr = open_db(d)
+        db = mgr.get()
+        db.cypher("CREATE (:E {d: date('2025-12-31')})")
+        if hasattr(db.engine, "snapshot"):
+            db.engine.snapshot()
+        mgr.close()
+        mgr2 = open_db(d)
+        assert mgr2.get().cypher(
+            "MATCH (e:E) RETURN toString(e.d)").rows == [["2025-12-31"]]
+        mgr2.close()
