@@ -273,6 +273,52 @@ LOCAL_DATETIME_TAG = 0x64  # 'd'
 DURATION_TAG = 0x45        # 'E' months, days, seconds, nanoseconds
 
 
+def temporal_from_struct(s: "Structure"):
+    """Inverse of temporal_struct: Bolt temporal structures (as sent in
+    driver parameters) -> cypher temporal values. Returns None if the
+    tag is not temporal."""
+    import datetime as _dt
+
+    from ..cypher import temporal as _tp
+
+    if s.tag == DURATION_TAG:
+        months, days, secs, nanos = s.fields
+        return _tp.CypherDuration(months, days, secs, nanos)
+    if s.tag == DATE_TAG:
+        return _tp.CypherDate(_dt.date(1970, 1, 1) +
+                              _dt.timedelta(days=s.fields[0]))
+    if s.tag == LOCAL_DATETIME_TAG:
+        secs, nanos = s.fields
+        return _tp.CypherDateTime(
+            _dt.datetime.utcfromtimestamp(secs) +
+            _dt.timedelta(microseconds=nanos // 1000))
+    if s.tag == DATETIME_TAG:
+        secs, nanos, off = s.fields
+        tz = _dt.timezone(_dt.timedelta(seconds=off))
+        return _tp.CypherDateTime(_dt.datetime.fromtimestamp(
+            secs + nanos / 1e9, tz))
+    if s.tag == DATETIME_LEGACY_TAG:
+        secs, nanos, off = s.fields
+        tz = _dt.timezone(_dt.timedelta(seconds=off))
+        return _tp.CypherDateTime(_dt.datetime.fromtimestamp(
+            secs - off + nanos / 1e9, tz))
+    if s.tag == LOCAL_TIME_TAG:
+        nanos = s.fields[0]
+        us = nanos // 1000
+        return _tp.make_time(_dt.time((us // 3600000000) % 24,
+                                      (us // 60000000) % 60,
+                                      (us // 1000000) % 60,
+                                      us % 1000000), local=True)
+    if s.tag == TIME_TAG:
+        nanos, off = s.fields
+        us = nanos // 1000
+        t = _dt.time((us // 3600000000) % 24, (us // 60000000) % 60,
+                     (us // 1000000) % 60, us % 1000000,
+                     tzinfo=_dt.timezone(_dt.timedelta(seconds=off)))
+        return _tp.CypherTime(t)
+    return None
+
+
 def temporal_struct(v, bolt5: bool = False):
     """Convert a cypher temporal value to its Bolt structure (or None)."""
     import datetime as _dt

@@ -244,9 +244,21 @@ class BoltSession:
         self.write_message(R_SUCCESS, {})
         return False
 
+    def from_bolt(self, v):
+        """Convert inbound parameter values: temporal structures ->
+        cypher temporal values; recurse containers."""
+        if isinstance(v, ps.Structure):
+            t = ps.temporal_from_struct(v)
+            return t if t is not None else v
+        if isinstance(v, list):
+            return [self.from_bolt(x) for x in v]
+        if isinstance(v, dict):
+            return {k: self.from_bolt(x) for k, x in v.items()}
+        return v
+
     def on_run(self, fields) -> bool:
         query = fields[0] if fields else ""
-        params = fields[1] if len(fields) > 1 else {}
+        params = self.from_bolt(fields[1] if len(fields) > 1 else {})
         extra = fields[2] if len(fields) > 2 else {}
         db = extra.get("db") or self.db
         try:

@@ -193,3 +193,29 @@ class TestNativeCodec:
         for v in vals:
             assert ps.pack(v) == ps.pack_py(v)
             assert ps.unpack_py(ps.pack(v)) == v
+
+
+class TestTemporalParams:
+    """Inbound Bolt temporal structures in parameters convert to cypher
+    temporal values (packstream.temporal_from_struct); outbound values
+    convert back (temporal_struct) — full driver roundtrip."""
+
+    def test_struct_roundtrips(self):
+        from nornicdb_amd.bolt import packstream as ps
+        from nornicdb_amd.cypher import temporal as tp
+        for v in (tp.make_date("2026-09-12"),
+                  tp.make_duration("P1Y2M3DT4H5M6S"),
+                  tp.make_datetime("2026-09-12T08:00:00+02:00"),
+                  tp.make_datetime("2026-09-12T08:00:00", local=True)):
+            s = ps.temporal_struct(v, bolt5=True)
+            back = ps.temporal_from_struct(s)
+            assert type(back) is type(v)
+            assert str(back)[:19] == str(v)[:19]
+
+    def test_bolt4_legacy_datetime(self):
+        from nornicdb_amd.bolt import packstream as ps
+        from nornicdb_amd.cypher import temporal as tp
+        v = tp.make_datetime("2026-09-12T08:00:00+02:00")
+        s4 = ps.temporal_struct(v, bolt5=False)
+        assert s4.tag == ps.DATETIME_LEGACY_TAG
+        assert str(ps.temporal_from_struct(s4))[:19] == str(v)[:19]
