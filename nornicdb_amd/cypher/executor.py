@@ -567,6 +567,20 @@ class Executor:
             base = self._eval(e.expr, row, params)
             if base is None:
                 return None
+            if hasattr(base, "component"):  # temporal value
+                props = {}
+                out = {}
+                for it in e.items:
+                    if it[0] == "prop":
+                        try:
+                            out[it[1]] = base.component(it[1])
+                        except KeyError:
+                            out[it[1]] = None
+                    elif it[0] == "kv":
+                        out[it[1]] = self._eval(it[2], row, params)
+                    elif it[0] == "var":
+                        out[it[1]] = row.get(it[1])
+                return out
             props = (dict(base.properties) if isinstance(base, (Node, Edge))
                      else dict(base or {}))
             out = {}

@@ -94,6 +94,12 @@ class CypherDuration:
                                   self.nanoseconds + o.nanoseconds)
         return NotImplemented
 
+    def __radd__(self, o):
+        # sum() starts from 0
+        if o == 0:
+            return self
+        return NotImplemented
+
     def __sub__(self, o):
         if isinstance(o, CypherDuration):
             return CypherDuration(self.months - o.months, self.days - o.days,

@@ -573,3 +573,29 @@ class TestProfile:
         ex = Executor(MemoryEngine())
         r = ex.execute("EXPLAIN MATCH (n) RETURN n")
         assert r.columns == ["plan"]
+
+
+class TestTemporalAggregation:
+    def _ex(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        return Executor(MemoryEngine())
+
+    def test_sum_avg_durations(self):
+        ex = self._ex()
+        r = ex.execute("UNWIND [duration('PT1H'), duration('PT2H')] AS x "
+                       "RETURN toString(sum(x)), toString(avg(x))")
+        assert r.rows == [["PT3H", "PT1H30M"]]
+
+    def test_min_max_order_distinct(self):
+        ex = self._ex()
+        r = ex.execute("UNWIND [date('2026-01-02'), date('2026-01-01'), "
+                       "date('2026-01-01')] AS d "
+                       "RETURN toString(min(d)), toString(max(d)), "
+                       "count(DISTINCT d)")
+        assert r.rows == [["2026-01-01", "2026-01-02", 2]]
+
+    def test_map_projection_on_temporal(self):
+        ex = self._ex()
+        r = ex.execute("WITH date('2026-01-01') AS d RETURN d {.year, .month}")
+        assert r.rows == [[{"year": 2026, "month": 1}]]
