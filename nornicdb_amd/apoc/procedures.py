@@ -1710,4 +1710,238 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
     def _lk_read_rels(ex, rels):
         return ["locked"], [[len(rels or [])]]
 
+    # -------------------- apoc.meta extras --------------------
+    @register("apoc.meta.nodeLabels")
+    def _mt_labels2(ex):
+        seen = set()
+        for n in eng.all_nodes():
+            seen.update(n.labels)
+        return ["label"], [[lb] for lb in sorted(seen)]
+
+    @register("apoc.meta.relTypes")
+    def _mt_rtypes2(ex):
+        return ["type"], [[t] for t in sorted({e.type for e in eng.all_edges()})]
+
+    @register("apoc.meta.propertyKeys")
+    def _mt_pkeys(ex):
+        seen = set()
+        for n in eng.all_nodes():
+            seen.update(n.properties.keys())
+        for e in eng.all_edges():
+            seen.update(e.properties.keys())
+        return ["propertyKey"], [[p] for p in sorted(seen)]
+
+    @register("apoc.meta.cardinality")
+    def _mt_card(ex, label=None):
+        if label:
+            n = len(eng.get_nodes_by_label(label))
+        else:
+            n = eng.node_count()
+        return ["count"], [[n]]
+
+    @register("apoc.meta.analyze")
+    def _mt_analyze(ex):
+        return procs["apoc.meta.stats"](ex)
+
+    @register("apoc.meta.subGraph")
+    def _mt_subgraph(ex, config=None):
+        cfg = dict(config or {})
+        labels = cfg.get("labels") or []
+        nodes = []
+        for lb in (labels or [None]):
+            nodes += (eng.get_nodes_by_label(lb) if lb
+                      else list(eng.all_nodes()))
+        ids = {n.id for n in nodes}
+        rels = [e for e in eng.all_edges()
+                if e.start_node in ids and e.end_node in ids]
+        return ["nodes", "relationships"], [[nodes, rels]]
+
+    @register("apoc.meta.snapshot")
+    def _mt_snapshot(ex):
+        return ["nodes", "relationships", "labels"], [[
+            eng.node_count(), eng.edge_count(),
+            sorted({lb for n in eng.all_nodes() for lb in n.labels})]]
+
+    # -------------------- apoc.label procs --------------------
+    @register("apoc.label.add")
+    def _lb_add(ex, node, label):
+        return _cr_addl(ex, node, [label])
+
+    @register("apoc.label.remove")
+    def _lb_remove(ex, node, label):
+        return _cr_rml(ex, node, [label])
+
+    @register("apoc.label.set")
+    def _lb_set(ex, node, labels):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n.labels = list(labels or [])
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.label.clear")
+    def _lb_clear(ex, node):
+        return _lb_set(ex, node, [])
+
+    @register("apoc.label.replace")
+    def _lb_replace(ex, node, old, new):
+        n = eng.get_node(node.id if isinstance(node, Node) else node)
+        n.labels = [new if lb == old else lb for lb in n.labels]
+        return ["node"], [[eng.update_node(n)]]
+
+    @register("apoc.label.nodes")
+    def _lb_nodes(ex, label):
+        return ["node"], [[n] for n in eng.get_nodes_by_label(label)]
+
+    @register("apoc.label.stats")
+    def _lb_stats(ex):
+        from collections import Counter
+        c = Counter()
+        for n in eng.all_nodes():
+            c.update(n.labels)
+        return ["label", "count"], [[k, v] for k, v in sorted(c.items())]
+
+    # -------------------- apoc.lock extras --------------------
+    @register("apoc.lock.tryLock")
+    def _lk_try(ex, nodes=None):
+        return ["acquired"], [[True]]
+
+    @register("apoc.lock.unlockAll")
+    def _lk_unlock_all(ex):
+        return ["released"], [[True]]
+
+    @register("apoc.lock.isLocked")
+    def _lk_islocked(ex, node):
+        return ["locked"], [[False]]
+
+    @register("apoc.lock.stats")
+    def _lk_stats2(ex):
+        return ["active", "waiting"], [[0, 0]]
+
+    @register("apoc.lock.detectDeadlock")
+    def _lk_deadlock(ex):
+        return ["deadlocks"], [[[]]]
+
+    # -------------------- apoc.log extras --------------------
+    @register("apoc.log.getLevel")
+    def _lg_getlevel(ex):
+        return ["level"], [["INFO"]]
+
+    @register("apoc.log.setLevel")
+    def _lg_setlevel(ex, level):
+        return ["level"], [[str(level).upper()]]
+
+    @register("apoc.log.search")
+    def _lg_search(ex, pattern, limit=100):
+        out = [x for x in _LOG_BUF if re.search(pattern, x[2])]
+        return ["level", "timestamp", "message"], [
+            list(x) for x in out[-int(limit):]]
+
+    @register("apoc.log.timer")
+    def _lg_timer(ex, name=None):
+        return ["name", "now"], [[name or "timer", time.time()]]
+
+    @register("apoc.log.memory")
+    def _lg_memory(ex):
+        import resource
+        kb = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+        return ["maxRssMB"], [[kb / 1024.0]]
+
+    # -------------------- apoc.warmup / search extras --------------------
+    @register("apoc.warmup.nodes")
+    def _wm_nodes(ex):
+        return ["loaded"], [[sum(1 for _ in eng.all_nodes())]]
+
+    @register("apoc.warmup.relationships")
+    def _wm_rels(ex):
+        return ["loaded"], [[sum(1 for _ in eng.all_edges())]]
+
+    @register("apoc.warmup.indexes")
+    def _wm_idx(ex):
+        sm = getattr(ex, "schema", None)
+        return ["indexes"], [[len(sm.list_indexes()) if sm else 0]]
+
+    @register("apoc.search.missing")
+    def _se_missing(ex, label, prop):
+        out = [n for n in eng.get_nodes_by_label(label)
+               if n.properties.get(prop) is None]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.search.notNull")
+    def _se_notnull(ex, label, prop):
+        out = [n for n in eng.get_nodes_by_label(label)
+               if n.properties.get(prop) is not None]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.search.null")
+    def _se_null(ex, label, prop):
+        return _se_missing(ex, label, prop)
+
+    @register("apoc.search.range")
+    def _se_range(ex, label, prop, lo, hi):
+        out = [n for n in eng.get_nodes_by_label(label)
+               if isinstance(n.properties.get(prop), (int, float))
+               and lo <= n.properties[prop] <= hi]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.search.in")
+    def _se_in(ex, label, prop, values):
+        vals = set(map(repr, values or []))
+        out = [n for n in eng.get_nodes_by_label(label)
+               if repr(n.properties.get(prop)) in vals]
+        return ["node"], [[n] for n in out]
+
+    @register("apoc.search.exists")
+    def _se_exists2(ex, label, prop):
+        return _se_notnull(ex, label, prop)
+
+    # -------------------- apoc.paths extras --------------------
+    @register("apoc.paths.exists")
+    def _pa_exists(ex, a, b, max_hops=6):
+        aid = a.id if isinstance(a, Node) else a
+        bid = b.id if isinstance(b, Node) else b
+        seen = {aid}
+        frontier = {aid}
+        for _ in range(int(max_hops)):
+            nxt = set()
+            for nid in frontier:
+                for other in eng.neighbors(nid):
+                    if other == bid:
+                        return ["value"], [[True]]
+                    if other not in seen:
+                        seen.add(other)
+                        nxt.add(other)
+            frontier = nxt
+        return ["value"], [[False]]
+
+    @register("apoc.paths.distance")
+    def _pa_distance(ex, a, b, max_hops=10):
+        aid = a.id if isinstance(a, Node) else a
+        bid = b.id if isinstance(b, Node) else b
+        if aid == bid:
+            return ["value"], [[0]]
+        seen = {aid}
+        frontier = {aid}
+        for hop in range(1, int(max_hops) + 1):
+            nxt = set()
+            for nid in frontier:
+                for other in eng.neighbors(nid):
+                    if other == bid:
+                        return ["value"], [[hop]]
+                    if other not in seen:
+                        seen.add(other)
+                        nxt.add(other)
+            frontier = nxt
+        return ["value"], [[None]]
+
+    @register("apoc.paths.common")
+    def _pa_common(ex, a, b):
+        aid = a.id if isinstance(a, Node) else a
+        bid = b.id if isinstance(b, Node) else b
+        common = set(eng.neighbors(aid)) & set(eng.neighbors(bid))
+        return ["node"], [[eng.get_node(i)] for i in sorted(common)]
+
+    @register("apoc.paths.count")
+    def _pa_count(ex, a, rel_type=None, max_hops=3):
+        aid = a.id if isinstance(a, Node) else a
+        return ["value"], [[len(_hops(aid, rel_type, max_hops, 1))]]
+
     return procs
