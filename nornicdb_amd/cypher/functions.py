@@ -238,3 +238,123 @@ class Aggregator:
         if fin is not None:
             return fin(self.values, extra)
         raise CypherRuntimeError(f"unknown aggregate {n}")
+
+
+# ---- Cypher 5 additions: OrNull casts, valueType, point, normalize ----
+register("char_length", lambda s: None if s is None else len(s))
+register("character_length", lambda s: None if s is None else len(s))
+register("btrim", lambda s, chars=None: None if s is None else
+         (s.strip() if chars is None else s.strip(chars)))
+register("normalize", lambda s, form="NFC": None if s is None else
+         __import__("unicodedata").normalize(form, s))
+register("isnan", lambda x: isinstance(x, float) and x != x)
+register("nullif", lambda a, b: None if a == b else a)
+register("tointegerornull", lambda x: _to_int(x))
+register("tofloatornull", lambda x: _to_float(x))
+register("tobooleanornull", lambda x: _to_bool(x))
+register("tostringornull", lambda x: None if not isinstance(
+    x, (str, bool, int, float)) else (str(x).lower() if isinstance(x, bool)
+                                      else str(x)))
+
+
+def _value_type(x):
+    if x is None:
+        return "NULL"
+    if isinstance(x, bool):
+        return "BOOLEAN NOT NULL"
+    if isinstance(x, int):
+        return "INTEGER NOT NULL"
+    if isinstance(x, float):
+        return "FLOAT NOT NULL"
+    if isinstance(x, str):
+        return "STRING NOT NULL"
+    if isinstance(x, list):
+        return "LIST<ANY> NOT NULL"
+    if isinstance(x, dict):
+        return "MAP NOT NULL"
+    if isinstance(x, Node):
+        return "NODE NOT NULL"
+    if isinstance(x, Edge):
+        return "RELATIONSHIP NOT NULL"
+    return type(x).__name__.upper() + " NOT NULL"
+
+
+register("valuetype", _value_type)
+
+
+class CypherPoint:
+    """2D/3D point (cartesian or WGS-84); reference supports points via
+    its Cypher layer. Accessors via component() like temporal values."""
+
+    __slots__ = ("x", "y", "z", "crs")
+
+    def __init__(self, m):
+        m = {k.lower(): v for k, v in dict(m).items()}
+        if "latitude" in m or "longitude" in m:
+            self.crs = "wgs-84"
+            self.x = float(m.get("longitude", 0.0))
+            self.y = float(m.get("latitude", 0.0))
+        else:
+            self.crs = m.get("crs", "cartesian")
+            self.x = float(m.get("x", 0.0))
+            self.y = float(m.get("y", 0.0))
+        self.z = float(m["z"]) if "z" in m else (
+            float(m["height"]) if "height" in m else None)
+
+    def component(self, key):
+        k = key.lower()
+        vals = {"x": self.x, "y": self.y, "z": self.z, "crs": self.crs,
+                "longitude": self.x, "latitude": self.y, "height": self.z,
+                "srid": 4326 if self.crs == "wgs-84" else 7203}
+        if k in vals:
+            return vals[k]
+        raise KeyError(key)
+
+    def __eq__(self, o):
+        return (isinstance(o, CypherPoint) and self.crs == o.crs
+                and (self.x, self.y, self.z) == (o.x, o.y, o.z))
+
+    def __hash__(self):
+        return hash(("point", self.crs, self.x, self.y, self.z))
+
+    def __str__(self):
+        z = f", z: {self.z}" if self.z is not None else ""
+        return f"point({{x: {self.x}, y: {self.y}{z}, crs: '{self.crs}'}})"
+
+    __repr__ = __str__
+
+
+def _point_distance(a, b):
+    if a is None or b is None:
+        return None
+    if a.crs == "wgs-84" and b.crs == "wgs-84":
+        # haversine metres
+        import math as _m
+        p1, p2 = _m.radians(a.y), _m.radians(b.y)
+        dp = _m.radians(b.y - a.y)
+        dl = _m.radians(b.x - a.x)
+        h = _m.sin(dp / 2) ** 2 + _m.cos(p1) * _m.cos(p2) * _m.sin(dl / 2) ** 2
+        return 2 * 6371008.8 * _m.asin(_m.sqrt(h))
+    dz = ((a.z or 0) - (b.z or 0)) ** 2 if (a.z is not None or
+                                            b.z is not None) else 0
+    return math.sqrt((a.x - b.x) ** 2 + (a.y - b.y) ** 2 + dz)
+
+
+register("point", lambda m: None if m is None else CypherPoint(m))
+register("point.distance", _point_distance)
+register("distance", _point_distance)
+register("point.withinbbox", lambda p, lo, hi: None if p is None else
+         (lo.x <= p.x <= hi.x and lo.y <= p.y <= hi.y))
+
+
+# Neo4j math semantics: out-of-domain returns NaN (not an error)
+register("sqrt", lambda x: None if x is None else (
+    float("nan") if x < 0 else math.sqrt(_num(x))))
+register("log", lambda x: None if x is None else (
+    float("nan") if x <= 0 else math.log(_num(x))))
+register("log10", lambda x: None if x is None else (
+    float("nan") if x <= 0 else math.log10(_num(x))))
+register("asin", lambda x: None if x is None else (
+    float("nan") if abs(x) > 1 else math.asin(_num(x))))
+register("acos", lambda x: None if x is None else (
+    float("nan") if abs(x) > 1 else math.acos(_num(x))))

@@ -599,3 +599,52 @@ class TestTemporalAggregation:
         ex = self._ex()
         r = ex.execute("WITH date('2026-01-01') AS d RETURN d {.year, .month}")
         assert r.rows == [[{"year": 2026, "month": 1}]]
+
+
+class TestCypher5Builtins:
+    """char_length/btrim/normalize/isNaN/nullIf/valueType/*OrNull casts and
+    the point() spatial type (Neo4j 5 builtin surface)."""
+
+    def _ex(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        return Executor(MemoryEngine())
+
+    def test_string_builtins(self):
+        ex = self._ex()
+        assert ex.execute("RETURN char_length('abc'), btrim('xxaxx', 'x'), "
+                          "normalize('abc')").rows == [[3, "a", "abc"]]
+
+    def test_null_helpers(self):
+        ex = self._ex()
+        assert ex.execute(
+            "RETURN nullIf(1, 1), nullIf('a', 'b'), isNaN(sqrt(-1.0)), "
+            "isNaN(1.0)").rows == [[None, "a", True, False]]
+        assert ex.execute(
+            "RETURN toIntegerOrNull('nope'), toIntegerOrNull('7'), "
+            "toFloatOrNull([1]), toStringOrNull(true)"
+        ).rows == [[None, 7, None, "true"]]
+
+    def test_value_type(self):
+        ex = self._ex()
+        assert ex.execute("RETURN valueType(1)").rows == [["INTEGER NOT NULL"]]
+        assert ex.execute("RETURN valueType(null)").rows == [["NULL"]]
+
+    def test_point_cartesian(self):
+        ex = self._ex()
+        r = ex.execute("WITH point({x: 3, y: 4}) AS p "
+                       "RETURN p.x, p.y, p.crs, "
+                       "point.distance(p, point({x: 0, y: 0}))")
+        assert r.rows == [[3.0, 4.0, "cartesian", 5.0]]
+
+    def test_point_wgs84(self):
+        ex = self._ex()
+        r = ex.execute(
+            "RETURN point.distance(point({latitude: 48.8566, longitude: 2.3522}), "
+            "point({latitude: 51.5074, longitude: -0.1278}))")
+        assert 330e3 < r.rows[0][0] < 350e3  # Paris-London
+
+    def test_math_domain_nan(self):
+        ex = self._ex()
+        assert ex.execute("RETURN isNaN(log(-1.0)), isNaN(asin(2.0))"
+                          ).rows == [[True, True]]
