@@ -321,6 +321,11 @@ class Executor:
                      "default", "home"],
                     [[n, "standard", "read-write", "primary", "online",
                       n == "neo4j", n == "neo4j"] for n in names])
+            if c.kind == "transactions":
+                # single current transaction (our executor is synchronous)
+                return Result(
+                    ["database", "transactionId", "currentQuery", "status"],
+                    [[self.current_database, "tx-0", "", "Running"]])
             if c.kind == "procedures":
                 return Result(["name", "description", "mode"],
                               [[n, (getattr(f, "__doc__", "") or "").strip()

@@ -356,7 +356,9 @@ class Parser:
                  "CONSTRAINT": "constraints", "CONSTRAINTS": "constraints",
                  "DATABASE": "databases", "DATABASES": "databases",
                  "PROCEDURE": "procedures", "PROCEDURES": "procedures",
-                 "FUNCTION": "functions", "FUNCTIONS": "functions"}
+                 "FUNCTION": "functions", "FUNCTIONS": "functions",
+                 "TRANSACTION": "transactions",
+                 "TRANSACTIONS": "transactions"}
         if word not in kinds:
             raise CypherSyntaxError(f"cannot SHOW {t.value!r}")
         # optional YIELD ... (accepted, ignored: full rows returned)

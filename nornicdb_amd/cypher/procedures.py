@@ -107,4 +107,101 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
         db.search.recluster(int(k) if k else None)
         return ["clusters"], [[db.search.clusters.k]]
 
+    @register("dbms.routing.getRoutingTable")
+    def _routing_table(ex, context=None, database=None):
+        # single-server topology (drivers use this for session routing)
+        addr = "localhost:7687"
+        servers = [{"addresses": [addr], "role": r}
+                   for r in ("WRITE", "READ", "ROUTE")]
+        return ["ttl", "servers"], [[300, servers]]
+
+    @register("db.schema.visualization")
+    def _schema_viz(ex):
+        from ..storage.types import Node as _N
+        labels = sorted({lb for n in db.engine.all_nodes()
+                         for lb in n.labels})
+        vnodes = {lb: _N(id=f"schema:{lb}", labels=[lb],
+                         properties={"name": lb}) for lb in labels}
+        vrels = []
+        seen = set()
+        for e in db.engine.all_edges():
+            try:
+                a = db.engine.get_node(e.start_node)
+                b = db.engine.get_node(e.end_node)
+            except Exception:
+                continue
+            for la in a.labels:
+                for lb2 in b.labels:
+                    key = (la, e.type, lb2)
+                    if key not in seen:
+                        seen.add(key)
+                        vrels.append({"type": e.type, "from": la, "to": lb2})
+        return ["nodes", "relationships"], [[list(vnodes.values()), vrels]]
+
+    @register("db.schema.nodeTypeProperties")
+    def _schema_ntp(ex):
+        from collections import defaultdict
+        props = defaultdict(set)
+        for n in db.engine.all_nodes():
+            for lb in n.labels:
+                for k, v in n.properties.items():
+                    props[(lb, k)].add(type(v).__name__)
+        return (["nodeType", "nodeLabels", "propertyName", "propertyTypes",
+                 "mandatory"],
+                [[f":`{lb}`", [lb], k, sorted(ts), False]
+                 for (lb, k), ts in sorted(props.items())])
+
+    @register("db.schema.relTypeProperties")
+    def _schema_rtp(ex):
+        from collections import defaultdict
+        props = defaultdict(set)
+        for e in db.engine.all_edges():
+            for k, v in e.properties.items():
+                props[(e.type, k)].add(type(v).__name__)
+        return (["relType", "propertyName", "propertyTypes", "mandatory"],
+                [[f":`{t}`", k, sorted(ts), False]
+                 for (t, k), ts in sorted(props.items())])
+
+    @register("db.awaitIndexes")
+    def _await_indexes(ex, timeout=300):
+        return [], []
+
+    @register("db.awaitIndex")
+    def _await_index(ex, name=None, timeout=300):
+        return [], []
+
+    @register("db.resampleIndex")
+    def _resample_index(ex, name=None):
+        return [], []
+
+    @register("db.resampleOutdatedIndexes")
+    def _resample_outdated(ex):
+        return [], []
+
+    @register("dbms.listConfig")
+    def _list_config(ex, search=None):
+        from ..utils.config import Config
+        cfg = Config()
+        rows = [[k, str(getattr(cfg, k))] for k in sorted(vars(cfg))
+                if not k.startswith("_")]
+        if search:
+            rows = [r for r in rows if search.lower() in r[0].lower()]
+        return ["name", "value"], rows
+
+    @register("dbms.info")
+    def _dbms_info(ex):
+        return ["id", "name", "creationDate"], [["nornicdb-amd", db.name, ""]]
+
+    @register("db.stats.retrieve")
+    def _stats_retrieve(ex, section="GRAPH COUNTS"):
+        from collections import Counter
+        lc = Counter()
+        for n in db.engine.all_nodes():
+            lc.update(n.labels)
+        tc = Counter(e.type for e in db.engine.all_edges())
+        data = {"nodes": db.engine.node_count(),
+                "relationships": db.engine.edge_count(),
+                "labels": dict(lc), "relTypes": dict(tc)}
+        return ["section", "data"], [[section, data]]
+
     return procs

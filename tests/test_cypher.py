@@ -648,3 +648,42 @@ class TestCypher5Builtins:
         ex = self._ex()
         assert ex.execute("RETURN isNaN(log(-1.0)), isNaN(asin(2.0))"
                           ).rows == [[True, True]]
+
+
+class TestDriverCompatProcedures:
+    """Driver-facing procedures: routing table, schema visualization,
+    db.stats, dbms.listConfig, SHOW TRANSACTIONS."""
+
+    def _db(self):
+        from nornicdb_amd.db import NornicDB
+        from nornicdb_amd.storage.memory import MemoryEngine
+        db = NornicDB(MemoryEngine(), auto_embed=False)
+        db.cypher("CREATE (a:P {name:'x'})-[:K]->(b:Q)")
+        return db
+
+    def test_routing_table(self):
+        r = self._db().cypher("CALL dbms.routing.getRoutingTable({}) "
+                              "YIELD ttl, servers RETURN ttl, servers")
+        assert r.rows[0][0] == 300
+        assert {s["role"] for s in r.rows[0][1]} == {"WRITE", "READ", "ROUTE"}
+
+    def test_schema_visualization(self):
+        r = self._db().cypher("CALL db.schema.visualization() "
+                              "YIELD nodes, relationships "
+                              "RETURN size(nodes), size(relationships)")
+        assert r.rows == [[2, 1]]
+
+    def test_node_type_properties(self):
+        r = self._db().cypher("CALL db.schema.nodeTypeProperties() "
+                              "YIELD nodeLabels, propertyName "
+                              "RETURN nodeLabels, propertyName")
+        assert [["P"], "name"] in r.rows
+
+    def test_show_transactions(self):
+        r = self._db().cypher("SHOW TRANSACTIONS")
+        assert r.rows[0][3] == "Running"
+
+    def test_stats_retrieve(self):
+        r = self._db().cypher("CALL db.stats.retrieve('GRAPH COUNTS') "
+                              "YIELD data RETURN data.nodes, data.relationships")
+        assert r.rows == [[2, 1]]
