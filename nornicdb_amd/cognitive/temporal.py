@@ -120,3 +120,58 @@ class QueryLoadTracker:
         """Back off decay cycles under load."""
         load = self.qps()
         return base * (1.0 + min(load / 10.0, 10.0))
+
+
+class RelationshipEvolution:
+    """Tracks how relationship strength evolves with co-access
+    (reference pkg/temporal/relationship_evolution.go): every co-access
+    of two connected nodes reinforces the edge; idle time decays it.
+    Strength feeds decay protection and link-prediction priors."""
+
+    def __init__(self, engine, reinforce: float = 0.1,
+                 half_life_s: float = 7 * 86400.0, now_fn=time.time):
+        self.engine = engine
+        self.reinforce = reinforce
+        self.half_life_s = half_life_s
+        self.now = now_fn
+        self._strength: Dict[str, float] = {}
+        self._last: Dict[str, float] = {}
+
+    def _decayed(self, eid: str) -> float:
+        s = self._strength.get(eid, 0.5)
+        last = self._last.get(eid)
+        if last is None:
+            return s
+        dt = max(0.0, self.now() - last)
+        return s * (0.5 ** (dt / self.half_life_s))
+
+    def record_coaccess(self, edge_id: str) -> float:
+        s = self._decayed(edge_id)
+        s = min(1.0, s + self.reinforce * (1.0 - s))
+        self._strength[edge_id] = s
+        self._last[edge_id] = self.now()
+        return s
+
+    def strength(self, edge_id: str) -> float:
+        return self._decayed(edge_id)
+
+    def evolution_class(self, edge_id: str) -> str:
+        s = self.strength(edge_id)
+        if s >= 0.8:
+            return "strengthening"
+        if s >= 0.4:
+            return "stable"
+        return "fading"
+
+    def persist(self) -> int:
+        """Write current strengths into edge properties (_strength)."""
+        n = 0
+        for eid in list(self._strength):
+            try:
+                e = self.engine.get_edge(eid)
+                e.properties["_strength"] = round(self.strength(eid), 4)
+                self.engine.update_edge(e)
+                n += 1
+            except Exception:
+                self._strength.pop(eid, None)
+        return n

@@ -285,3 +285,29 @@ class TestInferenceQCAndClusters:
         assert not ci.same_cluster("a0", "b0")
         assert ci.boost_for("a0", "a1") > 0
         assert ci.boost_for("a0", "b1") == 0.0
+
+
+class TestRelationshipEvolution:
+    """Co-access reinforcement with half-life decay (reference
+    pkg/temporal/relationship_evolution.go)."""
+
+    def test_reinforce_decay_persist(self):
+        from nornicdb_amd.cognitive.temporal import RelationshipEvolution
+        from nornicdb_amd.storage.memory import MemoryEngine
+        from nornicdb_amd.storage.types import Edge, Node
+        eng = MemoryEngine()
+        eng.create_node(Node(id="a", labels=[], properties={}))
+        eng.create_node(Node(id="b", labels=[], properties={}))
+        eng.create_edge(Edge(id="e1", type="K", start_node="a",
+                             end_node="b", properties={}))
+        t = [0.0]
+        ev = RelationshipEvolution(eng, now_fn=lambda: t[0])
+        for _ in range(10):
+            ev.record_coaccess("e1")
+        assert ev.strength("e1") > 0.8
+        assert ev.evolution_class("e1") == "strengthening"
+        t[0] = 30 * 86400  # a month idle -> fades past 4 half-lives
+        assert ev.strength("e1") < 0.1
+        assert ev.evolution_class("e1") == "fading"
+        assert ev.persist() == 1
+        assert "_strength" in eng.get_edge("e1").properties
