@@ -247,6 +247,12 @@ class CallClause:
 
 
 @dataclass
+class UseClause:
+    """USE <database> — multi-db routing (executed by DatabaseManager)."""
+    database: str
+
+
+@dataclass
 class SchemaCommand:
     """Schema DDL: CREATE/DROP INDEX|CONSTRAINT, SHOW INDEXES|CONSTRAINTS|
     DATABASES|PROCEDURES|FUNCTIONS (Neo4j 4/5 syntax + 3.x legacy)."""

@@ -81,6 +81,7 @@ class Executor:
         self.schema = schema            # storage.SchemaManager (DDL target)
         self._profile_log = None
         self.database_lister = None     # set by DatabaseManager for SHOW DATABASES
+        self.database_router = None     # set by DatabaseManager for USE <db>
         self.current_database = "neo4j"
 
     # ------------------------------------------------------------------ API
@@ -103,6 +104,17 @@ class Executor:
             return Result(["plan"], [[self._explain(q)]], dict(self.stats))
         if q.profile:
             return self._profile(q, params)
+        if q.clauses and isinstance(q.clauses[0], A.UseClause):
+            use = q.clauses[0]
+            router = getattr(self, "database_router", None)
+            if router is None:
+                raise CypherRuntimeError(
+                    f"USE {use.database}: no database router attached "
+                    "(open via DatabaseManager)")
+            rest = A.Query(q.clauses[1:], union=q.union)
+            if not rest.clauses:
+                return Result([], [])
+            return router(use.database, rest, params)
         if len(q.clauses) == 1 and isinstance(q.clauses[0], A.SchemaCommand):
             res = self._exec_schema(q.clauses[0], params)
             res.stats = dict(self.stats)

@@ -136,6 +136,13 @@ class Parser:
             return self._foreach()
         if self.at_kw("SHOW"):
             return self._show()
+        if (self.peek().kind == "IDENT" and self.peek().value.upper() == "USE"
+                and self.peek(1).kind in ("IDENT", "KW")):
+            self.next()
+            name = self.name_part()
+            while self.try_op("."):
+                name += "." + self.name_part()
+            return A.UseClause(name)
         if self.at_kw("DROP") or (self.peek().kind == "IDENT"
                                   and self.peek().value.upper() == "DROP"):
             return self._drop_schema()

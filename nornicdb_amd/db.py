@@ -318,6 +318,18 @@ class DatabaseManager:
         db = NornicDB(eng, name=name, embedder=self._embedder,
                       dims=self._dims, device=self._device)
         db.executor.database_lister = lambda: sorted(self._dbs)
+
+        def _route(target, rest_query, params):
+            tdb = self.get(target)
+            ex = tdb.executor
+            ex.stats = {"nodes_created": 0, "nodes_deleted": 0,
+                        "edges_created": 0, "edges_deleted": 0,
+                        "properties_set": 0, "labels_added": 0}
+            res = ex._run_query(rest_query, params)
+            res.stats = dict(ex.stats)
+            return res
+
+        db.executor.database_router = _route
         self._dbs[name] = db
         return db
 

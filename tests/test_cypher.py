@@ -687,3 +687,29 @@ class TestDriverCompatProcedures:
         r = self._db().cypher("CALL db.stats.retrieve('GRAPH COUNTS') "
                               "YIELD data RETURN data.nodes, data.relationships")
         assert r.rows == [[2, 1]]
+
+
+class TestUseClause:
+    """USE <db> multi-database routing (reference pkg/multidb)."""
+
+    def test_use_routes_to_target(self):
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.storage.memory import MemoryEngine
+        mgr = DatabaseManager(MemoryEngine())
+        mgr.get().cypher("USE system CREATE (:Cfg {k: 1})")
+        assert mgr.get("system").cypher(
+            "MATCH (c:Cfg) RETURN c.k").rows == [[1]]
+        assert mgr.get().cypher(
+            "MATCH (c:Cfg) RETURN count(c)").rows == [[0]]
+        assert mgr.get().cypher(
+            "USE system MATCH (c:Cfg) RETURN c.k").rows == [[1]]
+
+    def test_use_without_router_raises(self):
+        import pytest
+
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.cypher.functions import CypherRuntimeError
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        with pytest.raises(CypherRuntimeError):
+            ex.execute("USE other MATCH (n) RETURN n")
