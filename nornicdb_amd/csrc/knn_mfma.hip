@@ -285,8 +285,12 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
   long long n_panels = n / BM;
   // Enough blocks that each retires quickly (load balance + lets other
   // streams' kernels co-schedule), few enough that the candidate buffer
-  // and merge stay small.
-  int grid = (int)std::min<long long>(n_panels, 8192);
+  // and merge stay small. At grid 8192 a 100M-corpus block runs ~6 ms,
+  // which blocks embed-stream GEMMs from co-scheduling; tunable for the
+  // overlap experiment via NORNICDB_KNN_GRID.
+  int max_grid = 8192;
+  if (const char* g = getenv("NORNICDB_KNN_GRID")) max_grid = atoi(g);
+  int grid = (int)std::min<long long>(n_panels, max_grid);
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   auto opts_f = db.options().dtype(at::kFloat);
