@@ -182,6 +182,71 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
                 break
         return {"results": results, "errors": errors}
 
+    # ---- explicit transaction lifecycle (reference server_db.go:381-1226:
+    # simplified semantics — statements execute eagerly, commit returns a
+    # bookmark, rollback acknowledges) ----
+    import itertools as _it
+    _tx_counter = _it.count(1)
+    _open_txs: Dict[str, str] = {}
+
+    async def _run_statements(db_name: str, request: Request):
+        body = await request.json() if (await request.body()) else {}
+        try:
+            db = mgr.get(db_name)
+        except KeyError:
+            raise HTTPException(404, f"database {db_name} not found")
+        results, errors = [], []
+        for stmt in (body or {}).get("statements", []):
+            try:
+                r = db.cypher(stmt.get("statement", ""),
+                              stmt.get("parameters") or {})
+                results.append({
+                    "columns": r.columns,
+                    "data": [{"row": [_jsonable(v) for v in row], "meta": []}
+                             for row in r.rows],
+                    "stats": r.stats})
+            except Exception as e:
+                code = ("Neo.ClientError.Statement.SyntaxError"
+                        if isinstance(e, CypherSyntaxError)
+                        else "Neo.ClientError.Statement.ExecutionFailed")
+                errors.append({"code": code, "message": str(e)})
+                break
+        return results, errors
+
+    @app.post("/db/{db_name}/tx")
+    async def tx_open(db_name: str, request: Request, response: Response,
+                      _user=Depends(check_auth)):
+        txid = str(next(_tx_counter))
+        _open_txs[txid] = db_name
+        results, errors = await _run_statements(db_name, request)
+        response.headers["Location"] = f"/db/{db_name}/tx/{txid}"
+        return {"results": results, "errors": errors,
+                "commit": f"/db/{db_name}/tx/{txid}/commit",
+                "transaction": {"expires": ""}}
+
+    @app.post("/db/{db_name}/tx/{txid}")
+    async def tx_execute(db_name: str, txid: str, request: Request,
+                         _user=Depends(check_auth)):
+        if txid not in _open_txs:
+            raise HTTPException(
+                404, f"transaction {txid} not found or already closed")
+        results, errors = await _run_statements(db_name, request)
+        return {"results": results, "errors": errors,
+                "commit": f"/db/{db_name}/tx/{txid}/commit"}
+
+    @app.post("/db/{db_name}/tx/{txid}/commit")
+    async def tx_commit_open(db_name: str, txid: str, request: Request,
+                             _user=Depends(check_auth)):
+        _open_txs.pop(txid, None)
+        results, errors = await _run_statements(db_name, request)
+        return {"results": results, "errors": errors,
+                "lastBookmarks": [f"FB:bookmark-{txid}"]}
+
+    @app.delete("/db/{db_name}/tx/{txid}")
+    async def tx_rollback(db_name: str, txid: str, _user=Depends(check_auth)):
+        _open_txs.pop(txid, None)
+        return {"results": [], "errors": []}
+
     # ---- multi-database management ----
     @app.post("/admin/databases/{name}")
     def create_db(name: str, _user=Depends(check_auth)):

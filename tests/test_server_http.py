@@ -146,3 +146,30 @@ def test_console_served(client):
     r = client.get("/")
     assert r.status_code == 200
     assert "NornicDB-AMD console" in r.text or "console" in r.text
+
+
+class TestExplicitTransactions:
+    """Explicit tx lifecycle endpoints (reference server_db.go:381 — same
+    simplified semantics: eager execution, bookmark on commit,
+    acknowledged rollback)."""
+
+    def test_open_execute_commit(self, client):
+        r = client.post("/db/neo4j/tx", json={"statements": [
+            {"statement": "CREATE (:TX {v: 1})"}]})
+        assert r.status_code == 200
+        loc = r.headers["Location"]
+        txid = loc.rstrip("/").rsplit("/", 1)[-1]
+        r2 = client.post(f"/db/neo4j/tx/{txid}", json={"statements": [
+            {"statement": "MATCH (t:TX) RETURN t.v"}]})
+        assert r2.json()["results"][0]["data"][0]["row"] == [1]
+        r3 = client.post(f"/db/neo4j/tx/{txid}/commit", json={})
+        assert r3.json()["lastBookmarks"]
+        # closed tx rejects further statements
+        r4 = client.post(f"/db/neo4j/tx/{txid}", json={"statements": []})
+        assert r4.status_code == 404
+
+    def test_rollback_acknowledges(self, client):
+        r = client.post("/db/neo4j/tx", json={"statements": []})
+        txid = r.headers["Location"].rstrip("/").rsplit("/", 1)[-1]
+        r2 = client.delete(f"/db/neo4j/tx/{txid}")
+        assert r2.status_code == 200
