@@ -1944,4 +1944,69 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
         aid = a.id if isinstance(a, Node) else a
         return ["value"], [[len(_hops(aid, rel_type, max_hops, 1))]]
 
+    # -------------------- apoc.export/import graphml --------------------
+    @register("apoc.export.graphml.all")
+    def _ex_graphml(ex, config=None):
+        import xml.sax.saxutils as _sx
+        lines = ['<?xml version="1.0" encoding="UTF-8"?>',
+                 '<graphml xmlns="http://graphml.graphdrawing.org/xmlns">',
+                 '<graph id="G" edgedefault="directed">']
+        for n in eng.all_nodes():
+            labels = _sx.escape(":".join(n.labels))
+            lines.append(f'<node id="{_sx.escape(n.id)}" labels=":{labels}">')
+            for k, v in n.properties.items():
+                if k.startswith("_"):
+                    continue
+                lines.append(f'<data key="{_sx.escape(str(k))}">'
+                             f'{_sx.escape(str(v))}</data>')
+            lines.append("</node>")
+        for e in eng.all_edges():
+            lines.append(f'<edge id="{_sx.escape(e.id)}" '
+                         f'source="{_sx.escape(e.start_node)}" '
+                         f'target="{_sx.escape(e.end_node)}" '
+                         f'label="{_sx.escape(e.type)}">')
+            for k, v in e.properties.items():
+                if k.startswith("_"):
+                    continue
+                lines.append(f'<data key="{_sx.escape(str(k))}">'
+                             f'{_sx.escape(str(v))}</data>')
+            lines.append("</edge>")
+        lines.append("</graph></graphml>")
+        return ["data", "nodes", "relationships"], [[
+            "\n".join(lines), eng.node_count(), eng.edge_count()]]
+
+    @register("apoc.import.graphml")
+    def _im_graphml(ex, data, config=None):
+        import xml.etree.ElementTree as _ET
+        ns = {"g": "http://graphml.graphdrawing.org/xmlns"}
+        root = _ET.fromstring(str(data))
+        n_nodes = n_edges = 0
+        for el in root.iter():
+            tag = el.tag.split("}")[-1]
+            if tag == "node":
+                labels = [lb for lb in
+                          (el.get("labels", "").lstrip(":").split(":"))
+                          if lb]
+                props = {d.get("key"): d.text for d in el
+                         if d.tag.split("}")[-1] == "data"}
+                try:
+                    eng.create_node(Node(id=el.get("id") or new_id("n"),
+                                         labels=labels, properties=props))
+                    n_nodes += 1
+                except Exception:
+                    pass
+            elif tag == "edge":
+                props = {d.get("key"): d.text for d in el
+                         if d.tag.split("}")[-1] == "data"}
+                try:
+                    eng.create_edge(Edge(
+                        id=el.get("id") or new_id("e"),
+                        type=el.get("label", "RELATED"),
+                        start_node=el.get("source"),
+                        end_node=el.get("target"), properties=props))
+                    n_edges += 1
+                except Exception:
+                    pass
+        return ["nodes", "relationships"], [[n_nodes, n_edges]]
+
     return procs

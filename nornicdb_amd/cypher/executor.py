@@ -333,6 +333,12 @@ class Executor:
                      "default", "home"],
                     [[n, "standard", "read-write", "primary", "online",
                       n == "neo4j", n == "neo4j"] for n in names])
+            if c.kind == "settings":
+                proc = self.procedures.get("dbms.listconfig")
+                if proc:
+                    cols, rows = proc(self)
+                    return Result(cols, rows)
+                return Result(["name", "value"], [])
             if c.kind == "transactions":
                 # single current transaction (our executor is synchronous)
                 return Result(

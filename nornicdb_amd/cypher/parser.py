@@ -365,7 +365,8 @@ class Parser:
                  "PROCEDURE": "procedures", "PROCEDURES": "procedures",
                  "FUNCTION": "functions", "FUNCTIONS": "functions",
                  "TRANSACTION": "transactions",
-                 "TRANSACTIONS": "transactions"}
+                 "TRANSACTIONS": "transactions",
+                 "SETTING": "settings", "SETTINGS": "settings"}
         if word not in kinds:
             raise CypherSyntaxError(f"cannot SHOW {t.value!r}")
         # optional YIELD ... (accepted, ignored: full rows returned)

@@ -237,3 +237,18 @@ class TestApocBatch3:
         tree = r.rows[0][0]
         assert tree and tree[0]["name"] == "root"
         assert tree[0]["has"][0]["name"] == "leaf"
+
+
+class TestGraphML:
+    def test_roundtrip(self):
+        a = NornicDB(MemoryEngine(), auto_embed=False)
+        a.cypher("CREATE (x:G {name:'x'})-[:L {w: '2'}]->(y:H)")
+        data = a.cypher("CALL apoc.export.graphml.all() YIELD data "
+                        "RETURN data").rows[0][0]
+        assert "<graphml" in data
+        b = NornicDB(MemoryEngine(), auto_embed=False)
+        r = b.cypher("CALL apoc.import.graphml($d) YIELD nodes, relationships "
+                     "RETURN nodes, relationships", {"d": data})
+        assert r.rows == [[2, 1]]
+        assert b.cypher("MATCH (g:G)-[l:L]->(h:H) RETURN g.name, l.w"
+                        ).rows == [["x", "2"]]
