@@ -170,3 +170,26 @@ class TestFusedDecoder:
         graph_out = gd.generate(prompt.clone(), max_new_tokens=8,
                                 temperature=0.0)
         assert fused_out == graph_out
+
+
+@pytest.mark.gpu
+class TestManagerGPUIntegration:
+    """HeimdallManager on GPU: FusedDecoder is selected and generates;
+    streaming callback path works end-to-end."""
+
+    def test_manager_uses_fused_decoder(self):
+        from nornicdb_amd.heimdall import HeimdallManager
+        from nornicdb_amd.models.heimdall import FusedDecoder, HeimdallConfig
+        mgr = HeimdallManager(config=HeimdallConfig(num_layers=4),
+                              device="cuda", max_tokens_default=8)
+        assert isinstance(mgr._graphed, FusedDecoder)
+        out = mgr.generate("hello world", max_tokens=8)
+        assert out and mgr.stats["generations"] == 1
+
+    def test_manager_stream(self):
+        from nornicdb_amd.heimdall import HeimdallManager
+        from nornicdb_amd.models.heimdall import HeimdallConfig
+        mgr = HeimdallManager(config=HeimdallConfig(num_layers=4),
+                              device="cuda", max_tokens_default=8)
+        toks = list(mgr.generate_stream("stream me", max_tokens=6))
+        assert 1 <= len(toks) <= 6
