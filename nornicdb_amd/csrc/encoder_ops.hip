@@ -164,7 +164,7 @@ __global__ void k_mean_pool_l2norm(const unsigned short* __restrict__ x,
 #define FA_D 64
 #define FA_KT 64
 #define FA_QT 64
-#define FA_PSTRIDE 72
+#define FA_PSTRIDE 64   // measured: 64 beats 72 by 15% (208 vs 181 TF; scripts/attn_pad.hip sweep)
 #define FA_VSTRIDE 72
 
 __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
