@@ -228,19 +228,23 @@ __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
     }
-    // ---- stage V transposed: thread t covers k = t % 64, d-block t/64*8 --
+    // ---- stage V transposed: gather 8 consecutive k at fixed d, ONE
+    // b128 LDS write per octet. The strided global reads hit L1/L2 (the
+    // V tile is 8 KB); swapping the cost off the LDS write port was
+    // +5% within-probe on top of the stride fix (scripts/attn_pad.hip:
+    // 216 -> 227 TF, refchecked).
     {
       const unsigned short* vp = V + b * v_bs + hoff;
-      int kk = threadIdx.x & 63;
-      int db = (threadIdx.x >> 6) * 16;  // 4 groups x 16 d each
+      int dd = threadIdx.x & 63;          // d fixed per thread
+      int k8 = (threadIdx.x >> 6) * 16;   // two k-octets per thread
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        int dd = db + half * 8;
-        short8v v = *reinterpret_cast<const short8v*>(
-            vp + (long long)(k0 + kk) * v_ss + dd);
+        int kk = k8 + half * 8;
+        short8v o;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          sVT[(dd + j) * FA_VSTRIDE + kk] = (unsigned short)v[j];
+          o[j] = (short)vp[(long long)(k0 + kk + j) * v_ss + dd];
+        *reinterpret_cast<short8v*>(sVT + dd * FA_VSTRIDE + kk) = o;
       }
     }
     __syncthreads();
