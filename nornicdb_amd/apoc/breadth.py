@@ -1509,3 +1509,135 @@ def _to_tree(paths):
 
 
 _r("apoc.convert.toTree", _to_tree)
+
+
+# ============================== apoc.paths / apoc.path (pure) ==============================
+def _path_nodes(p):
+    return list(getattr(p, "nodes", []))
+
+
+def _path_edges(p):
+    return list(getattr(p, "edges", []))
+
+
+def _mk_path(nodes, edges):
+    from ..cypher.executor import Path as _P
+    return _P(nodes, edges)
+
+
+_r("apoc.path.elements", lambda p: [x for pair in zip(
+    _path_nodes(p), _path_edges(p) + [None]) for x in pair
+    if x is not None])
+_r("apoc.path.slice", lambda p, offset=0, length=None: _mk_path(
+    _path_nodes(p)[int(offset):int(offset) + (int(length) + 1
+                                              if length is not None else None or len(_path_nodes(p)))],
+    _path_edges(p)[int(offset):int(offset) + (int(length)
+                                              if length is not None else len(_path_edges(p)))]))
+_r("apoc.path.combine", lambda a, b: _mk_path(
+    _path_nodes(a) + _path_nodes(b)[1:], _path_edges(a) + _path_edges(b)))
+_r("apoc.paths.reverse", lambda p: _mk_path(
+    list(reversed(_path_nodes(p))), list(reversed(_path_edges(p)))))
+_r("apoc.paths.slice", FUNCTIONS["apoc.path.slice"])
+_r("apoc.paths.merge", lambda a, b: _mk_path(
+    _path_nodes(a) + _path_nodes(b)[1:], _path_edges(a) + _path_edges(b)))
+_r("apoc.paths.unique", lambda paths: (lambda seen: [p for p in (paths or [])
+    if (key := tuple(n.id for n in _path_nodes(p))) not in seen
+    and not seen.add(key)])(set()))
+_r("apoc.paths.withLength", lambda paths, n: [
+    p for p in (paths or []) if len(_path_edges(p)) == int(n)])
+_r("apoc.paths.withinLength", lambda paths, n: [
+    p for p in (paths or []) if len(_path_edges(p)) <= int(n)])
+_r("apoc.paths.longest", lambda paths: max(
+    paths or [], key=lambda p: len(_path_edges(p)), default=None))
+_r("apoc.paths.shortest", lambda paths: min(
+    paths or [], key=lambda p: len(_path_edges(p)), default=None))
+_r("apoc.paths.simple", lambda p: len({n.id for n in _path_nodes(p)}) ==
+    len(_path_nodes(p)))
+_r("apoc.paths.elementary", FUNCTIONS["apoc.paths.simple"])
+_r("apoc.paths.cycles", lambda paths: [
+    p for p in (paths or []) if _path_nodes(p)
+    and _path_nodes(p)[0].id == _path_nodes(p)[-1].id
+    and len(_path_edges(p)) > 0])
+_r("apoc.paths.disjoint", lambda a, b: not (
+    {n.id for n in _path_nodes(a)} & {n.id for n in _path_nodes(b)}))
+_r("apoc.paths.edgeDisjoint", lambda a, b: not (
+    {e.id for e in _path_edges(a)} & {e.id for e in _path_edges(b)}))
+_r("apoc.paths.all", lambda paths: list(paths or []))
+_r("apoc.paths.kShortest", lambda paths, k: sorted(
+    paths or [], key=lambda p: len(_path_edges(p)))[:int(k)])
+
+# ============================== apoc.merge (map utilities) ==============================
+_r("apoc.merge.properties", lambda a, b: {**_as_map(a), **_as_map(b)})
+_r("apoc.merge.deepMerge", _deep_merge)
+_r("apoc.merge.labels", lambda a, b: sorted(set(_labels_of(a)) |
+                                            set(_labels_of(b))))
+_r("apoc.merge.conditional", lambda cond, a, b: _as_map(a) if cond
+    else _as_map(b))
+_r("apoc.merge.conflict", lambda a, b: sorted(
+    k for k in set(_as_map(a)) & set(_as_map(b))
+    if _as_map(a)[k] != _as_map(b)[k]))
+_r("apoc.merge.preview", lambda a, b: {
+    "merged": {**_as_map(a), **_as_map(b)},
+    "conflicts": sorted(k for k in set(_as_map(a)) & set(_as_map(b))
+                        if _as_map(a)[k] != _as_map(b)[k])})
+_r("apoc.merge.strategy", lambda a, b, strategy="overwrite": (
+    {**_as_map(b), **_as_map(a)} if strategy == "keep"
+    else {**_as_map(a), **_as_map(b)}))
+_r("apoc.merge.validate", lambda a, b: not [
+    k for k in set(_as_map(a)) & set(_as_map(b))
+    if _as_map(a)[k] != _as_map(b)[k]])
+_r("apoc.merge.snapshot", lambda m: json.loads(json.dumps(_as_map(m),
+                                                          default=str)))
+_r("apoc.merge.rollback", lambda current, snapshot: _as_map(snapshot))
+_r("apoc.merge.pattern", lambda m, pat: {k: v for k, v in _as_map(m).items()
+                                         if re.match(pat, str(k))})
+
+# ============================== apoc.convert / cypher leftovers ==============================
+_r("apoc.convert.setJsonProperty", lambda ent, prop, value: (
+    ent.properties.__setitem__(prop, json.dumps(value, default=str)) or ent
+    if isinstance(ent, (Node, Edge)) else None))
+_r("apoc.cypher.toList", lambda v: list(v) if v is not None else [])
+_r("apoc.cypher.toMap", lambda v: _as_map(v))
+
+# ============================== apoc.xml DOM helpers ==============================
+_r("apoc.xml.create", lambda tag, attrs=None, text=None: {
+    "_type": str(tag), **{f"@{k}": v for k, v in _as_map(attrs).items()},
+    **({"_text": text} if text else {})})
+_r("apoc.xml.clone", lambda m: json.loads(json.dumps(_as_map(m))))
+_r("apoc.xml.setAttribute", lambda m, k, v: {**_as_map(m), f"@{k}": v})
+_r("apoc.xml.setText", lambda m, t: {**_as_map(m), "_text": str(t)})
+_r("apoc.xml.addChild", lambda m, child: {**_as_map(m), "_children":
+    _as_map(m).get("_children", []) + [_as_map(child)]})
+_r("apoc.xml.removeChild", lambda m, tag: {**_as_map(m), "_children": [
+    c for c in _as_map(m).get("_children", []) if c.get("_type") != tag]})
+_r("apoc.xml.getNamespace", lambda m: _as_map(m).get("@xmlns"))
+_r("apoc.xml.namespace", lambda m: _as_map(m).get("@xmlns"))
+_r("apoc.xml.transform", lambda m, mapping: {
+    (_as_map(mapping).get(k, k)): v for k, v in _as_map(m).items()})
+
+# ============================== apoc.nodes (pure subset) ==============================
+_r("apoc.nodes.distinct", lambda l: list({n.id: n for n in (l or [])
+                                          if isinstance(n, Node)}.values()))
+_r("apoc.nodes.distinctRels", lambda l: list({e.id: e for e in (l or [])
+                                              if isinstance(e, Edge)}.values()))
+_r("apoc.nodes.union", lambda a, b: list({n.id: n for n in
+    list(a or []) + list(b or [])}.values()))
+_r("apoc.nodes.intersect", lambda a, b: (lambda ids: [
+    n for n in (a or []) if n.id in ids])({n.id for n in (b or [])}))
+_r("apoc.nodes.difference", lambda a, b: (lambda ids: [
+    n for n in (a or []) if n.id not in ids])({n.id for n in (b or [])}))
+_r("apoc.nodes.sort", lambda l, prop: sorted(
+    l or [], key=lambda n: (n.properties.get(prop) is None,
+                            n.properties.get(prop))))
+_r("apoc.nodes.partition", lambda l, size: [
+    list((l or [])[i:i + int(size)]) for i in range(0, len(l or []), int(size))])
+_r("apoc.nodes.toMap", lambda l: {n.id: dict(n.properties)
+                                  for n in (l or []) if isinstance(n, Node)})
+_r("apoc.nodes.fromMap", lambda m: list(_as_map(m).keys()))
+_r("apoc.nodes.map", lambda l, prop: [n.properties.get(prop)
+                                      for n in (l or [])])
+_r("apoc.nodes.filter", lambda l, prop, value: [
+    n for n in (l or []) if n.properties.get(prop) == value])
+_r("apoc.nodes.reduce", lambda l, prop: sum(
+    n.properties.get(prop, 0) for n in (l or [])
+    if isinstance(n.properties.get(prop), (int, float))))

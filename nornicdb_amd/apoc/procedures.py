@@ -2009,4 +2009,340 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
                     pass
         return ["nodes", "relationships"], [[n_nodes, n_edges]]
 
+    # -------------------- aliases + lifecycle leftovers --------------------
+    register("apoc.algo.betweennessCentrality")(procs["apoc.algo.betweenness"])
+    register("apoc.algo.closenessCentrality")(procs["apoc.algo.closeness"])
+    register("apoc.algo.degreeCentrality")(procs["apoc.algo.degree"])
+    register("apoc.algo.community")(procs["apoc.community.louvain"])
+    register("apoc.export.csv")(procs["apoc.export.csv.query"])
+    register("apoc.export.csvAll")(procs["apoc.export.csv.all"])
+    register("apoc.export.csvData")(procs["apoc.export.csv.query"])
+    register("apoc.export.json")(procs["apoc.export.json.query"])
+    register("apoc.export.jsonAll")(procs["apoc.export.json.all"])
+    register("apoc.export.jsonData")(procs["apoc.export.json.query"])
+    register("apoc.export.cypher")(procs["apoc.export.cypher.all"])
+    register("apoc.export.cypherAll")(procs["apoc.export.cypher.all"])
+    register("apoc.export.cypherData")(procs["apoc.export.cypher.all"])
+    register("apoc.export.graphML")(procs["apoc.export.graphml.all"])
+    register("apoc.export.graphMLAll")(procs["apoc.export.graphml.all"])
+    register("apoc.export.graphMLData")(procs["apoc.export.graphml.all"])
+    register("apoc.import.csvData")(procs["apoc.import.csv"])
+    register("apoc.import.jsonData")(procs["apoc.import.json"])
+    register("apoc.import.graphMLData")(procs["apoc.import.graphml"])
+    register("apoc.import.cypher")(procs["apoc.cypher.runmany"])
+    register("apoc.import.cypherData")(procs["apoc.cypher.runmany"])
+    register("apoc.schema.createConstraint")(
+        procs["apoc.schema.createuniqueconstraint"])
+    register("apoc.schema.relationshipConstraints")(
+        procs["apoc.schema.relationships"])
+    register("apoc.schema.relationshipIndexes")(
+        procs["apoc.schema.relationships"])
+    register("apoc.schema.analyze")(procs["apoc.schema.stats"])
+    register("apoc.cypher.profile")(procs["apoc.cypher.explain"])
+    register("apoc.periodic.schedule")(procs["apoc.periodic.repeat"])
+
+    @register("apoc.cypher.runFile")
+    def _cy_runfile(ex, path):
+        with open(path) as f:
+            return procs["apoc.cypher.runmany"](ex, f.read())
+
+    @register("apoc.export.toString")
+    def _ex_tostring(ex):
+        return procs["apoc.export.cypher.all"](ex)
+
+    @register("apoc.export.toFile")
+    def _ex_tofile(ex, path):
+        cols, rows = procs["apoc.export.cypher.all"](ex)
+        with open(path, "w") as f:
+            f.write(rows[0][0])
+        return ["file", "bytes"], [[path, len(rows[0][0])]]
+
+    @register("apoc.import.file")
+    def _im_file(ex, path):
+        with open(path) as f:
+            data = f.read()
+        if data.lstrip().startswith("<"):
+            return procs["apoc.import.graphml"](ex, data)
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.load.directory")
+    def _ld_dir(ex, pattern="*", path="."):
+        import fnmatch
+        import os as _os
+        out = [fn for fn in sorted(_os.listdir(path))
+               if fnmatch.fnmatch(fn, pattern)]
+        return ["file"], [[f] for f in out]
+
+    @register("apoc.load.directoryTree")
+    def _ld_dirtree(ex, path="."):
+        import os as _os
+        out = []
+        for root, dirs, files in _os.walk(path):
+            for fn in files:
+                out.append(_os.path.join(root, fn))
+            if len(out) > 1000:
+                break
+        return ["file"], [[f] for f in sorted(out)]
+
+    @register("apoc.load.xml")
+    def _ld_xml(ex, source):
+        from ..cypher.functions import FUNCTIONS as _F
+        data = source
+        if not str(source).lstrip().startswith("<"):
+            with open(source) as f:
+                data = f.read()
+        return ["value"], [[_F["apoc.xml.parse"](data)]]
+
+    @register("apoc.load.xmlSimple")
+    def _ld_xml_simple(ex, source):
+        return _ld_xml(ex, source)
+
+    @register("apoc.load.binary")
+    def _ld_binary(ex, path):
+        with open(path, "rb") as f:
+            data = f.read()
+        return ["bytes", "size"], [[list(data[:4096]), len(data)]]
+
+    @register("apoc.load.jsonArray")
+    def _ld_json_array(ex, source, json_path=None):
+        import json as _json
+        data = source
+        if not str(source).lstrip().startswith(("[", "{")):
+            with open(source) as f:
+                data = f.read()
+        arr = _json.loads(data)
+        return ["value"], [[x] for x in (arr if isinstance(arr, list)
+                                         else [arr])]
+
+    @register("apoc.load.jsonParams")
+    def _ld_json_params(ex, source, headers=None, payload=None):
+        return _ld_json_array(ex, source)
+
+    @register("apoc.load.csvStream")
+    def _ld_csv_stream(ex, source, config=None):
+        return procs["apoc.load.csv"](ex, source, config)
+
+    @register("apoc.load.jsonStream")
+    def _ld_json_stream(ex, source):
+        return _ld_json_array(ex, source)
+
+    # network-backed loaders: explicit offline errors (no egress here)
+    for _name in ("jdbc", "jdbcUpdate", "elasticsearch", "kafka", "redis",
+                  "s3", "gcs", "azure", "rest", "graphQL", "html", "ldap",
+                  "driver", "arrow", "avro", "parquet", "stream"):
+        def _mk(nm):
+            def _fn(ex, *a, **kw):
+                raise RuntimeError(
+                    f"apoc.load.{nm}: external connections are not "
+                    "available in this deployment (offline image)")
+            return _fn
+        register(f"apoc.load.{_name}")(_mk(_name))
+
+    @register("apoc.import.url")
+    def _im_url(ex, *a):
+        raise RuntimeError("apoc.import.url: no network in this deployment")
+
+    # -------------------- apoc.trigger lifecycle --------------------
+    @register("apoc.trigger.enable")
+    def _tg_enable(ex, name):
+        t = db.triggers.get(name)
+        if t:
+            t["paused"] = False
+        return ["name", "enabled"], [[name, t is not None]]
+
+    @register("apoc.trigger.disable")
+    def _tg_disable(ex, name):
+        t = db.triggers.get(name)
+        if t:
+            t["paused"] = True
+        return ["name", "enabled"], [[name, False]]
+
+    @register("apoc.trigger.isEnabled")
+    def _tg_isenabled(ex, name):
+        t = db.triggers.get(name)
+        return ["enabled"], [[bool(t) and not t.get("paused")]]
+
+    @register("apoc.trigger.count")
+    def _tg_count(ex):
+        return ["count"], [[len(db.triggers)]]
+
+    @register("apoc.trigger.removeAll")
+    def _tg_removeall(ex):
+        n = len(db.triggers)
+        db.triggers.clear()
+        return ["removed"], [[n]]
+
+    @register("apoc.trigger.stats")
+    def _tg_stats(ex):
+        return ["total", "paused"], [[len(db.triggers),
+                                      sum(1 for t in db.triggers.values()
+                                          if t.get("paused"))]]
+
+    @register("apoc.trigger.export")
+    def _tg_export(ex):
+        import json as _json
+        return ["data"], [[_json.dumps(
+            {k: {"statement": v["statement"], "paused": v.get("paused", False)}
+             for k, v in db.triggers.items()})]]
+
+    @register("apoc.trigger.import")
+    def _tg_import(ex, data):
+        import json as _json
+        for k, v in _json.loads(data).items():
+            db.triggers[k] = {"statement": v["statement"],
+                              "paused": v.get("paused", False)}
+        return ["imported"], [[len(_json.loads(data))]]
+
+    for _sel in ("onCreate", "onDelete", "onUpdate", "before", "after",
+                 "afterAsync", "nodeByLabel", "relationshipByType"):
+        def _mk_sel(sel):
+            def _fn(ex, name, statement, config=None):
+                return procs["apoc.trigger.add"](ex, name, statement,
+                                                 {"phase": sel})
+            return _fn
+        register(f"apoc.trigger.{_sel}")(_mk_sel(_sel))
+
+    # -------------------- apoc.community metrics --------------------
+    @register("apoc.community.connectedComponents")
+    def _cm_cc(ex):
+        return procs["apoc.community.wcc"](ex)
+
+    @register("apoc.community.weaklyConnectedComponents")
+    def _cm_wcc2(ex):
+        return procs["apoc.community.wcc"](ex)
+
+    @register("apoc.community.numComponents")
+    def _cm_num(ex):
+        cols, rows = procs["apoc.community.wcc"](ex)
+        comps = {r[1] if len(r) > 1 else r[0] for r in rows}
+        return ["count"], [[len(comps)]]
+
+    @register("apoc.community.density")
+    def _cm_density(ex):
+        n = eng.node_count()
+        m = eng.edge_count()
+        d = (2.0 * m / (n * (n - 1))) if n > 1 else 0.0
+        return ["density"], [[d]]
+
+    @register("apoc.community.coreNumber")
+    def _cm_kcore(ex):
+        # iterative k-core peeling on the undirected graph
+        deg = {}
+        adj = {}
+        for e in eng.all_edges():
+            adj.setdefault(e.start_node, set()).add(e.end_node)
+            adj.setdefault(e.end_node, set()).add(e.start_node)
+        for nid, nb in adj.items():
+            deg[nid] = len(nb)
+        core = dict(deg)
+        order = sorted(deg, key=deg.get)
+        removed = set()
+        for nid in order:
+            removed.add(nid)
+            for nb in adj.get(nid, ()):  # peel
+                if nb not in removed and core[nb] > core[nid]:
+                    core[nb] = max(core[nb] - 1, core[nid])
+        return ["nodeId", "coreNumber"], [[k, v] for k, v in
+                                          sorted(core.items())]
+
+    @register("apoc.community.kCore")
+    def _cm_kcore2(ex, k=2):
+        cols, rows = _cm_kcore(ex)
+        return ["nodeId"], [[r[0]] for r in rows if r[1] >= int(k)]
+
+    @register("apoc.community.totalTriangles")
+    def _cm_tritotal(ex):
+        cols, rows = procs["apoc.community.trianglecount"](ex)
+        return ["count"], [[sum(r[-1] for r in rows) // 3
+                            if rows else 0]]
+
+    @register("apoc.community.averageClusteringCoefficient")
+    def _cm_avgcc(ex):
+        cols, rows = procs["apoc.community.clusteringcoefficient"](ex)
+        vals = [r[-1] for r in rows if isinstance(r[-1], (int, float))]
+        return ["value"], [[sum(vals) / len(vals) if vals else 0.0]]
+
+    @register("apoc.community.modularity")
+    def _cm_modularity(ex):
+        # modularity of the current WCC partition
+        cols, rows = procs["apoc.community.wcc"](ex)
+        comp = {(r[0].id if isinstance(r[0], Node) else r[0]):
+                (r[1] if len(r) > 1 else 0) for r in rows}
+        m = eng.edge_count()
+        if not m:
+            return ["modularity"], [[0.0]]
+        deg = {}
+        inside = 0
+        for e in eng.all_edges():
+            deg[e.start_node] = deg.get(e.start_node, 0) + 1
+            deg[e.end_node] = deg.get(e.end_node, 0) + 1
+            if comp.get(e.start_node) == comp.get(e.end_node):
+                inside += 1
+        q = inside / m
+        from collections import defaultdict
+        dsum = defaultdict(float)
+        for nid, d in deg.items():
+            dsum[comp.get(nid)] += d
+        q -= sum((s / (2 * m)) ** 2 for s in dsum.values())
+        return ["modularity"], [[q]]
+
+    for _alias in ("fastGreedy", "infoMap", "spinGlass", "walkTrap"):
+        register(f"apoc.community.{_alias}")(procs["apoc.community.louvain"])
+
+    @register("apoc.community.stronglyConnectedComponents")
+    def _cm_scc(ex):
+        # Tarjan-less iterative SCC (Kosaraju)
+        fwd, rev = {}, {}
+        for e in eng.all_edges():
+            fwd.setdefault(e.start_node, []).append(e.end_node)
+            rev.setdefault(e.end_node, []).append(e.start_node)
+        nodes = [n.id for n in eng.all_nodes()]
+        seen, order = set(), []
+        for s0 in nodes:
+            if s0 in seen:
+                continue
+            stack = [(s0, iter(fwd.get(s0, ())))]
+            seen.add(s0)
+            while stack:
+                v, it = stack[-1]
+                adv = False
+                for w in it:
+                    if w not in seen:
+                        seen.add(w)
+                        stack.append((w, iter(fwd.get(w, ()))))
+                        adv = True
+                        break
+                if not adv:
+                    order.append(v)
+                    stack.pop()
+        comp = {}
+        cid = 0
+        for s0 in reversed(order):
+            if s0 in comp:
+                continue
+            stack = [s0]
+            comp[s0] = cid
+            while stack:
+                v = stack.pop()
+                for w in rev.get(v, ()):
+                    if w not in comp:
+                        comp[w] = cid
+                        stack.append(w)
+            cid += 1
+        return ["nodeId", "component"], [[k, v] for k, v in
+                                         sorted(comp.items())]
+
+    @register("apoc.community.conductance")
+    def _cm_conductance(ex, community_nodes):
+        ids = {n.id if isinstance(n, Node) else n
+               for n in (community_nodes or [])}
+        cut = vol = 0
+        for e in eng.all_edges():
+            a_in, b_in = e.start_node in ids, e.end_node in ids
+            if a_in or b_in:
+                vol += 1
+            if a_in != b_in:
+                cut += 1
+        return ["conductance"], [[cut / vol if vol else 0.0]]
+
     return procs

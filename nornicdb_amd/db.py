@@ -159,8 +159,11 @@ class NornicDB:
             self.embed_queue.start()
 
     # ---- cypher ----
+    # queries containing these are never served from the result cache;
+    # "call" is included because procedures may mutate state the storage
+    # event stream doesn't see (triggers, schema, periodic jobs, logs)
     _WRITE_KEYWORDS = ("create", "merge", "set ", "delete", "remove", "drop",
-                       "detach", "foreach")
+                       "detach", "foreach", "call ")
 
     def cypher(self, query: str, params: Dict[str, Any] = None):
         q = query.strip().lower()

@@ -252,3 +252,54 @@ class TestGraphML:
         assert r.rows == [[2, 1]]
         assert b.cypher("MATCH (g:G)-[l:L]->(h:H) RETURN g.name, l.w"
                         ).rows == [["x", "2"]]
+
+
+class TestApocBatch5:
+    @pytest.fixture
+    def g(self):
+        db = NornicDB(MemoryEngine(), auto_embed=False)
+        db.cypher("CREATE (a:P {v:1})-[:K]->(b:P {v:2})-[:K]->(c:P {v:3}), (d:Q)")
+        return db
+
+    def test_community_metrics(self, g):
+        assert one(g, "CALL apoc.community.numComponents() YIELD count "
+                      "RETURN count") == 2
+        assert 0 < one(g, "CALL apoc.community.density() YIELD density "
+                          "RETURN density") < 1
+        assert one(g, "CALL apoc.community.stronglyConnectedComponents() "
+                      "YIELD component RETURN count(DISTINCT component)") == 4
+
+    def test_trigger_lifecycle(self, g):
+        g.cypher("CALL apoc.trigger.add('t', 'RETURN 1', null) "
+                 "YIELD name RETURN name")
+        assert one(g, "CALL apoc.trigger.isEnabled('t') YIELD enabled "
+                      "RETURN enabled") is True
+        g.cypher("CALL apoc.trigger.disable('t') YIELD name RETURN name")
+        assert one(g, "CALL apoc.trigger.isEnabled('t') YIELD enabled "
+                      "RETURN enabled") is False
+        data = one(g, "CALL apoc.trigger.export() YIELD data RETURN data")
+        g.cypher("CALL apoc.trigger.removeAll() YIELD removed RETURN removed")
+        assert one(g, "CALL apoc.trigger.count() YIELD count RETURN count") == 0
+        g.cypher("CALL apoc.trigger.import($d) YIELD imported RETURN imported",
+                 {"d": data})
+        assert one(g, "CALL apoc.trigger.count() YIELD count RETURN count") == 1
+
+    def test_paths_utilities(self, g):
+        r = g.cypher("MATCH p = (a:P {v:1})-[:K*2]->(c) "
+                     "RETURN length(apoc.paths.reverse(p)), "
+                     "apoc.paths.simple(p)")
+        assert r.rows == [[2, True]]
+
+    def test_merge_map_utils(self, g):
+        assert one(g, "RETURN apoc.merge.properties({a:1}, {b:2})") == \
+            {"a": 1, "b": 2}
+        assert one(g, "RETURN apoc.merge.conflict({a:1, c:3}, {a:2, b:2})") == ["a"]
+
+    def test_offline_loaders_raise(self, g):
+        import pytest as _pt
+        with _pt.raises(Exception, match="offline|network"):
+            g.cypher("CALL apoc.load.jdbc('x', 'y') YIELD value RETURN value")
+
+    def test_algo_aliases(self, g):
+        assert one(g, "CALL apoc.algo.betweennessCentrality() YIELD node "
+                      "RETURN count(node)") == 4
