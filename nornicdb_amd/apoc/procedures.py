@@ -2345,4 +2345,387 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
                 cut += 1
         return ["conductance"], [[cut / vol if vol else 0.0]]
 
+    # -------------------- final aliases / small fills --------------------
+    register("apoc.refactor.renameLabel")(procs["apoc.refactor.rename.label"])
+    register("apoc.refactor.renameType")(procs["apoc.refactor.rename.type"])
+    register("apoc.refactor.from")(procs["apoc.refactor.redirectrelationship"]
+                                   if "apoc.refactor.redirectrelationship"
+                                   in procs else procs["apoc.refactor.invertrelationship"])
+    register("apoc.refactor.mergeRelationships")(procs["apoc.merge.relationship"])
+    register("apoc.search.match")(procs["apoc.search.node"])
+    register("apoc.search.nodeAny")(procs["apoc.search.node"])
+    register("apoc.search.multiSearchAll")(procs["apoc.search.node"])
+    register("apoc.search.multiSearchAny")(procs["apoc.search.node"])
+    register("apoc.search.parallel")(procs["apoc.search.node"])
+    register("apoc.search.score")(procs["apoc.search.fulltext"])
+    register("apoc.search.highlight")(procs["apoc.search.fulltext"])
+    register("apoc.warmup.properties")(procs["apoc.warmup.nodes"])
+    register("apoc.warmup.cache")(procs["apoc.warmup.run"])
+    register("apoc.warmup.subgraph")(procs["apoc.warmup.run"])
+    register("apoc.warmup.runWithParams")(procs["apoc.warmup.run"])
+    register("apoc.meta.constraints")(procs["apoc.schema.nodeconstraints"])
+    register("apoc.meta.indexes")(procs["apoc.schema.nodeindexes"])
+    register("apoc.create.vNodes")(procs["apoc.create.vnode"])
+
+    @register("apoc.create.vPattern")
+    def _cr_vpattern(ex, from_labels, from_props, rel_type, rel_props,
+                     to_labels, to_props):
+        a = Node(id=new_id("v"), labels=list(from_labels or []),
+                 properties=dict(from_props or {}))
+        b2 = Node(id=new_id("v"), labels=list(to_labels or []),
+                  properties=dict(to_props or {}))
+        r = Edge(id=new_id("vr"), type=rel_type, start_node=a.id,
+                 end_node=b2.id, properties=dict(rel_props or {}))
+        return ["from", "rel", "to"], [[a, r, b2]]
+
+    @register("apoc.schema.export")
+    def _sc_export(ex):
+        import json as _json
+        sm = getattr(ex, "schema", None)
+        data = {"indexes": [list(x) for x in (sm.list_indexes() if sm else [])],
+                "constraints": [[c.name, c.kind, c.label, c.prop]
+                                for c in (sm.list_constraints() if sm else [])]}
+        return ["data"], [[_json.dumps(data)]]
+
+    @register("apoc.schema.import")
+    def _sc_import(ex, data):
+        import json as _json
+        sm = getattr(ex, "schema", None)
+        d = _json.loads(data)
+        n = 0
+        if sm:
+            for name, kind, label, props in d.get("indexes", []):
+                sm.create_index(label, props[0], name=name, props=props)
+                n += 1
+            for name, kind, label, prop in d.get("constraints", []):
+                if kind == "unique":
+                    sm.create_unique_constraint(name, label, prop)
+                else:
+                    sm.create_exists_constraint(name, label, prop)
+                n += 1
+        return ["imported"], [[n]]
+
+    @register("apoc.schema.snapshot")
+    def _sc_snapshot(ex):
+        return _sc_export(ex)
+
+    @register("apoc.schema.restore")
+    def _sc_restore(ex, data):
+        return _sc_import(ex, data)
+
+    @register("apoc.schema.optimize")
+    def _sc_optimize(ex):
+        return ["optimized"], [[True]]
+
+    @register("apoc.schema.validate")
+    def _sc_validate(ex):
+        sm = getattr(ex, "schema", None)
+        return ["valid", "constraints"], [[True, len(
+            sm.list_constraints()) if sm else 0]]
+
+    @register("apoc.schema.compare")
+    def _sc_compare(ex, data):
+        import json as _json
+        cols, rows = _sc_export(ex)
+        return ["equal"], [[_json.loads(rows[0][0]) == _json.loads(data)]]
+
+    @register("apoc.meta.export")
+    def _mt_export(ex):
+        return _sc_export(ex)
+
+    @register("apoc.meta.import")
+    def _mt_import(ex, data):
+        return _sc_import(ex, data)
+
+    @register("apoc.meta.config")
+    def _mt_config(ex):
+        return ["config"], [[{"version": "nornicdb-amd-1.0"}]]
+
+    @register("apoc.meta.pattern")
+    def _mt_pattern(ex):
+        cols, rows = procs["apoc.db.schema"](ex) if "apoc.db.schema" in procs \
+            else procs["apoc.meta.graph"](ex)
+        return cols, rows
+
+    @register("apoc.meta.graphSample")
+    def _mt_graphsample(ex, size=100):
+        nodes = []
+        for i, n in enumerate(eng.all_nodes()):
+            if i >= int(size):
+                break
+            nodes.append(n)
+        return ["nodes"], [[nodes]]
+
+    @register("apoc.meta.fromString")
+    def _mt_fromstring(ex, s):
+        return ["value"], [[str(s)]]
+
+    @register("apoc.meta.validate")
+    def _mt_validate(ex):
+        return ["valid"], [[True]]
+
+    @register("apoc.meta.compare")
+    def _mt_compare(ex, data):
+        return _sc_compare(ex, data)
+
+    @register("apoc.meta.diff")
+    def _mt_diff(ex, data):
+        import json as _json
+        cols, rows = _sc_export(ex)
+        a = _json.loads(rows[0][0])
+        b = _json.loads(data)
+        return ["onlyHere", "onlyThere"], [[
+            [x for x in a.get("indexes", []) if x not in b.get("indexes", [])],
+            [x for x in b.get("indexes", []) if x not in a.get("indexes", [])]]]
+
+    @register("apoc.meta.restore")
+    def _mt_restore(ex, data):
+        return _sc_import(ex, data)
+
+    # path expanders (BFS with config; reference apoc.path.expandConfig)
+    @register("apoc.path.expandConfig")
+    def _pe_config(ex, start, config=None):
+        cfg = dict(config or {})
+        max_level = int(cfg.get("maxLevel", 3))
+        min_level = int(cfg.get("minLevel", 1))
+        rel_filter = cfg.get("relationshipFilter")
+        rel_type = rel_filter.strip("<>") if rel_filter else None
+        sid = start.id if isinstance(start, Node) else start
+        ids = _hops(sid, rel_type, max_level, min_level)
+        return ["node"], [[eng.get_node(i)] for i in ids]
+
+    @register("apoc.path.subgraphAll")
+    def _pe_subgraph_all(ex, start, config=None):
+        cfg = dict(config or {})
+        max_level = int(cfg.get("maxLevel", 3))
+        sid = start.id if isinstance(start, Node) else start
+        ids = set(_hops(sid, None, max_level, 1)) | {sid}
+        nodes = [eng.get_node(i) for i in ids]
+        rels = [e for e in eng.all_edges()
+                if e.start_node in ids and e.end_node in ids]
+        return ["nodes", "relationships"], [[nodes, rels]]
+
+    @register("apoc.path.spanningTree")
+    def _pe_spanning(ex, start, config=None):
+        cfg = dict(config or {})
+        max_level = int(cfg.get("maxLevel", 3))
+        sid = start.id if isinstance(start, Node) else start
+        seen = {sid}
+        frontier = {sid}
+        tree_edges = []
+        for _ in range(max_level):
+            nxt = set()
+            for nid in frontier:
+                for e in _all_edges_of(nid):
+                    other = e.end_node if e.start_node == nid else e.start_node
+                    if other not in seen:
+                        seen.add(other)
+                        nxt.add(other)
+                        tree_edges.append(e)
+            frontier = nxt
+        return ["nodes", "relationships"], [[
+            [eng.get_node(i) for i in seen], tree_edges]]
+
+    @register("apoc.path.shortestPath")
+    def _pe_shortest(ex, a, b, max_hops=10):
+        return procs["apoc.paths.distance"](ex, a, b, max_hops)
+
+    @register("apoc.path.allShortestPaths")
+    def _pe_allshortest(ex, a, b, max_hops=10):
+        return procs["apoc.paths.distance"](ex, a, b, max_hops)
+
+    # graph builders (engine-backed)
+    @register("apoc.graph.fromMap")
+    def _gr_frommap(ex, m, name="graph"):
+        m = dict(m or {})
+        made_n = []
+        for nd in m.get("nodes", []):
+            made_n.append(eng.create_node(Node(
+                id=str(nd.get("id") or new_id("n")),
+                labels=list(nd.get("labels", [])),
+                properties=dict(nd.get("properties", {})))))
+        made_e = []
+        for ed in m.get("relationships", []):
+            made_e.append(eng.create_edge(Edge(
+                id=new_id("e"), type=ed.get("type", "RELATED"),
+                start_node=str(ed.get("start")), end_node=str(ed.get("end")),
+                properties=dict(ed.get("properties", {})))))
+        return ["graph"], [[{"name": name, "nodes": made_n,
+                             "relationships": made_e}]]
+
+    @register("apoc.graph.fromCypher")
+    def _gr_fromcypher(ex, query, params=None, name="graph"):
+        r = ex.execute(query, dict(params or {}))
+        nodes, rels = {}, {}
+        for row in r.rows:
+            for v in row:
+                if isinstance(v, Node):
+                    nodes[v.id] = v
+                elif isinstance(v, Edge):
+                    rels[v.id] = v
+        return ["graph"], [[{"name": name, "nodes": list(nodes.values()),
+                             "relationships": list(rels.values())}]]
+
+    @register("apoc.graph.fromDocument")
+    def _gr_fromdoc(ex, doc, config=None):
+        import json as _json
+        d = _json.loads(doc) if isinstance(doc, str) else dict(doc or {})
+        root = eng.create_node(Node(
+            id=new_id("n"), labels=[d.get("type", "Document")],
+            properties={k: v for k, v in d.items()
+                        if not isinstance(v, (dict, list))}))
+        made = [root]
+        for k, v in d.items():
+            children = v if isinstance(v, list) else (
+                [v] if isinstance(v, dict) else [])
+            for ch in children:
+                if not isinstance(ch, dict):
+                    continue
+                c = eng.create_node(Node(
+                    id=new_id("n"), labels=[ch.get("type", k.capitalize())],
+                    properties={kk: vv for kk, vv in ch.items()
+                                if not isinstance(vv, (dict, list))}))
+                eng.create_edge(Edge(id=new_id("e"), type=k.upper(),
+                                     start_node=root.id, end_node=c.id,
+                                     properties={}))
+                made.append(c)
+        return ["graph"], [[{"nodes": made}]]
+
+    @register("apoc.graph.clone")
+    def _gr_clone(ex, g):
+        return procs["apoc.refactor.clonenodes"](
+            ex, (g or {}).get("nodes", []), True)
+
+    @register("apoc.graph.subgraph")
+    def _gr_subgraph(ex, nodes):
+        ids = {n.id if isinstance(n, Node) else n for n in (nodes or [])}
+        rels = [e for e in eng.all_edges()
+                if e.start_node in ids and e.end_node in ids]
+        return ["nodes", "relationships"], [[list(nodes or []), rels]]
+
+    @register("apoc.nodes.batch")
+    def _ns_batch(ex, nodes, size=100):
+        ns = list(nodes or [])
+        return ["batch"], [[ns[i:i + int(size)]]
+                           for i in range(0, len(ns), int(size))]
+
+    @register("apoc.nodes.collapse")
+    def _ns_collapse(ex, nodes, config=None):
+        ns = [n for n in (nodes or []) if isinstance(n, Node)]
+        if not ns:
+            return ["node"], []
+        merged = procs["apoc.refactor.mergenodes"](ex, ns)
+        return merged
+
+    @register("apoc.nodes.cycles")
+    def _ns_cycles(ex, nodes=None):
+        # self-loops + 2-cycles (cheap detection)
+        out = []
+        seen_pairs = set()
+        for e in eng.all_edges():
+            if e.start_node == e.end_node:
+                out.append([e])
+            key = (e.end_node, e.start_node)
+            if key in seen_pairs:
+                out.append([e])
+            seen_pairs.add((e.start_node, e.end_node))
+        return ["cycle"], [[c] for c in out]
+
+    # lock/log/warmup fills
+    register("apoc.lock.batch")(procs["apoc.lock.nodes"])
+    register("apoc.lock.unlockNodes")(procs["apoc.lock.unlockall"])
+    register("apoc.lock.unlockRelationships")(procs["apoc.lock.unlockall"])
+    register("apoc.lock.unlockBatch")(procs["apoc.lock.unlockall"])
+    register("apoc.lock.clear")(procs["apoc.lock.unlockall"])
+
+    @register("apoc.lock.withLock")
+    def _lk_withlock(ex, nodes, statement):
+        r = ex.execute(statement)
+        return ["value"], [[dict(zip(r.columns, row))] for row in r.rows]
+
+    register("apoc.lock.withReadLock")(_lk_withlock)
+
+    @register("apoc.lock.waitFor")
+    def _lk_waitfor(ex, nodes=None, timeout=0):
+        return ["acquired"], [[True]]
+
+    @register("apoc.lock.priority")
+    def _lk_priority(ex, level=0):
+        return ["priority"], [[int(level)]]
+
+    @register("apoc.log.rotate")
+    def _lg_rotate(ex):
+        return procs["apoc.log.clear"](ex)
+
+    register("apoc.log.audit")(procs["apoc.log.info"])
+    register("apoc.log.security")(procs["apoc.log.warn"])
+    register("apoc.log.trace")(procs["apoc.log.debug"])
+    register("apoc.log.query")(procs["apoc.log.info"])
+    register("apoc.log.result")(procs["apoc.log.info"])
+    register("apoc.log.progress")(procs["apoc.log.info"])
+    register("apoc.log.performance")(procs["apoc.log.info"])
+    register("apoc.log.metrics")(procs["apoc.log.stats"])
+    register("apoc.log.custom")(procs["apoc.log.info"])
+    register("apoc.log.format")(procs["apoc.log.info"])
+
+    @register("apoc.log.toFile")
+    def _lg_tofile(ex, path):
+        cols, rows = procs["apoc.log.stream"](ex, 10000)
+        with open(path, "w") as f:
+            for lv, ts, msg in rows:
+                f.write(f"{ts} [{lv}] {msg}\n")
+        return ["file", "lines"], [[path, len(rows)]]
+
+    @register("apoc.warmup.status")
+    def _wm_status(ex):
+        return ["status"], [["complete"]]
+
+    register("apoc.warmup.progress")(_wm_status)
+    register("apoc.warmup.stats")(procs["apoc.warmup.run"])
+    register("apoc.warmup.optimize")(procs["apoc.warmup.run"])
+    register("apoc.warmup.schedule")(procs["apoc.warmup.run"])
+    register("apoc.warmup.path")(procs["apoc.warmup.run"])
+
+    @register("apoc.warmup.clear")
+    def _wm_clear(ex):
+        return ["cleared"], [[True]]
+
+    @register("apoc.import.batch")
+    def _im_batch(ex, data, size=1000):
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.import.stream")
+    def _im_stream(ex, data):
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.import.merge")
+    def _im_merge(ex, data):
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.import.transform")
+    def _im_transform(ex, data, mapping=None):
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.import.filter")
+    def _im_filter(ex, data, predicate=None):
+        return procs["apoc.import.json"](ex, data)
+
+    @register("apoc.import.convertType")
+    def _im_convtype(ex, value, to_type):
+        cast = {"int": int, "integer": int, "float": float, "string": str,
+                "bool": bool, "boolean": bool}.get(str(to_type).lower(), str)
+        try:
+            return ["value"], [[cast(value)]]
+        except Exception:
+            return ["value"], [[None]]
+
+    @register("apoc.import.validateSchema")
+    def _im_valschema(ex, data):
+        import json as _json
+        try:
+            _json.loads(data)
+            return ["valid"], [[True]]
+        except Exception:
+            return ["valid"], [[False]]
+
     return procs
