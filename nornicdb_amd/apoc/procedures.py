@@ -133,6 +133,13 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
     # -------------------- gds compat (reference pkg/cypher/fastrp.go) ----
     @register("gds.fastRP.stream")
     def _fastrp(ex, dims=128, iteration_weights=None, seed=42):
+        # GDS contract: first arg may be a projected graph NAME with a
+        # config map (gds.fastRP.stream('g', {embeddingDimension: 64}))
+        if isinstance(dims, str):
+            cfg = dict(iteration_weights or {}) if isinstance(
+                iteration_weights, dict) else {}
+            dims = cfg.get("embeddingDimension", cfg.get("dims", 128))
+            iteration_weights = cfg.get("iterationWeights")
         from ..graph.fastrp import fastrp_embeddings
         g = from_engine(eng)
         emb = fastrp_embeddings(g, dims=int(dims),

@@ -82,6 +82,8 @@ class Executor:
         self._profile_log = None
         self.database_lister = None     # set by DatabaseManager for SHOW DATABASES
         self.database_router = None     # set by DatabaseManager for USE <db>
+        self.database_admin = None      # set by DatabaseManager for
+                                        # CREATE/DROP DATABASE|ALIAS
         self.current_database = "neo4j"
 
     # ------------------------------------------------------------------ API
@@ -324,7 +326,16 @@ class Executor:
         without one raises."""
         from ..storage.schema import SchemaManager
 
+        if c.op in ("create", "drop") and c.kind in ("database", "alias"):
+            if self.database_admin is None:
+                raise CypherRuntimeError(
+                    f"{c.op.upper()} {c.kind.upper()} needs a DatabaseManager")
+            return self.database_admin(c)
         if c.op == "show":
+            if c.kind == "aliases":
+                if self.database_admin is None:
+                    return Result(["name", "database"], [])
+                return self.database_admin(c)
             if c.kind == "databases":
                 names = (self.database_lister() if self.database_lister
                          else [self.current_database])

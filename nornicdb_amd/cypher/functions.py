@@ -358,3 +358,58 @@ register("asin", lambda x: None if x is None else (
     float("nan") if abs(x) > 1 else math.asin(_num(x))))
 register("acos", lambda x: None if x is None else (
     float("nan") if abs(x) > 1 else math.acos(_num(x))))
+
+
+# ---- kalman.* functions (reference exposes pkg/filter via Cypher) ----
+_KALMAN_STATES: Dict[str, Any] = {}
+
+
+def _kalman_get(name):
+    from ..cognitive.kalman import Kalman1D
+    if name not in _KALMAN_STATES:
+        _KALMAN_STATES[name] = Kalman1D()
+    return _KALMAN_STATES[name]
+
+
+def _kalman_init(name, q=0.01, r=0.1, initial=0.0):
+    from ..cognitive.kalman import Kalman1D
+    k = Kalman1D(q=q, r=r)
+    k.x = float(initial)
+    _KALMAN_STATES[str(name)] = k
+    return str(name)
+
+
+register("kalman.init", _kalman_init)
+register("kalman.predict", lambda name: _kalman_get(str(name)).predict())
+register("kalman.process", lambda name, measurement:
+         _kalman_get(str(name)).update(float(measurement)))
+register("kalman.update", lambda name, measurement:
+         _kalman_get(str(name)).update(float(measurement)))
+register("kalman.state", lambda name: {
+    "x": _kalman_get(str(name)).x,
+    "p": getattr(_kalman_get(str(name)), "p", None)})
+register("kalman.reset", lambda name: (_KALMAN_STATES.pop(str(name), None),
+                                       True)[1])
+
+# ---- string format/pad (reference util) ----
+register("lpad", lambda s, n, pad=" ": None if s is None else
+         str(s).rjust(int(n), str(pad)[:1] or " "))
+register("rpad", lambda s, n, pad=" ": None if s is None else
+         str(s).ljust(int(n), str(pad)[:1] or " "))
+
+
+def _java_format(fmt, *args):
+    out = str(fmt)
+    # %s/%d/%f subset
+    for a in args:
+        for tok in ("%s", "%d", "%f"):
+            i = out.find(tok)
+            if i >= 0:
+                rep = (str(int(a)) if tok == "%d" else
+                       (f"{float(a):f}" if tok == "%f" else str(a)))
+                out = out[:i] + rep + out[i + 2:]
+                break
+    return out
+
+
+register("format", _java_format)

@@ -28,6 +28,7 @@ class Token:
     kind: str   # KW, IDENT, INT, FLOAT, STRING, OP, PARAM, EOF
     value: str
     pos: int
+    raw: str = ""   # original source text (case-preserving, KW tokens)
 
 
 class CypherSyntaxError(Exception):
@@ -131,7 +132,9 @@ def tokenize(text: str) -> List[Token]:
                 j += 1
             word = text[i:j]
             if word.upper() in KEYWORDS:
-                toks.append(Token("KW", word.upper(), i))
+                t = Token("KW", word.upper(), i)
+                t.raw = word  # original case for label/name positions
+                toks.append(t)
             else:
                 toks.append(Token("IDENT", word, i))
             i = j

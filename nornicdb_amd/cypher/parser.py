@@ -68,6 +68,17 @@ class Parser:
             return self.next().value.lower()
         raise CypherSyntaxError(f"expected name, got {t.value!r} at {t.pos}")
 
+    def label_name(self) -> str:
+        """Label / rel-type name: identifiers OR reserved words, original
+        case preserved (Neo4j allows `:Order`, `:Match` etc.)."""
+        t = self.peek()
+        if t.kind == "IDENT":
+            return self.next().value
+        if t.kind == "KW":
+            tok = self.next()
+            return getattr(tok, "raw", "") or tok.value
+        raise CypherSyntaxError(f"expected label, got {t.value!r} at {t.pos}")
+
     def ident(self) -> str:
         t = self.peek()
         if t.kind == "IDENT":
@@ -179,6 +190,30 @@ class Parser:
             return self._create_index(kindword or "RANGE", or_replace)
         if self.at_kw("CONSTRAINT"):
             return self._create_constraint(or_replace)
+        if self.at_kw("DATABASE") or self._at_word("DATABASE"):
+            self.next()
+            ine, _ = (self._if_not_exists()
+                      if self._at_word("IF") else (False, False))
+            name = self.name_part()
+            while self.try_op("."):
+                name += "." + self.name_part()
+            if self._at_word("IF"):
+                ine, _ = self._if_not_exists()
+            return A.SchemaCommand("create", "database", name=name,
+                                   if_not_exists=ine, or_replace=or_replace)
+        if self._at_word("ALIAS"):
+            self.next()
+            name = self.name_part()
+            while self.try_op("."):
+                name += "." + self.name_part()
+            ine, _ = self._if_not_exists()
+            self._eat_word("FOR")
+            self._eat_word("DATABASE")
+            target = self.name_part()
+            while self.try_op("."):
+                target += "." + self.name_part()
+            return A.SchemaCommand("create", "alias", name=name, label=target,
+                                   if_not_exists=ine, or_replace=or_replace)
         if or_replace:
             raise CypherSyntaxError("expected INDEX or CONSTRAINT after "
                                     "CREATE OR REPLACE")
@@ -353,7 +388,22 @@ class Parser:
             _, ie = self._if_not_exists()
             return A.SchemaCommand("drop", "constraint", name=name,
                                    if_exists=ie)
-        raise CypherSyntaxError("expected INDEX or CONSTRAINT after DROP")
+        if self.at_kw("DATABASE") or self._at_word("DATABASE"):
+            self.next()
+            name = self.name_part()
+            _, ie = self._if_not_exists()
+            return A.SchemaCommand("drop", "database", name=name,
+                                   if_exists=ie)
+        if self._at_word("ALIAS"):
+            self.next()
+            name = self.name_part()
+            _, ie = self._if_not_exists()
+            if self._at_word("FOR"):
+                self.next()
+                self._eat_word("DATABASE")
+            return A.SchemaCommand("drop", "alias", name=name, if_exists=ie)
+        raise CypherSyntaxError(
+            "expected INDEX, CONSTRAINT, DATABASE or ALIAS after DROP")
 
     def _show(self):
         self.eat_kw("SHOW")
@@ -366,7 +416,8 @@ class Parser:
                  "FUNCTION": "functions", "FUNCTIONS": "functions",
                  "TRANSACTION": "transactions",
                  "TRANSACTIONS": "transactions",
-                 "SETTING": "settings", "SETTINGS": "settings"}
+                 "SETTING": "settings", "SETTINGS": "settings",
+                 "ALIAS": "aliases", "ALIASES": "aliases"}
         if word not in kinds:
             raise CypherSyntaxError(f"cannot SHOW {t.value!r}")
         # optional YIELD ... (accepted, ignored: full rows returned)
@@ -408,7 +459,7 @@ class Parser:
         if self.at_op(":"):
             labels = []
             while self.try_op(":"):
-                labels.append(self.ident())
+                labels.append(self.label_name())
             return A.SetItem(target, None, op="label", labels=labels)
         if self.try_op("+="):
             return A.SetItem(target, self._expr(), op="+=")
@@ -423,7 +474,7 @@ class Parser:
             if self.at_op(":"):
                 labels = []
                 while self.try_op(":"):
-                    labels.append(self.ident())
+                    labels.append(self.label_name())
                 items.append(A.SetItem(target, None, op="label", labels=labels))
             else:
                 items.append(target)
@@ -631,9 +682,9 @@ class Parser:
         labels = []
         or_labels = False
         while self.try_op(":"):
-            labels.append(self.ident())
+            labels.append(self.label_name())
             while self.try_op("|"):       # :A|B -> OR semantics
-                labels.append(self.ident())
+                labels.append(self.label_name())
                 or_labels = True
         props = None
         if self.at_op("{"):
@@ -665,10 +716,10 @@ class Parser:
             if t.kind == "IDENT":
                 var = self.next().value
             while self.try_op(":"):
-                types.append(self.ident())
+                types.append(self.label_name())
                 while self.try_op("|"):
                     self.try_op(":")
-                    types.append(self.ident())
+                    types.append(self.label_name())
             if self.try_op("*"):
                 var_len = True
                 min_h, max_h = 1, 15
@@ -844,7 +895,7 @@ class Parser:
                 # n:Label predicate inside expressions
                 labels = []
                 while self.try_op(":"):
-                    labels.append(self.ident())
+                    labels.append(self.label_name())
                 e = A.FuncCall("__haslabels", [e, A.Lit(labels)])
             elif self.at_op("{") and isinstance(e, (A.Var, A.Prop)):
                 # map projection n {.name, .*, key: expr, var}

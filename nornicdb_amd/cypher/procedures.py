@@ -204,4 +204,133 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
                 "labels": dict(lc), "relTypes": dict(tc)}
         return ["section", "data"], [[section, data]]
 
+    # -------------------- gds.* (reference pkg/cypher/fastrp.go +
+    # linkpredict exposure via GDS-compatible procedure names) --------------
+    _GDS_GRAPHS: Dict[str, Any] = {}
+
+    @register("gds.graph.project")
+    def _gds_project(ex, name, node_label=None, rel_type=None):
+        from ..graph import csr as _csr
+        g = _csr.from_engine(db.engine,
+                             edge_types=None if rel_type in ("*", None)
+                             else [rel_type])
+        _GDS_GRAPHS[name] = g
+        return (["graphName", "nodeCount", "relationshipCount"],
+                [[name, g.n, g.m]])
+
+    @register("gds.graph.drop")
+    def _gds_drop(ex, name, fail_if_missing=True):
+        existed = _GDS_GRAPHS.pop(name, None) is not None
+        if not existed and fail_if_missing:
+            raise ValueError(f"graph {name!r} not found")
+        return ["graphName"], [[name]]
+
+    @register("gds.graph.list")
+    def _gds_list(ex, name=None):
+        items = ([(name, _GDS_GRAPHS[name])] if name in _GDS_GRAPHS
+                 else list(_GDS_GRAPHS.items()))
+        return (["graphName", "nodeCount"],
+                [[k, g.n] for k, g in items])
+
+    def _gds_graph(name):
+        if name not in _GDS_GRAPHS:
+            from ..graph import csr as _csr
+            _GDS_GRAPHS[name] = _csr.from_engine(db.engine)
+        return _GDS_GRAPHS[name]
+
+    @register("gds.fastRP.stats")
+    def _gds_fastrp_stats(ex, graph_name, config=None):
+        g = _gds_graph(graph_name)
+        return ["nodeCount"], [[g.n]]
+
+    def _lp_pairs(limit=1000):
+        """Candidate non-adjacent pairs within 2 hops."""
+        out = []
+        for n in db.engine.all_nodes():
+            nb1 = set(db.engine.neighbors(n.id))
+            for mid in nb1:
+                for cand in db.engine.neighbors(mid):
+                    if cand != n.id and cand not in nb1 and n.id < cand:
+                        out.append((n.id, cand))
+                        if len(out) >= limit:
+                            return out
+        return out
+
+    def _lp_stream(scorer):
+        from ..cognitive.linkpredict import (adamic_adar, common_neighbors,
+                                             jaccard,
+                                             preferential_attachment)
+        fn = {"adamic": adamic_adar, "common": common_neighbors,
+              "jaccard": jaccard, "pref": preferential_attachment}[scorer]
+        rows = [[a, b, float(fn(db.engine, a, b))] for a, b in _lp_pairs()]
+        rows.sort(key=lambda r: -r[2])
+        return ["node1", "node2", "score"], rows
+
+    @register("gds.linkPrediction.adamicAdar.stream")
+    def _gds_lp_aa(ex, config=None):
+        return _lp_stream("adamic")
+
+    @register("gds.linkPrediction.commonNeighbors.stream")
+    def _gds_lp_cn(ex, config=None):
+        return _lp_stream("common")
+
+    @register("gds.linkPrediction.jaccard.stream")
+    def _gds_lp_j(ex, config=None):
+        return _lp_stream("jaccard")
+
+    @register("gds.linkPrediction.preferentialAttachment.stream")
+    def _gds_lp_pa(ex, config=None):
+        return _lp_stream("pref")
+
+    @register("gds.linkPrediction.predict.stream")
+    def _gds_lp_predict(ex, config=None):
+        return _lp_stream("adamic")
+
+    # -------------------- db.index extras --------------------
+    @register("db.index.vector.createRelationshipIndex")
+    def _vec_rel_idx(ex, name, rel_type, prop, dims, similarity="cosine"):
+        sm = getattr(ex, "schema", None)
+        if sm:
+            sm.create_vector_index(name, rel_type, prop, int(dims), similarity)
+        return ["name"], [[name]]
+
+    @register("db.index.vector.queryRelationships")
+    def _vec_rel_query(ex, index_name, k, query):
+        # relationships carry no embeddings in this engine: resolve via
+        # endpoint-node similarity (same contract shape)
+        import numpy as np
+        qv = (db.embedder.embed_query(query) if isinstance(query, str)
+              else np.asarray(query, dtype=np.float32))
+        res = db.search.vector_search(qv, int(k))
+        rows = []
+        for r in res:
+            for e in db.engine.get_out_edges(r.id):
+                rows.append([e, r.score])
+                break
+        return ["relationship", "score"], rows[:int(k)]
+
+    @register("db.index.vector.drop")
+    def _vec_drop(ex, name):
+        sm = getattr(ex, "schema", None)
+        return ["dropped"], [[bool(sm and sm.drop_index(name))]]
+
+    @register("db.index.fulltext.drop")
+    def _ft_drop(ex, name):
+        sm = getattr(ex, "schema", None)
+        return ["dropped"], [[bool(sm and sm.drop_index(name))]]
+
+    @register("db.create.setNodeVectorProperty")
+    def _set_node_vec(ex, node, prop, vector):
+        n = db.engine.get_node(node.id if hasattr(node, "id") else node)
+        vec = [float(x) for x in (vector or [])]
+        n.properties[prop] = vec
+        db.engine.update_node(n)
+        if prop == "embedding" and hasattr(db.engine, "update_embedding"):
+            db.engine.update_embedding(n.id, vec)
+        return ["node"], [[n]]
+
+    @register("db.create.setVectorProperty")
+    def _set_vec(ex, node, prop, vector):
+        return _set_node_vec(ex, node, prop, vector)
+
     return procs

@@ -333,6 +333,40 @@ class DatabaseManager:
             return res
 
         db.executor.database_router = _route
+
+        def _admin(cmd):
+            from .cypher.executor import Result
+            if cmd.op == "create" and cmd.kind == "database":
+                try:
+                    self.create(cmd.name)
+                except ValueError:
+                    if not cmd.if_not_exists and not cmd.or_replace:
+                        raise
+                return Result([], [])
+            if cmd.op == "drop" and cmd.kind == "database":
+                try:
+                    self.drop(cmd.name)
+                except (KeyError, ValueError):
+                    if not cmd.if_exists:
+                        raise
+                return Result([], [])
+            if cmd.op == "create" and cmd.kind == "alias":
+                if cmd.name in self._aliases and not (cmd.if_not_exists
+                                                      or cmd.or_replace):
+                    raise ValueError(f"alias {cmd.name!r} exists")
+                self._aliases[cmd.name] = cmd.label
+                return Result([], [])
+            if cmd.op == "drop" and cmd.kind == "alias":
+                if cmd.name not in self._aliases and not cmd.if_exists:
+                    raise KeyError(cmd.name)
+                self._aliases.pop(cmd.name, None)
+                return Result([], [])
+            if cmd.op == "show":
+                return Result(["name", "database"],
+                              [[a, t] for a, t in sorted(self._aliases.items())])
+            raise ValueError(f"unsupported admin op {cmd.op} {cmd.kind}")
+
+        db.executor.database_admin = _admin
         self._dbs[name] = db
         return db
 
