@@ -2735,4 +2735,34 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
         except Exception:
             return ["valid"], [[False]]
 
+    register("apoc.algo.labelPropagation")(procs["apoc.community.labelpropagation"])
+    register("apoc.algo.louvain")(procs["apoc.community.louvain"])
+    register("apoc.algo.wcc")(procs["apoc.community.wcc"])
+
+    def _node_arg(x):
+        """Accept Node, id string, or {id: ...} map."""
+        if isinstance(x, Node):
+            return x.id
+        if isinstance(x, dict):
+            return str(x.get("id"))
+        return str(x)
+
+    _orig_spanning = procs["apoc.path.spanningtree"]
+
+    @register("apoc.path.spanningTree")
+    def _pe_spanning2(ex, start, config=None):
+        return _orig_spanning(ex, _node_arg(start), config)
+
+    _orig_subnodes = procs.get("apoc.path.subgraphnodes")
+    if _orig_subnodes is not None:
+        @register("apoc.path.subgraphNodes")
+        def _pe_subnodes2(ex, start, config=None):
+            return _orig_subnodes(ex, _node_arg(start), config)
+
+    _orig_suball = procs["apoc.path.subgraphall"]
+
+    @register("apoc.path.subgraphAll")
+    def _pe_suball2(ex, start, config=None):
+        return _orig_suball(ex, _node_arg(start), config)
+
     return procs

@@ -166,6 +166,24 @@ class Parser:
         pats = [self._pattern_path()]
         while self.try_op(","):
             pats.append(self._pattern_path())
+        # planner hints: USING INDEX/SCAN/JOIN ... — accepted and ignored
+        # (our executor picks indexes itself)
+        while self._at_word("USING"):
+            self.next()
+            t = self.next()  # INDEX | SCAN | JOIN | RANGE etc.
+            if t.value.upper() == "JOIN":
+                self._eat_word("ON")
+                self.ident()
+            else:
+                if t.value.upper() in ("INDEX", "SCAN") and self._at_word("SEEK"):
+                    self.next()
+                self.ident()          # variable
+                if self.try_op(":"):
+                    self.label_name()
+                    if self.try_op("("):
+                        while not self.at_op(")"):
+                            self.next()
+                        self.eat_op(")")
         where = None
         if self.try_kw("WHERE"):
             where = self._expr()
@@ -679,6 +697,13 @@ class Parser:
         t = self.peek()
         if t.kind == "IDENT":
             var = self.next().value
+        elif t.kind == "KW" and t.value in ("END", "START", "COUNT", "ALL",
+                                            "ANY", "NONE", "SINGLE", "INDEX",
+                                            "UNIQUE", "DATABASE", "SHOW",
+                                            "BY", "ON", "CONTAINS"):
+            # soft keywords usable as variable names (Neo4j allows them)
+            var = getattr(t, "raw", "") or t.value.lower()
+            self.next()
         labels = []
         or_labels = False
         while self.try_op(":"):
@@ -991,10 +1016,11 @@ class Parser:
                     self.eat_op(")")
                     return A.Quantifier(kind, var, src, wh)
             if t.value in ("COUNT", "ALL", "ANY", "NONE", "SINGLE", "INDEX",
-                           "UNIQUE", "DATABASE", "SHOW", "BY", "ON", "END"):
+                           "UNIQUE", "DATABASE", "SHOW", "BY", "ON", "END",
+                           "START", "CONTAINS"):
                 # soft keyword used as a plain variable name
                 self.next()
-                return A.Var(t.value.lower())
+                return A.Var(getattr(t, "raw", "") or t.value.lower())
         if t.kind == "OP" and t.value == "(":
             # pattern predicate like (n)-[:R]->(m) in boolean context
             save = self.i

@@ -333,4 +333,35 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
     def _set_vec(ex, node, prop, vector):
         return _set_node_vec(ex, node, prop, vector)
 
+    @register("dbms.procedures")
+    def _dbms_procs(ex):
+        return ["name", "signature"], [[n, ""] for n in sorted(procs)]
+
+    @register("dbms.functions")
+    def _dbms_fns(ex):
+        from .functions import FUNCTIONS
+        return ["name"], [[n] for n in sorted(FUNCTIONS)]
+
+    @register("db.constraints")
+    def _db_constraints(ex):
+        sm = getattr(ex, "schema", None)
+        return ["name", "description"], [
+            [c.name, f"CONSTRAINT ON (:{c.label}) {c.kind} {c.prop}"]
+            for c in (sm.list_constraints() if sm else [])]
+
+    @register("db.indexes")
+    def _db_indexes(ex):
+        sm = getattr(ex, "schema", None)
+        return ["name", "state", "type"], [
+            [n, "ONLINE", k.upper()]
+            for n, k, lb, ps in (sm.list_indexes() if sm else [])]
+
+    @register("db.index.fulltext.createRelationshipIndex")
+    def _ft_rel_idx(ex, name, types=None, props=None):
+        sm = getattr(ex, "schema", None)
+        if sm:
+            sm.create_index((types or ["REL"])[0], (props or ["text"])[0],
+                            name=name, kind="fulltext", props=props)
+        return ["name"], [[name]]
+
     return procs
