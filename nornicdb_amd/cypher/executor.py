@@ -326,7 +326,8 @@ class Executor:
         without one raises."""
         from ..storage.schema import SchemaManager
 
-        if c.op in ("create", "drop") and c.kind in ("database", "alias"):
+        if c.op in ("create", "drop") and c.kind in ("database", "alias",
+                                                     "composite"):
             if self.database_admin is None:
                 raise CypherRuntimeError(
                     f"{c.op.upper()} {c.kind.upper()} needs a DatabaseManager")

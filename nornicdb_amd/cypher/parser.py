@@ -208,6 +208,20 @@ class Parser:
             return self._create_index(kindword or "RANGE", or_replace)
         if self.at_kw("CONSTRAINT"):
             return self._create_constraint(or_replace)
+        if self._at_word("COMPOSITE"):
+            self.next()
+            self._eat_word("DATABASE")
+            name = self.name_part()
+            constituents = []
+            while self._at_word("ALIAS"):
+                self.next()
+                alias = self.name_part()
+                self._eat_word("FOR")
+                self._eat_word("DATABASE")
+                target = self.name_part()
+                constituents.append((alias, target))
+            return A.SchemaCommand("create", "composite", name=name,
+                                   props=[f"{a}:{t}" for a, t in constituents])
         if self.at_kw("DATABASE") or self._at_word("DATABASE"):
             self.next()
             ine, _ = (self._if_not_exists()

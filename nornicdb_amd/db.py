@@ -350,6 +350,25 @@ class DatabaseManager:
                     if not cmd.if_exists:
                         raise
                 return Result([], [])
+            if cmd.op == "create" and cmd.kind == "composite":
+                from .storage import CompositeEngine, NamespacedEngine
+                parts = {}
+                default = None
+                for spec in cmd.props:
+                    alias, target = spec.split(":", 1)
+                    if target not in self._dbs:
+                        self._open(target)
+                    parts[alias] = self._dbs[target].engine
+                    default = default or alias
+                if not parts:
+                    raise ValueError("composite database needs constituents")
+                eng = CompositeEngine(parts, default)
+                self._dbs[cmd.name] = NornicDB(
+                    eng, name=cmd.name, embedder=self._embedder,
+                    dims=self._dims, device=self._device)
+                self._dbs[cmd.name].executor.database_lister = \
+                    lambda: sorted(self._dbs)
+                return Result([], [])
             if cmd.op == "create" and cmd.kind == "alias":
                 if cmd.name in self._aliases and not (cmd.if_not_exists
                                                       or cmd.or_replace):

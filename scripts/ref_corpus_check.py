@@ -14,7 +14,11 @@ from nornicdb_amd.db import DatabaseManager
 from nornicdb_amd.storage.memory import MemoryEngine
 
 
-def main(path="/tmp/ref_queries.txt"):
+def main(path=None):
+    import os
+    if path is None:
+        path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "ref_queries.txt")
     qs = [l for l in open(path).read().splitlines() if l.strip()]
     db = DatabaseManager(MemoryEngine()).get()
     ok = pf = rf = 0

@@ -765,3 +765,23 @@ class TestReferenceSurfaceParity:
                       "YIELD node1, node2, score RETURN count(*)")
         assert r.rows[0][0] >= 1
         db.cypher("CALL gds.graph.drop('g') YIELD graphName RETURN graphName")
+
+
+class TestCompositeDatabase:
+    """CREATE COMPOSITE DATABASE (reference multi_database_test.go fabric
+    surface + pkg/storage/composite_engine.go)."""
+
+    def test_composite_reads_fan_out(self):
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.storage.memory import MemoryEngine
+        mgr = DatabaseManager(MemoryEngine())
+        db = mgr.get()
+        db.cypher("CREATE DATABASE db_a")
+        db.cypher("CREATE DATABASE db_b")
+        db.cypher("USE db_a CREATE (:A {v: 1})")
+        db.cypher("USE db_b CREATE (:B {v: 2})")
+        db.cypher("CREATE COMPOSITE DATABASE comp "
+                  "ALIAS pa FOR DATABASE db_a ALIAS pb FOR DATABASE db_b")
+        comp = mgr.get("comp")
+        assert comp.cypher("MATCH (n) RETURN n.v ORDER BY n.v").rows == [[1], [2]]
+        assert "comp" in [r[0] for r in db.cypher("SHOW DATABASES").rows]
