@@ -46,30 +46,26 @@ class TestMetrics:
 
 class TestCLI:
     def test_init_import_decay(self, tmp_path):
-        env = {"PYTHONPATH": "."}
+        # one subprocess for all three commands: each interpreter start
+        # pays the torch import (~10 s), so chaining keeps the suite fast
         import os
         env = {**os.environ, "PYTHONPATH": "."}
         d = str(tmp_path / "data")
-        r = subprocess.run([sys.executable, "-m", "nornicdb_amd", "init",
-                            "--data-dir", d], capture_output=True, text=True,
-                           env=env, timeout=60)
-        assert r.returncode == 0, r.stderr
-        assert "initialized" in r.stdout
-
         f = tmp_path / "imp.json"
         f.write_text(json.dumps({
             "nodes": [{"id": "a", "labels": ["P"], "properties": {"x": 1}},
                       {"id": "b", "labels": ["P"], "properties": {}}],
             "relationships": [{"id": "e1", "type": "R", "start": "a", "end": "b"}],
         }))
-        r = subprocess.run([sys.executable, "-m", "nornicdb_amd", "import",
-                            "--data-dir", d, "--file", str(f)],
-                           capture_output=True, text=True, env=env, timeout=60)
+        script = (
+            "from nornicdb_amd.__main__ import main\n"
+            f"main(['init', '--data-dir', {d!r}])\n"
+            f"main(['import', '--data-dir', {d!r}, '--file', {str(f)!r}])\n"
+            f"main(['decay', '--data-dir', {d!r}])\n")
+        r = subprocess.run([sys.executable, "-c", script],
+                           capture_output=True, text=True, env=env,
+                           timeout=120)
         assert r.returncode == 0, r.stderr
+        assert "initialized" in r.stdout
         assert "imported 2 nodes, 1 relationships" in r.stdout
-
-        r = subprocess.run([sys.executable, "-m", "nornicdb_amd", "decay",
-                            "--data-dir", d], capture_output=True, text=True,
-                           env=env, timeout=60)
-        assert r.returncode == 0, r.stderr
         assert "scored" in r.stdout
