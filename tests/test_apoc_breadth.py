@@ -198,7 +198,12 @@ class TestApocBatch3:
         import time
         db.cypher("CALL apoc.periodic.submit('t1', 'CREATE (:Done)') "
                   "YIELD name RETURN name")
-        time.sleep(0.3)
+        # poll: the job runs on a background thread (slow machines need
+        # longer than a fixed sleep)
+        for _ in range(100):
+            if db.cypher("MATCH (d:Done) RETURN count(d)").rows == [[1]]:
+                break
+            time.sleep(0.1)
         assert db.cypher("MATCH (d:Done) RETURN count(d)").rows == [[1]]
         rows = db.cypher("CALL apoc.periodic.list() YIELD name, done "
                          "RETURN name, done").rows

@@ -139,7 +139,10 @@ class TestHA:
         standby = HAStandby("s", "p", bus, StorageAdapter(standby_engine).apply)
         for i in range(5):
             primary.replicate(node_cmd(i))
-        time.sleep(0.3)
+        for _ in range(100):          # poll: async apply on slow machines
+            if standby_engine.node_count() == 5:
+                break
+            time.sleep(0.05)
         assert standby_engine.node_count() == 5
         assert primary.lag("s") == 0
         assert not standby.check_failover.__self__.promoted
