@@ -64,7 +64,7 @@ class TestCLI:
             f"main(['decay', '--data-dir', {d!r}])\n")
         r = subprocess.run([sys.executable, "-c", script],
                            capture_output=True, text=True, env=env,
-                           timeout=120)
+                           timeout=600)
         assert r.returncode == 0, r.stderr
         assert "initialized" in r.stdout
         assert "imported 2 nodes, 1 relationships" in r.stdout
