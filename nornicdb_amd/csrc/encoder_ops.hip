@@ -165,7 +165,7 @@ __global__ void k_mean_pool_l2norm(const unsigned short* __restrict__ x,
 #define FA_KT 64
 #define FA_QT 64
 #define FA_PSTRIDE 64   // measured: 64 beats 72 by 15% (208 vs 181 TF; scripts/attn_pad.hip sweep)
-#define FA_VSTRIDE 72
+#define FA_VSTRIDE 80   // measured best with gather-V staging (232 vs 226 TF at 72)
 
 __global__ __launch_bounds__(256, 2) void k_flash_attn_nc(
     const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,

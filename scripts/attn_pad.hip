@@ -275,6 +275,14 @@ int main() {
   printf("P64/V72/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
   ms = run<64, 66, 1>(Q, K, V, O2, B, H, S, 20);
   printf("P64/V66/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
+  ms = run<64, 68, 1>(Q, K, V, O2, B, H, S, 20);
+  printf("P64/V68/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
+  ms = run<64, 76, 1>(Q, K, V, O2, B, H, S, 20);
+  printf("P64/V76/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
+  ms = run<64, 80, 1>(Q, K, V, O2, B, H, S, 20);
+  printf("P64/V80/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
+  ms = run<64, 96, 1>(Q, K, V, O2, B, H, S, 20);
+  printf("P64/V96/gatherV: %6.3f ms  %5.0f TF\n", ms, fl / ms / 1e9);
   {
     run<64, 72, 0>(Q, K, V, O, B, H, S, 1);
     run<64, 72, 1>(Q, K, V, O2, B, H, S, 1);
