@@ -90,8 +90,11 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
 
     @register("db.info")
     def info(ex):
-        return (["name", "nodes", "relationships"],
-                [[db.name, db.engine.node_count(), db.engine.edge_count()]])
+        return (["id", "name", "creationDate", "nodes", "relationships",
+                 "nodeCount", "relationshipCount"],
+                [["nornicdb", db.name, "", db.engine.node_count(),
+                  db.engine.edge_count(), db.engine.node_count(),
+                  db.engine.edge_count()]])
 
     @register("db.ping")
     def ping(ex):
@@ -363,5 +366,16 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
             sm.create_index((types or ["REL"])[0], (props or ["text"])[0],
                             name=name, kind="fulltext", props=props)
         return ["name"], [[name]]
+
+    @register("db.create.setRelationshipVectorProperty")
+    def _set_rel_vec(ex, rel, prop, vector):
+        e = db.engine.get_edge(rel.id if hasattr(rel, "id") else rel)
+        e.properties[prop] = [float(x) for x in (vector or [])]
+        db.engine.update_edge(e)
+        return ["relationship"], [[e]]
+
+    @register("tx.setMetaData")
+    def _tx_meta(ex, meta=None):
+        return [], []
 
     return procs
