@@ -193,3 +193,26 @@ class TestManagerGPUIntegration:
                               device="cuda", max_tokens_default=8)
         toks = list(mgr.generate_stream("stream me", max_tokens=6))
         assert 1 <= len(toks) <= 6
+
+
+@pytest.mark.gpu
+class TestFusedDecoderFullSize:
+    def test_full_model_greedy_matches_graphed(self):
+        """Full Qwen2-0.5B-shape config (24 layers): fused decode must
+        track the graphed decoder exactly at production size."""
+        import torch
+
+        from nornicdb_amd.models.heimdall import (FusedDecoder,
+                                                  GraphedDecoder,
+                                                  HeimdallConfig,
+                                                  HeimdallModel)
+        torch.manual_seed(0)
+        cfg = HeimdallConfig()
+        m = HeimdallModel(cfg).init_small().to("cuda", torch.bfloat16).eval()
+        prompt = torch.randint(0, cfg.vocab_size, (1, 8), device="cuda")
+        fd = FusedDecoder(m, max_len=512)
+        fused = fd.generate(prompt.clone(), max_new_tokens=6, temperature=0.0)
+        gd = GraphedDecoder(m, max_len=512).capture()
+        graphed = gd.generate(prompt.clone(), max_new_tokens=6,
+                              temperature=0.0)
+        assert fused == graphed
