@@ -413,3 +413,22 @@ def _java_format(fmt, *args):
 
 
 register("format", _java_format)
+
+
+# Neo4j round(): optional precision and mode arguments
+def _round(x, precision=0, mode="HALF_UP"):
+    if x is None:
+        return None
+    import decimal
+    p = int(precision)
+    modes = {"UP": decimal.ROUND_UP, "DOWN": decimal.ROUND_DOWN,
+             "CEILING": decimal.ROUND_CEILING, "FLOOR": decimal.ROUND_FLOOR,
+             "HALF_UP": decimal.ROUND_HALF_UP,
+             "HALF_DOWN": decimal.ROUND_HALF_DOWN,
+             "HALF_EVEN": decimal.ROUND_HALF_EVEN}
+    m = modes.get(str(mode).upper(), decimal.ROUND_HALF_UP)
+    q = decimal.Decimal(1).scaleb(-p)
+    return float(decimal.Decimal(str(float(x))).quantize(q, rounding=m))
+
+
+register("round", _round)

@@ -785,3 +785,14 @@ class TestCompositeDatabase:
         comp = mgr.get("comp")
         assert comp.cypher("MATCH (n) RETURN n.v ORDER BY n.v").rows == [[1], [2]]
         assert "comp" in [r[0] for r in db.cypher("SHOW DATABASES").rows]
+
+
+class TestRoundPrecision:
+    def test_round_modes(self):
+        from nornicdb_amd.cypher.executor import Executor
+        from nornicdb_amd.storage.memory import MemoryEngine
+        ex = Executor(MemoryEngine())
+        assert ex.execute(
+            "RETURN round(3.14159, 2), round(3.5), "
+            "round(2.5, 0, 'HALF_EVEN'), round(-1.5)"
+        ).rows == [[3.14, 4.0, 2.0, -2.0]]
