@@ -1641,3 +1641,20 @@ _r("apoc.nodes.filter", lambda l, prop, value: [
 _r("apoc.nodes.reduce", lambda l, prop: sum(
     n.properties.get(prop, 0) for n in (l or [])
     if isinstance(n.properties.get(prop), (int, float))))
+
+
+def _flatten_deep(l):
+    out = []
+    for x in (l or []):
+        if isinstance(x, list):
+            out.extend(_flatten_deep(x))
+        else:
+            out.append(x)
+    return out
+
+
+_r("apoc.text.join", lambda l, sep="": str(sep).join(
+    str(x) for x in (l or []) if x is not None))
+_r("apoc.coll.flatten", lambda l, recursive=False: (
+    _flatten_deep(l) if recursive else
+    [y for x in (l or []) for y in (x if isinstance(x, list) else [x])]))

@@ -432,3 +432,9 @@ def _round(x, precision=0, mode="HALF_UP"):
 
 
 register("round", _round)
+
+
+register("ltrim", lambda s, chars=None: None if s is None else
+         (s.lstrip() if chars is None else s.lstrip(chars)))
+register("rtrim", lambda s, chars=None: None if s is None else
+         (s.rstrip() if chars is None else s.rstrip(chars)))

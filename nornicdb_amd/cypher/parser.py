@@ -1089,6 +1089,25 @@ class Parser:
         if t.kind == "OP" and t.value == "{":
             return self._map_lit()
         if t.kind == "IDENT":
+            # trim([BOTH|LEADING|TRAILING] [ch] FROM s) — SQL-style form
+            if (t.value.lower() in ("trim", "ltrim", "rtrim", "btrim")
+                    and self.peek(1).kind == "OP" and self.peek(1).value == "("
+                    and self.peek(2).kind in ("IDENT", "KW")
+                    and self.peek(2).value.upper() in ("BOTH", "LEADING",
+                                                       "TRAILING")):
+                self.next()
+                self.eat_op("(")
+                spec = self.next().value.upper()
+                ch = None
+                if not self._at_word("FROM"):
+                    ch = self._expr()
+                self._eat_word("FROM")
+                s = self._expr()
+                self.eat_op(")")
+                fname = {"BOTH": "btrim", "LEADING": "ltrim",
+                         "TRAILING": "rtrim"}[spec]
+                args = [s] + ([ch] if ch is not None else [])
+                return A.FuncCall(fname, args)
             # reduce(acc = init, x IN list | expr)
             if (t.value.lower() == "reduce" and self.peek(1).kind == "OP"
                     and self.peek(1).value == "("):

@@ -796,3 +796,14 @@ class TestRoundPrecision:
             "RETURN round(3.14159, 2), round(3.5), "
             "round(2.5, 0, 'HALF_EVEN'), round(-1.5)"
         ).rows == [[3.14, 4.0, 2.0, -2.0]]
+
+
+class TestTrimSpecForm:
+    def test_sql_style_trim(self):
+        from nornicdb_amd.db import NornicDB
+        from nornicdb_amd.storage.memory import MemoryEngine
+        db = NornicDB(MemoryEngine(), auto_embed=False)
+        assert db.cypher(
+            "RETURN trim(BOTH 'x' FROM 'xxaxx'), "
+            "trim(LEADING 'x' FROM 'xxa'), trim(TRAILING FROM 'a  ')"
+        ).rows == [["a", "a", "a"]]
