@@ -158,7 +158,7 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
             n = eng.get_node(g.node_ids[i])
             n.properties[prop] = [float(x) for x in emb[i]]
             eng.update_node(n)
-        return ["nodeCount"], [[g.n]]
+        return ["nodeCount", "nodesWritten"], [[g.n, g.n]]
 
     # -------------------- apoc.create --------------------
     @register("apoc.create.node")
