@@ -253,3 +253,56 @@ class TestPathsAndFunctions:
         ex.execute("CREATE (:EP {a: 1}), (:EP)")
         r = ex.execute("MATCH (n:EP) WHERE n.a IS NOT NULL RETURN count(n)")
         assert r.rows == [[1]]
+
+
+class TestConformanceBatch3:
+    """Probed corners locked in as regressions."""
+
+    @pytest.fixture
+    def ex(self):
+        from nornicdb_amd.cypher import Executor
+        from nornicdb_amd.storage import MemoryEngine
+        return Executor(MemoryEngine())
+
+    def test_with_star_passthrough(self, ex):
+        ex.execute("CREATE (a:P {v:1})-[:K]->(b:P {v:2})")
+        r = ex.execute("MATCH (a:P {v:1})-[:K]->(b) WITH * RETURN a.v, b.v")
+        assert r.rows == [[1, 2]]
+
+    def test_nested_list_comprehension(self, ex):
+        r = ex.execute("RETURN [x IN [1,2] | [y IN [3,4] | x*y]]")
+        assert r.rows == [[[[3, 4], [6, 8]]]]
+
+    def test_order_by_nulls_last(self, ex):
+        r = ex.execute("UNWIND [3, null, 1] AS x RETURN x ORDER BY x")
+        assert r.rows == [[1], [3], [None]]
+
+    def test_string_concat_null_propagates(self, ex):
+        assert ex.execute("RETURN 'a' + null").rows == [[None]]
+
+    def test_set_null_removes_property(self, ex):
+        ex.execute("CREATE (:N {a: 1, b: 2})")
+        ex.execute("MATCH (n:N) SET n.a = null")
+        r = ex.execute("MATCH (n:N) RETURN keys(n)")
+        assert r.rows == [[["b"]]]
+
+    def test_merge_full_pattern(self, ex):
+        for _ in range(2):
+            ex.execute("MERGE (a:MA {k:1})-[r:ML {p: 2}]->(b:MB {k:2})")
+        assert ex.execute("MATCH (:MA)-[r:ML]->(:MB) RETURN count(r)"
+                          ).rows == [[1]]
+
+    def test_with_aggregate_then_where(self, ex):
+        ex.execute("UNWIND range(1,4) AS i CREATE (:W {v: i})")
+        r = ex.execute("MATCH (n:W) WITH count(n) AS c WHERE c > 3 RETURN c")
+        assert r.rows == [[4]]
+
+    def test_union_three_way_distinct(self, ex):
+        r = ex.execute("RETURN 1 AS x UNION RETURN 2 AS x UNION RETURN 1 AS x")
+        assert sorted(v[0] for v in r.rows) == [1, 2]
+
+    def test_all_shortest_paths(self, ex):
+        ex.execute("CREATE (a:S {n:1})-[:R]->(b:S {n:2})-[:R]->(c:S {n:3})")
+        r = ex.execute("MATCH p = allShortestPaths((a:S {n:1})-[*..4]->"
+                       "(c:S {n:3})) RETURN length(p)")
+        assert r.rows == [[2]]
