@@ -223,3 +223,40 @@ class TestSearchService:
         assert len(svc.fulltext) == 0
         svc.build_indexes()
         assert len(svc.fulltext) == 3
+
+
+class TestVectorSpaceRegistry:
+    """pkg/vectorspace parity: named vector spaces per (db, entity, name)."""
+
+    def test_register_get_list_drop(self):
+        from nornicdb_amd.search.vectorspace import Registry, VectorSpace
+        reg = Registry()
+        reg.register(VectorSpace("neo4j", "Doc", "emb", 1024, "cosine"))
+        reg.register(VectorSpace("neo4j", "Chunk", "emb2", 512, "euclidean"))
+        assert reg.get("neo4j", "Doc", "emb").dims == 1024
+        assert len(reg.list("neo4j")) == 2
+        assert reg.drop("neo4j", "Doc", "emb") is True
+        assert reg.get("neo4j", "Doc", "emb") is None
+
+
+class TestHTTPEmbedderOffline:
+    """embed/http_providers: transport injection (no network here)."""
+
+    def test_injected_transport(self):
+        import numpy as np
+
+        from nornicdb_amd.embed.http_providers import HTTPEmbedder
+        calls = []
+
+        def fake_transport(url, payload, headers):
+            calls.append(url)
+            texts = payload.get("input") or payload.get("prompt") or []
+            if isinstance(texts, str):
+                texts = [texts]
+            return {"data": [{"embedding": [0.1] * 8} for _ in texts]}
+
+        e = HTTPEmbedder("openai", base_url="http://fake", dims=8,
+                         transport=fake_transport)
+        out = e.embed_batch(["a", "b"])
+        assert out.shape == (2, 8)
+        assert calls
