@@ -379,7 +379,9 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
   TORCH_CHECK(d % 128 == 0, "knn_gemv needs D % 128 == 0");
   TORCH_CHECK(k_out >= 1 && k_out <= KNN_K, "k_out must be <= ", KNN_K);
 
-  int blocks = (int)std::min<long long>((n + 1023) / 1024, 1280);
+  int cap = 1280;
+  if (const char* g = getenv("NORNICDB_GEMV_BLOCKS")) cap = atoi(g);
+  int blocks = (int)std::min<long long>((n + 1023) / 1024, cap);
   blocks = std::max(blocks, 1);
   long long waves = (long long)blocks * 4 * 4;  // 4 waves x 4 row-groups
 
