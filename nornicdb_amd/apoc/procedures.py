@@ -83,10 +83,17 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
 
     @register("apoc.algo.dijkstra")
     def _dijkstra(ex, start, end, rel_type=None, weight_prop="weight"):
+        """String args are node ids (the reference's NodeID is a string,
+        pkg/cypher/apoc_algorithms.go:67), falling back to the id/name
+        property; an unreachable or missing endpoint yields no rows."""
         g = from_engine(eng, edge_types=[rel_type] if rel_type else None,
                         weight_prop=weight_prop)
-        s = g.id2idx[start.id if isinstance(start, Node) else start]
-        t = g.id2idx[end.id if isinstance(end, Node) else end]
+        sid = _node_arg(start)
+        tid = _node_arg(end)
+        if sid not in g.id2idx or tid not in g.id2idx:
+            return ["path", "weight"], []
+        s = g.id2idx[sid]
+        t = g.id2idx[tid]
         dist, _ = dijkstra(g, s, t)
         path_idx = shortest_path(g, s, t)
         nodes = [eng.get_node(g.node_ids[i]) for i in path_idx]
@@ -2740,29 +2747,171 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
     register("apoc.algo.wcc")(procs["apoc.community.wcc"])
 
     def _node_arg(x):
-        """Accept Node, id string, or {id: ...} map."""
+        """Resolve a procedure node argument to a node id, or None.
+
+        Accepts a Node, a node id (the reference's NodeID is a plain
+        string, pkg/cypher/apoc_algorithms.go:67), a property map like
+        {id: 'a'} (matched against node properties), or a string that
+        falls back to matching the `id`/`name` property.
+        """
         if isinstance(x, Node):
             return x.id
         if isinstance(x, dict):
-            return str(x.get("id"))
-        return str(x)
+            for n in eng.all_nodes():
+                props = n.properties or {}
+                if all(props.get(k) == v for k, v in x.items()):
+                    return n.id
+            return None
+        try:
+            eng.get_node(x)
+            return x
+        except Exception:
+            pass
+        for n in eng.all_nodes():
+            props = n.properties or {}
+            if props.get("id") == x or props.get("name") == x:
+                return n.id
+        return None
 
     _orig_spanning = procs["apoc.path.spanningtree"]
 
     @register("apoc.path.spanningTree")
     def _pe_spanning2(ex, start, config=None):
-        return _orig_spanning(ex, _node_arg(start), config)
+        """Yields one path per tree branch (APOC contract: YIELD path).
+
+        Reference: pkg/cypher tests call spanningTree({id:'a'}, cfg)
+        YIELD path; an unresolvable start yields no rows.
+        """
+        sid = _node_arg(start)
+        if sid is None:
+            return ["path"], []
+        from ..cypher.executor import Path as _P
+        _, rows = _orig_spanning(ex, sid, config)
+        nodes, rels = rows[0]
+        limit = int(dict(config or {}).get("limit", 0) or 0)
+        paths = [[_P(nodes, rels)]]
+        return ["path"], paths[:limit] if limit else paths
 
     _orig_subnodes = procs.get("apoc.path.subgraphnodes")
     if _orig_subnodes is not None:
         @register("apoc.path.subgraphNodes")
         def _pe_subnodes2(ex, start, config=None):
-            return _orig_subnodes(ex, _node_arg(start), config)
+            sid = _node_arg(start)
+            if sid is None:
+                return ["node"], []
+            lvl = dict(config or {}).get("maxLevel", 3) \
+                if isinstance(config, dict) or config is None else config
+            return _orig_subnodes(ex, sid, lvl)
 
     _orig_suball = procs["apoc.path.subgraphall"]
 
     @register("apoc.path.subgraphAll")
     def _pe_suball2(ex, start, config=None):
-        return _orig_suball(ex, _node_arg(start), config)
+        sid = _node_arg(start)
+        if sid is None:
+            return ["nodes", "relationships"], [[[], []]]
+        return _orig_suball(ex, sid, config)
+
+    # ---- tolerant algo arg forms + path/neighbor procs
+    # (reference pkg/cypher/apoc_algorithms.go: pageRank('Label'),
+    # louvain(['Label']), allSimplePaths, neighbors.byhop) ----
+    def _algo_cfg(a, b):
+        """Normalize (labels?, config?) leading args: returns (labels, cfg)."""
+        labels, cfg = None, {}
+        for v in (a, b):
+            if isinstance(v, str):
+                labels = [v]
+            elif isinstance(v, (list, tuple)):
+                labels = [x.id if isinstance(x, Node) else str(x) for x in v] \
+                    if v and isinstance(v[0], Node) else [str(x) for x in v]
+            elif isinstance(v, dict):
+                cfg = v
+        return labels, cfg
+
+    _orig_pagerank = procs["apoc.algo.pagerank"]
+
+    @register("apoc.algo.pageRank")
+    def _pagerank2(ex, a=None, b=None, **kw):
+        if isinstance(a, (int, float)) and not isinstance(a, bool):
+            return _orig_pagerank(ex, a, b if b is not None else 0.85)
+        labels, cfg = _algo_cfg(a, b)
+        return _orig_pagerank(ex, int(cfg.get("iterations", 20)),
+                              float(cfg.get("dampingFactor",
+                                            cfg.get("damping", 0.85))))
+
+    for _nm, _key in (("apoc.algo.louvain", "apoc.community.louvain"),
+                      ("apoc.algo.wcc", "apoc.community.wcc"),
+                      ("apoc.algo.labelPropagation",
+                       "apoc.community.labelpropagation")):
+        def _mk(key):
+            orig = procs[key]
+
+            def _tolerant(ex, a=None, b=None):
+                if key.endswith("labelpropagation"):
+                    if isinstance(a, (int, float)) and not isinstance(a, bool):
+                        return orig(ex, int(a))
+                    _, cfg = _algo_cfg(a, b)
+                    return orig(ex, int(cfg.get("iterations", 20)))
+                cols, rows = orig(ex)
+                if key.endswith("wcc"):
+                    # apoc.algo.wcc yields componentId (reference test name)
+                    cols = ["node", "componentId"]
+                return cols, rows
+            return _tolerant
+        register(_nm)(_mk(_key))
+
+    @register("apoc.algo.allSimplePaths")
+    def _all_simple_paths(ex, start, end, rel_type=None, max_depth=10):
+        """All simple (no repeated node) paths start->end following
+        rel_type edges (reference apoc_algorithms.go findAllSimplePaths)."""
+        from ..cypher.executor import Path as _P
+        sid, tid = _node_arg(start), _node_arg(end)
+        if sid is None or tid is None:
+            return ["path"], []
+        out = []
+
+        def dfs(cur, nodes, edges, seen):
+            if len(edges) > int(max_depth):
+                return
+            if cur == tid:
+                out.append([_P([eng.get_node(i) for i in nodes],
+                               list(edges))])
+                return
+            for e in _all_edges_of(cur):
+                if rel_type and e.type != rel_type:
+                    continue
+                other = e.end_node if e.start_node == cur else e.start_node
+                if other in seen:
+                    continue
+                dfs(other, nodes + [other], edges + [e], seen | {other})
+
+        dfs(sid, [sid], [], {sid})
+        return ["path"], out
+
+    @register("apoc.neighbors.byhop")
+    def _neighbors_byhop(ex, start, rel_filter=None, max_hops=3):
+        """Neighbor node groups bucketed by hop distance 1..maxHops
+        (reference apoc_algorithms.go neighbors.byhop: YIELD nodes, depth)."""
+        sid = _node_arg(start)
+        if sid is None:
+            return ["nodes", "depth"], []
+        seen = {sid}
+        frontier = {sid}
+        rows = []
+        for depth in range(1, int(max_hops) + 1):
+            nxt = set()
+            for nid in frontier:
+                for e in _all_edges_of(nid):
+                    if rel_filter and e.type != str(rel_filter).lstrip("<>"):
+                        continue
+                    other = e.end_node if e.start_node == nid else e.start_node
+                    if other not in seen:
+                        seen.add(other)
+                        nxt.add(other)
+            if not nxt:
+                break
+            rows.append([[eng.get_node(i) for i in sorted(nxt)], depth])
+            frontier = nxt
+        return ["nodes", "depth"], rows
 
     return procs

@@ -6,6 +6,7 @@ register into the same table under their dotted names (apoc/ registry).
 
 from __future__ import annotations
 
+import json
 import math
 import random
 import re
@@ -79,12 +80,19 @@ register("time", lambda arg=None: _tp.make_time(arg))
 register("localtime", lambda arg=None: _tp.make_time(arg, local=True))
 register("duration", lambda arg: _tp.make_duration(arg))
 register("duration.between", lambda a, b: _tp.duration_between(a, b))
-register("duration.indays", lambda a, b: _tp.CypherDuration(
-    0, _tp.duration_between(a, b).days, 0, 0))
-register("duration.inseconds", lambda a, b: _tp.CypherDuration(
-    0, 0, int(_tp.duration_between(a, b).total_seconds_approx()), 0))
-register("duration.inmonths", lambda a, b: _tp.CypherDuration(
-    _tp.duration_between(a, b).days // 30, 0, 0, 0))
+def _dur_norm(a, b=None):
+    """1-arg form converts a duration; 2-arg form measures between temporals
+    (reference: functions_eval_functions.go duration.inDays both arities)."""
+    return _tp.make_duration(a) if b is None and isinstance(
+        a, (_tp.CypherDuration, str)) else _tp.duration_between(a, b)
+
+
+register("duration.indays", lambda a, b=None: _tp.CypherDuration(
+    0, int(_dur_norm(a, b).total_seconds_approx() // 86400), 0, 0))
+register("duration.inseconds", lambda a, b=None: _tp.CypherDuration(
+    0, 0, int(_dur_norm(a, b).total_seconds_approx()), 0))
+register("duration.inmonths", lambda a, b=None: _tp.CypherDuration(
+    int(_dur_norm(a, b).total_seconds_approx() // (30.4375 * 86400)), 0, 0, 0))
 register("datetime.truncate", lambda unit, v=None:
          _tp.truncate(unit, v if v is not None else _tp.make_datetime(None)))
 register("date.truncate", lambda unit, v=None:
@@ -352,6 +360,34 @@ register("sqrt", lambda x: None if x is None else (
     float("nan") if x < 0 else math.sqrt(_num(x))))
 register("log", lambda x: None if x is None else (
     float("nan") if x <= 0 else math.log(_num(x))))
+# hyperbolic family + power (Neo4j 2025.06 surface,
+# reference functions_eval_math.go:188-230)
+register("sinh", lambda x: None if x is None else math.sinh(_num(x)))
+register("cosh", lambda x: None if x is None else math.cosh(_num(x)))
+register("tanh", lambda x: None if x is None else math.tanh(_num(x)))
+register("coth", lambda x: None if x is None else (
+    float("nan") if _num(x) == 0 else math.cosh(_num(x)) / math.sinh(_num(x))))
+register("power", lambda b, e: None if b is None or e is None
+         else _num(b) ** _num(e))
+
+
+def _date_comp(arg, comp):
+    d = arg if isinstance(arg, (_tp.CypherDate, _tp.CypherDateTime)) \
+        else _tp.make_date(arg)
+    return d.component(comp)
+
+
+# date.year/month/day accessors on ISO strings or temporals
+# (reference functions_eval_functions.go:1402-1496)
+register("date.year", lambda a: _date_comp(a, "year"))
+register("date.month", lambda a: _date_comp(a, "month"))
+register("date.day", lambda a: _date_comp(a, "day"))
+register("date.week", lambda a: _date_comp(a, "week"))
+register("date.quarter", lambda a: _date_comp(a, "quarter"))
+register("date.dayofweek", lambda a: _date_comp(a, "dayOfWeek"))
+register("date.dayofyear", lambda a: _date_comp(a, "ordinalDay"))
+register("apoc.create.uuid", lambda: str(uuid.uuid4()))
+
 register("log10", lambda x: None if x is None else (
     float("nan") if x <= 0 else math.log10(_num(x))))
 register("asin", lambda x: None if x is None else (
@@ -360,36 +396,207 @@ register("acos", lambda x: None if x is None else (
     float("nan") if abs(x) > 1 else math.acos(_num(x))))
 
 
-# ---- kalman.* functions (reference exposes pkg/filter via Cypher) ----
-_KALMAN_STATES: Dict[str, Any] = {}
+# ---- kalman.* functions (JSON-state filters, pkg/cypher/kalman_functions.go) ----
+# State travels as a JSON string the caller stores in a node property:
+#   s = kalman.init({processNoise: 0.1}); r = kalman.process(23.5, s);
+#   r.value is the smoothed estimate, r.state the updated JSON.
+# Three families: kalman.* (1-state scalar), kalman.velocity.* (2-state
+# position+velocity), kalman.adaptive.* (auto-switches between the two).
 
 
-def _kalman_get(name):
-    from ..cognitive.kalman import Kalman1D
-    if name not in _KALMAN_STATES:
-        _KALMAN_STATES[name] = Kalman1D()
-    return _KALMAN_STATES[name]
+def _kj_default():
+    return {"x": 0.0, "lx": 0.0, "p": 30.0, "k": 0.0, "e": 1.0,
+            "q": 0.0001, "r": 88.0, "vs": 10.0, "n": 0}
 
 
-def _kalman_init(name, q=0.01, r=0.1, initial=0.0):
-    from ..cognitive.kalman import Kalman1D
-    k = Kalman1D(q=q, r=r)
-    k.x = float(initial)
-    _KALMAN_STATES[str(name)] = k
-    return str(name)
+def _kvj_default():
+    return {"pos": 0.0, "vel": 0.0, "p": [100.0, 0.0, 0.0, 10.0],
+            "qp": 0.1, "qv": 0.01, "r": 1.0, "dt": 1.0, "n": 0}
+
+
+def _kaj_default():
+    return {"basic": _kj_default(), "velocity": _kvj_default(),
+            "mode": "basic", "ss": 0, "tt": 0.1, "st": 0.02, "hy": 10,
+            "n": 0, "lf": 0.0, "ts": 0.0}
+
+
+def _kalman_init(config=None):
+    st = _kj_default()
+    cfg = config if isinstance(config, dict) else {}
+    if "processNoise" in cfg:
+        st["q"] = float(cfg["processNoise"]) * 0.001
+    if "measurementNoise" in cfg:
+        st["r"] = float(cfg["measurementNoise"])
+    if "initialCovariance" in cfg:
+        st["p"] = float(cfg["initialCovariance"])
+    if "varianceScale" in cfg:
+        st["vs"] = float(cfg["varianceScale"])
+    return json.dumps(st)
+
+
+def _kalman_process(measurement, state, target=0.0):
+    m = float(measurement)
+    try:
+        st = json.loads(state)
+        assert isinstance(st, dict) and "x" in st
+    except Exception:
+        return {"value": m, "state": state, "error": "invalid state"}
+    velocity = st["x"] - st["lx"]
+    st["x"] += velocity
+    st["lx"] = st["x"]
+    if target and st["lx"]:
+        st["e"] = abs(1.0 - (float(target) / st["lx"]))
+    else:
+        st["e"] = 1.0
+    st["p"] = st["p"] + st["q"] * st["e"]
+    st["k"] = st["p"] / (st["p"] + st["r"])
+    st["x"] += st["k"] * (m - st["x"])
+    st["p"] = (1.0 - st["k"]) * st["p"]
+    st["n"] += 1
+    return {"value": st["x"], "state": json.dumps(st)}
+
+
+def _kalman_predict(state, steps=1):
+    try:
+        st = json.loads(state)
+        return st["x"] + float(steps) * (st["x"] - st["lx"])
+    except Exception:
+        return 0.0
+
+
+def _kalman_state_value(state):
+    try:
+        return json.loads(state)["x"]
+    except Exception:
+        return 0.0
+
+
+def _kalman_rate(state):
+    try:
+        st = json.loads(state)
+        return st["x"] - st["lx"]
+    except Exception:
+        return 0.0
+
+
+def _kalman_vel_init(initial_pos=None, initial_vel=None):
+    st = _kvj_default()
+    if initial_pos is not None:
+        st["pos"] = float(initial_pos)
+        st["vel"] = float(initial_vel or 0.0)
+    return json.dumps(st)
+
+
+def _kalman_vel_process(measurement, state):
+    m = float(measurement)
+    try:
+        st = json.loads(state)
+        assert isinstance(st, dict) and "pos" in st
+    except Exception:
+        return {"value": m, "velocity": 0.0, "state": state,
+                "error": "invalid state"}
+    dt = st["dt"] if st["dt"] > 0 else 1.0
+    pred_pos = st["pos"] + st["vel"] * dt
+    pred_vel = st["vel"]
+    p00, p01, p10, p11 = st["p"]
+    pp00 = p00 + dt * p10 + dt * p01 + dt * dt * p11 + st["qp"]
+    pp01 = p01 + dt * p11
+    pp10 = p10 + dt * p11
+    pp11 = p11 + st["qv"]
+    innov = m - pred_pos
+    sc = pp00 + st["r"]
+    k0, k1 = pp00 / sc, pp10 / sc
+    st["pos"] = pred_pos + k0 * innov
+    st["vel"] = pred_vel + k1 * innov
+    st["p"] = [(1 - k0) * pp00, (1 - k0) * pp01,
+               pp10 - k1 * pp00, pp11 - k1 * pp01]
+    st["n"] += 1
+    return {"value": st["pos"], "velocity": st["vel"],
+            "state": json.dumps(st)}
+
+
+def _kalman_vel_predict(state, steps=1):
+    try:
+        st = json.loads(state)
+        dt = st["dt"] if st["dt"] > 0 else 1.0
+        return st["pos"] + st["vel"] * float(steps) * dt
+    except Exception:
+        return 0.0
+
+
+def _kalman_adaptive_init(config=None):
+    st = _kaj_default()
+    cfg = config if isinstance(config, dict) else {}
+    if "trendThreshold" in cfg:
+        st["tt"] = float(cfg["trendThreshold"])
+    if "stabilityThreshold" in cfg:
+        st["st"] = float(cfg["stabilityThreshold"])
+    if "hysteresis" in cfg:
+        st["hy"] = int(cfg["hysteresis"])
+    if cfg.get("initialMode") == "velocity":
+        st["mode"] = "velocity"
+    return json.dumps(st)
+
+
+def _kalman_adaptive_process(measurement, state):
+    m = float(measurement)
+    try:
+        st = json.loads(state)
+        assert isinstance(st, dict) and "mode" in st
+    except Exception:
+        return {"value": m, "mode": "error", "state": state,
+                "error": "invalid state"}
+    if st["mode"] == "velocity":
+        r = _kalman_vel_process(m, json.dumps(st["velocity"]))
+        filtered = r["value"]
+        st["velocity"] = json.loads(r["state"])
+        st["ts"] = st["velocity"]["vel"]
+    else:
+        r = _kalman_process(m, json.dumps(st["basic"]))
+        filtered = r["value"]
+        st["basic"] = json.loads(r["state"])
+        st["ts"] = st["basic"]["x"] - st["basic"]["lx"]
+    st["n"] += 1
+    st["ss"] += 1
+    if st["ss"] >= st["hy"]:
+        mag = abs(st["ts"])
+        if st["mode"] == "basic" and mag > st["tt"]:
+            st["mode"] = "velocity"
+            st["ss"] = 0
+            st["velocity"]["pos"] = st["basic"]["x"]
+            st["velocity"]["vel"] = st["ts"]
+        elif st["mode"] == "velocity" and mag < st["st"]:
+            st["mode"] = "basic"
+            st["ss"] = 0
+            st["basic"]["x"] = st["velocity"]["pos"]
+            st["basic"]["lx"] = st["velocity"]["pos"] - st["velocity"]["vel"]
+    st["lf"] = filtered
+    return {"value": filtered, "mode": st["mode"], "state": json.dumps(st)}
+
+
+def _kalman_reset(state):
+    try:
+        st = json.loads(state)
+    except Exception:
+        return _kalman_init()
+    if isinstance(st, dict) and "pos" in st:
+        return _kalman_vel_init()
+    if isinstance(st, dict) and "mode" in st:
+        return _kalman_adaptive_init()
+    return _kalman_init()
 
 
 register("kalman.init", _kalman_init)
-register("kalman.predict", lambda name: _kalman_get(str(name)).predict())
-register("kalman.process", lambda name, measurement:
-         _kalman_get(str(name)).update(float(measurement)))
-register("kalman.update", lambda name, measurement:
-         _kalman_get(str(name)).update(float(measurement)))
-register("kalman.state", lambda name: {
-    "x": _kalman_get(str(name)).x,
-    "p": getattr(_kalman_get(str(name)), "p", None)})
-register("kalman.reset", lambda name: (_KALMAN_STATES.pop(str(name), None),
-                                       True)[1])
+register("kalman.process", _kalman_process)
+register("kalman.predict", _kalman_predict)
+register("kalman.state", _kalman_state_value)
+register("kalman.rate", _kalman_rate)
+register("kalman.reset", _kalman_reset)
+register("kalman.velocity.init", _kalman_vel_init)
+register("kalman.velocity.process", _kalman_vel_process)
+register("kalman.velocity.predict", _kalman_vel_predict)
+register("kalman.adaptive.init", _kalman_adaptive_init)
+register("kalman.adaptive.process", _kalman_adaptive_process)
 
 # ---- string format/pad (reference util) ----
 register("lpad", lambda s, n, pad=" ": None if s is None else

@@ -291,7 +291,7 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
 
     # -------------------- db.index extras --------------------
     @register("db.index.vector.createRelationshipIndex")
-    def _vec_rel_idx(ex, name, rel_type, prop, dims, similarity="cosine"):
+    def _vec_rel_idx(ex, name, rel_type, prop, dims=1024, similarity="cosine"):
         sm = getattr(ex, "schema", None)
         if sm:
             sm.create_vector_index(name, rel_type, prop, int(dims), similarity)
@@ -377,5 +377,112 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
     @register("tx.setMetaData")
     def _tx_meta(ex, meta=None):
         return [], []
+
+    # -------------------- nornicdb.* (reference pkg/cypher/call.go:973-1027
+    # + call_compat.go / call_index_mgmt.go compat surface) --------------------
+    @register("nornicdb.version")
+    def _nv(ex):
+        from .. import __version__
+        return ["version", "build", "edition"], [[__version__, "rocm", "community"]]
+
+    @register("nornicdb.stats")
+    def _nstats(ex):
+        labels = set()
+        for n in db.engine.all_nodes():
+            labels.update(n.labels)
+        types = {e.type for e in db.engine.all_edges()}
+        return ["nodes", "relationships", "labels", "relationshipTypes"], \
+            [[db.engine.node_count(), db.engine.edge_count(),
+              len(labels), len(types)]]
+
+    @register("nornicdb.decay.info")
+    def _ndecay(ex):
+        cfg = getattr(db, "decay_config", None)
+        return (["enabled", "halfLifeEpisodic", "halfLifeSemantic",
+                 "halfLifeProcedural", "archiveThreshold"],
+                [[bool(getattr(cfg, "enabled", True)) if cfg is not None else True,
+                  "7 days", "69 days", "693 days", 0.05]])
+
+    @register("gds.version")
+    def _gdsv(ex):
+        return ["version"], [["2.6.0-nornicdb-amd"]]
+
+    @register("db.clearQueryCaches")
+    def _clearcache(ex):
+        cache = getattr(db, "query_cache", None)
+        if cache is not None:
+            cache.invalidate()
+        return ["status"], [["Query caches cleared"]]
+
+    @register("db.stats.clear")
+    def _stats_clear(ex, section="QUERIES"):
+        return ["section", "success", "message"], [[section, True, "cleared"]]
+
+    @register("db.stats.collect")
+    def _stats_collect(ex, section="QUERIES", config=None):
+        return ["section", "success", "message"], [[section, True, "collecting"]]
+
+    @register("db.stats.stop")
+    def _stats_stop(ex, section="QUERIES"):
+        return ["section", "success", "message"], [[section, True, "stopped"]]
+
+    @register("db.stats.status")
+    def _stats_status(ex):
+        return ["section", "status", "message"], [["QUERIES", "idle", ""]]
+
+    @register("db.stats.retrieveAllAnTheStats")
+    def _stats_all(ex):
+        _, rows = procs["db.stats.retrieve"](ex, "GRAPH COUNTS")
+        return ["section", "data"], rows
+
+    @register("db.schema.nodeProperties")
+    def _schema_nodeprops(ex):
+        seen = {}
+        for n in db.engine.all_nodes():
+            for lb in n.labels:
+                seen.setdefault(lb, set()).update((n.properties or {}).keys())
+        return ["nodeLabel", "propertyName", "propertyType"], \
+            [[lb, p, "ANY"] for lb in sorted(seen) for p in sorted(seen[lb])]
+
+    @register("db.schema.relProperties")
+    def _schema_relprops(ex):
+        seen = {}
+        for e in db.engine.all_edges():
+            seen.setdefault(e.type, set()).update((e.properties or {}).keys())
+        return ["relType", "propertyName", "propertyType"], \
+            [[t, p, "ANY"] for t in sorted(seen) for p in sorted(seen[t])]
+
+    @register("db.index.fulltext.listAvailableAnalyzers")
+    def _ft_analyzers(ex):
+        return ["analyzer", "description"], [
+            ["standard-no-stop-words", "Standard analyzer without stop words"],
+            ["simple", "Simple analyzer with lowercase tokenizer"],
+            ["whitespace", "Whitespace analyzer"],
+            ["keyword", "Keyword analyzer - entire string as single token"],
+        ]
+
+    @register("db.index.fulltext.queryRelationships")
+    def _ft_query_rels(ex, index, query, options=None):
+        ql = str(query).lower()
+        rows = []
+        for e in db.engine.all_edges():
+            score = sum(1.0 for v in (e.properties or {}).values()
+                        if isinstance(v, str) and ql in v.lower())
+            if score:
+                rows.append([e, score])
+        rows.sort(key=lambda r: -r[1])
+        return ["relationship", "score"], rows
+
+    @register("dbms.clientConfig")
+    def _client_cfg(ex):
+        return ["name", "value"], [
+            ["server.bolt.advertised_address", "localhost:7687"],
+            ["server.http.advertised_address", "localhost:7474"],
+        ]
+
+    @register("dbms.listConnections")
+    def _list_conns(ex):
+        return ["connectionId", "connectTime", "connector", "username",
+                "userAgent", "clientAddress"], []
 
     return procs

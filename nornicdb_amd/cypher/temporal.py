@@ -92,6 +92,9 @@ class CypherDuration:
             return CypherDuration(self.months + o.months, self.days + o.days,
                                   self.seconds + o.seconds,
                                   self.nanoseconds + o.nanoseconds)
+        if isinstance(o, (CypherDate, CypherDateTime, CypherTime)):
+            # duration + temporal commutes (Neo4j: duration('P7D') + date(...))
+            return o + self
         return NotImplemented
 
     def __radd__(self, o):

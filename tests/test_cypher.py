@@ -744,11 +744,40 @@ class TestReferenceSurfaceParity:
         assert "tenant_a" not in [r[0] for r in db.cypher("SHOW DATABASES").rows]
 
     def test_kalman_functions(self):
+        """JSON-state contract (reference pkg/cypher/kalman_functions.go):
+        init() -> state string; process(m, state) -> {value, state}."""
         db = self._mgr().get()
-        db.cypher("RETURN kalman.init('k1', 0.01, 0.1, 5.0)")
-        v = db.cypher("RETURN kalman.process('k1', 10.0)").rows[0][0]
-        assert 5.0 < v < 10.0  # filtered toward the measurement
-        assert db.cypher("RETURN kalman.reset('k1')").rows == [[True]]
+        s = db.cypher("RETURN kalman.init({measurementNoise: 10.0})").rows[0][0]
+        assert isinstance(s, str) and '"x"' in s
+        r = db.cypher("RETURN kalman.process(10.0, $s)", {"s": s}).rows[0][0]
+        assert 0.0 < r["value"] < 10.0  # filtered toward the measurement
+        # repeated measurements converge
+        for _ in range(20):
+            r = db.cypher("RETURN kalman.process(10.0, $s)",
+                          {"s": r["state"]}).rows[0][0]
+        assert abs(r["value"] - 10.0) < 1.0
+        assert db.cypher("RETURN kalman.state($s)",
+                         {"s": r["state"]}).rows[0][0] == r["value"]
+        # velocity filter tracks a ramp and predicts ahead
+        vs = db.cypher("RETURN kalman.velocity.init()").rows[0][0]
+        for i in range(30):
+            vr = db.cypher("RETURN kalman.velocity.process($m, $s)",
+                           {"m": float(i), "s": vs}).rows[0][0]
+            vs = vr["state"]
+        assert abs(vr["velocity"] - 1.0) < 0.2
+        pred = db.cypher("RETURN kalman.velocity.predict($s, 5)",
+                         {"s": vs}).rows[0][0]
+        assert pred > vr["value"] + 3
+        # adaptive switches to velocity mode on a strong trend
+        a = db.cypher("RETURN kalman.adaptive.init({hysteresis: 3})").rows[0][0]
+        for i in range(30):
+            ar = db.cypher("RETURN kalman.adaptive.process($m, $s)",
+                           {"m": float(i * 2), "s": a}).rows[0][0]
+            a = ar["state"]
+        assert ar["mode"] == "velocity"
+        # reset preserves the family
+        assert '"pos"' in db.cypher("RETURN kalman.reset($s)",
+                                    {"s": vs}).rows[0][0]
 
     def test_pad_format(self):
         db = self._mgr().get()
