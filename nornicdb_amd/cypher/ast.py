@@ -236,14 +236,16 @@ class WithClause(ReturnClause):
 class UnwindClause:
     expr: Any
     alias: str
+    where: Optional[Any] = None  # reference allows UNWIND ... AS x WHERE p
 
 
 @dataclass
 class CallClause:
     proc: str
     args: List[Any]
-    yields: List[Tuple[str, Optional[str]]]  # (name, alias)
+    yields: List[Tuple[str, Optional[str]]]  # (name, alias); ("*", None) = all
     where: Optional[Any] = None
+    limit: Optional[Any] = None
 
 
 @dataclass
@@ -266,6 +268,7 @@ class SchemaCommand:
     constraint_kind: Optional[str] = None   # "unique" | "exists" | "node_key"
     options: Optional[Any] = None           # MapLit AST for OPTIONS {...}
     if_not_exists: bool = False
+    type_filter: Optional[str] = None       # SHOW VECTOR|FULLTEXT|... INDEXES
     if_exists: bool = False
     or_replace: bool = False
 

@@ -371,6 +371,8 @@ class Executor:
                 rows = []
                 for i, (name, kind, label, props) in enumerate(
                         self.schema.list_indexes()):
+                    if c.type_filter and kind.upper() != c.type_filter:
+                        continue
                     rows.append([i + 1, name, "ONLINE", 100.0, kind.upper(),
                                  "NODE", [label], props])
                 return Result(["id", "name", "state", "populationPercent",
@@ -705,7 +707,8 @@ class Executor:
             if l is None or r is None:
                 return None
             return self._cy_eq(l, r)
-        if op == "<>":
+        if op in ("<>", "!="):
+            # != accepted as <> (reference executor_mutations.go:995)
             if l is None or r is None:
                 return None
             return not self._cy_eq(l, r)
@@ -1117,6 +1120,9 @@ class Executor:
             for item in v:
                 r2 = dict(row)
                 r2[c.alias] = item
+                if c.where is not None and \
+                        self._eval(c.where, r2, params) is not True:
+                    continue
                 out.append(r2)
         return out
 
@@ -1547,14 +1553,18 @@ class Executor:
         if proc is None:
             raise CypherRuntimeError(f"unknown procedure {c.proc}")
         out_rows = []
+        yields = c.yields
         for row in rows:
             args = [self._eval(a, row, params) for a in c.args]
             cols, prows = proc(self, *args)
+            if yields and yields[0][0] == "*":
+                # YIELD * expands to every procedure column
+                yields = [(col, None) for col in cols]
             for pr in prows:
                 rec = dict(zip(cols, pr)) if isinstance(pr, (list, tuple)) else dict(pr)
                 r2 = dict(row)
-                if c.yields:
-                    for (yname, alias) in c.yields:
+                if yields:
+                    for (yname, alias) in yields:
                         if yname not in rec:
                             raise CypherRuntimeError(
                                 f"procedure {c.proc} does not yield {yname}")
@@ -1564,14 +1574,17 @@ class Executor:
                 if c.where is not None and self._eval(c.where, r2, params) is not True:
                     continue
                 out_rows.append(r2)
-        if standalone and not c.yields:
+        if getattr(c, "limit", None) is not None:
+            lim = self._eval(c.limit, {}, params)
+            out_rows = out_rows[:int(lim)]
+        if standalone and not yields:
             # standalone CALL returns all procedure columns
             if out_rows:
                 cols = [k for k in out_rows[0].keys()]
                 return out_rows, (cols, [[r[k] for k in cols] for r in out_rows])
             return out_rows, ([], [])
-        if standalone and c.yields:
-            cols = [alias or y for y, alias in c.yields]
+        if standalone and yields:
+            cols = [alias or y for y, alias in yields]
             return out_rows, (cols, [[r[k] for k in cols] for r in out_rows])
         return out_rows, None
 

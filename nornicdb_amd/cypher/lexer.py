@@ -140,7 +140,7 @@ def tokenize(text: str) -> List[Token]:
             i = j
             continue
         # operators
-        for op in ("<>", "<=", ">=", "=~", "..", "+=", "->", "<-"):
+        for op in ("<>", "!=", "<=", ">=", "=~", "..", "+=", "->", "<-"):
             if text.startswith(op, i):
                 toks.append(Token("OP", op, i))
                 i += len(op)

@@ -441,6 +441,13 @@ class Parser:
         self.eat_kw("SHOW")
         t = self.next()
         word = t.value.upper()
+        type_filter = None
+        if word in ("VECTOR", "FULLTEXT", "TEXT", "RANGE", "POINT",
+                    "LOOKUP", "BTREE"):
+            # SHOW VECTOR INDEXES etc. — type-filtered index listing
+            type_filter = word
+            t = self.next()
+            word = t.value.upper()
         kinds = {"INDEX": "indexes", "INDEXES": "indexes",
                  "CONSTRAINT": "constraints", "CONSTRAINTS": "constraints",
                  "DATABASE": "databases", "DATABASES": "databases",
@@ -450,7 +457,7 @@ class Parser:
                  "TRANSACTIONS": "transactions",
                  "SETTING": "settings", "SETTINGS": "settings",
                  "ALIAS": "aliases", "ALIASES": "aliases"}
-        if word not in kinds:
+        if word not in kinds or (type_filter and kinds[word] != "indexes"):
             raise CypherSyntaxError(f"cannot SHOW {t.value!r}")
         # optional YIELD ... (accepted, ignored: full rows returned)
         if self.try_kw("YIELD"):
@@ -458,7 +465,8 @@ class Parser:
                 self.next()
                 if not self.try_op(","):
                     break
-        return A.SchemaCommand("show", kinds[word])
+        return A.SchemaCommand("show", kinds[word],
+                               type_filter=type_filter)
 
     def _merge(self):
         self.eat_kw("MERGE")
@@ -585,7 +593,9 @@ class Parser:
         self.eat_kw("UNWIND")
         e = self._expr()
         self.eat_kw("AS")
-        return A.UnwindClause(e, self.ident())
+        alias = self.ident()
+        where = self._expr() if self.try_kw("WHERE") else None
+        return A.UnwindClause(e, alias, where)
 
     def _call(self):
         self.eat_kw("CALL")
@@ -618,24 +628,45 @@ class Parser:
         args = []
         if self.try_op("("):
             if not self.at_op(")"):
-                args.append(self._expr())
-                while self.try_op(","):
+                # implicit-map call form: CALL p(key: v, key2: v2) — the
+                # reference's regex parser accepts bare config entries
+                # (gds tests: stream(sourceNode: 'x', topK: 5))
+                t0, t1 = self.peek(), self.peek(1)
+                if t0.kind in ("IDENT", "KW") and t1.kind == "OP" \
+                        and t1.value == ":":
+                    items = []
+                    while True:
+                        k = self.next().value
+                        self.eat_op(":")
+                        items.append((k, self._expr()))
+                        if not self.try_op(","):
+                            break
+                    args.append(A.MapLit(items))
+                else:
                     args.append(self._expr())
+                    while self.try_op(","):
+                        args.append(self._expr())
             self.eat_op(")")
         yields = []
         where = None
+        limit = None
         if self.try_kw("YIELD"):
-            while True:
-                y = self.ident()
-                alias = None
-                if self.try_kw("AS"):
-                    alias = self.ident()
-                yields.append((y, alias))
-                if not self.try_op(","):
-                    break
+            if self.try_op("*"):
+                yields.append(("*", None))
+            else:
+                while True:
+                    y = self.ident()
+                    alias = None
+                    if self.try_kw("AS"):
+                        alias = self.ident()
+                    yields.append((y, alias))
+                    if not self.try_op(","):
+                        break
             if self.try_kw("WHERE"):
                 where = self._expr()
-        return A.CallClause(name, args, yields, where)
+            if self.try_kw("LIMIT"):
+                limit = self._expr()
+        return A.CallClause(name, args, yields, where, limit)
 
     def _braced_query(self) -> A.Query:
         """Parse `{ <clauses> }`; a bare pattern (EXISTS shorthand) is
@@ -794,6 +825,10 @@ class Parser:
                     k = self.next().value.lower()
                 else:
                     raise CypherSyntaxError(f"bad map key at {t.pos}")
+                # dotted keys: OPTIONS {indexConfig: {vector.dimensions: N}}
+                while t.kind != "STRING" and self.at_op("."):
+                    self.next()
+                    k += "." + self.name_part()
                 self.eat_op(":")
                 items.append((k, self._expr()))
                 if not self.try_op(","):
@@ -836,7 +871,8 @@ class Parser:
         ops = []
         while True:
             t = self.peek()
-            if t.kind == "OP" and t.value in ("=", "<>", "<", ">", "<=", ">=", "=~"):
+            if t.kind == "OP" and t.value in ("=", "<>", "!=", "<", ">",
+                                              "<=", ">=", "=~"):
                 op = self.next().value
                 rhs = self._addsub()
                 ops.append((op, rhs))
@@ -1124,6 +1160,13 @@ class Parser:
                 body = self._expr()
                 self.eat_op(")")
                 return A.Reduce(acc, init, var, source, body)
+            # COLLECT { ... } subquery (Cypher 5; executor returns the
+            # first return column per row)
+            if (t.value.lower() == "collect" and self.peek(1).kind == "OP"
+                    and self.peek(1).value == "{"):
+                self.next()
+                return A.SubqueryExpr(
+                    "COLLECT", self._ensure_return(self._braced_query()))
             # function call?
             if self.peek(1).kind == "OP" and self.peek(1).value == "(":
                 name = self.next().value

@@ -263,8 +263,10 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
         from ..cognitive.linkpredict import (adamic_adar, common_neighbors,
                                              jaccard,
                                              preferential_attachment)
+        from ..cognitive.linkpredict import resource_allocation
         fn = {"adamic": adamic_adar, "common": common_neighbors,
-              "jaccard": jaccard, "pref": preferential_attachment}[scorer]
+              "jaccard": jaccard, "pref": preferential_attachment,
+              "resource": resource_allocation}[scorer]
         rows = [[a, b, float(fn(db.engine, a, b))] for a, b in _lp_pairs()]
         rows.sort(key=lambda r: -r[2])
         return ["node1", "node2", "score"], rows
@@ -280,6 +282,10 @@ def build_procedures(db: "NornicDB") -> Dict[str, Any]:
     @register("gds.linkPrediction.jaccard.stream")
     def _gds_lp_j(ex, config=None):
         return _lp_stream("jaccard")
+
+    @register("gds.linkPrediction.resourceAllocation.stream")
+    def _gds_lp_ra(ex, config=None):
+        return _lp_stream("resource")
 
     @register("gds.linkPrediction.preferentialAttachment.stream")
     def _gds_lp_pa(ex, config=None):

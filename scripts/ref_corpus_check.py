@@ -20,6 +20,11 @@ def main(path=None):
         path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                             "ref_queries.txt")
     qs = [l for l in open(path).read().splitlines() if l.strip()]
+    # extraction kept Go escape sequences literally; restore the real
+    # whitespace the reference executed (whitespaceVariations tests)
+    qs = [q.replace("\\n", "\n").replace("\\t", "\t").replace("\\r", "\r")
+            .replace("\\\\", "\\")
+          if "\\" in q else q for q in qs]
     db = DatabaseManager(MemoryEngine()).get()
     ok = pf = rf = 0
     for q in qs:

@@ -836,3 +836,89 @@ class TestTrimSpecForm:
             "RETURN trim(BOTH 'x' FROM 'xxaxx'), "
             "trim(LEADING 'x' FROM 'xxa'), trim(TRAILING FROM 'a  ')"
         ).rows == [["a", "a", "a"]]
+
+
+class TestGrammarExtensions:
+    """Reference-parity grammar forms (round-1 corpus batch 4):
+    != (executor_mutations.go:995), UNWIND..WHERE, COLLECT{} subquery,
+    YIELD * / WHERE / LIMIT, implicit-map CALL args, dotted OPTIONS keys."""
+
+    def _db(self):
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.storage.memory import MemoryEngine
+        return DatabaseManager(MemoryEngine()).get()
+
+    def test_not_equals_operator(self):
+        db = self._db()
+        db.cypher("CREATE (:P {name:'a'}), (:P {name:'b'})")
+        assert db.cypher("MATCH (n:P) WHERE n.name != 'a' "
+                         "RETURN n.name").rows == [["b"]]
+        assert db.cypher("RETURN 1 != 2, 1 != 1").rows == [[True, False]]
+
+    def test_unwind_where(self):
+        db = self._db()
+        assert db.cypher("UNWIND [1,2,3,4] AS x WHERE x > 2 "
+                         "RETURN collect(x)").rows == [[[3, 4]]]
+
+    def test_collect_subquery(self):
+        db = self._db()
+        db.cypher("CREATE (a:P {name:'a'})-[:KNOWS]->(:P {name:'b'}), "
+                  "(a)-[:KNOWS]->(:P {name:'c'})")
+        r = db.cypher("MATCH (p:P {name:'a'}) RETURN collect { MATCH "
+                      "(p)-[:KNOWS]->(f) RETURN f.name ORDER BY f.name }")
+        assert r.rows == [[["b", "c"]]]
+
+    def test_yield_star_where_limit(self):
+        db = self._db()
+        db.cypher("CREATE (:L), (:M), (:N)")
+        assert db.cypher("CALL db.labels() YIELD *").rows == [
+            ["L"], ["M"], ["N"]]
+        assert db.cypher("CALL db.labels() YIELD label LIMIT 2").rows == [
+            ["L"], ["M"]]
+        assert db.cypher("CALL db.labels() YIELD * WHERE label = 'M' "
+                         "RETURN label").rows == [["M"]]
+
+    def test_implicit_map_call_args(self):
+        db = self._db()
+        r = db.cypher("CALL gds.linkPrediction.adamicAdar.stream("
+                      "sourceNode: 'x', topK: 5) YIELD node1 "
+                      "RETURN count(*)")
+        assert r.rows == [[0]]
+
+    def test_dotted_options_map_keys(self):
+        db = self._db()
+        db.cypher("CREATE VECTOR INDEX vvi IF NOT EXISTS FOR (n:Doc) "
+                  "ON (n.embedding) OPTIONS {indexConfig: "
+                  "{ vector.dimensions: 4, "
+                  "vector.similarity_function: 'cosine' }}")
+        assert [r[1] for r in db.cypher("SHOW VECTOR INDEXES").rows] == ["vvi"]
+
+    def test_duration_date_commute_and_single_arg_forms(self):
+        db = self._db()
+        assert str(db.cypher("RETURN duration('P7D') + date('2025-01-01')")
+                   .rows[0][0]) == "2025-01-08"
+        assert str(db.cypher("RETURN duration.inDays(duration('P10D'))")
+                   .rows[0][0]) == "P10D"
+        assert str(db.cypher("RETURN duration.inSeconds(duration('PT1H'))")
+                   .rows[0][0]) == "PT1H"
+
+    def test_new_procedures_surface(self):
+        db = self._db()
+        db.cypher("CREATE (a:N {name:'A'})-[:CONNECTS {weight: 1.0}]->"
+                  "(b:N {name:'B'})-[:CONNECTS {weight: 1.0}]->"
+                  "(d:N {name:'D'})")
+        assert db.cypher("CALL nornicdb.version() YIELD version "
+                         "RETURN version").rows[0][0]
+        assert db.cypher("CALL nornicdb.stats()").rows == [[3, 2, 1, 1]]
+        assert db.cypher("CALL gds.version()").rows[0][0].startswith("2.6")
+        # string node refs resolve by name property; missing -> empty
+        assert db.cypher("CALL apoc.algo.dijkstra('A', 'D', 'CONNECTS', "
+                         "'weight') YIELD weight RETURN weight").rows == [[2.0]]
+        assert db.cypher("CALL apoc.algo.dijkstra('A', 'ZZZ', 'CONNECTS', "
+                         "'weight') YIELD weight RETURN weight").rows == []
+        assert db.cypher("CALL apoc.algo.allSimplePaths('A', 'D', "
+                         "'CONNECTS', 10) YIELD path RETURN count(path)"
+                         ).rows == [[1]]
+        rows = db.cypher("CALL apoc.neighbors.byhop('A', 'CONNECTS', 3) "
+                         "YIELD nodes, depth RETURN depth, size(nodes)").rows
+        assert rows == [[1, 1], [2, 1]]
