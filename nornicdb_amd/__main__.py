@@ -54,6 +54,12 @@ def cmd_serve(args):
     bolt = BoltServer(lambda db: mgr.get(db).executor,
                       host=cfg.bolt_host, port=args.bolt_port or cfg.bolt_port,
                       authenticator=auth)
+    grpc_server = None
+    if getattr(args, "grpc_port", None):
+        from .server.nornic_grpc import serve as grpc_serve
+        grpc_server, gport = grpc_serve(mgr, host=cfg.http_host,
+                                        port=args.grpc_port)
+        print(f"gRPC (NornicSearch) listening on {cfg.http_host}:{gport}")
 
     async def main():
         await bolt.start()
@@ -70,6 +76,8 @@ def cmd_serve(args):
     except KeyboardInterrupt:
         pass
     finally:
+        if grpc_server is not None:
+            grpc_server.stop(0)
         mgr.close()
 
 
@@ -217,6 +225,8 @@ def main(argv=None):
     sp.add_argument("--data-dir", default=None)
     sp.add_argument("--bolt-port", type=int, default=None)
     sp.add_argument("--http-port", type=int, default=None)
+    sp.add_argument("--grpc-port", type=int, default=None,
+                    help="enable the native NornicSearch gRPC API")
     sp.add_argument("--auth", action="store_true")
     sp.set_defaults(fn=cmd_serve)
 
