@@ -60,6 +60,14 @@ def cmd_serve(args):
         grpc_server, gport = grpc_serve(mgr, host=cfg.http_host,
                                         port=args.grpc_port)
         print(f"gRPC (NornicSearch) listening on {cfg.http_host}:{gport}")
+    qdrant_grpc_server = None
+    if getattr(args, "qdrant_grpc_port", None):
+        # share the HTTP app's registry so REST + gRPC see one store
+        from .server.qdrant_grpc import serve as qg_serve
+        qdrant_grpc_server, qport, _ = qg_serve(
+            app.state.qdrant, host=cfg.http_host,
+            port=args.qdrant_grpc_port)
+        print(f"gRPC (qdrant compat) listening on {cfg.http_host}:{qport}")
 
     async def main():
         await bolt.start()
@@ -78,6 +86,8 @@ def cmd_serve(args):
     finally:
         if grpc_server is not None:
             grpc_server.stop(0)
+        if qdrant_grpc_server is not None:
+            qdrant_grpc_server.stop(0)
         mgr.close()
 
 
@@ -227,6 +237,9 @@ def main(argv=None):
     sp.add_argument("--http-port", type=int, default=None)
     sp.add_argument("--grpc-port", type=int, default=None,
                     help="enable the native NornicSearch gRPC API")
+    sp.add_argument("--qdrant-grpc-port", type=int, default=None,
+                    help="enable the Qdrant-compatible gRPC endpoint "
+                         "(Qdrant default: 6334)")
     sp.add_argument("--auth", action="store_true")
     sp.set_defaults(fn=cmd_serve)
 
