@@ -343,8 +343,9 @@ class GraphQLExecutor:
                     for i in eng.neighbors(args["id"])]
         if name == "neighborhood":
             depth = int(args.get("depth", 1))
-            seen = {args["id"]}
-            frontier = {args["id"]}
+            start = args.get("nodeId", args.get("id"))
+            seen = {start}
+            frontier = {start}
             for _ in range(depth):
                 nxt = set()
                 for nid in frontier:
@@ -357,15 +358,30 @@ class GraphQLExecutor:
             return {"nodes": [self._node(n, None) for n in nodes],
                     "relationships": []}
         if name == "shortestPath":
+            a = args.get("startNodeId", args.get("from"))
+            b = args.get("endNodeId", args.get("to"))
+            depth = int(args.get("maxDepth", 10))
             r = self.db.cypher(
-                "MATCH p = shortestPath((a)-[*1..10]->(b)) "
-                "WHERE id(a) = $a AND id(b) = $b RETURN p",
-                {"a": args["from"], "b": args["to"]})
+                "MATCH (a), (b) WHERE id(a) = $a AND id(b) = $b "
+                f"MATCH p = shortestPath((a)-[*1..{depth}]->(b)) RETURN p",
+                {"a": a, "b": b})
             if not r.rows:
                 return None
             p = r.rows[0][0]
             return {"nodes": [self._node(n, None) for n in p.nodes],
                     "relationships": [self._rel(e) for e in p.edges]}
+        if name == "allPaths":
+            # schema.graphql allPaths: [[Node!]!]! — lists of node paths
+            a = args.get("startNodeId", args.get("from"))
+            b = args.get("endNodeId", args.get("to"))
+            depth = int(args.get("maxDepth", 5))
+            limit = int(args.get("limit", 10))
+            r = self.db.cypher(
+                "MATCH (a), (b) WHERE id(a) = $a AND id(b) = $b "
+                f"MATCH p = (a)-[*1..{depth}]->(b) RETURN p LIMIT $lim",
+                {"a": a, "b": b, "lim": limit})
+            return [[self._node(n, None) for n in row[0].nodes]
+                    for row in r.rows]
         if name == "search":
             qtext = args.get("query") or args.get("text", "")
             k = int(args.get("limit", args.get("k", 10)))

@@ -244,3 +244,41 @@ class TestQdrantExtended:
             "filter": {"must": [{"key": "n", "range": {"gte": 2}}]}
         }).json()["result"]["count"]
         assert r == 2
+
+
+class TestGraphQLPaths:
+    """shortestPath/allPaths/neighborhood with the reference schema's
+    argument names (schema.graphql: startNodeId/endNodeId/maxDepth)."""
+
+    def _gq(self):
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.storage.memory import MemoryEngine
+        from nornicdb_amd.server.graphql import GraphQLExecutor
+        db = DatabaseManager(MemoryEngine()).get()
+        db.cypher("CREATE (a:P {name:'a'})-[:R]->(b:P {name:'b'})"
+                  "-[:R]->(c:P {name:'c'})")
+        db.cypher("MATCH (a:P {name:'a'}), (c:P {name:'c'}) "
+                  "CREATE (a)-[:R]->(c)")
+        ids = {r[1]: r[0] for r in
+               db.cypher("MATCH (n:P) RETURN id(n), n.name").rows}
+        return GraphQLExecutor(db), ids
+
+    def test_shortest_path(self):
+        gq, ids = self._gq()
+        r = gq.execute('query($a: ID!, $b: ID!) { shortestPath('
+                       'startNodeId: $a, endNodeId: $b) { nodes { id } } }',
+                       {"a": ids["a"], "b": ids["c"]})
+        assert len(r["data"]["shortestPath"]["nodes"]) == 2
+
+    def test_all_paths(self):
+        gq, ids = self._gq()
+        r = gq.execute('query($a: ID!, $b: ID!) { allPaths(startNodeId: $a, '
+                       'endNodeId: $b, maxDepth: 4) { id } }',
+                       {"a": ids["a"], "b": ids["c"]})
+        assert len(r["data"]["allPaths"]) == 2  # direct + via b
+
+    def test_neighborhood_node_id_arg(self):
+        gq, ids = self._gq()
+        r = gq.execute('query($n: ID!) { neighborhood(nodeId: $n, depth: 1) '
+                       '{ nodes { id } } }', {"n": ids["a"]})
+        assert len(r["data"]["neighborhood"]["nodes"]) == 3
