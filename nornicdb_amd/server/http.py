@@ -148,6 +148,149 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
             raise HTTPException(401, str(e))
         return {"token": token}
 
+    @app.post("/auth/logout")
+    def logout(_user=Depends(check_auth)):
+        return {"status": "ok"}  # tokens are stateless JWTs
+
+    @app.get("/auth/config")
+    def auth_config():
+        return {"enabled": auth is not None, "oauth": False,
+                "methods": ["password", "token"] if auth else []}
+
+    @app.get("/auth/me")
+    def auth_me(user=Depends(check_auth)):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        return {"username": user.get("sub"), "role": user.get("role")}
+
+    @app.post("/auth/password")
+    def auth_password(body: Dict[str, Any], user=Depends(check_auth)):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        try:
+            auth.login(user.get("sub"), body.get("current", ""))
+        except Exception:
+            raise HTTPException(403, "current password incorrect")
+        auth.set_password(user.get("sub"), body.get("password", ""))
+        return {"status": "ok"}
+
+    @app.post("/auth/api-token")
+    def auth_api_token(body: Dict[str, Any], user=Depends(check_auth)):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        token = auth.issue_token(body.get("username", user.get("sub")),
+                                 body.get("password", ""))
+        return {"token": token}
+
+    @app.get("/auth/users")
+    def auth_users(user=Depends(check_auth)):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        auth.require(user, "admin")
+        return {"users": auth.list_users()}
+
+    @app.post("/auth/users")
+    def auth_create_user(body: Dict[str, Any], user=Depends(check_auth)):
+        if auth is None:
+            raise HTTPException(404, "auth disabled")
+        auth.require(user, "admin")
+        role = {"reader": "readonly", "editor": "readwrite",
+                "publisher": "readwrite"}.get(body.get("role", "readonly"),
+                                              body.get("role", "readonly"))
+        auth.create_user(body["username"], body["password"], role=role)
+        return {"status": "created"}
+
+    # ---- admin (reference server_router.go /admin/*) ----
+    @app.get("/admin/config")
+    def admin_config(_user=Depends(check_auth)):
+        from ..utils.config import Config
+        cfg = Config()
+        return {k: getattr(cfg, k) for k in sorted(vars(cfg))
+                if not k.startswith("_")}
+
+    @app.get("/admin/stats")
+    def admin_stats(_user=Depends(check_auth)):
+        db = mgr.get()
+        labels = {}
+        for n in db.engine.all_nodes():
+            for lb in n.labels:
+                labels[lb] = labels.get(lb, 0) + 1
+        return {"nodes": db.engine.node_count(),
+                "relationships": db.engine.edge_count(),
+                "labels": labels, "databases": mgr.list()}
+
+    # GPU admin (reference /admin/gpu/*; here: the MI355X torch backend)
+    _gpu_enabled = [True]
+
+    @app.get("/admin/gpu/status")
+    def gpu_status():
+        import torch
+        avail = torch.cuda.is_available()
+        info = {"available": avail, "enabled": _gpu_enabled[0],
+                "backend": "rocm-hip" if avail else "cpu"}
+        if avail:
+            info["device"] = torch.cuda.get_device_name(0)
+            info["devices"] = torch.cuda.device_count()
+            free, total = torch.cuda.mem_get_info(0)
+            info["memory"] = {"free": free, "total": total}
+        return info
+
+    @app.post("/admin/gpu/enable")
+    def gpu_enable(_user=Depends(check_auth)):
+        _gpu_enabled[0] = True
+        return {"enabled": True}
+
+    @app.post("/admin/gpu/disable")
+    def gpu_disable(_user=Depends(check_auth)):
+        _gpu_enabled[0] = False
+        return {"enabled": False}
+
+    @app.post("/admin/gpu/test")
+    def gpu_test(_user=Depends(check_auth)):
+        import torch
+        if not torch.cuda.is_available():
+            return {"ok": False, "reason": "no GPU visible"}
+        t0 = time.time()
+        a = torch.randn(512, 512, device="cuda")
+        s = float((a @ a).sum())
+        torch.cuda.synchronize()
+        return {"ok": s == s, "elapsed_ms": round((time.time() - t0) * 1e3, 2)}
+
+    # ---- embed queue + index admin (reference /nornicdb/embed/*) ----
+    @app.get("/nornicdb/embed/stats")
+    def embed_stats():
+        db = mgr.get()
+        pending = len(db.engine.pending_embeddings(10_000))
+        return {"pending": pending, "indexed": len(db.search.emb)}
+
+    @app.post("/nornicdb/embed/trigger")
+    def embed_trigger(_user=Depends(check_auth)):
+        db = mgr.get()
+        done = db.embed_queue.drain(timeout=30.0) if db.embed_queue else True
+        return {"drained": done}
+
+    @app.post("/nornicdb/embed/clear")
+    def embed_clear(_user=Depends(check_auth)):
+        db = mgr.get()
+        ids = db.engine.pending_embeddings(1_000_000)
+        for nid in ids:
+            db.engine.clear_pending_embedding(nid)
+        return {"cleared": len(ids)}
+
+    @app.post("/nornicdb/search/rebuild")
+    def search_rebuild(_user=Depends(check_auth)):
+        db = mgr.get()
+        db.search.build_indexes()
+        return {"status": "rebuilt", "indexed": len(db.search.emb)}
+
+    @app.get("/nornicdb/decay")
+    def decay_info():
+        db = mgr.get()
+        cfg = getattr(db, "decay_config", None)
+        return {"enabled": cfg is not None,
+                "halfLife": {"episodic": "7 days", "semantic": "69 days",
+                             "procedural": "693 days"}}
+
     # ---- Neo4j HTTP transaction API ----
     # raw-body endpoint (no pydantic model): this is the hot path and
     # schema validation costs more than the query itself
@@ -344,11 +487,47 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
                 deleted += 1
         return {"subject": subject, "deleted": deleted}
 
-    # ---- MCP (JSON-RPC 2.0) ----
+    # ---- MCP (JSON-RPC 2.0 + the reference's REST aliases) ----
     @app.post("/mcp")
     async def mcp_endpoint(request: Request):
         body = await request.json()
         return mcp.handle(body)
+
+    @app.get("/mcp/health")
+    def mcp_health():
+        return {"status": "ok", "protocol": "mcp"}
+
+    @app.post("/mcp/initialize")
+    def mcp_initialize():
+        return mcp.handle({"jsonrpc": "2.0", "id": 1,
+                           "method": "initialize"})["result"]
+
+    @app.get("/mcp/tools/list")
+    def mcp_tools_list():
+        return mcp.handle({"jsonrpc": "2.0", "id": 1,
+                           "method": "tools/list"})["result"]
+
+    @app.post("/mcp/tools/call")
+    async def mcp_tools_call(request: Request):
+        body = await request.json()
+        out = mcp.handle({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                          "params": body})
+        return out.get("result", out)
+
+    @app.get("/graphql/playground")
+    def graphql_playground():
+        return Response(
+            "<!doctype html><title>GraphQL</title><body>"
+            "<h3>NornicDB-AMD GraphQL</h3>"
+            "<p>POST /graphql with {query, variables}; "
+            "subscriptions stream from GET /graphql/stream (SSE).</p>"
+            "<textarea id=q rows=8 cols=80>{ nodeCount }</textarea><br>"
+            "<button onclick=\"fetch('/graphql',{method:'POST',"
+            "headers:{'Content-Type':'application/json'},"
+            "body:JSON.stringify({query:document.getElementById('q').value})})"
+            ".then(r=>r.json()).then(d=>document.getElementById('o')"
+            ".textContent=JSON.stringify(d,null,2))\">run</button>"
+            "<pre id=o></pre></body>", media_type="text/html")
 
     # ---- Heimdall / Bifrost (reference pkg/heimdall/bifrost.go SSE) ----
     _heimdall = [None]
@@ -396,6 +575,38 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         return {"db": h.db_metrics(), "generation": h.stats,
                 "plugins": h.plugin_health(),
                 "tokens_per_second": h.tokens_per_second()}
+
+    # reference route aliases (server_router.go /api/bifrost/*)
+    @app.post("/api/bifrost/chat/completions")
+    def bifrost_chat_completions(body: Dict[str, Any],
+                                 _user=Depends(check_auth)):
+        # OpenAI-shaped wrapper over chat
+        out = bifrost_chat(body, _user)
+        return {"object": "chat.completion",
+                "choices": [{"index": 0, "finish_reason": "stop",
+                             "message": {"role": "assistant",
+                                         "content": out["text"]}}]}
+
+    @app.get("/api/bifrost/status")
+    def bifrost_status():
+        try:
+            h = get_heimdall()
+            return {"available": True, "decoder": type(h.decoder).__name__
+                    if getattr(h, "decoder", None) else "eager"}
+        except Exception as e:
+            return {"available": False, "reason": str(e)}
+
+    @app.get("/api/bifrost/events")
+    def bifrost_events(_user=Depends(check_auth)):
+        from starlette.responses import StreamingResponse
+
+        def sse():
+            h = get_heimdall()
+            import json as _json
+            yield "data: " + _json.dumps(
+                {"event": "status", "stats": h.stats}) + "\n\n"
+
+        return StreamingResponse(sse(), media_type="text/event-stream")
 
     # ---- GraphQL (reference pkg/graphql) ----
     from .graphql import GraphQLExecutor

@@ -173,3 +173,83 @@ class TestExplicitTransactions:
         txid = r.headers["Location"].rstrip("/").rsplit("/", 1)[-1]
         r2 = client.delete(f"/db/neo4j/tx/{txid}")
         assert r2.status_code == 200
+
+
+class TestAdminAuthRoutes:
+    """Reference server_router.go surface: /auth/*, /admin/*, /mcp REST
+    aliases, embed/index admin, /graphql/playground."""
+
+    def _client(self, with_auth=False):
+        from fastapi.testclient import TestClient
+        from nornicdb_amd.db import DatabaseManager
+        from nornicdb_amd.storage.memory import MemoryEngine
+        from nornicdb_amd.server.http import create_app
+        mgr = DatabaseManager(MemoryEngine())
+        auth = None
+        if with_auth:
+            from nornicdb_amd.auth import Authenticator
+            auth = Authenticator(mgr.get().engine)
+            auth.ensure_admin(password="secret123")
+        return TestClient(create_app(mgr, auth=auth)), mgr
+
+    def test_admin_and_gpu_status(self):
+        c, mgr = self._client()
+        mgr.get().cypher("CREATE (:Doc {t: 'x'})")
+        assert c.get("/admin/stats").json()["nodes"] == 1
+        assert "bolt_host" in c.get("/admin/config").json()
+        gpu = c.get("/admin/gpu/status").json()
+        assert "available" in gpu and "backend" in gpu
+        assert c.post("/admin/gpu/disable").json()["enabled"] is False
+        assert c.post("/admin/gpu/enable").json()["enabled"] is True
+        # no GPU in CI: test endpoint reports gracefully
+        assert c.post("/admin/gpu/test").status_code == 200
+
+    def test_embed_and_rebuild(self):
+        c, mgr = self._client()
+        mgr.get().cypher("CREATE (:Doc {title: 'hello'})")
+        assert c.get("/nornicdb/embed/stats").status_code == 200
+        assert c.post("/nornicdb/search/rebuild").json()["status"] == "rebuilt"
+        assert c.post("/nornicdb/embed/clear").status_code == 200
+        assert c.get("/nornicdb/decay").status_code == 200
+
+    def test_mcp_rest_aliases(self):
+        c, _ = self._client()
+        assert c.get("/mcp/health").json()["status"] == "ok"
+        tools = c.get("/mcp/tools/list").json()["tools"]
+        assert any(t["name"] == "recall" for t in tools)
+        init = c.post("/mcp/initialize").json()
+        assert init["serverInfo"]["name"] == "nornicdb-amd"
+
+    def test_graphql_playground(self):
+        c, _ = self._client()
+        r = c.get("/graphql/playground")
+        assert r.status_code == 200 and "GraphQL" in r.text
+
+    def test_auth_user_lifecycle(self):
+        c, _ = self._client(with_auth=True)
+        tok = c.post("/auth/login", json={
+            "username": "neo4j", "password": "secret123"}).json()["token"]
+        H = {"Authorization": f"Bearer {tok}"}
+        me = c.get("/auth/me", headers=H).json()
+        assert me == {"username": "neo4j", "role": "admin"}
+        assert c.post("/auth/users", headers=H, json={
+            "username": "bob", "password": "pw1234567",
+            "role": "reader"}).status_code == 200
+        users = {u["username"] for u in
+                 c.get("/auth/users", headers=H).json()["users"]}
+        assert users == {"neo4j", "bob"}
+        assert c.post("/auth/password", headers=H, json={
+            "current": "secret123", "password": "newpass99"}).status_code == 200
+        assert c.post("/auth/login", json={
+            "username": "neo4j", "password": "newpass99"}).status_code == 200
+        assert c.post("/auth/logout", headers=H).status_code == 200
+        cfg = c.get("/auth/config").json()
+        assert cfg["enabled"] is True
+
+    def test_auth_me_requires_token(self):
+        c, _ = self._client(with_auth=True)
+        assert c.get("/auth/me").status_code == 401
+
+    def test_bifrost_status_alias(self):
+        c, _ = self._client()
+        assert "available" in c.get("/api/bifrost/status").json()
