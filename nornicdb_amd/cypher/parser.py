@@ -1221,14 +1221,60 @@ class Parser:
             test = self._expr()
         whens = []
         while self.try_kw("WHEN"):
-            cond = self._expr()
+            if test is not None:
+                # Extended simple form (Neo4j 5): each WHEN is a
+                # comma-separated list of candidates, each either a value
+                # (equality) or a comparison applied to the operand.
+                # Desugared here to a general-form boolean condition so
+                # null-semantics match Cypher `=` (WHEN null never matches).
+                cond = self._case_alt(test)
+                while self.try_op(","):
+                    cond = A.BinOp("OR", cond, self._case_alt(test))
+            else:
+                cond = self._expr()
             self.eat_kw("THEN")
             whens.append((cond, self._expr()))
         default = None
         if self.try_kw("ELSE"):
             default = self._expr()
         self.eat_kw("END")
-        return A.Case(test, whens, default)
+        return A.Case(None, whens, default) if test is not None \
+            else A.Case(test, whens, default)
+
+    def _case_alt(self, test):
+        """One alternative of an extended simple-form CASE WHEN."""
+        t = self.peek()
+        if t.kind == "OP" and t.value in ("=", "<>", "!=", "<", ">",
+                                          "<=", ">=", "=~"):
+            return A.BinOp(self.next().value, test, self._addsub())
+        if self.at_kw("IS"):
+            self.next()
+            neg = self.try_kw("NOT")
+            if self.at_op("::") or self.at_op(":"):
+                if not self.try_op("::"):
+                    self.eat_op(":")
+                    self.eat_op(":")
+                tname = self.next().value.upper()
+                if tname in ("LOCAL", "ZONED"):
+                    tname += " " + self.next().value.upper()
+                return A.TypePredicate(test, tname, negated=neg)
+            self.eat_kw("NULL")
+            return A.UnOp("IS NOT NULL" if neg else "IS NULL", test)
+        if self.at_kw("STARTS"):
+            self.next()
+            self.eat_kw("WITH")
+            return A.BinOp("STARTS WITH", test, self._addsub())
+        if self.at_kw("ENDS"):
+            self.next()
+            self.eat_kw("WITH")
+            return A.BinOp("ENDS WITH", test, self._addsub())
+        if self.at_kw("CONTAINS"):
+            self.next()
+            return A.BinOp("CONTAINS", test, self._addsub())
+        if self.at_kw("IN"):
+            self.next()
+            return A.BinOp("IN", test, self._addsub())
+        return A.BinOp("=", test, self._addsub())
 
     def _expr_atom_chain(self):
         """Left side of SET: variable with optional property chain."""

@@ -203,6 +203,27 @@ class TestExpressions:
             "RETURN CASE WHEN x = 1 THEN 'one' WHEN x = 2 THEN 'two' ELSE 'many' END")
         assert [x[0] for x in r.rows] == ["one", "two", "many"]
 
+    def test_case_extended_simple_form(self, ex):
+        # Neo4j 5 extended simple CASE: candidate lists and comparisons
+        # applied to the operand (reference: pkg/cypher expression tests).
+        r = ex.execute("RETURN CASE 2 WHEN 1, 2 THEN 'low' ELSE 'high' END")
+        assert r.rows == [["low"]]
+        r = ex.execute(
+            "RETURN CASE 7 WHEN 1, 2 THEN 'low' WHEN > 5 THEN 'big' END")
+        assert r.rows == [["big"]]
+        r = ex.execute(
+            "RETURN CASE 'abc' WHEN STARTS WITH 'a' THEN 1 ELSE 0 END")
+        assert r.rows == [[1]]
+        # simple-form equality is Cypher `=`: WHEN null never matches
+        r = ex.execute("RETURN CASE null WHEN null THEN 'm' ELSE 'no' END")
+        assert r.rows == [["no"]]
+        r = ex.execute("RETURN CASE null WHEN IS NULL THEN 'y' ELSE 'n' END")
+        assert r.rows == [["y"]]
+        r = ex.execute("RETURN CASE 3 WHEN IN [1,3,5] THEN 'odd' ELSE 'x' END")
+        assert r.rows == [["odd"]]
+        r = ex.execute("RETURN CASE 5 WHEN IS :: INTEGER THEN 'i' ELSE 'n' END")
+        assert r.rows == [["i"]]
+
     def test_null_semantics(self, ex):
         r = ex.execute("RETURN null = null, null IS NULL, 1 + null, coalesce(null, 5)")
         assert r.rows == [[None, True, None, 5]]
