@@ -561,14 +561,15 @@ class Executor:
                 return None
         if isinstance(e, A.Case):
             if e.test is not None:
-                t = self._eval(e.test, row, params)
-                for w, r in e.whens:
-                    if self._eval(w, row, params) == t:
-                        return self._eval(r, row, params)
-            else:
-                for w, r in e.whens:
-                    if self._eval(w, row, params) is True:
-                        return self._eval(r, row, params)
+                # Simple form: the operand is evaluated EXACTLY ONCE and
+                # bound to the synthetic __case__ variable the parser's
+                # desugared WHEN conditions reference (Neo4j semantics for
+                # non-deterministic operands like rand()).
+                row = dict(row)
+                row["__case__"] = self._eval(e.test, row, params)
+            for w, r in e.whens:
+                if self._eval(w, row, params) is True:
+                    return self._eval(r, row, params)
             return self._eval(e.default, row, params) if e.default else None
         if isinstance(e, A.ListComp):
             src = self._eval(e.source, row, params) or []
@@ -1425,6 +1426,8 @@ class Executor:
         if isinstance(e, A.MapLit):
             return any(self._contains_aggregate(v) for _, v in e.items)
         if isinstance(e, A.Case):
+            if e.test is not None and self._contains_aggregate(e.test):
+                return True
             return any(self._contains_aggregate(w) or self._contains_aggregate(t)
                        for w, t in e.whens)
         return False

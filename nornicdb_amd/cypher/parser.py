@@ -1220,16 +1220,20 @@ class Parser:
         if not self.at_kw("WHEN"):
             test = self._expr()
         whens = []
+        # Extended simple form (Neo4j 5): each WHEN is a comma-separated
+        # list of candidates, each either a value (equality) or a
+        # comparison applied to the operand. Desugared to general-form
+        # boolean conditions over the synthetic variable __case__, which
+        # the executor binds to the operand value evaluated EXACTLY ONCE
+        # (Neo4j semantics; a non-deterministic operand like rand() must
+        # not be re-evaluated per alternative). Null-semantics match
+        # Cypher `=` (WHEN null never matches).
+        operand = A.Var("__case__") if test is not None else None
         while self.try_kw("WHEN"):
             if test is not None:
-                # Extended simple form (Neo4j 5): each WHEN is a
-                # comma-separated list of candidates, each either a value
-                # (equality) or a comparison applied to the operand.
-                # Desugared here to a general-form boolean condition so
-                # null-semantics match Cypher `=` (WHEN null never matches).
-                cond = self._case_alt(test)
+                cond = self._case_alt(operand)
                 while self.try_op(","):
-                    cond = A.BinOp("OR", cond, self._case_alt(test))
+                    cond = A.BinOp("OR", cond, self._case_alt(operand))
             else:
                 cond = self._expr()
             self.eat_kw("THEN")
@@ -1238,11 +1242,11 @@ class Parser:
         if self.try_kw("ELSE"):
             default = self._expr()
         self.eat_kw("END")
-        return A.Case(None, whens, default) if test is not None \
-            else A.Case(test, whens, default)
+        return A.Case(test, whens, default)
 
     def _case_alt(self, test):
-        """One alternative of an extended simple-form CASE WHEN."""
+        """One alternative of an extended simple-form CASE WHEN; `test`
+        is the synthetic __case__ variable, never the operand AST."""
         t = self.peek()
         if t.kind == "OP" and t.value in ("=", "<>", "!=", "<", ">",
                                           "<=", ">=", "=~"):

@@ -203,6 +203,15 @@ class TestExpressions:
             "RETURN CASE WHEN x = 1 THEN 'one' WHEN x = 2 THEN 'two' ELSE 'many' END")
         assert [x[0] for x in r.rows] == ["one", "two", "many"]
 
+    def test_case_operand_evaluated_once(self, ex):
+        # a non-deterministic simple-CASE operand must be
+        # evaluated exactly once, so complementary alternatives always
+        # cover it (Neo4j semantics).
+        for _ in range(40):
+            r = ex.execute(
+                "RETURN CASE rand() WHEN < 0.5 THEN 'lo' WHEN >= 0.5 THEN 'hi' END AS x")
+            assert r.rows[0][0] in ("lo", "hi")
+
     def test_case_extended_simple_form(self, ex):
         # Neo4j 5 extended simple CASE: candidate lists and comparisons
         # applied to the operand (reference: pkg/cypher expression tests).
