@@ -433,14 +433,43 @@ class DatabaseManager:
 
 def open_db(data_dir: Optional[str] = None, embedder: Embedder = None,
             dims: int = None, device: str = None, durable_sync: bool = False,
+            engine: str = "disk", encryption_passphrase: str = "",
             **kw) -> DatabaseManager:
     """Open a NornicDB instance (reference nornicdb.Open, db.go:750).
 
-    data_dir=None -> in-memory. Engine stack: Persistent(WAL+snapshots)
-    [-> Async] -> Namespaced per database.
+    data_dir=None -> in-memory. Otherwise `engine` picks the stack:
+      "disk" (default) -> DiskEngine: LSM on-disk store (Badger-parity;
+              datasets may exceed RAM, restart cost O(active log));
+      "wal"  -> PersistentEngine: RAM + WAL + snapshots (round-1 engine,
+              fastest when the working set fits memory).
+    encryption_passphrase != "" seals every block/log/backup on disk
+    (reference pkg/nornicdb/db.go:775-808 Badger at-rest encryption).
     """
     if data_dir:
-        base = PersistentEngine(data_dir, sync_on_write=durable_sync)
+        crypt = None
+        if encryption_passphrase:
+            from .utils.encryption import EncryptionManager
+            import hashlib
+            import os as _os
+            # deterministic per-datadir salt, persisted next to the store
+            _os.makedirs(data_dir, exist_ok=True)
+            salt_p = _os.path.join(data_dir, "SALT")
+            if _os.path.exists(salt_p):
+                salt = open(salt_p, "rb").read()
+            else:
+                salt = _os.urandom(16)
+                with open(salt_p, "wb") as f:
+                    f.write(salt)
+            crypt = EncryptionManager(encryption_passphrase, salt=salt)
+        if engine == "wal":
+            if crypt is not None:
+                raise ValueError(
+                    "encryption_passphrase requires the disk engine")
+            base = PersistentEngine(data_dir, sync_on_write=durable_sync)
+        else:
+            from .storage import DiskEngine
+            base = DiskEngine(data_dir, sync_on_write=durable_sync,
+                              encryption=crypt)
     else:
         base = MemoryEngine()
     return DatabaseManager(base, embedder=embedder, dims=dims, device=device)

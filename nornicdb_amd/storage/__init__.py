@@ -1,7 +1,10 @@
 """Storage engines.
 
 Stack (mirrors reference assembly pkg/nornicdb/db.go:762-915):
-    PersistentEngine (RAM state + WAL + snapshots, replaces Badger+WALEngine)
+    DiskEngine (LSM on-disk store, replaces BadgerEngine — the default
+                persistent engine; datasets may exceed RAM)
+    PersistentEngine (RAM state + WAL + snapshots — the fast in-RAM
+                durable engine, kept for workloads that fit memory)
       -> AsyncEngine (write-behind, optional)
         -> NamespacedEngine (multi-tenant prefixing)
 """
@@ -15,11 +18,13 @@ from .namespaced import NamespacedEngine
 from .wal import WAL, WALCorruption, WALDegraded
 from .schema import Constraint, SchemaManager, VectorIndexMeta
 from .composite import CompositeEngine
+from .disk import DiskEngine, DiskTransaction
+from .lsm import LSMStore
 
 __all__ = [
     "Node", "Edge", "Engine", "EventType", "StorageError", "NotFoundError",
     "ConstraintViolation", "new_id", "MemoryEngine", "PersistentEngine",
     "Transaction", "AsyncEngine", "NamespacedEngine", "WAL", "WALCorruption",
     "SchemaManager", "Constraint", "VectorIndexMeta", "CompositeEngine",
-    "WALDegraded",
+    "WALDegraded", "DiskEngine", "DiskTransaction", "LSMStore",
 ]
