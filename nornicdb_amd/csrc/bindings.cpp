@@ -11,6 +11,10 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
 std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out);
 
+// gemm.hip
+at::Tensor gemm_nt(at::Tensor a, at::Tensor w, c10::optional<at::Tensor> bias,
+                   long long act);
+
 // graph.hip
 at::Tensor pagerank_contrib(at::Tensor rank, at::Tensor outdeg);
 at::Tensor pagerank_gather(at::Tensor row_ptr, at::Tensor col_idx,
@@ -55,6 +59,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused MFMA cosine score + top-k, 256-query batches (bf16 db)",
         py::arg("db"), py::arg("q"), py::arg("row_base") = 0,
         py::arg("k_out") = 10);
+  m.def("gemm_nt", &gemm_nt,
+        "C = A[M,K] @ W[N,K]^T + bias (+GELU), bf16 MFMA 256x256 8-phase",
+        py::arg("a"), py::arg("w"), py::arg("bias") = c10::nullopt,
+        py::arg("act") = 0);
   m.def("add_layernorm", &add_layernorm, "LN(a+b)*gamma+beta fused (bf16)",
         py::arg("a"), py::arg("b"), py::arg("gamma"), py::arg("beta"),
         py::arg("eps") = 1e-5);

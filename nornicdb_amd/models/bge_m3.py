@@ -70,7 +70,12 @@ class _EncoderLayer(nn.Module):
 
     def forward(self, x, attn_bias):
         b, s, h = x.shape
-        qkv = self.qkv(x).view(b, s, 3, self.num_heads, self.head_dim)
+        if _fused_ok(x):
+            from nornicdb_amd.ops.gemm import linear_act
+            qkv = linear_act(x, self.qkv.weight, self.qkv.bias)
+        else:
+            qkv = self.qkv(x)
+        qkv = qkv.view(b, s, 3, self.num_heads, self.head_dim)
         q, k, v = qkv.unbind(dim=2)  # [b, s, nh, hd] strided views
         from nornicdb_amd.ops.attention import (flash_attention_bshd,
                                                 flash_attention_nc)
@@ -84,11 +89,17 @@ class _EncoderLayer(nn.Module):
             a = a.transpose(1, 2).reshape(b, s, h)
         if _fused_ok(x):
             from nornicdb_amd.ops import encoder as eops
+            from nornicdb_amd.ops.gemm import ACT_GELU, linear_act
 
-            x = eops.add_layernorm(x, self.attn_out(a), self.ln1.weight,
-                                   self.ln1.bias, self.ln1.eps)
-            f = self.ffn_out(eops.bias_gelu(
-                F.linear(x, self.ffn_in.weight), self.ffn_in.bias))
+            # all four GEMMs are the hand-written MFMA kernel, with bias
+            # (and the FFN GELU) fused into the epilogue — zero Tensile
+            # kernels on the serve path.
+            x = eops.add_layernorm(
+                x, linear_act(a, self.attn_out.weight, self.attn_out.bias),
+                self.ln1.weight, self.ln1.bias, self.ln1.eps)
+            f = linear_act(
+                linear_act(x, self.ffn_in.weight, self.ffn_in.bias, ACT_GELU),
+                self.ffn_out.weight, self.ffn_out.bias)
             x = eops.add_layernorm(x, f, self.ln2.weight, self.ln2.bias,
                                    self.ln2.eps)
         else:
