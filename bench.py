@@ -111,6 +111,82 @@ def search_step(shard, queries, k, row_base, world):
     return merge_global_topk(s, i, k, world)
 
 
+def graph_bench(args, rank, world, local_rank):
+    """BASELINE config #4: PageRank on a 1B-edge synthetic CSR, sharded
+    across ranks with halo-exchange contrib traffic (parallel/graph.py).
+    Prints one JSON line (its own metric, separate from the kNN headline).
+    """
+    from nornicdb_amd.parallel.graph import pagerank_sharded, shard_rows
+
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    device = torch.device(args.device or ("cuda:%d" % local_rank if use_cuda else "cpu"))
+    n, m = args.graph_nodes, args.graph_edges
+    if device.type == "cpu" and m > 10_000_000:
+        n, m = 100_000, 1_000_000  # CPU dev mode
+    deg = m // n
+    lo, hi = shard_rows(n, rank, world)
+    n_local = hi - lo
+    g = torch.Generator(device=device).manual_seed(4242 + rank)
+    log(rank, f"[graph] building shard: {n_local} rows x deg {deg} "
+              f"({n_local * deg * 4 / 1e9:.1f} GB col_idx) ...")
+    row_ptr = torch.arange(n_local + 1, device=device, dtype=torch.int64) * deg
+    col_idx = torch.randint(0, n, (n_local * deg,), device=device,
+                            generator=g, dtype=torch.int32)
+    # global out-degree histogram (once; all-reduced across shards)
+    outdeg = torch.zeros(n, device=device, dtype=torch.float32)
+    ones = torch.ones(1, device=device).expand(col_idx.numel())
+    outdeg.scatter_add_(0, col_idx.long(), ones)
+    if world > 1:
+        dist.all_reduce(outdeg)
+    log(rank, "[graph] warmup ...")
+    pagerank_sharded(row_ptr, col_idx, outdeg, n, lo, iters=max(args.warmup, 1),
+                     tol=0.0)
+    if world > 1:
+        dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.time()
+    r = pagerank_sharded(row_ptr, col_idx, outdeg, n, lo, iters=args.steps,
+                         tol=0.0)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.time() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    # sanity: ranks sum to ~1
+    total = float(r.sum()) if world == 1 else float(r.sum())
+    iters_per_s = args.steps / elapsed
+    if rank == 0:
+        out = {
+            "metric": "PageRank iterations/s on 1B-edge synthetic CSR (GPU-sharded, halo exchange)",
+            "value": round(iters_per_s, 3),
+            "unit": "iters/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (uniform random CSR, fixed out-degree)",
+            "config": {
+                "model": "apoc.algo.pageRank (BASELINE config #4)",
+                "nodes": n, "edges": m, "parallelism": f"row-sharded CSR x{world}, halo all_to_all",
+                "edges_per_s": round(m * args.steps / elapsed / 1e9, 2),
+                "unit_edges_per_s": "GTEPS",
+                "rank_sum": round(total, 6),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -129,9 +205,16 @@ def main():
                    help="disable embed/search stream overlap (sequential steps)")
     p.add_argument("--recall-queries", type=int, default=8)
     p.add_argument("--device", default=None)
+    p.add_argument("--graph", action="store_true",
+                   help="run the 1B-edge PageRank bench (BASELINE config #4) instead")
+    p.add_argument("--graph-nodes", type=int, default=100_000_000)
+    p.add_argument("--graph-edges", type=int, default=1_000_000_000)
     args = p.parse_args()
 
     rank, world, local_rank = setup_dist(args)
+    if args.graph:
+        graph_bench(args, rank, world, local_rank)
+        return
     if args.gpus > 1 and world == 1:
         print("ERROR: --gpus>1 requires torchrun (WORLD_SIZE env)", file=sys.stderr)
         sys.exit(2)

@@ -15,6 +15,18 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
 at::Tensor gemm_nt(at::Tensor a, at::Tensor w, c10::optional<at::Tensor> bias,
                    long long act);
 
+// kmeans.hip
+std::tuple<at::Tensor, at::Tensor> kmeans_assign(at::Tensor x, at::Tensor cent,
+                                                 at::Tensor cnorm2);
+std::tuple<at::Tensor, at::Tensor> kmeans_accum(at::Tensor x, at::Tensor assign,
+                                                long long k);
+std::tuple<at::Tensor, at::Tensor> kmeans_finalize(at::Tensor sums,
+                                                   at::Tensor counts,
+                                                   at::Tensor old_c);
+void kmeanspp_update(at::Tensor x, at::Tensor c, double cn2, at::Tensor d2);
+void kmeans_point_update(at::Tensor cent, at::Tensor counts, at::Tensor xv,
+                         long long c, long long sign);
+
 // graph.hip
 at::Tensor pagerank_contrib(at::Tensor rank, at::Tensor outdeg);
 at::Tensor pagerank_gather(at::Tensor row_ptr, at::Tensor col_idx,
@@ -79,6 +91,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("n_layers"), py::arg("hidden"), py::arg("n_heads"),
         py::arg("n_kv"), py::arg("hd"), py::arg("inter"),
         py::arg("max_len"), py::arg("rms_eps"), py::arg("pos"));
+  m.def("kmeans_assign", &kmeans_assign,
+        "fused distance+argmin assignment (bf16 x/centroids)");
+  m.def("kmeans_accum", &kmeans_accum, "atomic centroid accumulate");
+  m.def("kmeans_finalize", &kmeans_finalize,
+        "sums/counts -> centroids + drift^2 (empty keep old)");
+  m.def("kmeanspp_update", &kmeanspp_update,
+        "d2 = min(d2, dist2(x, new_seed))");
+  m.def("kmeans_point_update", &kmeans_point_update,
+        "incremental single-point centroid update (+1 add / -1 remove)");
   m.def("pagerank_contrib", &pagerank_contrib, "rank/outdeg elementwise");
   m.def("pagerank_gather", &pagerank_gather,
         "CSR pull-gather pagerank iteration (wave per row)");

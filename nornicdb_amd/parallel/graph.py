@@ -2,11 +2,15 @@
 
 The CSR adjacency is ROW-SHARDED across ranks (rank owns rows
 [base, base+n_local)); column indices stay global. Per iteration:
-  - PageRank: every rank holds the full rank vector; computes new values
-    for its rows from all-gathered contributions; all-gather over xGMI
-    reassembles the vector (bandwidth-bound -> big contiguous segments).
+  - PageRank: the rank vector stays SHARDED; each rank exchanges only
+    the halo contrib values its local CSR references (HaloExchange,
+    one all_to_all of per-owner segments built from a one-time halo
+    index) plus two scalar all-reduces. xGMI is point-to-point, so the
+    all_to_all maps onto direct links rather than a ring of the full
+    vector.
   - BFS / WCC: all-reduce(MIN) of the global dist/component vector per
-    level (xGMI ring all-reduce).
+    level (xGMI ring all-reduce; frontier vectors are int32, one per
+    level, not per edge).
 Works with the gloo backend on CPU for tests (world_size > 1 in-process /
 multi-process), with the HIP kernels on MI355X under nccl(=RCCL).
 
@@ -27,6 +31,28 @@ from ..ops import native_or_none
 from ..graph.csr import CSRGraph
 
 
+def _all_to_all(recv: list, send: list, rank: int, world: int):
+    """dist.all_to_all with a gloo fallback (gloo lacks alltoall): post
+    all irecvs, then isends, wait — self-slot copied locally."""
+    if dist.get_backend() == "nccl":
+        dist.all_to_all(recv, send)
+        return
+    works = []
+    for p in range(world):
+        if p == rank:
+            continue
+        if recv[p].numel():
+            works.append(dist.irecv(recv[p], src=p))
+    for p in range(world):
+        if p == rank:
+            continue
+        if send[p].numel():
+            works.append(dist.isend(send[p].contiguous(), dst=p))
+    recv[rank].copy_(send[rank])
+    for w in works:
+        w.wait()
+
+
 def shard_rows(n: int, rank: int, world: int) -> Tuple[int, int]:
     per = (n + world - 1) // world
     lo = min(rank * per, n)
@@ -34,14 +60,56 @@ def shard_rows(n: int, rank: int, world: int) -> Tuple[int, int]:
     return lo, hi
 
 
-def _all_gather_rank_vector(local: torch.Tensor, n: int, world: int):
-    """Gather per-rank row slices into the full [n] vector."""
-    per = (n + world - 1) // world
-    padded = torch.zeros(per, dtype=local.dtype, device=local.device)
-    padded[: local.numel()] = local
-    out = [torch.empty_like(padded) for _ in range(world)]
-    dist.all_gather(out, padded)
-    return torch.cat(out)[:n]
+class HaloExchange:
+    """Per-shard slice exchange for row-sharded CSR iteration.
+
+    Built ONCE from the local column set: each rank needs contrib values
+    only for the (unique, sorted) global columns its local CSR touches.
+    Because row ownership is by contiguous ranges, the sorted unique
+    column array splits into per-owner contiguous segments, so each
+    iteration is ONE all_to_all of exactly those values — no rank ever
+    ships the full vector (the previous all-gather
+    moved the whole n-vector per iteration; at 100M+ nodes that is GBs
+    over xGMI each round).
+    """
+
+    def __init__(self, col_idx_local: torch.Tensor, n: int, world: int,
+                 rank: int, device):
+        self.world = world
+        self.rank = rank
+        per = (n + world - 1) // world
+        uniq = torch.unique(col_idx_local.long())  # sorted
+        owner_bounds = torch.arange(1, world + 1, device=uniq.device) * per
+        # segment p = needed cols owned by rank p
+        seg_ends = torch.searchsorted(uniq, owner_bounds.to(uniq.dtype))
+        seg_starts = torch.cat([torch.zeros(1, dtype=seg_ends.dtype,
+                                            device=seg_ends.device),
+                                seg_ends[:-1]])
+        self.uniq = uniq
+        self.recv_counts = (seg_ends - seg_starts).tolist()
+        # tell every owner which of ITS local indices we need
+        req = [
+            (uniq[int(seg_starts[p]):int(seg_ends[p])] - p * per).to(device)
+            for p in range(world)
+        ]
+        # exchange request sizes, then the request index lists
+        send_sizes = torch.tensor(self.recv_counts, dtype=torch.long)
+        all_sizes = [torch.zeros(world, dtype=torch.long) for _ in range(world)]
+        dist.all_gather(all_sizes, send_sizes)
+        self.send_counts = [int(all_sizes[p][rank]) for p in range(world)]
+        send_idx = [torch.zeros(c, dtype=torch.long, device=device)
+                    for c in self.send_counts]
+        _all_to_all(send_idx, req, rank, world)
+        self.send_idx = send_idx      # per peer: OUR local indices to send
+        self.device = device
+
+    def exchange(self, local_vals: torch.Tensor) -> torch.Tensor:
+        """Returns the halo value buffer aligned with self.uniq."""
+        send = [local_vals[idx] for idx in self.send_idx]
+        recv = [torch.empty(c, dtype=local_vals.dtype, device=self.device)
+                for c in self.recv_counts]
+        _all_to_all(recv, send, self.rank, self.world)
+        return torch.cat(recv) if recv else local_vals.new_empty(0)
 
 
 def pagerank_sharded(row_ptr_local: torch.Tensor, col_idx_local: torch.Tensor,
@@ -49,47 +117,70 @@ def pagerank_sharded(row_ptr_local: torch.Tensor, col_idx_local: torch.Tensor,
                      damping: float = 0.85, iters: int = 20,
                      tol: float = 1e-6) -> torch.Tensor:
     """Distributed PageRank. Inputs are the LOCAL in-edge CSR rows; returns
-    the full rank vector (identical on every rank).
+    the full rank vector (identical on every rank; gathered once at END).
 
-    On GPU the per-iteration gather uses the HIP wave-per-row kernel; on
-    CPU (gloo tests) a torch index_add implements the same contraction.
+    The rank vector lives SHARDED: each iteration computes the local row
+    slice with the HIP wave-per-row kernel (CPU: torch index_add) and
+    exchanges only halo contrib values (HaloExchange) plus two scalar
+    all-reduces (dangling mass, convergence delta).
     """
     world = dist.get_world_size() if dist.is_initialized() else 1
+    rank = dist.get_rank() if dist.is_initialized() else 0
     device = row_ptr_local.device
     nat = native_or_none() if device.type == "cuda" else None
     n_local = row_ptr_local.numel() - 1
+    per = (n + world - 1) // world
 
-    rank_vec = torch.full((n,), 1.0 / n, device=device)
-    dangling_mask = outdeg_global == 0
+    outdeg_local = outdeg_global[row_base:row_base + n_local]
+    rank_local = torch.full((n_local,), 1.0 / n, device=device)
+    dangling_local = outdeg_local == 0
+
+    halo = None
+    col_halo = col_idx_local
+    if world > 1:
+        halo = HaloExchange(col_idx_local, n, world, rank, device)
+        col_halo = torch.searchsorted(
+            halo.uniq, col_idx_local.long()).to(col_idx_local.dtype)
 
     if nat is None:
-        # CPU contraction precompute: local edge -> destination row map
         counts = (row_ptr_local[1:] - row_ptr_local[:-1])
         rows_local = torch.repeat_interleave(
             torch.arange(n_local, device=device), counts)
 
     for _ in range(iters):
-        contrib = torch.where(outdeg_global > 0,
-                              rank_vec / outdeg_global.clamp_min(1),
-                              torch.zeros((), device=device))
-        dangling = float(rank_vec[dangling_mask].sum()) / n
+        contrib_local = torch.where(
+            outdeg_local > 0, rank_local / outdeg_local.clamp_min(1),
+            torch.zeros((), device=device)).float()
+        d = rank_local[dangling_local].sum().reshape(1)
+        if world > 1:
+            vals = halo.exchange(contrib_local)
+            dist.all_reduce(d)
+        else:
+            vals = contrib_local
+        dangling = float(d) / n
         base = (1 - damping) / n + damping * dangling
         if nat is not None:
-            new_local = nat.pagerank_gather(row_ptr_local, col_idx_local,
-                                            contrib.float(), damping, base)
+            new_local = nat.pagerank_gather(row_ptr_local, col_halo,
+                                            vals, damping, base)
         else:
             new_local = torch.full((n_local,), base, device=device)
             new_local.index_add_(0, rows_local,
-                                 damping * contrib[col_idx_local.long()])
+                                 damping * vals[col_halo.long()])
+        delta = (new_local - rank_local).abs().sum()
         if world > 1:
-            new = _all_gather_rank_vector(new_local, n, world)
-        else:
-            new = new_local
-        delta = float((new - rank_vec).abs().sum())
-        rank_vec = new
-        if delta < tol:
+            dist.all_reduce(delta)
+        rank_local = new_local
+        if float(delta) < tol:
             break
-    return rank_vec
+
+    if world == 1:
+        return rank_local
+    # reassemble the full vector once at the end
+    padded = torch.zeros(per, dtype=rank_local.dtype, device=device)
+    padded[:n_local] = rank_local
+    out = [torch.empty_like(padded) for _ in range(world)]
+    dist.all_gather(out, padded)
+    return torch.cat(out)[:n]
 
 
 def bfs_sharded(row_ptr_local: torch.Tensor, col_idx_local: torch.Tensor,

@@ -2,10 +2,12 @@
 
 Parity: reference pkg/gpu/kmeans.go ClusterIndex (:144) — k-means++ init
 (:364), GPU assignment (:491), incremental updates (:910-1009),
-optimal k = sqrt(n/2) (:323). Here assignment is a single GEMM + argmax
-(hipBLASLt on MI355X) and the update is an index_add scatter — the
-distance-matrix / assign / accumulate kernel suite of the reference's
-Metal/CUDA backends collapses into two tensor ops on the CDNA4 path.
+optimal k = sqrt(n/2) (:323). On MI355X the inner ops are the
+hand-written HIP suite in csrc/kmeans.hip (fused distance+argmin assign,
+atomic accumulate + finalize-with-drift, k-means++ min-distance update,
+single-point incremental reassign — the CDNA4 replacement for the
+reference's 9-kernel Metal suite, kmeans_kernels_darwin.metal:71-370);
+the torch expressions below remain the CPU path and numerics oracle.
 """
 
 from __future__ import annotations
@@ -16,6 +18,16 @@ from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch
+
+from ..ops import native_or_none
+
+
+def _native_for(x: torch.Tensor):
+    """HIP kernel suite usable? (CUDA + d % 512 == 0, d <= 4096)."""
+    nat = native_or_none()
+    if nat is not None and x.is_cuda and x.shape[-1] % 512 == 0             and x.shape[-1] <= 4096:
+        return nat
+    return None
 
 
 def optimal_k(n: int) -> int:
@@ -35,6 +47,23 @@ def kmeans(x: torch.Tensor, k: int, iters: int = 25, tol: float = 1e-4,
     # k-means++ init (on a sample for large n)
     sample = xf if n <= 100_000 else xf[torch.randperm(n, generator=g)[:100_000].to(x.device)]
     c = _kmeanspp(sample, k, g)
+
+    nat = _native_for(x)
+    if nat is not None:
+        xb = x if x.dtype == torch.bfloat16 else xf.to(torch.bfloat16)
+        prev_inertia = None
+        assign = None
+        for it in range(iters):
+            cb = c.to(torch.bfloat16)
+            a32, d2 = nat.kmeans_assign(xb, cb, (c * c).sum(-1))
+            inertia = float(d2.sum())
+            sums, counts = nat.kmeans_accum(xb, a32, k)
+            c, _drift2 = nat.kmeans_finalize(sums, counts, c)
+            assign = a32.long()
+            if prev_inertia is not None and                     abs(prev_inertia - inertia) <= tol * max(prev_inertia, 1e-12):
+                break
+            prev_inertia = inertia
+        return c, assign
 
     x_sq = (xf * xf).sum(-1, keepdim=True)  # [n,1]
     prev_inertia = None
@@ -73,6 +102,24 @@ def _kmeanspp(x: torch.Tensor, k: int, g) -> torch.Tensor:
     host sync (the reference's init is the same algorithm on CPU,
     kmeans.go:364)."""
     n = x.shape[0]
+    nat = _native_for(x)
+    if nat is not None:
+        xb = x.to(torch.bfloat16)
+        first = int(torch.randint(n, (1,), generator=g))
+        idxs = [torch.as_tensor([first], device=x.device)]
+        d2 = torch.full((n,), 3.4e38, device=x.device)
+        cvec = xb[first].contiguous()
+        nat.kmeanspp_update(xb, cvec, float((x[first] ** 2).sum()), d2)
+        for _ in range(1, k):
+            w = d2.clamp_min(1e-12)
+            cdf = torch.cumsum(w, 0)
+            r = torch.rand(1, device=x.device) * cdf[-1]
+            idx = torch.searchsorted(cdf, r).clamp_max_(n - 1)
+            idxs.append(idx)
+            cvec = xb[idx[0]].contiguous()
+            nat.kmeanspp_update(xb, cvec, float((x[idx[0]] ** 2).sum()), d2)
+        return x[torch.cat(idxs)].clone().float()
+
     x_sq = (x * x).sum(-1)
     first = int(torch.randint(n, (1,), generator=g))
     idxs = [torch.as_tensor([first], device=x.device)]
@@ -101,6 +148,7 @@ class ClusterIndex:
         self.centroids: Optional[torch.Tensor] = None
         self.members: List[List[str]] = []
         self._id2cluster: Dict[str, int] = {}
+        self.counts: Optional[torch.Tensor] = None
         self.nprobe = nprobe
         self._drift = 0
         self._drift_limit_ratio = reassign_drift
@@ -124,29 +172,60 @@ class ClusterIndex:
             for i, id_ in enumerate(ids):
                 self.members[a[i]].append(id_)
                 self._id2cluster[id_] = a[i]
+            self.counts = torch.bincount(assign, minlength=c.shape[0]) \
+                .to(dtype=torch.int32, device=c.device)
             self._drift = 0
 
     def add(self, id_: str, vec) -> None:
-        """Incremental assignment to nearest centroid."""
+        """Incremental assignment to nearest centroid, updating the
+        centroid itself (reference kmeans.go:910-1009 incremental path;
+        HIP single-point kernel on MI355X)."""
         with self._lock:
             if self.centroids is None:
                 return
             v = torch.as_tensor(np.asarray(vec, dtype=np.float32),
                                 device=self.centroids.device)
-            d2 = ((self.centroids - v) ** 2).sum(-1)
-            cl = int(d2.argmin())
+            nat = _native_for(self.centroids)
+            if nat is not None:
+                vb = v.to(torch.bfloat16).contiguous()
+                a32, _ = nat.kmeans_assign(
+                    vb[None, :], self.centroids.to(torch.bfloat16),
+                    (self.centroids * self.centroids).sum(-1))
+                cl = int(a32[0])
+            else:
+                d2 = ((self.centroids - v) ** 2).sum(-1)
+                cl = int(d2.argmin())
             old = self._id2cluster.get(id_)
             if old is not None and old != cl:
                 try:
                     self.members[old].remove(id_)
                 except ValueError:
                     pass
+                self._point_update(old, v, -1)
             if old != cl:
                 self.members[cl].append(id_)
                 self._id2cluster[id_] = cl
+                self._point_update(cl, v, +1)
             self._drift += 1
 
-    def remove(self, id_: str) -> None:
+    def _point_update(self, cl: int, v: torch.Tensor, sign: int) -> None:
+        """c = (c*cnt +/- v) / (cnt +/- 1) — incremental centroid move."""
+        if getattr(self, "counts", None) is None:
+            return
+        nat = _native_for(self.centroids)
+        if nat is not None:
+            nat.kmeans_point_update(self.centroids, self.counts,
+                                    v.to(torch.bfloat16).contiguous(), cl, sign)
+            return
+        cnt = int(self.counts[cl])
+        new = cnt + sign
+        if new <= 0:
+            self.counts[cl] = 0
+            return
+        self.centroids[cl] = (self.centroids[cl] * cnt + sign * v) / new
+        self.counts[cl] = new
+
+    def remove(self, id_: str, vec=None) -> None:
         with self._lock:
             cl = self._id2cluster.pop(id_, None)
             if cl is not None:
@@ -154,6 +233,10 @@ class ClusterIndex:
                     self.members[cl].remove(id_)
                 except ValueError:
                     pass
+                if vec is not None:
+                    v = torch.as_tensor(np.asarray(vec, dtype=np.float32),
+                                        device=self.centroids.device)
+                    self._point_update(cl, v, -1)
 
     def needs_recluster(self) -> bool:
         with self._lock:

@@ -260,3 +260,100 @@ class TestHTTPEmbedderOffline:
         out = e.embed_batch(["a", "b"])
         assert out.shape == (2, 8)
         assert calls
+
+
+# ---------------------------------------------------------------------------
+# HIP k-means kernel suite (csrc/kmeans.hip) vs torch oracle
+# ---------------------------------------------------------------------------
+
+def test_cluster_index_incremental_updates_cpu():
+    import torch
+    from nornicdb_amd.search.kmeans import ClusterIndex
+    torch.manual_seed(0)
+    ci = ClusterIndex()
+    mat = torch.cat([torch.randn(20, 8) + 4, torch.randn(20, 8) - 4])
+    ids = [f"v{i}" for i in range(40)]
+    ci.cluster(ids, mat, k=2, seed=1)
+    c_before = ci.centroids.clone()
+    cnt_before = ci.counts.clone()
+    # adding a point near cluster of v0 must move that centroid toward it
+    v = mat[0] + 0.5
+    ci.add("new", v)
+    cl = ci._id2cluster["new"]
+    assert int(ci.counts[cl]) == int(cnt_before[cl]) + 1
+    moved = (ci.centroids[cl] - c_before[cl]).abs().sum()
+    assert float(moved) > 0
+    # removing it restores the centroid (same arithmetic inverted)
+    ci.remove("new", v)
+    assert torch.allclose(ci.centroids[cl], c_before[cl], atol=1e-4)
+    assert int(ci.counts[cl]) == int(cnt_before[cl])
+
+
+@pytest.mark.gpu
+def test_kmeans_kernels_match_torch_gpu():
+    import torch
+    from nornicdb_amd.ops import require_native
+    nat = require_native()
+    torch.manual_seed(3)
+    n, d, k = 5000, 512, 32
+    x = torch.randn(n, d, device="cuda")
+    xb = x.to(torch.bfloat16)
+    cent = torch.randn(k, d, device="cuda")
+    cb = cent.to(torch.bfloat16)
+    a, d2 = nat.kmeans_assign(xb, cb, (cent * cent).sum(-1))
+    # oracle in the SAME bf16 precision
+    xf, cf = xb.float(), cb.float()
+    ref_d2 = ((xf * xf).sum(-1, keepdim=True) + (cf * cf).sum(-1)[None]
+              - 2 * xf @ cf.T)
+    ref_m, ref_a = ref_d2.min(dim=1)
+    agree = (a.long() == ref_a).float().mean().item()
+    assert agree > 0.99, agree  # ties may flip at bf16
+    match = a.long() == ref_a
+    assert (d2[match] - ref_m[match].clamp_min(0)).abs().max().item() < 0.5
+    # accumulate + finalize
+    sums, counts = nat.kmeans_accum(xb, a, k)
+    ref_counts = torch.bincount(a.long(), minlength=k).int()
+    assert (counts == ref_counts).all()
+    ref_sums = torch.zeros(k, d, device="cuda")
+    ref_sums.index_add_(0, a.long(), xf)
+    assert (sums - ref_sums).abs().max().item() < 0.05
+    newc, drift2 = nat.kmeans_finalize(sums, counts, cent)
+    ref_new = ref_sums / ref_counts.clamp_min(1)[:, None].float()
+    ref_new[ref_counts == 0] = cent[ref_counts == 0]
+    assert (newc - ref_new).abs().max().item() < 0.05
+    ref_drift = ((ref_new - cent) ** 2).sum(-1)
+    assert (drift2 - ref_drift).abs().max().item() < 0.5
+    # k-means++ min-distance update
+    d2b = torch.full((n,), 3.4e38, device="cuda")
+    nat.kmeanspp_update(xb, cb[0].contiguous(), float((cf[0] ** 2).sum()), d2b)
+    ref_pp = ((xf - cf[0]) ** 2).sum(-1)
+    assert (d2b - ref_pp).abs().max().item() < 0.5
+    # single-point incremental update
+    cents = cent.clone()
+    cnts = ref_counts.clone()
+    v = xb[0].contiguous()
+    nat.kmeans_point_update(cents, cnts, v, 3, 1)
+    cnt3 = int(ref_counts[3])
+    ref_upd = (cent[3] * cnt3 + xf[0]) / (cnt3 + 1)
+    assert (cents[3] - ref_upd).abs().max().item() < 1e-2
+    assert int(cnts[3]) == cnt3 + 1
+
+
+@pytest.mark.gpu
+def test_kmeans_gpu_end_to_end_quality():
+    """Full kmeans() on the HIP path converges to the same inertia
+    ballpark as the torch oracle path."""
+    import torch
+    from nornicdb_amd.search.kmeans import kmeans
+    torch.manual_seed(5)
+    centers = torch.randn(8, 512, device="cuda") * 5
+    x = (centers.repeat_interleave(500, 0)
+         + 0.3 * torch.randn(4000, 512, device="cuda"))
+    c, a = kmeans(x, 8, iters=20, seed=2)
+    assert c.shape == (8, 512)
+    # every true center has one recovered centroid nearby
+    dists = ((centers[:, None, :] - c[None]) ** 2).sum(-1).min(dim=1).values
+    assert float(dists.max()) < 30.0, float(dists.max())
+    # assignment is consistent with nearest centroid
+    d2 = ((x[:100, None, :] - c[None]) ** 2).sum(-1)
+    assert (d2.argmin(1) == a[:100]).float().mean().item() > 0.97
