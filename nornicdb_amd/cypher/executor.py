@@ -130,6 +130,15 @@ class Executor:
         res.stats = dict(self.stats)
         return res
 
+    @property
+    def _colstore(self):
+        cs = getattr(self, "_colstore_inst", None)
+        if cs is None:
+            from .columnar import ColumnStore
+            cs = ColumnStore(self.engine)
+            self._colstore_inst = cs
+        return cs
+
     def _try_fast_count(self, q: A.Query, params):
         """Fast path for `MATCH (n:L [{props}]) [WHERE ...] RETURN count(..)`
         over the raw label index — no node copies, no row pipeline
@@ -197,6 +206,15 @@ class Executor:
                         hits = None
                     if hits is not None:
                         return Result([name], [[len(hits)]])
+
+        # vectorized count over cached numpy columns (columnar.py)
+        if (m.where is not None and label and not extra_labels and not props
+                and np_.var):
+            from .columnar import columnar_filter
+            got = columnar_filter(self._colstore, label, np_.var, m.where,
+                                  params)
+            if got is not None:
+                return Result([name], [[int(got[1].sum())]])
 
         npred = None
         if m.where is not None:
@@ -1075,7 +1093,41 @@ class Executor:
                         q.append((other, eacc + [e], nacc + [onode]))
 
     # -------------------------------------------------------------- clauses
+    def _columnar_match(self, c: A.MatchClause, rows, params):
+        """Vectorized WHERE over a single-label scan: evaluates the
+        predicate as numpy masks on cached columns and copies only the
+        surviving nodes (columnar.py). Returns rows or None."""
+        if (c.optional or c.where is None or len(c.patterns) != 1
+                or len(rows) != 1 or rows[0]):
+            return None
+        pat = c.patterns[0]
+        if pat.var or len(pat.elements) != 1:
+            return None
+        el = pat.elements[0]
+        if (not getattr(el, "var", None) or not el.labels
+                or getattr(el, "or_labels", False)
+                or getattr(el, "where", None) or el.props is not None):
+            return None
+        from .columnar import columnar_filter
+        got = columnar_filter(self._colstore, el.labels[0], el.var, c.where,
+                              params)
+        if got is None:
+            return None
+        nodes, mask = got
+        extra = el.labels[1:]
+        out = []
+        import numpy as _np
+        for i in _np.nonzero(mask)[0]:
+            n = nodes[int(i)]
+            if extra and not all(lb in n.labels for lb in extra):
+                continue
+            out.append({el.var: n.copy()})
+        return out
+
     def _exec_match(self, c: A.MatchClause, rows, params):
+        fast = self._columnar_match(c, rows, params)
+        if fast is not None:
+            return fast
         out = []
         for row in rows:
             matched = [row]
