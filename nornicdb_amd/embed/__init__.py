@@ -64,7 +64,7 @@ class BgeM3Embedder(Embedder):
                  layers: int = 24):
         import torch
         from ..models import BgeM3Config, BgeM3Encoder
-        from .tokenizer import HashTokenizer
+        from .tokenizer import default_tokenizer
 
         self.dims = dims
         self.max_tokens = max_tokens
@@ -75,7 +75,7 @@ class BgeM3Embedder(Embedder):
         if weights_path:
             sd = torch.load(weights_path, map_location=self.device)
             self.model.load_state_dict(sd)
-        self.tokenizer = HashTokenizer(cfg.vocab_size, max_tokens)
+        self.tokenizer = default_tokenizer(cfg.vocab_size, max_tokens)
         self._torch = torch
 
     def embed_batch(self, texts: Sequence[str]) -> np.ndarray:

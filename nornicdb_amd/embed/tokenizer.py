@@ -70,3 +70,21 @@ class HFTokenizer:
             ids.append(e + [PAD] * pad)
             mask.append([1] * len(e) + [0] * pad)
         return ids, mask
+
+
+DEFAULT_VOCAB = __file__.rsplit("/", 1)[0] + "/vocab/nornic_bpe.json"
+
+
+def default_tokenizer(vocab_size: int = 250002, max_tokens: int = 512):
+    """The embed queue's tokenizer: NORNICDB_TOKENIZER env path (e.g. a
+    real bge-m3 tokenizer.json) > the shipped trained BPE artifact
+    (scripts/train_tokenizer.py; XLM-R special-token layout, ids fit the
+    bge-m3 embedding table) > hash fallback (no artifact present)."""
+    import os
+    path = os.environ.get("NORNICDB_TOKENIZER") or DEFAULT_VOCAB
+    if os.path.exists(path):
+        try:
+            return HFTokenizer(path, max_tokens)
+        except Exception:
+            pass
+    return HashTokenizer(vocab_size, max_tokens)

@@ -184,3 +184,24 @@ class TestQueryCache:
         db.cypher("CREATE (:W1)")
         db.cypher("CREATE (:W1)")
         assert db.cypher("MATCH (n:W1) RETURN count(n)").rows == [[2]]
+
+
+def test_default_tokenizer_is_trained_bpe():
+    """HashTokenizer must no longer be the default —
+    the shipped BPE artifact drives the text->tokens->vector path."""
+    from nornicdb_amd.embed.tokenizer import (HFTokenizer, default_tokenizer,
+                                              BOS, EOS, PAD)
+    tok = default_tokenizer()
+    assert isinstance(tok, HFTokenizer), type(tok)
+    ids = tok.encode("NornicDB stores graph memories with vector search.")
+    assert ids[0] == BOS and ids[-1] == EOS
+    assert all(0 <= i < 250002 for i in ids)
+    # subword merges learned: common English words are single tokens
+    short = tok.encode("the")
+    assert len(short) <= 4
+    batch_ids, mask = tok.encode_batch(["a tiny text", "a much longer text "
+                                        "with several additional words"])
+    assert len(batch_ids[0]) == len(batch_ids[1])
+    assert mask[0][-1] == 0 and batch_ids[0][-1] == PAD
+    # deterministic
+    assert tok.encode("same input") == tok.encode("same input")
