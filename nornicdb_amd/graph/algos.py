@@ -504,7 +504,9 @@ def _louvain_local_moving(src, dst, w, k, n, two_m, max_passes, resolution,
             comm[move_u] = move_c
             next_active[move_u] = True
             total_moved += int(move_u.numel())
-        if total_moved == 0 or total_moved < max(32, n // 200):
+        # fractional convergence threshold only at scale — small graphs
+        # legitimately move < 32 nodes per pass and must keep iterating
+        if total_moved == 0 or (n > 10000 and total_moved < n // 200):
             break
         touched = next_active[td]
         next_active[ts[touched]] = True
