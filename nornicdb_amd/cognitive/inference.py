@@ -14,8 +14,13 @@ from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
 from ..storage.types import Edge, Engine, Node, NotFoundError, new_id
+from .evidence import (CooldownTable, EvidenceBuffer, EvidenceThreshold)
 from .linkpredict import adamic_adar
 from .temporal import AccessTracker
+
+# suggestion reason -> evidence label (reference signal taxonomy)
+_REASON_LABEL = {"similarity": "similar_to", "co-access": "coaccess",
+                 "temporal": "relates_to", "transitive": "topology"}
 
 
 @dataclass
@@ -40,16 +45,30 @@ class InferenceEngine:
         self.tracker = tracker or AccessTracker(now_fn=now_fn)
         self.cfg = config or InferenceConfig()
         self.now = now_fn
-        self._cooldown: Dict[Tuple[str, str], float] = {}
-        self._evidence: Dict[Tuple[str, str], int] = defaultdict(int)
+        # evidence buffering + per-label cooldown (reference evidence.go /
+        # cooldown.go). evidence_required <= 1 collapses every threshold
+        # to a single signal (the facade's immediate-link mode).
+        if self.cfg.evidence_required <= 1:
+            thr = EvidenceThreshold(1, self.cfg.min_confidence, 0, 86400.0)
+            self.evidence = EvidenceBuffer(
+                {lb: thr for lb in _REASON_LABEL.values()}, now_fn=now_fn)
+        else:
+            base = EvidenceThreshold(self.cfg.evidence_required,
+                                     self.cfg.min_confidence, 1, 86400.0)
+            self.evidence = EvidenceBuffer(
+                {lb: base for lb in _REASON_LABEL.values()}, now_fn=now_fn)
+        self.cooldowns = CooldownTable(
+            {lb: self.cfg.cooldown_s for lb in _REASON_LABEL.values()},
+            now_fn=now_fn)
 
     # ---- main hook ----
-    def on_store(self, node: Node) -> List[Edge]:
+    def on_store(self, node: Node, session_id: str = "") -> List[Edge]:
         """Run inference for a freshly stored/embedded node; returns created edges."""
         suggestions = self.suggest(node)
         created = []
         for target, confidence, reason in suggestions:
-            e = self._maybe_link(node.id, target, confidence, reason)
+            e = self._maybe_link(node.id, target, confidence, reason,
+                                 session_id)
             if e is not None:
                 created.append(e)
         return created
@@ -107,20 +126,24 @@ class InferenceEngine:
         return scores[:5]
 
     def _maybe_link(self, a: str, b: str, confidence: float,
-                    reason: str) -> Optional[Edge]:
+                    reason: str, session_id: str = "") -> Optional[Edge]:
         if confidence < self.cfg.min_confidence:
             return None
-        key = tuple(sorted((a, b)))
-        now = self.now()
-        if now - self._cooldown.get(key, 0) < self.cfg.cooldown_s:
+        s, d = sorted((a, b))
+        label = _REASON_LABEL.get(reason, "relates_to")
+        # cooldown first: a recently-materialized pair accumulates no new
+        # evidence churn (reference cooldown.go CanMaterialize)
+        if not self.cooldowns.can_materialize(s, d, label):
             return None
-        self._evidence[key] += 1
-        if self._evidence[key] < self.cfg.evidence_required:
+        if not self.evidence.add_evidence(s, d, label, confidence, reason,
+                                          session_id):
             return None
         # already linked?
         if b in self.engine.neighbors(a):
+            self.cooldowns.record_materialization(s, d, label)
             return None
-        self._cooldown[key] = now
+        self.cooldowns.record_materialization(s, d, label)
+        now = self.now()
         e = Edge(id=new_id("inf"), type=self.cfg.edge_type, start_node=a,
                  end_node=b,
                  properties={"confidence": confidence, "inferred": True,

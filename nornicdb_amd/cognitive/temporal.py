@@ -78,6 +78,23 @@ class AccessTracker:
         with self._lock:
             return list(self._sessions)
 
+    @property
+    def session_id(self) -> str:
+        """Stable identifier of the CURRENT session (start timestamp);
+        feeds evidence buffering's unique-session counting."""
+        with self._lock:
+            if not self._sessions:
+                return "s0"
+            return f"s{int(self._sessions[-1][0])}"
+
+    def is_session_boundary(self, ts: float = None) -> bool:
+        """True when the next access at `ts` would start a new session
+        (reference tracker.go IsSessionBoundary: inactivity gap)."""
+        ts = ts if ts is not None else self.now()
+        with self._lock:
+            return (self._last_access_time is None
+                    or ts - self._last_access_time > SESSION_GAP)
+
     def detect_period(self, node_id: str) -> Optional[float]:
         """Dominant access period via mean/variance test on intervals."""
         with self._lock:

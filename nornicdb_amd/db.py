@@ -157,6 +157,18 @@ class NornicDB:
         self.embed_queue = EmbedQueue(self, workers=max(embed_workers, 1))
         if embed_workers > 0:
             self.embed_queue.start()
+        # cognitive pipeline (reference db.go:957-973 + inference.go OnStore):
+        # access tracking feeds co-access/temporal link sources, session
+        # boundaries scope evidence counting, query load adapts decay.
+        from .cognitive import AccessTracker, InferenceConfig, InferenceEngine
+        from .cognitive.patterns import PatternDetector, QueryLoadPredictor
+        self.tracker = AccessTracker()
+        self.patterns = PatternDetector()
+        self.query_load = QueryLoadPredictor()
+        self.auto_link = False  # NORNICDB_AUTO_TLP-equivalent toggle
+        self.inference = InferenceEngine(
+            engine, search_service=self.search, tracker=self.tracker,
+            config=InferenceConfig(evidence_required=2))
 
     # ---- cypher ----
     # queries containing these are never served from the result cache;
@@ -215,12 +227,20 @@ class NornicDB:
         self.engine.create_node(node)
         if embed if embed is not None else self.auto_embed:
             self.engine.mark_pending_embedding(mid)
+        self.tracker.record(mid)
+        self.patterns.record_access(mid)
+        if self.auto_link:
+            try:
+                self.inference.on_store(node, self.tracker.session_id)
+            except Exception:
+                pass
         return Memory(id=mid, content=content, title=title,
                       memory_type=memory_type, importance=importance,
                       tags=list(tags), metadata=metadata or {}, created_at=now)
 
     def recall(self, query: str, limit: int = 10,
                memory_type: str = None) -> List[Memory]:
+        self.query_load.record_query()
         qv = self.embedder.embed_query(query)
         labels = [memory_type.capitalize()] if memory_type else ["Memory"]
         res = self.search.search(query=query, query_vec=qv, k=limit, labels=labels)
@@ -228,6 +248,8 @@ class NornicDB:
         for r in res:
             m = self._to_memory(r.node)
             self._touch(r.node.id)
+            self.tracker.record(r.node.id)
+            self.patterns.record_access(r.node.id)
             out.append(m)
         return out
 

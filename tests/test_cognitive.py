@@ -311,3 +311,223 @@ class TestRelationshipEvolution:
         assert ev.evolution_class("e1") == "fading"
         assert ev.persist() == 1
         assert "_strength" in eng.get_edge("e1").properties
+
+
+# ---------------------------------------------------------------------------
+# reference-behavior tests for the deepened cognitive layer
+# ---------------------------------------------------------------------------
+
+class FakeClock:
+    def __init__(self, t=1_700_000_000.0):
+        self.t = t
+
+    def __call__(self):
+        return self.t
+
+    def advance(self, dt):
+        self.t += dt
+
+
+class TestEvidenceBuffer:
+    def test_three_signal_progression(self):
+        """The reference's doc example: two signals accumulate, the third
+        (from a distinct session) crosses the relates_to threshold."""
+        from nornicdb_amd.cognitive.evidence import EvidenceBuffer
+        clock = FakeClock()
+        eb = EvidenceBuffer(now_fn=clock)
+        assert eb.add_evidence("A", "B", "relates_to", 0.8, "coaccess", "s1") is False
+        ok, reason = eb.check_threshold("A", "B", "relates_to")
+        assert not ok and "more signal" in reason
+        assert eb.add_evidence("A", "B", "relates_to", 0.7, "coaccess", "s2") is False
+        assert eb.add_evidence("A", "B", "relates_to", 0.9, "similarity", "s3") is True
+        # materialized entries leave the buffer
+        assert eb.get_evidence("A", "B", "relates_to") is None
+        assert eb.stats()["materialized"] == 1
+
+    def test_session_requirement_blocks_single_session(self):
+        from nornicdb_amd.cognitive.evidence import EvidenceBuffer
+        eb = EvidenceBuffer(now_fn=FakeClock())
+        for _ in range(5):
+            assert eb.add_evidence("A", "B", "relates_to", 0.9, "x", "same") is False
+        ok, reason = eb.check_threshold("A", "B", "relates_to")
+        assert not ok and "session" in reason
+
+    def test_low_scores_block(self):
+        from nornicdb_amd.cognitive.evidence import EvidenceBuffer
+        eb = EvidenceBuffer(now_fn=FakeClock())
+        for i in range(4):
+            assert eb.add_evidence("A", "B", "relates_to", 0.1, "x", f"s{i}") is False
+        ok, reason = eb.check_threshold("A", "B", "relates_to")
+        assert not ok and "score" in reason
+
+    def test_evidence_expiry(self):
+        from nornicdb_amd.cognitive.evidence import EvidenceBuffer
+        clock = FakeClock()
+        eb = EvidenceBuffer(now_fn=clock)
+        eb.add_evidence("A", "B", "relates_to", 0.9, "x", "s1")
+        eb.add_evidence("A", "B", "relates_to", 0.9, "x", "s2")
+        clock.advance(25 * 3600)  # past relates_to MaxAge (24h)
+        # stale evidence resets; this is signal 1 of a fresh window
+        assert eb.add_evidence("A", "B", "relates_to", 0.9, "x", "s3") is False
+        assert eb.get_evidence("A", "B", "relates_to").count == 1
+        assert eb.stats()["expired"] == 1
+
+    def test_per_label_thresholds(self):
+        from nornicdb_amd.cognitive.evidence import EvidenceBuffer
+        eb = EvidenceBuffer(now_fn=FakeClock())
+        # similar_to: 2 signals, 1 session, score 0.7
+        assert eb.add_evidence("A", "B", "similar_to", 0.9, "sim", "s1") is False
+        assert eb.add_evidence("A", "B", "similar_to", 0.8, "sim", "s1") is True
+        # coaccess: 5 signals, 3 sessions
+        for i in range(4):
+            assert eb.add_evidence("A", "B", "coaccess", 0.9, "co", f"s{i}") is False
+        assert eb.add_evidence("A", "B", "coaccess", 0.9, "co", "s9") is True
+
+
+class TestCooldownTable:
+    def test_window_blocks_then_allows(self):
+        from nornicdb_amd.cognitive.evidence import CooldownTable
+        clock = FakeClock()
+        ct = CooldownTable(now_fn=clock)
+        assert ct.can_materialize("A", "B", "coaccess") is True
+        ct.record_materialization("A", "B", "coaccess")
+        assert ct.can_materialize("A", "B", "coaccess") is False
+        assert 0 < ct.time_until_allowed("A", "B", "coaccess") <= 5 * 60
+        clock.advance(5 * 60 + 1)  # coaccess cooldown is 5 min
+        assert ct.can_materialize("A", "B", "coaccess") is True
+
+    def test_per_label_windows_and_cleanup(self):
+        from nornicdb_amd.cognitive.evidence import CooldownTable
+        clock = FakeClock()
+        ct = CooldownTable(now_fn=clock)
+        ct.record_materialization("A", "B", "coaccess")    # 5 min
+        ct.record_materialization("A", "B", "similar_to")  # 30 min
+        clock.advance(6 * 60)
+        assert ct.can_materialize("A", "B", "coaccess")
+        assert not ct.can_materialize("A", "B", "similar_to")
+        assert ct.cleanup() == 1
+        assert len(ct) == 1
+
+
+class TestInferenceDedupAndCooldown:
+    def _engine_pair(self):
+        from nornicdb_amd.storage import MemoryEngine, Node
+        eng = MemoryEngine()
+        eng.create_node(Node("a", ["Memory"], {"created_at": 0.0}))
+        eng.create_node(Node("b", ["Memory"], {"created_at": 1.0}))
+        return eng
+
+    def test_no_duplicate_edges_on_repeat_store(self):
+        from nornicdb_amd.cognitive import InferenceConfig, InferenceEngine
+        clock = FakeClock()
+        eng = self._engine_pair()
+        inf = InferenceEngine(eng, config=InferenceConfig(
+            evidence_required=1, temporal_window_s=10), now_fn=clock)
+        n = eng.get_node("b")
+        created1 = inf.on_store(n, "s1")
+        assert len(created1) == 1  # temporal proximity link a<->b
+        # repeat within cooldown: suggestion recurs, edge does not
+        created2 = inf.on_store(n, "s1")
+        assert created2 == []
+        assert eng.edge_count() == 1
+
+    def test_evidence_required_two_sessions(self):
+        from nornicdb_amd.cognitive import InferenceConfig, InferenceEngine
+        clock = FakeClock()
+        eng = self._engine_pair()
+        inf = InferenceEngine(eng, config=InferenceConfig(
+            evidence_required=2, temporal_window_s=10, cooldown_s=0.0),
+            now_fn=clock)
+        n = eng.get_node("b")
+        assert inf.on_store(n, "s1") == []      # 1 signal: buffered
+        clock.advance(1)
+        assert len(inf.on_store(n, "s2")) == 1  # 2nd signal, 2nd session
+        assert eng.edge_count() == 1
+
+
+class TestPatternDetector:
+    def test_daily_pattern(self):
+        import time as _time
+        from nornicdb_amd.cognitive.patterns import DAILY, PatternDetector
+        clock = FakeClock()
+        pd = PatternDetector(now_fn=clock)
+        # accesses concentrated at one wall-clock hour over 2 weeks
+        base = clock.t - (clock.t % 86400)
+        for day in range(14):
+            pd.record_access("n", base + day * 86400 + 9 * 3600 + 100)
+        pats = pd.detect_patterns("n")
+        daily = [p for p in pats if p.type == DAILY]
+        assert daily and daily[0].confidence > 0.5
+        hour, _, conf = pd.peak_access_time("n")
+        expect_hour = _time.localtime(base + 9 * 3600 + 100).tm_hour
+        assert hour == expect_hour
+
+    def test_burst_pattern(self):
+        from nornicdb_amd.cognitive.patterns import BURST, PatternDetector
+        clock = FakeClock()
+        pd = PatternDetector(now_fn=clock)
+        for i in range(12):
+            pd.record_access("n", clock.t - 30 + i * 2)
+        assert pd.has_pattern("n", BURST)
+
+    def test_trend_patterns_from_velocity(self):
+        from nornicdb_amd.cognitive.patterns import (DECAYING, GROWING,
+                                                     PatternDetector)
+        pd = PatternDetector(now_fn=FakeClock())
+        assert pd.has_pattern("unseen", GROWING, velocity=0.2)
+        assert pd.has_pattern("unseen", DECAYING, velocity=-0.2)
+        assert not pd.has_pattern("unseen", GROWING, velocity=0.0)
+
+
+class TestQueryLoadAdaptiveDecay:
+    def test_trend_and_adaptive_interval(self):
+        from nornicdb_amd.cognitive.patterns import QueryLoadPredictor
+        clock = FakeClock()
+        qlp = QueryLoadPredictor(now_fn=clock)
+        # idle: base decay interval
+        idle_interval = qlp.decay_interval(base=300.0)
+        assert abs(idle_interval - 300.0) < 30.0
+        # ramp load: 100 QPS for 30 seconds
+        for s in range(30):
+            qlp.record_queries(100)
+            clock.advance(1.0)
+        p = qlp.prediction()
+        assert p.current_qps > 20
+        assert p.total_queries == 3000
+        # A/B: decay sweep stretches under load (reference query-load
+        # adaptive decay behavior)
+        loaded_interval = qlp.decay_interval(base=300.0)
+        assert loaded_interval > idle_interval * 3
+
+    def test_spike_anomaly(self):
+        from nornicdb_amd.cognitive.patterns import QueryLoadPredictor
+        clock = FakeClock()
+        qlp = QueryLoadPredictor(now_fn=clock, spike_threshold=2.0)
+        for s in range(10):
+            qlp.record_queries(1)
+            clock.advance(1.0)
+        for s in range(5):
+            qlp.record_queries(200)
+            clock.advance(1.0)
+        p = qlp.prediction()
+        assert p.trend == "increasing"
+        assert p.is_anomaly and p.anomaly_type == "spike"
+
+
+class TestSessionBoundaries:
+    def test_session_id_changes_after_gap(self):
+        from nornicdb_amd.cognitive.temporal import AccessTracker, SESSION_GAP
+        clock = FakeClock()
+        tr = AccessTracker(now_fn=clock)
+        assert tr.is_session_boundary()
+        tr.record("n1")
+        s1 = tr.session_id
+        clock.advance(10)
+        tr.record("n2")
+        assert tr.session_id == s1            # same session
+        assert not tr.is_session_boundary()
+        clock.advance(SESSION_GAP + 1)
+        assert tr.is_session_boundary()
+        tr.record("n3")
+        assert tr.session_id != s1            # new session
+        assert len(tr.sessions()) == 2
