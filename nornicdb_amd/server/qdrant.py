@@ -10,8 +10,12 @@ own vectors; NornicDB does not auto-embed through this API.
 
 from __future__ import annotations
 
+import os
+import tempfile
 import threading
 import time
+
+import msgpack
 from typing import Any, Dict, List, Optional
 
 from fastapi import APIRouter, HTTPException
@@ -50,6 +54,84 @@ class QdrantRegistry:
         if c is None:
             raise KeyError(name)
         return c
+
+    # ---- snapshots (reference pkg/qdrantgrpc/snapshots_service.go) ----
+    @property
+    def snapshot_dir(self) -> str:
+        d = getattr(self, "_snapshot_dir", None)
+        if d is None:
+            d = os.environ.get("NORNICDB_QDRANT_SNAPSHOT_DIR",
+                               os.path.join(tempfile.gettempdir(),
+                                            "nornicdb-qdrant-snapshots"))
+            self._snapshot_dir = d
+        return d
+
+    def set_snapshot_dir(self, d: str):
+        self._snapshot_dir = d
+
+    def _snap_coll_dir(self, name: str) -> str:
+        return os.path.join(self.snapshot_dir, "collections", name)
+
+    def snapshot_create(self, name: str) -> Dict[str, Any]:
+        c = self.get(name)
+        d = self._snap_coll_dir(name)
+        os.makedirs(d, exist_ok=True)
+        ts = time.time()
+        snap_name = f"{name}-{int(ts * 1e9)}.snapshot"
+        blob = msgpack.packb({
+            "version": "qdrant-compat-1.0",
+            "collection": name,
+            "config": {"size": c.size, "distance": c.distance},
+            "points": {pid: {"vector": c.vectors.get(pid),
+                             "payload": c.payloads.get(pid, {})}
+                       for pid in c.payloads},
+            "timestamp": ts,
+        }, use_bin_type=True)
+        path = os.path.join(d, snap_name)
+        with open(path, "wb") as f:
+            f.write(blob)
+        return {"name": snap_name, "creation_time": ts,
+                "size": os.path.getsize(path)}
+
+    def snapshot_list(self, name: str) -> List[Dict[str, Any]]:
+        self.get(name)
+        d = self._snap_coll_dir(name)
+        out = []
+        if os.path.isdir(d):
+            for fn in sorted(os.listdir(d)):
+                if fn.endswith(".snapshot"):
+                    p = os.path.join(d, fn)
+                    st = os.stat(p)
+                    out.append({"name": fn, "creation_time": st.st_mtime,
+                                "size": st.st_size})
+        return out
+
+    def snapshot_path(self, name: str, snap: str) -> str:
+        p = os.path.join(self._snap_coll_dir(name), os.path.basename(snap))
+        if not os.path.exists(p):
+            raise KeyError(snap)
+        return p
+
+    def snapshot_delete(self, name: str, snap: str):
+        os.remove(self.snapshot_path(name, snap))
+
+    def snapshot_recover(self, name: str, location: str):
+        """Rebuild collection `name` from a snapshot file (file:// URI or
+        local path; reference recover semantics: replace contents)."""
+        path = location[7:] if location.startswith("file://") else location
+        with open(path, "rb") as f:
+            data = msgpack.unpackb(f.read(), raw=False)
+        cfg = data["config"]
+        with self._lock:
+            self.collections.pop(name, None)
+        self.create(name, int(cfg["size"]), cfg["distance"])
+        c = self.get(name)
+        for pid, rec in data["points"].items():
+            vec = rec.get("vector")
+            if vec is not None:
+                c.index.add(pid, vec)
+                c.vectors[pid] = list(vec)
+            c.payloads[pid] = rec.get("payload") or {}
 
 
 def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
@@ -303,6 +385,50 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
         for pid in body.get("points", []):
             c.index.remove(str(pid))
         return ok({"operation_id": 0, "status": "completed"})
+
+
+    # ---- snapshots (reference snapshots_service.go; Qdrant REST parity) ----
+    @r.post("/collections/{name}/snapshots")
+    @r.put("/collections/{name}/snapshots")
+    def create_snapshot(name: str):
+        try:
+            return ok(reg.snapshot_create(name))
+        except KeyError:
+            raise HTTPException(404, f"collection {name} not found")
+
+    @r.get("/collections/{name}/snapshots")
+    def list_snapshots(name: str):
+        try:
+            return ok(reg.snapshot_list(name))
+        except KeyError:
+            raise HTTPException(404, f"collection {name} not found")
+
+    @r.delete("/collections/{name}/snapshots/{snap}")
+    def delete_snapshot(name: str, snap: str):
+        try:
+            reg.snapshot_delete(name, snap)
+            return ok(True)
+        except KeyError:
+            raise HTTPException(404, "snapshot not found")
+
+    @r.get("/collections/{name}/snapshots/{snap}")
+    def download_snapshot(name: str, snap: str):
+        from fastapi.responses import FileResponse
+        try:
+            return FileResponse(reg.snapshot_path(name, snap),
+                                media_type="application/octet-stream",
+                                filename=snap)
+        except KeyError:
+            raise HTTPException(404, "snapshot not found")
+
+    @r.put("/collections/{name}/snapshots/recover")
+    def recover_snapshot(name: str, body: Dict[str, Any]):
+        loc = (body or {}).get("location", "")
+        try:
+            reg.snapshot_recover(name, loc)
+            return ok(True)
+        except (KeyError, OSError) as e:
+            raise HTTPException(404, f"recover failed: {e}")
 
     return r
 

@@ -13,6 +13,7 @@ of a collection are identical.
 
 Implemented RPCs (the reference's COMPAT.md core set):
   Collections: Create, Get, List, Delete, CollectionExists
+  Snapshots: Create, List, Delete (REST adds download + recover)
   Points: Upsert, Get, Delete, Count, Search, SearchBatch, Scroll,
           SetPayload, OverwritePayload, DeletePayload, ClearPayload
 """
@@ -270,6 +271,37 @@ def _build_pool():
         _f("collection_name", 1, _T.TYPE_STRING),
         _f("points", 3, _T.TYPE_MESSAGE, type_name=".qdrant.PointsSelector")])
 
+    # ---- snapshots.proto subset (official field numbers) ----
+    from google.protobuf import timestamp_pb2
+    ts_fd = descriptor_pb2.FileDescriptorProto()
+    timestamp_pb2.DESCRIPTOR.CopyToProto(ts_fd)
+    pool.Add(ts_fd)
+    fd.dependency.append("google/protobuf/timestamp.proto")
+    _msg(fd, "CreateSnapshotRequest", [
+        _f("collection_name", 1, _T.TYPE_STRING)])
+    _msg(fd, "ListSnapshotsRequest", [
+        _f("collection_name", 1, _T.TYPE_STRING)])
+    _msg(fd, "DeleteSnapshotRequest", [
+        _f("collection_name", 1, _T.TYPE_STRING),
+        _f("snapshot_name", 2, _T.TYPE_STRING)])
+    _msg(fd, "SnapshotDescription", [
+        _f("name", 1, _T.TYPE_STRING),
+        _f("creation_time", 2, _T.TYPE_MESSAGE,
+           type_name=".google.protobuf.Timestamp"),
+        _f("size", 3, _T.TYPE_INT64),
+        _f("checksum", 4, _T.TYPE_STRING)])
+    _msg(fd, "CreateSnapshotResponse", [
+        _f("snapshot_description", 1, _T.TYPE_MESSAGE,
+           type_name=".qdrant.SnapshotDescription"),
+        _f("time", 2, _T.TYPE_DOUBLE)])
+    _msg(fd, "ListSnapshotsResponse", [
+        _f("snapshot_descriptions", 1, _T.TYPE_MESSAGE,
+           label=_T.LABEL_REPEATED,
+           type_name=".qdrant.SnapshotDescription"),
+        _f("time", 2, _T.TYPE_DOUBLE)])
+    _msg(fd, "DeleteSnapshotResponse", [
+        _f("time", 1, _T.TYPE_DOUBLE)])
+
     pool.Add(fd)
     return pool
 
@@ -293,7 +325,10 @@ M = {n: _cls(n) for n in (
     "CountPoints", "CountResponse", "SearchPoints", "SearchResponse",
     "SearchBatchPoints", "SearchBatchResponse", "ScrollPoints",
     "ScrollResponse", "SetPayloadPoints", "DeletePayloadPoints",
-    "ClearPayloadPoints")}
+    "ClearPayloadPoints", "CreateSnapshotRequest", "ListSnapshotsRequest",
+    "DeleteSnapshotRequest", "SnapshotDescription",
+    "CreateSnapshotResponse", "ListSnapshotsResponse",
+    "DeleteSnapshotResponse")}
 
 _DISTANCES = {0: "cosine", 1: "cosine", 2: "euclid", 3: "dot", 4: "manhattan"}
 _DIST_NUM = {"cosine": 1, "euclid": 2, "dot": 3, "manhattan": 4}
@@ -575,6 +610,52 @@ class QdrantGrpc:
                 c.payloads[pid] = {}
         return self._op_ok()
 
+    # ---- qdrant.Snapshots (reference snapshots_service.go) ----
+    def SnapshotCreate(self, req, context):
+        import time as _time
+        t0 = _time.time()
+        try:
+            d = self.reg.snapshot_create(req.collection_name)
+        except KeyError:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"collection {req.collection_name!r} not found")
+        r = M["CreateSnapshotResponse"]()
+        r.snapshot_description.name = d["name"]
+        r.snapshot_description.size = d["size"]
+        r.snapshot_description.creation_time.FromSeconds(
+            int(d["creation_time"]))
+        r.time = _time.time() - t0
+        return r
+
+    def SnapshotList(self, req, context):
+        import time as _time
+        t0 = _time.time()
+        try:
+            snaps = self.reg.snapshot_list(req.collection_name)
+        except KeyError:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"collection {req.collection_name!r} not found")
+        r = M["ListSnapshotsResponse"]()
+        for d in snaps:
+            sd = r.snapshot_descriptions.add()
+            sd.name = d["name"]
+            sd.size = d["size"]
+            sd.creation_time.FromSeconds(int(d["creation_time"]))
+        r.time = _time.time() - t0
+        return r
+
+    def SnapshotDelete(self, req, context):
+        import time as _time
+        t0 = _time.time()
+        try:
+            self.reg.snapshot_delete(req.collection_name,
+                                          req.snapshot_name)
+        except KeyError:
+            context.abort(grpc.StatusCode.NOT_FOUND, "snapshot not found")
+        r = M["DeleteSnapshotResponse"]()
+        r.time = _time.time() - t0
+        return r
+
 
 def _handler(fn, req_cls):
     def call(request_bytes, context):
@@ -607,9 +688,16 @@ def _service_handlers(svc: QdrantGrpc):
                                   M["DeletePayloadPoints"]),
         "ClearPayload": _handler(svc.ClearPayload, M["ClearPayloadPoints"]),
     }
+    snapshots = {
+        "Create": _handler(svc.SnapshotCreate, M["CreateSnapshotRequest"]),
+        "List": _handler(svc.SnapshotList, M["ListSnapshotsRequest"]),
+        "Delete": _handler(svc.SnapshotDelete, M["DeleteSnapshotRequest"]),
+    }
     return (grpc.method_handlers_generic_handler("qdrant.Collections",
                                                  collections),
-            grpc.method_handlers_generic_handler("qdrant.Points", points))
+            grpc.method_handlers_generic_handler("qdrant.Points", points),
+            grpc.method_handlers_generic_handler("qdrant.Snapshots",
+                                                 snapshots))
 
 
 def serve(registry: QdrantRegistry = None, host: str = "127.0.0.1",
