@@ -173,6 +173,31 @@ class Authenticator:
                             hashlib.sha256).digest())
         return f"{header}.{payload}.{sig}"
 
+    def get_user(self, username: str) -> Dict:
+        try:
+            n = self.engine.get_node(self._user_id(username))
+        except NotFoundError:
+            raise AuthError(f"no such user {username!r}")
+        return {"username": username, "role": n.properties["role"],
+                "suspended": n.properties.get("suspended", False)}
+
+    def issue_token_for(self, username: str) -> str:
+        """Server-side token issuance WITHOUT a password — the OAuth
+        callback path, where the identity was established by the
+        provider (reference oauth.go HandleCallback issues the local
+        JWT the same way)."""
+        info = self.get_user(username)
+        if info.get("suspended"):
+            raise AuthError("account suspended")
+        header = _b64(json.dumps({"alg": "HS256", "typ": "JWT"}).encode())
+        payload = _b64(json.dumps({
+            "sub": info["username"], "role": info["role"],
+            "iat": int(self.now()), "exp": int(self.now() + self.ttl),
+        }).encode())
+        sig = _b64(hmac.new(self.secret, f"{header}.{payload}".encode(),
+                            hashlib.sha256).digest())
+        return f"{header}.{payload}.{sig}"
+
     def validate_token(self, token: str) -> Dict:
         try:
             header, payload, sig = token.split(".")

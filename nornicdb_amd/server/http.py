@@ -16,6 +16,7 @@ from fastapi import Depends, FastAPI, HTTPException, Request, Response
 from starlette.responses import StreamingResponse
 from pydantic import BaseModel
 
+from ..auth import AuthError
 from ..cypher import CypherRuntimeError, CypherSyntaxError
 from ..cypher.executor import Path
 from ..db import DatabaseManager
@@ -152,10 +153,109 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     def logout(_user=Depends(check_auth)):
         return {"status": "ok"}  # tokens are stateless JWTs
 
+    # ---- OAuth2/OIDC (reference pkg/auth/oauth.go + cmd/oauth-provider) --
+    oauth_provider = None
+    oauth_client = None
+    if auth is not None:
+        import os as _os
+        from ..auth.oauth import OAuthClientManager, OAuthProvider
+        if _os.environ.get("NORNICDB_OAUTH_PROVIDER_ENABLED", "").lower() \
+                in ("1", "true", "yes"):
+            oauth_provider = OAuthProvider(
+                _os.environ.get("NORNICDB_OAUTH_CLIENT_ID", "nornicdb"),
+                _os.environ.get("NORNICDB_OAUTH_CLIENT_SECRET", "secret"),
+                _os.environ.get("NORNICDB_OAUTH_ISSUER",
+                                "http://127.0.0.1:7474"),
+                authenticator=auth)
+        if _os.environ.get("NORNICDB_AUTH_PROVIDER") == "oauth":
+            oauth_client = OAuthClientManager(
+                auth,
+                _os.environ.get("NORNICDB_OAUTH_ISSUER", ""),
+                _os.environ.get("NORNICDB_OAUTH_CLIENT_ID", ""),
+                _os.environ.get("NORNICDB_OAUTH_CLIENT_SECRET", ""),
+                _os.environ.get("NORNICDB_OAUTH_CALLBACK_URL", ""))
+
+    @app.get("/.well-known/oauth-authorization-server")
+    @app.get("/.well-known/openid-configuration")
+    def oauth_discovery():
+        if oauth_provider is None:
+            raise HTTPException(404, "OAuth provider disabled")
+        return oauth_provider.discovery()
+
+    @app.get("/oauth2/v1/authorize")
+    def oauth_authorize(request: Request):
+        if oauth_provider is None:
+            raise HTTPException(404, "OAuth provider disabled")
+        status, body = oauth_provider.authorize(dict(request.query_params))
+        if status != 200:
+            raise HTTPException(status, body.get("error", "invalid_request"))
+        return body
+
+    async def _form(request: Request) -> dict:
+        # urlencoded parse without the python-multipart dependency
+        import urllib.parse as _up
+        raw = (await request.body()).decode()
+        return {k: v[0] for k, v in _up.parse_qs(raw).items()}
+
+    @app.post("/oauth2/v1/authorize/consent")
+    async def oauth_consent(request: Request):
+        if oauth_provider is None:
+            raise HTTPException(404, "OAuth provider disabled")
+        form = await _form(request)
+        status, body = oauth_provider.consent(
+            form.get("username", ""), form.get("password", ""),
+            form.get("redirect_uri", ""), form.get("state", ""),
+            form.get("scope", "openid profile"))
+        if status == 302:
+            from fastapi.responses import RedirectResponse
+            return RedirectResponse(body["location"], status_code=302)
+        raise HTTPException(status, body.get("error_description",
+                                             body.get("error", "denied")))
+
+    @app.post("/oauth2/v1/token")
+    async def oauth_token(request: Request):
+        if oauth_provider is None:
+            raise HTTPException(404, "OAuth provider disabled")
+        form = await _form(request)
+        status, body = oauth_provider.token(form)
+        if status != 200:
+            from fastapi.responses import JSONResponse
+            return JSONResponse(body, status_code=status)
+        return body
+
+    @app.get("/oauth2/v1/userinfo")
+    def oauth_userinfo(request: Request):
+        if oauth_provider is None:
+            raise HTTPException(404, "OAuth provider disabled")
+        status, body = oauth_provider.userinfo(
+            request.headers.get("authorization", ""))
+        if status != 200:
+            raise HTTPException(status, body.get("error", "invalid_token"))
+        return body
+
+    @app.get("/auth/oauth/login")
+    def oauth_login_url():
+        if oauth_client is None or not oauth_client.is_configured():
+            raise HTTPException(404, "OAuth login not configured")
+        url, state = oauth_client.generate_auth_url()
+        return {"auth_url": url, "state": state}
+
+    @app.get("/auth/oauth/callback")
+    def oauth_callback(code: str, state: str):
+        if oauth_client is None:
+            raise HTTPException(404, "OAuth login not configured")
+        try:
+            return oauth_client.handle_callback(code, state)
+        except AuthError as e:
+            raise HTTPException(401, str(e))
+
     @app.get("/auth/config")
     def auth_config():
-        return {"enabled": auth is not None, "oauth": False,
-                "methods": ["password", "token"] if auth else []}
+        return {"enabled": auth is not None,
+                "oauth": oauth_client is not None or oauth_provider is not None,
+                "methods": (["password", "token"]
+                            + (["oauth"] if oauth_client else []))
+                if auth else []}
 
     @app.get("/auth/me")
     def auth_me(user=Depends(check_auth)):
