@@ -43,6 +43,8 @@ SUPPORTED = {(5, m) for m in range(0, 9)} | {(4, m) for m in range(1, 5)}
 
 
 class BoltSession:
+    _tx_wrote = False
+
     def __init__(self, server: "BoltServer", reader, writer):
         self.server = server
         self.reader = reader
@@ -174,7 +176,7 @@ class BoltSession:
             self.write_message(R_IGNORED, {})
             return False
         if tag == M_RUN:
-            return self.on_run(msg.fields)
+            return await self.on_run(msg.fields)
         if tag == M_PULL:
             return self.on_pull(msg.fields[0] if msg.fields else {})
         if tag == M_DISCARD:
@@ -184,12 +186,23 @@ class BoltSession:
         if tag == M_BEGIN:
             meta = msg.fields[0] if msg.fields else {}
             self.db = meta.get("db", self.db)
+            if not await self.server.await_bookmarks(
+                    self.db, meta.get("bookmarks")):
+                self.write_message(R_FAILURE, {
+                    "code": "Neo.TransientError.Transaction.BookmarkTimeout",
+                    "message": "bookmark version not reached"})
+                return False
             self.in_tx = True
+            self._tx_wrote = False
             self.write_message(R_SUCCESS, {})
             return False
         if tag == M_COMMIT:
             self.in_tx = False
-            self.write_message(R_SUCCESS, {"bookmark": f"bm-{id(self):x}"})
+            if self._tx_wrote:
+                self.server.bump_version(self.db)
+                self._tx_wrote = False
+            self.write_message(R_SUCCESS,
+                               {"bookmark": self.server.bookmark(self.db)})
             return False
         if tag == M_ROLLBACK:
             # note: implicit-apply semantics — statements already applied.
@@ -256,11 +269,18 @@ class BoltSession:
             return {k: self.from_bolt(x) for k, x in v.items()}
         return v
 
-    def on_run(self, fields) -> bool:
+    async def on_run(self, fields) -> bool:
         query = fields[0] if fields else ""
         params = self.from_bolt(fields[1] if len(fields) > 1 else {})
         extra = fields[2] if len(fields) > 2 else {}
         db = extra.get("db") or self.db
+        if not self.in_tx and not await self.server.await_bookmarks(
+                db, extra.get("bookmarks")):
+            self.failed = True
+            self.write_message(R_FAILURE, {
+                "code": "Neo.TransientError.Transaction.BookmarkTimeout",
+                "message": "bookmark version not reached"})
+            return False
         try:
             result = self.server.execute(db, query, params or {})
         except (CypherSyntaxError,) as e:
@@ -277,6 +297,14 @@ class BoltSession:
             return False
         self.pending = result
         self.pending_pos = 0
+        wrote = any((result.stats or {}).get(k) for k in
+                    ("nodes_created", "nodes_deleted", "edges_created",
+                     "edges_deleted", "properties_set", "labels_added"))
+        if wrote:
+            if self.in_tx:
+                self._tx_wrote = True
+            else:
+                self.server.bump_version(db)
         self.write_message(R_SUCCESS, {"fields": result.columns,
                                        "t_first": 0,
                                        "qid": 0})
@@ -308,6 +336,9 @@ class BoltSession:
                     "db": self.db or "neo4j"}
             if counters:
                 meta["stats"] = counters
+            if not self.in_tx:
+                # autocommit summary carries the causal bookmark
+                meta["bookmark"] = self.server.bookmark(self.db)
             self.pending = None
             self.write_message(R_SUCCESS, meta)
         else:
@@ -318,9 +349,56 @@ class BoltSession:
 class BoltServer:
     """TCP server exposing executors per database name."""
 
+    def version_of(self, db: str) -> int:
+        return self._versions.get(db or "neo4j", 0)
+
+    def bump_version(self, db: str) -> int:
+        db = db or "neo4j"
+        self._versions[db] = self._versions.get(db, 0) + 1
+        return self._versions[db]
+
+    def bookmark(self, db: str) -> str:
+        db = db or "neo4j"
+        return f"ndb:{db}:{self.version_of(db)}"
+
+    @staticmethod
+    def parse_bookmark(bm: str):
+        """-> (db, version) or None for foreign/legacy bookmarks."""
+        parts = bm.split(":")
+        if len(parts) == 3 and parts[0] == "ndb":
+            try:
+                return parts[1], int(parts[2])
+            except ValueError:
+                return None
+        return None
+
+    async def await_bookmarks(self, db: str, bookmarks) -> bool:
+        """Causal consistency: block until this server has applied at
+        least the bookmarked version (satisfied immediately on a
+        single instance; bounded wait covers replicated catch-up)."""
+        want = 0
+        for bm in bookmarks or ():
+            parsed = self.parse_bookmark(bm)
+            if parsed and parsed[0] == (db or "neo4j"):
+                want = max(want, parsed[1])
+        if want <= self.version_of(db):
+            return True
+        deadline = asyncio.get_event_loop().time() + self.BOOKMARK_WAIT_TIMEOUT
+        while asyncio.get_event_loop().time() < deadline:
+            await asyncio.sleep(0.01)
+            if want <= self.version_of(db):
+                return True
+        return False
+
+    BOOKMARK_WAIT_TIMEOUT = 3.0
+
     def __init__(self, executor_for_db: Callable[[str], Any],
                  host: str = "127.0.0.1", port: int = 7687,
                  authenticator=None, version_str: str = "0.1.0"):
+        # causal bookmarks: per-db monotonically increasing commit
+        # version; bookmark = "ndb:<db>:<version>" (reference
+        # server.go:1617-1650 bookmark lifecycle)
+        self._versions: Dict[str, int] = {}
         self.executor_for_db = executor_for_db
         self.host = host
         self.port = port

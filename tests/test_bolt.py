@@ -219,3 +219,84 @@ class TestTemporalParams:
         s4 = ps.temporal_struct(v, bolt5=False)
         assert s4.tag == ps.DATETIME_LEGACY_TAG
         assert str(ps.temporal_from_struct(s4))[:19] == str(v)[:19]
+
+
+def test_causal_bookmarks():
+    """Bookmarks are causal tokens (ndb:<db>:<version>), not fabricated
+    ids: commits advance a per-db version, the
+    autocommit PULL summary carries the bookmark, and RUN waits for (or
+    times out on) future versions."""
+    async def run():
+        srv, port = await _start_server()
+        srv.BOOKMARK_WAIT_TIMEOUT = 0.15
+        try:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            c = _Client(reader, writer)
+            await c.handshake()
+            await c.send(M_HELLO, {"user_agent": "t", "scheme": "none"})
+            await c.recv()
+
+            async def run_pull(q, extra=None):
+                await c.send(M_RUN, q, {}, extra or {})
+                r1 = await c.recv()
+                if r1.tag == R_FAILURE:
+                    return r1.fields[0], None, []
+                await c.send(M_PULL, {"n": -1})
+                recs = []
+                while True:
+                    m = await c.recv()
+                    if m.tag == R_RECORD:
+                        recs.append(m.fields[0])
+                    else:
+                        return None, m.fields[0], recs
+
+            # write advances the version; summary carries the bookmark
+            _, s1, _ = await run_pull("CREATE (:B1)")
+            bm1 = s1["bookmark"]
+            assert bm1.startswith("ndb:neo4j:")
+            v1 = int(bm1.split(":")[2])
+            assert v1 >= 1
+            _, s2, _ = await run_pull("CREATE (:B2)")
+            assert int(s2["bookmark"].split(":")[2]) == v1 + 1
+            # reads do not advance
+            _, s3, _ = await run_pull("RETURN 1")
+            assert int(s3["bookmark"].split(":")[2]) == v1 + 1
+            # presenting a satisfied bookmark proceeds
+            _, s4, recs = await run_pull("MATCH (n:B1) RETURN count(n)",
+                                         {"bookmarks": [s2["bookmark"]]})
+            assert recs[0][0] == 1
+            # a future bookmark times out with a transient error
+            fail, _, _ = await run_pull("RETURN 1",
+                                        {"bookmarks": ["ndb:neo4j:999999"]})
+            assert fail and "Bookmark" in fail["code"]
+            await c.send(0x0F)  # RESET
+            await c.recv()
+        finally:
+            srv.close()
+    asyncio.run(run())
+
+
+def test_explicit_tx_commit_bookmark():
+    async def run():
+        srv, port = await _start_server()
+        try:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            c = _Client(reader, writer)
+            await c.handshake()
+            await c.send(M_HELLO, {"user_agent": "t", "scheme": "none"})
+            await c.recv()
+            await c.send(0x01, {})              # BEGIN
+            assert (await c.recv()).tag == R_SUCCESS
+            await c.send(M_RUN, "CREATE (:TX1)", {}, {})
+            await c.recv()
+            await c.send(M_PULL, {"n": -1})
+            while (await c.recv()).tag == R_RECORD:
+                pass
+            await c.send(0x02)                  # COMMIT
+            done = await c.recv()
+            assert done.tag == R_SUCCESS
+            bm = done.fields[0]["bookmark"]
+            assert bm.startswith("ndb:neo4j:") and int(bm.split(":")[2]) == 1
+        finally:
+            srv.close()
+    asyncio.run(run())
