@@ -285,14 +285,14 @@ def test_explicit_tx_commit_bookmark():
             await c.handshake()
             await c.send(M_HELLO, {"user_agent": "t", "scheme": "none"})
             await c.recv()
-            await c.send(0x01, {})              # BEGIN
+            await c.send(0x11, {})              # BEGIN
             assert (await c.recv()).tag == R_SUCCESS
             await c.send(M_RUN, "CREATE (:TX1)", {}, {})
             await c.recv()
             await c.send(M_PULL, {"n": -1})
             while (await c.recv()).tag == R_RECORD:
                 pass
-            await c.send(0x02)                  # COMMIT
+            await c.send(0x12)                  # COMMIT
             done = await c.recv()
             assert done.tag == R_SUCCESS
             bm = done.fields[0]["bookmark"]
