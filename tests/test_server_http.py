@@ -253,3 +253,16 @@ class TestAdminAuthRoutes:
     def test_bifrost_status_alias(self):
         c, _ = self._client()
         assert "available" in c.get("/api/bifrost/status").json()
+
+
+def test_admin_console_served_and_endpoints(client):
+    """The single-file admin SPA (reference ui/src/ parity) is served at /
+    and every endpoint it calls exists."""
+    html = client.get("/").text
+    assert "NornicDB-AMD Console" in html
+    for tab in ("tab-query", "tab-search", "tab-memory", "tab-databases",
+                "tab-admin"):
+        assert tab in html
+    for path in ("/admin/databases", "/admin/stats", "/nornicdb/embed/stats",
+                 "/nornicdb/decay", "/status"):
+        assert client.get(path).status_code == 200, path
