@@ -77,6 +77,80 @@ class LoginRequest(BaseModel):
     password: str
 
 
+class _TxFastPath:
+    """Raw-ASGI fast path for POST /db/{name}/tx/commit — the cross-
+    protocol hot route. Skips FastAPI routing/DI/validation (~0.5 ms per
+    request of the reference-gap ASGI tax): parse the
+    body, run the shared statement runner, emit json. Everything else
+    (auth present, other paths, errors) falls through to the FastAPI app.
+    """
+
+    def __init__(self, app, auth):
+        self.app = app
+        self.auth = auth
+
+    def __getattr__(self, item):  # delegate state/router/etc. for tests
+        return getattr(self.app, item)
+
+    async def __call__(self, scope, receive, send):
+        if (scope.get("type") == "http" and scope.get("method") == "POST"
+                and self.auth is None):
+            path = scope.get("path", "")
+            if path.startswith("/db/") and path.endswith("/tx/commit"):
+                db_name = path[4:-len("/tx/commit")]
+                if "/" not in db_name:
+                    await self._fast(db_name, receive, send)
+                    return
+            if path == "/graphql":
+                await self._fast_gql(receive, send)
+                return
+        await self.app(scope, receive, send)
+
+    async def _fast(self, db_name, receive, send):
+        chunks = []
+        while True:
+            ev = await receive()
+            chunks.append(ev.get("body", b""))
+            if not ev.get("more_body"):
+                break
+        status = 200
+        try:
+            body = json.loads(b"".join(chunks) or b"{}")
+            out = self.app._tx_commit_body(db_name, body)
+        except KeyError:
+            status, out = 404, {"detail": f"database {db_name} not found"}
+        except Exception as e:
+            out = {"results": [], "errors": [
+                {"code": "Neo.ClientError.Request.InvalidFormat",
+                 "message": str(e)}]}
+        await self._respond(send, status, out)
+
+    async def _fast_gql(self, receive, send):
+        chunks = []
+        while True:
+            ev = await receive()
+            chunks.append(ev.get("body", b""))
+            if not ev.get("more_body"):
+                break
+        try:
+            body = json.loads(b"".join(chunks) or b"{}")
+            out = self.app._gql_for(None).execute(
+                body.get("query", ""), body.get("variables"))
+            status = 200
+        except Exception as e:
+            status, out = 200, {"data": None, "errors": [{"message": str(e)}]}
+        await self._respond(send, status, out)
+
+    @staticmethod
+    async def _respond(send, status, out):
+        payload = json.dumps(out, default=str).encode()
+        await send({"type": "http.response.start", "status": status,
+                    "headers": [(b"content-type", b"application/json"),
+                                (b"content-length",
+                                 str(len(payload)).encode())]})
+        await send({"type": "http.response.body", "body": payload})
+
+
 def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastAPI:
     app = FastAPI(title="NornicDB-AMD", version=version)
     metrics = MetricsRegistry()
@@ -394,14 +468,8 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     # ---- Neo4j HTTP transaction API ----
     # raw-body endpoint (no pydantic model): this is the hot path and
     # schema validation costs more than the query itself
-    @app.post("/db/{db_name}/tx/commit")
-    async def tx_commit(db_name: str, request: Request,
-                        _user=Depends(check_auth)):
-        body = await request.json()
-        try:
-            db = mgr.get(db_name)
-        except KeyError:
-            raise HTTPException(404, f"database {db_name} not found")
+    def _tx_commit_body(db_name: str, body: dict) -> dict:
+        db = mgr.get(db_name)  # KeyError -> 404 at both call sites
         results, errors = [], []
         for stmt in body.get("statements", []):
             t0 = time.time()
@@ -424,6 +492,15 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
                                "message": str(e)})
                 break
         return {"results": results, "errors": errors}
+
+    @app.post("/db/{db_name}/tx/commit")
+    async def tx_commit(db_name: str, request: Request,
+                        _user=Depends(check_auth)):
+        body = await request.json()
+        try:
+            return _tx_commit_body(db_name, body)
+        except KeyError:
+            raise HTTPException(404, f"database {db_name} not found")
 
     # ---- explicit transaction lifecycle (reference server_db.go:381-1226:
     # simplified semantics — statements execute eagerly, commit returns a
@@ -758,4 +835,6 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     app.state.qdrant = QdrantRegistry()
     app.include_router(qdrant_router(app.state.qdrant))
 
-    return app
+    app._tx_commit_body = _tx_commit_body  # reused by the ASGI fast path
+    app._gql_for = _gql_for
+    return _TxFastPath(app, auth)
