@@ -31,7 +31,9 @@ def _open(args, cfg):
               file=sys.stderr)
         emb = create_embedder("mock", dims=cfg.embedding_dims)
     return open_db(args.data_dir or cfg.data_dir or None, embedder=emb,
-                   dims=cfg.embedding_dims)
+                   dims=cfg.embedding_dims,
+                   engine=getattr(cfg, "storage_engine", "disk") or "disk",
+                   encryption_passphrase=cfg.encryption_passphrase)
 
 
 def cmd_serve(args):

@@ -492,6 +492,174 @@ __global__ __launch_bounds__(GM_NTHREADS, 1) void k_gemm_nt(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Variant B: 96x256 tile, 4 waves, 3 workgroups/CU (the knn_mfma geometry).
+//
+// The 256x256 8-phase kernel above is a single-workgroup-per-CU design;
+// hardware ablation (scripts/gemm_ablate.py) showed its costs are
+// ADDITIVE — staging DMA, LDS fragment reads and the C store all
+// serialize with the MFMA stream because nothing else is resident on
+// the CU to overlap them. This variant trades per-tile efficiency for
+// CO-RESIDENCY: 3 small workgroups per CU (44 KB LDS each, 12 waves)
+// run phase-shifted, so one workgroup's staging and epilogue overlap
+// another's MFMA cluster — the same occupancy lesson the kNN kernel
+// established in round 1 (multi-WG occupancy IS the memory pipeline).
+// ---------------------------------------------------------------------------
+
+#define GB_BM 96
+#define GB_BN 256
+#define GB_BK 64
+#define GB_NT 256
+#define GB_MW (GB_BM / 16)
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GB_NT, 3) void k_gemm_nt96(
+    const unsigned short* __restrict__ A,     // [M,K] bf16
+    const unsigned short* __restrict__ W,     // [N,K] bf16
+    const unsigned short* __restrict__ bias,  // [N] or nullptr
+    unsigned short* __restrict__ C,           // [M,N] bf16
+    int tiles_m, int tiles_n, long long K, long long N) {
+  using namespace gemm_nt_detail;
+  __shared__ __align__(16) char smem[GB_BM * GB_BK * 2 + GB_BN * GB_BK * 2];
+  unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + GB_BM * GB_BK * 2);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;  // wave col 0..3 (64 output cols each)
+
+  // XCD-bijective remap + band-major walk (same rationale as variant A:
+  // a band of tm panels sweeps all tn so A stays L2/LLC resident).
+  int nwg = tiles_m * tiles_n;
+  int wg = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wg & 7, slot = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
+  }
+  int tm, tn;
+  {
+    const int BAND = 16;  // 16 x 96 rows = 3 MB A-band at K=1024
+    int nb_full = tiles_m / BAND;
+    int full_total = nb_full * BAND * tiles_n;
+    if (wg < full_total) {
+      int band = wg / (BAND * tiles_n);
+      int r2 = wg % (BAND * tiles_n);
+      tn = r2 / BAND;
+      tm = band * BAND + r2 % BAND;
+    } else {
+      int r2 = wg - full_total;
+      int bh = tiles_m - nb_full * BAND;
+      tn = r2 / bh;
+      tm = nb_full * BAND + r2 % bh;
+    }
+  }
+  const long long brow = (long long)tm * GB_BM;
+  const long long bcol = (long long)tn * GB_BN;
+  const long long ld = K * 2;
+  const char* gA = (const char*)A + brow * ld;
+  const char* gW = (const char*)W + bcol * ld;
+
+  float4v acc[GB_MW][4];
+#pragma unroll
+  for (int m = 0; m < GB_MW; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long long kt = 0; kt < K; kt += GB_BK) {
+    // stage A (96x64 = 12 KB: 12 chunks) + B (256x64 = 32 KB: 32 chunks)
+    // with the pre-swizzled-source global_load_lds pattern (knn_mfma).
+#pragma unroll
+    for (int it = 0; it < 3; ++it) {
+      int chunk = wc * 3 + it;
+      int x = chunk * 1024 + lane * 16;
+      int p = swz(x);
+      const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+          gA + (long long)(p >> 7) * ld + kt * 2 + (p & 127));
+      L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+      __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+    }
+#pragma unroll
+    for (int it = 0; it < 8; ++it) {
+      int chunk = wc * 8 + it;
+      int x = chunk * 1024 + lane * 16;
+      int p = swz(x);
+      const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+          gW + (long long)(p >> 7) * ld + kt * 2 + (p & 127));
+      L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+      __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) {
+        int c = wc * 64 + nn * 16 + (lane & 15);
+        bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+            (const char*)sB + swz(c * 128 + kb)));
+      }
+#pragma unroll
+      for (int m = 0; m < GB_MW; ++m) {
+        int r = m * 16 + (lane & 15);
+        bf16x8 af = (bf16x8)(*reinterpret_cast<const short8v*>(
+            (const char*)sA + swz(r * 128 + kb)));
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn)
+          acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr[nn], acc[m][nn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: bias (+GELU), bf16, LDS bounce in two 48-row half-passes
+  // ([48][256] bf16 = 24 KB, fits the freed staging space) for 16 B
+  // coalesced global stores.
+  float bvals[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    bvals[n] = HAS_BIAS
+        ? bf16_bits_to_f32(bias[bcol + wc * 64 + n * 16 + (lane & 15)])
+        : 0.0f;
+  }
+  unsigned short* sC = (unsigned short*)smem;  // [48][256]
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+#pragma unroll
+    for (int m = 0; m < 3; ++m) {
+      const int mi = half * 3 + m;
+      const int row0 = m * 16 + (lane >> 4) * 4;
+      const int col = wc * 64 + (lane & 15);
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          float v = acc[mi][n][e] + bvals[n];
+          if (ACT == 1) v = gelu_erf(v);
+          sC[(row0 + e) * 256 + col + n * 16] = f32_to_bf16_bits(v);
+        }
+      }
+    }
+    __syncthreads();
+    // store 48x256 bf16 = 24 KB: each thread 6 x 16 B
+    const long long grow0 = brow + half * 48;
+#pragma unroll
+    for (int j = 0; j < 6; ++j) {
+      int off = j * 4096 + tid * 16;          // 512 B per LDS row
+      int row = off >> 9;
+      int colb = off & 511;
+      uint4v v = *reinterpret_cast<const uint4v*>((const char*)sC + off);
+      *reinterpret_cast<uint4v*>((char*)C + (grow0 + row) * N * 2 +
+                                 bcol * 2 + colb) = v;
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------------------------------------------------------------------
 // host wrapper
 // ---------------------------------------------------------------------------
@@ -505,9 +673,15 @@ at::Tensor gemm_nt(at::Tensor a, at::Tensor w, c10::optional<at::Tensor> bias,
               "gemm_nt: W must be contiguous 2D bf16 CUDA");
   long long M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
-  TORCH_CHECK(M % GM_BM == 0, "gemm_nt: M % 256 == 0 (python pads)");
+  // kernel choice: variant B (96x256, 3 WGs/CU) by default — co-residency
+  // beats the 1-WG 256x256 template on the encoder shapes (ablation in
+  // scripts/gemm_ablate.py); NORNICDB_GEMM_KERNEL=256 forces variant A.
+  const char* kv = getenv("NORNICDB_GEMM_KERNEL");
+  bool use96 = !(kv && atoi(kv) == 256) && (M % GB_BM == 0);
+  TORCH_CHECK(use96 ? (M % GB_BM == 0) : (M % GM_BM == 0),
+              "gemm_nt: M tiling (python pads)");
   TORCH_CHECK(N % GM_BN == 0, "gemm_nt: N % 256 == 0 (python pads)");
-  TORCH_CHECK(K % (2 * GM_BK) == 0, "gemm_nt: K % 128 == 0");
+  TORCH_CHECK(K % (use96 ? GB_BK : 2 * GM_BK) == 0, "gemm_nt: K tiling");
   const unsigned short* bptr = nullptr;
   if (bias.has_value() && bias->defined() && bias->numel() > 0) {
     TORCH_CHECK(bias->is_cuda() && bias->is_contiguous() &&
@@ -518,13 +692,31 @@ at::Tensor gemm_nt(at::Tensor a, at::Tensor w, c10::optional<at::Tensor> bias,
   TORCH_CHECK(act <= 1, "gemm_nt: act in {0: none, 1: gelu, <0: ablation probe}");
 
   auto c = at::empty({M, N}, a.options());
-  int tiles_m = (int)(M / GM_BM), tiles_n = (int)(N / GM_BN);
-  dim3 grid(tiles_m * tiles_n);
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   const unsigned short* ap = (const unsigned short*)a.data_ptr();
   const unsigned short* wp = (const unsigned short*)w.data_ptr();
   unsigned short* cp = (unsigned short*)c.data_ptr();
+
+  if (use96 && act >= 0) {
+    int tm96 = (int)(M / GB_BM), tn96 = (int)(N / GB_BN);
+    dim3 g96(tm96 * tn96);
+    auto launch96 = [&](auto kern) {
+      hipLaunchKernelGGL(kern, g96, dim3(GB_NT), 0, stream, ap, wp, bptr, cp,
+                         tm96, tn96, K, N);
+    };
+    if (act == 1) {
+      if (bptr) launch96(k_gemm_nt96<1, true>);
+      else launch96(k_gemm_nt96<1, false>);
+    } else {
+      if (bptr) launch96(k_gemm_nt96<0, true>);
+      else launch96(k_gemm_nt96<0, false>);
+    }
+    HIP_CHECK_LAST();
+    return c;
+  }
+  int tiles_m = (int)(M / GM_BM), tiles_n = (int)(N / GM_BN);
+  dim3 grid(tiles_m * tiles_n);
 
   if (act < 0) {
     // ablation probe (timing only; output undefined for abl != 0):

@@ -36,6 +36,8 @@ class Config:
     snapshot_interval_s: float = 300.0
     async_writes: bool = True
     encryption_passphrase: str = ""
+    # "disk" (LSM on-disk engine, default) or "wal" (RAM + WAL snapshot)
+    storage_engine: str = "disk"
     # search
     hnsw_m: int = 16
     hnsw_ef_construction: int = 200
