@@ -52,10 +52,24 @@ def cmd_serve(args):
         if pw:
             print(f"initial admin user 'neo4j' password: {pw}")
 
+    # TLS (reference pkg/security TLS middleware): --tls enables both
+    # bolt+s and https with the configured or a generated self-signed pair
+    ssl_ctx = None
+    cert = key = None
+    if getattr(args, "tls", False) or getattr(args, "tls_cert", None):
+        from .utils.tls import ensure_self_signed, make_ssl_context
+        cert = getattr(args, "tls_cert", None)
+        key = getattr(args, "tls_key", None)
+        if not (cert and key):
+            cert, key = ensure_self_signed(
+                args.data_dir or cfg.data_dir or ".")
+            print(f"TLS: using self-signed pair {cert}")
+        ssl_ctx = make_ssl_context(cert, key)
+
     app = create_app(mgr, auth=auth)
     bolt = BoltServer(lambda db: mgr.get(db).executor,
                       host=cfg.bolt_host, port=args.bolt_port or cfg.bolt_port,
-                      authenticator=auth)
+                      authenticator=auth, ssl_context=ssl_ctx)
     grpc_server = None
     if getattr(args, "grpc_port", None):
         from .server.nornic_grpc import serve as grpc_serve
@@ -76,7 +90,8 @@ def cmd_serve(args):
         print(f"Bolt listening on {cfg.bolt_host}:{bolt.port}")
         config = uvicorn.Config(app, host=cfg.http_host,
                                 port=args.http_port or cfg.http_port,
-                                log_level="warning")
+                                log_level="warning",
+                                ssl_certfile=cert, ssl_keyfile=key)
         server = uvicorn.Server(config)
         print(f"HTTP listening on {cfg.http_host}:{config.port}")
         await server.serve()
@@ -243,6 +258,10 @@ def main(argv=None):
                     help="enable the Qdrant-compatible gRPC endpoint "
                          "(Qdrant default: 6334)")
     sp.add_argument("--auth", action="store_true")
+    sp.add_argument("--tls", action="store_true",
+                    help="enable TLS for Bolt and HTTP (self-signed if no cert)")
+    sp.add_argument("--tls-cert", default=None)
+    sp.add_argument("--tls-key", default=None)
     sp.set_defaults(fn=cmd_serve)
 
     for name, fn in (("init", cmd_init), ("decay", cmd_decay)):

@@ -394,12 +394,14 @@ class BoltServer:
 
     def __init__(self, executor_for_db: Callable[[str], Any],
                  host: str = "127.0.0.1", port: int = 7687,
-                 authenticator=None, version_str: str = "0.1.0"):
+                 authenticator=None, version_str: str = "0.1.0",
+                 ssl_context=None):
         # causal bookmarks: per-db monotonically increasing commit
         # version; bookmark = "ndb:<db>:<version>" (reference
         # server.go:1617-1650 bookmark lifecycle)
         self._versions: Dict[str, int] = {}
         self.executor_for_db = executor_for_db
+        self.ssl_context = ssl_context
         self.host = host
         self.port = port
         self.authenticator = authenticator
@@ -426,7 +428,7 @@ class BoltServer:
 
     async def start(self):
         self._server = await asyncio.start_server(
-            self._on_conn, self.host, self.port)
+            self._on_conn, self.host, self.port, ssl=self.ssl_context)
         if self.port == 0 and self._server.sockets:
             self.port = self._server.sockets[0].getsockname()[1]
         return self

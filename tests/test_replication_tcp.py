@@ -210,3 +210,66 @@ def test_ha_two_process_failover(tmp_path):
     r = json.load(open(out))
     assert r["promoted"] is True
     assert len(r["applied"]) == 5
+
+
+def test_bolt_over_tls(tmp_path):
+    """Bolt with TLS (reference pkg/security TLS middleware): handshake
+    + HELLO + RUN over an ssl-wrapped loopback connection."""
+    import asyncio
+    import ssl
+    import struct as st
+
+    from nornicdb_amd.bolt import packstream as ps
+    from nornicdb_amd.bolt.server import BoltServer
+    from nornicdb_amd.cypher.executor import Executor
+    from nornicdb_amd.storage import MemoryEngine
+    from nornicdb_amd.utils.tls import ensure_self_signed, make_ssl_context
+
+    cert, key = ensure_self_signed(str(tmp_path))
+    server_ctx = make_ssl_context(cert, key)
+
+    async def run():
+        ex = Executor(MemoryEngine())
+        srv = BoltServer(lambda db: ex, host="127.0.0.1", port=0,
+                         ssl_context=server_ctx)
+        await srv.start()
+        try:
+            cctx = ssl.create_default_context()
+            cctx.check_hostname = False
+            cctx.verify_mode = ssl.CERT_NONE
+            reader, writer = await asyncio.open_connection(
+                "127.0.0.1", srv.port, ssl=cctx)
+            writer.write(st.pack(">I", 0x6060B017)
+                         + bytes([0, 0, 4, 4]) + bytes(12))
+            await writer.drain()
+            resp = await reader.readexactly(4)
+            assert resp[3] == 4  # negotiated major
+
+            def send(tag, *fields):
+                data = ps.pack(ps.Structure(tag, list(fields)))
+                writer.write(st.pack(">H", len(data)) + data + b"\x00\x00")
+
+            async def recv():
+                buf = b""
+                while True:
+                    size = st.unpack(">H", await reader.readexactly(2))[0]
+                    if size == 0:
+                        if buf:
+                            return ps.unpack(buf)
+                        continue
+                    buf += await reader.readexactly(size)
+
+            send(0x01, {"user_agent": "tls-test", "scheme": "none"})
+            await writer.drain()
+            hello = await recv()
+            assert hello.tag == 0x70
+            send(0x10, "RETURN 42 AS x", {}, {})
+            send(0x3F, {"n": -1})
+            await writer.drain()
+            assert (await recv()).tag == 0x70
+            rec = await recv()
+            assert rec.tag == 0x71 and rec.fields[0] == [42]
+            writer.close()
+        finally:
+            srv.close()
+    asyncio.run(run())
