@@ -60,6 +60,8 @@ class InferenceEngine:
         self.cooldowns = CooldownTable(
             {lb: self.cfg.cooldown_s for lb in _REASON_LABEL.values()},
             now_fn=now_fn)
+        from ..storage.node_config import NodeConfigStore
+        self.node_configs = NodeConfigStore(engine)
 
     # ---- main hook ----
     def on_store(self, node: Node, session_id: str = "") -> List[Edge]:
@@ -129,6 +131,14 @@ class InferenceEngine:
                     reason: str, session_id: str = "") -> Optional[Edge]:
         if confidence < self.cfg.min_confidence:
             return None
+        # per-node overrides: deny/pin lists, trust-adjusted confidence
+        # bar, edge caps (reference pkg/storage/node_config.go)
+        if self.node_configs is not None:
+            ok, _why = self.node_configs.is_edge_allowed(
+                a, b, self.cfg.edge_type, confidence,
+                self.cfg.min_confidence)
+            if not ok:
+                return None
         s, d = sorted((a, b))
         label = _REASON_LABEL.get(reason, "relates_to")
         # cooldown first: a recently-materialized pair accumulates no new

@@ -20,11 +20,14 @@ from .schema import Constraint, SchemaManager, VectorIndexMeta
 from .composite import CompositeEngine
 from .disk import DiskEngine, DiskTransaction
 from .lsm import LSMStore
+from .node_config import (LabelConfig, NodeConfig, NodeConfigStore,
+                          TRUST_DEFAULT, TRUST_HIGH, TRUST_LOW,
+                          TRUST_VERIFIED)
 
 __all__ = [
     "Node", "Edge", "Engine", "EventType", "StorageError", "NotFoundError",
     "ConstraintViolation", "new_id", "MemoryEngine", "PersistentEngine",
     "Transaction", "AsyncEngine", "NamespacedEngine", "WAL", "WALCorruption",
     "SchemaManager", "Constraint", "VectorIndexMeta", "CompositeEngine",
-    "WALDegraded", "DiskEngine", "DiskTransaction", "LSMStore",
+    "WALDegraded", "DiskEngine", "DiskTransaction", "LSMStore", "NodeConfig", "NodeConfigStore", "LabelConfig",
 ]

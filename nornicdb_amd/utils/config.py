@@ -61,6 +61,13 @@ class Config:
         "inference": False,
         "temporal_tracking": True,
         "heimdall": False,
+        # reference feature_flags.go parity set
+        "evidence_buffering": True,   # NORNICDB_EVIDENCE_BUFFERING_ENABLED
+        "cooldown": True,             # NORNICDB_COOLDOWN_ENABLED
+        "edge_provenance": False,
+        "per_node_config": False,
+        "auto_tlp_llm_qc": False,     # HeimdallQC gate for auto links
+        "gpu_clustering_auto": True,
     })
 
     def flag(self, name: str, default: bool = False) -> bool:
@@ -100,6 +107,9 @@ def load_config(path: Optional[str] = None, env: Dict[str, str] = None,
         name = key[len(ENV_PREFIX):].lower()
         if name.startswith("flag_"):
             cfg.flags[name[5:]] = _coerce(value, bool)
+        elif name.endswith("_enabled"):
+            # reference-style flags: NORNICDB_EVIDENCE_BUFFERING_ENABLED=1
+            cfg.flags[name[:-8]] = _coerce(value, bool)
         elif hasattr(cfg, name):
             cur = getattr(cfg, name)
             setattr(cfg, name, _coerce(value, type(cur)))

@@ -531,3 +531,57 @@ class TestSessionBoundaries:
         tr.record("n3")
         assert tr.session_id != s1            # new session
         assert len(tr.sessions()) == 2
+
+
+class TestNodeConfig:
+    """Per-node overrides gating auto-links (reference node_config.go)."""
+
+    def _inf(self):
+        from nornicdb_amd.cognitive import InferenceConfig, InferenceEngine
+        from nornicdb_amd.storage import MemoryEngine, Node
+        clock = FakeClock()
+        eng = MemoryEngine()
+        eng.create_node(Node("a", ["Memory"], {"created_at": 0.0}))
+        eng.create_node(Node("b", ["Memory"], {"created_at": 1.0}))
+        inf = InferenceEngine(eng, config=InferenceConfig(
+            evidence_required=1, temporal_window_s=10), now_fn=clock)
+        return eng, inf
+
+    def test_denied_pair_never_links(self):
+        eng, inf = self._inf()
+        inf.node_configs.get_or_create("a").add_deny("b")
+        assert inf.on_store(eng.get_node("b"), "s1") == []
+        assert eng.edge_count() == 0
+
+    def test_low_trust_raises_confidence_bar(self):
+        eng, inf = self._inf()
+        from nornicdb_amd.storage import TRUST_LOW
+        inf.node_configs.get_or_create("b").trust_level = TRUST_LOW
+        # temporal suggestion confidence ~0.5-0.8 < 0.5+0.2 bar -> blocked
+        created = inf.on_store(eng.get_node("b"), "s1")
+        assert created == [] or all(
+            e.properties["confidence"] >= 0.7 for e in created)
+
+    def test_pinned_always_allowed(self):
+        eng, inf = self._inf()
+        from nornicdb_amd.storage import TRUST_LOW
+        c = inf.node_configs.get_or_create("b")
+        c.trust_level = TRUST_LOW
+        c.add_pin("a")
+        assert len(inf.on_store(eng.get_node("b"), "s1")) == 1
+
+    def test_label_cap(self):
+        from nornicdb_amd.storage import (LabelConfig, MemoryEngine, Node,
+                                          NodeConfigStore)
+        from nornicdb_amd.storage.types import Edge
+        eng = MemoryEngine()
+        for nid in ("x", "y", "z"):
+            eng.create_node(Node(nid, [], {}))
+        eng.create_edge(Edge("e1", "REL", "x", "y", {}))
+        store = NodeConfigStore(eng)
+        c = store.get_or_create("x")
+        c.label_configs["REL"] = LabelConfig(max_edges=1)
+        ok, why = store.is_edge_allowed("x", "z", "REL", 1.0, 0.0)
+        assert not ok and "max capacity" in why
+        ok, _ = store.is_edge_allowed("x", "z", "OTHER", 1.0, 0.0)
+        assert ok
