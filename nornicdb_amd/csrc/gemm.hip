@@ -602,32 +602,15 @@ __global__ __launch_bounds__(GB_NT, 3) void k_gemm_nt96(
         bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
             (const char*)sB + swz(c * 128 + kb)));
       }
-      // A reads software-pipelined one m-group ahead via inline-asm
-      // ds_read + counted lgkmcnt (see knn_mfma.hip: the compiler
-      // coalesces a source-level double buffer back into one register)
-      {
-        const unsigned lA = (unsigned)(unsigned long long)(L_AS char*)sA;
-        bf16x8 af[2];
-        asm volatile("ds_read_b128 %0, %1"
-                     : "=&v"(af[0])
-                     : "v"(lA + swz((lane & 15) * 128 + kb)));
 #pragma unroll
-        for (int m = 0; m < GB_MW; ++m) {
-          if (m + 1 < GB_MW) {
-            asm volatile("ds_read_b128 %0, %1"
-                         : "=&v"(af[(m + 1) & 1])
-                         : "v"(lA + swz(((m + 1) * 16 + (lane & 15)) * 128
-                                        + kb)));
-            asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");
-          } else {
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-          }
-          __builtin_amdgcn_sched_barrier(0);  // guide rule 18
+      for (int m = 0; m < GB_MW; ++m) {
+        int r = m * 16 + (lane & 15);
+        bf16x8 af = (bf16x8)(*reinterpret_cast<const short8v*>(
+            (const char*)sA + swz(r * 128 + kb)));
 #pragma unroll
-          for (int nn = 0; nn < 4; ++nn)
-            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[m & 1], bfr[nn], acc[m][nn], 0, 0, 0);
-        }
+        for (int nn = 0; nn < 4; ++nn)
+          acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr[nn], acc[m][nn], 0, 0, 0);
       }
     }
     __syncthreads();

@@ -112,45 +112,25 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
       __syncthreads();
 
       // ---- MFMA over the staged tile: 2 k-steps of 32 ----
-      // A-fragment reads are software-pipelined one m-group ahead via
-      // inline-asm ds_read_b128 + counted lgkmcnt(1) waits: the naive
-      // form compiles to ds_read -> lgkmcnt(0) -> 4 MFMA per group into
-      // ONE register (the compiler coalesces a source-level double
-      // buffer back into an anti-dependency), exposing the full LDS
-      // latency against ~20 cy of math every group.
-      {
-        const unsigned lA = (unsigned)(unsigned long long)(L_AS char*)sA;
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
-          bf16x8 bfr[4];
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+        bf16x8 bfr[4];
 #pragma unroll
-          for (int nn = 0; nn < 4; ++nn) {
-            int c = wc * 64 + nn * 16 + (lane & 15);
-            bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                (const char*)sB + swz(c * 128 + kb)));
-          }
-          bf16x8 af[2];
-          asm volatile("ds_read_b128 %0, %1"
-                       : "=&v"(af[0])
-                       : "v"(lA + swz((lane & 15) * 128 + kb)));
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sB + swz(c * 128 + kb)));
+        }
 #pragma unroll
-          for (int m = 0; m < MW; ++m) {
-            if (m + 1 < MW) {
-              asm volatile("ds_read_b128 %0, %1"
-                           : "=&v"(af[(m + 1) & 1])
-                           : "v"(lA + swz(((m + 1) * 16 + (lane & 15)) * 128
-                                          + kb)));
-              asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");
-            } else {
-              asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            }
-            __builtin_amdgcn_sched_barrier(0);  // guide rule 18
+        for (int m = 0; m < MW; ++m) {
+          int r = m * 16 + (lane & 15);
+          bf16x8 af = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sA + swz(r * 128 + kb)));
 #pragma unroll
-            for (int nn = 0; nn < 4; ++nn)
-              acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  af[m & 1], bfr[nn], acc[m][nn], 0, 0, 0);
-          }
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
         }
       }
       __syncthreads();
