@@ -96,6 +96,14 @@ def cmd_serve(args):
         print(f"HTTP listening on {cfg.http_host}:{config.port}")
         await server.serve()
 
+    # graceful SIGTERM (systemd/docker stop, test harnesses): without
+    # this the process dies mid-buffer and the finally never runs
+    import signal
+
+    def _term(_sig, _frm):
+        raise KeyboardInterrupt
+
+    signal.signal(signal.SIGTERM, _term)
     try:
         asyncio.run(main())
     except KeyboardInterrupt:

@@ -301,7 +301,8 @@ class LSMStore:
     def __init__(self, path: str, sync_on_write: bool = False,
                  memtable_bytes: int = 8 << 20, max_tables: int = 12,
                  cache_bytes: int = 32 << 20, crypt=None,
-                 compact_interval: float = 2.0):
+                 compact_interval: float = 2.0,
+                 sync_interval: float = 0.05):
         self.path = path
         os.makedirs(path, exist_ok=True)
         self.crypt = crypt
@@ -322,6 +323,13 @@ class LSMStore:
         self._compactor = threading.Thread(target=self._compact_loop,
                                            args=(compact_interval,), daemon=True)
         self._compactor.start()
+        # batched-fsync loop (reference wal.go:377): commits are pushed to
+        # the OS on every batch (flush) and to disk every sync_interval.
+        self._syncer = None
+        if not sync_on_write and sync_interval > 0:
+            self._syncer = threading.Thread(target=self._sync_loop,
+                                            args=(sync_interval,), daemon=True)
+            self._syncer.start()
 
     # ------------------------------------------------------------------
     # open / manifest / log replay
@@ -455,6 +463,11 @@ class LSMStore:
                 self._log.flush()
                 os.fsync(self._log.fileno())
                 self._log_dirty = False
+            else:
+                # push the record out of the PYTHON buffer on every batch:
+                # a killed process must lose at most the fsync window, not
+                # whole buffered batches (serve-integration regression)
+                self._log.flush()
             for k, v in puts:
                 self._mem_put(k, v)
             for k in dels:
@@ -467,6 +480,13 @@ class LSMStore:
 
     def delete(self, k: bytes):
         self.write_batch(dels=[k])
+
+    def _sync_loop(self, interval: float):
+        while not self._stop.wait(interval):
+            try:
+                self.sync()
+            except Exception:
+                pass
 
     def sync(self):
         with self._lock:

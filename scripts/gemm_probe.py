@@ -15,12 +15,13 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from nornicdb_amd.ops import require_native  # noqa: E402
 
 # encoder shapes at bench operating point: M = 256 batch x 256 seq tokens
-SHAPES = [
-    ("qkv", 65536, 3072, 1024, 0),
-    ("attn_out", 65536, 1024, 1024, 0),
-    ("ffn_up", 65536, 4096, 1024, 1),
-    ("ffn_down", 65536, 1024, 4096, 0),
-]
+def shapes(m):
+    return [
+        ("qkv", m, 3072, 1024, 0),
+        ("attn_out", m, 1024, 1024, 0),
+        ("ffn_up", m, 4096, 1024, 1),
+        ("ffn_down", m, 1024, 4096, 0),
+    ]
 
 
 def bench(fn, iters, warmup=10):
@@ -37,13 +38,15 @@ def bench(fn, iters, warmup=10):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--m", type=int, default=65536,
+                    help="65280 = divisible by both 96 and 256 tiles")
     ap.add_argument("--refcheck", action="store_true", default=True)
     args = ap.parse_args()
     nat = require_native()
     dev = "cuda"
     out = {"shapes": []}
 
-    for name, m, n, k, act in SHAPES:
+    for name, m, n, k, act in shapes(args.m):
         torch.manual_seed(1234)
         x = (torch.randn(m, k, device=dev) / k ** 0.25).to(torch.bfloat16)
         w = (torch.randn(n, k, device=dev) / k ** 0.25).to(torch.bfloat16)

@@ -169,3 +169,29 @@ def test_linear_act_autograd_gpu(monkeypatch):
         d = (g.float() - gr).abs().max().item()
         scale = max(gr.abs().max().item(), 1e-3)
         assert d / scale < 0.05
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("m,n,k,act", [
+    (960, 512, 512, 0),      # variant B path: M % 96 == 0
+    (1920, 1024, 1024, 1),   # variant B + GELU
+    (96, 256, 128, 0),       # single variant-B tile
+])
+def test_gemm_nt96_variant_gpu(m, n, k, act):
+    """The 96x256 3-WG/CU kernel (selected when M % 96 == 0) — the
+    earlier suite's shapes all silently fell back to variant A."""
+    from nornicdb_amd.ops.gemm import gemm_nt
+    torch.manual_seed(m + n + k)
+    x = (torch.randn(m, k, device="cuda") / k ** 0.25).to(torch.bfloat16)
+    w = (torch.randn(n, k, device="cuda") / k ** 0.25).to(torch.bfloat16)
+    b = torch.randn(n, device="cuda").to(torch.bfloat16)
+    y = gemm_nt(x, w, b, act)
+    ref = F.linear(x.float(), w.float(), b.float())
+    if act == 1:
+        ref = F.gelu(ref)
+    torch.cuda.synchronize()
+    rel = (y.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1e-6)
+    assert rel < 0.02, rel
+    i, j = m // 2, n // 3
+    assert abs(y[i, j].float().item() - ref[i, j].item()) \
+        < 0.05 * max(1.0, abs(ref[i, j].item()))
