@@ -354,3 +354,28 @@ class TestLouvainMultilevel:
                       np.array([1], np.int32))
         c = louvain(g1)
         assert c[0] == c[1]  # two connected nodes merge
+
+
+@pytest.mark.gpu
+def test_louvain_gpu_local_moving():
+    """device='cuda' runs the local-moving phase on GPU with identical
+    community quality."""
+    from nornicdb_amd.graph.algos import louvain, modularity
+    rng = np.random.default_rng(5)
+    nc, per = 8, 50
+    srcs, dsts = [], []
+    for c in range(nc):
+        base = c * per
+        srcs.append(rng.integers(base, base + per, 400))
+        dsts.append(rng.integers(base, base + per, 400))
+    srcs.append(rng.integers(0, nc * per, 200))
+    dsts.append(rng.integers(0, nc * per, 200))
+    s = np.concatenate(srcs); d = np.concatenate(dsts)
+    o = np.argsort(s, kind="stable"); s, d = s[o], d[o]
+    rp = np.searchsorted(s, np.arange(nc * per + 1))
+    from nornicdb_amd.graph.csr import CSRGraph
+    g = CSRGraph([str(i) for i in range(nc * per)], rp.astype(np.int64),
+                 d.astype(np.int32))
+    c_gpu = louvain(g, device="cuda")
+    c_cpu = louvain(g, device="cpu")
+    assert abs(modularity(g, c_gpu) - modularity(g, c_cpu)) < 0.05

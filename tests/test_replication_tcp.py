@@ -273,3 +273,58 @@ def test_bolt_over_tls(tmp_path):
         finally:
             srv.close()
     asyncio.run(run())
+
+
+def test_cluster_node_replicated_engine_three_members():
+    """Full cluster story in-process over REAL TCP: three ClusterNodes
+    with their own engines; a write through a FOLLOWER's
+    ReplicatedEngine is forwarded to the leader, committed by Raft, and
+    readable on every member (reference serve --cluster semantics)."""
+    import time as _t
+
+    from nornicdb_amd.replication.cluster import ClusterNode
+    from nornicdb_amd.storage import MemoryEngine, Node
+
+    ports = free_ports(3)
+    ids = ["c0", "c1", "c2"]
+    peers = {i: ("127.0.0.1", p) for i, p in zip(ids, ports)}
+    nodes = [ClusterNode(i, peers, MemoryEngine()).start() for i in ids]
+    try:
+        deadline = _t.time() + 10
+        leader = None
+        while _t.time() < deadline and leader is None:
+            _t.sleep(0.05)
+            for n in nodes:
+                if n.raft.is_leader:
+                    leader = n
+        assert leader is not None, [n.health() for n in nodes]
+        follower = next(n for n in nodes if n is not leader)
+
+        # write through the LEADER's replicated engine
+        leader.replicated.create_node(Node("L1", ["C"], {"v": 1}))
+        # write through a FOLLOWER (forwarded to the leader)
+        follower.replicated.create_node(Node("F1", ["C"], {"v": 2}))
+
+        deadline = _t.time() + 5
+        while _t.time() < deadline:
+            if all(n.engine.has_node("L1") and n.engine.has_node("F1")
+                   for n in nodes):
+                break
+            _t.sleep(0.05)
+        for n in nodes:
+            assert n.engine.get_node("L1").properties["v"] == 1, n.id
+            assert n.engine.get_node("F1").properties["v"] == 2, n.id
+
+        # follower reads its own forwarded write locally
+        assert follower.replicated.get_node("F1").properties["v"] == 2
+        # delete replicates too
+        leader.replicated.delete_node("L1")
+        deadline = _t.time() + 5
+        while _t.time() < deadline:
+            if all(not n.engine.has_node("L1") for n in nodes):
+                break
+            _t.sleep(0.05)
+        assert all(not n.engine.has_node("L1") for n in nodes)
+    finally:
+        for n in nodes:
+            n.stop()
