@@ -44,7 +44,33 @@ def cmd_serve(args):
     import uvicorn
 
     cfg = load_config(args.config)
-    mgr = _open(args, cfg)
+    cluster_node = None
+    if getattr(args, "cluster_peers", None):
+        # replicated mode (reference serve cluster flags + :7688 transport):
+        # --cluster-id me --cluster-peers me=host:7688,n2=host2:7688,...
+        from .db import DatabaseManager
+        from .embed import create_embedder
+        from .replication.cluster import ClusterNode
+        from .storage import DiskEngine, MemoryEngine
+        peers = {}
+        for part in args.cluster_peers.split(","):
+            nid, addr = part.split("=", 1)
+            host, port = addr.rsplit(":", 1)
+            peers[nid.strip()] = (host.strip(), int(port))
+        me = args.cluster_id or sorted(peers)[0]
+        data_dir = args.data_dir or cfg.data_dir
+        base = DiskEngine(data_dir) if data_dir else MemoryEngine()
+        cluster_node = ClusterNode(me, peers, base).start()
+        try:
+            emb = create_embedder(cfg.embedder, dims=cfg.embedding_dims)
+        except Exception:
+            emb = create_embedder("mock", dims=cfg.embedding_dims)
+        mgr = DatabaseManager(cluster_node.replicated, embedder=emb,
+                              dims=cfg.embedding_dims)
+        print(f"cluster member {me} of {sorted(peers)} "
+              f"(transport {peers[me][0]}:{peers[me][1]})")
+    else:
+        mgr = _open(args, cfg)
     auth = None
     if cfg.auth_enabled or args.auth:
         auth = Authenticator(mgr.get("system").engine)
@@ -113,6 +139,8 @@ def cmd_serve(args):
             grpc_server.stop(0)
         if qdrant_grpc_server is not None:
             qdrant_grpc_server.stop(0)
+        if cluster_node is not None:
+            cluster_node.stop()
         mgr.close()
 
 
@@ -270,6 +298,10 @@ def main(argv=None):
                     help="enable TLS for Bolt and HTTP (self-signed if no cert)")
     sp.add_argument("--tls-cert", default=None)
     sp.add_argument("--tls-key", default=None)
+    sp.add_argument("--cluster-id", default=None,
+                    help="this member's id in --cluster-peers")
+    sp.add_argument("--cluster-peers", default=None,
+                    help="id=host:port,... Raft cluster over TCP")
     sp.set_defaults(fn=cmd_serve)
 
     for name, fn in (("init", cmd_init), ("decay", cmd_decay)):

@@ -107,7 +107,12 @@ def build_apoc_procedures(db) -> Dict[str, Any]:
     @register("apoc.community.louvain")
     def _louvain(ex):
         g = from_engine(eng, undirected=True)
-        comm = louvain(g)
+        # large graphs route the local-moving phase to the GPU (10M edges:
+        # ~1.5 s vs ~20 s CPU — profiles/README.md)
+        import torch as _torch
+        dev = "cuda" if (g.m > 2_000_000
+                         and _torch.cuda.is_available()) else "cpu"
+        comm = louvain(g, device=dev)
         return ["node", "community"], [[eng.get_node(g.node_ids[i]), int(comm[i])]
                                        for i in range(g.n)]
 

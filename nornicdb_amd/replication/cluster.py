@@ -64,14 +64,11 @@ class ClusterNode:
     def _on_message(self, msg: Dict[str, Any]):
         t = msg.get("type")
         if t == "fwd_write":
-            ok = False
-            if self.raft.is_leader:
-                ok = self._propose_and_wait(msg["command"],
-                                            cmd_id=msg["cmd_id"])
-            self.transport.send(msg["from"], {
-                "type": "fwd_ack", "from": self.id,
-                "cmd_id": msg["cmd_id"], "ok": ok,
-                "leader": self.raft.leader_id})
+            # handle OFF the transport reader thread: propose-and-wait
+            # needs the reader to keep delivering the follower's
+            # append-acks, or the commit (and this handler) deadlock
+            threading.Thread(target=self._handle_forward, args=(msg,),
+                             daemon=True).start()
         elif t == "fwd_ack":
             with self._lock:
                 ev = self._applied.get("fwd:" + msg["cmd_id"])
@@ -80,6 +77,15 @@ class ClusterNode:
                 ev.set()
         else:
             self.raft.on_message(msg)
+
+    def _handle_forward(self, msg: Dict[str, Any]):
+        ok = False
+        if self.raft.is_leader:
+            ok = self._propose_and_wait(msg["command"], cmd_id=msg["cmd_id"])
+        self.transport.send(msg["from"], {
+            "type": "fwd_ack", "from": self.id,
+            "cmd_id": msg["cmd_id"], "ok": ok,
+            "leader": self.raft.leader_id})
 
     def _apply(self, command: Dict[str, Any]):
         self.adapter.apply(command)

@@ -328,3 +328,79 @@ def test_cluster_node_replicated_engine_three_members():
     finally:
         for n in nodes:
             n.stop()
+
+
+def test_serve_cluster_two_processes(tmp_path):
+    """TWO real `serve --cluster` processes: a write over HTTP against
+    one member becomes readable on the OTHER member (Raft over the
+    cluster TCP port + follower write forwarding end to end)."""
+    import urllib.request
+
+    ports = free_ports(6)  # 2x (cluster, bolt, http)
+    cl = {f"n{i}": ("127.0.0.1", ports[i]) for i in range(2)}
+    peers_arg = ",".join(f"{k}={h}:{p}" for k, (h, p) in cl.items())
+    procs = []
+    env = dict(os.environ, PYTHONPATH=REPO)
+    try:
+        for i in range(2):
+            procs.append(subprocess.Popen(
+                [sys.executable, "-m", "nornicdb_amd", "serve",
+                 "--cluster-id", f"n{i}", "--cluster-peers", peers_arg,
+                 "--bolt-port", str(ports[2 + i]),
+                 "--http-port", str(ports[4 + i])],
+                cwd=REPO, env=env, stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT))
+        # wait for both HTTP endpoints
+        deadline = time.time() + 60
+        for i in range(2):
+            while time.time() < deadline:
+                try:
+                    urllib.request.urlopen(
+                        f"http://127.0.0.1:{ports[4+i]}/health", timeout=1)
+                    break
+                except Exception:
+                    time.sleep(0.2)
+            else:
+                raise AssertionError(
+                    f"member {i} never came up: "
+                    f"{procs[i].stdout.read(3000)}")
+        time.sleep(1.0)  # election settle
+
+        def tx(port, stmt):
+            body = json.dumps({"statements": [{"statement": stmt}]}).encode()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/db/neo4j/tx/commit", data=body,
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=10) as r:
+                return json.loads(r.read())
+
+        # write via member 0 (leader OR follower — forwarding covers both)
+        out = tx(ports[4], "CREATE (:Clu {who: 'm0'}) RETURN 1")
+        assert not out["errors"], out
+        # visible on member 1
+        deadline = time.time() + 10
+        n = 0
+        while time.time() < deadline:
+            out = tx(ports[5], "MATCH (c:Clu) RETURN count(c)")
+            n = out["results"][0]["data"][0]["row"][0]
+            if n == 1:
+                break
+            time.sleep(0.2)
+        assert n == 1, out
+        # and a write via member 1 lands on member 0
+        tx(ports[5], "CREATE (:Clu {who: 'm1'}) RETURN 1")
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            out = tx(ports[4], "MATCH (c:Clu) RETURN count(c)")
+            if out["results"][0]["data"][0]["row"][0] == 2:
+                break
+            time.sleep(0.2)
+        assert out["results"][0]["data"][0]["row"][0] == 2, out
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
