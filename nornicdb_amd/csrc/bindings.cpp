@@ -8,6 +8,8 @@ std::tuple<at::Tensor, at::Tensor> knn_gemv(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out);
 
 // knn_mfma.hip
+long long knn_bm_value();
+#define KNN_BM_VALUE knn_bm_value()
 std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
                                             long long row_base, int k_out);
 
@@ -67,6 +69,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused cosine score + top-k for <=16 queries (bf16 db)",
         py::arg("db"), py::arg("q"), py::arg("row_base") = 0,
         py::arg("k_out") = 10);
+  m.attr("KNN_BM") = KNN_BM_VALUE;
   m.def("knn_mfma", &knn_mfma,
         "Fused MFMA cosine score + top-k, 256-query batches (bf16 db)",
         py::arg("db"), py::arg("q"), py::arg("row_base") = 0,

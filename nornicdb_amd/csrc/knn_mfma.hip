@@ -35,7 +35,10 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define L_AS __attribute__((address_space(3)))
 
 // tile geometry
-#define BM 96
+#ifndef KNN_BM
+#define KNN_BM 96
+#endif
+#define BM KNN_BM
 #define BN 256
 #define BK 64
 #define NTHREADS 256
@@ -315,3 +318,5 @@ std::tuple<at::Tensor, at::Tensor> knn_mfma(at::Tensor db, at::Tensor q,
   HIP_CHECK_LAST();
   return {out_s, out_i};
 }
+
+long long knn_bm_value() { return BM; }

@@ -22,6 +22,12 @@ import torch
 from . import native_or_none, require_native
 
 
+def _knn_bm() -> int:
+    """Panel row count compiled into the kernel (KNN_BM, default 96)."""
+    nat = native_or_none()
+    return int(getattr(nat, "KNN_BM", 96)) if nat is not None else 96
+
+
 def knn_search_exact(
     db: torch.Tensor, q: torch.Tensor, k: int, row_base: int = 0
 ) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -85,7 +91,7 @@ def knn_search(
         k <= 10
         and db.dtype == torch.bfloat16
         and db.shape[1] % 64 == 0
-        and db.shape[0] >= 96
+        and db.shape[0] >= _knn_bm()
     ):
         if q.shape[0] <= 256:
             return _knn_mfma(nat, db, q, k, row_base)
@@ -111,7 +117,7 @@ def knn_search(
 
 
 def _knn_mfma(nat, db, q, k, row_base):
-    """Fused MFMA score+topk over full 96-row panels + torch-scored tail."""
+    """Fused MFMA score+topk over full BM-row panels + torch-scored tail."""
     qn = q.shape[0]
     qq = q.to(torch.bfloat16)
     if qn < 256:
@@ -120,7 +126,8 @@ def _knn_mfma(nat, db, q, k, row_base):
         )
     qq = qq.contiguous()
     n = db.shape[0]
-    n_main = (n // 96) * 96
+    bm = _knn_bm()
+    n_main = (n // bm) * bm
     s, i = nat.knn_mfma(db.narrow(0, 0, n_main), qq, row_base, k)
     s, i = s[:qn], i[:qn]
     if n_main < n:

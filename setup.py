@@ -32,7 +32,11 @@ setup(
             sources=sources,
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17"],
+                # NORNICDB_KNN_BM overrides the kNN panel-tile rows
+                # (experiment knob; default 96 in knn_mfma.hip)
+                "nvcc": ["-O3", "-std=c++17"] + (
+                    ["-DKNN_BM=" + os.environ["NORNICDB_KNN_BM"]]
+                    if os.environ.get("NORNICDB_KNN_BM") else []),
             },
         )
     ],
