@@ -114,13 +114,27 @@ def cmd_serve(args):
     async def main():
         await bolt.start()
         print(f"Bolt listening on {cfg.bolt_host}:{bolt.port}")
-        config = uvicorn.Config(app, host=cfg.http_host,
-                                port=args.http_port or cfg.http_port,
-                                log_level="warning",
-                                ssl_certfile=cert, ssl_keyfile=key)
-        server = uvicorn.Server(config)
-        print(f"HTTP listening on {cfg.http_host}:{config.port}")
-        await server.serve()
+        http_port = args.http_port or cfg.http_port
+        if os.environ.get("NORNICDB_HTTP_SERVER") == "uvicorn":
+            config = uvicorn.Config(app, host=cfg.http_host, port=http_port,
+                                    log_level="warning",
+                                    ssl_certfile=cert, ssl_keyfile=key)
+            server = uvicorn.Server(config)
+            print(f"HTTP listening on {cfg.http_host}:{config.port}")
+            await server.serve()
+        else:
+            # default: in-repo asyncio HTTP server (server/fasthttp.py) —
+            # ~2x uvicorn/h11 throughput on the tx/GraphQL hot routes
+            from .server.fasthttp import start_http_server
+            sctx = None
+            if cert and key:
+                from .utils.tls import make_ssl_context
+                sctx = make_ssl_context(cert, key)
+            srv = await start_http_server(app, cfg.http_host, http_port,
+                                          ssl_context=sctx)
+            print(f"HTTP listening on {cfg.http_host}:{http_port}")
+            async with srv:
+                await srv.serve_forever()
 
     # graceful SIGTERM (systemd/docker stop, test harnesses): without
     # this the process dies mid-buffer and the finally never runs

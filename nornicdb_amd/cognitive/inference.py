@@ -203,7 +203,9 @@ class HeimdallQC:
             prompt = (f"Should '{a_node.properties.get('name', a_node.id)}' "
                       f"link to '{b_node.properties.get('name', b_node.id)}' "
                       f"because {reason}? Answer yes or no.")
-            out = self.manager.generate(prompt, max_tokens=4)
+            # greedy: deterministic verdicts, and temperature<=0 takes the
+            # in-kernel multi-token decode path (models/heimdall.py)
+            out = self.manager.generate(prompt, max_tokens=4, temperature=0.0)
             ok = "no" not in out.lower().split()
         except Exception:
             ok = True

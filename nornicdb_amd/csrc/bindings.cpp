@@ -58,6 +58,17 @@ void decode_step(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
                  long long n_heads, long long n_kv, long long hd,
                  long long inter, long long max_len, double rms_eps,
                  long long pos);
+double sync_bench(long long iters, long long which, long long grid,
+                  at::Tensor scratch);
+void decode_tokens(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
+                   at::Tensor attn, at::Tensor h, at::Tensor rope_cos,
+                   at::Tensor rope_sin, at::Tensor embed_w,
+                   at::Tensor norm_w, at::Tensor lm_w, at::Tensor out,
+                   at::Tensor n_done, at::Tensor pmax, at::Tensor pidx,
+                   long long n_layers, long long hidden, long long n_heads,
+                   long long n_kv, long long hd, long long inter,
+                   long long max_len, double rms_eps, long long pos,
+                   long long start_tok, long long n_toks, long long eos);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "NornicDB-AMD CDNA4 (gfx950) native kernels";
@@ -94,6 +105,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("n_layers"), py::arg("hidden"), py::arg("n_heads"),
         py::arg("n_kv"), py::arg("hd"), py::arg("inter"),
         py::arg("max_len"), py::arg("rms_eps"), py::arg("pos"));
+  m.def("decode_tokens", &decode_tokens,
+        "fused cooperative multi-token greedy decode (embed + layers + "
+        "lm_head argmax, whole generation loop in one launch)",
+        py::arg("layer_ptrs"), py::arg("x"), py::arg("q"), py::arg("attn"),
+        py::arg("h"), py::arg("rope_cos"), py::arg("rope_sin"),
+        py::arg("embed_w"), py::arg("norm_w"), py::arg("lm_w"),
+        py::arg("out"), py::arg("n_done"), py::arg("pmax"), py::arg("pidx"),
+        py::arg("n_layers"), py::arg("hidden"), py::arg("n_heads"),
+        py::arg("n_kv"), py::arg("hd"), py::arg("inter"),
+        py::arg("max_len"), py::arg("rms_eps"), py::arg("pos"),
+        py::arg("start_tok"), py::arg("n_toks"), py::arg("eos") = -1);
+  m.def("sync_bench", &sync_bench,
+        "grid-barrier microbenchmark: ms for `iters` barriers "
+        "(which=0 cg::grid.sync, 1 two-level custom)",
+        py::arg("iters"), py::arg("which"), py::arg("grid"),
+        py::arg("scratch"));
   m.def("kmeans_assign", &kmeans_assign,
         "fused distance+argmin assignment (bf16 x/centroids)");
   m.def("kmeans_accum", &kmeans_accum, "atomic centroid accumulate");

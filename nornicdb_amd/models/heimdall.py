@@ -347,6 +347,14 @@ class FusedDecoder:
         self.attn = torch.zeros_like(self.x)
         self.hbuf = torch.zeros(c.intermediate_size, device=dev,
                                 dtype=torch.float32)
+        # multi-token greedy path (decode_tokens): weights + scratch
+        self.embed_w = model.embed.weight.data.contiguous()
+        self.norm_w = model.norm.weight.data.contiguous()
+        self.lm_w = model.lm_head.weight.data.contiguous()
+        self.tok_out = torch.zeros(512, dtype=torch.int32, device=dev)
+        self.n_done = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.pmax = torch.zeros(512, dtype=torch.float32, device=dev)
+        self.pidx = torch.zeros(512, dtype=torch.int32, device=dev)
 
     @torch.no_grad()
     def step_logits(self, token_id: torch.Tensor, pos: int) -> torch.Tensor:
@@ -364,6 +372,30 @@ class FusedDecoder:
         return (xn.to(torch.bfloat16) @ m.lm_head.weight.T).float()
 
     @torch.no_grad()
+    def decode_greedy(self, start_tok: int, pos0: int, n_toks: int,
+                      eos_id=None) -> list:
+        """Up to n_toks greedy tokens in ONE cooperative launch.
+
+        The kernel embeds start_tok at pos0, runs all layers, takes the
+        lm_head argmax in-kernel, feeds it back, and repeats — no host
+        round-trip per token. Returns the emitted token ids (shorter
+        than n_toks iff EOS was emitted)."""
+        c = self.cfg
+        n = min(n_toks, self.max_len - pos0, self.tok_out.numel())
+        if n <= 0:
+            return []
+        self.n_done.zero_()
+        self.nat.decode_tokens(
+            self.layer_ptrs, self.x, self.qs, self.attn, self.hbuf,
+            self.rope_cos, self.rope_sin, self.embed_w, self.norm_w,
+            self.lm_w, self.tok_out, self.n_done, self.pmax, self.pidx,
+            c.num_layers, c.hidden_size, c.num_heads, c.num_kv_heads,
+            self.hd, c.intermediate_size, self.max_len, c.rms_eps, pos0,
+            start_tok, n, -1 if eos_id is None else int(eos_id))
+        k = int(self.n_done.item())
+        return self.tok_out[:k].tolist()
+
+    @torch.no_grad()
     def generate(self, token_ids: torch.Tensor, max_new_tokens: int = 32,
                  temperature: float = 0.8, top_k: int = 40,
                  eos_id=None, stream_cb=None):
@@ -378,6 +410,34 @@ class FusedDecoder:
         last = logits[:, -1, :].float().view(-1)
         out = []
         pos = s
+        import os as _os
+        if temperature <= 0 and _os.environ.get("NORNICDB_DECODE_INKERNEL") == "1":
+            # whole greedy loop in-kernel (decode_tokens, one launch per
+            # chunk). Measured SLOWER than the host loop on one stream
+            # (445 vs 511 tok/s: the in-kernel lm_head runs on the
+            # 64-WG cooperative grid while torch's GEMV uses the full
+            # chip), but it frees the host entirely — opt-in for
+            # concurrent serving. Chunked so stream_cb stays responsive.
+            tok = int(last.argmax().item())
+            out.append(tok)
+            if stream_cb is not None:
+                stream_cb(tok)
+            if eos_id is not None and tok == eos_id:
+                return out
+            chunk = 16 if stream_cb is not None else 256
+            while len(out) < max_new_tokens and pos < self.max_len:
+                want = min(max_new_tokens - len(out), chunk)
+                toks = self.decode_greedy(out[-1], pos, want, eos_id=eos_id)
+                if not toks:
+                    break
+                pos += len(toks)
+                for t2 in toks:
+                    out.append(t2)
+                    if stream_cb is not None:
+                        stream_cb(t2)
+                if eos_id is not None and toks[-1] == eos_id:
+                    break
+            return out
         cur = None
         for _ in range(max_new_tokens):
             if temperature <= 0:

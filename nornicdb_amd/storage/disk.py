@@ -438,6 +438,14 @@ class DiskEngine(Engine):
                 out.append(n)
         return out
 
+    def iter_nodes_by_label(self, label: str) -> Iterator[Node]:
+        pref = _pk(P_LABEL, label.encode()) + SEP
+        ids = [k[len(pref):].decode() for k, _ in self._kv.scan(pref)]
+        for nid in ids:
+            n = self._get_or_none(nid)
+            if n is not None:
+                yield n
+
     def all_nodes(self) -> Iterator[Node]:
         for k, v in self._kv.scan(P_NODE):
             yield self._load_node(k[1:].decode(), v)

@@ -151,6 +151,19 @@ class MemoryEngine(Engine):
                 return len(self._label_index.get(label, ()))
             return self._label_ns_counts.get((label, ns), 0)
 
+    def iter_nodes_by_label(self, label: str):
+        """Lazy copying label scan: snapshot the id set (cheap), then copy
+        nodes one at a time — pagination (GraphQL nodes(limit:)) stops
+        after `limit` copies instead of copying the whole label set."""
+        with self._lock:
+            ids = list(self._label_index.get(label, ()))
+        for i in ids:
+            with self._lock:
+                n = self._nodes.get(i)
+                c = n.copy() if n is not None else None
+            if c is not None:
+                yield c
+
     def iter_nodes_raw(self, label: str = None):
         """Yield LIVE node objects without copying — read-only fast paths
         (reference storage_fastpaths.go). Callers must not mutate."""

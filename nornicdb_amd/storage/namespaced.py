@@ -70,6 +70,13 @@ class NamespacedEngine(Engine):
         return [self._unwrap_node(n) for n in self.inner.get_nodes_by_label(label)
                 if self._mine(n.id)]
 
+    def iter_nodes_by_label(self, label: str):
+        # inner yields fresh copies (Engine read contract), so unwrapping
+        # in place is safe
+        for n in self.inner.iter_nodes_by_label(label):
+            if self._mine(n.id):
+                yield self._unwrap_node(n)
+
     def all_nodes(self):
         for n in self.inner.all_nodes():
             if self._mine(n.id):

@@ -93,6 +93,19 @@ class Engine:
     def get_nodes_by_label(self, label: str) -> List[Node]: raise NotImplementedError
     def all_nodes(self) -> Iterator[Node]: raise NotImplementedError
     def node_count(self) -> int: raise NotImplementedError
+
+    # lazy label-scan primitives: engines override to avoid copying the
+    # whole label set when the caller wants a page or a count (GraphQL
+    # nodes(limit:)/nodeCount resolvers). Defaults fall back to the
+    # eager list so every Engine keeps working unchanged.
+    def iter_nodes_by_label(self, label: str) -> Iterator[Node]:
+        return iter(self.get_nodes_by_label(label))
+
+    def node_ids_by_label(self, label: str) -> List[str]:
+        return [n.id for n in self.get_nodes_by_label(label)]
+
+    def node_count_by_label(self, label: str) -> int:
+        return len(self.node_ids_by_label(label))
     def has_node(self, node_id: str) -> bool:
         try:
             self.get_node(node_id)
