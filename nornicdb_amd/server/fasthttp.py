@@ -169,6 +169,12 @@ class AsgiHttpProtocol(asyncio.Protocol):
                 sent["v"] = True
                 return {"type": "http.request", "body": body,
                         "more_body": False}
+            # starlette's StreamingResponse awaits a second receive() as a
+            # disconnect listener — block until the client actually goes
+            # away (returning http.disconnect immediately aborts streams)
+            while not self._closed:
+                self._wake.clear()
+                await self._wake.wait()
             return {"type": "http.disconnect"}
 
         state = {"started": False, "clen": None, "chunked_out": False,
@@ -272,6 +278,6 @@ async def start_http_server(app, host: str, port: int,
     """Bind and serve the ASGI app; returns the asyncio server (use
     server.sockets[0].getsockname()[1] for the bound port)."""
     loop = asyncio.get_event_loop()
-    return await loop.start_server(
+    return await loop.create_server(
         lambda: AsgiHttpProtocol(app, is_tls=ssl_context is not None),
         host, port, ssl=ssl_context, backlog=512, reuse_address=True)

@@ -135,19 +135,37 @@ def main():
         loop.run_forever()
 
     app = create_app(mgr)
-    http_cfg = uvicorn.Config(app, host="127.0.0.1", port=0, log_level="error")
-    http_srv = uvicorn.Server(http_cfg)
+    use_uvicorn = os.environ.get("NORNICDB_HTTP_SERVER") == "uvicorn"
+    if use_uvicorn:
+        http_cfg = uvicorn.Config(app, host="127.0.0.1", port=0,
+                                  log_level="error")
+        http_srv = uvicorn.Server(http_cfg)
 
-    def http_thread():
-        asyncio.new_event_loop()
-        http_srv.run()
+        def http_thread():
+            asyncio.new_event_loop()
+            http_srv.run()
+    else:
+        from nornicdb_amd.server.fasthttp import start_http_server
+
+        def http_thread():
+            loop = asyncio.new_event_loop()
+            asyncio.set_event_loop(loop)
+            srv = loop.run_until_complete(
+                start_http_server(app, "127.0.0.1", 0))
+            ports["http"] = srv.sockets[0].getsockname()[1]
+            loop.run_forever()
 
     threading.Thread(target=bolt_thread, daemon=True).start()
     threading.Thread(target=http_thread, daemon=True).start()
     t0 = time.time()
-    while ("bolt" not in ports or not http_srv.started) and time.time() - t0 < 15:
+
+    def _http_up():
+        return http_srv.started if use_uvicorn else "http" in ports
+
+    while ("bolt" not in ports or not _http_up()) and time.time() - t0 < 15:
         time.sleep(0.05)
-    http_port = http_srv.servers[0].sockets[0].getsockname()[1]
+    http_port = (http_srv.servers[0].sockets[0].getsockname()[1]
+                 if use_uvicorn else ports["http"])
 
     results = {}
 

@@ -255,7 +255,7 @@ class TestInferenceQCAndClusters:
         from nornicdb_amd.storage.types import Node
 
         class StubMgr:
-            def generate(self, prompt, max_tokens=4):
+            def generate(self, prompt, max_tokens=4, temperature=0.0):
                 return "no , unrelated"
         qc = HeimdallQC(manager=StubMgr())
         a = Node(id="a", labels=[], properties={})
