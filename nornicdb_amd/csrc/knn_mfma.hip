@@ -57,15 +57,11 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
     int d,               // inner dim, % BK == 0
     long long row_base, float* __restrict__ cand_score,
     int* __restrict__ cand_idx) {
-  // K-loop uses sA (12 KB) only — the query tile (B) is NOT staged: it
-  // depends on kt, not on the panel, so with tens of thousands of panels
-  // it is L2-resident and re-staging it burned 96 KB of LDS traffic per
-  // tile (32 KB DMA write + 32 KB fragment read + the issue slots). The
-  // MFMA loop loads B fragments straight from global (L2-hot) instead.
-  // Epilogue reuses the space as the [256][36] fp32 transposed chunk
-  // (36.9 KB, the high-water mark).
-  __shared__ __align__(16) char smem[BN * 36 * 4];
+  // 44 KB: K-loop uses sA (12 KB) + sB (32 KB); epilogue reuses the same
+  // space as the [256][36] fp32 transposed chunk (36.9 KB).
+  __shared__ __align__(16) char smem[BM * BK * 2 + BN * BK * 2];
   unsigned short* sA = (unsigned short*)smem;
+  unsigned short* sB = (unsigned short*)(smem + BM * BK * 2);
   float* sT = (float*)smem;  // epilogue alias [256][36]
 
   const int tid = threadIdx.x;
@@ -105,25 +101,30 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
         L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
         __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
       }
-      // B fragments straight from global (L2-hot query block): all 8
-      // fragment loads (4 cols x 2 k-steps) are issued before the first
-      // MFMA so they overlap the sA DMA wait
-      bf16x8 bfr[2][4];
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-        for (int nn = 0; nn < 4; ++nn) {
-          int c = wc * 64 + nn * 16 + (lane & 15);
-          int kq = kt + ks * 32 + (lane >> 4) * 8;
-          bfr[ks][nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
-              qs + (long long)c * d + kq));
-        }
+      for (int it = 0; it < 8; ++it) {  // B: 32 chunks, 4 waves x 8
+        int chunk = wc * 8 + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)(p >> 7) * d2 + (long long)kt * 2 +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
       __syncthreads();
 
       // ---- MFMA over the staged tile: 2 k-steps of 32 ----
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const int kb = (ks * 32 + (lane >> 4) * 8) * 2;
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          bfr[nn] = (bf16x8)(*reinterpret_cast<const short8v*>(
+              (const char*)sB + swz(c * 128 + kb)));
+        }
 #pragma unroll
         for (int m = 0; m < MW; ++m) {
           int r = m * 16 + (lane & 15);
@@ -132,7 +133,7 @@ __global__ __launch_bounds__(NTHREADS, 3) void k_knn_mfma(
 #pragma unroll
           for (int nn = 0; nn < 4; ++nn)
             acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af, bfr[ks][nn], acc[m][nn], 0, 0, 0);
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
         }
       }
       __syncthreads();
