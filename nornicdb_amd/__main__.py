@@ -96,7 +96,9 @@ def cmd_serve(args):
     app = create_app(mgr, auth=auth)
     bolt = BoltServer(lambda db: mgr.get(db).executor,
                       host=cfg.bolt_host, port=args.bolt_port or cfg.bolt_port,
-                      authenticator=auth, ssl_context=ssl_ctx)
+                      authenticator=auth, ssl_context=ssl_ctx,
+                      log_queries=(getattr(args, "log_queries", False)
+                                   or cfg.log_queries))
     grpc_server = None
     if getattr(args, "grpc_port", None):
         from .server.nornic_grpc import serve as grpc_serve
@@ -309,6 +311,8 @@ def main(argv=None):
                     help="enable the Qdrant-compatible gRPC endpoint "
                          "(Qdrant default: 6334)")
     sp.add_argument("--auth", action="store_true")
+    sp.add_argument("--log-queries", action="store_true",
+                    help="log every Cypher query with duration to stdout")
     sp.add_argument("--tls", action="store_true",
                     help="enable TLS for Bolt and HTTP (self-signed if no cert)")
     sp.add_argument("--tls-cert", default=None)

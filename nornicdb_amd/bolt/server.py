@@ -395,7 +395,7 @@ class BoltServer:
     def __init__(self, executor_for_db: Callable[[str], Any],
                  host: str = "127.0.0.1", port: int = 7687,
                  authenticator=None, version_str: str = "0.1.0",
-                 ssl_context=None):
+                 ssl_context=None, log_queries: bool = False):
         # causal bookmarks: per-db monotonically increasing commit
         # version; bookmark = "ndb:<db>:<version>" (reference
         # server.go:1617-1650 bookmark lifecycle)
@@ -406,6 +406,8 @@ class BoltServer:
         self.port = port
         self.authenticator = authenticator
         self.version_str = version_str
+        # reference pkg/bolt/server.go:440 LogQueries — stdout query log
+        self.log_queries = log_queries
         self._server: Optional[asyncio.AbstractServer] = None
 
     def authenticate(self, auth: Dict[str, Any]):
@@ -424,7 +426,21 @@ class BoltServer:
 
     def execute(self, db: str, query: str, params: Dict[str, Any]):
         ex = self.executor_for_db(db or "neo4j")
-        return ex.execute(query, params)
+        if not self.log_queries:
+            return ex.execute(query, params)
+        import time as _t
+        t0 = _t.perf_counter()
+        try:
+            r = ex.execute(query, params)
+        except Exception as e:
+            print(f"[query] db={db or 'neo4j'} FAILED "
+                  f"({(_t.perf_counter() - t0) * 1e3:.1f} ms): "
+                  f"{query!r} err={e}", flush=True)
+            raise
+        print(f"[query] db={db or 'neo4j'} "
+              f"{(_t.perf_counter() - t0) * 1e3:.1f} ms rows={len(r.rows)}: "
+              f"{query!r}", flush=True)
+        return r
 
     async def start(self):
         self._server = await asyncio.start_server(

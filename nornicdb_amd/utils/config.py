@@ -44,6 +44,8 @@ class Config:
     hnsw_ef_search: int = 100
     brute_force_max: int = 5000
     kmeans_min: int = 100000
+    # observability
+    log_queries: bool = False   # reference --log-queries / bolt LogQueries
     # auth
     auth_enabled: bool = False
     initial_admin_password: str = ""

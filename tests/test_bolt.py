@@ -300,3 +300,18 @@ def test_explicit_tx_commit_bookmark():
         finally:
             srv.close()
     asyncio.run(run())
+
+
+def test_log_queries_stdout(capsys):
+    """--log-queries / cfg.log_queries: every executed query is logged
+    with duration and row count (reference pkg/bolt/server.go LogQueries)."""
+    from nornicdb_amd.bolt.server import BoltServer
+    from nornicdb_amd.cypher import Executor
+    from nornicdb_amd.storage.memory import MemoryEngine
+
+    ex = Executor(MemoryEngine())
+    srv = BoltServer(lambda db: ex, log_queries=True)
+    r = srv.execute("neo4j", "RETURN 1 AS one", {})
+    assert r.rows == [[1]]
+    out = capsys.readouterr().out
+    assert "[query]" in out and "RETURN 1 AS one" in out and "rows=1" in out
