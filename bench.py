@@ -289,7 +289,14 @@ def main():
     # pipeline just has one batch of latency (steady-state serving shape).
     pipelined = device.type == "cuda" and not args.no_pipeline
     if pipelined:
-        s_embed = torch.cuda.Stream()
+        # NOTE (measured, MI355X/ROCm 7.2): the two phases do NOT overlap
+        # in practice — step == embed + search exactly, with or without
+        # stream priority, and capping the search grid to free CUs only
+        # slows the search without admitting encoder workgroups. Both
+        # phases are CU-limited, so there is no free capacity to overlap
+        # into; the pipeline is kept for its serving shape (1-batch
+        # latency) and for hardware where dispatch interleaves.
+        s_embed = torch.cuda.Stream(priority=-1)
         s_search = torch.cuda.Stream()
 
         def pipe_step(prev_q):
