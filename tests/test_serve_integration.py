@@ -117,3 +117,74 @@ def test_serve_boots_disk_engine_and_answers(tmp_path):
     r = mgr.get("neo4j").execute_cypher("MATCH (b:Boot) RETURN count(b)")
     assert r.rows == [[1]]
     mgr.close()
+
+
+@pytest.mark.timeout(120)
+def test_serve_auth_login_and_protected_routes(tmp_path):
+    """--auth: boot prints the initial admin password; login yields a
+    token; protected routes 401 without it and work with it — all over
+    the fasthttp server."""
+    bolt_port, http_port = _free_ports(2)
+    env = dict(os.environ, PYTHONPATH=REPO,
+               NORNICDB_INITIAL_ADMIN_PASSWORD="s3cret-pw")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "nornicdb_amd", "serve", "--auth",
+         "--data-dir", str(tmp_path / "data"),
+         "--bolt-port", str(bolt_port), "--http-port", str(http_port)],
+        cwd=REPO, env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        deadline = time.time() + 60
+        up = False
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{http_port}/health", timeout=1) as r:
+                    if r.status == 200:
+                        up = True
+                        break
+            except Exception:
+                time.sleep(0.2)
+        assert up, proc.stdout.read(4000)
+
+        base = f"http://127.0.0.1:{http_port}"
+        # protected route without a token -> 401
+        req = urllib.request.Request(base + "/admin/stats")
+        try:
+            urllib.request.urlopen(req, timeout=5)
+            assert False, "expected 401"
+        except urllib.error.HTTPError as e:
+            assert e.code == 401
+
+        # login
+        body = json.dumps({"username": "neo4j",
+                           "password": "s3cret-pw"}).encode()
+        req = urllib.request.Request(
+            base + "/auth/login", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=5) as r:
+            out = json.loads(r.read())
+        token = out.get("token") or out.get("access_token")
+        assert token, out
+
+        hdr = {"Authorization": f"Bearer {token}",
+               "Content-Type": "application/json"}
+        req = urllib.request.Request(base + "/auth/me", headers=hdr)
+        with urllib.request.urlopen(req, timeout=5) as r:
+            me = json.loads(r.read())
+        assert me.get("username") == "neo4j" or me.get("user"), me
+
+        # authenticated tx round-trip (full FastAPI path: the fast path
+        # is disabled when auth is on)
+        body = json.dumps({"statements": [
+            {"statement": "RETURN 42 AS v"}]}).encode()
+        req = urllib.request.Request(
+            base + "/db/neo4j/tx/commit", data=body, headers=hdr)
+        with urllib.request.urlopen(req, timeout=5) as r:
+            out = json.loads(r.read())
+        assert out["results"][0]["data"][0]["row"] == [42]
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
