@@ -74,6 +74,32 @@ class DecayManager:
         self.engine.update_node(node)
         return imp
 
+    def start(self, interval_s: float = 3600.0) -> None:
+        """Background recalculation ticker (reference pkg/decay
+        decay.go:643 Manager.Start, RecalculateInterval default 1h).
+        Errors are swallowed per cycle; call stop() before close."""
+        import threading
+        if getattr(self, "_bg", None) is not None:
+            return
+        self._stop = threading.Event()
+
+        def loop():
+            while not self._stop.wait(interval_s):
+                try:
+                    self.run_cycle()
+                except Exception:
+                    pass   # keep ticking (reference: log and continue)
+
+        self._bg = threading.Thread(target=loop, daemon=True,
+                                    name="decay-recalc")
+        self._bg.start()
+
+    def stop(self) -> None:
+        if getattr(self, "_bg", None) is not None:
+            self._stop.set()
+            self._bg.join(timeout=5)
+            self._bg = None
+
     def run_cycle(self) -> Dict[str, int]:
         """Score all Memory nodes; archive / delete below thresholds."""
         now = self.now()

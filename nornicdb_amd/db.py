@@ -169,6 +169,15 @@ class NornicDB:
         self.inference = InferenceEngine(
             engine, search_service=self.search, tracker=self.tracker,
             config=InferenceConfig(evidence_required=2))
+        # background decay recalculation (reference pkg/decay Manager.Start,
+        # hourly ticker) — enabled by the edge_decay feature flag or
+        # NORNICDB_DECAY_INTERVAL_S; stopped by close()
+        from .cognitive import DecayManager
+        self.decay = DecayManager(engine)
+        import os as _os
+        _di = _os.environ.get("NORNICDB_DECAY_INTERVAL_S")
+        if _di:
+            self.decay.start(float(_di))
 
     # ---- cypher ----
     # queries containing these are never served from the result cache;
@@ -312,6 +321,7 @@ class NornicDB:
                       last_accessed=p.get("last_accessed", 0.0))
 
     def close(self):
+        self.decay.stop()
         self.embed_queue.stop()
         self.engine.flush()
 

@@ -585,3 +585,30 @@ class TestNodeConfig:
         assert not ok and "max capacity" in why
         ok, _ = store.is_edge_allowed("x", "z", "OTHER", 1.0, 0.0)
         assert ok
+
+
+def test_decay_background_ticker():
+    """DecayManager.start/stop (reference pkg/decay Manager.Start): the
+    ticker runs cycles at the interval and stop() joins cleanly."""
+    import time
+
+    from nornicdb_amd.cognitive import DecayManager
+    from nornicdb_amd.storage.memory import MemoryEngine
+    from nornicdb_amd.storage.types import Node
+
+    eng = MemoryEngine()
+    eng.create_node(Node(id="m1", labels=["Memory"],
+                         properties={"importance": 0.9,
+                                     "created_at": time.time(),
+                                     "last_accessed": time.time()}))
+    dm = DecayManager(eng)
+    calls = []
+    orig = dm.run_cycle
+    dm.run_cycle = lambda: calls.append(orig())
+    dm.start(interval_s=0.05)
+    time.sleep(0.3)
+    dm.stop()
+    assert len(calls) >= 2
+    n = len(calls)
+    time.sleep(0.15)
+    assert len(calls) == n  # stopped for real
