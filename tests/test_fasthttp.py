@@ -152,3 +152,44 @@ def test_graphql_roundtrip(server):
     assert st == 200
     out = json.loads(body)
     assert out["data"]["nodes"] and out["data"]["nodes"][0]["labels"] == ["Person"]
+
+
+def test_https_serving(tmp_path):
+    """fasthttp over TLS: self-signed pair + ssl context, GET /health."""
+    import ssl
+    import urllib.request
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+    from nornicdb_amd.server import create_app
+    from nornicdb_amd.server.fasthttp import start_http_server
+    from nornicdb_amd.utils.tls import ensure_self_signed, make_ssl_context
+
+    cert, key = ensure_self_signed(str(tmp_path))
+    sctx = make_ssl_context(cert, key)
+    mgr = open_db(embedder=MockEmbedder(8), dims=8)
+    app = create_app(mgr, auth=None)
+    ports = {}
+    loops = {}
+
+    def run():
+        loop = asyncio.new_event_loop()
+        asyncio.set_event_loop(loop)
+        loops["loop"] = loop
+        srv = loop.run_until_complete(
+            start_http_server(app, "127.0.0.1", 0, ssl_context=sctx))
+        ports["p"] = srv.sockets[0].getsockname()[1]
+        loop.run_forever()
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    t0 = time.time()
+    while "p" not in ports and time.time() - t0 < 10:
+        time.sleep(0.02)
+    cctx = ssl.create_default_context()
+    cctx.check_hostname = False
+    cctx.verify_mode = ssl.CERT_NONE
+    with urllib.request.urlopen(f"https://127.0.0.1:{ports['p']}/health",
+                                context=cctx, timeout=5) as r:
+        assert r.status == 200 and b"ok" in r.read()
+    loops["loop"].call_soon_threadsafe(loops["loop"].stop)
