@@ -322,12 +322,13 @@ __device__ void run_layers(const DecodeArgs& a, cg::grid_group& grid,
       const float scale = rsqrtf((float)HD);
       float mx = -1e30f;
       // scores: lane handles t = lane, lane+64, ...
-      // static LDS: 8 waves x 2048 positions = 64 KB on top of the
-      // ~20 KB dynamic r_lds — gfx950 allows >64 KB per WG for static
-      // allocations, while 64 KB DYNAMIC makes the cooperative launch
-      // fail with invalid argument (measured)
-      static __shared__ float s_scores[(DWG / WAVE) * 2048];
-      float* sc = s_scores + wid * 2048;   // per-wave region (<=2048 pos)
+      // static LDS: 8 waves x 4096 positions = 128 KB on top of the
+      // ~20 KB dynamic r_lds — gfx950 allows up to 160 KB per WG for
+      // STATIC allocations (128 KB measured fine in the GEMM kernels),
+      // while >=64 KB DYNAMIC makes the cooperative launch fail with
+      // invalid argument (measured). 4096 = the Qwen2 max_position.
+      static __shared__ float s_scores[(DWG / WAVE) * 4096];
+      float* sc = s_scores + wid * 4096;   // per-wave region (<=4096 pos)
       for (int t = lane; t < T; t += WAVE) {
         const unsigned short* kr = K + (long long)t * HD;
         float s = 0.f;
@@ -699,7 +700,7 @@ void decode_step(at::Tensor layer_ptrs,  // [n_layers, 14] int64 (LayerPtrs)
   TORCH_CHECK(x.scalar_type() == at::kFloat && x.is_contiguous());
   TORCH_CHECK(rope_cos.scalar_type() == at::kFloat &&
               rope_cos.is_contiguous());
-  TORCH_CHECK(max_len <= 2048, "fused decode supports max_len <= 2048");
+  TORCH_CHECK(max_len <= 4096, "fused decode supports max_len <= 4096");
   TORCH_CHECK(hd == 64 && hidden % 2 == 0);
 
   DecodeArgs a;
@@ -720,7 +721,7 @@ void decode_step(at::Tensor layer_ptrs,  // [n_layers, 14] int64 (LayerPtrs)
   a.h = h.data_ptr<float>();
   a.pos = (int)pos;
 
-  // LDS: max(hidden, inter, 4 waves * 2048 scores) floats
+  // LDS (dynamic): max(hidden, inter) floats; scores are static
   size_t lds = sizeof(float) *
       std::max<long long>(hidden, inter);
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -761,7 +762,7 @@ void decode_tokens(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
   TORCH_CHECK(out.scalar_type() == at::kInt && out.is_contiguous() &&
               out.numel() >= n_toks);
   TORCH_CHECK(n_done.scalar_type() == at::kInt && n_done.numel() >= 1);
-  TORCH_CHECK(max_len <= 2048, "fused decode supports max_len <= 2048");
+  TORCH_CHECK(max_len <= 4096, "fused decode supports max_len <= 4096");
   TORCH_CHECK(hd == 64 && hidden % 2 == 0);
 
   DecodeTokArgs t;

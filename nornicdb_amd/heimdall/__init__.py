@@ -67,7 +67,7 @@ class HeimdallManager:
                 from ..models.heimdall import FusedDecoder
                 self._graphed = FusedDecoder(self.model,
                                              max_len=min(cfg.max_position,
-                                                         2048))
+                                                         4096))
             except Exception:
                 try:
                     from ..models.heimdall import GraphedDecoder

@@ -303,10 +303,10 @@ class FusedDecoder:
     (grid-wide syncs between GEMV/attention stages); only the final norm,
     lm_head GEMV and sampling remain as torch ops (~4 launches/token vs
     ~180 for the eager loop). Falls back is handled by HeimdallManager.
-    Requires head_dim 64 and max_len <= 2048.
+    Requires head_dim 64 and max_len <= 4096.
     """
 
-    def __init__(self, model: HeimdallModel, max_len: int = 2048):
+    def __init__(self, model: HeimdallModel, max_len: int = 4096):
         from ..ops import require_native
         self.nat = require_native()
         self.m = model
@@ -317,7 +317,7 @@ class FusedDecoder:
         self.dev = dev
         self.hd = c.hidden_size // c.num_heads
         assert self.hd == 64, "fused decode supports head_dim 64"
-        self.max_len = min(max_len, c.max_position, 2048)
+        self.max_len = min(max_len, c.max_position, 4096)
         dt = next(model.parameters()).dtype
         assert dt == torch.bfloat16, "fused decode expects bf16 weights"
         self.cache_k = [torch.zeros(c.num_kv_heads, self.max_len, self.hd,

@@ -264,3 +264,26 @@ class TestFusedDecoderFullSize:
         graphed = gd.generate(prompt.clone(), max_new_tokens=6,
                               temperature=0.0)
         assert fused == graphed
+
+
+@pytest.mark.gpu
+class TestFusedDecoderLongContext:
+    def test_decode_beyond_2048(self):
+        """Fused decode past the old 2048-position cap (score LDS now
+        covers the full 4096 max_position): prefill 2100 tokens, then
+        fused greedy must match the eager model exactly."""
+        import torch
+
+        from nornicdb_amd.models.heimdall import (FusedDecoder,
+                                                  HeimdallConfig,
+                                                  HeimdallModel)
+        torch.manual_seed(3)
+        cfg = HeimdallConfig(num_layers=2, max_position=4096)
+        m = HeimdallModel(cfg).init_small().to("cuda", torch.bfloat16).eval()
+        prompt = torch.randint(0, cfg.vocab_size, (1, 2100), device="cuda")
+        fd = FusedDecoder(m, max_len=4096)
+        assert fd.max_len == 4096
+        fused = fd.generate(prompt.clone(), max_new_tokens=6,
+                            temperature=0.0)
+        eager = m.generate(prompt.clone(), max_new_tokens=6, temperature=0)
+        assert fused == eager, (fused, eager)
