@@ -29,7 +29,7 @@ class _Disconnect(Exception):
 
 class AsgiHttpProtocol(asyncio.Protocol):
     __slots__ = ("app", "transport", "buf", "task", "peer", "is_tls",
-                 "_closed", "_wake")
+                 "_closed", "_wake", "_can_write")
 
     def __init__(self, app, is_tls: bool = False):
         self.app = app
@@ -52,11 +52,22 @@ class AsgiHttpProtocol(asyncio.Protocol):
             pass
         self.peer = transport.get_extra_info("peername") or ("", 0)
         self._wake = asyncio.Event()
+        self._can_write = asyncio.Event()
+        self._can_write.set()
         self.task = asyncio.get_event_loop().create_task(self._serve())
 
     def data_received(self, data):
         self.buf += data
         self._wake.set()
+
+    # asyncio flow control: the loop calls these when the socket send
+    # buffer fills — streaming responses to slow clients block in
+    # _drain instead of growing the transport buffer unboundedly
+    def pause_writing(self):
+        self._can_write.clear()
+
+    def resume_writing(self):
+        self._can_write.set()
 
     def eof_received(self):
         self._closed = True
@@ -140,6 +151,8 @@ class AsgiHttpProtocol(asyncio.Protocol):
                 chunked = True
             elif k == b"connection":
                 connection = v.lower()
+            elif k == b"expect" and v.lower() == b"100-continue":
+                self.transport.write(b"HTTP/1.1 100 Continue\r\n\r\n")
         if b"?" in target:
             raw_path, _, qs = target.partition(b"?")
         else:
@@ -214,9 +227,10 @@ class AsgiHttpProtocol(asyncio.Protocol):
                         w(chunk)
                 if self.transport.is_closing():
                     raise _Disconnect
-                # give the loop a chance to flush large streams
-                if msg.get("more_body"):
-                    await _drain(self.transport)
+                # flow control: block while the socket send buffer is
+                # over the high-water mark (pause_writing fired)
+                if not self._can_write.is_set():
+                    await self._can_write.wait()
 
         try:
             await self.app(scope, receive, send)
@@ -254,12 +268,6 @@ class AsgiHttpProtocol(asyncio.Protocol):
             if n == 0:
                 break
         return bytes(out)
-
-
-async def _drain(transport):
-    # cooperative backpressure: yield to the loop; asyncio pauses the
-    # protocol via write buffer limits
-    await asyncio.sleep(0)
 
 
 _REASON = {
