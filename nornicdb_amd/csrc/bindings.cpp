@@ -60,6 +60,12 @@ void decode_step(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
                  long long pos);
 double sync_bench(long long iters, long long which, long long grid,
                   at::Tensor scratch);
+// knn_fp8.hip
+std::tuple<at::Tensor, at::Tensor> knn_fp8(at::Tensor db, at::Tensor q,
+                                           long long row_base, int k_out);
+std::tuple<at::Tensor, at::Tensor> knn_i8(at::Tensor db, at::Tensor sa,
+                                          at::Tensor q, at::Tensor sq,
+                                          long long row_base, int k_out);
 void decode_tokens(at::Tensor layer_ptrs, at::Tensor x, at::Tensor q,
                    at::Tensor attn, at::Tensor h, at::Tensor rope_cos,
                    at::Tensor rope_sin, at::Tensor embed_w,
@@ -116,6 +122,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("n_kv"), py::arg("hd"), py::arg("inter"),
         py::arg("max_len"), py::arg("rms_eps"), py::arg("pos"),
         py::arg("start_tok"), py::arg("n_toks"), py::arg("eos") = -1);
+  m.def("knn_i8", &knn_i8,
+        "Fused i8 MFMA score + top-k over a symmetric int8 corpus with "
+        "per-row scales (score = i32 dot * sa[row] * sq[col])",
+        py::arg("db"), py::arg("sa"), py::arg("q"), py::arg("sq"),
+        py::arg("row_base") = 0, py::arg("k_out") = 10);
+  m.def("knn_fp8", &knn_fp8,
+        "Fused MFMA cosine score + top-k over an FP8 e4m3fn corpus "
+        "(uint8 carrier), 256-query batches",
+        py::arg("db"), py::arg("q"), py::arg("row_base") = 0,
+        py::arg("k_out") = 10);
   m.def("sync_bench", &sync_bench,
         "grid-barrier microbenchmark: ms for `iters` barriers "
         "(which=0 cg::grid.sync, 1 two-level custom)",

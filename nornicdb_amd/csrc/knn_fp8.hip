@@ -1,0 +1,482 @@
+// Fused MFMA score + top-k kNN over an FP8 (OCP e4m3fn) corpus.
+//
+// Opt-in quantized search mode: the corpus is stored at 1 B/element —
+// HALF the HBM/LDS traffic of bf16 and 2x the per-GPU capacity (200M+
+// 1024-d vectors in 288 GB) — scored with the gfx950 fp8 MFMA
+// (mfma_f32_16x16x32_fp8_fp8; non-scaled fp8 runs at the bf16 MFMA rate,
+// so the win is bandwidth, not math). Scores accumulate in fp32; recall
+// loss from e4m3 quantization is ~1% @ k=10 on normalized vectors
+// (measured in tests/test_ops_vector.py). The reference has no
+// quantized mode (pkg/gpu scores fp32 only) — this is MI355X-native
+// headroom, gated behind EmbeddingIndex(quant="fp8").
+//
+// Geometry mirrors csrc/knn_mfma.hip exactly at the BYTE level: a tile
+// row is 128 B = 128 fp8 K-elements (vs 64 bf16), so the staging loops,
+// row-XOR swizzle and epilogue are identical; only the MFMA loop reads
+// 8-byte i64 fragments and runs 4 k-steps of 32 per tile.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+#define G_AS __attribute__((address_space(1)))
+#define L_AS __attribute__((address_space(3)))
+
+#define F8_BM 96
+#define F8_BN 256
+#define F8_BK 128          // fp8 K-elements per tile (= 128 B rows)
+#define F8_NTHREADS 256
+#define F8_KCAND 10
+#define F8_MW (F8_BM / 16)
+
+__device__ __forceinline__ int swz8(int b) {
+  return (b & ~127) | ((b & 127) ^ (((b >> 7) & 7) << 4));
+}
+
+__global__ __launch_bounds__(F8_NTHREADS, 3) void k_knn_fp8(
+    const unsigned char* __restrict__ db, const unsigned char* __restrict__ qs,
+    long long n_panels, int d, long long row_base,
+    float* __restrict__ cand_score, int* __restrict__ cand_idx) {
+  // 44 KB: sA (12 KB) + sB (32 KB); epilogue aliases [256][36] fp32.
+  __shared__ __align__(16) char smem[F8_BM * F8_BK + F8_BN * F8_BK];
+  unsigned char* sA = (unsigned char*)smem;
+  unsigned char* sB = (unsigned char*)(smem + F8_BM * F8_BK);
+  float* sT = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[F8_KCAND];
+  int ti[F8_KCAND];
+#pragma unroll
+  for (int i = 0; i < F8_KCAND; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const long long ld = d;   // row stride in bytes (1 B/elem)
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * F8_BM;
+
+    float4v acc[F8_MW][4];
+#pragma unroll
+    for (int m = 0; m < F8_MW; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < d; kt += F8_BK) {
+      // ---- stage A (96x128 B = 12 KB) and B (256x128 B = 32 KB) ----
+      // identical byte pattern to the bf16 kernel: 1 KB per
+      // global_load_lds chunk, global source pre-swizzled
+#pragma unroll
+      for (int it = 0; it < F8_BM / 32; ++it) {
+        int chunk = wc * (F8_BM / 32) + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz8(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + (p >> 7)) * ld + (long long)kt +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {
+        int chunk = wc * 8 + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz8(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)(p >> 7) * ld + (long long)kt +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+      __syncthreads();
+
+      // ---- MFMA: 4 k-steps of 32 fp8 over the staged 128 B rows ----
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const int kb = ks * 32 + (lane >> 4) * 8;   // byte == elem offset
+        long bfr[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          bfr[nn] = *reinterpret_cast<const long*>(
+              (const char*)sB + swz8(c * 128 + kb));
+        }
+#pragma unroll
+        for (int m = 0; m < F8_MW; ++m) {
+          int r = m * 16 + (lane & 15);
+          long af = *reinterpret_cast<const long*>(
+              (const char*)sA + swz8(r * 128 + kb));
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: identical to the bf16 kernel ----
+#pragma unroll
+    for (int h = 0; h < F8_BM / 32; ++h) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int col = wc * 64 + nn * 16 + (lane & 15);
+          int srow = mi * 16 + (lane >> 4) * 4;
+          *reinterpret_cast<float4v*>(sT + col * 36 + srow) = acc[m][nn];
+        }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * 32;
+#pragma unroll
+      for (int j = 0; j < 32; j += 4) {
+        float4v v = *reinterpret_cast<const float4v*>(sT + tid * 36 + j);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          float s = v[e];
+          if (s > tv[F8_KCAND - 1]) {
+            float cs = s;
+            int ci = (int)(grow0 + j + e);
+#pragma unroll
+            for (int i = 0; i < F8_KCAND; ++i) {
+              bool ins = cs > tv[i];
+              float ts = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i];
+              ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * F8_BN + tid;
+#pragma unroll
+  for (int i = 0; i < F8_KCAND; ++i) {
+    cand_score[slot * F8_KCAND + i] = tv[i];
+    cand_idx[slot * F8_KCAND + i] = ti[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// INT8 symmetric-quantized variant (the recommended quant mode): per-row
+// absmax scales give ~7 effective bits (recall@10 ~0.98-0.99 even on
+// worst-case gaussian corpora, vs ~0.91 for e4m3), and the gfx950 i8
+// MFMA (mfma_i32_16x16x64_i8, K=64/instr) runs at 2x the bf16 rate.
+// score = i32_dot * sa[row] * sq[col], applied in the epilogue.
+// ---------------------------------------------------------------------------
+
+typedef int int4v_ __attribute__((ext_vector_type(4)));
+
+__global__ __launch_bounds__(F8_NTHREADS, 3) void k_knn_i8(
+    const signed char* __restrict__ db, const float* __restrict__ sa_g,
+    const signed char* __restrict__ qs, const float* __restrict__ sq_g,
+    long long n_panels, int d, long long row_base,
+    float* __restrict__ cand_score, int* __restrict__ cand_idx) {
+  __shared__ __align__(16) char smem[F8_BM * F8_BK + F8_BN * F8_BK];
+  signed char* sA = (signed char*)smem;
+  signed char* sB = (signed char*)(smem + F8_BM * F8_BK);
+  float* sT = (float*)smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wc = tid / WAVE;
+
+  float tv[F8_KCAND];
+  int ti[F8_KCAND];
+#pragma unroll
+  for (int i = 0; i < F8_KCAND; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+
+  const float sq_own = sq_g[tid];   // this thread's query column scale
+  const long long ld = d;
+
+  for (long long panel = blockIdx.x; panel < n_panels; panel += gridDim.x) {
+    const long long prow = panel * F8_BM;
+
+    int4v_ acc[F8_MW][4];
+#pragma unroll
+    for (int m = 0; m < F8_MW; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) acc[m][nn] = {0, 0, 0, 0};
+
+    for (int kt = 0; kt < d; kt += F8_BK) {
+#pragma unroll
+      for (int it = 0; it < F8_BM / 32; ++it) {
+        int chunk = wc * (F8_BM / 32) + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz8(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)db + (prow + (p >> 7)) * ld + (long long)kt +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sA + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {
+        int chunk = wc * 8 + it;
+        int x = chunk * 1024 + lane * 16;
+        int p = swz8(x);
+        const G_AS unsigned int* gp = (const G_AS unsigned int*)(
+            (const char*)qs + (long long)(p >> 7) * ld + (long long)kt +
+            (p & 127));
+        L_AS unsigned int* lp = (L_AS unsigned int*)((char*)sB + chunk * 1024);
+        __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+      }
+      __syncthreads();
+
+      // ---- i8 MFMA: 2 k-steps of 64 over the staged 128 B rows ----
+      // fragment = 16 B per lane, the same 16 B-slot pattern the
+      // row-XOR swizzle was built for
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kb = ks * 64 + (lane >> 4) * 16;
+        int4v_ bfr[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int c = wc * 64 + nn * 16 + (lane & 15);
+          bfr[nn] = *reinterpret_cast<const int4v_*>(
+              (const char*)sB + swz8(c * 128 + kb));
+        }
+#pragma unroll
+        for (int m = 0; m < F8_MW; ++m) {
+          int r = m * 16 + (lane & 15);
+          int4v_ af = *reinterpret_cast<const int4v_*>(
+              (const char*)sA + swz8(r * 128 + kb));
+#pragma unroll
+          for (int nn = 0; nn < 4; ++nn)
+            acc[m][nn] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                af, bfr[nn], acc[m][nn], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: i32 accums through the same transposed LDS block,
+    // scaled to float on the way into the top-k lists ----
+#pragma unroll
+    for (int h = 0; h < F8_BM / 32; ++h) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        int m = h * 2 + mi;
+#pragma unroll
+        for (int nn = 0; nn < 4; ++nn) {
+          int col = wc * 64 + nn * 16 + (lane & 15);
+          int srow = mi * 16 + (lane >> 4) * 4;
+          *reinterpret_cast<int4v_*>((int*)sT + col * 36 + srow) = acc[m][nn];
+        }
+      }
+      __syncthreads();
+      const long long grow0 = prow + (long long)h * 32;
+#pragma unroll
+      for (int j = 0; j < 32; j += 4) {
+        int4v_ v = *reinterpret_cast<const int4v_*>((const int*)sT + tid * 36 + j);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          long long row = grow0 + j + e;
+          float s = (float)v[e] * sa_g[row] * sq_own;
+          if (s > tv[F8_KCAND - 1]) {
+            float cs = s;
+            int ci = (int)row;
+#pragma unroll
+            for (int i = 0; i < F8_KCAND; ++i) {
+              bool ins = cs > tv[i];
+              float ts = tv[i]; int tj = ti[i];
+              tv[i] = ins ? cs : tv[i];
+              ti[i] = ins ? ci : ti[i];
+              cs = ins ? ts : cs; ci = ins ? tj : ci;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  long long slot = (long long)blockIdx.x * F8_BN + tid;
+#pragma unroll
+  for (int i = 0; i < F8_KCAND; ++i) {
+    cand_score[slot * F8_KCAND + i] = tv[i];
+    cand_idx[slot * F8_KCAND + i] = ti[i];
+  }
+}
+
+// identical merge to knn_mfma.hip's template (weak/implicit instantiation
+// there; re-declared here as its own symbol to avoid cross-TU device links)
+template <int K>
+__global__ void k_topk_merge_f8(const float* __restrict__ cand_score,
+                                const int* __restrict__ cand_idx,
+                                long long w, int q_stride, int k_out,
+                                long long row_base,
+                                float* __restrict__ out_score,
+                                long long* __restrict__ out_idx) {
+  const int q = blockIdx.x;
+  const long long total = w * K;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  __shared__ unsigned long long s_best[8];
+
+  float tv[K];
+  int ti[K];
+#pragma unroll
+  for (int i = 0; i < K; ++i) { tv[i] = -1e30f; ti[i] = -1; }
+  for (long long j = threadIdx.x; j < total; j += blockDim.x) {
+    long long src = ((j / K) * q_stride + q) * K + (j % K);
+    float s = cand_score[src];
+    if (s > tv[K - 1]) {
+      int ci = cand_idx[src];
+      float cs = s;
+#pragma unroll
+      for (int i = 0; i < K; ++i) {
+        bool ins = cs > tv[i];
+        float ts = tv[i]; int tj = ti[i];
+        tv[i] = ins ? cs : tv[i];
+        ti[i] = ins ? ci : ti[i];
+        cs = ins ? ts : cs; ci = ins ? tj : ci;
+      }
+    }
+  }
+
+  unsigned int mono;
+  int head = 0;
+  for (int r = 0; r < k_out; ++r) {
+    float hv = -1e30f;
+#pragma unroll
+    for (int i = 0; i < K; ++i)
+      if (i == head) hv = tv[i];
+    if (head >= K) hv = -1e30f;
+    {
+      unsigned int u = __float_as_uint(hv);
+      mono = (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+    }
+    unsigned long long packed =
+        ((unsigned long long)mono << 32) | (unsigned int)threadIdx.x;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      unsigned long long o = __shfl_xor(packed, off, WAVE);
+      if (o > packed) packed = o;
+    }
+    if (lane == 0) s_best[wid] = packed;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned long long b = s_best[0];
+      for (int i = 1; i < (int)(blockDim.x / WAVE); ++i)
+        if (s_best[i] > b) b = s_best[i];
+      s_best[0] = b;
+    }
+    __syncthreads();
+    int winner = (int)(s_best[0] & 0xffffffffu);
+    if (threadIdx.x == winner) {
+      float wv = -1e30f; int wi = -1;
+#pragma unroll
+      for (int i = 0; i < K; ++i)
+        if (i == head) { wv = tv[i]; wi = ti[i]; }
+      out_score[(long long)q * k_out + r] = wv;
+      out_idx[(long long)q * k_out + r] = (wi < 0) ? -1 : row_base + wi;
+      head++;
+    }
+    __syncthreads();
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor> knn_i8(at::Tensor db, at::Tensor sa,
+                                          at::Tensor q, at::Tensor sq,
+                                          long long row_base, int k_out) {
+  TORCH_CHECK(db.is_cuda() && db.dim() == 2 && db.is_contiguous() &&
+                  db.scalar_type() == at::kChar,
+              "knn_i8: db must be contiguous 2D int8 CUDA");
+  TORCH_CHECK(q.is_cuda() && q.dim() == 2 && q.is_contiguous() &&
+                  q.scalar_type() == at::kChar,
+              "knn_i8: q must be contiguous 2D int8 CUDA");
+  long long n = db.size(0);
+  int d = (int)db.size(1);
+  TORCH_CHECK(sa.scalar_type() == at::kFloat && sa.is_contiguous() &&
+              sa.numel() == n, "knn_i8: sa must be fp32 [N]");
+  TORCH_CHECK(sq.scalar_type() == at::kFloat && sq.is_contiguous() &&
+              sq.numel() == F8_BN, "knn_i8: sq must be fp32 [256]");
+  TORCH_CHECK(q.size(0) == F8_BN, "knn_i8: q must be padded to ", F8_BN);
+  TORCH_CHECK(q.size(1) == d, "dim mismatch");
+  TORCH_CHECK(d % F8_BK == 0, "knn_i8 needs D % 128 == 0");
+  TORCH_CHECK(n % F8_BM == 0, "knn_i8 needs N % ", F8_BM, " == 0");
+  TORCH_CHECK(n < (1LL << 31), "shard too large for int32 local rows");
+  TORCH_CHECK(k_out >= 1 && k_out <= F8_KCAND);
+
+  long long n_panels = n / F8_BM;
+  int max_grid = 7680;
+  if (const char* g = getenv("NORNICDB_KNN_GRID")) max_grid = atoi(g);
+  int grid = (int)std::min<long long>(n_panels, max_grid);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+
+  auto opts_f = db.options().dtype(at::kFloat);
+  at::Tensor cand_s = at::empty({grid, F8_BN, F8_KCAND}, opts_f);
+  at::Tensor cand_i = at::empty({grid, F8_BN, F8_KCAND},
+                                db.options().dtype(at::kInt));
+
+  hipLaunchKernelGGL(k_knn_i8, dim3(grid), dim3(F8_NTHREADS), 0, stream,
+                     (const signed char*)db.data_ptr(),
+                     sa.data_ptr<float>(),
+                     (const signed char*)q.data_ptr(),
+                     sq.data_ptr<float>(), n_panels, d, row_base,
+                     cand_s.data_ptr<float>(), cand_i.data_ptr<int>());
+  HIP_CHECK_LAST();
+
+  at::Tensor out_s = at::empty({F8_BN, k_out}, opts_f);
+  at::Tensor out_i = at::empty({F8_BN, k_out},
+                               db.options().dtype(at::kLong));
+  hipLaunchKernelGGL((k_topk_merge_f8<F8_KCAND>), dim3(F8_BN), dim3(256), 0,
+                     stream, cand_s.data_ptr<float>(), cand_i.data_ptr<int>(),
+                     grid, F8_BN, k_out, row_base, out_s.data_ptr<float>(),
+                     reinterpret_cast<long long*>(out_i.data_ptr<int64_t>()));
+  HIP_CHECK_LAST();
+  return {out_s, out_i};
+}
+
+std::tuple<at::Tensor, at::Tensor> knn_fp8(at::Tensor db, at::Tensor q,
+                                           long long row_base, int k_out) {
+  TORCH_CHECK(db.is_cuda() && db.dim() == 2 && db.is_contiguous() &&
+                  db.scalar_type() == at::kByte,
+              "knn_fp8: db must be contiguous 2D uint8 (e4m3fn bits) CUDA");
+  TORCH_CHECK(q.is_cuda() && q.dim() == 2 && q.is_contiguous() &&
+                  q.scalar_type() == at::kByte,
+              "knn_fp8: q must be contiguous 2D uint8 (e4m3fn bits) CUDA");
+  long long n = db.size(0);
+  int d = (int)db.size(1);
+  TORCH_CHECK(q.size(0) == F8_BN, "knn_fp8: q must be padded to ", F8_BN);
+  TORCH_CHECK(q.size(1) == d, "dim mismatch");
+  TORCH_CHECK(d % F8_BK == 0, "knn_fp8 needs D % 128 == 0");
+  TORCH_CHECK(n % F8_BM == 0, "knn_fp8 needs N % ", F8_BM, " == 0");
+  TORCH_CHECK(n < (1LL << 31), "shard too large for int32 local rows");
+  TORCH_CHECK(k_out >= 1 && k_out <= F8_KCAND);
+
+  long long n_panels = n / F8_BM;
+  int max_grid = 7680;
+  if (const char* g = getenv("NORNICDB_KNN_GRID")) max_grid = atoi(g);
+  int grid = (int)std::min<long long>(n_panels, max_grid);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+
+  auto opts_f = db.options().dtype(at::kFloat);
+  at::Tensor cand_s = at::empty({grid, F8_BN, F8_KCAND}, opts_f);
+  at::Tensor cand_i = at::empty({grid, F8_BN, F8_KCAND},
+                                db.options().dtype(at::kInt));
+
+  hipLaunchKernelGGL(k_knn_fp8, dim3(grid), dim3(F8_NTHREADS), 0, stream,
+                     (const unsigned char*)db.data_ptr(),
+                     (const unsigned char*)q.data_ptr(), n_panels, d,
+                     row_base, cand_s.data_ptr<float>(),
+                     cand_i.data_ptr<int>());
+  HIP_CHECK_LAST();
+
+  at::Tensor out_s = at::empty({F8_BN, k_out}, opts_f);
+  at::Tensor out_i = at::empty({F8_BN, k_out},
+                               db.options().dtype(at::kLong));
+  hipLaunchKernelGGL((k_topk_merge_f8<F8_KCAND>), dim3(F8_BN), dim3(256), 0,
+                     stream, cand_s.data_ptr<float>(), cand_i.data_ptr<int>(),
+                     grid, F8_BN, k_out, row_base, out_s.data_ptr<float>(),
+                     reinterpret_cast<long long*>(out_i.data_ptr<int64_t>()));
+  HIP_CHECK_LAST();
+  return {out_s, out_i};
+}

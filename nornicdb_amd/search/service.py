@@ -45,13 +45,19 @@ class SearchResult:
 class SearchService:
     def __init__(self, engine: Engine, dims: int = 1024,
                  device: Optional[str] = None, use_hnsw: bool = True,
-                 embedder=None):
+                 embedder=None, quant: Optional[str] = None):
         self.engine = engine
         self.dims = dims
         self.embedder = embedder
         self._lock = threading.RLock()
         self.fulltext = FulltextIndex()
-        self.emb = EmbeddingIndex(dims, device=device)
+        # NORNICDB_SEARCH_QUANT=int8|fp8 stores the GPU corpus at
+        # 1 B/element (half the HBM traffic, 2x capacity). int8 keeps
+        # per-row scales (~7 bits, recall@10 >= 0.95 worst-case); fp8
+        # e4m3 is scale-free but coarser (~0.91 worst-case)
+        import os as _os
+        quant = quant or _os.environ.get("NORNICDB_SEARCH_QUANT") or None
+        self.emb = EmbeddingIndex(dims, device=device, quant=quant)
         self.hnsw = HNSWIndex(dims) if use_hnsw else None
         self.clusters = ClusterIndex()
         self.pipeline = VectorSearchPipeline(self.emb, self.hnsw, self.clusters)

@@ -152,3 +152,130 @@ def test_knn_mfma_large_query_batch():
     s_ref, i_ref = knn_search_exact(dbb[:, :].float(), q[:64].float(), 10)
     match = (i[:64] == i_ref).float().mean().item()
     assert match > 0.9, match
+
+
+def test_fp8_quantize_cpu_roundtrip():
+    """e4m3fn quantization error on normalized vectors is small and the
+    CPU (dequantized-exact) search path handles fp8 corpora."""
+    from nornicdb_amd.ops.knn import knn_search, quantize_fp8
+    torch.manual_seed(0)
+    db = torch.nn.functional.normalize(torch.randn(200, 128), dim=-1)
+    db8 = quantize_fp8(db)
+    err = (db8.float() - db).abs().max()
+    assert float(err) < 0.05, err
+    q = db[:4]
+    s, i = knn_search(db8, q, 5)
+    assert (i[:, 0] == torch.arange(4)).all()
+
+
+@pytest.mark.gpu
+class TestKnnFp8GPU:
+    def test_scores_match_dequantized_oracle(self):
+        """Kernel fp8 x fp8 fp32-accum scores == torch fp32 matmul over
+        the SAME e4m3 values (both sides see identical inputs)."""
+        from nornicdb_amd.ops import require_native
+        from nornicdb_amd.ops.knn import quantize_fp8
+        nat = require_native()
+        torch.manual_seed(1)
+        db = torch.nn.functional.normalize(
+            torch.randn(96 * 13, 1024, device="cuda"), dim=-1)
+        db8 = quantize_fp8(db)
+        q = db[:17]
+        q8 = q.to(torch.float8_e4m3fn)
+        qpad = torch.cat([q8, torch.zeros(256 - 17, 1024, device="cuda",
+                                          dtype=torch.float8_e4m3fn)])
+        s, i = nat.knn_fp8(db8.view(torch.uint8),
+                           qpad.contiguous().view(torch.uint8), 0, 10)
+        ref = (q8.float() @ db8.float().T)
+        rs, ri = torch.topk(ref, 10, dim=-1)
+        assert torch.allclose(s[:17], rs, atol=1e-3), (s[:17] - rs).abs().max()
+        # ties can permute equal-score indices; top-1 must agree
+        assert (i[:17, 0] == ri[:, 0]).all()
+
+    def test_recall_vs_fp32_exact(self):
+        """fp8 e4m3 recall@10 vs fp32 exact on WORST-CASE gaussian
+        vectors (near-uniform score gaps): measured ~0.91. Real embedding
+        corpora (clustered) fare much better; int8 mode (below) holds
+        >= 0.95 even here."""
+        from nornicdb_amd.ops.knn import knn_search, knn_search_exact, quantize_fp8
+        torch.manual_seed(2)
+        db = torch.nn.functional.normalize(
+            torch.randn(96 * 500, 1024, device="cuda"), dim=-1)
+        q = torch.nn.functional.normalize(
+            torch.randn(64, 1024, device="cuda"), dim=-1)
+        es, ei = knn_search_exact(db, q, 10)
+        s, i = knn_search(quantize_fp8(db), q, 10)
+        hit = 0
+        for r in range(64):
+            hit += len(set(i[r].tolist()) & set(ei[r].tolist()))
+        recall = hit / (64 * 10)
+        assert recall >= 0.85, recall
+        # self-match: query = db row must return itself top-1
+        s2, i2 = knn_search(quantize_fp8(db), db[:32], 10)
+        assert (i2[:, 0] == torch.arange(32, device="cuda")).float().mean() > 0.95
+
+    def test_embedding_index_fp8_gpu(self):
+        import numpy as np
+
+        from nornicdb_amd.search.embedding_index import EmbeddingIndex
+        idx = EmbeddingIndex(128, device="cuda", quant="fp8")
+        rng = np.random.default_rng(3)
+        vecs = rng.normal(size=(500, 128)).astype("float32")
+        vecs /= np.linalg.norm(vecs, axis=1, keepdims=True)
+        idx.add_batch([f"n{i}" for i in range(500)], vecs)
+        hits = idx.search(vecs[42], k=5)
+        assert hits[0][0] == "n42" and hits[0][1] > 0.97
+
+
+@pytest.mark.gpu
+class TestKnnInt8GPU:
+    def test_scores_match_dequantized_oracle(self):
+        """i8 kernel (i32 dot * sa * sq) == fp32 matmul over the same
+        dequantized values — i32 dots are exact, so agreement is to
+        fp32 rounding."""
+        from nornicdb_amd.ops import require_native
+        from nornicdb_amd.ops.knn import knn_search_int8, quantize_int8
+        torch.manual_seed(4)
+        db = torch.nn.functional.normalize(
+            torch.randn(96 * 13, 1024, device="cuda"), dim=-1)
+        db8, sa = quantize_int8(db)
+        q = db[:17]
+        s, i = knn_search_int8(db8, sa, q, 10)
+        qi, sq = quantize_int8(q)
+        ref = (qi.float() * sq[:, None]) @ (db8.float() * sa[:, None]).T
+        rs, ri = torch.topk(ref, 10, dim=-1)
+        assert torch.allclose(s, rs, atol=1e-4), (s - rs).abs().max()
+        assert (i[:, 0] == ri[:, 0]).all()
+
+    def test_recall_vs_fp32_exact(self):
+        """int8 per-row-scale recall@10 >= 0.95 even on worst-case
+        gaussian corpora — the quantized-mode contract."""
+        from nornicdb_amd.ops.knn import (knn_search_exact, knn_search_int8,
+                                          quantize_int8)
+        torch.manual_seed(5)
+        db = torch.nn.functional.normalize(
+            torch.randn(96 * 500, 1024, device="cuda"), dim=-1)
+        q = torch.nn.functional.normalize(
+            torch.randn(64, 1024, device="cuda"), dim=-1)
+        es, ei = knn_search_exact(db, q, 10)
+        db8, sa = quantize_int8(db)
+        s, i = knn_search_int8(db8, sa, q, 10)
+        hit = 0
+        for r in range(64):
+            hit += len(set(i[r].tolist()) & set(ei[r].tolist()))
+        recall = hit / (64 * 10)
+        assert recall >= 0.95, recall
+        s2, i2 = knn_search_int8(db8, sa, db[:32], 10)
+        assert (i2[:, 0] == torch.arange(32, device="cuda")).all()
+
+    def test_embedding_index_int8_gpu(self):
+        import numpy as np
+
+        from nornicdb_amd.search.embedding_index import EmbeddingIndex
+        idx = EmbeddingIndex(128, device="cuda", quant="int8")
+        rng = np.random.default_rng(6)
+        vecs = rng.normal(size=(500, 128)).astype("float32")
+        vecs /= np.linalg.norm(vecs, axis=1, keepdims=True)
+        idx.add_batch([f"n{i}" for i in range(500)], vecs)
+        hits = idx.search(vecs[42], k=5)
+        assert hits[0][0] == "n42" and hits[0][1] > 0.99
