@@ -69,6 +69,56 @@ def build_corpus(n_rows, dim, device, row_base, seed=0x6E6F726E):
     return shard
 
 
+def exact_fp32_topk_regen(n_rows, dim, device, q, k, row_base,
+                          chunk=1 << 20):
+    """fp32 exact top-k against the TRUE corpus values, regenerated
+    deterministically per chunk (used by --quant modes, where the stored
+    corpus is quantized and the bf16 original is not kept)."""
+    from nornicdb_amd.ops import fill_random_unit_
+    best_s = best_i = None
+    qf = q.float()
+    tmp = torch.empty(chunk, dim, device=device, dtype=torch.bfloat16)
+    for s in range(0, n_rows, chunk):
+        e = min(s + chunk, n_rows)
+        t = tmp[:e - s]
+        fill_random_unit_(t, row_base=row_base + s)
+        sc = qf @ t.float().T
+        kk = min(k, e - s)
+        bs, bi = torch.topk(sc, kk, dim=-1)
+        bi = bi + (row_base + s)
+        if best_s is None:
+            best_s, best_i = bs, bi
+        else:
+            cs = torch.cat([best_s, bs], -1)
+            ci = torch.cat([best_i, bi], -1)
+            best_s, sel = torch.topk(cs, min(k, cs.shape[-1]), -1)
+            best_i = torch.gather(ci, -1, sel)
+    return best_s, best_i
+
+
+def build_corpus_int8(n_rows, dim, device, row_base):
+    """Build the shard directly as symmetric int8 + per-row scales —
+    slice-wise so the transient bf16 stays bounded (~32 GB) and the
+    1 B/element corpus fits alongside it."""
+    from nornicdb_amd.ops import fill_random_unit_
+    from nornicdb_amd.ops.knn import quantize_int8
+    shard = torch.empty(n_rows, dim, device=device, dtype=torch.int8)
+    scales = torch.empty(n_rows, device=device, dtype=torch.float32)
+    step = 16 << 20
+    tmp = torch.empty(min(step, n_rows), dim, device=device,
+                      dtype=torch.bfloat16)
+    for s in range(0, n_rows, step):
+        e = min(s + step, n_rows)
+        t = tmp[:e - s]
+        fill_random_unit_(t, row_base=row_base + s)
+        qi, sc = quantize_int8(t)
+        shard[s:e] = qi
+        scales[s:e] = sc
+    del tmp
+    torch.cuda.empty_cache()
+    return shard, scales
+
+
 def exact_fp32_topk(shard, q, k, row_base, chunk=1 << 20):
     """fp32 exact top-k of q against the local bf16 shard (upcast per chunk)."""
     best_s = best_i = None
@@ -103,11 +153,17 @@ def merge_global_topk(local_s, local_i, k, world):
     return s, torch.gather(ci, -1, sel)
 
 
-def search_step(shard, queries, k, row_base, world):
-    """Score queries (already gathered, [Qg, D]) vs local shard, merge top-k."""
+def search_step(shard, queries, k, row_base, world, scales=None):
+    """Score queries (already gathered, [Qg, D]) vs local shard, merge top-k.
+    scales!=None -> shard is a symmetric int8 corpus (--quant int8)."""
     from nornicdb_amd.ops import knn_search
 
-    s, i = knn_search(shard, queries, k, row_base=row_base)
+    if scales is not None:
+        from nornicdb_amd.ops.knn import knn_search_int8
+        s, i = knn_search_int8(shard, scales, queries.float(), k,
+                               row_base=row_base)
+    else:
+        s, i = knn_search(shard, queries, k, row_base=row_base)
     return merge_global_topk(s, i, k, world)
 
 
@@ -200,6 +256,10 @@ def main():
     p.add_argument("--k", type=int, default=10)
     p.add_argument("--layers", type=int, default=24)
     p.add_argument("--skip-recall", action="store_true")
+    p.add_argument("--quant", choices=["none", "int8"], default="none",
+                   help="int8: 1 B/elem corpus scored by the i8 MFMA "
+                        "kernel (2x capacity; NOT the headline config — "
+                        "dtype is reported as int8)")
     p.add_argument("--timing", action="store_true", help="print phase breakdown")
     p.add_argument("--no-pipeline", action="store_true",
                    help="disable embed/search stream overlap (sequential steps)")
@@ -249,7 +309,12 @@ def main():
               f"({shard_rows * args.dim * 2 / 1e9:.1f} GB/GPU) ...")
     t0 = time.time()
     if device.type == "cuda":
-        shard = build_corpus(shard_rows, args.dim, device, row_base)
+        if args.quant == "int8":
+            shard, shard_scales = build_corpus_int8(
+                shard_rows, args.dim, device, row_base)
+        else:
+            shard = build_corpus(shard_rows, args.dim, device, row_base)
+            shard_scales = None
         torch.cuda.synchronize()
     else:
         shard = torch.empty(shard_rows, args.dim, dtype=torch.float32)
@@ -257,6 +322,10 @@ def main():
         bf = torch.empty(shard_rows, args.dim, dtype=torch.bfloat16)
         fill_random_unit_(bf, row_base=row_base)
         shard = bf.float() if device.type == "cpu" else bf
+        shard_scales = None
+        if args.quant == "int8":
+            from nornicdb_amd.ops.knn import quantize_int8
+            shard, shard_scales = quantize_int8(shard)
     log(rank, f"[bench] corpus ready in {time.time()-t0:.1f}s")
 
     vocab = cfg.vocab_size
@@ -270,7 +339,8 @@ def main():
         tokens = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
         with torch.no_grad():
             q = model(tokens)  # [B, H] fp32 normalized
-        return q.to(shard.dtype)
+        return q.to(torch.bfloat16 if shard_scales is not None
+                    else shard.dtype)
 
     def gather_queries(q):
         if world > 1:
@@ -281,7 +351,8 @@ def main():
 
     def one_step():
         qg = gather_queries(embed_batch())
-        s, i = search_step(shard, qg, args.k, row_base, world)
+        s, i = search_step(shard, qg, args.k, row_base, world,
+                           scales=shard_scales)
         return qg, s, i
 
     # serving pipeline: embed batch t on one HIP stream while batch t-1 is
@@ -304,7 +375,8 @@ def main():
                 q_new = embed_batch()
             results = [None]
             with torch.cuda.stream(s_search):
-                results[0] = search_step(shard, prev_q, args.k, row_base, world)
+                results[0] = search_step(shard, prev_q, args.k, row_base,
+                                         world, scales=shard_scales)
             torch.cuda.synchronize()
             return gather_queries(q_new), results[0]
 
@@ -332,7 +404,8 @@ def main():
             qf = model(tok).to(shard.dtype)
 
             ms_embed = _t(lambda: model(tok))
-            ms_search = _t(lambda: search_step(shard, qf, args.k, row_base, world))
+            ms_search = _t(lambda: search_step(shard, qf, args.k, row_base,
+                                               world, scales=shard_scales))
         log(rank, f"[timing] embed {ms_embed:.1f} ms/batch ({args.batch/ms_embed*1000:.0f} docs/s)  "
                   f"search {ms_search:.1f} ms/batch ({args.batch/ms_search*1000:.0f} qps; "
                   f"{shard_rows*args.dim*2/ms_search/1e9:.2f} TB/s shard read)")
@@ -348,8 +421,13 @@ def main():
             qr = model(tokens).to(shard.dtype)
         if world > 1:
             dist.broadcast(qr, src=0)
-        ps, pi = search_step(shard, qr, args.k, row_base, world)
-        es, ei = exact_fp32_topk(shard, qr, args.k, row_base)
+        ps, pi = search_step(shard, qr, args.k, row_base, world,
+                             scales=shard_scales)
+        if shard_scales is not None:
+            es, ei = exact_fp32_topk_regen(shard.shape[0], args.dim, device,
+                                           qr, args.k, row_base)
+        else:
+            es, ei = exact_fp32_topk(shard, qr, args.k, row_base)
         es, ei = merge_global_topk(es, ei, args.k, world)
         hit = 0
         for r in range(nq):
@@ -398,7 +476,9 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(qps / baseline_qps, 3) if device.type == "cuda" else None,
-            "dtype": "bf16" if device.type == "cuda" else "fp32 (cpu dev mode)",
+            "dtype": ("int8 (opt-in quantized corpus; headline config is bf16)"
+                      if args.quant == "int8" else
+                      "bf16" if device.type == "cuda" else "fp32 (cpu dev mode)"),
             "data": "synthetic (deterministic unit-norm corpus; random token ids; random-init bge-m3 weights)",
             "config": {
                 "model": "bge-m3 (XLM-R-large shape: 24L/1024h/16heads, random init)",
