@@ -104,7 +104,8 @@ def build_corpus_int8(n_rows, dim, device, row_base):
     from nornicdb_amd.ops.knn import quantize_int8
     shard = torch.empty(n_rows, dim, device=device, dtype=torch.int8)
     scales = torch.empty(n_rows, device=device, dtype=torch.float32)
-    step = 16 << 20
+    # small slices: quantize_int8 makes fp32 transients (~3x slice bytes)
+    step = 2 << 20
     tmp = torch.empty(min(step, n_rows), dim, device=device,
                       dtype=torch.bfloat16)
     for s in range(0, n_rows, step):
@@ -323,9 +324,6 @@ def main():
         fill_random_unit_(bf, row_base=row_base)
         shard = bf.float() if device.type == "cpu" else bf
         shard_scales = None
-        if args.quant == "int8":
-            from nornicdb_amd.ops.knn import quantize_int8
-            shard, shard_scales = quantize_int8(shard)
     log(rank, f"[bench] corpus ready in {time.time()-t0:.1f}s")
 
     vocab = cfg.vocab_size
@@ -334,6 +332,9 @@ def main():
     if device.type == "cpu":
         args.dim = h
         shard = shard[:, :h].contiguous() if shard.shape[1] >= h else torch.randn(shard_rows, h)
+    if args.quant == "int8" and shard_scales is None:
+        from nornicdb_amd.ops.knn import quantize_int8
+        shard, shard_scales = quantize_int8(shard)
 
     def embed_batch():
         tokens = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
@@ -401,7 +402,8 @@ def main():
 
         tok = torch.randint(0, vocab, (args.batch, args.seq_len), device=device)
         with torch.no_grad():
-            qf = model(tok).to(shard.dtype)
+            qf = model(tok).to(torch.bfloat16 if shard_scales is not None
+                               else shard.dtype)
 
             ms_embed = _t(lambda: model(tok))
             ms_search = _t(lambda: search_step(shard, qf, args.k, row_base,
@@ -418,14 +420,22 @@ def main():
         tokens = torch.randint(0, vocab, (nq, args.seq_len), device=device,
                                generator=torch.Generator(device=device).manual_seed(99))
         with torch.no_grad():
-            qr = model(tokens).to(shard.dtype)
+            qr = model(tokens).to(torch.bfloat16 if shard_scales is not None
+                                  else shard.dtype)
         if world > 1:
             dist.broadcast(qr, src=0)
         ps, pi = search_step(shard, qr, args.k, row_base, world,
                              scales=shard_scales)
-        if shard_scales is not None:
+        if shard_scales is not None and device.type == "cuda":
+            # true pre-quantization values, regenerated deterministically
+            # (the GPU fill kernel is row-indexed, so regen == builder)
             es, ei = exact_fp32_topk_regen(shard.shape[0], args.dim, device,
                                            qr, args.k, row_base)
+        elif shard_scales is not None:
+            # CPU dev mode: the fill fallback is not slice-invariant, so
+            # score against the dequantized stored corpus instead
+            deq = shard.float() * shard_scales[:, None]
+            es, ei = exact_fp32_topk(deq, qr, args.k, row_base)
         else:
             es, ei = exact_fp32_topk(shard, qr, args.k, row_base)
         es, ei = merge_global_topk(es, ei, args.k, world)
