@@ -24,6 +24,10 @@ def _open(args, cfg):
     from .db import open_db
     from .embed import create_embedder
 
+    if getattr(cfg, "search_quant", ""):
+        # SearchService reads the env at index construction
+        os.environ.setdefault("NORNICDB_SEARCH_QUANT", cfg.search_quant)
+
     provider = "mock" if cfg.embedder == "mock" else cfg.embedder
     try:
         emb = create_embedder(provider, dims=cfg.embedding_dims)

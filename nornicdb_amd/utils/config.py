@@ -44,6 +44,9 @@ class Config:
     hnsw_ef_search: int = 100
     brute_force_max: int = 5000
     kmeans_min: int = 100000
+    # "" | "int8" | "fp8": 1 B/element GPU corpus (2x capacity; int8
+    # keeps recall@10 >= 0.95 worst-case). Also NORNICDB_SEARCH_QUANT.
+    search_quant: str = ""
     # observability
     log_queries: bool = False   # reference --log-queries / bolt LogQueries
     # auth
