@@ -107,3 +107,63 @@ def test_store_embed_recall_on_gpu(tmp_path):
             proc.wait(timeout=20)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_store_embed_recall_quantized(tmp_path):
+    """Same end-to-end path with the int8 quantized corpus mode
+    (NORNICDB_SEARCH_QUANT=int8): store -> GPU embed -> search must
+    still retrieve the stored node top-1."""
+    bolt_port, http_port = _free_ports(2)
+    env = dict(os.environ, PYTHONPATH=REPO, NORNICDB_EMBEDDER="bge-m3",
+               NORNICDB_EMBEDDING_DIMS="1024",
+               NORNICDB_SEARCH_QUANT="int8")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "nornicdb_amd", "serve",
+         "--data-dir", str(tmp_path / "data"),
+         "--bolt-port", str(bolt_port), "--http-port", str(http_port)],
+        cwd=REPO, env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    base = f"http://127.0.0.1:{http_port}"
+    try:
+        deadline = time.time() + 180
+        up = False
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(base + "/health", timeout=2) as r:
+                    if r.status == 200:
+                        up = True
+                        break
+            except Exception:
+                time.sleep(0.5)
+        assert up, proc.stdout.read(4000)
+
+        def post(path, payload):
+            req = urllib.request.Request(
+                base + path, data=json.dumps(payload).encode(),
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=60) as r:
+                return json.loads(r.read())
+
+        ids = {}
+        for key, text in (("a", "alpha particle physics"),
+                          ("b", "beta distribution statistics")):
+            ids[key] = post("/nornicdb/store", {"content": text})["id"]
+        post("/nornicdb/embed/trigger", {})
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            with urllib.request.urlopen(base + "/nornicdb/embed/stats",
+                                        timeout=10) as r:
+                if json.loads(r.read()).get("pending", 1) == 0:
+                    break
+            time.sleep(0.5)
+        hits = post("/nornicdb/search",
+                    {"query": "beta distribution statistics", "limit": 2})
+        got = hits["results"]
+        assert got and got[0]["id"] == ids["b"], (got, ids)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
