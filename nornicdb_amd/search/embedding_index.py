@@ -210,4 +210,9 @@ class EmbeddingIndex:
 
     def matrix(self) -> torch.Tensor:
         with self._lock:
-            return self._buf[:self._n]
+            m = self._buf[:self._n]
+            if self._scales is not None:
+                # int8 storage: return dequantized values — callers
+                # (recluster/k-means) expect real vector magnitudes
+                return m.float() * self._scales[:self._n, None]
+            return m
