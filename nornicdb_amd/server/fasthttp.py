@@ -8,7 +8,7 @@ bolt/server.py for Bolt) and drives the SAME ASGI app, so every route —
 REST, GraphQL, SSE streams, the console SPA — is served unchanged.
 
 Replaces the reference's Go net/http serving layer (cmd/nornicdb serve,
-pkg/api/server.go) with an asyncio-native one. Supports: keep-alive,
+pkg/server/server_router.go) with an asyncio-native one. Supports: keep-alive,
 Content-Length bodies, chunked request decoding, streamed (chunked or
 close-delimited) responses for SSE, TLS via loop.start_server(ssl=...).
 uvicorn remains available behind NORNICDB_HTTP_SERVER=uvicorn.
