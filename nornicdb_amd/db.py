@@ -139,6 +139,14 @@ class NornicDB:
         self.dims = dims or self.embedder.dims
         self.search = SearchService(engine, dims=self.dims, device=device,
                                     embedder=self.embedder)
+        # reopening a persisted store: rebuild the GPU/embedding indexes
+        # from stored embeddings (reference pkg/gpu auto-sync on boot /
+        # search_services BuildIndexes) — without this, vector search
+        # starts empty after a restart
+        try:
+            self.search.build_indexes()   # no-op scan on a fresh store
+        except Exception:
+            pass
         from .apoc import build_apoc_procedures
         from .storage import SchemaManager
         procs = build_procedures(self)

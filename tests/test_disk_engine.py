@@ -352,3 +352,30 @@ class TestDiskEngine:
         mgr2.close()
         with pytest.raises(Exception):
             open_db(d, dims=8, encryption_passphrase="wrong")
+
+
+def test_vector_search_survives_reopen(tmp_path):
+    """Reopening a persisted store rebuilds the embedding/fulltext
+    indexes from stored embeddings — recall works immediately after a
+    restart (regression: the index used to start empty)."""
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    d = str(tmp_path / "store")
+    mgr = open_db(d, embedder=MockEmbedder(16), dims=16)
+    db = mgr.get()
+    m = db.store("persistent memory about graph databases")
+    db.store("unrelated note about cooking")
+    db.embed_queue.drain()
+    assert db.recall("graph databases", limit=1)[0].id == m.id
+    mgr.close()
+
+    mgr2 = open_db(d, embedder=MockEmbedder(16), dims=16)
+    after = mgr2.get().recall("graph databases", limit=1)
+    assert after and after[0].id == m.id
+    # fulltext survives too
+    ft = mgr2.get().cypher(
+        "CALL db.index.fulltext.queryNodes('x', 'cooking') "
+        "YIELD node RETURN node.content")
+    assert any("cooking" in r[0] for r in ft.rows)
+    mgr2.close()
