@@ -50,6 +50,20 @@ class NamespacedEngine(Engine):
         e.end_node = self._unwrap(e.end_node)
         return e
 
+    # --- validators (constraint enforcement) ---
+    def add_validator(self, fn):
+        """Register on the PHYSICAL engine (the wrapper is never consulted
+        by inner write paths), scoped to this namespace and with the node
+        translated back to namespace-local ids so the validator's
+        duplicate lookups compare like with like."""
+        def wrapped(node: Node, is_update: bool):
+            if not self._mine(node.id):
+                return
+            c = node.copy()
+            c.id = self._unwrap(c.id)
+            fn(c, is_update)
+        self.inner.add_validator(wrapped)
+
     # --- nodes ---
     def create_node(self, node: Node) -> Node:
         return self._unwrap_node(self.inner.create_node(self._wrap_node(node)))

@@ -8,6 +8,8 @@ node create/update).
 
 from __future__ import annotations
 
+import json
+import os
 import threading
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional
@@ -40,8 +42,77 @@ class SchemaManager:
         self.vector_indexes: Dict[str, VectorIndexMeta] = {}
         self.property_indexes: List[tuple] = []
         self.named_indexes: Dict[str, tuple] = {}  # name -> (kind, label, props)
+        # schema survives restarts: definitions live in a sidecar
+        # schema.json next to the engine's store (reference persists
+        # schema in badger, pkg/storage/schema.go); keyed by namespace
+        # so every logical database keeps its own schema. Memory engines
+        # have no data dir -> memory-only, matching their semantics.
+        self._load()
         if hasattr(engine, "add_validator"):
             engine.add_validator(self._validate)
+
+    # ---- persistence ----
+    def _schema_file(self):
+        eng = self.engine
+        ns = getattr(eng, "ns", "") or "default"
+        inner = eng
+        while hasattr(inner, "inner"):
+            inner = inner.inner
+        d = getattr(inner, "data_dir", None)
+        return (os.path.join(d, "schema.json") if d else None), ns
+
+    def _persist(self):
+        path, ns = self._schema_file()
+        if path is None:
+            return
+        try:
+            all_ = {}
+            if os.path.exists(path):
+                with open(path) as f:
+                    all_ = json.load(f)
+            all_[ns] = {
+                "constraints": [[c.name, c.kind, c.label, c.prop]
+                                for c in self.constraints.values()],
+                "vector_indexes": [[v.name, v.label, v.prop, v.dims,
+                                    v.similarity]
+                                   for v in self.vector_indexes.values()],
+                "named_indexes": [[n, k, lb, list(ps)] for n, (k, lb, ps)
+                                  in self.named_indexes.items()],
+            }
+            tmp = path + ".tmp"
+            with open(tmp, "w") as f:
+                json.dump(all_, f)
+                f.flush()
+                os.fsync(f.fileno())
+            os.replace(tmp, path)
+        except (OSError, ValueError):
+            pass
+
+    def _load(self):
+        path, ns = self._schema_file()
+        if path is None or not os.path.exists(path):
+            return
+        try:
+            with open(path) as f:
+                payload = json.load(f).get(ns)
+            if not payload:
+                return
+            for nm, kind, lb, pr in payload.get("constraints", []):
+                self.constraints[nm] = Constraint(nm, kind, lb, pr)
+                if kind == "unique" and hasattr(self.engine,
+                                                "create_property_index"):
+                    self.engine.create_property_index(lb, pr)
+            for nm, lb, pr, dims, sim in payload.get("vector_indexes", []):
+                self.vector_indexes[nm] = VectorIndexMeta(nm, lb, pr,
+                                                          dims, sim)
+            for nm, k, lb, ps in payload.get("named_indexes", []):
+                self.named_indexes[nm] = (k, lb, list(ps))
+                for pp in ps:
+                    self.property_indexes.append((lb, pp))
+                    if hasattr(self.engine, "create_property_index"):
+                        self.engine.create_property_index(lb, pp)
+        except (OSError, ValueError, KeyError):
+            pass
 
     # ---- constraints ----
     def create_unique_constraint(self, name: str, label: str, prop: str):
@@ -62,6 +133,7 @@ class SchemaManager:
         # back the constraint with an exact index
         if hasattr(self.engine, "create_property_index"):
             self.engine.create_property_index(label, prop)
+        self._persist()
 
     def create_exists_constraint(self, name: str, label: str, prop: str):
         with self._lock:
@@ -71,10 +143,12 @@ class SchemaManager:
                         f"cannot create constraint {name}: node {n.id} "
                         f"missing {label}.{prop}")
             self.constraints[name] = Constraint(name, "exists", label, prop)
+        self._persist()
 
     def drop_constraint(self, name: str):
         with self._lock:
             self.constraints.pop(name, None)
+        self._persist()
 
     def list_constraints(self) -> List[Constraint]:
         with self._lock:
@@ -93,6 +167,7 @@ class SchemaManager:
             self.property_indexes.append((label, p))
             if hasattr(self.engine, "create_property_index"):
                 self.engine.create_property_index(label, p)
+        self._persist()
         return nm
 
     def drop_index(self, name: str) -> bool:
@@ -108,6 +183,7 @@ class SchemaManager:
                 pass
             if hasattr(self.engine, "drop_property_index"):
                 self.engine.drop_property_index(label, p)
+        self._persist()
         return True
 
     def list_indexes(self):
@@ -124,6 +200,7 @@ class SchemaManager:
         with self._lock:
             self.vector_indexes[name] = VectorIndexMeta(name, label, prop,
                                                         dims, similarity)
+        self._persist()
 
     # ---- enforcement hook ----
     def _validate(self, node: Node, is_update: bool):

@@ -379,3 +379,33 @@ def test_vector_search_survives_reopen(tmp_path):
         "YIELD node RETURN node.content")
     assert any("cooking" in r[0] for r in ft.rows)
     mgr2.close()
+
+
+def test_schema_survives_reopen(tmp_path):
+    """Constraints/indexes persist (sidecar schema.json) AND enforce
+    after a restart; enforcement also works through the DatabaseManager
+    facade (regression: the namespaced wrapper dropped validators)."""
+    import pytest as _pt
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+    from nornicdb_amd.storage.types import ConstraintViolation
+
+    d = str(tmp_path / "store")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8)
+    db = mgr.get()
+    db.cypher("CREATE CONSTRAINT uq FOR (n:P) REQUIRE n.x IS UNIQUE")
+    db.cypher("CREATE INDEX pidx FOR (n:P) ON (n.x)")
+    db.cypher("CREATE (:P {x: 1})")
+    with _pt.raises(Exception):
+        db.cypher("CREATE (:P {x: 1})")
+    mgr.close()
+
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    db2 = mgr2.get()
+    assert len(db2.cypher("SHOW CONSTRAINTS").rows) == 1
+    assert len(db2.cypher("SHOW INDEXES").rows) >= 1
+    with _pt.raises(Exception):
+        db2.cypher("CREATE (:P {x: 1})")
+    db2.cypher("CREATE (:P {x: 2})")   # non-duplicates still fine
+    mgr2.close()
