@@ -832,7 +832,13 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
 
     # ---- Qdrant-compatible REST (reference pkg/qdrantgrpc) ----
     from .qdrant import QdrantRegistry, qdrant_router
-    app.state.qdrant = QdrantRegistry()
+    # engine-backed: collections/points persist as _QdrantCollection /
+    # _QdrantPoint_* nodes and reload on boot (reference registry.go)
+    try:
+        _qeng = mgr.get().engine
+    except Exception:
+        _qeng = None
+    app.state.qdrant = QdrantRegistry(engine=_qeng)
     app.include_router(qdrant_router(app.state.qdrant))
 
     app._tx_commit_body = _tx_commit_body  # reused by the ASGI fast path

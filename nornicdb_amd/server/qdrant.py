@@ -25,7 +25,7 @@ from ..search.vectorspace import COSINE, GLOBAL, VectorSpace
 
 
 class _Collection:
-    def __init__(self, name: str, size: int, distance: str):
+    def __init__(self, name: str, size: int, distance: str, eng=None):
         self.name = name
         self.size = size
         self.distance = distance
@@ -33,21 +33,116 @@ class _Collection:
         self.payloads: Dict[str, dict] = {}
         self.vectors: Dict[str, list] = {}
         self.created = time.time()
+        self.eng = eng   # storage engine for durability (reference
+                         # registry.go _QdrantCollection/QdrantPoint nodes)
+
+    # ---- durability (write-through to the storage engine) ----
+    def _pt_id(self, pid: str) -> str:
+        return f"_qdr:{self.name}:{pid}"
+
+    def persist_point(self, pid: str):
+        if self.eng is None:
+            return
+        from ..storage.types import Node
+        n = Node(id=self._pt_id(pid),
+                 labels=[f"_QdrantPoint_{self.name}"],
+                 properties={"payload": self.payloads.get(pid, {}),
+                             "vector": self.vectors.get(pid)})
+        try:
+            self.eng.update_node(n)
+        except Exception:
+            try:
+                self.eng.create_node(n)
+            except Exception:
+                pass
+
+    def unpersist_point(self, pid: str):
+        if self.eng is None:
+            return
+        try:
+            self.eng.delete_node(self._pt_id(pid))
+        except Exception:
+            pass
 
 
 class QdrantRegistry:
-    def __init__(self, db_name: str = "neo4j"):
+    def __init__(self, db_name: str = "neo4j", engine=None):
         self._lock = threading.Lock()
         self.collections: Dict[str, _Collection] = {}
         self.db_name = db_name
+        self.engine = engine
+        if engine is not None:
+            self._load()
+
+    def _load(self):
+        """Rebuild collections + points from the storage engine
+        (reference pkg/qdrantgrpc/registry.go loads _QdrantCollection
+        metadata nodes + per-point nodes on boot)."""
+        try:
+            metas = self.engine.get_nodes_by_label("_QdrantCollection")
+        except Exception:
+            return
+        for m in metas:
+            props = m.properties
+            name = props.get("name")
+            if not name or name in self.collections:
+                continue
+            c = _Collection(name, int(props.get("size", 0)),
+                            props.get("distance", "Cosine"),
+                            eng=self.engine)
+            self.collections[name] = c
+            try:
+                GLOBAL.register(VectorSpace(self.db_name, "qdrant", name,
+                                            c.size, c.distance.lower()))
+            except Exception:
+                pass   # already registered by a previous open in-process
+            try:
+                pts = self.engine.get_nodes_by_label(
+                    f"_QdrantPoint_{name}")
+            except Exception:
+                pts = []
+            ids, vecs = [], []
+            for pn in pts:
+                pid = pn.id.split(":", 2)[2] if pn.id.count(":") >= 2 \
+                    else pn.id
+                vec = pn.properties.get("vector")
+                c.payloads[pid] = pn.properties.get("payload") or {}
+                if vec is not None:
+                    c.vectors[pid] = list(vec)
+                    ids.append(pid)
+                    vecs.append(vec)
+            if ids:
+                c.index.add_batch(ids, vecs)
 
     def create(self, name, size, distance):
         with self._lock:
             if name in self.collections:
                 raise KeyError(name)
-            self.collections[name] = _Collection(name, size, distance)
+            self.collections[name] = _Collection(name, size, distance,
+                                                 eng=self.engine)
             GLOBAL.register(VectorSpace(self.db_name, "qdrant", name, size,
                                         distance.lower()))
+            if self.engine is not None:
+                from ..storage.types import Node
+                try:
+                    self.engine.create_node(Node(
+                        id=f"_qdrcol:{name}", labels=["_QdrantCollection"],
+                        properties={"name": name, "size": int(size),
+                                    "distance": distance}))
+                except Exception:
+                    pass
+
+    def drop(self, name: str):
+        """Remove a collection and its persisted nodes."""
+        with self._lock:
+            c = self.collections.pop(name, None)
+        if c is not None and self.engine is not None:
+            try:
+                self.engine.delete_node(f"_qdrcol:{name}")
+            except Exception:
+                pass
+            for pid in list(c.payloads):
+                c.unpersist_point(pid)
 
     def get(self, name) -> _Collection:
         c = self.collections.get(name)
@@ -122,8 +217,7 @@ class QdrantRegistry:
         with open(path, "rb") as f:
             data = msgpack.unpackb(f.read(), raw=False)
         cfg = data["config"]
-        with self._lock:
-            self.collections.pop(name, None)
+        self.drop(name)
         self.create(name, int(cfg["size"]), cfg["distance"])
         c = self.get(name)
         for pid, rec in data["points"].items():
@@ -132,6 +226,7 @@ class QdrantRegistry:
                 c.index.add(pid, vec)
                 c.vectors[pid] = list(vec)
             c.payloads[pid] = rec.get("payload") or {}
+            c.persist_point(pid)
 
 
 def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
@@ -174,7 +269,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
 
     @r.delete("/collections/{name}")
     def delete_collection(name: str):
-        reg.collections.pop(name, None)
+        reg.drop(name)
         return ok(True)
 
     @r.put("/collections/{name}/points")
@@ -196,6 +291,8 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
             c.vectors[pid] = list(vec)
         if ids:
             c.index.add_batch(ids, vecs)
+        for pid in ids:
+            c.persist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     @r.post("/collections/{name}/points/search")
@@ -257,6 +354,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
             c.index.remove(pid)
             c.payloads.pop(pid, None)
             c.vectors.pop(pid, None)
+            c.unpersist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     @r.post("/collections/{name}/points")
@@ -297,6 +395,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
         for pid in _sel_ids(c, body):
             if pid in c.payloads:
                 c.payloads[pid].update(body.get("payload", {}))
+                c.persist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     @r.put("/collections/{name}/points/payload")
@@ -305,6 +404,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
         for pid in _sel_ids(c, body):
             if pid in c.payloads:
                 c.payloads[pid] = dict(body.get("payload", {}))
+                c.persist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     @r.post("/collections/{name}/points/payload/delete")
@@ -314,6 +414,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
         for pid in _sel_ids(c, body):
             for k in keys:
                 c.payloads.get(pid, {}).pop(k, None)
+            c.persist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     @r.post("/collections/{name}/points/payload/clear")
@@ -321,6 +422,7 @@ def qdrant_router(registry: QdrantRegistry = None) -> APIRouter:
         c = _coll(name)
         for pid in _sel_ids(c, body or {}):
             c.payloads[pid] = {}
+            c.persist_point(pid)
         return ok({"operation_id": 0, "status": "completed"})
 
     # ---- count / exists / query (qdrant >=1.10 universal query) ----

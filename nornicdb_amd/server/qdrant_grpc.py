@@ -459,7 +459,7 @@ class QdrantGrpc:
 
     def Delete(self, req, context):
         existed = req.collection_name in self.reg.collections
-        self.reg.collections.pop(req.collection_name, None)
+        self.reg.drop(req.collection_name)
         return M["CollectionOperationResponse"](result=existed)
 
     def CollectionExists(self, req, context):
@@ -483,6 +483,8 @@ class QdrantGrpc:
             c.vectors[pid] = vec
         if ids:
             c.index.add_batch(ids, vecs)
+        for pid in ids:
+            c.persist_point(pid)
         return self._op_ok()
 
     def _points_get(self, req, context):
@@ -506,6 +508,7 @@ class QdrantGrpc:
             c.payloads.pop(pid, None)
             c.vectors.pop(pid, None)
             c.index.remove(pid)
+            c.unpersist_point(pid)
         return self._op_ok()
 
     def Count(self, req, context):
@@ -589,6 +592,7 @@ class QdrantGrpc:
                     c.payloads[pid] = dict(patch)
                 else:
                     c.payloads[pid].update(patch)
+                c.persist_point(pid)
         return self._op_ok()
 
     def OverwritePayload(self, req, context):
@@ -601,6 +605,7 @@ class QdrantGrpc:
             if pl:
                 for k in req.keys:
                     pl.pop(k, None)
+                c.persist_point(pid)
         return self._op_ok()
 
     def ClearPayload(self, req, context):
@@ -608,6 +613,7 @@ class QdrantGrpc:
         for pid in self._selected(c, req.points):
             if pid in c.payloads:
                 c.payloads[pid] = {}
+                c.persist_point(pid)
         return self._op_ok()
 
     # ---- qdrant.Snapshots (reference snapshots_service.go) ----

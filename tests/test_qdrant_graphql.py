@@ -282,3 +282,60 @@ class TestGraphQLPaths:
         r = gq.execute('query($n: ID!) { neighborhood(nodeId: $n, depth: 1) '
                        '{ nodes { id } } }', {"n": ids["a"]})
         assert len(r["data"]["neighborhood"]["nodes"]) == 3
+
+
+def test_qdrant_collections_survive_restart(tmp_path):
+    """Collections/points persist as _QdrantCollection/_QdrantPoint_*
+    nodes in the storage engine and reload on boot (reference
+    pkg/qdrantgrpc/registry.go) — create, upsert, restart, search,
+    delete, restart again."""
+    from fastapi.testclient import TestClient
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+    from nornicdb_amd.server import create_app
+
+    d = str(tmp_path / "store")
+
+    def client(mgr):
+        app = create_app(mgr, auth=None)
+        return TestClient(app.app if hasattr(app, "app") else app)
+
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8)
+    c = client(mgr)
+    assert c.put("/collections/docs",
+                 json={"vectors": {"size": 4, "distance": "Cosine"}}
+                 ).json()["status"] == "ok"
+    assert c.put("/collections/docs/points", json={"points": [
+        {"id": 1, "vector": [1, 0, 0, 0], "payload": {"t": "a"}},
+        {"id": 2, "vector": [0, 1, 0, 0], "payload": {"t": "b"}}]}
+        ).json()["status"] == "ok"
+    mgr.close()
+
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    c2 = client(mgr2)
+    names = [x["name"] for x in
+             c2.get("/collections").json()["result"]["collections"]]
+    assert "docs" in names
+    hits = c2.post("/collections/docs/points/search",
+                   json={"vector": [1, 0, 0, 0], "limit": 1}).json()["result"]
+    assert str(hits[0]["id"]) == "1"
+    pl = c2.post("/collections/docs/points",
+                 json={"ids": [2]}).json()["result"]
+    assert pl[0]["payload"] == {"t": "b"}
+    c2.post("/collections/docs/points/delete", json={"points": [1]})
+    mgr2.close()
+
+    mgr3 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    c3 = client(mgr3)
+    assert c3.post("/collections/docs/points/count",
+                   json={}).json()["result"]["count"] == 1
+    # collection drop removes persisted state too
+    c3.delete("/collections/docs")
+    mgr3.close()
+    mgr4 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    c4 = client(mgr4)
+    names = [x["name"] for x in
+             c4.get("/collections").json()["result"]["collections"]]
+    assert "docs" not in names
+    mgr4.close()
