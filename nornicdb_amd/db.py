@@ -355,6 +355,23 @@ class DatabaseManager:
         self._aliases: Dict[str, str] = {}
         for name in (self.DEFAULT, self.SYSTEM):
             self._open(name)
+        # database registry persists in the system namespace (_Database /
+        # _DbAlias nodes — Neo4j keeps db metadata in the system db;
+        # reference pkg/multidb/manager.go persists the registry): without
+        # this, user-created databases vanish from the manager on restart
+        # while their namespaced data stays orphaned on disk
+        try:
+            sys_eng = self._dbs[self.SYSTEM].engine
+            for n in sys_eng.get_nodes_by_label("_Database"):
+                nm = n.properties.get("name")
+                if nm and nm not in self._dbs:
+                    self._open(nm)
+            for n in sys_eng.get_nodes_by_label("_DbAlias"):
+                a, t = n.properties.get("alias"), n.properties.get("target")
+                if a and t:
+                    self._aliases[a] = t
+        except Exception:
+            pass
 
     def _open(self, name: str) -> NornicDB:
         eng = NamespacedEngine(self._base, name)
@@ -441,7 +458,15 @@ class DatabaseManager:
         with self._lock:
             if name in self._dbs:
                 raise ValueError(f"database {name} already exists")
-            return self._open(name)
+            db = self._open(name)
+        try:
+            from .storage.types import Node as _N
+            self._dbs[self.SYSTEM].engine.create_node(_N(
+                id=f"_db:{name}", labels=["_Database"],
+                properties={"name": name}))
+        except Exception:
+            pass
+        return db
 
     def drop(self, name: str):
         if name in (self.SYSTEM, self.DEFAULT):
@@ -456,10 +481,25 @@ class DatabaseManager:
                 except NotFoundError:
                     pass
             db.close()
+        try:
+            self._dbs[self.SYSTEM].engine.delete_node(f"_db:{name}")
+        except Exception:
+            pass
 
     def alias(self, alias: str, target: str):
         with self._lock:
             self._aliases[alias] = target
+        try:
+            from .storage.types import Node as _N
+            eng = self._dbs[self.SYSTEM].engine
+            n = _N(id=f"_dbalias:{alias}", labels=["_DbAlias"],
+                   properties={"alias": alias, "target": target})
+            try:
+                eng.update_node(n)
+            except Exception:
+                eng.create_node(n)
+        except Exception:
+            pass
 
     def list(self) -> List[str]:
         with self._lock:

@@ -409,3 +409,29 @@ def test_schema_survives_reopen(tmp_path):
         db2.cypher("CREATE (:P {x: 1})")
     db2.cypher("CREATE (:P {x: 2})")   # non-duplicates still fine
     mgr2.close()
+
+
+def test_database_registry_survives_reopen(tmp_path):
+    """User-created databases + aliases persist (system-namespace
+    _Database/_DbAlias nodes) — without this, created databases vanished
+    from the manager on restart while their data stayed orphaned."""
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    d = str(tmp_path / "store")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8)
+    mgr.create("analytics")
+    mgr.alias("stats", "analytics")
+    mgr.get("analytics").cypher("CREATE (:T {x: 1})")
+    mgr.close()
+
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    assert "analytics" in mgr2.list()
+    assert mgr2.get("stats").cypher(
+        "MATCH (n:T) RETURN count(n)").rows == [[1]]
+    mgr2.drop("analytics")
+    mgr2.close()
+
+    mgr3 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    assert "analytics" not in mgr3.list()
+    mgr3.close()
