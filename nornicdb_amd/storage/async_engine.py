@@ -181,6 +181,16 @@ class AsyncEngine(Engine):
     def register_callback(self, cb):
         self.inner.register_callback(cb)
 
+    def add_validator(self, fn):
+        # same forwarding rule as events: enforcement lives in the
+        # physical engine's write path
+        if hasattr(self.inner, "add_validator"):
+            self.inner.add_validator(fn)
+
+    def iter_nodes_by_label(self, label):
+        self.flush()
+        return self.inner.iter_nodes_by_label(label)
+
     def create_property_index(self, label, prop):
         self.flush()
         return self.inner.create_property_index(label, prop)
