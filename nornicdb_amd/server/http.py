@@ -589,11 +589,21 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         return {"databases": mgr.list()}
 
     @app.post("/admin/backup")
-    def backup(_user=Depends(check_auth)):
+    def backup(body: Dict[str, Any] = None, _user=Depends(check_auth)):
+        """Online backup. {"path": "/backups/x"} streams a consistent
+        copy there (disk/LSM engine); without a path, snapshot-capable
+        engines write their snapshot in place."""
         base = mgr._base
+        path = (body or {}).get("path")
+        if path and hasattr(base, "backup"):
+            base.backup(path)
+            return {"status": "backup written", "path": path}
         if hasattr(base, "snapshot"):
             base.snapshot()
             return {"status": "snapshot written"}
+        if hasattr(base, "backup") and path is None:
+            return {"status": "disk engine: POST {\"path\": ...} for an "
+                              "online backup"}
         return {"status": "in-memory engine, nothing to snapshot"}
 
     # ---- NornicDB memory/search API ----

@@ -30,6 +30,7 @@ durable and crash-consistent as the data they describe.
 
 from __future__ import annotations
 
+import os
 import struct
 import threading
 import time
@@ -684,10 +685,19 @@ class DiskEngine(Engine):
     # ------------------------------------------------------------------
     def backup(self, dest: str):
         self._kv.backup(dest)
+        # sidecar metadata (schema definitions) travels with the backup
+        import shutil
+        sp = os.path.join(self.data_dir, "schema.json")
+        if os.path.exists(sp):
+            shutil.copyfile(sp, dest + ".schema.json")
 
     @staticmethod
     def restore(backup_path: str, target_dir: str, encryption=None) -> "DiskEngine":
         LSMStore.restore(backup_path, target_dir, crypt=encryption).close()
+        import shutil
+        sj = backup_path + ".schema.json"
+        if os.path.exists(sj):
+            shutil.copyfile(sj, os.path.join(target_dir, "schema.json"))
         return DiskEngine(target_dir, encryption=encryption)
 
     def compact(self):

@@ -435,3 +435,32 @@ def test_database_registry_survives_reopen(tmp_path):
     mgr3 = open_db(d, embedder=MockEmbedder(8), dims=8)
     assert "analytics" not in mgr3.list()
     mgr3.close()
+
+
+def test_backup_restore_carries_schema(tmp_path):
+    """Online backup + restore preserves constraints (sidecar
+    schema.json travels with the backup file)."""
+    import pytest as _pt
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+    from nornicdb_amd.storage.disk import DiskEngine
+
+    d = str(tmp_path / "src")
+    b = str(tmp_path / "backup.snap")
+    r = str(tmp_path / "restored")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8)
+    db = mgr.get()
+    db.cypher("CREATE CONSTRAINT u FOR (n:P) REQUIRE n.x IS UNIQUE")
+    db.cypher("CREATE (:P {x: 1})")
+    mgr._base.backup(b)
+    mgr.close()
+
+    DiskEngine.restore(b, r).close()
+    mgr2 = open_db(r, embedder=MockEmbedder(8), dims=8)
+    db2 = mgr2.get()
+    assert db2.cypher("MATCH (n:P) RETURN count(n)").rows == [[1]]
+    assert len(db2.cypher("SHOW CONSTRAINTS").rows) == 1
+    with _pt.raises(Exception):
+        db2.cypher("CREATE (:P {x: 1})")
+    mgr2.close()
