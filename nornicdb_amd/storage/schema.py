@@ -61,6 +61,35 @@ class SchemaManager:
         d = getattr(inner, "data_dir", None)
         return (os.path.join(d, "schema.json") if d else None), ns
 
+    def _crypt(self):
+        """Engine's at-rest cipher, if any — the sidecar must not leak
+        plaintext next to an encrypted store."""
+        inner = self.engine
+        while hasattr(inner, "inner"):
+            inner = inner.inner
+        kv = getattr(inner, "_kv", None)
+        return getattr(kv, "crypt", None)
+
+    def _read_file(self, path):
+        with open(path, "rb") as f:
+            raw = f.read()
+        crypt = self._crypt()
+        if crypt is not None:
+            raw = crypt.decrypt(raw, aad=b"schema")
+        return json.loads(raw.decode())
+
+    def _write_file(self, path, obj):
+        raw = json.dumps(obj).encode()
+        crypt = self._crypt()
+        if crypt is not None:
+            raw = crypt.encrypt(raw, aad=b"schema")
+        tmp = path + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(raw)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, path)
+
     def _persist(self):
         path, ns = self._schema_file()
         if path is None:
@@ -68,8 +97,7 @@ class SchemaManager:
         try:
             all_ = {}
             if os.path.exists(path):
-                with open(path) as f:
-                    all_ = json.load(f)
+                all_ = self._read_file(path)
             all_[ns] = {
                 "constraints": [[c.name, c.kind, c.label, c.prop]
                                 for c in self.constraints.values()],
@@ -79,12 +107,7 @@ class SchemaManager:
                 "named_indexes": [[n, k, lb, list(ps)] for n, (k, lb, ps)
                                   in self.named_indexes.items()],
             }
-            tmp = path + ".tmp"
-            with open(tmp, "w") as f:
-                json.dump(all_, f)
-                f.flush()
-                os.fsync(f.fileno())
-            os.replace(tmp, path)
+            self._write_file(path, all_)
         except (OSError, ValueError):
             pass
 
@@ -93,8 +116,7 @@ class SchemaManager:
         if path is None or not os.path.exists(path):
             return
         try:
-            with open(path) as f:
-                payload = json.load(f).get(ns)
+            payload = self._read_file(path).get(ns)
             if not payload:
                 return
             for nm, kind, lb, pr in payload.get("constraints", []):

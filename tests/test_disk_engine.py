@@ -464,3 +464,25 @@ def test_backup_restore_carries_schema(tmp_path):
     with _pt.raises(Exception):
         db2.cypher("CREATE (:P {x: 1})")
     mgr2.close()
+
+
+def test_schema_sidecar_encrypted_at_rest(tmp_path):
+    """With encryption on, the schema sidecar is ciphertext (no label
+    names leak next to the encrypted store) and reloads correctly."""
+    import os
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    d = str(tmp_path / "store")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8,
+                  encryption_passphrase="pw")
+    mgr.get().cypher(
+        "CREATE CONSTRAINT u FOR (n:Secret) REQUIRE n.k IS UNIQUE")
+    mgr.close()
+    raw = open(os.path.join(d, "schema.json"), "rb").read()
+    assert b"Secret" not in raw
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8,
+                   encryption_passphrase="pw")
+    assert len(mgr2.get().cypher("SHOW CONSTRAINTS").rows) == 1
+    mgr2.close()
