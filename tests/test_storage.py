@@ -433,3 +433,31 @@ class TestTemporalPersistence:
         assert mgr2.get().cypher(
             "MATCH (e:E) RETURN toString(e.d)").rows == [["2025-12-31"]]
         mgr2.close()
+
+
+def test_wal_engine_durability_suite(tmp_path):
+    """The round-1 WAL engine gets the same restart guarantees as the
+    disk engine: schema + enforcement, vector recall, db registry."""
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    d = str(tmp_path / "wal")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8, engine="wal")
+    db = mgr.get()
+    db.cypher("CREATE CONSTRAINT u FOR (n:P) REQUIRE n.x IS UNIQUE")
+    m = db.store("wal engine memory")
+    db.embed_queue.drain()
+    mgr.create("side")
+    mgr.close()
+
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8, engine="wal")
+    db2 = mgr2.get()
+    assert len(db2.cypher("SHOW CONSTRAINTS").rows) == 1
+    db2.cypher("CREATE (:P {x: 1})")
+    import pytest as _pt
+    with _pt.raises(Exception):
+        db2.cypher("CREATE (:P {x: 1})")
+    hits = db2.recall("wal engine", limit=1)
+    assert hits and hits[0].id == m.id
+    assert "side" in mgr2.list()
+    mgr2.close()
