@@ -102,7 +102,8 @@ def cmd_serve(args):
                       host=cfg.bolt_host, port=args.bolt_port or cfg.bolt_port,
                       authenticator=auth, ssl_context=ssl_ctx,
                       log_queries=(getattr(args, "log_queries", False)
-                                   or cfg.log_queries))
+                                   or cfg.log_queries),
+                      tx_factory=lambda name: mgr.get(name).begin_tx())
     grpc_server = None
     if getattr(args, "grpc_port", None):
         from .server.nornic_grpc import serve as grpc_serve

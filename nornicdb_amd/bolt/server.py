@@ -57,6 +57,7 @@ class BoltSession:
         self.pending_pos = 0
         self.failed = False
         self.in_tx = False
+        self._txpair = None   # (executor, recorder) when tx_factory is set
         self.tx_queries: List = []
 
     @property
@@ -166,6 +167,9 @@ class BoltSession:
         if tag == M_RESET:
             self.failed = False
             self.pending = None
+            if self.in_tx and self._txpair is not None:
+                self._txpair[1].rollback()
+            self._txpair = None
             self.in_tx = False
             self.write_message(R_SUCCESS, {})
             return False
@@ -194,10 +198,14 @@ class BoltSession:
                 return False
             self.in_tx = True
             self._tx_wrote = False
+            self._txpair = self.server.make_tx(self.db)
             self.write_message(R_SUCCESS, {})
             return False
         if tag == M_COMMIT:
             self.in_tx = False
+            if self._txpair is not None:
+                self._txpair[1].commit()
+                self._txpair = None
             if self._tx_wrote:
                 self.server.bump_version(self.db)
                 self._tx_wrote = False
@@ -205,8 +213,12 @@ class BoltSession:
                                {"bookmark": self.server.bookmark(self.db)})
             return False
         if tag == M_ROLLBACK:
-            # note: implicit-apply semantics — statements already applied.
             self.in_tx = False
+            if self._txpair is not None:
+                # undo every applied statement (reference
+                # pkg/cypher/transaction.go handleRollback)
+                self._txpair[1].rollback()
+                self._txpair = None
             self.write_message(R_SUCCESS, {})
             return False
         if tag == M_ROUTE:
@@ -282,7 +294,10 @@ class BoltSession:
                 "message": "bookmark version not reached"})
             return False
         try:
-            result = self.server.execute(db, query, params or {})
+            if self.in_tx and self._txpair is not None:
+                result = self._txpair[0].execute(query, params or {})
+            else:
+                result = self.server.execute(db, query, params or {})
         except (CypherSyntaxError,) as e:
             self.failed = True
             self.write_message(R_FAILURE, {
@@ -395,7 +410,8 @@ class BoltServer:
     def __init__(self, executor_for_db: Callable[[str], Any],
                  host: str = "127.0.0.1", port: int = 7687,
                  authenticator=None, version_str: str = "0.1.0",
-                 ssl_context=None, log_queries: bool = False):
+                 ssl_context=None, log_queries: bool = False,
+                 tx_factory: Callable[[str], Any] = None):
         # causal bookmarks: per-db monotonically increasing commit
         # version; bookmark = "ndb:<db>:<version>" (reference
         # server.go:1617-1650 bookmark lifecycle)
@@ -408,6 +424,9 @@ class BoltServer:
         self.version_str = version_str
         # reference pkg/bolt/server.go:440 LogQueries — stdout query log
         self.log_queries = log_queries
+        # tx_factory(db) -> (executor, recorder): real BEGIN/ROLLBACK
+        # support (undo-recording); None keeps implicit-apply semantics
+        self.tx_factory = tx_factory
         self._server: Optional[asyncio.AbstractServer] = None
 
     def authenticate(self, auth: Dict[str, Any]):
@@ -423,6 +442,14 @@ class BoltServer:
             return True, None
         except Exception as e:
             return False, str(e)
+
+    def make_tx(self, db: str):
+        if self.tx_factory is None:
+            return None
+        try:
+            return self.tx_factory(db or "neo4j")
+        except Exception:
+            return None
 
     def execute(self, db: str, query: str, params: Dict[str, Any]):
         ex = self.executor_for_db(db or "neo4j")

@@ -187,6 +187,23 @@ class NornicDB:
         if _di:
             self.decay.start(float(_di))
 
+    def begin_tx(self):
+        """Explicit transaction: (executor, recorder). Statements run on
+        the returned executor apply immediately (read-your-writes) while
+        the recorder captures inverses; recorder.rollback() undoes them
+        (reference pkg/cypher/transaction.go semantics; isolation is
+        read-uncommitted — see storage/txrecorder.py)."""
+        from .cypher import Executor as _Ex
+        from .storage.txrecorder import TxRecorder
+        rec = TxRecorder(self.engine)
+        ex = _Ex(rec, procedures=self.executor.procedures,
+                 schema=self.schema)
+        ex.current_database = self.name
+        ex.database_lister = self.executor.database_lister
+        ex.database_router = self.executor.database_router
+        ex.database_admin = getattr(self.executor, "database_admin", None)
+        return ex, rec
+
     # ---- cypher ----
     # queries containing these are never served from the result cache;
     # "call" is included because procedures may mutate state the storage

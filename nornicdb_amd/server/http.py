@@ -507,19 +507,20 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     # bookmark, rollback acknowledges) ----
     import itertools as _it
     _tx_counter = _it.count(1)
-    _open_txs: Dict[str, str] = {}
+    _open_txs: Dict[str, tuple] = {}   # txid -> (db_name, executor, recorder)
 
-    async def _run_statements(db_name: str, request: Request):
+    async def _run_statements(db_name: str, request: Request, executor=None):
         body = await request.json() if (await request.body()) else {}
         try:
             db = mgr.get(db_name)
         except KeyError:
             raise HTTPException(404, f"database {db_name} not found")
         results, errors = [], []
+        runner = executor.execute if executor is not None else db.cypher
         for stmt in (body or {}).get("statements", []):
             try:
-                r = db.cypher(stmt.get("statement", ""),
-                              stmt.get("parameters") or {})
+                r = runner(stmt.get("statement", ""),
+                           stmt.get("parameters") or {})
                 results.append({
                     "columns": r.columns,
                     "data": [{"row": [_jsonable(v) for v in row], "meta": []}
@@ -537,8 +538,12 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
     async def tx_open(db_name: str, request: Request, response: Response,
                       _user=Depends(check_auth)):
         txid = str(next(_tx_counter))
-        _open_txs[txid] = db_name
-        results, errors = await _run_statements(db_name, request)
+        try:
+            ex, rec = mgr.get(db_name).begin_tx()
+        except KeyError:
+            raise HTTPException(404, f"database {db_name} not found")
+        _open_txs[txid] = (db_name, ex, rec)
+        results, errors = await _run_statements(db_name, request, executor=ex)
         response.headers["Location"] = f"/db/{db_name}/tx/{txid}"
         return {"results": results, "errors": errors,
                 "commit": f"/db/{db_name}/tx/{txid}/commit",
@@ -550,21 +555,31 @@ def create_app(mgr: DatabaseManager, auth=None, version: str = "0.1.0") -> FastA
         if txid not in _open_txs:
             raise HTTPException(
                 404, f"transaction {txid} not found or already closed")
-        results, errors = await _run_statements(db_name, request)
+        _, ex, _rec = _open_txs[txid]
+        results, errors = await _run_statements(db_name, request, executor=ex)
         return {"results": results, "errors": errors,
                 "commit": f"/db/{db_name}/tx/{txid}/commit"}
 
     @app.post("/db/{db_name}/tx/{txid}/commit")
     async def tx_commit_open(db_name: str, txid: str, request: Request,
                              _user=Depends(check_auth)):
-        _open_txs.pop(txid, None)
-        results, errors = await _run_statements(db_name, request)
+        ent = _open_txs.pop(txid, None)
+        if ent is not None:
+            _, ex, rec = ent
+            results, errors = await _run_statements(db_name, request,
+                                                    executor=ex)
+            rec.commit()
+        else:
+            results, errors = await _run_statements(db_name, request)
         return {"results": results, "errors": errors,
                 "lastBookmarks": [f"FB:bookmark-{txid}"]}
 
     @app.delete("/db/{db_name}/tx/{txid}")
     async def tx_rollback(db_name: str, txid: str, _user=Depends(check_auth)):
-        _open_txs.pop(txid, None)
+        ent = _open_txs.pop(txid, None)
+        if ent is not None:
+            ent[2].rollback()   # undo applied statements (reference
+                                # BadgerTransaction.Rollback semantics)
         return {"results": [], "errors": []}
 
     # ---- multi-database management ----

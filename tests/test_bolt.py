@@ -315,3 +315,83 @@ def test_log_queries_stdout(capsys):
     assert r.rows == [[1]]
     out = capsys.readouterr().out
     assert "[query]" in out and "RETURN 1 AS one" in out and "rows=1" in out
+
+
+def test_explicit_tx_rollback_undoes_writes():
+    """BEGIN .. RUN(CREATE) .. ROLLBACK leaves no state (reference
+    pkg/cypher/transaction.go handleRollback — real rollback, not
+    implicit-apply); COMMIT keeps it; RESET mid-tx also rolls back."""
+    import asyncio
+
+    from nornicdb_amd.bolt.server import BoltServer
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    mgr = open_db(embedder=MockEmbedder(8), dims=8)
+    srv = BoltServer(lambda db: mgr.get(db).executor,
+                     tx_factory=lambda name: mgr.get(name).begin_tx())
+
+    async def run():
+        import struct
+
+        from nornicdb_amd.bolt import packstream as ps
+        await srv.start()
+        r, w = await asyncio.open_connection("127.0.0.1", srv.port)
+        w.write(struct.pack(">I", 0x6060B017) + bytes([0, 0, 4, 4]) +
+                bytes(12))
+        await r.read(4)
+
+        async def send(tag, *fields):
+            data = ps.pack(ps.Structure(tag, list(fields)))
+            out = b""
+            for i in range(0, len(data), 65535):
+                chunk = data[i:i + 65535]
+                out += struct.pack(">H", len(chunk)) + chunk
+            w.write(out + b"\x00\x00")
+            await w.drain()
+
+        async def recv():
+            buf = b""
+            while True:
+                hdr = await r.readexactly(2)
+                size = struct.unpack(">H", hdr)[0]
+                if size == 0:
+                    break
+                buf += await r.readexactly(size)
+            return ps.unpack(buf)
+
+        await send(0x01, {"scheme": "none"})
+        await recv()
+        # tx 1: create then ROLLBACK
+        await send(0x11, {})
+        await recv()
+        await send(0x10, "CREATE (:TxB {x: 1})", {}, {})
+        await recv()
+        await send(0x3F, {"n": -1})
+        await recv()
+        await send(0x13)   # ROLLBACK
+        await recv()
+        # tx 2: create then COMMIT
+        await send(0x11, {})
+        await recv()
+        await send(0x10, "CREATE (:TxB {x: 2})", {}, {})
+        await recv()
+        await send(0x3F, {"n": -1})
+        await recv()
+        await send(0x12)   # COMMIT
+        await recv()
+        # tx 3: create then RESET (must roll back too)
+        await send(0x11, {})
+        await recv()
+        await send(0x10, "CREATE (:TxB {x: 3})", {}, {})
+        await recv()
+        await send(0x3F, {"n": -1})
+        await recv()
+        await send(0x0F)   # RESET
+        await recv()
+        w.close()
+        await srv._server.wait_closed() if False else None
+
+    asyncio.new_event_loop().run_until_complete(run())
+    rows = mgr.get().cypher("MATCH (n:TxB) RETURN n.x ORDER BY n.x").rows
+    assert rows == [[2]], rows

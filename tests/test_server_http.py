@@ -266,3 +266,33 @@ def test_admin_console_served_and_endpoints(client):
     for path in ("/admin/databases", "/admin/stats", "/nornicdb/embed/stats",
                  "/nornicdb/decay", "/status"):
         assert client.get(path).status_code == 200, path
+
+
+def test_http_tx_rollback_undoes_writes():
+    """POST /db/x/tx (open, run CREATE) then DELETE (rollback) leaves
+    no state; the commit path keeps it."""
+    from fastapi.testclient import TestClient
+
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+    from nornicdb_amd.server import create_app
+
+    mgr = open_db(embedder=MockEmbedder(8), dims=8)
+    app = create_app(mgr, auth=None)
+    c = TestClient(app.app if hasattr(app, "app") else app)
+
+    r = c.post("/db/neo4j/tx", json={"statements": [
+        {"statement": "CREATE (:HT {x: 1})"}]})
+    txid = r.headers["Location"].rsplit("/", 1)[1]
+    # visible inside the tx (read-your-writes), then rolled back
+    r2 = c.post(f"/db/neo4j/tx/{txid}", json={"statements": [
+        {"statement": "MATCH (n:HT) RETURN count(n)"}]})
+    assert r2.json()["results"][0]["data"][0]["row"] == [1]
+    c.delete(f"/db/neo4j/tx/{txid}")
+    assert mgr.get().cypher("MATCH (n:HT) RETURN count(n)").rows == [[0]]
+
+    r = c.post("/db/neo4j/tx", json={"statements": [
+        {"statement": "CREATE (:HT {x: 2})"}]})
+    txid = r.headers["Location"].rsplit("/", 1)[1]
+    c.post(f"/db/neo4j/tx/{txid}/commit", json={})
+    assert mgr.get().cypher("MATCH (n:HT) RETURN count(n)").rows == [[1]]
