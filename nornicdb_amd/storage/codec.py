@@ -17,6 +17,12 @@ def default(obj):
         return {"__t__": "time", "v": str(obj)}
     if isinstance(obj, tp.CypherDuration):
         return {"__t__": "duration", "v": str(obj)}
+    from ..cypher.functions import CypherPoint
+    if isinstance(obj, CypherPoint):
+        d = {"__t__": "point", "x": obj.x, "y": obj.y, "crs": obj.crs}
+        if getattr(obj, "z", None) is not None:
+            d["z"] = obj.z
+        return d
     raise TypeError(f"cannot serialize {type(obj).__name__}")
 
 
@@ -25,6 +31,15 @@ def object_hook(m):
     tag = m.get("__t__") if isinstance(m, dict) else None
     if tag is None:
         return m
+    if tag == "point":
+        from ..cypher.functions import CypherPoint
+        src = {"x": m["x"], "y": m["y"]}
+        if m.get("crs") == "wgs-84":
+            src = {"longitude": m["x"], "latitude": m["y"]}
+        if m.get("z") is not None:
+            src["z"] = m["z"]
+        src["crs"] = m.get("crs", "cartesian")
+        return CypherPoint(src)
     from ..cypher import temporal as tp
     v = m["v"]
     if tag == "date":

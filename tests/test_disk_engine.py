@@ -486,3 +486,29 @@ def test_schema_sidecar_encrypted_at_rest(tmp_path):
                    encryption_passphrase="pw")
     assert len(mgr2.get().cypher("SHOW CONSTRAINTS").rows) == 1
     mgr2.close()
+
+
+def test_typed_properties_roundtrip_reopen(tmp_path):
+    """Temporal, spatial (cartesian + wgs-84), nested and unicode
+    property values survive the disk codec and a reopen (regression:
+    point() properties could not be stored at all)."""
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    d = str(tmp_path / "store")
+    mgr = open_db(d, embedder=MockEmbedder(8), dims=8)
+    mgr.get().cypher(
+        "CREATE (:X {dt: datetime('2026-01-02T03:04:05Z'), "
+        "pt: point({x: 1.5, y: 2.5}), "
+        "gp: point({longitude: 9.9, latitude: 48.4}), "
+        "m: {a: [1, 2, {b: 'c'}]}, s: 'ünïcode ✓', f: 1.25})")
+    mgr.close()
+    mgr2 = open_db(d, embedder=MockEmbedder(8), dims=8)
+    r = mgr2.get().cypher(
+        "MATCH (n:X) RETURN n.dt.year, n.pt.x, n.gp.latitude, "
+        "n.m.a[2].b, n.s, n.f")
+    assert r.rows == [[2026, 1.5, 48.4, "c", "ünïcode ✓", 1.25]]
+    r2 = mgr2.get().cypher(
+        "MATCH (n:X) RETURN point.distance(n.pt, point({x: 1.5, y: 0.5}))")
+    assert abs(r2.rows[0][0] - 2.0) < 1e-9
+    mgr2.close()
