@@ -319,6 +319,40 @@ def temporal_from_struct(s: "Structure"):
     return None
 
 
+POINT2D_TAG = 0x58   # 'X'
+POINT3D_TAG = 0x59   # 'Y'
+
+
+def point_struct(v):
+    """CypherPoint -> Bolt Point2D/Point3D structure (srid, x, y[, z]).
+    SRIDs per Neo4j: 7203/9157 cartesian 2D/3D, 4326/4979 wgs-84."""
+    from ..cypher.functions import CypherPoint
+    if not isinstance(v, CypherPoint):
+        return None
+    wgs = v.crs == "wgs-84"
+    if v.z is None:
+        return Structure(POINT2D_TAG, [4326 if wgs else 7203,
+                                       float(v.x), float(v.y)])
+    return Structure(POINT3D_TAG, [4979 if wgs else 9157,
+                                   float(v.x), float(v.y), float(v.z)])
+
+
+def point_from_struct(st):
+    if not isinstance(st, Structure) or st.tag not in (POINT2D_TAG,
+                                                       POINT3D_TAG):
+        return None
+    from ..cypher.functions import CypherPoint
+    srid = st.fields[0]
+    wgs = srid in (4326, 4979)
+    if wgs:
+        src = {"longitude": st.fields[1], "latitude": st.fields[2]}
+    else:
+        src = {"x": st.fields[1], "y": st.fields[2]}
+    if st.tag == POINT3D_TAG:
+        src["z"] = st.fields[3]
+    return CypherPoint(src)
+
+
 def temporal_struct(v, bolt5: bool = False):
     """Convert a cypher temporal value to its Bolt structure (or None)."""
     import datetime as _dt

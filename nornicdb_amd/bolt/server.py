@@ -124,6 +124,9 @@ class BoltSession:
         ts = ps.temporal_struct(v, self.bolt5)
         if ts is not None:
             return ts
+        pt = ps.point_struct(v)
+        if pt is not None:
+            return pt
         return v
 
     # ---- message handlers ----
@@ -274,7 +277,10 @@ class BoltSession:
         cypher temporal values; recurse containers."""
         if isinstance(v, ps.Structure):
             t = ps.temporal_from_struct(v)
-            return t if t is not None else v
+            if t is not None:
+                return t
+            p = ps.point_from_struct(v)
+            return p if p is not None else v
         if isinstance(v, list):
             return [self.from_bolt(x) for x in v]
         if isinstance(v, dict):

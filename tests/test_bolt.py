@@ -395,3 +395,64 @@ def test_explicit_tx_rollback_undoes_writes():
     asyncio.new_event_loop().run_until_complete(run())
     rows = mgr.get().cypher("MATCH (n:TxB) RETURN n.x ORDER BY n.x").rows
     assert rows == [[2]], rows
+
+
+def test_point_values_over_bolt():
+    """RETURN point(...) encodes as Bolt Point2D/Point3D structures and
+    inbound point parameters decode back (regression: returning a point
+    crashed the session — native packer rejected CypherPoint)."""
+    import asyncio
+    import struct
+
+    from nornicdb_amd.bolt import packstream as ps
+    from nornicdb_amd.bolt.server import BoltServer
+    from nornicdb_amd.db import open_db
+    from nornicdb_amd.embed import MockEmbedder
+
+    mgr = open_db(embedder=MockEmbedder(8), dims=8)
+    srv = BoltServer(lambda db: mgr.get(db).executor)
+
+    async def run():
+        await srv.start()
+        r, w = await asyncio.open_connection("127.0.0.1", srv.port)
+        w.write(struct.pack(">I", 0x6060B017) + bytes([0, 0, 4, 4]) +
+                bytes(12))
+        await r.read(4)
+
+        async def send(tag, *fields):
+            data = ps.pack(ps.Structure(tag, list(fields)))
+            out = b""
+            for i in range(0, len(data), 65535):
+                ch = data[i:i + 65535]
+                out += struct.pack(">H", len(ch)) + ch
+            w.write(out + b"\x00\x00")
+            await w.drain()
+
+        async def recv():
+            buf = b""
+            while True:
+                size = struct.unpack(">H", await r.readexactly(2))[0]
+                if size == 0:
+                    break
+                buf += await r.readexactly(size)
+            return ps.unpack(buf)
+
+        await send(0x01, {"scheme": "none"})
+        await recv()
+        # round-trip: a Point2D parameter comes back out
+        await send(0x10, "RETURN $p AS p, point({longitude: 9.0, "
+                         "latitude: 48.0, height: 10.0}) AS g",
+                   {"p": ps.Structure(0x58, [7203, 1.5, 2.5])}, {})
+        await recv()
+        await send(0x3F, {"n": -1})
+        rec = await recv()
+        assert rec.tag == 0x71, rec
+        p, g = rec.fields[0]
+        assert isinstance(p, ps.Structure) and p.tag == 0x58
+        assert p.fields == [7203, 1.5, 2.5]
+        assert isinstance(g, ps.Structure) and g.tag == 0x59
+        assert g.fields[0] == 4979 and g.fields[1] == 9.0
+        await recv()   # summary
+        w.close()
+
+    asyncio.new_event_loop().run_until_complete(run())
